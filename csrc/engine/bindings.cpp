@@ -1,0 +1,362 @@
+// pybind11 bindings: windflow_amd._core
+//
+// Exposes the native engine (Engine/OpSpec/EdgeSpec), the native logic
+// catalog, and per-batch Python-callback logic.  Python callbacks receive
+// zero-copy numpy views of the batch columns (valid only during the call).
+#include <pybind11/functional.h>
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "engine.hpp"
+
+namespace py = pybind11;
+using namespace wfa;
+
+// ---- numpy view helpers ----
+static py::array col_view(DType dt, void* data, int64_t n) {
+    switch (dt) {
+        case DType::I64: return py::array_t<int64_t>({n}, {8}, (int64_t*)data, py::none());
+        case DType::F64: return py::array_t<double>({n}, {8}, (double*)data, py::none());
+        case DType::F32: return py::array_t<float>({n}, {4}, (float*)data, py::none());
+        case DType::U64: return py::array_t<uint64_t>({n}, {8}, (uint64_t*)data, py::none());
+        case DType::I32: return py::array_t<int32_t>({n}, {4}, (int32_t*)data, py::none());
+        case DType::U16: return py::array_t<uint16_t>({n}, {2}, (uint16_t*)data, py::none());
+        case DType::U8: return py::array_t<uint8_t>({n}, {1}, (uint8_t*)data, py::none());
+    }
+    throw std::runtime_error("bad dtype");
+}
+
+static py::dict batch_views(Batch* b) {
+    py::dict d;
+    d["ts"] = col_view(DType::I64, b->ts, b->count);
+    d["key"] = col_view(DType::U64, b->key, b->count);
+    for (size_t c = 0; c < b->schema.payload.size(); ++c)
+        d[py::str("c" + std::to_string(c))] = col_view(b->schema.payload[c], b->cols[c], b->count);
+    d["watermark"] = b->watermark;
+    return d;
+}
+
+// copy a python dict-of-arrays into fresh batches (splitting on capacity)
+static void emit_pydict(py::dict out_d, EmitCtx& out, int64_t wm) {
+    if (out_d.size() == 0) return;
+    py::array ts = out_d.contains("ts") ? out_d["ts"].cast<py::array>() : py::array();
+    py::array key = out_d.contains("key") ? out_d["key"].cast<py::array>() : py::array();
+    std::vector<py::array> cols;
+    for (size_t c = 0;; ++c) {
+        std::string nm = "c" + std::to_string(c);
+        if (!out_d.contains(nm.c_str())) break;
+        cols.push_back(out_d[nm.c_str()].cast<py::array>());
+    }
+    int64_t n = 0;
+    if (cols.size()) n = cols[0].shape(0);
+    else if (ts.ndim()) n = ts.shape(0);
+    int64_t done = 0;
+    while (done < n) {
+        Batch* o = out.new_batch();
+        int64_t take = std::min<int64_t>(o->capacity, n - done);
+        if (ts.ndim() == 1) {
+            auto a = ts.cast<py::array_t<int64_t>>();
+            memcpy(o->ts, a.data() + done, take * 8);
+        } else {
+            for (int64_t i = 0; i < take; ++i) o->ts[i] = done + i;
+        }
+        if (key.ndim() == 1) {
+            auto a = key.cast<py::array_t<uint64_t>>();
+            memcpy(o->key, a.data() + done, take * 8);
+        } else {
+            memset(o->key, 0, take * 8);
+        }
+        for (size_t c = 0; c < cols.size() && c < o->schema.payload.size(); ++c) {
+            size_t es = dsize(o->schema.payload[c]);
+            py::buffer_info bi = cols[c].request();
+            if ((size_t)bi.itemsize != es) throw std::runtime_error("dtype mismatch on c" + std::to_string(c));
+            memcpy(o->cols[c], (char*)bi.ptr + done * es, take * es);
+        }
+        o->count = take;
+        o->watermark = wm;
+        out.emit(o);
+        done += take;
+    }
+}
+
+// ---- Python-callback logic ----
+struct PyMapLogic : OpLogic {  // in-place mutate
+    py::function fn;
+    explicit PyMapLogic(py::function f) : fn(std::move(f)) {}
+    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+        if (b->refcnt.load(std::memory_order_acquire) > 1) {
+            Batch* c = clone(b, *b->pool);
+            release(b);
+            b = c;
+        }
+        {
+            py::gil_scoped_acquire gil;
+            fn(batch_views(b));
+        }
+        out.emit(b);
+    }
+};
+
+struct PyTransformLogic : OpLogic {  // returns new column dict (map/flatmap)
+    py::function fn;
+    explicit PyTransformLogic(py::function f) : fn(std::move(f)) {}
+    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+        int64_t wm = b->watermark;
+        py::gil_scoped_acquire gil;
+        py::object r = fn(batch_views(b));
+        release(b);
+        if (!r.is_none()) emit_pydict(r.cast<py::dict>(), out, wm);
+    }
+};
+
+struct PyFilterLogic : OpLogic {  // returns bool mask
+    py::function fn;
+    explicit PyFilterLogic(py::function f) : fn(std::move(f)) {}
+    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+        if (b->refcnt.load(std::memory_order_acquire) > 1) {
+            Batch* c = clone(b, *b->pool);
+            release(b);
+            b = c;
+        }
+        std::vector<char> keep;
+        {
+            py::gil_scoped_acquire gil;
+            py::object r = fn(batch_views(b));
+            auto mask = r.cast<py::array_t<bool>>();
+            keep.assign(mask.data(), mask.data() + mask.shape(0));
+        }
+        const size_t np_ = b->schema.payload.size();
+        int64_t w = 0;
+        for (int64_t i = 0; i < b->count; ++i) {
+            if (!keep[i]) continue;
+            if (w != i) {
+                b->ts[w] = b->ts[i];
+                b->key[w] = b->key[i];
+                for (size_t cc = 0; cc < np_; ++cc) {
+                    size_t es = dsize(b->schema.payload[cc]);
+                    memcpy((char*)b->cols[cc] + w * es, (char*)b->cols[cc] + i * es, es);
+                }
+            }
+            ++w;
+        }
+        b->count = w;
+        if (w)
+            out.emit(b);
+        else {
+            int64_t wm = b->watermark;
+            release(b);
+            for (auto* e : out.emitters) e->punct(wm);
+        }
+    }
+};
+
+struct PySinkLogic : OpLogic {
+    py::function fn;
+    explicit PySinkLogic(py::function f) : fn(std::move(f)) {}
+    void process(Batch* b, EmitCtx&, RuntimeCtx&) override {
+        {
+            py::gil_scoped_acquire gil;
+            fn(batch_views(b));
+        }
+        release(b);
+    }
+    void on_eos(EmitCtx&, RuntimeCtx&) override {
+        py::gil_scoped_acquire gil;
+        if (py::hasattr(fn, "on_eos")) fn.attr("on_eos")();
+    }
+};
+
+struct PySourceLogic : OpLogic {
+    py::function fn;  // fn(replica, parallelism) -> dict|None
+    explicit PySourceLogic(py::function f) : fn(std::move(f)) {}
+    bool is_source() const override { return true; }
+    bool source_step(EmitCtx& out, RuntimeCtx& ctx) override {
+        py::gil_scoped_acquire gil;
+        py::object r = fn(ctx.replica, ctx.parallelism);
+        if (r.is_none()) return false;
+        py::dict d = r.cast<py::dict>();
+        int64_t wm = d.contains("watermark") ? d["watermark"].cast<int64_t>() : 0;
+        emit_pydict(d, out, wm);
+        return true;
+    }
+};
+
+// key-extractor map: computes the key column from a payload column
+struct KeyByColLogic : OpLogic {
+    int col;
+    explicit KeyByColLogic(int c) : col(c) {}
+    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+        if (b->refcnt.load(std::memory_order_acquire) > 1) {
+            Batch* c2 = clone(b, *b->pool);
+            release(b);
+            b = c2;
+        }
+        switch (b->schema.payload[col]) {
+            case DType::I64:
+                for (int64_t i = 0; i < b->count; ++i) b->key[i] = (uint64_t)b->col<int64_t>(col)[i];
+                break;
+            case DType::U64:
+                for (int64_t i = 0; i < b->count; ++i) b->key[i] = b->col<uint64_t>(col)[i];
+                break;
+            case DType::I32:
+                for (int64_t i = 0; i < b->count; ++i) b->key[i] = (uint64_t)b->col<int32_t>(col)[i];
+                break;
+            default:
+                throw std::runtime_error("keyby col must be integer");
+        }
+        out.emit(b);
+    }
+};
+
+static StageSpec make_stage(Engine& e, int id, const std::string& kind, const std::string& spec,
+                            std::vector<double> fp, std::vector<int64_t> ip,
+                            std::vector<int> out_schema, int64_t out_batch, py::object pyfn) {
+    StageSpec st;
+    for (int d : out_schema) st.out_schema.payload.push_back((DType)d);
+    st.out_batch = out_batch;
+    if (kind == "sink") {
+        e.sink_acc_i64[id].store(0);
+        e.sink_tuples[id].store(0);
+    }
+    if (!pyfn.is_none()) {
+        auto fn = pyfn.cast<py::function>();
+        if (kind == "map")
+            st.factory = [fn] { return std::make_shared<PyMapLogic>(fn); };
+        else if (kind == "transform" || kind == "flatmap")
+            st.factory = [fn] { return std::make_shared<PyTransformLogic>(fn); };
+        else if (kind == "filter")
+            st.factory = [fn] { return std::make_shared<PyFilterLogic>(fn); };
+        else if (kind == "sink")
+            st.factory = [fn] { return std::make_shared<PySinkLogic>(fn); };
+        else if (kind == "source")
+            st.factory = [fn] { return std::make_shared<PySourceLogic>(fn); };
+        else
+            throw std::runtime_error("no python form for op kind " + kind);
+    } else if (kind == "keyby_col") {
+        int col = (int)ip.at(0);
+        st.factory = [col] { return std::make_shared<KeyByColLogic>(col); };
+    } else {
+        Engine* ep = &e;
+        st.factory = [kind, spec, fp, ip, ep, id] {
+            return make_native_logic(kind, spec, fp, ip, ep, id);
+        };
+    }
+    return st;
+}
+
+PYBIND11_MODULE(_core, m) {
+    m.doc() = "windflow_amd native engine (MI355X-native WindFlow re-design)";
+
+    py::enum_<ExecMode>(m, "ExecMode")
+        .value("DEFAULT", ExecMode::DEFAULT)
+        .value("DETERMINISTIC", ExecMode::DETERMINISTIC)
+        .value("PROBABILISTIC", ExecMode::PROBABILISTIC);
+    py::enum_<TimePolicy>(m, "TimePolicy")
+        .value("INGRESS_TIME", TimePolicy::INGRESS_TIME)
+        .value("EVENT_TIME", TimePolicy::EVENT_TIME);
+    py::enum_<Routing>(m, "Routing")
+        .value("FORWARD", Routing::FORWARD)
+        .value("KEYBY", Routing::KEYBY)
+        .value("BROADCAST", Routing::BROADCAST)
+        .value("REBALANCING", Routing::REBALANCING);
+    py::enum_<CollectorKind>(m, "CollectorKind")
+        .value("WATERMARK", CollectorKind::WATERMARK)
+        .value("ORDERING", CollectorKind::ORDERING)
+        .value("KSLACK", CollectorKind::KSLACK)
+        .value("JOIN", CollectorKind::JOIN);
+    py::enum_<DType>(m, "DType")
+        .value("I64", DType::I64)
+        .value("F64", DType::F64)
+        .value("F32", DType::F32)
+        .value("U64", DType::U64)
+        .value("I32", DType::I32)
+        .value("U16", DType::U16)
+        .value("U8", DType::U8);
+
+    py::class_<Engine>(m, "Engine")
+        .def(py::init<>())
+        .def_readwrite("mode", &Engine::mode)
+        .def_readwrite("time_policy", &Engine::time_policy)
+        .def_readwrite("queue_capacity", &Engine::queue_capacity)
+        .def_readwrite("pin_threads", &Engine::pin_threads)
+        .def("add_op",
+             [](Engine& e, const std::string& name, int par, const std::string& kind,
+                const std::string& spec, std::vector<double> fp, std::vector<int64_t> ip,
+                std::vector<int> out_schema, int64_t out_batch, py::object pyfn,
+                int device) {
+                 OpSpec op;
+                 op.id = (int)e.ops.size();
+                 op.name = name;
+                 op.parallelism = par;
+                 op.device = device;
+                 int id = op.id;
+                 op.stages.push_back(make_stage(e, id, kind, spec, fp, ip, out_schema,
+                                                out_batch, pyfn));
+                 e.ops.push_back(std::move(op));
+                 return id;
+             },
+             py::arg("name"), py::arg("parallelism"), py::arg("kind"), py::arg("spec") = "",
+             py::arg("fparams") = std::vector<double>{}, py::arg("iparams") = std::vector<int64_t>{},
+             py::arg("out_schema") = std::vector<int>{}, py::arg("out_batch") = 1024,
+             py::arg("pyfn") = py::none(), py::arg("device") = -1)
+        .def("chain_stage",
+             [](Engine& e, int op_id, const std::string& kind, const std::string& spec,
+                std::vector<double> fp, std::vector<int64_t> ip, std::vector<int> out_schema,
+                int64_t out_batch, py::object pyfn) {
+                 e.ops.at(op_id).stages.push_back(
+                     make_stage(e, op_id, kind, spec, fp, ip, out_schema, out_batch, pyfn));
+             },
+             py::arg("op_id"), py::arg("kind"), py::arg("spec") = "",
+             py::arg("fparams") = std::vector<double>{}, py::arg("iparams") = std::vector<int64_t>{},
+             py::arg("out_schema") = std::vector<int>{}, py::arg("out_batch") = 1024,
+             py::arg("pyfn") = py::none())
+        .def("add_edge",
+             [](Engine& e, int from, int to, Routing r, CollectorKind ck, int tag) {
+                 e.edges.push_back(EdgeSpec{from, to, r, ck, tag});
+             },
+             py::arg("from_op"), py::arg("to_op"), py::arg("routing") = Routing::FORWARD,
+             py::arg("collector") = CollectorKind::WATERMARK, py::arg("stream_tag") = -1)
+        .def("run",
+             [](Engine& e) {
+                 py::gil_scoped_release rel;
+                 e.run();
+             })
+        .def("start",
+             [](Engine& e) {
+                 e.build();
+                 e.start();
+             })
+        .def("wait",
+             [](Engine& e) {
+                 py::gil_scoped_release rel;
+                 e.wait();
+             })
+        .def("abort_now", [](Engine& e) { e.abort.store(true); })
+        .def("sink_sum", [](Engine& e, int op) { return e.sink_acc_i64.at(op).load(); })
+        .def("sink_count", [](Engine& e, int op) { return e.sink_tuples.at(op).load(); })
+        .def("dropped", [](Engine& e) { return e.dropped_tuples.load(); })
+        .def("stats", [](Engine& e) {
+            py::list out;
+            for (auto& r : e.replicas) {
+                py::dict d;
+                d["op"] = r->op_id;
+                d["name"] = e.ops[r->op_id].name;
+                d["replica"] = r->idx;
+                d["inputs"] = r->stats.inputs_received;
+                d["tuples_in"] = r->stats.tuples_received;
+                d["outputs"] = r->stats.outputs_sent;
+                d["tuples_out"] = r->stats.tuples_sent;
+                d["svc_us_ewma"] = r->stats.service_time_us_ewma;
+                d["start_us"] = r->stats.start_us;
+                d["end_us"] = r->stats.end_us;
+                d["kernels"] = r->stats.num_kernels;
+                d["bytes_h2d"] = r->stats.bytes_h2d;
+                d["bytes_d2h"] = r->stats.bytes_d2h;
+                out.append(d);
+            }
+            return out;
+        });
+
+    m.def("hash_key", [](uint64_t k) { return KeyByEmitter::mix(k); });
+}

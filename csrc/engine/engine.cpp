@@ -1,0 +1,476 @@
+#include "engine.hpp"
+
+#include <algorithm>
+#include <cassert>
+
+#ifdef __linux__
+#include <pthread.h>
+#include <sched.h>
+#endif
+
+namespace wfa {
+
+void OpLogic::process(Batch* in, EmitCtx&, RuntimeCtx&) { release(in); }
+
+// ===================== KeyByEmitter =====================
+
+void KeyByEmitter::flush_dest(size_t d) {
+    Batch* b = open[d];
+    if (!b || b->count == 0) return;
+    b->watermark = (open_wm[d] == WM_MAX) ? cur_wm : open_wm[d];
+    account(b);
+    dests[d]->push(b, abort);
+    open[d] = nullptr;
+    open_wm[d] = WM_MAX;
+    sent_recently[d] = true;
+}
+
+void KeyByEmitter::emit(Batch* b) {
+    const size_t n = dests.size();
+    cur_wm = std::max(cur_wm, b->watermark);
+    if (n == 1) {
+        // single destination: pass through untouched
+        account(b);
+        dests[0]->push(b, abort);
+        sent_recently[0] = true;
+    } else {
+        const int64_t cnt_in = b->count;
+        // count per destination
+        std::fill(cnt.begin(), cnt.end(), 0u);
+        // small stack scratch for destination of each row
+        static thread_local std::vector<uint32_t> dof;
+        dof.resize(cnt_in);
+        for (int64_t i = 0; i < cnt_in; ++i) {
+            uint32_t d = (uint32_t)(mix(b->key[i]) % n);
+            dof[i] = d;
+            cnt[d]++;
+        }
+        const size_t np = b->schema.payload.size();
+        for (size_t d = 0; d < n; ++d) {
+            if (!cnt[d]) continue;
+            int64_t remaining = cnt[d];
+            // append rows for dest d, splitting over open batches
+            int64_t i = 0;
+            while (remaining > 0) {
+                if (!open[d]) open[d] = out_pool->get();
+                Batch* o = open[d];
+                int64_t space = o->capacity - o->count;
+                int64_t take = std::min(space, remaining);
+                // gather rows with dof[i]==d
+                int64_t w = o->count;
+                int64_t taken = 0;
+                for (; i < cnt_in && taken < take; ++i) {
+                    if (dof[i] != d) continue;
+                    o->ts[w] = b->ts[i];
+                    o->key[w] = b->key[i];
+                    for (size_t c = 0; c < np; ++c) {
+                        size_t es = dsize(b->schema.payload[c]);
+                        memcpy((char*)o->cols[c] + w * es, (char*)b->cols[c] + i * es, es);
+                    }
+                    ++w;
+                    ++taken;
+                }
+                o->count = w;
+                open_wm[d] = std::min(open_wm[d], b->watermark);
+                remaining -= taken;
+                if (o->count >= o->capacity || o->count >= out_batch) flush_dest(d);
+            }
+        }
+        release(b);
+    }
+    // watermark cadence: flush opens + keep idle destinations alive
+    int64_t t = now_us();
+    if (t - last_cadence_us >= cadence_us) {
+        for (size_t d = 0; d < n; ++d) flush_dest(d);
+        for (size_t d = 0; d < n; ++d) {
+            if (!sent_recently[d]) {
+                Batch* p = punct_pool->get();
+                p->punct = true;
+                p->watermark = cur_wm;
+                dests[d]->push(p, abort);
+            }
+            sent_recently[d] = false;
+        }
+        last_cadence_us = t;
+    }
+}
+
+void KeyByEmitter::punct(int64_t wm) {
+    cur_wm = std::max(cur_wm, wm);
+    for (size_t d = 0; d < dests.size(); ++d) flush_dest(d);
+    for (auto* q : dests) {
+        Batch* p = punct_pool->get();
+        p->punct = true;
+        p->watermark = cur_wm;
+        q->push(p, abort);
+    }
+}
+
+// ===================== collectors =====================
+
+Batch* Collector::next() {
+    int spins = 0;
+    while (n_open > 0) {
+        if (abort && abort->load(std::memory_order_relaxed)) return nullptr;
+        bool got_any = false;
+        for (size_t k = 0; k < chans.size(); ++k) {
+            size_t c = (rr + k) % chans.size();
+            if (!open[c]) continue;
+            Batch* b = chans[c]->try_pop();
+            if (!b) continue;
+            rr = c + 1;
+            got_any = true;
+            if (b == EOS_TAG) {
+                open[c] = false;
+                n_open--;
+                break;  // re-evaluate loop condition
+            }
+            if (b->stream_tag < 0) b->stream_tag = chan_tag[c];
+            chan_wm[c] = std::max(chan_wm[c], b->watermark);
+            int64_t m = min_wm();
+            b->watermark = m;
+            if (b->punct) {
+                if (m > last_fwd_wm) {
+                    last_fwd_wm = m;
+                    return b;
+                }
+                release(b);
+                break;
+            }
+            if (m > last_fwd_wm) last_fwd_wm = m;
+            return b;
+        }
+        if (!got_any) SpscQueue::backoff(spins);
+    }
+    return nullptr;
+}
+
+Batch* OrderingCollector::next() {
+    if (pend.empty()) pend.resize(chans.size());
+    int spins = 0;
+    for (;;) {
+        if (abort && abort->load(std::memory_order_relaxed)) return nullptr;
+        // drain channels into pending queues
+        for (size_t c = 0; c < chans.size(); ++c) {
+            if (!open[c]) continue;
+            Batch* b;
+            while ((b = chans[c]->try_pop()) != nullptr) {
+                if (b == EOS_TAG) {
+                    open[c] = false;
+                    n_open--;
+                    break;
+                }
+                if (b->stream_tag < 0) b->stream_tag = chan_tag[c];
+                chan_wm[c] = std::max(chan_wm[c], b->watermark);
+                pend[c].push_back(b);
+            }
+        }
+        // release the batch with the smallest head timestamp, but only when
+        // every open channel has something pending (or is closed)
+        bool all_ready = true;
+        int best = -1;
+        int64_t best_ts = WM_MAX;
+        for (size_t c = 0; c < chans.size(); ++c) {
+            if (pend[c].empty()) {
+                if (open[c]) all_ready = false;
+                continue;
+            }
+            Batch* h = pend[c].front();
+            int64_t hts = h->punct ? h->watermark : (h->count ? h->ts[0] : h->watermark);
+            if (hts < best_ts) {
+                best_ts = hts;
+                best = (int)c;
+            }
+        }
+        if (best >= 0 && all_ready) {
+            Batch* b = pend[best].front();
+            pend[best].pop_front();
+            b->watermark = min_wm();
+            if (b->punct && b->watermark <= last_fwd_wm) {
+                release(b);
+                continue;
+            }
+            last_fwd_wm = std::max(last_fwd_wm, b->watermark);
+            return b;
+        }
+        if (n_open == 0) {
+            // flush remaining in ts order
+            best = -1;
+            best_ts = WM_MAX;
+            for (size_t c = 0; c < chans.size(); ++c) {
+                if (pend[c].empty()) continue;
+                Batch* h = pend[c].front();
+                int64_t hts = h->punct ? h->watermark : (h->count ? h->ts[0] : h->watermark);
+                if (hts < best_ts) {
+                    best_ts = hts;
+                    best = (int)c;
+                }
+            }
+            if (best < 0) return nullptr;
+            Batch* b = pend[best].front();
+            pend[best].pop_front();
+            b->watermark = min_wm();
+            return b;
+        }
+        SpscQueue::backoff(spins);
+    }
+}
+
+Batch* KSlackCollector::next() {
+    // Batched K-slack: buffer (first_ts, batch); release when
+    // first_ts <= t_curr - K.  K adapts to max observed disorder.
+    int spins = 0;
+    for (;;) {
+        if (abort && abort->load(std::memory_order_relaxed)) return nullptr;
+        for (size_t c = 0; c < chans.size(); ++c) {
+            if (!open[c]) continue;
+            Batch* b;
+            while ((b = chans[c]->try_pop()) != nullptr) {
+                if (b == EOS_TAG) {
+                    open[c] = false;
+                    n_open--;
+                    break;
+                }
+                if (b->stream_tag < 0) b->stream_tag = chan_tag[c];
+                chan_wm[c] = std::max(chan_wm[c], b->watermark);
+                if (b->punct) {
+                    release(b);
+                    continue;
+                }
+                int64_t bts = b->count ? b->ts[0] : b->watermark;
+                if (bts < t_curr) {
+                    int64_t d = t_curr - bts;
+                    if (d > K) K = d;  // adapt slack
+                }
+                t_curr = std::max(t_curr, bts);
+                buf.emplace_back(bts, b);
+            }
+        }
+        if (!buf.empty()) {
+            // release oldest if past slack horizon (or everything at EOS)
+            auto it = std::min_element(buf.begin(), buf.end(),
+                                       [](auto& a, auto& b) { return a.first < b.first; });
+            if (n_open == 0 || it->first <= t_curr - K) {
+                Batch* b = it->second;
+                buf.erase(it);
+                b->watermark = std::max(last_fwd_wm, b->watermark);
+                last_fwd_wm = b->watermark;
+                return b;
+            }
+        } else if (n_open == 0) {
+            return nullptr;
+        }
+        SpscQueue::backoff(spins);
+    }
+}
+
+// ===================== chained stages =====================
+
+struct ChainLogic::StageEmitter : Emitter {
+    OpLogic* next = nullptr;
+    EmitCtx* next_ctx = nullptr;
+    RuntimeCtx* rctx = nullptr;
+    void emit(Batch* b) override { next->process(b, *next_ctx, *rctx); }
+    void punct(int64_t wm) override {
+        for (auto* e : next_ctx->emitters) e->punct(wm);
+    }
+    void flush() override {}
+    void eos() override {}
+};
+
+void ChainLogic::wire(const std::vector<Pool*>& pools, EmitCtx& final_ctx, RuntimeCtx& rctx) {
+    size_t n = stages.size();
+    ctxs.resize(n);
+    for (size_t i = 0; i + 1 < n; ++i) {
+        auto se = std::make_unique<StageEmitter>();
+        se->next = stages[i + 1].get();
+        se->next_ctx = &ctxs[i + 1];
+        se->rctx = &rctx;
+        ctxs[i].out_pool = pools[i];
+        glue.push_back(std::move(se));
+        ctxs[i].emitters = {glue.back().get()};
+    }
+    ctxs[n - 1] = final_ctx;
+}
+
+void ChainLogic::on_eos(EmitCtx&, RuntimeCtx& ctx) {
+    for (size_t i = 0; i < stages.size(); ++i) stages[i]->on_eos(ctxs[i], ctx);
+}
+
+// ===================== replica loop =====================
+
+void Replica::run() {
+#ifdef __linux__
+    if (engine->pin_threads) {
+        cpu_set_t cs;
+        CPU_ZERO(&cs);
+        unsigned ncpu = std::thread::hardware_concurrency();
+        CPU_SET((op_id * 13 + idx) % (ncpu ? ncpu : 1), &cs);
+        pthread_setaffinity_np(pthread_self(), sizeof(cs), &cs);
+    }
+#endif
+    stats.start_us = now_us();
+    if (logic->is_source()) {
+        while (!engine->abort.load(std::memory_order_relaxed)) {
+            if (!logic->source_step(ectx, rctx)) break;
+        }
+    } else {
+        for (;;) {
+            Batch* b = collector->next();
+            if (!b) break;
+            int64_t t0 = now_us();
+            stats.inputs_received++;
+            stats.tuples_received += b->count;
+            rctx.current_wm = b->watermark;
+            if (b->count) rctx.current_ts = b->ts[b->count - 1];
+            if (b->punct) {
+                int64_t wm = b->watermark;
+                release(b);
+                for (auto& e : emitters) e->punct(wm);
+            } else {
+                logic->process(b, ectx, rctx);
+            }
+            double dt = (double)(now_us() - t0);
+            stats.service_time_us_ewma = stats.service_time_us_ewma * 0.95 + dt * 0.05;
+        }
+    }
+    logic->on_eos(ectx, rctx);
+    for (auto& e : emitters) {
+        e->flush();
+        e->eos();
+    }
+    stats.end_us = now_us();
+}
+
+// ===================== engine build/run =====================
+
+void Engine::build() {
+    replicas.clear();
+    queues.clear();
+
+    // adjacency
+    std::vector<std::vector<int>> out_edges(ops.size());
+    std::vector<std::vector<int>> in_edges(ops.size());
+    for (size_t e = 0; e < edges.size(); ++e) {
+        out_edges[edges[e].from].push_back((int)e);
+        in_edges[edges[e].to].push_back((int)e);
+    }
+
+    // per-op, per-stage pools
+    std::vector<std::vector<Pool*>> op_pools(ops.size());
+    std::vector<Pool*> op_pool(ops.size());  // final-stage pool (feeds emitters)
+    Schema punct_schema;  // empty payload
+    Pool* punct_pool = make_pool(punct_schema, 1, false);
+    for (size_t i = 0; i < ops.size(); ++i) {
+        for (auto& st : ops[i].stages)
+            op_pools[i].push_back(
+                make_pool(st.out_schema, std::max<int64_t>(st.out_batch, 1), ops[i].pinned_out));
+        op_pool[i] = op_pools[i].back();
+    }
+
+    // create replicas
+    std::vector<std::vector<Replica*>> op_reps(ops.size());
+    for (size_t i = 0; i < ops.size(); ++i) {
+        for (int r = 0; r < ops[i].parallelism; ++r) {
+            auto rep = std::make_unique<Replica>();
+            rep->engine = this;
+            rep->op_id = (int)i;
+            rep->idx = r;
+            if (ops[i].stages.size() == 1) {
+                rep->logic = ops[i].stages[0].factory();
+            } else {
+                auto ch = std::make_shared<ChainLogic>();
+                for (auto& st : ops[i].stages) ch->stages.push_back(st.factory());
+                rep->logic = ch;
+            }
+            rep->rctx.replica = r;
+            rep->rctx.parallelism = ops[i].parallelism;
+            rep->rctx.engine = this;
+            rep->rctx.op_id = (int)i;
+            rep->ectx.out_pool = op_pool[i];
+            op_reps[i].push_back(rep.get());
+            replicas.push_back(std::move(rep));
+        }
+    }
+
+    // queues + collectors: per edge, per (src replica, dst replica) queue.
+    // Channel list at each consumer replica aggregates over all in-edges.
+    std::vector<std::vector<SpscQueue*>> in_chans(replicas.size());
+    std::vector<std::vector<int>> in_tags(replicas.size());
+    auto rep_gid = [&](int op, int r) {
+        int g = 0;
+        for (int i = 0; i < op; ++i) g += ops[i].parallelism;
+        return g + r;
+    };
+
+    for (auto& es : edges) {
+        auto& src_reps = op_reps[es.from];
+        auto& dst_reps = op_reps[es.to];
+        for (auto* sr : src_reps) {
+            std::vector<SpscQueue*> dq;
+            for (auto* dr : dst_reps) {
+                queues.push_back(std::make_unique<SpscQueue>(queue_capacity));
+                SpscQueue* q = queues.back().get();
+                dq.push_back(q);
+                int g = rep_gid(es.to, dr->idx);
+                in_chans[g].push_back(q);
+                in_tags[g].push_back(es.stream_tag);
+            }
+            std::unique_ptr<Emitter> em;
+            switch (es.routing) {
+                case Routing::KEYBY:
+                    em = std::make_unique<KeyByEmitter>(dq, op_pool[es.from], punct_pool,
+                                                        ops[es.from].last().out_batch);
+                    break;
+                case Routing::BROADCAST:
+                    em = std::make_unique<BroadcastEmitter>(dq, punct_pool);
+                    break;
+                default:
+                    em = std::make_unique<ForwardEmitter>(dq, punct_pool);
+            }
+            em->abort = &abort;
+            em->stats = &sr->stats;
+            sr->emitters.push_back(std::move(em));
+        }
+    }
+    // wire EmitCtx + collectors
+    for (auto& rep : replicas) {
+        for (auto& e : rep->emitters) rep->ectx.emitters.push_back(e.get());
+        if (auto* ch = dynamic_cast<ChainLogic*>(rep->logic.get()))
+            ch->wire(op_pools[rep->op_id], rep->ectx, rep->rctx);
+        if (!rep->logic->is_source()) {
+            CollectorKind ck = CollectorKind::WATERMARK;
+            for (auto& es : edges)
+                if (es.to == rep->op_id) ck = es.collector;
+            int g = rep_gid(rep->op_id, rep->idx);
+            switch (ck) {
+                case CollectorKind::ORDERING:
+                    rep->collector = std::make_unique<OrderingCollector>(in_chans[g], in_tags[g]);
+                    break;
+                case CollectorKind::KSLACK: {
+                    auto ks = std::make_unique<KSlackCollector>(in_chans[g], in_tags[g]);
+                    ks->dropped = &dropped_tuples;
+                    rep->collector = std::move(ks);
+                    break;
+                }
+                default:
+                    rep->collector = std::make_unique<Collector>(in_chans[g], in_tags[g]);
+            }
+            rep->collector->abort = &abort;
+        }
+    }
+}
+
+void Engine::start() {
+    for (auto& r : replicas) {
+        Replica* rp = r.get();
+        rp->th = std::thread([rp] { rp->run(); });
+    }
+}
+
+void Engine::wait() {
+    for (auto& r : replicas)
+        if (r->th.joinable()) r->th.join();
+}
+
+}  // namespace wfa

@@ -1,0 +1,251 @@
+// Native (compiled) operator-logic catalog for the CPU path.
+//
+// The reference takes user C++ lambdas (API file: SOURCE/MAP/FILTER/...).
+// Our Python front end offers two forms: per-batch Python callbacks
+// (py_logic in bindings.cpp, GIL-amortized over batches) and this catalog
+// of compiled functors for hot paths/benchmarks — parameterized versions
+// of the reference test fixtures (tests/graph_tests/graph_common.hpp).
+#include <cmath>
+#include <random>
+
+#include "engine.hpp"
+
+namespace wfa {
+
+// ----- helpers -----
+static Batch* ensure_exclusive(Batch* b, EmitCtx& out) {
+    if (b->refcnt.load(std::memory_order_acquire) > 1) {
+        Batch* c = clone(b, *b->pool);
+        release(b);
+        return c;
+    }
+    return b;
+}
+
+// ----- source: deterministic keyed integer sequence -----
+// iparams: [stream_len, n_keys, batch_size, value_offset]
+// Emits values v = 1..stream_len, key = v % n_keys, ts = v (event time),
+// watermark = last ts.  Mirrors the synthetic sources of the reference's
+// differential tests (tests/graph_tests/graph_common.hpp:60-130).
+struct SeqSource : OpLogic {
+    int64_t len, n_keys, bsz, voff;
+    int64_t pos = 0;
+    SeqSource(int64_t l, int64_t k, int64_t b, int64_t vo)
+        : len(l), n_keys(k), bsz(b), voff(vo) {}
+    bool is_source() const override { return true; }
+    bool source_step(EmitCtx& out, RuntimeCtx& ctx) override {
+        if (pos >= len) return false;
+        Batch* b = out.new_batch();
+        int64_t n = std::min<int64_t>(std::min<int64_t>(bsz, b->capacity), len - pos);
+        for (int64_t i = 0; i < n; ++i) {
+            int64_t v = voff + pos + i + 1;
+            b->ts[i] = pos + i + 1;
+            b->key[i] = (uint64_t)(v % n_keys);
+            b->col<int64_t>(0)[i] = v;
+        }
+        b->count = n;
+        pos += n;
+        b->watermark = pos;  // DEFAULT mode: wm = max emitted ts
+        out.emit(b);
+        return pos < len;
+    }
+};
+
+// ----- source: random keyed stream with value column -----
+// iparams: [stream_len, n_keys, batch_size, seed]
+struct RandSource : OpLogic {
+    int64_t len, n_keys, bsz;
+    uint64_t seed;
+    int64_t pos = 0;
+    std::mt19937_64 rng;
+    RandSource(int64_t l, int64_t k, int64_t b, uint64_t s)
+        : len(l), n_keys(k), bsz(b), seed(s), rng(s) {}
+    bool is_source() const override { return true; }
+    bool source_step(EmitCtx& out, RuntimeCtx& ctx) override {
+        if (pos >= len) return false;
+        Batch* b = out.new_batch();
+        int64_t n = std::min<int64_t>(std::min<int64_t>(bsz, b->capacity), len - pos);
+        for (int64_t i = 0; i < n; ++i) {
+            b->ts[i] = pos + i + 1;
+            b->key[i] = rng() % (uint64_t)n_keys;
+            b->col<int64_t>(0)[i] = (int64_t)(rng() % 1000);
+        }
+        b->count = n;
+        pos += n;
+        b->watermark = pos;
+        out.emit(b);
+        return pos < len;
+    }
+};
+
+// ----- map: x = a*x + b on an i64 column, in place -----
+struct AffineMapI64 : OpLogic {
+    int col;
+    int64_t a, bb;
+    AffineMapI64(int c, int64_t a_, int64_t b_) : col(c), a(a_), bb(b_) {}
+    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+        b = ensure_exclusive(b, out);
+        int64_t* x = b->col<int64_t>(col);
+        const int64_t n = b->count;
+        for (int64_t i = 0; i < n; ++i) x[i] = a * x[i] + bb;
+        out.emit(b);
+    }
+};
+
+// ----- filter: keep rows where (x % m != c) (or ==, via keep_eq) -----
+struct ModFilterI64 : OpLogic {
+    int col;
+    int64_t m, c;
+    bool keep_eq;
+    ModFilterI64(int col_, int64_t m_, int64_t c_, bool ke) : col(col_), m(m_), c(c_), keep_eq(ke) {}
+    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+        b = ensure_exclusive(b, out);
+        int64_t* x = b->col<int64_t>(col);
+        const size_t np = b->schema.payload.size();
+        int64_t w = 0;
+        for (int64_t i = 0; i < b->count; ++i) {
+            bool keep = ((x[i] % m) == c) == keep_eq;
+            if (!keep) continue;
+            if (w != i) {
+                b->ts[w] = b->ts[i];
+                b->key[w] = b->key[i];
+                for (size_t cc = 0; cc < np; ++cc) {
+                    size_t es = dsize(b->schema.payload[cc]);
+                    memcpy((char*)b->cols[cc] + w * es, (char*)b->cols[cc] + i * es, es);
+                }
+            }
+            ++w;
+        }
+        b->count = w;
+        if (w > 0)
+            out.emit(b);
+        else {
+            // all dropped: keep watermarks flowing (reference filter.hpp:151)
+            int64_t wm = b->watermark;
+            release(b);
+            for (auto* e : out.emitters) e->punct(wm);
+        }
+    }
+};
+
+// ----- flatmap: emit each input row `k` times (value gets +j per copy) ---
+struct DupFlatMapI64 : OpLogic {
+    int64_t k;
+    explicit DupFlatMapI64(int64_t k_) : k(k_) {}
+    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+        Batch* o = out.new_batch();
+        const size_t np = b->schema.payload.size();
+        for (int64_t i = 0; i < b->count; ++i) {
+            for (int64_t j = 0; j < k; ++j) {
+                if (o->count == o->capacity) {
+                    o->watermark = b->watermark;
+                    out.emit(o);
+                    o = out.new_batch();
+                }
+                int64_t w = o->count++;
+                o->ts[w] = b->ts[i];
+                o->key[w] = b->key[i];
+                for (size_t cc = 0; cc < np; ++cc) {
+                    size_t es = dsize(b->schema.payload[cc]);
+                    memcpy((char*)o->cols[cc] + w * es, (char*)b->cols[cc] + i * es, es);
+                }
+            }
+        }
+        o->watermark = b->watermark;
+        if (o->count)
+            out.emit(o);
+        else
+            release(o);
+        release(b);
+    }
+};
+
+// ----- reduce: keyed running sum, emits updated (key, acc) per input ----
+// (reference: wf/reduce.hpp — KEYBY routing, per-key state in replica map)
+struct KeyedSumReduceI64 : OpLogic {
+    int col;
+    std::unordered_map<uint64_t, int64_t> acc;
+    explicit KeyedSumReduceI64(int c) : col(c) { acc.reserve(1 << 12); }
+    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+        Batch* o = out.new_batch();
+        int64_t* x = b->col<int64_t>(col);
+        for (int64_t i = 0; i < b->count; ++i) {
+            int64_t& a = acc[b->key[i]];
+            a += x[i];
+            if (o->count == o->capacity) {
+                o->watermark = b->watermark;
+                out.emit(o);
+                o = out.new_batch();
+            }
+            int64_t w = o->count++;
+            o->ts[w] = b->ts[i];
+            o->key[w] = b->key[i];
+            o->col<int64_t>(0)[w] = a;
+        }
+        o->watermark = b->watermark;
+        if (o->count)
+            out.emit(o);
+        else
+            release(o);
+        release(b);
+    }
+};
+
+// ----- sink: sum an i64 column into the engine accumulator -----
+struct SumSinkI64 : OpLogic {
+    Engine* eng;
+    int op_id;
+    int col;
+    int64_t local = 0, tuples = 0;
+    SumSinkI64(Engine* e, int id, int c) : eng(e), op_id(id), col(c) {}
+    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+        int64_t* x = b->col<int64_t>(col);
+        for (int64_t i = 0; i < b->count; ++i) local += x[i];
+        tuples += b->count;
+        release(b);
+    }
+    void on_eos(EmitCtx&, RuntimeCtx&) override {
+        eng->sink_acc_i64[op_id].fetch_add(local, std::memory_order_relaxed);
+        eng->sink_tuples[op_id].fetch_add(tuples, std::memory_order_relaxed);
+    }
+};
+
+// ----- sink: count only (throughput benchmarks) -----
+struct CountSink : OpLogic {
+    Engine* eng;
+    int op_id;
+    int64_t tuples = 0;
+    CountSink(Engine* e, int id) : eng(e), op_id(id) {}
+    void process(Batch* b, EmitCtx&, RuntimeCtx&) override {
+        tuples += b->count;
+        release(b);
+    }
+    void on_eos(EmitCtx&, RuntimeCtx&) override {
+        eng->sink_tuples[op_id].fetch_add(tuples, std::memory_order_relaxed);
+    }
+};
+
+std::shared_ptr<OpLogic> make_native_logic(const std::string& kind, const std::string& spec,
+                                           const std::vector<double>& fp,
+                                           const std::vector<int64_t>& ip, Engine* eng,
+                                           int op_id) {
+    if (kind == "source" && spec == "seq")
+        return std::make_shared<SeqSource>(ip[0], ip[1], ip[2], ip.size() > 3 ? ip[3] : 0);
+    if (kind == "source" && spec == "rand")
+        return std::make_shared<RandSource>(ip[0], ip[1], ip[2], ip.size() > 3 ? ip[3] : 42);
+    if (kind == "map" && spec == "affine_i64")
+        return std::make_shared<AffineMapI64>((int)ip[0], ip[1], ip[2]);
+    if (kind == "filter" && spec == "mod_i64")
+        return std::make_shared<ModFilterI64>((int)ip[0], ip[1], ip[2], ip[3] != 0);
+    if (kind == "flatmap" && spec == "dup_i64")
+        return std::make_shared<DupFlatMapI64>(ip[0]);
+    if (kind == "reduce" && spec == "sum_by_key_i64")
+        return std::make_shared<KeyedSumReduceI64>((int)ip[0]);
+    if (kind == "sink" && spec == "sum_i64")
+        return std::make_shared<SumSinkI64>(eng, op_id, (int)ip[0]);
+    if (kind == "sink" && spec == "count")
+        return std::make_shared<CountSink>(eng, op_id);
+    throw std::runtime_error("unknown native logic: " + kind + "/" + spec);
+}
+
+}  // namespace wfa

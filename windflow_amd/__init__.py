@@ -1,0 +1,11 @@
+"""windflow_amd — MI355X-native streaming dataflow engine.
+
+A from-scratch re-design of the WindFlow programming model (MultiPipe /
+PipeGraph builder API, CPU+GPU operator set, FlatFAT windowed aggregation)
+for AMD MI355X: a native C++ runtime (pinned threads + SPSC queues), SoA
+micro-batches as the universal message unit, hand-written HIP/CDNA4 kernels
+for the GPU operators, and RCCL-over-xGMI for multi-GPU shuffles.
+"""
+from .basic import (ExecutionMode, TimePolicy, WinType, JoinMode, RoutingMode, DType)  # noqa: F401
+
+__version__ = "0.1.0"
