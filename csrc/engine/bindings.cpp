@@ -319,8 +319,12 @@ PYBIND11_MODULE(_core, m) {
              py::arg("collector") = CollectorKind::WATERMARK, py::arg("stream_tag") = -1)
         .def("run",
              [](Engine& e) {
-                 py::gil_scoped_release rel;
-                 e.run();
+                 e.build();  // copies py::function logic — needs the GIL
+                 {
+                     py::gil_scoped_release rel;
+                     e.start();
+                     e.wait();
+                 }
              })
         .def("start",
              [](Engine& e) {

@@ -225,6 +225,55 @@ struct CountSink : OpLogic {
     }
 };
 
+// ----- split: route rows to branches (reference: wf/splitting_emitter.hpp) --
+// Gathers rows per branch into fresh batches and emits on that branch edge.
+struct SplitModI64 : OpLogic {
+    int col;
+    explicit SplitModI64(int c) : col(c) {}
+    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+        const size_t nb = out.n_branches();
+        const size_t np = b->schema.payload.size();
+        int64_t* x = b->col<int64_t>(col);
+        for (size_t br = 0; br < nb; ++br) {
+            Batch* o = nullptr;
+            for (int64_t i = 0; i < b->count; ++i) {
+                size_t d = (size_t)(((x[i] % (int64_t)nb) + (int64_t)nb) % (int64_t)nb);
+                if (d != br) continue;
+                if (!o) o = out.new_batch();
+                if (o->count == o->capacity) {
+                    o->watermark = b->watermark;
+                    out.emit_to(br, o);
+                    o = out.new_batch();
+                }
+                int64_t w = o->count++;
+                o->ts[w] = b->ts[i];
+                o->key[w] = b->key[i];
+                for (size_t cc = 0; cc < np; ++cc) {
+                    size_t es = dsize(b->schema.payload[cc]);
+                    memcpy((char*)o->cols[cc] + w * es, (char*)b->cols[cc] + i * es, es);
+                }
+            }
+            if (o) {
+                o->watermark = b->watermark;
+                if (o->count)
+                    out.emit_to(br, o);
+                else
+                    release(o);
+            }
+        }
+        release(b);
+    }
+};
+
+// round-robin whole batches across branches (split_gpu-style replication-free)
+struct SplitRR : OpLogic {
+    size_t rr = 0;
+    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+        out.emit_to(rr, b);
+        rr = (rr + 1) % out.n_branches();
+    }
+};
+
 std::shared_ptr<OpLogic> make_native_logic(const std::string& kind, const std::string& spec,
                                            const std::vector<double>& fp,
                                            const std::vector<int64_t>& ip, Engine* eng,
@@ -245,6 +294,10 @@ std::shared_ptr<OpLogic> make_native_logic(const std::string& kind, const std::s
         return std::make_shared<SumSinkI64>(eng, op_id, (int)ip[0]);
     if (kind == "sink" && spec == "count")
         return std::make_shared<CountSink>(eng, op_id);
+    if (kind == "split" && spec == "mod_i64")
+        return std::make_shared<SplitModI64>((int)(ip.empty() ? 0 : ip[0]));
+    if (kind == "split" && spec == "rr")
+        return std::make_shared<SplitRR>();
     throw std::runtime_error("unknown native logic: " + kind + "/" + spec);
 }
 

@@ -1,0 +1,135 @@
+"""Merge/split DAG topologies (reference: tests/graph_tests, merge_tests,
+split_tests) with the same randomized differential invariant."""
+import random
+
+import numpy as np
+
+import windflow_amd as wf
+from windflow_amd import native
+
+
+def test_merge_two_pipes():
+    # Source1->Map \
+    #               +--> Filter -> Sink    (test_graph_1 shape, single sink)
+    # Source2->Map /
+    rng = random.Random(7)
+    stream_len = 15000
+    exp_one = [3 * v + 1 for v in range(1, stream_len + 1)]
+    exp = sum(x for x in exp_one if x % 5 != 0) * 2  # two sources par=1
+    for _ in range(3):
+        g = wf.PipeGraph("merge")
+        s1 = (wf.Source_Builder(native.seq_source(stream_len, 11, 128))
+              .withParallelism(1).withOutputSchema([0]).build())
+        s2 = (wf.Source_Builder(native.seq_source(stream_len, 11, 128))
+              .withParallelism(1).withOutputSchema([0]).build())
+        mp1 = g.add_source(s1)
+        mp2 = g.add_source(s2)
+        mp1.add(wf.Map_Builder(native.affine_map(0, 3, 1))
+                .withParallelism(rng.randint(1, 3)).withOutputSchema([0]).build())
+        mp2.add(wf.Map_Builder(native.affine_map(0, 3, 1))
+                .withParallelism(rng.randint(1, 3)).withOutputSchema([0]).build())
+        mp = mp1.merge(mp2)
+        mp.add(wf.Filter_Builder(native.mod_filter(0, 5, 0))
+               .withParallelism(rng.randint(1, 3)).withOutputSchema([0]).build())
+        snk = wf.Sink_Builder(native.sum_sink(0)).withParallelism(rng.randint(1, 2)).build()
+        mp.add_sink(snk)
+        g.run()
+        assert g.sink_sum(snk) == exp
+
+
+def test_split_two_branches():
+    # Source -> split(v%2) -> [branch0 -> SinkA, branch1 -> Map -> SinkB]
+    stream_len = 10000
+    g = wf.PipeGraph("split")
+    src = (wf.Source_Builder(native.seq_source(stream_len, 5, 64))
+           .withParallelism(1).withOutputSchema([0]).build())
+    mp = g.add_source(src)
+    mp.split(native.split_mod(0), 2)
+    b0 = mp.select(0)
+    b1 = mp.select(1)
+    snk_a = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+    b0.add_sink(snk_a)
+    b1.add(wf.Map_Builder(native.affine_map(0, 1, 100)).withParallelism(2)
+           .withOutputSchema([0]).build())
+    snk_b = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+    b1.add_sink(snk_b)
+    g.run()
+    evens = sum(v for v in range(1, stream_len + 1) if v % 2 == 0)
+    odds = sum(v + 100 for v in range(1, stream_len + 1) if v % 2 == 1)
+    assert g.sink_sum(snk_a) == evens
+    assert g.sink_sum(snk_b) == odds
+
+
+def test_python_callbacks_pipeline():
+    """Per-batch numpy callbacks: source, map, filter, flatmap, sink."""
+    stream_len = 5000
+    state = dict(pos=0, total=0, n=0)
+
+    def pysource(replica, par):
+        if state['pos'] >= stream_len:
+            return None
+        n = min(512, stream_len - state['pos'])
+        v = np.arange(state['pos'] + 1, state['pos'] + n + 1, dtype=np.int64)
+        state['pos'] += n
+        return dict(c0=v, ts=v, key=(v % 9).astype(np.uint64),
+                    watermark=int(v[-1]))
+
+    def pymap(cols):
+        cols['c0'][:] = cols['c0'] * 2 + 1
+
+    def pyfilter(cols):
+        return (cols['c0'] % 3 != 0)
+
+    def pyflat(cols):
+        v = np.repeat(cols['c0'], 2)
+        return dict(c0=v, ts=np.repeat(cols['ts'], 2), key=np.repeat(cols['key'], 2))
+
+    def pysink(cols):
+        state['total'] += int(cols['c0'].sum())
+        state['n'] += len(cols['c0'])
+
+    g = wf.PipeGraph("py")
+    mp = g.add_source(wf.Source_Builder(pysource).withParallelism(1)
+                      .withOutputSchema([0]).build())
+    mp.add(wf.Map_Builder(pymap).withParallelism(1).withOutputSchema([0]).build())
+    mp.add(wf.Filter_Builder(pyfilter).withParallelism(1).withOutputSchema([0]).build())
+    mp.add(wf.FlatMap_Builder(pyflat).withParallelism(1).withOutputSchema([0]).build())
+    mp.add_sink(wf.Sink_Builder(pysink).withParallelism(1).build())
+    g.run()
+    vals = [2 * v + 1 for v in range(1, stream_len + 1)]
+    keep = [x for x in vals if x % 3 != 0]
+    assert state['total'] == sum(keep) * 2
+    assert state['n'] == len(keep) * 2
+
+
+def test_broadcast_routing():
+    """BROADCAST: every replica of the consumer sees every tuple."""
+    from windflow_amd.operators import Operator
+    stream_len = 3000
+    g = wf.PipeGraph("bc")
+    src = (wf.Source_Builder(native.seq_source(stream_len, 3, 128))
+           .withParallelism(1).withOutputSchema([0]).build())
+    mp = g.add_source(src)
+    snk_op = wf.Sink_Builder(native.sum_sink(0)).withParallelism(3).build()
+    snk_op.broadcast_input = True
+    mp.add_sink(snk_op)
+    g.run()
+    exp = sum(range(1, stream_len + 1)) * 3
+    assert g.sink_sum(snk_op) == exp
+
+
+def test_stats_records():
+    g = wf.PipeGraph("st")
+    src = (wf.Source_Builder(native.seq_source(1000, 3, 128))
+           .withParallelism(2).withOutputSchema([0]).build())
+    mp = g.add_source(src)
+    snk = wf.Sink_Builder(native.count_sink()).withParallelism(1).build()
+    mp.add_sink(snk)
+    g.run()
+    st = g.stats()
+    assert len(st) == 3
+    tuples_in = sum(r['tuples_in'] for r in st if r['name'] == 'sink')
+    assert tuples_in == 2000
+    assert g.sink_count(snk) == 2000
+    js = g.stats_json()
+    assert 'operators' in js

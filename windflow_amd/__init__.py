@@ -9,3 +9,11 @@ for the GPU operators, and RCCL-over-xGMI for multi-GPU shuffles.
 from .basic import (ExecutionMode, TimePolicy, WinType, JoinMode, RoutingMode, DType)  # noqa: F401
 
 __version__ = "0.1.0"
+
+from .builders import (  # noqa: F401,E402
+    Source_Builder, Map_Builder, Filter_Builder, FlatMap_Builder,
+    Reduce_Builder, Sink_Builder, Keyed_Windows_Builder,
+    Parallel_Windows_Builder, Paned_Windows_Builder,
+    MapReduce_Windows_Builder, Ffat_Windows_Builder, Interval_Join_Builder)
+from .pipegraph import PipeGraph, MultiPipe  # noqa: F401,E402
+from . import native  # noqa: F401,E402

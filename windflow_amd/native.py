@@ -1,0 +1,53 @@
+"""Native (compiled) logic specs — the fast path for sources/ops/sinks.
+
+These map 1:1 onto the C++ catalog in csrc/engine/native_logic.cpp (CPU)
+and csrc/hip/kernels.hip (GPU device functors).  Python callables are the
+general path; these are the zero-interpreter hot paths the benchmarks use.
+"""
+from .operators import NativeLogic
+
+I64, F64, F32, U64, I32, U16, U8 = range(7)
+
+
+def seq_source(stream_len, n_keys=1, batch=1024, value_offset=0):
+    """Deterministic keyed integer sequence: v=1..len, key=v%n_keys, ts=v."""
+    return NativeLogic("source", "seq", [], [stream_len, n_keys, batch, value_offset])
+
+
+def rand_source(stream_len, n_keys=1, batch=1024, seed=42):
+    """Random keyed stream: key ~ U[0,n_keys), value ~ U[0,1000)."""
+    return NativeLogic("source", "rand", [], [stream_len, n_keys, batch, seed])
+
+
+def affine_map(col=0, a=1, b=0):
+    """x = a*x + b on i64 column `col` (in place)."""
+    return NativeLogic("map", "affine_i64", [], [col, a, b])
+
+
+def mod_filter(col=0, m=2, c=0, keep_eq=False):
+    """keep rows where (x % m != c); keep_eq flips to ==."""
+    return NativeLogic("filter", "mod_i64", [], [col, m, c, 1 if keep_eq else 0])
+
+
+def dup_flatmap(k=2):
+    """emit each input row k times."""
+    return NativeLogic("flatmap", "dup_i64", [], [k])
+
+
+def keyed_sum_reduce(col=0):
+    """per-key running sum; emits the updated (key, acc) per input tuple."""
+    return NativeLogic("reduce", "sum_by_key_i64", [], [col])
+
+
+def sum_sink(col=0):
+    """accumulate sum of i64 column into engine accumulator (invariant tests)."""
+    return NativeLogic("sink", "sum_i64", [], [col])
+
+
+def count_sink():
+    return NativeLogic("sink", "count", [], [])
+
+
+def split_mod(col=0):
+    """split branch = x % n_branches on i64 column."""
+    return NativeLogic("split", "mod_i64", [], [col])
