@@ -1,0 +1,166 @@
+#!/usr/bin/env python3
+"""windflow_amd flagship benchmark — driver contract.
+
+Headline metric (BASELINE.json): tuples/sec (whole node) on a keyed FFAT
+sliding window, count-based win=1000 slide=100, bf16 tuple values with
+random keys, synthetic GPU-resident stream.  One rank per GPU; for N>1
+the driver launches via torch.distributed.run (RCCL backend).
+
+A "step" = one micro-batch of --batch tuples through the pipeline
+(device source -> keyed FFAT window -> sink).  W warmup steps run first
+(untimed), then exactly K steps are timed, bracketed by barrier +
+torch.cuda.synchronize on both sides; value = whole-job aggregate
+tuples/sec (max step time over ranks).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, ROOT)
+
+
+def build_ffat_graph(n_tuples, batch, n_keys, win, slide, rank, world, device):
+    import windflow_amd as wf
+    from windflow_amd import native_gpu
+    from windflow_amd.builders_gpu import (Source_GPU_Builder,
+                                           Ffat_Windows_GPU_Builder,
+                                           Sink_GPU_Builder)
+    src = (Source_GPU_Builder(
+        native_gpu.gpu_source(n_tuples, n_keys, batch, vdt=5, seed=42 + rank))
+        .withOutputSchema([5]).withOutputBatchSize(batch)
+        .withDevice(device).build())
+    ff = (Ffat_Windows_GPU_Builder(
+        native_gpu.gpu_ffat_windows(native_gpu.COMB_SUM, 0, win, slide,
+                                    max_keys=2 * n_keys))
+        .withOutputSchema([2]).withOutputBatchSize(batch)
+        .withDevice(device).build())
+    snk = (Sink_GPU_Builder(native_gpu.gpu_count_sink())
+           .withDevice(device).build())
+    g = wf.PipeGraph("bench_ffat")
+    mp = g.add_source(src)
+    mp.chain(ff)
+    mp.chain_sink(snk)
+    return g, snk
+
+
+def build_cpu_graph(n_tuples, batch):
+    import windflow_amd as wf
+    from windflow_amd import native
+    src = (wf.Source_Builder(native.seq_source(n_tuples, 1000, batch))
+           .withParallelism(1).withOutputSchema([0])
+           .withOutputBatchSize(batch).build())
+    g = wf.PipeGraph("bench_cpu")
+    mp = g.add_source(src)
+    mp.chain(wf.Map_Builder(native.affine_map(0, 3, 1)).withParallelism(1)
+             .withOutputSchema([0]).withOutputBatchSize(batch).build())
+    mp.chain(wf.Filter_Builder(native.mod_filter(0, 5, 0)).withParallelism(1)
+             .withOutputSchema([0]).withOutputBatchSize(batch).build())
+    snk = wf.Sink_Builder(native.count_sink()).withParallelism(1).build()
+    mp.add_sink(snk)
+    return g, snk
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=4_000_000)
+    ap.add_argument("--keys", type=int, default=1 << 16,
+                    help="distinct keys per rank")
+    ap.add_argument("--win", type=int, default=1000)
+    ap.add_argument("--slide", type=int, default=100)
+    ap.add_argument("--config", choices=["ffat", "cpu"], default="ffat")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    dist = None
+    torch = None
+    if args.config == "ffat":
+        import torch  # noqa: F811
+        if world > 1:
+            import torch.distributed as dist  # noqa: F811
+            dist.init_process_group("nccl", rank=rank, world_size=world)
+            torch.cuda.set_device(local_rank)
+
+    def sync():
+        if torch is not None and torch.cuda.is_available():
+            torch.cuda.synchronize()
+
+    def barrier():
+        if dist is not None:
+            dist.barrier()
+
+    K, W, B = args.steps, args.warmup, args.batch
+
+    if args.config == "cpu":
+        B = min(B, 65536)
+        gw, _ = build_cpu_graph(W * B, B)
+        gw.run()
+        g, snk = build_cpu_graph(K * B, B)
+        t0 = time.time()
+        g.run()
+        dt = time.time() - t0
+        n_gpus = 0
+    else:
+        # warmup engine (also JIT-warms pools/streams/arena)
+        if W > 0:
+            gw, _ = build_ffat_graph(W * B, B, args.keys, args.win, args.slide,
+                                     rank, world, local_rank)
+            gw.run()
+        sync()
+        barrier()
+        t0 = time.time()
+        g, snk = build_ffat_graph(K * B, B, args.keys, args.win, args.slide,
+                                  rank, world, local_rank)
+        g.run()
+        sync()
+        t1 = time.time()
+        barrier()
+        dt = t1 - t0
+        n_gpus = world
+
+    # max step time over ranks
+    if dist is not None:
+        t = torch.tensor([dt], dtype=torch.float64, device="cuda")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        dt = float(t.item())
+
+    total_tuples = K * B * max(world, 1)
+    value = total_tuples / dt
+    if rank == 0:
+        out = {
+            "metric": "tuples_per_sec",
+            "value": value,
+            "unit": "tuples/s",
+            "n_gpus": n_gpus if args.config == "ffat" else 0,
+            "steps": K,
+            "warmup": W,
+            "ms_per_step": dt / K * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if args.config == "ffat" else "int64",
+            "data": "synthetic",
+            "config": {
+                "model": "keyed_ffat_cb_window" if args.config == "ffat"
+                         else "cpu_source_map_filter_sink",
+                "global_batch": B * max(world, 1),
+                "win": args.win,
+                "slide": args.slide,
+                "keys_per_rank": args.keys,
+                "parallelism": f"keyed-dp{max(world,1)}",
+            },
+        }
+        print(json.dumps(out))
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

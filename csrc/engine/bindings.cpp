@@ -211,13 +211,22 @@ struct KeyByColLogic : OpLogic {
 
 static StageSpec make_stage(Engine& e, int id, const std::string& kind, const std::string& spec,
                             std::vector<double> fp, std::vector<int64_t> ip,
-                            std::vector<int> out_schema, int64_t out_batch, py::object pyfn) {
+                            std::vector<int> out_schema, int64_t out_batch, py::object pyfn,
+                            int device = -1) {
     StageSpec st;
     for (int d : out_schema) st.out_schema.payload.push_back((DType)d);
     st.out_batch = out_batch;
     if (kind == "sink") {
         e.sink_acc_i64[id].store(0);
         e.sink_tuples[id].store(0);
+    }
+    if (kind.rfind("gpu_", 0) == 0) {
+        Engine* ep = &e;
+        Schema os = st.out_schema;
+        st.factory = [kind, spec, fp, ip, ep, id, device, os, out_batch] {
+            return make_gpu_logic(kind, spec, fp, ip, ep, id, device, os, out_batch);
+        };
+        return st;
     }
     if (!pyfn.is_none()) {
         auto fn = pyfn.cast<py::function>();
@@ -292,7 +301,7 @@ PYBIND11_MODULE(_core, m) {
                  op.device = device;
                  int id = op.id;
                  op.stages.push_back(make_stage(e, id, kind, spec, fp, ip, out_schema,
-                                                out_batch, pyfn));
+                                                out_batch, pyfn, device));
                  e.ops.push_back(std::move(op));
                  return id;
              },
@@ -305,7 +314,8 @@ PYBIND11_MODULE(_core, m) {
                 std::vector<double> fp, std::vector<int64_t> ip, std::vector<int> out_schema,
                 int64_t out_batch, py::object pyfn) {
                  e.ops.at(op_id).stages.push_back(
-                     make_stage(e, op_id, kind, spec, fp, ip, out_schema, out_batch, pyfn));
+                     make_stage(e, op_id, kind, spec, fp, ip, out_schema, out_batch, pyfn,
+                                e.ops.at(op_id).device));
              },
              py::arg("op_id"), py::arg("kind"), py::arg("spec") = "",
              py::arg("fparams") = std::vector<double>{}, py::arg("iparams") = std::vector<int64_t>{},

@@ -37,6 +37,7 @@ void host_free(void* p, bool pinned) {
 }
 
 static Batch* alloc_batch(const Schema& s, int64_t cap, bool pinned) {
+    (void)s;
     Batch* b = new Batch();
     b->capacity = cap;
     b->schema = s;
@@ -54,8 +55,10 @@ static void free_batch(Batch* b) {
         host_free(b->ts, b->pinned);
         host_free(b->key, b->pinned);
         for (size_t i = 0; i < b->cols.size(); ++i) host_free(b->cols[i], b->pinned);
+        delete b;
+    } else {
+        gpu_free_batch(b);
     }
-    delete b;
 }
 
 Pool::~Pool() {
@@ -76,7 +79,8 @@ Batch* Pool::get() {
             return b;
         }
     }
-    Batch* b = alloc_batch(schema, capacity, pinned);
+    Batch* b = (loc == Loc::DEVICE) ? gpu_alloc_batch(*this)
+                                    : alloc_batch(schema, capacity, pinned);
     b->pool = this;
     live.fetch_add(1, std::memory_order_relaxed);
     return b;

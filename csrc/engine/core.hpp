@@ -114,6 +114,11 @@ struct Pool {
 // drop one reference; recycle into pool when it hits zero
 void release(Batch* b);
 
+// device-batch hooks (gpu_ops.cpp): one contiguous HBM slab per batch from
+// the per-GPU arena allocator — the recycling_gpu redesign for 288 GB HBM.
+Batch* gpu_alloc_batch(Pool& pool);
+void gpu_free_batch(Batch* b);
+
 // Deep-copy b (same pool class) — used for copy-on-write under broadcast.
 Batch* clone(Batch* b, Pool& pool);
 

@@ -387,6 +387,7 @@ void Engine::build() {
             rep->rctx.parallelism = ops[i].parallelism;
             rep->rctx.engine = this;
             rep->rctx.op_id = (int)i;
+            rep->rctx.stats = &rep->stats;
             rep->ectx.out_pool = op_pool[i];
             op_reps[i].push_back(rep.get());
             replicas.push_back(std::move(rep));

@@ -55,6 +55,7 @@ struct RuntimeCtx {
     int64_t current_wm = 0;
     Engine* engine = nullptr;
     int op_id = 0;
+    StatsRecord* stats = nullptr;
 };
 
 // ===== emitters =====
@@ -360,5 +361,13 @@ std::shared_ptr<OpLogic> make_native_logic(const std::string& kind,
                                            const std::vector<double>& fparams,
                                            const std::vector<int64_t>& iparams,
                                            Engine* eng, int op_id);
+
+// GPU logic factory (gpu_ops.cpp) — kinds: gpu_source/gpu_map/gpu_filter/
+// gpu_reduce/gpu_ffat/gpu_to_host/gpu_count_sink
+std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::string& spec,
+                                        const std::vector<double>& fp,
+                                        const std::vector<int64_t>& ip, Engine* eng,
+                                        int op_id, int device, const Schema& os,
+                                        int64_t out_batch);
 
 }  // namespace wfa

@@ -1,0 +1,586 @@
+// GPU operator logic — host side.  Each GPU operator replica owns a HIP
+// stream on its device and drives the gfx950 kernels in csrc/hip/ over
+// device-resident SoA batches.  Chained GPU stages (map->filter->ffat)
+// share one replica thread and one stream; queue edges between GPU ops
+// synchronize through per-batch hipEvents.
+//
+// MI355X-native replacements (redesigns, not translations):
+//   recycling_gpu.hpp pool      -> DeviceArena (size-bucketed HBM slabs,
+//                                  one allocation per batch, 288 GB budget)
+//   batch_gpu_t.hpp             -> Batch{loc=DEVICE} single-slab layout
+//   map_gpu/filter_gpu/reduce_gpu/ffat_windows_gpu operator classes
+//                               -> Gpu{Map,Filter,Reduce,Ffat}Logic below
+#include <cstring>
+#include <map>
+
+#include "engine.hpp"
+
+#ifdef WFA_WITH_HIP
+#include <hip/hip_runtime.h>
+
+#include "../hip/wfa_kernels.h"
+
+namespace wfa {
+
+#define HIPCHK(x)                                                                  \
+    do {                                                                           \
+        hipError_t err_ = (x);                                                     \
+        if (err_ != hipSuccess)                                                    \
+            throw std::runtime_error(std::string("HIP error: ") +                  \
+                                     hipGetErrorString(err_) + " at " #x);         \
+    } while (0)
+
+// ===== per-device arena allocator =====
+struct DeviceArena {
+    std::mutex mu;
+    std::map<size_t, std::vector<void*>> free_by_size;
+    int device;
+    size_t allocated = 0;
+
+    void* get(size_t bytes) {
+        bytes = (bytes + 255) & ~size_t(255);
+        {
+            std::lock_guard<std::mutex> g(mu);
+            auto it = free_by_size.find(bytes);
+            if (it != free_by_size.end() && !it->second.empty()) {
+                void* p = it->second.back();
+                it->second.pop_back();
+                return p;
+            }
+        }
+        void* p = nullptr;
+        HIPCHK(hipSetDevice(device));
+        HIPCHK(hipMalloc(&p, bytes));
+        std::lock_guard<std::mutex> g(mu);
+        allocated += bytes;
+        return p;
+    }
+    void put(void* p, size_t bytes) {
+        bytes = (bytes + 255) & ~size_t(255);
+        std::lock_guard<std::mutex> g(mu);
+        free_by_size[bytes].push_back(p);
+    }
+};
+
+static DeviceArena g_arena[64];
+
+DeviceArena& arena(int dev) {
+    g_arena[dev].device = dev;
+    return g_arena[dev];
+}
+
+// ===== device batch alloc (core.cpp hooks) =====
+static size_t dev_batch_bytes(const Schema& s, int64_t cap) {
+    size_t b = 16 * cap;  // ts + key
+    for (auto d : s.payload) b += ((dsize(d) * cap + 255) & ~size_t(255));
+    return b;
+}
+
+Batch* gpu_alloc_batch(Pool& pool) {
+    Batch* b = new Batch();
+    b->capacity = pool.capacity;
+    b->schema = pool.schema;
+    b->loc = Loc::DEVICE;
+    b->device = pool.device;
+    size_t total = dev_batch_bytes(pool.schema, pool.capacity);
+    char* base = (char*)arena(pool.device).get(total);
+    b->ts = (int64_t*)base;
+    b->key = (uint64_t*)(base + 8 * pool.capacity);
+    char* p = base + 16 * pool.capacity;
+    b->cols.resize(pool.schema.payload.size());
+    for (size_t i = 0; i < b->cols.size(); ++i) {
+        b->cols[i] = p;
+        p += (dsize(pool.schema.payload[i]) * pool.capacity + 255) & ~size_t(255);
+    }
+    hipEvent_t ev;
+    HIPCHK(hipSetDevice(pool.device));
+    HIPCHK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    b->ready_event = ev;
+    return b;
+}
+
+void gpu_free_batch(Batch* b) {
+    arena(b->device).put(b->ts, dev_batch_bytes(b->schema, b->capacity));
+    if (b->ready_event) (void)hipEventDestroy((hipEvent_t)b->ready_event);
+    delete b;
+}
+
+// ===== base for GPU logics =====
+struct GpuLogicBase : OpLogic {
+    int device = 0;
+    hipStream_t stream = nullptr;
+    std::unique_ptr<Pool> dev_pool;  // device batches this logic emits
+    Schema out_schema;
+    int64_t out_cap = 1 << 20;
+    int64_t* h_count = nullptr;      // pinned readback
+    bool inited = false;
+
+    virtual void init_device() {}
+    void ensure_init() {
+        if (inited) return;
+        HIPCHK(hipSetDevice(device));
+        HIPCHK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+        HIPCHK(hipHostMalloc((void**)&h_count, 64, hipHostMallocDefault));
+        dev_pool = std::make_unique<Pool>(out_schema, out_cap, false);
+        dev_pool->loc = Loc::DEVICE;
+        dev_pool->device = device;
+        init_device();
+        inited = true;
+    }
+    ~GpuLogicBase() override {
+        if (h_count) (void)hipHostFree(h_count);
+        if (stream) (void)hipStreamDestroy(stream);
+    }
+
+    // make the producing stream's work visible to our stream
+    void wait_ready(Batch* b) {
+        if (b->loc == Loc::DEVICE && b->ready_event)
+            HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)b->ready_event, 0));
+    }
+    void record_ready(Batch* b) {
+        b->stream = stream;
+        HIPCHK(hipEventRecord((hipEvent_t)b->ready_event, stream));
+    }
+
+    // H2D staging: host batch -> fresh device batch (reference
+    // forward_emitter_gpu.hpp CPU->GPU path, redesigned: whole-SoA copies)
+    Batch* to_device(Batch* hb, RuntimeCtx& ctx) {
+        Batch* db = dev_pool->get();
+        int64_t n = hb->count;
+        HIPCHK(hipMemcpyAsync(db->ts, hb->ts, 8 * n, hipMemcpyHostToDevice, stream));
+        HIPCHK(hipMemcpyAsync(db->key, hb->key, 8 * n, hipMemcpyHostToDevice, stream));
+        size_t bytes = 16 * n;
+        for (size_t c = 0; c < hb->cols.size() && c < db->cols.size(); ++c) {
+            size_t es = dsize(hb->schema.payload[c]);
+            HIPCHK(hipMemcpyAsync(db->cols[c], hb->cols[c], es * n,
+                                  hipMemcpyHostToDevice, stream));
+            bytes += es * n;
+        }
+        db->count = n;
+        db->watermark = hb->watermark;
+        db->stream_tag = hb->stream_tag;
+        if (ctx.stats) ctx.stats->bytes_h2d += bytes;
+        release(hb);
+        return db;
+    }
+
+    Batch* input_on_device(Batch* b, RuntimeCtx& ctx) {
+        if (b->loc == Loc::HOST) return to_device(b, ctx);
+        wait_ready(b);
+        return b;
+    }
+};
+
+// ===== device source (GPU-resident synthetic generator) =====
+struct GpuSourceLogic : GpuLogicBase {
+    int64_t len, n_keys, bsz;
+    int vdt;
+    uint64_t seed;
+    int64_t pos = 0;
+    GpuSourceLogic(int64_t l, int64_t k, int64_t b, int v, uint64_t sd, int dev,
+                   Schema os) {
+        len = l; n_keys = k; bsz = b; vdt = v; seed = sd;
+        device = dev;
+        out_schema = os;
+        out_cap = b;
+    }
+    bool is_source() const override { return true; }
+    bool source_step(EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        if (pos >= len) return false;
+        Batch* db = dev_pool->get();
+        int64_t n = std::min<int64_t>(bsz, len - pos);
+        wfa_gen_batch(stream, db->ts, db->key, db->cols[0], vdt, n, pos, seed, n_keys);
+        db->count = n;
+        pos += n;
+        db->watermark = pos - 1;
+        record_ready(db);
+        if (ctx.stats) ctx.stats->num_kernels++;
+        out.emit(db);
+        return pos < len;
+    }
+};
+
+// ===== stateless Map_GPU =====
+struct GpuMapLogic : GpuLogicBase {
+    int spec, col;
+    double a, b;
+    GpuMapLogic(int sp, int c, double a_, double b_, int dev, Schema os) {
+        spec = sp; col = c; a = a_; b = b_;
+        device = dev;
+        out_schema = os;
+    }
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        Batch* db = input_on_device(in, ctx);
+        wfa_map_apply(stream, spec, db->cols[col], (int)db->schema.payload[col],
+                      db->count, a, b);
+        record_ready(db);
+        if (ctx.stats) ctx.stats->num_kernels++;
+        out.emit(db);
+    }
+};
+
+// ===== stateless Filter_GPU with on-device compaction =====
+struct GpuFilterLogic : GpuLogicBase {
+    int spec, col;
+    double a, b;
+    uint32_t* d_flags = nullptr;
+    uint32_t* d_scan = nullptr;
+    int64_t* d_cnt = nullptr;
+    void** d_colptrs = nullptr;  // [2*ncols] in,out
+    int* d_esize = nullptr;
+    GpuFilterLogic(int sp, int c, double a_, double b_, int dev, Schema os,
+                   int64_t cap) {
+        spec = sp; col = c; a = a_; b = b_;
+        device = dev;
+        out_schema = os;
+        out_cap = cap;
+    }
+    void init_device() override {
+        d_flags = (uint32_t*)arena(device).get(4 * out_cap);
+        d_scan = (uint32_t*)arena(device).get(4 * (out_cap / 2048 + 2));
+        d_cnt = (int64_t*)arena(device).get(64);
+        size_t nc = out_schema.payload.size();
+        d_colptrs = (void**)arena(device).get(16 * (nc + 1));
+        d_esize = (int*)arena(device).get(4 * (nc + 1));
+        std::vector<int> es;
+        for (auto d : out_schema.payload) es.push_back((int)dsize(d));
+        HIPCHK(hipMemcpy(d_esize, es.data(), 4 * nc, hipMemcpyHostToDevice));
+    }
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        Batch* db = input_on_device(in, ctx);
+        int64_t n = db->count;
+        if (n > out_cap)
+            throw std::runtime_error("filter input batch > out_batch capacity");
+        Batch* ob = dev_pool->get();
+        wfa_filter_flags(stream, spec, db->cols[col], (int)db->schema.payload[col], n,
+                         a, b, d_flags);
+        // column pointer table (in then out)
+        size_t nc = db->cols.size();
+        std::vector<void*> ptrs(2 * nc);
+        for (size_t c = 0; c < nc; ++c) {
+            ptrs[c] = db->cols[c];
+            ptrs[nc + c] = ob->cols[c];
+        }
+        HIPCHK(hipMemcpyAsync(d_colptrs, ptrs.data(), 8 * 2 * nc,
+                              hipMemcpyHostToDevice, stream));
+        wfa_compact(stream, n, d_flags, d_scan, db->ts, ob->ts, db->key, ob->key,
+                    (const void* const*)d_colptrs, (void* const*)(d_colptrs + nc),
+                    d_esize, (int)nc, d_cnt);
+        HIPCHK(hipMemcpyAsync(h_count, d_cnt, 8, hipMemcpyDeviceToHost, stream));
+        HIPCHK(hipStreamSynchronize(stream));
+        ob->count = *h_count;
+        ob->watermark = db->watermark;
+        ob->stream_tag = db->stream_tag;
+        if (ctx.stats) ctx.stats->num_kernels += 3;
+        release(db);
+        if (ob->count) {
+            record_ready(ob);
+            out.emit(ob);
+        } else {
+            int64_t wm = ob->watermark;
+            release(ob);
+            for (auto* e : out.emitters) e->punct(wm);
+        }
+    }
+};
+
+// ===== shared keyed front half: slot -> sort -> gather -> segments =====
+struct KeyedScratch {
+    uint64_t* tkeys = nullptr;
+    uint32_t* tslots = nullptr;
+    uint32_t* d_nslots = nullptr;
+    uint64_t* slot_to_key = nullptr;
+    uint32_t *slot = nullptr, *idx = nullptr, *slot_t = nullptr, *idx_t = nullptr;
+    uint32_t* hist = nullptr;
+    uint32_t *seg_start = nullptr, *seg_slot = nullptr;
+    int64_t* d_nseg = nullptr;
+    float* v_sorted = nullptr;
+    int64_t* ts_sorted = nullptr;
+    float* v_f32 = nullptr;  // cast buffer
+    int64_t table_cap = 0;
+    int64_t max_keys = 0;
+    int64_t cap = 0;
+    int bits = 20;
+
+    void alloc(int dev, int64_t cap_, int64_t mk, hipStream_t s) {
+        cap = cap_;
+        max_keys = mk;
+        table_cap = 1;
+        while (table_cap < 2 * mk) table_cap <<= 1;
+        bits = 1;
+        while ((1ll << bits) < mk + 1) ++bits;
+        auto& A = arena(dev);
+        tkeys = (uint64_t*)A.get(8 * table_cap);
+        tslots = (uint32_t*)A.get(4 * table_cap);
+        d_nslots = (uint32_t*)A.get(64);
+        slot_to_key = (uint64_t*)A.get(8 * mk);
+        slot = (uint32_t*)A.get(4 * cap);
+        idx = (uint32_t*)A.get(4 * cap);
+        slot_t = (uint32_t*)A.get(4 * cap);
+        idx_t = (uint32_t*)A.get(4 * cap);
+        hist = (uint32_t*)A.get(4 * (16 * (cap / 2048 + 2) + 16));
+        seg_start = (uint32_t*)A.get(4 * cap);
+        seg_slot = (uint32_t*)A.get(4 * cap);
+        d_nseg = (int64_t*)A.get(64);
+        v_sorted = (float*)A.get(8 * cap);   // also holds i64 when needed
+        ts_sorted = (int64_t*)A.get(8 * cap);
+        v_f32 = (float*)A.get(4 * cap);
+        wfa_fill_u64(s, tkeys, ~0ULL, table_cap);
+        wfa_fill_u32(s, tslots, ~0u, table_cap);
+        wfa_fill_u32(s, d_nslots, 0, 1);
+    }
+
+    // returns sorted (slot,idx) and fills segments; v column cast to f32
+    void group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx) {
+        int64_t n = db->count;
+        if (n > cap)
+            throw std::runtime_error("batch larger than keyed scratch capacity — "
+                                     "set the GPU op's out_batch >= upstream batch");
+        wfa_key_to_slot(s, db->key, n, tkeys, tslots, d_nslots, table_cap, slot,
+                        slot_to_key);
+        wfa_iota_u32(s, idx, n);
+        uint32_t *os, *oi;
+        wfa_sort_pairs(s, slot, idx, slot_t, idx_t, hist, n, bits, &os, &oi);
+        // gather values (as f32) + ts into segment order
+        int vdt = (int)db->schema.payload[vcol];
+        const void* vsrc = db->cols[vcol];
+        if (vdt != 2) {  // cast to f32 first (i64/bf16 lifted)
+            wfa_cast(s, vsrc, vdt, v_f32, 2, n);
+            vsrc = v_f32;
+        }
+        wfa_gather(s, oi, n, vsrc, v_sorted, 4, db->ts, ts_sorted);
+        wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg);
+        if (ctx.stats) ctx.stats->num_kernels += 6 + 3 * ((bits + 3) / 4);
+    }
+};
+
+// ===== Reduce_GPU: per-batch keyed reduction =====
+struct GpuReduceLogic : GpuLogicBase {
+    int comb, vcol;
+    int64_t max_keys;
+    KeyedScratch ks;
+    uint64_t* d_okey = nullptr;
+    float* d_oval = nullptr;
+    int64_t* d_ots = nullptr;
+    int64_t* d_on = nullptr;
+    GpuReduceLogic(int comb_, int vc, int64_t mk, int dev, Schema os, int64_t cap) {
+        comb = comb_; vcol = vc; max_keys = mk;
+        device = dev;
+        out_schema = os;  // payload [F32]
+        out_cap = cap;
+    }
+    void init_device() override {
+        ks.alloc(device, out_cap, max_keys, stream);
+        auto& A = arena(device);
+        d_okey = (uint64_t*)A.get(8 * out_cap);
+        d_oval = (float*)A.get(4 * out_cap);
+        d_ots = (int64_t*)A.get(8 * out_cap);
+        d_on = (int64_t*)A.get(64);
+    }
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        Batch* db = input_on_device(in, ctx);
+        int64_t n = db->count;
+        ks.group(stream, db, vcol, ctx);
+        Batch* ob = dev_pool->get();
+        wfa_segment_reduce(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
+                           ks.v_sorted, ks.ts_sorted, 2, comb, ks.slot_to_key,
+                           ob->key, ob->cols[0], ob->ts, d_on);
+        HIPCHK(hipMemcpyAsync(h_count, d_on, 8, hipMemcpyDeviceToHost, stream));
+        HIPCHK(hipStreamSynchronize(stream));
+        ob->count = *h_count;
+        ob->watermark = db->watermark;
+        if (ctx.stats) ctx.stats->num_kernels += 1;
+        release(db);
+        record_ready(ob);
+        out.emit(ob);
+    }
+};
+
+// ===== Ffat_Windows_GPU: keyed CB sliding window =====
+struct GpuFfatLogic : GpuLogicBase {
+    int comb, vcol;
+    int64_t win, slide, max_keys;
+    bool use_tree;
+    int64_t pane_len, P, S;
+    int ring_log2;
+    KeyedScratch ks;
+    // state arenas
+    int64_t* st_count = nullptr;
+    uint32_t* st_fill = nullptr;
+    float* st_acc = nullptr;
+    float* ring_or_tree = nullptr;
+    uint32_t* st_head = nullptr;
+    float* st_wsum = nullptr;
+    int64_t* d_on = nullptr;
+
+    GpuFfatLogic(int comb_, int vc, int64_t w, int64_t sl, int64_t mk, bool tree,
+                 int dev, Schema os, int64_t cap) {
+        comb = comb_; vcol = vc; win = w; slide = sl; max_keys = mk; use_tree = tree;
+        device = dev;
+        out_schema = os;  // payload [F32]
+        out_cap = cap;
+        pane_len = std::__gcd(win, slide);
+        P = win / pane_len;
+        S = slide / pane_len;
+        ring_log2 = 1;
+        while ((1ll << ring_log2) < P + 2) ++ring_log2;
+    }
+    void init_device() override {
+        ks.alloc(device, out_cap, max_keys, stream);
+        auto& A = arena(device);
+        int64_t R = 1ll << ring_log2;
+        st_count = (int64_t*)A.get(8 * max_keys);
+        st_fill = (uint32_t*)A.get(4 * max_keys);
+        st_acc = (float*)A.get(4 * max_keys);
+        ring_or_tree = (float*)A.get(4 * max_keys * (use_tree ? 2 * R : R));
+        st_head = (uint32_t*)A.get(4 * max_keys);
+        st_wsum = (float*)A.get(4 * max_keys);
+        d_on = (int64_t*)A.get(64);
+        HIPCHK(hipMemsetAsync(st_count, 0, 8 * max_keys, stream));
+        HIPCHK(hipMemsetAsync(st_fill, 0, 4 * max_keys, stream));
+        HIPCHK(hipMemsetAsync(st_head, 0, 4 * max_keys, stream));
+        HIPCHK(hipMemsetAsync(st_wsum, 0, 4 * max_keys, stream));
+        float ident = comb == 0 ? 0.f : (comb == 1 ? INFINITY : -INFINITY);
+        wfa_fill_f32(stream, st_acc, ident, max_keys);
+        wfa_fill_f32(stream, ring_or_tree, ident,
+                     max_keys * (use_tree ? 2 * (1ll << ring_log2) : (1ll << ring_log2)));
+    }
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        Batch* db = input_on_device(in, ctx);
+        int64_t n = db->count;
+        ks.group(stream, db, vcol, ctx);
+        Batch* ob = dev_pool->get();
+        HIPCHK(hipMemsetAsync(d_on, 0, 8, stream));
+        if (use_tree)
+            wfa_ffat_tree_fold(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
+                               ks.v_sorted, ks.ts_sorted, pane_len, P, S, comb,
+                               ring_log2, st_count, st_fill, st_acc, ring_or_tree,
+                               st_head, ks.slot_to_key, ob->key, (float*)ob->cols[0],
+                               ob->ts, d_on, ob->capacity);
+        else
+            wfa_ffat_cb_fold(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
+                             ks.v_sorted, ks.ts_sorted, pane_len, P, S, comb,
+                             ring_log2, st_count, st_fill, st_acc, ring_or_tree,
+                             st_head, st_wsum, ks.slot_to_key, ob->key,
+                             (float*)ob->cols[0], ob->ts, d_on, ob->capacity);
+        HIPCHK(hipMemcpyAsync(h_count, d_on, 8, hipMemcpyDeviceToHost, stream));
+        HIPCHK(hipStreamSynchronize(stream));
+        ob->count = std::min<int64_t>(*h_count, ob->capacity);
+        ob->watermark = db->watermark;
+        if (ctx.stats) ctx.stats->num_kernels += 1;
+        release(db);
+        if (ob->count) {
+            record_ready(ob);
+            out.emit(ob);
+        } else {
+            int64_t wm = ob->watermark;
+            release(ob);
+            for (auto* e : out.emitters) e->punct(wm);
+        }
+    }
+};
+
+// ===== GPU -> host stage (D2H) =====
+struct GpuToHostLogic : GpuLogicBase {
+    explicit GpuToHostLogic(int dev) { device = dev; }
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        if (in->loc == Loc::HOST) {
+            out.emit(in);
+            return;
+        }
+        ensure_init();
+        wait_ready(in);
+        Batch* hb = out.new_batch();
+        int64_t n = std::min(in->count, hb->capacity);
+        HIPCHK(hipMemcpyAsync(hb->ts, in->ts, 8 * n, hipMemcpyDeviceToHost, stream));
+        HIPCHK(hipMemcpyAsync(hb->key, in->key, 8 * n, hipMemcpyDeviceToHost, stream));
+        size_t bytes = 16 * n;
+        for (size_t c = 0; c < hb->cols.size() && c < in->cols.size(); ++c) {
+            size_t es = dsize(in->schema.payload[c]);
+            HIPCHK(hipMemcpyAsync(hb->cols[c], in->cols[c], es * n,
+                                  hipMemcpyDeviceToHost, stream));
+            bytes += es * n;
+        }
+        HIPCHK(hipStreamSynchronize(stream));
+        hb->count = n;
+        hb->watermark = in->watermark;
+        hb->stream_tag = in->stream_tag;
+        if (ctx.stats) ctx.stats->bytes_d2h += bytes;
+        release(in);
+        out.emit(hb);
+    }
+};
+
+// ===== GPU sink: consume device batches, count tuples (bench path) =====
+struct GpuCountSink : GpuLogicBase {
+    Engine* eng;
+    int op_id;
+    int64_t tuples = 0;
+    GpuCountSink(Engine* e, int id, int dev) : eng(e), op_id(id) { device = dev; }
+    void process(Batch* b, EmitCtx&, RuntimeCtx& ctx) override {
+        ensure_init();
+        wait_ready(b);
+        tuples += b->count;
+        release(b);
+    }
+    void on_eos(EmitCtx&, RuntimeCtx&) override {
+        if (stream) hipStreamSynchronize(stream);
+        eng->sink_tuples[op_id].fetch_add(tuples, std::memory_order_relaxed);
+    }
+};
+
+std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::string& spec,
+                                        const std::vector<double>& fp,
+                                        const std::vector<int64_t>& ip, Engine* eng,
+                                        int op_id, int device, const Schema& os,
+                                        int64_t out_batch) {
+    if (kind == "gpu_source")
+        // ip: [len, n_keys, batch, vdt, seed]
+        return std::make_shared<GpuSourceLogic>(ip[0], ip[1], ip[2], (int)ip[3],
+                                                (uint64_t)ip[4], device, os);
+    if (kind == "gpu_map")
+        // ip: [spec, col]; fp: [a, b]
+        return std::make_shared<GpuMapLogic>((int)ip[0], (int)ip[1], fp[0], fp[1],
+                                             device, os);
+    if (kind == "gpu_filter")
+        return std::make_shared<GpuFilterLogic>((int)ip[0], (int)ip[1], fp[0], fp[1],
+                                                device, os, out_batch);
+    if (kind == "gpu_reduce")
+        // ip: [comb, vcol, max_keys]
+        return std::make_shared<GpuReduceLogic>((int)ip[0], (int)ip[1], ip[2], device,
+                                                os, out_batch);
+    if (kind == "gpu_ffat")
+        // ip: [comb, vcol, win, slide, max_keys, use_tree]
+        return std::make_shared<GpuFfatLogic>((int)ip[0], (int)ip[1], ip[2], ip[3],
+                                              ip[4], ip[5] != 0, device, os, out_batch);
+    if (kind == "gpu_to_host")
+        return std::make_shared<GpuToHostLogic>(device);
+    if (kind == "gpu_count_sink") {
+        eng->sink_acc_i64[op_id].store(0);
+        eng->sink_tuples[op_id].store(0);
+        return std::make_shared<GpuCountSink>(eng, op_id, device);
+    }
+    throw std::runtime_error("unknown gpu logic: " + kind);
+}
+
+}  // namespace wfa
+
+#else  // !WFA_WITH_HIP
+
+namespace wfa {
+Batch* gpu_alloc_batch(Pool&) { throw std::runtime_error("built without HIP"); }
+void gpu_free_batch(Batch*) {}
+std::shared_ptr<OpLogic> make_gpu_logic(const std::string&, const std::string&,
+                                        const std::vector<double>&,
+                                        const std::vector<int64_t>&, Engine*, int, int,
+                                        const Schema&, int64_t) {
+    throw std::runtime_error("built without HIP");
+}
+}  // namespace wfa
+
+#endif

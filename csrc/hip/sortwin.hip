@@ -1,0 +1,513 @@
+// gfx950 keyed-stream machinery: key->dense-slot hash table, stable 4-bit
+// LSD radix sort, segment extraction, segmented reduce, and the batched
+// multi-key sliding-window fold.
+//
+// MI355X-native redesign of the reference's keyed GPU path:
+//   Extract_Dests/Compute_Mapping + thrust sort/unique
+//       (keyby_emitter_gpu.hpp:68-99,518-583)        -> hash table + radix sort
+//   Reduce_GPU thrust sort_by_key/reduce_by_key
+//       (reduce_gpu.hpp:239-269)                     -> wfa_segment_reduce
+//   Ffat_Replica_GPU per-key host loop + per-key FlatFAT_GPU streams
+//       (ffat_replica_gpu.hpp:423-1047, flatfat_gpu.hpp) ->
+//       wfa_ffat_cb_fold / wfa_ffat_tree_fold: ONE kernel advances every
+//       key's window state over the batch's segments (the reference's
+//       per-key host loop is its main scaling flaw — SURVEY.md §7 step 6).
+#include <hip/hip_runtime.h>
+
+#include "wfa_kernels.h"
+
+#define WFA_THREADS 256
+#define WFA_MAX_BLOCKS 2048
+#define RS_IPT 8
+#define RS_PER_BLOCK (WFA_THREADS * RS_IPT)
+
+static inline int64_t nblk(int64_t n, int64_t per_thread = 1) {
+    int64_t b = (n + WFA_THREADS * per_thread - 1) / (WFA_THREADS * per_thread);
+    return b < 1 ? 1 : (b > WFA_MAX_BLOCKS ? WFA_MAX_BLOCKS : b);
+}
+
+__device__ __forceinline__ uint64_t mix64s(uint64_t k) {
+    k += 0x9e3779b97f4a7c15ULL;
+    k = (k ^ (k >> 30)) * 0xbf58476d1ce4e5b9ULL;
+    k = (k ^ (k >> 27)) * 0x94d049bb133111ebULL;
+    return k ^ (k >> 31);
+}
+
+// ===== key -> dense slot (open addressing, device-scope atomics) =====
+#define WFA_EMPTY_KEY (~0ULL)
+
+__global__ void k_key_to_slot(const uint64_t* key, int64_t n, uint64_t* tkeys,
+                              uint32_t* tslots, uint32_t* n_slots, int64_t cap,
+                              uint32_t* slot_out, uint64_t* slot_to_key) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * (int64_t)blockDim.x) {
+        uint64_t k = key[i];
+        uint64_t p = mix64s(k) % (uint64_t)cap;
+        for (;;) {
+            uint64_t cur = __hip_atomic_load(&tkeys[p], __ATOMIC_RELAXED,
+                                             __HIP_MEMORY_SCOPE_AGENT);
+            if (cur == k) break;
+            if (cur == WFA_EMPTY_KEY) {
+                uint64_t expected = WFA_EMPTY_KEY;
+                bool won = __hip_atomic_compare_exchange_strong(
+                    &tkeys[p], &expected, k, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+                    __HIP_MEMORY_SCOPE_AGENT);
+                if (won) {  // we inserted
+                    uint32_t slot = atomicAdd(n_slots, 1u);
+                    slot_to_key[slot] = k;
+                    __hip_atomic_store(&tslots[p], slot, __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+                    break;
+                }
+                if (expected == k) break;  // raced: same key inserted by other
+                // raced: different key took the cell — keep probing
+            }
+            p = (p + 1) % (uint64_t)cap;
+        }
+        // wait for the slot id to be published
+        uint32_t s;
+        do {
+            s = __hip_atomic_load(&tslots[p], __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
+        } while (s == ~0u);
+        slot_out[i] = s;
+    }
+}
+
+extern "C" void wfa_key_to_slot(wfa_stream_t s, const uint64_t* key, int64_t n,
+                                uint64_t* table_keys, uint32_t* table_slots,
+                                uint32_t* n_slots, int64_t table_cap,
+                                uint32_t* slot_out, uint64_t* slot_to_key) {
+    hipLaunchKernelGGL(k_key_to_slot, dim3(nblk(n)), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, key, n, table_keys, table_slots, n_slots,
+                       table_cap, slot_out, slot_to_key);
+}
+
+// ===== stable LSD radix sort, 4-bit digits =====
+extern "C" int64_t wfa_sort_nblocks(int64_t n) {
+    return (n + RS_PER_BLOCK - 1) / RS_PER_BLOCK;
+}
+
+__global__ void k_rs_hist(const uint32_t* keys, int64_t n, int shift, uint32_t* hist,
+                          int64_t nblocks) {
+    __shared__ uint32_t h[16];
+    if (threadIdx.x < 16) h[threadIdx.x] = 0;
+    __syncthreads();
+    int64_t base = (int64_t)blockIdx.x * RS_PER_BLOCK + (int64_t)threadIdx.x * RS_IPT;
+#pragma unroll
+    for (int j = 0; j < RS_IPT; ++j) {
+        int64_t i = base + j;
+        if (i < n) atomicAdd(&h[(keys[i] >> shift) & 15], 1u);
+    }
+    __syncthreads();
+    if (threadIdx.x < 16) hist[(int64_t)threadIdx.x * nblocks + blockIdx.x] = h[threadIdx.x];
+}
+
+__global__ void k_rs_scan(uint32_t* hist, int64_t m) {  // single block, 1024 thr
+    __shared__ uint32_t carry;
+    __shared__ uint32_t buf[1024];
+    if (threadIdx.x == 0) carry = 0;
+    __syncthreads();
+    for (int64_t base = 0; base < m; base += 1024) {
+        int64_t i = base + threadIdx.x;
+        uint32_t v = (i < m) ? hist[i] : 0;
+        buf[threadIdx.x] = v;
+        __syncthreads();
+        for (int off = 1; off < 1024; off <<= 1) {
+            uint32_t t = (threadIdx.x >= off) ? buf[threadIdx.x - off] : 0;
+            __syncthreads();
+            buf[threadIdx.x] += t;
+            __syncthreads();
+        }
+        if (i < m) hist[i] = carry + buf[threadIdx.x] - v;
+        __syncthreads();
+        if (threadIdx.x == 1023) carry += buf[1023];
+        __syncthreads();
+    }
+}
+
+// Stable scatter: per-thread contiguous items; LDS vector-scan of 16-digit
+// count vectors across the block's threads.
+__global__ void k_rs_scatter(const uint32_t* keys, const uint32_t* vals, int64_t n,
+                             int shift, const uint32_t* hist, int64_t nblocks,
+                             uint32_t* keys_out, uint32_t* vals_out) {
+    __shared__ uint32_t tc[WFA_THREADS][17];  // [thread][digit] padded
+    __shared__ uint32_t gbase[16];
+    int64_t base = (int64_t)blockIdx.x * RS_PER_BLOCK + (int64_t)threadIdx.x * RS_IPT;
+    uint32_t k_[RS_IPT];
+    uint32_t v_[RS_IPT];
+    int cnt_local[16];
+#pragma unroll
+    for (int d = 0; d < 16; ++d) cnt_local[d] = 0;
+    int nit = 0;
+#pragma unroll
+    for (int j = 0; j < RS_IPT; ++j) {
+        int64_t i = base + j;
+        if (i < n) {
+            k_[j] = keys[i];
+            v_[j] = vals[i];
+            cnt_local[(k_[j] >> shift) & 15]++;
+            nit = j + 1;
+        }
+    }
+#pragma unroll
+    for (int d = 0; d < 16; ++d) tc[threadIdx.x][d] = cnt_local[d];
+    if (threadIdx.x < 16)
+        gbase[threadIdx.x] = hist[(int64_t)threadIdx.x * nblocks + blockIdx.x];
+    __syncthreads();
+    // Hillis-Steele inclusive scan over threads of the 16-vector
+    for (int off = 1; off < WFA_THREADS; off <<= 1) {
+        uint32_t t[16];
+#pragma unroll
+        for (int d = 0; d < 16; ++d)
+            t[d] = (threadIdx.x >= off) ? tc[threadIdx.x - off][d] : 0;
+        __syncthreads();
+#pragma unroll
+        for (int d = 0; d < 16; ++d) tc[threadIdx.x][d] += t[d];
+        __syncthreads();
+    }
+    // exclusive prefix for this thread = inclusive - own count
+    uint32_t pre[16];
+#pragma unroll
+    for (int d = 0; d < 16; ++d) pre[d] = tc[threadIdx.x][d] - cnt_local[d];
+    // scatter in order; rank among own items recomputed (static indexing)
+    for (int j = 0; j < nit; ++j) {
+        uint32_t d = (k_[j] >> shift) & 15;
+        uint32_t own = 0;
+        for (int j2 = 0; j2 < j; ++j2) own += (((k_[j2] >> shift) & 15) == d);
+        uint32_t pos = gbase[d] + pre[d] + own;
+        keys_out[pos] = k_[j];
+        vals_out[pos] = v_[j];
+    }
+}
+
+extern "C" void wfa_sort_pairs(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
+                               uint32_t* slot_tmp, uint32_t* idx_tmp, uint32_t* hist,
+                               int64_t n, int bits, uint32_t** out_slot,
+                               uint32_t** out_idx) {
+    hipStream_t st = (hipStream_t)s;
+    int64_t nblocks = wfa_sort_nblocks(n);
+    uint32_t *ka = slot, *va = idx, *kb = slot_tmp, *vb = idx_tmp;
+    int passes = (bits + 3) / 4;
+    for (int p = 0; p < passes; ++p) {
+        int shift = 4 * p;
+        hipLaunchKernelGGL(k_rs_hist, dim3(nblocks), dim3(WFA_THREADS), 0, st, ka, n,
+                           shift, hist, nblocks);
+        hipLaunchKernelGGL(k_rs_scan, dim3(1), dim3(1024), 0, st, hist, 16 * nblocks);
+        hipLaunchKernelGGL(k_rs_scatter, dim3(nblocks), dim3(WFA_THREADS), 0, st, ka,
+                           va, n, shift, hist, nblocks, kb, vb);
+        uint32_t* t;
+        t = ka; ka = kb; kb = t;
+        t = va; va = vb; vb = t;
+    }
+    *out_slot = ka;
+    *out_idx = va;
+}
+
+// ===== segment extraction =====
+__global__ void k_seg_count(const uint32_t* slot, int64_t n, uint32_t* blk_cnt) {
+    int64_t base = (int64_t)blockIdx.x * RS_PER_BLOCK + (int64_t)threadIdx.x * RS_IPT;
+    uint32_t c = 0;
+#pragma unroll
+    for (int j = 0; j < RS_IPT; ++j) {
+        int64_t i = base + j;
+        if (i < n) c += (i == 0) || (slot[i] != slot[i - 1]);
+    }
+    __shared__ uint32_t red[WFA_THREADS / 64];
+    for (int off = 32; off; off >>= 1) c += __shfl_down(c, off, 64);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = c;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        uint32_t t = 0;
+        for (int w = 0; w < WFA_THREADS / 64; ++w) t += red[w];
+        blk_cnt[blockIdx.x] = t;
+    }
+}
+
+__global__ void k_seg_scatter(const uint32_t* slot, int64_t n, const uint32_t* blk_off,
+                              uint32_t* seg_start, uint32_t* seg_slot) {
+    int64_t base = (int64_t)blockIdx.x * RS_PER_BLOCK + (int64_t)threadIdx.x * RS_IPT;
+    uint32_t c = 0;
+#pragma unroll
+    for (int j = 0; j < RS_IPT; ++j) {
+        int64_t i = base + j;
+        if (i < n) c += (i == 0) || (slot[i] != slot[i - 1]);
+    }
+    __shared__ uint32_t tc[WFA_THREADS];
+    tc[threadIdx.x] = c;
+    __syncthreads();
+    for (int off = 1; off < WFA_THREADS; off <<= 1) {
+        uint32_t t = (threadIdx.x >= off) ? tc[threadIdx.x - off] : 0;
+        __syncthreads();
+        tc[threadIdx.x] += t;
+        __syncthreads();
+    }
+    uint32_t w = blk_off[blockIdx.x] + tc[threadIdx.x] - c;
+    for (int j = 0; j < RS_IPT; ++j) {
+        int64_t i = base + j;
+        if (i >= n) continue;
+        if ((i == 0) || (slot[i] != slot[i - 1])) {
+            seg_start[w] = (uint32_t)i;
+            seg_slot[w] = slot[i];
+            ++w;
+        }
+    }
+}
+
+__global__ void k_nseg_total(const uint32_t* slot_sorted, int64_t n,
+                             const uint32_t* scan_tmp, int64_t nb, int64_t* d_nseg) {
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        // recount boundaries of the LAST block; add its exclusive base
+        int64_t base = (nb - 1) * (int64_t)RS_PER_BLOCK;
+        uint32_t c = 0;
+        for (int64_t i = base; i < n; ++i)
+            c += (i == 0) || (slot_sorted[i] != slot_sorted[i - 1]);
+        *d_nseg = (int64_t)scan_tmp[nb - 1] + c;
+    }
+}
+
+extern "C" void wfa_segments(wfa_stream_t s, const uint32_t* slot_sorted, int64_t n,
+                             uint32_t* scan_tmp, uint32_t* seg_start,
+                             uint32_t* seg_slot, int64_t* d_nseg) {
+    hipStream_t st = (hipStream_t)s;
+    int64_t nb = (n + RS_PER_BLOCK - 1) / RS_PER_BLOCK;
+    hipLaunchKernelGGL(k_seg_count, dim3(nb), dim3(WFA_THREADS), 0, st, slot_sorted, n,
+                       scan_tmp);
+    hipLaunchKernelGGL(k_rs_scan, dim3(1), dim3(1024), 0, st, scan_tmp, nb);
+    hipLaunchKernelGGL(k_seg_scatter, dim3(nb), dim3(WFA_THREADS), 0, st, slot_sorted,
+                       n, scan_tmp, seg_start, seg_slot);
+    hipLaunchKernelGGL(k_nseg_total, dim3(1), dim3(1), 0, st, slot_sorted, n, scan_tmp,
+                       nb, d_nseg);
+}
+
+// ===== segmented reduce (Reduce_GPU per-batch semantics) =====
+__global__ void k_seg_reduce(const uint32_t* seg_start, const uint32_t* seg_slot,
+                             const int64_t* d_nseg, int64_t n, const void* v_sorted,
+                             const int64_t* ts_sorted, int vdt, int comb,
+                             const uint64_t* slot_to_key, uint64_t* out_key,
+                             void* out_val, int64_t* out_ts, int64_t* d_out_n) {
+    int64_t nseg = *d_nseg;
+    for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nseg;
+         j += gridDim.x * (int64_t)blockDim.x) {
+        int64_t b = seg_start[j];
+        int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        int64_t tmax = ts_sorted ? ts_sorted[b] : 0;
+        if (vdt == 0) {
+            const int64_t* v = (const int64_t*)v_sorted;
+            int64_t acc = (comb == 3) ? 0 : v[b];
+            if (comb == 3) acc = e - b;
+            else
+                for (int64_t i = b + 1; i < e; ++i) {
+                    int64_t x = v[i];
+                    acc = (comb == 0) ? acc + x : (comb == 1 ? min(acc, x) : max(acc, x));
+                }
+            ((int64_t*)out_val)[j] = acc;
+        } else {
+            const float* v = (const float*)v_sorted;
+            float acc = (comb == 3) ? (float)(e - b) : v[b];
+            if (comb != 3)
+                for (int64_t i = b + 1; i < e; ++i) {
+                    float x = v[i];
+                    acc = (comb == 0) ? acc + x : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
+                }
+            ((float*)out_val)[j] = acc;
+        }
+        if (ts_sorted)
+            for (int64_t i = b + 1; i < e; ++i) tmax = max(tmax, ts_sorted[i]);
+        out_key[j] = slot_to_key ? slot_to_key[seg_slot[j]] : (uint64_t)seg_slot[j];
+        if (out_ts) out_ts[j] = tmax;
+    }
+    if (blockIdx.x == 0 && threadIdx.x == 0) *d_out_n = nseg;
+}
+
+extern "C" void wfa_segment_reduce(wfa_stream_t s, const uint32_t* seg_start,
+                                   const uint32_t* seg_slot, const int64_t* d_nseg,
+                                   int64_t n, const void* v_sorted,
+                                   const int64_t* ts_sorted, int vdt, int comb,
+                                   const uint64_t* slot_to_key, uint64_t* out_key,
+                                   void* out_val, int64_t* out_ts, int64_t* d_out_n) {
+    hipLaunchKernelGGL(k_seg_reduce, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_sorted,
+                       ts_sorted, vdt, comb, slot_to_key, out_key, out_val, out_ts,
+                       d_out_n);
+}
+
+// ===== batched multi-key CB sliding-window fold (pane ring) =====
+// One thread advances one key's window state over its segment.  Sum uses a
+// running window total (invertible); min/max recombine the P-pane ring on
+// fire.  All keys of the batch progress in ONE kernel launch — the
+// reference's per-key host loop + per-key stream (ffat_replica_gpu.hpp:
+// 829-867) becomes a dense device-side state machine.
+__global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
+                          const int64_t* d_nseg, int64_t n, const float* v_sorted,
+                          const int64_t* ts_sorted, int64_t pane_len, int64_t P,
+                          int64_t S, int comb, int ring_log2, int64_t* st_count,
+                          uint32_t* st_fill, float* st_acc, float* ring,
+                          uint32_t* st_head, float* st_wsum,
+                          const uint64_t* slot_to_key, uint64_t* out_key,
+                          float* out_val, int64_t* out_ts, int64_t* d_out_n,
+                          int64_t out_cap) {
+    const int64_t nseg = *d_nseg;
+    const uint32_t R = 1u << ring_log2;
+    const uint32_t Rm = R - 1;
+    const float ident = (comb == 0) ? 0.f : (comb == 1 ? INFINITY : -INFINITY);
+    for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nseg;
+         j += gridDim.x * (int64_t)blockDim.x) {
+        const uint32_t slot = seg_slot[j];
+        int64_t i = seg_start[j];
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        uint32_t fill = st_fill[slot];
+        float acc = st_acc[slot];
+        uint32_t head = st_head[slot];
+        float wsum = st_wsum[slot];
+        float* rg = ring + (size_t)slot * R;
+        for (; i < e; ++i) {
+            float x = v_sorted[i];
+            acc = (comb == 0) ? acc + x : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
+            if (++fill == (uint32_t)pane_len) {
+                // pane complete
+                if (comb == 0) {
+                    wsum += acc;
+                    if (head >= (uint32_t)P) wsum -= rg[(head - (uint32_t)P) & Rm];
+                }
+                rg[head & Rm] = acc;
+                acc = ident;
+                fill = 0;
+                ++head;
+                if (head >= (uint32_t)P && ((head - (uint32_t)P) % (uint32_t)S) == 0) {
+                    float res;
+                    if (comb == 0) {
+                        res = wsum;
+                    } else {
+                        res = rg[(head - 1) & Rm];
+                        for (uint32_t q = 2; q <= (uint32_t)P; ++q) {
+                            float pv = rg[(head - q) & Rm];
+                            res = (comb == 1) ? fminf(res, pv) : fmaxf(res, pv);
+                        }
+                    }
+                    int64_t pos = atomicAdd((unsigned long long*)d_out_n, 1ull);
+                    if (pos < out_cap) {
+                        out_key[pos] = slot_to_key[slot];
+                        out_val[pos] = res;
+                        out_ts[pos] = ts_sorted ? ts_sorted[i] : 0;
+                    }
+                }
+            }
+        }
+        st_fill[slot] = fill;
+        st_acc[slot] = acc;
+        st_head[slot] = head;
+        st_wsum[slot] = wsum;
+        st_count[slot] += e - seg_start[j];
+    }
+}
+
+extern "C" void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
+                                 const uint32_t* seg_slot, const int64_t* d_nseg,
+                                 int64_t n, const float* v_sorted,
+                                 const int64_t* ts_sorted, int64_t pane_len, int64_t P,
+                                 int64_t S, int comb, int ring_log2, int64_t* st_count,
+                                 uint32_t* st_fill, float* st_acc, float* ring,
+                                 uint32_t* st_head, float* st_wsum,
+                                 const uint64_t* slot_to_key, uint64_t* out_key,
+                                 float* out_val, int64_t* out_ts, int64_t* d_out_n,
+                                 int64_t out_cap) {
+    hipLaunchKernelGGL(k_ffat_cb, dim3(WFA_MAX_BLOCKS / 4), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_sorted,
+                       ts_sorted, pane_len, P, S, comb, ring_log2, st_count, st_fill,
+                       st_acc, ring, st_head, st_wsum, slot_to_key, out_key, out_val,
+                       out_ts, d_out_n, out_cap);
+}
+
+// ===== FlatFAT-tree fold: O(log R) window query for large P =====
+// Per-slot complete binary tree over R = 2^ring_log2 circular pane leaves
+// (tree[slot*2R + node], node 1 = root, leaves at R..2R-1).  On each pane:
+// write leaf (head & Rm), bubble up log R parents; on fire: circular range
+// query of the last P leaves via the classic two-pointer FlatFAT walk
+// (Tangwongsan VLDB'15 — reference flatfat.hpp:311-337 getResult).
+__global__ void k_ffat_tree(const uint32_t* seg_start, const uint32_t* seg_slot,
+                            const int64_t* d_nseg, int64_t n, const float* v_sorted,
+                            const int64_t* ts_sorted, int64_t pane_len, int64_t P,
+                            int64_t S, int comb, int ring_log2, int64_t* st_count,
+                            uint32_t* st_fill, float* st_acc, float* tree,
+                            uint32_t* st_head, const uint64_t* slot_to_key,
+                            uint64_t* out_key, float* out_val, int64_t* out_ts,
+                            int64_t* d_out_n, int64_t out_cap) {
+    const int64_t nseg = *d_nseg;
+    const uint32_t R = 1u << ring_log2;
+    const uint32_t Rm = R - 1;
+    const float ident = (comb == 0) ? 0.f : (comb == 1 ? INFINITY : -INFINITY);
+#define TCOMB(a, b) ((comb == 0) ? (a) + (b) : (comb == 1 ? fminf(a, b) : fmaxf(a, b)))
+    for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nseg;
+         j += gridDim.x * (int64_t)blockDim.x) {
+        const uint32_t slot = seg_slot[j];
+        int64_t i = seg_start[j];
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        uint32_t fill = st_fill[slot];
+        float acc = st_acc[slot];
+        uint32_t head = st_head[slot];
+        float* tr = tree + (size_t)slot * 2 * R;
+        for (; i < e; ++i) {
+            float x = v_sorted[i];
+            acc = TCOMB(acc, x);
+            if (++fill == (uint32_t)pane_len) {
+                uint32_t leaf = (head & Rm) + R;
+                tr[leaf] = acc;
+                for (uint32_t nd = leaf >> 1; nd >= 1; nd >>= 1)
+                    tr[nd] = TCOMB(tr[2 * nd], tr[2 * nd + 1]);
+                acc = ident;
+                fill = 0;
+                ++head;
+                if (head >= (uint32_t)P && ((head - (uint32_t)P) % (uint32_t)S) == 0) {
+                    // window = circular leaves [head-P, head)
+                    uint32_t lo = (head - (uint32_t)P) & Rm;
+                    uint32_t hi = (head - 1) & Rm;  // inclusive
+                    float res = ident;
+                    // classic FlatFAT inclusive range query [l, r] on one
+                    // contiguous range; split circular into <=2 ranges
+                    auto range_q = [&](uint32_t l, uint32_t r) {
+                        uint32_t a = l + R, b2 = r + R;
+                        float left = ident, right = ident;
+                        while (a <= b2) {
+                            if (a & 1) left = TCOMB(left, tr[a++]);
+                            if (!(b2 & 1)) right = TCOMB(tr[b2--], right);
+                            if (a > b2) break;
+                            a >>= 1;
+                            b2 >>= 1;
+                        }
+                        return TCOMB(left, right);
+                    };
+                    if (lo <= hi)
+                        res = range_q(lo, hi);
+                    else
+                        res = TCOMB(range_q(lo, Rm), range_q(0, hi));
+                    int64_t pos = atomicAdd((unsigned long long*)d_out_n, 1ull);
+                    if (pos < out_cap) {
+                        out_key[pos] = slot_to_key[slot];
+                        out_val[pos] = res;
+                        out_ts[pos] = ts_sorted ? ts_sorted[i] : 0;
+                    }
+                }
+            }
+        }
+        st_fill[slot] = fill;
+        st_acc[slot] = acc;
+        st_head[slot] = head;
+        st_count[slot] += e - seg_start[j];
+    }
+#undef TCOMB
+}
+
+extern "C" void wfa_ffat_tree_fold(
+    wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
+    const int64_t* d_nseg, int64_t n, const float* v_sorted, const int64_t* ts_sorted,
+    int64_t pane_len, int64_t P, int64_t S, int comb, int ring_log2, int64_t* st_count,
+    uint32_t* st_fill, float* st_acc, float* tree, uint32_t* st_head,
+    const uint64_t* slot_to_key, uint64_t* out_key, float* out_val, int64_t* out_ts,
+    int64_t* d_out_n, int64_t out_cap) {
+    hipLaunchKernelGGL(k_ffat_tree, dim3(WFA_MAX_BLOCKS / 4), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_sorted,
+                       ts_sorted, pane_len, P, S, comb, ring_log2, st_count, st_fill,
+                       st_acc, tree, st_head, slot_to_key, out_key, out_val, out_ts,
+                       d_out_n, out_cap);
+}

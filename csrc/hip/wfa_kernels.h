@@ -1,0 +1,122 @@
+// C ABI between the host engine (gpu_ops.cpp, compiled by g++) and the
+// gfx950 device code (kernels.hip / sortwin.hip, compiled by hipcc).
+#pragma once
+#include <cstdint>
+
+// opaque stream handle (hipStream_t)
+typedef void* wfa_stream_t;
+
+extern "C" {
+
+// ----- elementwise / generator -----
+// device-resident synthetic source: ts=start+i, key=mix(i)%n_keys,
+// value by dtype: I64 -> mix%1000, F32 -> uniform[0,1), U16(bf16) -> same
+void wfa_gen_batch(wfa_stream_t s, int64_t* ts, uint64_t* key, void* val, int vdt,
+                   int64_t n, int64_t start, uint64_t seed, uint64_t n_keys);
+
+// map functor catalog (spec ids; see MapSpec in kernels.hip)
+//  1 affine_i64  2 affine_f32  3 affine_bf16  4 square_i64  5 exp_decay_f32
+void wfa_map_apply(wfa_stream_t s, int spec, void* col, int dt, int64_t n,
+                   double a, double b);
+
+// filter predicate -> flags (1 keep)
+//  1 mod_ne_i64(m=a,c=b)  2 gt_f32(thr=a)  3 ge_bf16(thr=a)
+void wfa_filter_flags(wfa_stream_t s, int spec, const void* col, int dt, int64_t n,
+                      double a, double b, uint32_t* flags);
+
+// stream compaction: scatter rows with flag==1 preserving order.
+// cols_in/out: arrays of column base pointers (device-visible), esize bytes each.
+// n_cols counts payload cols; ts/key handled as two extra i64/u64 columns.
+// d_count receives the kept count (device int64).
+void wfa_compact(wfa_stream_t s, int64_t n, const uint32_t* flags,
+                 uint32_t* scan_tmp,  // >= (nblocks+1) u32
+                 const int64_t* ts_in, int64_t* ts_out,
+                 const uint64_t* key_in, uint64_t* key_out,
+                 const void* const* cols_in, void* const* cols_out,
+                 const int* col_esize, int n_cols, int64_t* d_count);
+
+// ----- key -> dense slot hash table (open addressing, u64 keys) -----
+// table_keys: u64[table_cap] init to EMPTY(~0); table_slots: u32[table_cap];
+// n_slots: device counter of allocated slots.  slot_out[i] = dense id of key[i].
+void wfa_key_to_slot(wfa_stream_t s, const uint64_t* key, int64_t n,
+                     uint64_t* table_keys, uint32_t* table_slots,
+                     uint32_t* n_slots, int64_t table_cap, uint32_t* slot_out,
+                     uint64_t* slot_to_key);
+
+// ----- stable LSD radix sort of (slot, iota idx) pairs, 4-bit digits -----
+// bits: how many low bits of slot to sort on. tmp arrays sized n (u32 each).
+// hist sized >= 16 * nblocks(n) + 16.  After return: slot_sorted, idx_sorted.
+void wfa_sort_pairs(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
+                    uint32_t* slot_tmp, uint32_t* idx_tmp, uint32_t* hist,
+                    int64_t n, int bits,
+                    uint32_t** out_slot, uint32_t** out_idx);
+int64_t wfa_sort_nblocks(int64_t n);
+
+// gather rows by permutation idx (values f32/i64 + ts)
+void wfa_gather(wfa_stream_t s, const uint32_t* idx, int64_t n,
+                const void* v_in, void* v_out, int esize,
+                const int64_t* ts_in, int64_t* ts_out);
+
+// segment boundaries of sorted slot array: seg_start[j] = first index of
+// segment j, seg_slot[j] = its slot; d_nseg = #segments.
+void wfa_segments(wfa_stream_t s, const uint32_t* slot_sorted, int64_t n,
+                  uint32_t* scan_tmp, uint32_t* seg_start, uint32_t* seg_slot,
+                  int64_t* d_nseg);
+
+// ----- per-batch keyed reduction (Reduce_GPU semantics) -----
+// comb: 0 sum 1 min 2 max 3 count ; vdt: dtype of v (F32 or I64 accum f64/i64)
+void wfa_segment_reduce(wfa_stream_t s, const uint32_t* seg_start,
+                        const uint32_t* seg_slot, const int64_t* d_nseg, int64_t n,
+                        const void* v_sorted, const int64_t* ts_sorted, int vdt,
+                        int comb, const uint64_t* slot_to_key,
+                        uint64_t* out_key, void* out_val, int64_t* out_ts,
+                        int64_t* d_out_n);
+
+// ----- FFAT/pane sliding-window state machine (CB) -----
+// Batched multi-key redesign of the reference's per-key FlatFAT_GPU
+// (SURVEY.md §7 step 6): all keys' folds advance in ONE kernel over the
+// batch's segments; per-slot state lives in dense arenas.
+// State arrays (slot-indexed, cap = max_keys):
+//   st_count i64, st_fill u32, st_acc f32, ring f32[cap*ring_sz], st_head u32
+// Window: win = P panes of pane_len tuples, fires every S panes.
+// comb: 0 sum(invertible running total) 1 min 2 max  (min/max combine ring)
+// Output appended to out_* at atomic cursor d_out_n.
+void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
+                      const uint32_t* seg_slot, const int64_t* d_nseg, int64_t n,
+                      const float* v_sorted, const int64_t* ts_sorted,
+                      int64_t pane_len, int64_t P, int64_t S, int comb,
+                      int ring_log2,
+                      int64_t* st_count, uint32_t* st_fill, float* st_acc,
+                      float* ring, uint32_t* st_head, float* st_wsum,
+                      const uint64_t* slot_to_key,
+                      uint64_t* out_key, float* out_val, int64_t* out_ts,
+                      int64_t* d_out_n, int64_t out_cap);
+
+// ----- FlatFAT arena path (non-invertible combines over many panes) -----
+// Per-slot complete binary tree over ring of 2^ring_log2 pane leaves,
+// stored slot-major: tree[slot * 2*R + node].  Incremental leaf update +
+// O(log R) range query, one thread per segment, batched over all keys.
+void wfa_ffat_tree_fold(wfa_stream_t s, const uint32_t* seg_start,
+                        const uint32_t* seg_slot, const int64_t* d_nseg, int64_t n,
+                        const float* v_sorted, const int64_t* ts_sorted,
+                        int64_t pane_len, int64_t P, int64_t S, int comb,
+                        int ring_log2,
+                        int64_t* st_count, uint32_t* st_fill, float* st_acc,
+                        float* tree, uint32_t* st_head,
+                        const uint64_t* slot_to_key,
+                        uint64_t* out_key, float* out_val, int64_t* out_ts,
+                        int64_t* d_out_n, int64_t out_cap);
+
+// ----- misc -----
+void wfa_fill_u64(wfa_stream_t s, uint64_t* p, uint64_t v, int64_t n);
+void wfa_fill_u32(wfa_stream_t s, uint32_t* p, uint32_t v, int64_t n);
+void wfa_fill_f32(wfa_stream_t s, float* p, float v, int64_t n);
+void wfa_iota_u32(wfa_stream_t s, uint32_t* p, int64_t n);
+void wfa_cast(wfa_stream_t s, const void* in, int dt_in, void* out, int dt_out,
+              int64_t n);
+
+// keyby bucketing for RCCL all-to-all: dest = mix(key) % world (stable order)
+void wfa_bucket_by_key(wfa_stream_t s, const uint64_t* key, int64_t n, int world,
+                       uint32_t* dest_out);
+
+}  // extern "C"

@@ -1,0 +1,54 @@
+"""GPU native logic specs — map onto csrc/engine/gpu_ops.cpp factories and
+the gfx950 kernels in csrc/hip/.  comb ids: 0 sum, 1 min, 2 max, 3 count.
+Value dtypes: 0 I64, 2 F32, 5 U16(bf16)."""
+from .operators import NativeLogic
+
+COMB_SUM, COMB_MIN, COMB_MAX, COMB_COUNT = 0, 1, 2, 3
+
+
+def gpu_source(stream_len, n_keys=1024, batch=1 << 20, vdt=5, seed=42):
+    """Device-resident synthetic source (ts, key, value[vdt])."""
+    return NativeLogic("gpu_source", "", [], [stream_len, n_keys, batch, vdt, seed])
+
+
+def gpu_affine_map(col=0, a=1.0, b=0.0, dtype=5):
+    """x = a*x + b on device; dtype picks the kernel (i64/f32/bf16)."""
+    spec = {0: 1, 2: 2, 5: 3}[dtype]
+    return NativeLogic("gpu_map", "", [float(a), float(b)], [spec, col])
+
+
+def gpu_square_map(col=0):
+    return NativeLogic("gpu_map", "", [0.0, 0.0], [4, col])
+
+
+def gpu_mod_filter(col=0, m=3, c=0):
+    """keep x % m != c (i64)."""
+    return NativeLogic("gpu_filter", "", [float(m), float(c)], [1, col])
+
+
+def gpu_gt_filter(col=0, thr=0.5, dtype=2):
+    """keep x > thr (f32) / x >= thr (bf16)."""
+    spec = 2 if dtype == 2 else 3
+    return NativeLogic("gpu_filter", "", [float(thr), 0.0], [spec, col])
+
+
+def gpu_keyed_reduce(comb=COMB_SUM, col=0, max_keys=1 << 16):
+    """per-batch keyed reduction -> one (key, agg, ts_max) per distinct key."""
+    return NativeLogic("gpu_reduce", "", [], [comb, col, max_keys])
+
+
+def gpu_ffat_windows(comb=COMB_SUM, col=0, win=1000, slide=100,
+                     max_keys=1 << 16, use_tree=False):
+    """Keyed CB sliding window over panes (pane = gcd(win, slide));
+    use_tree selects the FlatFAT-arena path (O(log) combine for large
+    win/slide ratios and non-invertible combines)."""
+    return NativeLogic("gpu_ffat", "", [],
+                       [comb, col, win, slide, max_keys, 1 if use_tree else 0])
+
+
+def gpu_count_sink():
+    return NativeLogic("gpu_count_sink", "", [], [])
+
+
+def gpu_to_host():
+    return NativeLogic("gpu_to_host", "", [], [])
