@@ -69,7 +69,7 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=4_000_000)
-    ap.add_argument("--keys", type=int, default=1 << 16,
+    ap.add_argument("--keys", type=int, default=8192,
                     help="distinct keys per rank")
     ap.add_argument("--win", type=int, default=1000)
     ap.add_argument("--slide", type=int, default=100)
