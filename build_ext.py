@@ -92,10 +92,24 @@ def build(force=False):
     out = os.path.join(ROOT, "windflow_amd", "_core.so")
     if newer(out, objs) or force:
         print("  LINK", out, flush=True)
-        link = ["g++", "-shared", "-o", out] + objs + [
-            f"-L{ROCM}/lib", "-lamdhip64", "-pthread",
-            f"-Wl,-rpath,{ROCM}/lib",
-        ]
+        # PyTorch-ROCm bundles its own libamdhip64.so (DT_NEEDED without the
+        # .so.7 suffix).  Linking ours against /opt/rocm's .so.7 loads TWO
+        # HIP runtimes when torch is imported in the same process — which
+        # crashes.  Link against torch's copy so exactly one runtime exists.
+        torchlib = None
+        try:
+            import torch
+            cand = os.path.join(os.path.dirname(torch.__file__), "lib")
+            if os.path.exists(os.path.join(cand, "libamdhip64.so")):
+                torchlib = cand
+        except Exception:
+            pass
+        if torchlib:
+            hip_link = [f"-L{torchlib}", "-l:libamdhip64.so",
+                        f"-Wl,-rpath,{torchlib}"]
+        else:
+            hip_link = [f"-L{ROCM}/lib", "-lamdhip64", f"-Wl,-rpath,{ROCM}/lib"]
+        link = ["g++", "-shared", "-o", out] + objs + hip_link + ["-pthread"]
         subprocess.check_call(link)
     return out
 

@@ -6,6 +6,16 @@ for AMD MI355X: a native C++ runtime (pinned threads + SPSC queues), SoA
 micro-batches as the universal message unit, hand-written HIP/CDNA4 kernels
 for the GPU operators, and RCCL-over-xGMI for multi-GPU shuffles.
 """
+# PyTorch-ROCm bundles its own HIP runtime (libamdhip64.so, SONAME .so.7).
+# If our engine loads /opt/rocm's copy first, a later `import torch` loads a
+# SECOND runtime into the process and the GPU context corrupts.  Importing
+# torch first makes every later libamdhip64.so.7 request (ours included)
+# resolve to the one already-loaded runtime.
+try:  # pragma: no cover
+    import torch  # noqa: F401
+except ImportError:
+    pass
+
 from .basic import (ExecutionMode, TimePolicy, WinType, JoinMode, RoutingMode, DType)  # noqa: F401
 
 __version__ = "0.1.0"
