@@ -77,6 +77,10 @@ struct Batch {
 
     int64_t* ts = nullptr;      // [capacity]
     uint64_t* key = nullptr;    // [capacity]
+    // Deferred count: device producers may emit before the row count is
+    // known on host (count == -1).  The count lands in *lazy_count (pinned)
+    // once ready_event fires; resolve with gpu_resolve_count().
+    int64_t* lazy_count = nullptr;
     std::vector<void*> cols;    // payload columns per schema
     Schema schema;
     bool pinned = false;
@@ -118,6 +122,7 @@ void release(Batch* b);
 // the per-GPU arena allocator — the recycling_gpu redesign for 288 GB HBM.
 Batch* gpu_alloc_batch(Pool& pool);
 void gpu_free_batch(Batch* b);
+void gpu_resolve_count(Batch* b);  // blocks on ready_event if count == -1
 
 // Deep-copy b (same pool class) — used for copy-on-write under broadcast.
 Batch* clone(Batch* b, Pool& pool);

@@ -320,7 +320,7 @@ void Replica::run() {
             if (!b) break;
             int64_t t0 = now_us();
             stats.inputs_received++;
-            stats.tuples_received += b->count;
+            if (b->count > 0) stats.tuples_received += b->count;
             rctx.current_wm = b->watermark;
             if (b->count) rctx.current_ts = b->ts[b->count - 1];
             if (b->punct) {

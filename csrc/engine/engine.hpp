@@ -72,7 +72,7 @@ struct Emitter {
     void account(Batch* b) {
         if (!stats) return;
         stats->outputs_sent++;
-        stats->tuples_sent += b->count;
+        if (b->count > 0) stats->tuples_sent += b->count;
     }
 };
 

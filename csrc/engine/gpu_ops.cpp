@@ -11,6 +11,7 @@
 //   map_gpu/filter_gpu/reduce_gpu/ffat_windows_gpu operator classes
 //                               -> Gpu{Map,Filter,Reduce,Ffat}Logic below
 #include <cstring>
+#include <deque>
 #include <map>
 
 #include "engine.hpp"
@@ -96,12 +97,20 @@ Batch* gpu_alloc_batch(Pool& pool) {
     HIPCHK(hipSetDevice(pool.device));
     HIPCHK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
     b->ready_event = ev;
+    HIPCHK(hipHostMalloc((void**)&b->lazy_count, 8, hipHostMallocDefault));
     return b;
+}
+
+void gpu_resolve_count(Batch* b) {
+    if (b->count >= 0) return;
+    HIPCHK(hipEventSynchronize((hipEvent_t)b->ready_event));
+    b->count = std::min<int64_t>(*b->lazy_count, b->capacity);
 }
 
 void gpu_free_batch(Batch* b) {
     arena(b->device).put(b->ts, dev_batch_bytes(b->schema, b->capacity));
     if (b->ready_event) (void)hipEventDestroy((hipEvent_t)b->ready_event);
+    if (b->lazy_count) (void)hipHostFree(b->lazy_count);
     delete b;
 }
 
@@ -166,6 +175,7 @@ struct GpuLogicBase : OpLogic {
 
     Batch* input_on_device(Batch* b, RuntimeCtx& ctx) {
         if (b->loc == Loc::HOST) return to_device(b, ctx);
+        gpu_resolve_count(b);
         wait_ready(b);
         return b;
     }
@@ -269,21 +279,14 @@ struct GpuFilterLogic : GpuLogicBase {
         wfa_compact(stream, n, d_flags, d_scan, db->ts, ob->ts, db->key, ob->key,
                     (const void* const*)d_colptrs, (void* const*)(d_colptrs + nc),
                     d_esize, (int)nc, d_cnt);
-        HIPCHK(hipMemcpyAsync(h_count, d_cnt, 8, hipMemcpyDeviceToHost, stream));
-        HIPCHK(hipStreamSynchronize(stream));
-        ob->count = *h_count;
+        HIPCHK(hipMemcpyAsync(ob->lazy_count, d_cnt, 8, hipMemcpyDeviceToHost, stream));
+        ob->count = -1;
         ob->watermark = db->watermark;
         ob->stream_tag = db->stream_tag;
         if (ctx.stats) ctx.stats->num_kernels += 3;
         release(db);
-        if (ob->count) {
-            record_ready(ob);
-            out.emit(ob);
-        } else {
-            int64_t wm = ob->watermark;
-            release(ob);
-            for (auto* e : out.emitters) e->punct(wm);
-        }
+        record_ready(ob);
+        out.emit(ob);
     }
 };
 
@@ -389,9 +392,8 @@ struct GpuReduceLogic : GpuLogicBase {
         wfa_segment_reduce(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                            ks.v_sorted, ks.ts_sorted, 2, comb, ks.slot_to_key,
                            ob->key, ob->cols[0], ob->ts, d_on);
-        HIPCHK(hipMemcpyAsync(h_count, d_on, 8, hipMemcpyDeviceToHost, stream));
-        HIPCHK(hipStreamSynchronize(stream));
-        ob->count = *h_count;
+        HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
+        ob->count = -1;
         ob->watermark = db->watermark;
         if (ctx.stats) ctx.stats->num_kernels += 1;
         release(db);
@@ -468,20 +470,13 @@ struct GpuFfatLogic : GpuLogicBase {
                              ring_log2, st_count, st_fill, st_acc, ring_or_tree,
                              st_head, st_wsum, ks.slot_to_key, ob->key,
                              (float*)ob->cols[0], ob->ts, d_on, ob->capacity);
-        HIPCHK(hipMemcpyAsync(h_count, d_on, 8, hipMemcpyDeviceToHost, stream));
-        HIPCHK(hipStreamSynchronize(stream));
-        ob->count = std::min<int64_t>(*h_count, ob->capacity);
+        HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
+        ob->count = -1;  // resolved by the consumer via gpu_resolve_count
         ob->watermark = db->watermark;
         if (ctx.stats) ctx.stats->num_kernels += 1;
         release(db);
-        if (ob->count) {
-            record_ready(ob);
-            out.emit(ob);
-        } else {
-            int64_t wm = ob->watermark;
-            release(ob);
-            for (auto* e : out.emitters) e->punct(wm);
-        }
+        record_ready(ob);
+        out.emit(ob);
     }
 };
 
@@ -494,6 +489,7 @@ struct GpuToHostLogic : GpuLogicBase {
             return;
         }
         ensure_init();
+        gpu_resolve_count(in);
         wait_ready(in);
         Batch* hb = out.new_batch();
         int64_t n = std::min(in->count, hb->capacity);
@@ -521,15 +517,22 @@ struct GpuCountSink : GpuLogicBase {
     Engine* eng;
     int op_id;
     int64_t tuples = 0;
+    std::deque<Batch*> pending;
     GpuCountSink(Engine* e, int id, int dev) : eng(e), op_id(id) { device = dev; }
-    void process(Batch* b, EmitCtx&, RuntimeCtx& ctx) override {
-        ensure_init();
-        wait_ready(b);
+    void drain_one() {
+        Batch* b = pending.front();
+        pending.pop_front();
+        gpu_resolve_count(b);
         tuples += b->count;
         release(b);
     }
+    void process(Batch* b, EmitCtx&, RuntimeCtx& ctx) override {
+        pending.push_back(b);
+        while (pending.size() > 4) drain_one();
+    }
     void on_eos(EmitCtx&, RuntimeCtx&) override {
-        if (stream) hipStreamSynchronize(stream);
+        while (!pending.empty()) drain_one();
+        if (stream) (void)hipStreamSynchronize(stream);
         eng->sink_tuples[op_id].fetch_add(tuples, std::memory_order_relaxed);
     }
 };
@@ -575,6 +578,7 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
 namespace wfa {
 Batch* gpu_alloc_batch(Pool&) { throw std::runtime_error("built without HIP"); }
 void gpu_free_batch(Batch*) {}
+void gpu_resolve_count(Batch*) {}
 std::shared_ptr<OpLogic> make_gpu_logic(const std::string&, const std::string&,
                                         const std::vector<double>&,
                                         const std::vector<int64_t>&, Engine*, int, int,

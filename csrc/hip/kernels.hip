@@ -169,31 +169,29 @@ __global__ void k_cp_count(const uint32_t* flags, int64_t n, uint32_t* blk_cnt) 
     }
 }
 
-// single-block exclusive scan over m entries (m <= a few 100k: loop chunks)
+// single-block exclusive scan (chunk-per-thread; see k_rs_scan note)
 __global__ void k_scan_exclusive(uint32_t* a, int64_t m, int64_t* d_total) {
-    __shared__ uint32_t carry;
-    if (threadIdx.x == 0) carry = 0;
+    __shared__ uint32_t tot[1024];
+    const int64_t chunk = (m + 1023) / 1024;
+    const int64_t b0 = (int64_t)threadIdx.x * chunk;
+    const int64_t b1 = min(m, b0 + chunk);
+    uint32_t s = 0;
+    for (int64_t i = b0; i < b1; ++i) s += a[i];
+    tot[threadIdx.x] = s;
     __syncthreads();
-    __shared__ uint32_t buf[1024];
-    for (int64_t base = 0; base < m; base += 1024) {
-        int64_t i = base + threadIdx.x;
-        uint32_t v = (i < m && threadIdx.x < 1024) ? a[i] : 0;
-        // Hillis-Steele inclusive scan in LDS
-        buf[threadIdx.x] = v;
+    for (int off = 1; off < 1024; off <<= 1) {
+        uint32_t t = (threadIdx.x >= off) ? tot[threadIdx.x - off] : 0;
         __syncthreads();
-        for (int off = 1; off < 1024; off <<= 1) {
-            uint32_t t = (threadIdx.x >= off) ? buf[threadIdx.x - off] : 0;
-            __syncthreads();
-            buf[threadIdx.x] += t;
-            __syncthreads();
-        }
-        uint32_t incl = buf[threadIdx.x];
-        if (i < m) a[i] = carry + incl - v;  // exclusive
-        __syncthreads();
-        if (threadIdx.x == 1023) carry += buf[1023];
+        tot[threadIdx.x] += t;
         __syncthreads();
     }
-    if (threadIdx.x == 0 && d_total) *d_total = carry;
+    uint32_t run = tot[threadIdx.x] - s;
+    for (int64_t i = b0; i < b1; ++i) {
+        uint32_t v = a[i];
+        a[i] = run;
+        run += v;
+    }
+    if (threadIdx.x == 1023 && d_total) *d_total = tot[1023];
 }
 
 __global__ void k_cp_scatter(const uint32_t* flags, int64_t n, const uint32_t* blk_off,

@@ -103,26 +103,30 @@ __global__ void k_rs_hist(const uint32_t* keys, int64_t n, int shift, uint32_t* 
     if (threadIdx.x < 16) hist[(int64_t)threadIdx.x * nblocks + blockIdx.x] = h[threadIdx.x];
 }
 
-__global__ void k_rs_scan(uint32_t* hist, int64_t m) {  // single block, 1024 thr
-    __shared__ uint32_t carry;
-    __shared__ uint32_t buf[1024];
-    if (threadIdx.x == 0) carry = 0;
+// single-block exclusive scan: each thread owns a contiguous chunk
+// (sequential sum -> 1024-wide block scan -> sequential rewrite).  Two
+// passes over the data at full thread parallelism, vs the old per-1024
+// Hillis-Steele chunk loop that cost 44 us on 31K entries.
+__global__ void k_rs_scan(uint32_t* a, int64_t m) {  // single block, 1024 thr
+    __shared__ uint32_t tot[1024];
+    const int64_t chunk = (m + 1023) / 1024;
+    const int64_t b0 = (int64_t)threadIdx.x * chunk;
+    const int64_t b1 = min(m, b0 + chunk);
+    uint32_t s = 0;
+    for (int64_t i = b0; i < b1; ++i) s += a[i];
+    tot[threadIdx.x] = s;
     __syncthreads();
-    for (int64_t base = 0; base < m; base += 1024) {
-        int64_t i = base + threadIdx.x;
-        uint32_t v = (i < m) ? hist[i] : 0;
-        buf[threadIdx.x] = v;
+    for (int off = 1; off < 1024; off <<= 1) {
+        uint32_t t = (threadIdx.x >= off) ? tot[threadIdx.x - off] : 0;
         __syncthreads();
-        for (int off = 1; off < 1024; off <<= 1) {
-            uint32_t t = (threadIdx.x >= off) ? buf[threadIdx.x - off] : 0;
-            __syncthreads();
-            buf[threadIdx.x] += t;
-            __syncthreads();
-        }
-        if (i < m) hist[i] = carry + buf[threadIdx.x] - v;
+        tot[threadIdx.x] += t;
         __syncthreads();
-        if (threadIdx.x == 1023) carry += buf[1023];
-        __syncthreads();
+    }
+    uint32_t run = tot[threadIdx.x] - s;  // exclusive base for this chunk
+    for (int64_t i = b0; i < b1; ++i) {
+        uint32_t v = a[i];
+        a[i] = run;
+        run += v;
     }
 }
 
@@ -256,13 +260,14 @@ __global__ void k_seg_scatter(const uint32_t* slot, int64_t n, const uint32_t* b
 
 __global__ void k_nseg_total(const uint32_t* slot_sorted, int64_t n,
                              const uint32_t* scan_tmp, int64_t nb, int64_t* d_nseg) {
-    if (blockIdx.x == 0 && threadIdx.x == 0) {
-        // recount boundaries of the LAST block; add its exclusive base
+    // one wave recounts the last block's boundaries in parallel
+    if (blockIdx.x == 0 && threadIdx.x < 64) {
         int64_t base = (nb - 1) * (int64_t)RS_PER_BLOCK;
         uint32_t c = 0;
-        for (int64_t i = base; i < n; ++i)
+        for (int64_t i = base + threadIdx.x; i < n; i += 64)
             c += (i == 0) || (slot_sorted[i] != slot_sorted[i - 1]);
-        *d_nseg = (int64_t)scan_tmp[nb - 1] + c;
+        for (int off = 32; off; off >>= 1) c += __shfl_down(c, off, 64);
+        if (threadIdx.x == 0) *d_nseg = (int64_t)scan_tmp[nb - 1] + c;
     }
 }
 
@@ -402,6 +407,99 @@ __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
     }
 }
 
+// Wave-per-segment variant: 64 lanes cooperatively stream one key's
+// segment in coalesced 256 B chunks; per-chunk pane partials come from
+// masked wave reduces; the (tiny) per-pane state machine runs redundantly
+// lane-uniform, with lane 0 doing the writes.  For pane_len >= 32 this
+// replaces the thread-per-segment kernel (which at 8 K keys puts only 32
+// waves on 256 CUs — 1.5 % occupancy, latency-bound).
+__global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_slot,
+                               const int64_t* d_nseg, int64_t n,
+                               const float* v_sorted, const int64_t* ts_sorted,
+                               int64_t pane_len, int64_t P, int64_t S, int comb,
+                               int ring_log2, int64_t* st_count, uint32_t* st_fill,
+                               float* st_acc, float* ring, uint32_t* st_head,
+                               float* st_wsum, const uint64_t* slot_to_key,
+                               uint64_t* out_key, float* out_val, int64_t* out_ts,
+                               int64_t* d_out_n, int64_t out_cap) {
+    const int64_t nseg = *d_nseg;
+    const uint32_t R = 1u << ring_log2;
+    const uint32_t Rm = R - 1;
+    const float ident = (comb == 0) ? 0.f : (comb == 1 ? INFINITY : -INFINITY);
+    const int lane = threadIdx.x & 63;
+    const int64_t wid = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
+    const uint32_t L = (uint32_t)pane_len;
+#define WCOMB(a, b) ((comb == 0) ? (a) + (b) : (comb == 1 ? fminf(a, b) : fmaxf(a, b)))
+    for (int64_t j = wid; j < nseg; j += nw) {
+        const uint32_t slot = seg_slot[j];
+        const int64_t i0 = seg_start[j];
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        uint32_t fill = st_fill[slot];
+        float acc = st_acc[slot];
+        uint32_t head = st_head[slot];
+        float wsum = st_wsum[slot];
+        float* rg = ring + (size_t)slot * R;
+        for (int64_t pos = i0; pos < e; pos += 64) {
+            const uint32_t nchunk = (uint32_t)min((int64_t)64, e - pos);
+            float v = (lane < (int)nchunk) ? v_sorted[pos + lane] : ident;
+            uint32_t rel = (lane < (int)nchunk) ? (fill + (uint32_t)lane) / L : ~0u;
+            const uint32_t maxrel = (fill + nchunk - 1) / L;
+            const uint32_t ncomplete = (fill + nchunk) / L;  // panes finished here
+            for (uint32_t r = 0; r <= maxrel; ++r) {
+                float pv = (rel == r) ? v : ident;
+                for (int o = 32; o; o >>= 1) pv = WCOMB(pv, __shfl_xor(pv, o, 64));
+                acc = WCOMB(acc, pv);
+                if (r < ncomplete) {  // pane boundary crossed inside this chunk
+                    if (lane == 0) rg[head & Rm] = acc;
+                    if (comb == 0) {
+                        wsum += acc;
+                        if (head >= (uint32_t)P) wsum -= rg[(head - (uint32_t)P) & Rm];
+                    }
+                    ++head;
+                    const float closed = acc;
+                    acc = ident;
+                    if (head >= (uint32_t)P &&
+                        ((head - (uint32_t)P) % (uint32_t)S) == 0) {
+                        float res;
+                        if (comb == 0) {
+                            res = wsum;
+                        } else {
+                            // lane-parallel recombine of the last P panes
+                            float part = ident;
+                            for (uint32_t q = lane; q < (uint32_t)P; q += 64) {
+                                float x = (q == 0) ? closed : rg[(head - 1 - q) & Rm];
+                                part = WCOMB(part, x);
+                            }
+                            for (int o = 32; o; o >>= 1)
+                                part = WCOMB(part, __shfl_xor(part, o, 64));
+                            res = part;
+                        }
+                        if (lane == 0) {
+                            int64_t p2 = atomicAdd((unsigned long long*)d_out_n, 1ull);
+                            if (p2 < out_cap) {
+                                out_key[p2] = slot_to_key[slot];
+                                out_val[p2] = res;
+                                int64_t last = pos + (int64_t)((r + 1) * L - fill) - 1;
+                                out_ts[p2] = ts_sorted ? ts_sorted[last] : 0;
+                            }
+                        }
+                    }
+                }
+            }
+            fill = fill + nchunk - ncomplete * L;
+        }
+        if (lane == 0) {
+            st_fill[slot] = fill;
+            st_acc[slot] = acc;
+            st_head[slot] = head;
+            st_wsum[slot] = wsum;
+            st_count[slot] += e - i0;
+        }
+    }
+#undef WCOMB
+}
+
 extern "C" void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                                  const uint32_t* seg_slot, const int64_t* d_nseg,
                                  int64_t n, const float* v_sorted,
@@ -412,11 +510,18 @@ extern "C" void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                                  const uint64_t* slot_to_key, uint64_t* out_key,
                                  float* out_val, int64_t* out_ts, int64_t* d_out_n,
                                  int64_t out_cap) {
-    hipLaunchKernelGGL(k_ffat_cb, dim3(WFA_MAX_BLOCKS / 4), dim3(WFA_THREADS), 0,
-                       (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_sorted,
-                       ts_sorted, pane_len, P, S, comb, ring_log2, st_count, st_fill,
-                       st_acc, ring, st_head, st_wsum, slot_to_key, out_key, out_val,
-                       out_ts, d_out_n, out_cap);
+    if (pane_len >= 32)
+        hipLaunchKernelGGL(k_ffat_cb_wave, dim3(WFA_MAX_BLOCKS), dim3(WFA_THREADS), 0,
+                           (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_sorted,
+                           ts_sorted, pane_len, P, S, comb, ring_log2, st_count,
+                           st_fill, st_acc, ring, st_head, st_wsum, slot_to_key,
+                           out_key, out_val, out_ts, d_out_n, out_cap);
+    else
+        hipLaunchKernelGGL(k_ffat_cb, dim3(WFA_MAX_BLOCKS / 4), dim3(WFA_THREADS), 0,
+                           (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_sorted,
+                           ts_sorted, pane_len, P, S, comb, ring_log2, st_count,
+                           st_fill, st_acc, ring, st_head, st_wsum, slot_to_key,
+                           out_key, out_val, out_ts, d_out_n, out_cap);
 }
 
 // ===== FlatFAT-tree fold: O(log R) window query for large P =====
