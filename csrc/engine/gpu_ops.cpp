@@ -151,10 +151,29 @@ struct GpuLogicBase : OpLogic {
         HIPCHK(hipEventRecord((hipEvent_t)b->ready_event, stream));
     }
 
+    // Releasing a device batch whose contents our ASYNC kernels still read:
+    // re-record its event on our stream ("uses done") so whoever pulls it
+    // from the pool next waits before overwriting.  (Single-consumer only;
+    // device-batch broadcast would need per-consumer events.)
+    void release_after_use(Batch* b) {
+        if (b->loc == Loc::DEVICE && b->ready_event &&
+            b->refcnt.load(std::memory_order_acquire) == 1)
+            HIPCHK(hipEventRecord((hipEvent_t)b->ready_event, stream));
+        release(b);
+    }
+
+    // pool get + wait for the previous user's recorded event
+    Batch* get_dev() {
+        Batch* b = dev_pool->get();
+        if (b->ready_event)
+            HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)b->ready_event, 0));
+        return b;
+    }
+
     // H2D staging: host batch -> fresh device batch (reference
     // forward_emitter_gpu.hpp CPU->GPU path, redesigned: whole-SoA copies)
     Batch* to_device(Batch* hb, RuntimeCtx& ctx) {
-        Batch* db = dev_pool->get();
+        Batch* db = get_dev();
         int64_t n = hb->count;
         HIPCHK(hipMemcpyAsync(db->ts, hb->ts, 8 * n, hipMemcpyHostToDevice, stream));
         HIPCHK(hipMemcpyAsync(db->key, hb->key, 8 * n, hipMemcpyHostToDevice, stream));
@@ -169,6 +188,9 @@ struct GpuLogicBase : OpLogic {
         db->watermark = hb->watermark;
         db->stream_tag = hb->stream_tag;
         if (ctx.stats) ctx.stats->bytes_h2d += bytes;
+        // the host batch may be recycled by a CPU producer the moment we
+        // release it — the copies must have landed first
+        HIPCHK(hipStreamSynchronize(stream));
         release(hb);
         return db;
     }
@@ -198,7 +220,7 @@ struct GpuSourceLogic : GpuLogicBase {
     bool source_step(EmitCtx& out, RuntimeCtx& ctx) override {
         ensure_init();
         if (pos >= len) return false;
-        Batch* db = dev_pool->get();
+        Batch* db = get_dev();
         int64_t n = std::min<int64_t>(bsz, len - pos);
         wfa_gen_batch(stream, db->ts, db->key, db->cols[0], vdt, n, pos, seed, n_keys);
         db->count = n;
@@ -264,7 +286,7 @@ struct GpuFilterLogic : GpuLogicBase {
         int64_t n = db->count;
         if (n > out_cap)
             throw std::runtime_error("filter input batch > out_batch capacity");
-        Batch* ob = dev_pool->get();
+        Batch* ob = get_dev();
         wfa_filter_flags(stream, spec, db->cols[col], (int)db->schema.payload[col], n,
                          a, b, d_flags);
         // column pointer table (in then out)
@@ -284,7 +306,7 @@ struct GpuFilterLogic : GpuLogicBase {
         ob->watermark = db->watermark;
         ob->stream_tag = db->stream_tag;
         if (ctx.stats) ctx.stats->num_kernels += 3;
-        release(db);
+        release_after_use(db);
         record_ready(ob);
         out.emit(ob);
     }
@@ -336,7 +358,11 @@ struct KeyedScratch {
         wfa_fill_u32(s, d_nslots, 0, 1);
     }
 
-    // returns sorted (slot,idx) and fills segments; v column cast to f32
+    uint32_t* idx_sorted = nullptr;  // valid after group()
+    const float* v_as_f32 = nullptr;
+
+    // sorts (slot, idx) pairs and fills segments; values stay unsorted and
+    // are read through idx_sorted (saves the gather round trip)
     void group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx) {
         int64_t n = db->count;
         if (n > cap)
@@ -347,16 +373,16 @@ struct KeyedScratch {
         wfa_iota_u32(s, idx, n);
         uint32_t *os, *oi;
         wfa_sort_pairs(s, slot, idx, slot_t, idx_t, hist, n, bits, &os, &oi);
-        // gather values (as f32) + ts into segment order
+        idx_sorted = oi;
         int vdt = (int)db->schema.payload[vcol];
         const void* vsrc = db->cols[vcol];
-        if (vdt != 2) {  // cast to f32 first (i64/bf16 lifted)
+        if (vdt != 2) {  // cast to f32 once (i64/bf16 lifted)
             wfa_cast(s, vsrc, vdt, v_f32, 2, n);
             vsrc = v_f32;
         }
-        wfa_gather(s, oi, n, vsrc, v_sorted, 4, db->ts, ts_sorted);
+        v_as_f32 = (const float*)vsrc;
         wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg);
-        if (ctx.stats) ctx.stats->num_kernels += 6 + 3 * ((bits + 3) / 4);
+        if (ctx.stats) ctx.stats->num_kernels += 5 + 3 * ((bits + 3) / 4);
     }
 };
 
@@ -388,15 +414,15 @@ struct GpuReduceLogic : GpuLogicBase {
         Batch* db = input_on_device(in, ctx);
         int64_t n = db->count;
         ks.group(stream, db, vcol, ctx);
-        Batch* ob = dev_pool->get();
+        Batch* ob = get_dev();
         wfa_segment_reduce(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                           ks.v_sorted, ks.ts_sorted, 2, comb, ks.slot_to_key,
-                           ob->key, ob->cols[0], ob->ts, d_on);
+                           ks.v_as_f32, ks.idx_sorted, db->ts, 2, comb,
+                           ks.slot_to_key, ob->key, ob->cols[0], ob->ts, d_on);
         HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
         ob->count = -1;
         ob->watermark = db->watermark;
         if (ctx.stats) ctx.stats->num_kernels += 1;
-        release(db);
+        release_after_use(db);
         record_ready(ob);
         out.emit(ob);
     }
@@ -456,20 +482,23 @@ struct GpuFfatLogic : GpuLogicBase {
         Batch* db = input_on_device(in, ctx);
         int64_t n = db->count;
         ks.group(stream, db, vcol, ctx);
-        Batch* ob = dev_pool->get();
-        HIPCHK(hipMemsetAsync(d_on, 0, 8, stream));
+        Batch* ob = get_dev();
+        // deterministic output offsets (no atomic cursor)
+        uint32_t* nf = (uint32_t*)ks.v_sorted;  // reuse freed scratch
+        wfa_ffat_fire_offsets(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
+                              pane_len, P, S, st_fill, st_head, nf, d_on);
         if (use_tree)
             wfa_ffat_tree_fold(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                               ks.v_sorted, ks.ts_sorted, pane_len, P, S, comb,
-                               ring_log2, st_count, st_fill, st_acc, ring_or_tree,
-                               st_head, ks.slot_to_key, ob->key, (float*)ob->cols[0],
-                               ob->ts, d_on, ob->capacity);
+                               ks.v_as_f32, ks.idx_sorted, db->ts, pane_len, P, S,
+                               comb, ring_log2, st_count, st_fill, st_acc,
+                               ring_or_tree, st_head, ks.slot_to_key, nf, ob->key,
+                               (float*)ob->cols[0], ob->ts, ob->capacity);
         else
             wfa_ffat_cb_fold(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                             ks.v_sorted, ks.ts_sorted, pane_len, P, S, comb,
-                             ring_log2, st_count, st_fill, st_acc, ring_or_tree,
-                             st_head, st_wsum, ks.slot_to_key, ob->key,
-                             (float*)ob->cols[0], ob->ts, d_on, ob->capacity);
+                             ks.v_as_f32, ks.idx_sorted, db->ts, pane_len, P, S,
+                             comb, ring_log2, st_count, st_fill, st_acc,
+                             ring_or_tree, st_head, st_wsum, ks.slot_to_key, nf,
+                             ob->key, (float*)ob->cols[0], ob->ts, ob->capacity);
         HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
         ob->count = -1;  // resolved by the consumer via gpu_resolve_count
         ob->watermark = db->watermark;

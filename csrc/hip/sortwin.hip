@@ -103,10 +103,37 @@ __global__ void k_rs_hist(const uint32_t* keys, int64_t n, int shift, uint32_t* 
     if (threadIdx.x < 16) hist[(int64_t)threadIdx.x * nblocks + blockIdx.x] = h[threadIdx.x];
 }
 
+// per-digit exclusive scan: 16 independent blocks, block d scans its own
+// hist[d*nblocks .. d*nblocks+nblocks) and records the digit total; the
+// scatter kernel turns digit totals into global digit bases with a
+// 16-element prefix.  Replaces one serial global scan of 16*nblocks.
+__global__ void k_rs_scan16(uint32_t* hist, int64_t nblocks, uint32_t* dt) {
+    uint32_t* a = hist + (int64_t)blockIdx.x * nblocks;
+    __shared__ uint32_t tot[256];
+    const int64_t chunk = (nblocks + 255) / 256;
+    const int64_t b0 = (int64_t)threadIdx.x * chunk;
+    const int64_t b1 = min(nblocks, b0 + chunk);
+    uint32_t s = 0;
+    for (int64_t i = b0; i < b1; ++i) s += a[i];
+    tot[threadIdx.x] = s;
+    __syncthreads();
+    for (int off = 1; off < 256; off <<= 1) {
+        uint32_t t = (threadIdx.x >= off) ? tot[threadIdx.x - off] : 0;
+        __syncthreads();
+        tot[threadIdx.x] += t;
+        __syncthreads();
+    }
+    uint32_t run = tot[threadIdx.x] - s;
+    for (int64_t i = b0; i < b1; ++i) {
+        uint32_t v = a[i];
+        a[i] = run;
+        run += v;
+    }
+    if (threadIdx.x == 255) dt[blockIdx.x] = tot[255];
+}
+
 // single-block exclusive scan: each thread owns a contiguous chunk
-// (sequential sum -> 1024-wide block scan -> sequential rewrite).  Two
-// passes over the data at full thread parallelism, vs the old per-1024
-// Hillis-Steele chunk loop that cost 44 us on 31K entries.
+// (sequential sum -> 1024-wide block scan -> sequential rewrite).
 __global__ void k_rs_scan(uint32_t* a, int64_t m) {  // single block, 1024 thr
     __shared__ uint32_t tot[1024];
     const int64_t chunk = (m + 1023) / 1024;
@@ -134,9 +161,18 @@ __global__ void k_rs_scan(uint32_t* a, int64_t m) {  // single block, 1024 thr
 // count vectors across the block's threads.
 __global__ void k_rs_scatter(const uint32_t* keys, const uint32_t* vals, int64_t n,
                              int shift, const uint32_t* hist, int64_t nblocks,
-                             uint32_t* keys_out, uint32_t* vals_out) {
+                             const uint32_t* dt, uint32_t* keys_out,
+                             uint32_t* vals_out) {
     __shared__ uint32_t tc[WFA_THREADS][17];  // [thread][digit] padded
     __shared__ uint32_t gbase[16];
+    __shared__ uint32_t dbase[16];
+    if (threadIdx.x == 0) {
+        uint32_t run = 0;
+        for (int d = 0; d < 16; ++d) {
+            dbase[d] = run;
+            run += dt[d];
+        }
+    }
     int64_t base = (int64_t)blockIdx.x * RS_PER_BLOCK + (int64_t)threadIdx.x * RS_IPT;
     uint32_t k_[RS_IPT];
     uint32_t v_[RS_IPT];
@@ -158,6 +194,8 @@ __global__ void k_rs_scatter(const uint32_t* keys, const uint32_t* vals, int64_t
     for (int d = 0; d < 16; ++d) tc[threadIdx.x][d] = cnt_local[d];
     if (threadIdx.x < 16)
         gbase[threadIdx.x] = hist[(int64_t)threadIdx.x * nblocks + blockIdx.x];
+    __syncthreads();
+    if (threadIdx.x < 16) gbase[threadIdx.x] += dbase[threadIdx.x];
     __syncthreads();
     // Hillis-Steele inclusive scan over threads of the 16-vector
     for (int off = 1; off < WFA_THREADS; off <<= 1) {
@@ -191,15 +229,16 @@ extern "C" void wfa_sort_pairs(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                                uint32_t** out_idx) {
     hipStream_t st = (hipStream_t)s;
     int64_t nblocks = wfa_sort_nblocks(n);
+    uint32_t* dt = hist + 16 * nblocks;  // hist is sized 16*nblocks + 16
     uint32_t *ka = slot, *va = idx, *kb = slot_tmp, *vb = idx_tmp;
     int passes = (bits + 3) / 4;
     for (int p = 0; p < passes; ++p) {
         int shift = 4 * p;
         hipLaunchKernelGGL(k_rs_hist, dim3(nblocks), dim3(WFA_THREADS), 0, st, ka, n,
                            shift, hist, nblocks);
-        hipLaunchKernelGGL(k_rs_scan, dim3(1), dim3(1024), 0, st, hist, 16 * nblocks);
+        hipLaunchKernelGGL(k_rs_scan16, dim3(16), dim3(256), 0, st, hist, nblocks, dt);
         hipLaunchKernelGGL(k_rs_scatter, dim3(nblocks), dim3(WFA_THREADS), 0, st, ka,
-                           va, n, shift, hist, nblocks, kb, vb);
+                           va, n, shift, hist, nblocks, dt, kb, vb);
         uint32_t* t;
         t = ka; ka = kb; kb = t;
         t = va; va = vb; vb = t;
@@ -281,14 +320,15 @@ extern "C" void wfa_segments(wfa_stream_t s, const uint32_t* slot_sorted, int64_
     hipLaunchKernelGGL(k_rs_scan, dim3(1), dim3(1024), 0, st, scan_tmp, nb);
     hipLaunchKernelGGL(k_seg_scatter, dim3(nb), dim3(WFA_THREADS), 0, st, slot_sorted,
                        n, scan_tmp, seg_start, seg_slot);
-    hipLaunchKernelGGL(k_nseg_total, dim3(1), dim3(1), 0, st, slot_sorted, n, scan_tmp,
+    hipLaunchKernelGGL(k_nseg_total, dim3(1), dim3(64), 0, st, slot_sorted, n, scan_tmp,
                        nb, d_nseg);
 }
 
 // ===== segmented reduce (Reduce_GPU per-batch semantics) =====
 __global__ void k_seg_reduce(const uint32_t* seg_start, const uint32_t* seg_slot,
-                             const int64_t* d_nseg, int64_t n, const void* v_sorted,
-                             const int64_t* ts_sorted, int vdt, int comb,
+                             const int64_t* d_nseg, int64_t n, const void* v_orig,
+                             const uint32_t* idx_sorted,
+                             const int64_t* ts_orig, int vdt, int comb,
                              const uint64_t* slot_to_key, uint64_t* out_key,
                              void* out_val, int64_t* out_ts, int64_t* d_out_n) {
     int64_t nseg = *d_nseg;
@@ -296,29 +336,29 @@ __global__ void k_seg_reduce(const uint32_t* seg_start, const uint32_t* seg_slot
          j += gridDim.x * (int64_t)blockDim.x) {
         int64_t b = seg_start[j];
         int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
-        int64_t tmax = ts_sorted ? ts_sorted[b] : 0;
+        int64_t tmax = ts_orig ? ts_orig[idx_sorted[b]] : 0;
         if (vdt == 0) {
-            const int64_t* v = (const int64_t*)v_sorted;
-            int64_t acc = (comb == 3) ? 0 : v[b];
+            const int64_t* v = (const int64_t*)v_orig;
+            int64_t acc = (comb == 3) ? 0 : v[idx_sorted[b]];
             if (comb == 3) acc = e - b;
             else
                 for (int64_t i = b + 1; i < e; ++i) {
-                    int64_t x = v[i];
+                    int64_t x = v[idx_sorted[i]];
                     acc = (comb == 0) ? acc + x : (comb == 1 ? min(acc, x) : max(acc, x));
                 }
             ((int64_t*)out_val)[j] = acc;
         } else {
-            const float* v = (const float*)v_sorted;
-            float acc = (comb == 3) ? (float)(e - b) : v[b];
+            const float* v = (const float*)v_orig;
+            float acc = (comb == 3) ? (float)(e - b) : v[idx_sorted[b]];
             if (comb != 3)
                 for (int64_t i = b + 1; i < e; ++i) {
-                    float x = v[i];
+                    float x = v[idx_sorted[i]];
                     acc = (comb == 0) ? acc + x : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
                 }
             ((float*)out_val)[j] = acc;
         }
-        if (ts_sorted)
-            for (int64_t i = b + 1; i < e; ++i) tmax = max(tmax, ts_sorted[i]);
+        if (ts_orig)
+            for (int64_t i = b + 1; i < e; ++i) tmax = max(tmax, ts_orig[idx_sorted[i]]);
         out_key[j] = slot_to_key ? slot_to_key[seg_slot[j]] : (uint64_t)seg_slot[j];
         if (out_ts) out_ts[j] = tmax;
     }
@@ -327,14 +367,80 @@ __global__ void k_seg_reduce(const uint32_t* seg_start, const uint32_t* seg_slot
 
 extern "C" void wfa_segment_reduce(wfa_stream_t s, const uint32_t* seg_start,
                                    const uint32_t* seg_slot, const int64_t* d_nseg,
-                                   int64_t n, const void* v_sorted,
-                                   const int64_t* ts_sorted, int vdt, int comb,
+                                   int64_t n, const void* v_orig,
+                                   const uint32_t* idx_sorted,
+                                   const int64_t* ts_orig, int vdt, int comb,
                                    const uint64_t* slot_to_key, uint64_t* out_key,
                                    void* out_val, int64_t* out_ts, int64_t* d_out_n) {
     hipLaunchKernelGGL(k_seg_reduce, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
-                       (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_sorted,
-                       ts_sorted, vdt, comb, slot_to_key, out_key, out_val, out_ts,
-                       d_out_n);
+                       (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_orig,
+                       idx_sorted, ts_orig, vdt, comb, slot_to_key, out_key, out_val,
+                       out_ts, d_out_n);
+}
+
+
+// ===== deterministic window-output offsets =====
+// Counting fires per segment is closed-form from the per-slot pane state,
+// so output positions come from an exclusive scan instead of 40K serial
+// atomicAdds on one cursor word (~88 atomics/us saturation — measured
+// 455 us per 4M-tuple batch before this pass existed).
+__global__ void k_fire_count(const uint32_t* seg_start, const uint32_t* seg_slot,
+                             const int64_t* d_nseg, int64_t n, int64_t pane_len,
+                             int64_t P, int64_t S, const uint32_t* st_fill,
+                             const uint32_t* st_head, uint32_t* nf) {
+    const int64_t nseg = *d_nseg;
+    for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nseg;
+         j += gridDim.x * (int64_t)blockDim.x) {
+        const uint32_t slot = seg_slot[j];
+        const int64_t b = seg_start[j];
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        const uint64_t fill = st_fill[slot];
+        const uint64_t head0 = st_head[slot];
+        const uint64_t ncomp = (fill + (uint64_t)(e - b)) / (uint64_t)pane_len;
+        auto F = [&](uint64_t x) {
+            return x < (uint64_t)P ? 0ull : (x - (uint64_t)P) / (uint64_t)S + 1ull;
+        };
+        nf[j] = (uint32_t)(F(head0 + ncomp) - F(head0));
+    }
+}
+
+// exclusive scan of nf[0..*d_nseg) + total -> *d_out_n (single block)
+__global__ void k_fire_scan(uint32_t* nf, const int64_t* d_nseg, int64_t* d_out_n) {
+    const int64_t m = *d_nseg;
+    __shared__ uint32_t tot[1024];
+    const int64_t chunk = (m + 1023) / 1024;
+    const int64_t b0 = (int64_t)threadIdx.x * chunk;
+    const int64_t b1 = min(m, b0 + chunk);
+    uint32_t s = 0;
+    for (int64_t i = b0; i < b1; ++i) s += nf[i];
+    tot[threadIdx.x] = s;
+    __syncthreads();
+    for (int off = 1; off < 1024; off <<= 1) {
+        uint32_t t = (threadIdx.x >= off) ? tot[threadIdx.x - off] : 0;
+        __syncthreads();
+        tot[threadIdx.x] += t;
+        __syncthreads();
+    }
+    uint32_t run = tot[threadIdx.x] - s;
+    for (int64_t i = b0; i < b1; ++i) {
+        uint32_t v = nf[i];
+        nf[i] = run;
+        run += v;
+    }
+    if (threadIdx.x == 1023 && d_out_n) *d_out_n = tot[1023];
+}
+
+extern "C" void wfa_ffat_fire_offsets(wfa_stream_t s, const uint32_t* seg_start,
+                                      const uint32_t* seg_slot, const int64_t* d_nseg,
+                                      int64_t n, int64_t pane_len, int64_t P,
+                                      int64_t S, const uint32_t* st_fill,
+                                      const uint32_t* st_head, uint32_t* nf,
+                                      int64_t* d_out_n) {
+    hipStream_t st = (hipStream_t)s;
+    hipLaunchKernelGGL(k_fire_count, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
+                       st, seg_start, seg_slot, d_nseg, n, pane_len, P, S, st_fill,
+                       st_head, nf);
+    hipLaunchKernelGGL(k_fire_scan, dim3(1), dim3(1024), 0, st, nf, d_nseg, d_out_n);
 }
 
 // ===== batched multi-key CB sliding-window fold (pane ring) =====
@@ -344,13 +450,15 @@ extern "C" void wfa_segment_reduce(wfa_stream_t s, const uint32_t* seg_start,
 // reference's per-key host loop + per-key stream (ffat_replica_gpu.hpp:
 // 829-867) becomes a dense device-side state machine.
 __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
-                          const int64_t* d_nseg, int64_t n, const float* v_sorted,
-                          const int64_t* ts_sorted, int64_t pane_len, int64_t P,
+                          const int64_t* d_nseg, int64_t n, const float* v_f32,
+                          const uint32_t* idx_sorted, const int64_t* ts_orig,
+                          int64_t pane_len, int64_t P,
                           int64_t S, int comb, int ring_log2, int64_t* st_count,
                           uint32_t* st_fill, float* st_acc, float* ring,
                           uint32_t* st_head, float* st_wsum,
-                          const uint64_t* slot_to_key, uint64_t* out_key,
-                          float* out_val, int64_t* out_ts, int64_t* d_out_n,
+                          const uint64_t* slot_to_key, const uint32_t* fire_base,
+                          uint64_t* out_key,
+                          float* out_val, int64_t* out_ts,
                           int64_t out_cap) {
     const int64_t nseg = *d_nseg;
     const uint32_t R = 1u << ring_log2;
@@ -365,9 +473,10 @@ __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
         float acc = st_acc[slot];
         uint32_t head = st_head[slot];
         float wsum = st_wsum[slot];
+        int64_t w = fire_base[j];
         float* rg = ring + (size_t)slot * R;
         for (; i < e; ++i) {
-            float x = v_sorted[i];
+            float x = v_f32[idx_sorted[i]];
             acc = (comb == 0) ? acc + x : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
             if (++fill == (uint32_t)pane_len) {
                 // pane complete
@@ -390,12 +499,12 @@ __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
                             res = (comb == 1) ? fminf(res, pv) : fmaxf(res, pv);
                         }
                     }
-                    int64_t pos = atomicAdd((unsigned long long*)d_out_n, 1ull);
-                    if (pos < out_cap) {
-                        out_key[pos] = slot_to_key[slot];
-                        out_val[pos] = res;
-                        out_ts[pos] = ts_sorted ? ts_sorted[i] : 0;
+                    if (w < out_cap) {
+                        out_key[w] = slot_to_key[slot];
+                        out_val[w] = res;
+                        out_ts[w] = ts_orig ? ts_orig[idx_sorted[i]] : 0;
                     }
+                    ++w;
                 }
             }
         }
@@ -415,13 +524,15 @@ __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
 // waves on 256 CUs — 1.5 % occupancy, latency-bound).
 __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_slot,
                                const int64_t* d_nseg, int64_t n,
-                               const float* v_sorted, const int64_t* ts_sorted,
+                               const float* v_f32, const uint32_t* idx_sorted,
+                               const int64_t* ts_orig,
                                int64_t pane_len, int64_t P, int64_t S, int comb,
                                int ring_log2, int64_t* st_count, uint32_t* st_fill,
                                float* st_acc, float* ring, uint32_t* st_head,
                                float* st_wsum, const uint64_t* slot_to_key,
+                               const uint32_t* fire_base,
                                uint64_t* out_key, float* out_val, int64_t* out_ts,
-                               int64_t* d_out_n, int64_t out_cap) {
+                               int64_t out_cap) {
     const int64_t nseg = *d_nseg;
     const uint32_t R = 1u << ring_log2;
     const uint32_t Rm = R - 1;
@@ -439,10 +550,11 @@ __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_sl
         float acc = st_acc[slot];
         uint32_t head = st_head[slot];
         float wsum = st_wsum[slot];
+        int64_t w = fire_base[j];
         float* rg = ring + (size_t)slot * R;
         for (int64_t pos = i0; pos < e; pos += 64) {
             const uint32_t nchunk = (uint32_t)min((int64_t)64, e - pos);
-            float v = (lane < (int)nchunk) ? v_sorted[pos + lane] : ident;
+            float v = (lane < (int)nchunk) ? v_f32[idx_sorted[pos + lane]] : ident;
             uint32_t rel = (lane < (int)nchunk) ? (fill + (uint32_t)lane) / L : ~0u;
             const uint32_t maxrel = (fill + nchunk - 1) / L;
             const uint32_t ncomplete = (fill + nchunk) / L;  // panes finished here
@@ -475,15 +587,13 @@ __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_sl
                                 part = WCOMB(part, __shfl_xor(part, o, 64));
                             res = part;
                         }
-                        if (lane == 0) {
-                            int64_t p2 = atomicAdd((unsigned long long*)d_out_n, 1ull);
-                            if (p2 < out_cap) {
-                                out_key[p2] = slot_to_key[slot];
-                                out_val[p2] = res;
-                                int64_t last = pos + (int64_t)((r + 1) * L - fill) - 1;
-                                out_ts[p2] = ts_sorted ? ts_sorted[last] : 0;
-                            }
+                        if (lane == 0 && w < out_cap) {
+                            out_key[w] = slot_to_key[slot];
+                            out_val[w] = res;
+                            int64_t last = pos + (int64_t)((r + 1) * L - fill) - 1;
+                            out_ts[w] = ts_orig ? ts_orig[idx_sorted[last]] : 0;
                         }
+                        ++w;
                     }
                 }
             }
@@ -502,26 +612,27 @@ __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_sl
 
 extern "C" void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                                  const uint32_t* seg_slot, const int64_t* d_nseg,
-                                 int64_t n, const float* v_sorted,
-                                 const int64_t* ts_sorted, int64_t pane_len, int64_t P,
+                                 int64_t n, const float* v_f32,
+                                 const uint32_t* idx_sorted, const int64_t* ts_orig,
+                                 int64_t pane_len, int64_t P,
                                  int64_t S, int comb, int ring_log2, int64_t* st_count,
                                  uint32_t* st_fill, float* st_acc, float* ring,
                                  uint32_t* st_head, float* st_wsum,
-                                 const uint64_t* slot_to_key, uint64_t* out_key,
-                                 float* out_val, int64_t* out_ts, int64_t* d_out_n,
-                                 int64_t out_cap) {
+                                 const uint64_t* slot_to_key,
+                                 const uint32_t* fire_base, uint64_t* out_key,
+                                 float* out_val, int64_t* out_ts, int64_t out_cap) {
     if (pane_len >= 32)
         hipLaunchKernelGGL(k_ffat_cb_wave, dim3(WFA_MAX_BLOCKS), dim3(WFA_THREADS), 0,
-                           (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_sorted,
-                           ts_sorted, pane_len, P, S, comb, ring_log2, st_count,
-                           st_fill, st_acc, ring, st_head, st_wsum, slot_to_key,
-                           out_key, out_val, out_ts, d_out_n, out_cap);
+                           (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_f32,
+                           idx_sorted, ts_orig, pane_len, P, S, comb, ring_log2,
+                           st_count, st_fill, st_acc, ring, st_head, st_wsum,
+                           slot_to_key, fire_base, out_key, out_val, out_ts, out_cap);
     else
         hipLaunchKernelGGL(k_ffat_cb, dim3(WFA_MAX_BLOCKS / 4), dim3(WFA_THREADS), 0,
-                           (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_sorted,
-                           ts_sorted, pane_len, P, S, comb, ring_log2, st_count,
-                           st_fill, st_acc, ring, st_head, st_wsum, slot_to_key,
-                           out_key, out_val, out_ts, d_out_n, out_cap);
+                           (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_f32,
+                           idx_sorted, ts_orig, pane_len, P, S, comb, ring_log2,
+                           st_count, st_fill, st_acc, ring, st_head, st_wsum,
+                           slot_to_key, fire_base, out_key, out_val, out_ts, out_cap);
 }
 
 // ===== FlatFAT-tree fold: O(log R) window query for large P =====
@@ -531,13 +642,15 @@ extern "C" void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
 // query of the last P leaves via the classic two-pointer FlatFAT walk
 // (Tangwongsan VLDB'15 — reference flatfat.hpp:311-337 getResult).
 __global__ void k_ffat_tree(const uint32_t* seg_start, const uint32_t* seg_slot,
-                            const int64_t* d_nseg, int64_t n, const float* v_sorted,
-                            const int64_t* ts_sorted, int64_t pane_len, int64_t P,
+                            const int64_t* d_nseg, int64_t n, const float* v_f32,
+                            const uint32_t* idx_sorted, const int64_t* ts_orig,
+                            int64_t pane_len, int64_t P,
                             int64_t S, int comb, int ring_log2, int64_t* st_count,
                             uint32_t* st_fill, float* st_acc, float* tree,
                             uint32_t* st_head, const uint64_t* slot_to_key,
+                            const uint32_t* fire_base,
                             uint64_t* out_key, float* out_val, int64_t* out_ts,
-                            int64_t* d_out_n, int64_t out_cap) {
+                            int64_t out_cap) {
     const int64_t nseg = *d_nseg;
     const uint32_t R = 1u << ring_log2;
     const uint32_t Rm = R - 1;
@@ -551,9 +664,10 @@ __global__ void k_ffat_tree(const uint32_t* seg_start, const uint32_t* seg_slot,
         uint32_t fill = st_fill[slot];
         float acc = st_acc[slot];
         uint32_t head = st_head[slot];
+        int64_t w = fire_base[j];
         float* tr = tree + (size_t)slot * 2 * R;
         for (; i < e; ++i) {
-            float x = v_sorted[i];
+            float x = v_f32[idx_sorted[i]];
             acc = TCOMB(acc, x);
             if (++fill == (uint32_t)pane_len) {
                 uint32_t leaf = (head & Rm) + R;
@@ -586,12 +700,12 @@ __global__ void k_ffat_tree(const uint32_t* seg_start, const uint32_t* seg_slot,
                         res = range_q(lo, hi);
                     else
                         res = TCOMB(range_q(lo, Rm), range_q(0, hi));
-                    int64_t pos = atomicAdd((unsigned long long*)d_out_n, 1ull);
-                    if (pos < out_cap) {
-                        out_key[pos] = slot_to_key[slot];
-                        out_val[pos] = res;
-                        out_ts[pos] = ts_sorted ? ts_sorted[i] : 0;
+                    if (w < out_cap) {
+                        out_key[w] = slot_to_key[slot];
+                        out_val[w] = res;
+                        out_ts[w] = ts_orig ? ts_orig[idx_sorted[i]] : 0;
                     }
+                    ++w;
                 }
             }
         }
@@ -605,14 +719,15 @@ __global__ void k_ffat_tree(const uint32_t* seg_start, const uint32_t* seg_slot,
 
 extern "C" void wfa_ffat_tree_fold(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
-    const int64_t* d_nseg, int64_t n, const float* v_sorted, const int64_t* ts_sorted,
+    const int64_t* d_nseg, int64_t n, const float* v_f32, const uint32_t* idx_sorted,
+    const int64_t* ts_orig,
     int64_t pane_len, int64_t P, int64_t S, int comb, int ring_log2, int64_t* st_count,
     uint32_t* st_fill, float* st_acc, float* tree, uint32_t* st_head,
-    const uint64_t* slot_to_key, uint64_t* out_key, float* out_val, int64_t* out_ts,
-    int64_t* d_out_n, int64_t out_cap) {
+    const uint64_t* slot_to_key, const uint32_t* fire_base, uint64_t* out_key,
+    float* out_val, int64_t* out_ts, int64_t out_cap) {
     hipLaunchKernelGGL(k_ffat_tree, dim3(WFA_MAX_BLOCKS / 4), dim3(WFA_THREADS), 0,
-                       (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_sorted,
-                       ts_sorted, pane_len, P, S, comb, ring_log2, st_count, st_fill,
-                       st_acc, tree, st_head, slot_to_key, out_key, out_val, out_ts,
-                       d_out_n, out_cap);
+                       (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_f32,
+                       idx_sorted, ts_orig, pane_len, P, S, comb, ring_log2, st_count,
+                       st_fill, st_acc, tree, st_head, slot_to_key, fire_base,
+                       out_key, out_val, out_ts, out_cap);
 }

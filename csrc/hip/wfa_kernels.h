@@ -67,10 +67,19 @@ void wfa_segments(wfa_stream_t s, const uint32_t* slot_sorted, int64_t n,
 // comb: 0 sum 1 min 2 max 3 count ; vdt: dtype of v (F32 or I64 accum f64/i64)
 void wfa_segment_reduce(wfa_stream_t s, const uint32_t* seg_start,
                         const uint32_t* seg_slot, const int64_t* d_nseg, int64_t n,
-                        const void* v_sorted, const int64_t* ts_sorted, int vdt,
+                        const void* v_orig, const uint32_t* idx_sorted,
+                        const int64_t* ts_orig, int vdt,
                         int comb, const uint64_t* slot_to_key,
                         uint64_t* out_key, void* out_val, int64_t* out_ts,
                         int64_t* d_out_n);
+
+// closed-form per-segment window-fire counts -> exclusive output offsets
+// (nf) + total (*d_out_n); run BEFORE the fold (reads pristine state)
+void wfa_ffat_fire_offsets(wfa_stream_t s, const uint32_t* seg_start,
+                           const uint32_t* seg_slot, const int64_t* d_nseg,
+                           int64_t n, int64_t pane_len, int64_t P, int64_t S,
+                           const uint32_t* st_fill, const uint32_t* st_head,
+                           uint32_t* nf, int64_t* d_out_n);
 
 // ----- FFAT/pane sliding-window state machine (CB) -----
 // Batched multi-key redesign of the reference's per-key FlatFAT_GPU
@@ -83,14 +92,15 @@ void wfa_segment_reduce(wfa_stream_t s, const uint32_t* seg_start,
 // Output appended to out_* at atomic cursor d_out_n.
 void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                       const uint32_t* seg_slot, const int64_t* d_nseg, int64_t n,
-                      const float* v_sorted, const int64_t* ts_sorted,
+                      const float* v_f32, const uint32_t* idx_sorted,
+                      const int64_t* ts_orig,
                       int64_t pane_len, int64_t P, int64_t S, int comb,
                       int ring_log2,
                       int64_t* st_count, uint32_t* st_fill, float* st_acc,
                       float* ring, uint32_t* st_head, float* st_wsum,
-                      const uint64_t* slot_to_key,
+                      const uint64_t* slot_to_key, const uint32_t* fire_base,
                       uint64_t* out_key, float* out_val, int64_t* out_ts,
-                      int64_t* d_out_n, int64_t out_cap);
+                      int64_t out_cap);
 
 // ----- FlatFAT arena path (non-invertible combines over many panes) -----
 // Per-slot complete binary tree over ring of 2^ring_log2 pane leaves,
@@ -98,14 +108,15 @@ void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
 // O(log R) range query, one thread per segment, batched over all keys.
 void wfa_ffat_tree_fold(wfa_stream_t s, const uint32_t* seg_start,
                         const uint32_t* seg_slot, const int64_t* d_nseg, int64_t n,
-                        const float* v_sorted, const int64_t* ts_sorted,
+                        const float* v_f32, const uint32_t* idx_sorted,
+                        const int64_t* ts_orig,
                         int64_t pane_len, int64_t P, int64_t S, int comb,
                         int ring_log2,
                         int64_t* st_count, uint32_t* st_fill, float* st_acc,
                         float* tree, uint32_t* st_head,
-                        const uint64_t* slot_to_key,
+                        const uint64_t* slot_to_key, const uint32_t* fire_base,
                         uint64_t* out_key, float* out_val, int64_t* out_ts,
-                        int64_t* d_out_n, int64_t out_cap);
+                        int64_t out_cap);
 
 // ----- misc -----
 void wfa_fill_u64(wfa_stream_t s, uint64_t* p, uint64_t v, int64_t n);
