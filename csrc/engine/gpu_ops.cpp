@@ -65,6 +65,35 @@ struct DeviceArena {
 
 static DeviceArena g_arena[64];
 
+// hipEventCreate + hipHostMalloc cost ~0.1-1 ms each; recycle them across
+// batches AND engines (a fresh PipeGraph would otherwise re-pin per batch
+// inside the timed region).
+struct AuxPool {
+    std::mutex mu;
+    std::vector<std::pair<void*, int64_t*>> free_list;  // (event, pinned slot)
+    std::pair<void*, int64_t*> get(int device) {
+        {
+            std::lock_guard<std::mutex> g(mu);
+            if (!free_list.empty()) {
+                auto p = free_list.back();
+                free_list.pop_back();
+                return p;
+            }
+        }
+        hipEvent_t ev;
+        int64_t* slot;
+        HIPCHK(hipSetDevice(device));
+        HIPCHK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+        HIPCHK(hipHostMalloc((void**)&slot, 8, hipHostMallocDefault));
+        return {ev, slot};
+    }
+    void put(void* ev, int64_t* slot) {
+        std::lock_guard<std::mutex> g(mu);
+        free_list.push_back({ev, slot});
+    }
+};
+static AuxPool g_aux;
+
 DeviceArena& arena(int dev) {
     g_arena[dev].device = dev;
     return g_arena[dev];
@@ -93,11 +122,9 @@ Batch* gpu_alloc_batch(Pool& pool) {
         b->cols[i] = p;
         p += (dsize(pool.schema.payload[i]) * pool.capacity + 255) & ~size_t(255);
     }
-    hipEvent_t ev;
-    HIPCHK(hipSetDevice(pool.device));
-    HIPCHK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
-    b->ready_event = ev;
-    HIPCHK(hipHostMalloc((void**)&b->lazy_count, 8, hipHostMallocDefault));
+    auto aux = g_aux.get(pool.device);
+    b->ready_event = aux.first;
+    b->lazy_count = aux.second;
     return b;
 }
 
@@ -109,8 +136,7 @@ void gpu_resolve_count(Batch* b) {
 
 void gpu_free_batch(Batch* b) {
     arena(b->device).put(b->ts, dev_batch_bytes(b->schema, b->capacity));
-    if (b->ready_event) (void)hipEventDestroy((hipEvent_t)b->ready_event);
-    if (b->lazy_count) (void)hipHostFree(b->lazy_count);
+    if (b->ready_event) g_aux.put(b->ready_event, b->lazy_count);
     delete b;
 }
 
@@ -350,8 +376,7 @@ struct KeyedScratch {
         seg_start = (uint32_t*)A.get(4 * cap);
         seg_slot = (uint32_t*)A.get(4 * cap);
         d_nseg = (int64_t*)A.get(64);
-        v_sorted = (float*)A.get(8 * cap);   // also holds i64 when needed
-        ts_sorted = (int64_t*)A.get(8 * cap);
+        v_sorted = (float*)A.get(4 * cap);   // reused as fire-offset scratch
         v_f32 = (float*)A.get(4 * cap);
         wfa_fill_u64(s, tkeys, ~0ULL, table_cap);
         wfa_fill_u32(s, tslots, ~0u, table_cap);
