@@ -341,6 +341,20 @@ PYBIND11_MODULE(_core, m) {
                  e.build();
                  e.start();
              })
+        .def("start_gated",
+             [](Engine& e) {
+                 e.build();
+                 {
+                     py::gil_scoped_release rel;
+                     e.start_gated();
+                 }
+             })
+        .def("open_gate_and_wait",
+             [](Engine& e) {
+                 py::gil_scoped_release rel;
+                 e.open_gate();
+                 e.wait();
+             })
         .def("wait",
              [](Engine& e) {
                  py::gil_scoped_release rel;

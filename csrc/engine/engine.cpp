@@ -297,6 +297,10 @@ void ChainLogic::on_eos(EmitCtx&, RuntimeCtx& ctx) {
     for (size_t i = 0; i < stages.size(); ++i) stages[i]->on_eos(ctxs[i], ctx);
 }
 
+void ChainLogic::warm(RuntimeCtx& ctx) {
+    for (auto& st : stages) st->warm(ctx);
+}
+
 // ===================== replica loop =====================
 
 void Replica::run() {
@@ -309,6 +313,14 @@ void Replica::run() {
         pthread_setaffinity_np(pthread_self(), sizeof(cs), &cs);
     }
 #endif
+    logic->warm(rctx);
+    if (engine->use_gate) {
+        engine->warmed.fetch_add(1, std::memory_order_acq_rel);
+        int spins = 0;
+        while (!engine->gate.load(std::memory_order_acquire) &&
+               !engine->abort.load(std::memory_order_relaxed))
+            SpscQueue::backoff(spins);
+    }
     stats.start_us = now_us();
     if (logic->is_source()) {
         while (!engine->abort.load(std::memory_order_relaxed)) {
@@ -467,6 +479,14 @@ void Engine::start() {
         Replica* rp = r.get();
         rp->th = std::thread([rp] { rp->run(); });
     }
+}
+
+void Engine::start_gated() {
+    use_gate = true;
+    start();
+    int spins = 0;
+    while (warmed.load(std::memory_order_acquire) < (int)replicas.size())
+        SpscQueue::backoff(spins);
 }
 
 void Engine::wait() {

@@ -240,6 +240,9 @@ struct EmitCtx;  // defined below
 
 struct OpLogic {
     virtual ~OpLogic() = default;
+    // called once in the replica thread before any processing — GPU logics
+    // create streams/pools here so a gated start measures steady state only
+    virtual void warm(RuntimeCtx& ctx) {}
     // Take ownership of `in` (release or re-emit it).
     virtual void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx);
     virtual void on_eos(EmitCtx& out, RuntimeCtx& ctx) {}
@@ -300,6 +303,7 @@ struct ChainLogic : OpLogic {
         stages.front()->process(in, ctxs[0], ctx);
     }
     void on_eos(EmitCtx&, RuntimeCtx& ctx) override;
+    void warm(RuntimeCtx& ctx) override;
 };
 
 struct EdgeSpec {
@@ -336,6 +340,10 @@ struct Engine {
     std::vector<std::unique_ptr<SpscQueue>> queues;
     std::atomic<bool> abort{false};
     std::atomic<int64_t> dropped_tuples{0};
+    // gated start: threads spawn + logics warm, then block until open_gate()
+    std::atomic<int> gate{0};
+    std::atomic<int> warmed{0};
+    bool use_gate = false;
     // per-op sink accumulators (differential-test invariant support)
     std::unordered_map<int, std::atomic<int64_t>> sink_acc_i64;
     std::unordered_map<int, double> sink_acc_f64;
@@ -353,6 +361,8 @@ struct Engine {
     void start();       // spawn threads
     void wait();        // join all
     void run() { build(); start(); wait(); }
+    void start_gated();  // spawn + warm, block replicas at the gate
+    void open_gate() { gate.store(1, std::memory_order_release); }
 };
 
 // native logic factory (native_logic.cpp)

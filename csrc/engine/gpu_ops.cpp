@@ -140,8 +140,15 @@ void gpu_free_batch(Batch* b) {
     delete b;
 }
 
+static bool wfa_prof() {
+    static int v = -1;
+    if (v < 0) v = getenv("WFA_PROF") ? 1 : 0;
+    return v;
+}
+
 // ===== base for GPU logics =====
 struct GpuLogicBase : OpLogic {
+    int64_t prof_host_us = 0, prof_calls = 0;
     int device = 0;
     hipStream_t stream = nullptr;
     std::unique_ptr<Pool> dev_pool;  // device batches this logic emits
@@ -166,6 +173,8 @@ struct GpuLogicBase : OpLogic {
         if (h_count) (void)hipHostFree(h_count);
         if (stream) (void)hipStreamDestroy(stream);
     }
+
+    void warm(RuntimeCtx&) override { ensure_init(); }
 
     // make the producing stream's work visible to our stream
     void wait_ready(Batch* b) {
@@ -244,9 +253,13 @@ struct GpuSourceLogic : GpuLogicBase {
     }
     bool is_source() const override { return true; }
     bool source_step(EmitCtx& out, RuntimeCtx& ctx) override {
+        int64_t t0 = now_us();
         ensure_init();
         if (pos >= len) return false;
         Batch* db = get_dev();
+        int64_t tg = now_us() - t0;
+        if (wfa_prof() && tg > 50)
+            fprintf(stderr, "[prof] source get_dev wait %ld us\n", (long)tg);
         int64_t n = std::min<int64_t>(bsz, len - pos);
         wfa_gen_batch(stream, db->ts, db->key, db->cols[0], vdt, n, pos, seed, n_keys);
         db->count = n;
@@ -503,6 +516,7 @@ struct GpuFfatLogic : GpuLogicBase {
                      max_keys * (use_tree ? 2 * (1ll << ring_log2) : (1ll << ring_log2)));
     }
     void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        int64_t t0 = now_us();
         ensure_init();
         Batch* db = input_on_device(in, ctx);
         int64_t n = db->count;
@@ -573,10 +587,13 @@ struct GpuCountSink : GpuLogicBase {
     int64_t tuples = 0;
     std::deque<Batch*> pending;
     GpuCountSink(Engine* e, int id, int dev) : eng(e), op_id(id) { device = dev; }
+    int64_t wait_us = 0;
     void drain_one() {
         Batch* b = pending.front();
         pending.pop_front();
+        int64_t t0 = now_us();
         gpu_resolve_count(b);
+        wait_us += now_us() - t0;
         tuples += b->count;
         release(b);
     }
@@ -587,6 +604,8 @@ struct GpuCountSink : GpuLogicBase {
     void on_eos(EmitCtx&, RuntimeCtx&) override {
         while (!pending.empty()) drain_one();
         if (stream) (void)hipStreamSynchronize(stream);
+        if (wfa_prof())
+            fprintf(stderr, "[prof] sink event-wait total: %ld us\n", (long)wait_us);
         eng->sink_tuples[op_id].fetch_add(tuples, std::memory_order_relaxed);
     }
 };
