@@ -42,6 +42,7 @@ ENGINE_SRCS = [
     "csrc/engine/core.cpp",
     "csrc/engine/engine.cpp",
     "csrc/engine/native_logic.cpp",
+    "csrc/engine/windows.cpp",
     "csrc/engine/gpu_ops.cpp",
     "csrc/engine/bindings.cpp",
 ]

@@ -220,6 +220,28 @@ static StageSpec make_stage(Engine& e, int id, const std::string& kind, const st
         e.sink_acc_i64[id].store(0);
         e.sink_tuples[id].store(0);
     }
+    if (kind.rfind("win_", 0) == 0 || kind == "interval_join") {
+        Engine* ep = &e;
+        WindowFn wf = nullptr;
+        if (!pyfn.is_none()) {
+            auto f = pyfn.cast<py::function>();
+            wf = [f](const WinRows& wr) -> double {
+                py::gil_scoped_acquire gil;
+                py::dict d;
+                d["ts"] = py::array_t<int64_t>({wr.n}, {(int64_t)8}, wr.ts, py::none());
+                d["key"] = wr.key;
+                d["gwid"] = wr.gwid;
+                for (size_t c = 0; c < wr.schema->payload.size(); ++c)
+                    d[py::str("c" + std::to_string(c))] =
+                        col_view(wr.schema->payload[c], (void*)wr.cols[c], wr.n);
+                return f(d).cast<double>();
+            };
+        }
+        st.factory = [kind, fp, ip, ep, id, wf] {
+            return make_window_logic(kind, fp, ip, ep, id, wf);
+        };
+        return st;
+    }
     if (kind.rfind("gpu_", 0) == 0) {
         Engine* ep = &e;
         Schema os = st.out_schema;

@@ -145,8 +145,16 @@ Batch* Collector::next() {
     return nullptr;
 }
 
+int64_t OrderingCollector::released_wm() const {
+    int64_t m = WM_MAX;
+    for (size_t c = 0; c < chans.size(); ++c)
+        if (open[c] || !pend[c].empty()) m = std::min(m, rel_wm[c]);
+    return m == WM_MAX ? chan_max_wm() : m;
+}
+
 Batch* OrderingCollector::next() {
     if (pend.empty()) pend.resize(chans.size());
+    if (rel_wm.empty()) rel_wm.assign(chans.size(), 0);
     int spins = 0;
     for (;;) {
         if (abort && abort->load(std::memory_order_relaxed)) return nullptr;
@@ -185,7 +193,8 @@ Batch* OrderingCollector::next() {
         if (best >= 0 && all_ready) {
             Batch* b = pend[best].front();
             pend[best].pop_front();
-            b->watermark = min_wm();
+            rel_wm[best] = std::max(rel_wm[best], b->watermark);
+            b->watermark = released_wm();
             if (b->punct && b->watermark <= last_fwd_wm) {
                 release(b);
                 continue;
@@ -209,7 +218,8 @@ Batch* OrderingCollector::next() {
             if (best < 0) return nullptr;
             Batch* b = pend[best].front();
             pend[best].pop_front();
-            b->watermark = min_wm();
+            rel_wm[best] = std::max(rel_wm[best], b->watermark);
+            b->watermark = released_wm();
             return b;
         }
         SpscQueue::backoff(spins);
@@ -272,7 +282,8 @@ struct ChainLogic::StageEmitter : Emitter {
     RuntimeCtx* rctx = nullptr;
     void emit(Batch* b) override { next->process(b, *next_ctx, *rctx); }
     void punct(int64_t wm) override {
-        for (auto* e : next_ctx->emitters) e->punct(wm);
+        if (!next->on_punct(wm, *next_ctx, *rctx))
+            for (auto* e : next_ctx->emitters) e->punct(wm);
     }
     void flush() override {}
     void eos() override {}
@@ -295,6 +306,12 @@ void ChainLogic::wire(const std::vector<Pool*>& pools, EmitCtx& final_ctx, Runti
 
 void ChainLogic::on_eos(EmitCtx&, RuntimeCtx& ctx) {
     for (size_t i = 0; i < stages.size(); ++i) stages[i]->on_eos(ctxs[i], ctx);
+}
+
+bool ChainLogic::on_punct(int64_t wm, EmitCtx&, RuntimeCtx& ctx) {
+    if (!stages[0]->on_punct(wm, ctxs[0], ctx))
+        for (auto* e : ctxs[0].emitters) e->punct(wm);
+    return true;
 }
 
 void ChainLogic::warm(RuntimeCtx& ctx) {
@@ -338,7 +355,8 @@ void Replica::run() {
             if (b->punct) {
                 int64_t wm = b->watermark;
                 release(b);
-                for (auto& e : emitters) e->punct(wm);
+                if (!logic->on_punct(wm, ectx, rctx))
+                    for (auto& e : emitters) e->punct(wm);
             } else {
                 logic->process(b, ectx, rctx);
             }

@@ -219,7 +219,12 @@ struct Collector {
 // before anything is released (reference: wf/ordering_collector.hpp).
 struct OrderingCollector : Collector {
     std::vector<std::deque<Batch*>> pend;
+    // wm of the last *released* batch per channel: a released batch must not
+    // carry watermark knowledge from batches still buffered in pend (that
+    // would fire downstream windows before their content arrives)
+    std::vector<int64_t> rel_wm;
     using Collector::Collector;
+    int64_t released_wm() const;
     Batch* next() override;
 };
 
@@ -245,6 +250,9 @@ struct OpLogic {
     virtual void warm(RuntimeCtx& ctx) {}
     // Take ownership of `in` (release or re-emit it).
     virtual void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx);
+    // Watermark-only punctuation: stateful logics (windows, joins) override
+    // to advance firing; return true to suppress default re-propagation.
+    virtual bool on_punct(int64_t wm, EmitCtx& out, RuntimeCtx& ctx) { return false; }
     virtual void on_eos(EmitCtx& out, RuntimeCtx& ctx) {}
     virtual bool is_source() const { return false; }
     // Source: fill-and-emit loop; return false when exhausted.
@@ -302,6 +310,7 @@ struct ChainLogic : OpLogic {
     void process(Batch* in, EmitCtx&, RuntimeCtx& ctx) override {
         stages.front()->process(in, ctxs[0], ctx);
     }
+    bool on_punct(int64_t wm, EmitCtx&, RuntimeCtx& ctx) override;
     void on_eos(EmitCtx&, RuntimeCtx& ctx) override;
     void warm(RuntimeCtx& ctx) override;
 };
@@ -371,6 +380,27 @@ std::shared_ptr<OpLogic> make_native_logic(const std::string& kind,
                                            const std::vector<double>& fparams,
                                            const std::vector<int64_t>& iparams,
                                            Engine* eng, int op_id);
+
+// ----- CPU window / join suite (windows.cpp) -----
+// Contiguous view over one window's rows, handed to user (Python) window
+// functions for the non-incremental path (reference wf/iterable.hpp).
+struct WinRows {
+    int64_t n = 0;
+    const int64_t* ts = nullptr;
+    uint64_t key = 0;
+    int64_t gwid = 0;
+    const Schema* schema = nullptr;
+    std::vector<const char*> cols;  // base pointer per payload column
+};
+using WindowFn = std::function<double(const WinRows&)>;
+
+// kinds: win_keyed, win_parallel, win_plq, win_wlq, win_mr_map,
+// win_mr_reduce, win_ffat, interval_join  (iparam layouts in windows.cpp)
+std::shared_ptr<OpLogic> make_window_logic(const std::string& kind,
+                                           const std::vector<double>& fp,
+                                           const std::vector<int64_t>& ip,
+                                           Engine* eng, int op_id,
+                                           WindowFn userfn = nullptr);
 
 // GPU logic factory (gpu_ops.cpp) — kinds: gpu_source/gpu_map/gpu_filter/
 // gpu_reduce/gpu_ffat/gpu_to_host/gpu_count_sink

@@ -1,0 +1,290 @@
+"""CPU window/join suite vs brute-force numpy oracles.
+
+Mirrors the reference's win_tests / join_tests differential strategy
+ but with exact per-window oracles, which the reference
+lacks: every operator's fired (key, value) multiset is compared against
+a plain recomputation over the generated stream.
+"""
+import math
+from collections import Counter, defaultdict
+
+import numpy as np
+import pytest
+
+import windflow_amd as wf
+from windflow_amd import native
+from windflow_amd.builders import (Keyed_Windows_Builder, Parallel_Windows_Builder,
+                                   Paned_Windows_Builder, MapReduce_Windows_Builder,
+                                   Ffat_Windows_Builder, Interval_Join_Builder,
+                                   Source_Builder, Sink_Builder)
+
+
+class Collect:
+    """Python sink callable collecting (ts, key, c0[, c1]) rows."""
+
+    def __init__(self):
+        self.rows = []
+
+    def __call__(self, cols):
+        n = len(cols['ts'])
+        cs = [cols[f'c{i}'] for i in range(8) if f'c{i}' in cols]
+        for i in range(n):
+            self.rows.append((int(cols['ts'][i]), int(cols['key'][i]),
+                              *[c[i].item() for c in cs]))
+
+
+def run_graph(win_op, stream_len=3000, n_keys=7, batch=128, src_par=1,
+              mode=wf.ExecutionMode.DEFAULT):
+    g = wf.PipeGraph("t", mode)
+    src = (Source_Builder(native.seq_source(stream_len, n_keys, batch))
+           .withParallelism(src_par).withOutputSchema([0])
+           .withOutputBatchSize(batch).build())
+    mp = g.add_source(src)
+    mp.add(win_op)
+    col = Collect()
+    snk = Sink_Builder(col).withParallelism(1).build()
+    mp.add_sink(snk)
+    g.run()
+    return col.rows
+
+
+def seq_stream(stream_len, n_keys):
+    """key -> list of (ts, value) in arrival order (v=1..N, key=v%K, ts=v)."""
+    per = defaultdict(list)
+    for v in range(1, stream_len + 1):
+        per[v % n_keys].append((v, v))
+    return per
+
+
+AGGS = {"sum": sum, "max": max, "min": min,
+        "count": len, "avg": lambda xs: sum(xs) / len(xs)}
+
+
+def oracle_cb(per, win, slide, agg="sum"):
+    """All windows incl. EOS partials: per key, window w = rows [w*s, w*s+win)."""
+    f = AGGS[agg]
+    out = Counter()
+    for k, rows in per.items():
+        vals = [v for _, v in rows]
+        w = 0
+        while w * slide < len(vals):
+            seg = vals[w * slide: w * slide + win]
+            out[(k, f(seg))] += 1
+            w += 1
+    return out
+
+
+def oracle_tb(per, win, slide, agg="sum"):
+    """Windows [wS, wS+win) on ts; from the first window containing the
+    key's first tuple up to the last containing its max ts."""
+    f = AGGS[agg]
+    out = Counter()
+    for k, rows in per.items():
+        ts = [t for t, _ in rows]
+        vals = {t: v for t, v in rows}
+        t0, tmax = ts[0], ts[-1]
+        w0 = max(0, -(-(t0 - win + 1) // slide))
+        for w in range(w0, tmax // slide + 1):
+            seg = [vals[t] for t in ts if w * slide <= t < w * slide + win]
+            out[(k, f(seg) if seg else 0)] += 1
+    return out
+
+
+def got_counter(rows):
+    return Counter((k, v) for _, k, v in rows)
+
+
+# ---------------- keyed windows ----------------
+@pytest.mark.parametrize("agg", ["sum", "max", "count"])
+def test_keyed_cb(agg):
+    win, slide = 40, 10
+    rows = run_graph(Keyed_Windows_Builder(func=(agg, 0)).withCBWindows(win, slide)
+                     .withOutputSchema([0]).build())
+    assert got_counter(rows) == oracle_cb(seq_stream(3000, 7), win, slide, agg)
+
+
+@pytest.mark.parametrize("agg", ["sum", "min"])
+def test_keyed_tb(agg):
+    win, slide = 100, 25
+    rows = run_graph(Keyed_Windows_Builder(func=(agg, 0)).withTBWindows(win, slide)
+                     .withOutputSchema([0]).build())
+    assert got_counter(rows) == oracle_tb(seq_stream(3000, 7), win, slide, agg)
+
+
+def test_keyed_cb_parallel_replicas():
+    win, slide = 30, 30  # tumbling
+    rows = run_graph(Keyed_Windows_Builder(func=("sum", 0))
+                     .withCBWindows(win, slide).withParallelism(3)
+                     .withOutputSchema([0]).build())
+    assert got_counter(rows) == oracle_cb(seq_stream(3000, 7), win, slide)
+
+
+def test_keyed_avg_f64():
+    win, slide = 50, 50
+    rows = run_graph(Keyed_Windows_Builder(func=("avg", 0)).withCBWindows(win, slide)
+                     .withOutputSchema([1]).build())
+    exp = oracle_cb(seq_stream(3000, 7), win, slide, "avg")
+    got = Counter((k, round(v, 9)) for _, k, v in rows)
+    assert got == Counter({(k, round(v, 9)): c for (k, v), c in exp.items()})
+
+
+# ---------------- python (non-incremental) windows ----------------
+def test_py_window_fn_matches_native():
+    win, slide = 25, 5
+
+    def median(w):
+        return float(np.median(w['c0'])) if len(w['c0']) else 0.0
+
+    rows = run_graph(Keyed_Windows_Builder(func=median).withCBWindows(win, slide)
+                     .withOutputSchema([1]).build(), stream_len=1500)
+    per = seq_stream(1500, 7)
+    exp = Counter()
+    for k, r in per.items():
+        vals = [v for _, v in r]
+        w = 0
+        while w * slide < len(vals):
+            exp[(k, float(np.median(vals[w * slide: w * slide + win])))] += 1
+            w += 1
+    assert Counter((k, v) for _, k, v in rows) == exp
+
+
+def test_py_window_tb():
+    win, slide = 80, 20
+
+    def total(w):
+        return float(w['c0'].sum())
+
+    rows = run_graph(Keyed_Windows_Builder(func=total).withTBWindows(win, slide)
+                     .withOutputSchema([0]).build(), stream_len=2000)
+    assert got_counter(rows) == oracle_tb(seq_stream(2000, 7), win, slide, "sum")
+
+
+# ---------------- parallel / paned / mapreduce ----------------
+def test_parallel_windows_cb():
+    win, slide = 40, 10
+    rows = run_graph(Parallel_Windows_Builder(func=("sum", 0))
+                     .withCBWindows(win, slide).withParallelism(4)
+                     .withOutputSchema([0]).build())
+    assert got_counter(rows) == oracle_cb(seq_stream(3000, 7), win, slide)
+
+
+def test_parallel_windows_tb():
+    win, slide = 60, 15
+    rows = run_graph(Parallel_Windows_Builder(func=("max", 0))
+                     .withTBWindows(win, slide).withParallelism(3)
+                     .withOutputSchema([0]).build())
+    assert got_counter(rows) == oracle_tb(seq_stream(3000, 7), win, slide, "max")
+
+
+@pytest.mark.parametrize("par", [1, 3])
+def test_paned_windows_tb(par):
+    win, slide = 100, 20  # pane = 20
+    rows = run_graph(Paned_Windows_Builder(plq_func=("sum", 0))
+                     .withTBWindows(win, slide).withParallelism(par)
+                     .withOutputSchema([0]).build())
+    assert got_counter(rows) == oracle_tb(seq_stream(3000, 7), win, slide)
+
+
+def test_mapreduce_windows_cb():
+    win, slide = 40, 10
+    rows = run_graph(MapReduce_Windows_Builder(map_func=("sum", 0))
+                     .withCBWindows(win, slide).withParallelism(3)
+                     .withOutputSchema([0]).build())
+    assert got_counter(rows) == oracle_cb(seq_stream(3000, 7), win, slide)
+
+
+def test_mapreduce_windows_tb():
+    win, slide = 90, 30
+    rows = run_graph(MapReduce_Windows_Builder(map_func=("sum", 0))
+                     .withTBWindows(win, slide).withParallelism(2)
+                     .withOutputSchema([0]).build())
+    assert got_counter(rows) == oracle_tb(seq_stream(3000, 7), win, slide)
+
+
+# ---------------- FFAT (FlatFAT) ----------------
+@pytest.mark.parametrize("agg", ["sum", "max"])
+def test_ffat_cb_matches_keyed(agg):
+    win, slide = 64, 16
+    ff = (Ffat_Windows_Builder(comb=(agg, 0)).withCBWindows(win, slide)
+          .withOutputSchema([0]).build())
+    rows = run_graph(ff)
+    assert got_counter(rows) == oracle_cb(seq_stream(3000, 7), win, slide, agg)
+
+
+@pytest.mark.parametrize("agg", ["sum", "max"])
+def test_ffat_tb_matches_keyed(agg):
+    win, slide = 120, 30
+    ff = (Ffat_Windows_Builder(comb=(agg, 0)).withTBWindows(win, slide)
+          .withOutputSchema([0]).build())
+    rows = run_graph(ff)
+    assert got_counter(rows) == oracle_tb(seq_stream(3000, 7), win, slide, agg)
+
+
+def test_ffat_cb_nonpow2_win():
+    win, slide = 100, 10
+    ff = (Ffat_Windows_Builder(comb=("min", 0)).withCBWindows(win, slide)
+          .withOutputSchema([0]).build())
+    rows = run_graph(ff, stream_len=5000, n_keys=11)
+    assert got_counter(rows) == oracle_cb(seq_stream(5000, 11), win, slide, "min")
+
+
+# ---------------- interval join ----------------
+def join_graph(mode_builder, n=2000, keys=5, lower=-10, upper=10, batch=64,
+               mode=wf.ExecutionMode.DEFAULT):
+    g = wf.PipeGraph("j", mode)
+    srcA = (Source_Builder(native.seq_source(n, keys, batch))
+            .withParallelism(1).withOutputSchema([0]).withOutputBatchSize(batch)
+            .build())
+    srcB = (Source_Builder(native.seq_source(n, keys, batch, value_offset=100000))
+            .withParallelism(1).withOutputSchema([0]).withOutputBatchSize(batch)
+            .build())
+    mpA = g.add_source(srcA)
+    mpB = g.add_source(srcB)
+    mp = mpA.merge(mpB)
+    jb = Interval_Join_Builder().withBoundaries(lower, upper).withValueCols(0)
+    jb = mode_builder(jb)
+    mp.add(jb.withOutputSchema([0, 0]).build())
+    col = Collect()
+    mp.add_sink(Sink_Builder(col).withParallelism(1).build())
+    g.run()
+    return col.rows
+
+
+def oracle_join(n, keys, lower, upper):
+    """(a from A, b from B) with same key and b.ts - a.ts in [lower, upper].
+    A: v=1..n (ts=v, val=v); B: same ts, val=v+100000."""
+    out = Counter()
+    for v in range(1, n + 1):         # a
+        k = v % keys
+        for bts in range(v + lower, v + upper + 1):
+            if 1 <= bts <= n and bts % keys == k:
+                out[(k, v, bts + 100000)] += 1
+    return out
+
+
+def test_interval_join_kp():
+    rows = join_graph(lambda b: b.withKPMode(), n=1500)
+    got = Counter((k, a, b) for _, k, a, b in rows)
+    assert got == oracle_join(1500, 5, -10, 10)
+
+
+def test_interval_join_kp_parallel():
+    rows = join_graph(lambda b: b.withKPMode().withParallelism(3), n=1500)
+    got = Counter((k, a, b) for _, k, a, b in rows)
+    assert got == oracle_join(1500, 5, -10, 10)
+
+
+def test_interval_join_dp():
+    rows = join_graph(lambda b: b.withDPMode().withParallelism(3), n=1200)
+    got = Counter((k, a, b) for _, k, a, b in rows)
+    assert got == oracle_join(1200, 5, -10, 10)
+
+
+# ---------------- deterministic mode ----------------
+def test_windows_deterministic_mode():
+    win, slide = 40, 10
+    rows = run_graph(Keyed_Windows_Builder(func=("sum", 0))
+                     .withCBWindows(win, slide).withParallelism(2)
+                     .withOutputSchema([0]).build(),
+                     mode=wf.ExecutionMode.DETERMINISTIC)
+    assert got_counter(rows) == oracle_cb(seq_stream(3000, 7), win, slide)

@@ -180,7 +180,7 @@ class Interval_Join_Builder(_BasicBuilder):
 
     def __init__(self, func=None):
         super().__init__(func)
-        self._op.join = dict(mode=JoinMode.KP, lower=0, upper=0)
+        self._op.join = dict(mode=JoinMode.KP, lower=0, upper=0, colA=0, colB=0)
 
     def withBoundaries(self, lower_us, upper_us):
         self._op.join.update(lower=int(lower_us), upper=int(upper_us))
@@ -188,8 +188,19 @@ class Interval_Join_Builder(_BasicBuilder):
 
     def withKPMode(self):
         self._op.join['mode'] = JoinMode.KP
+        self._op.broadcast_input = False
         return self
 
     def withDPMode(self):
+        """Data-parallel join: both streams broadcast, each replica stores
+        its round-robin slice (reference interval_join.hpp DP +
+        join_collector.hpp)."""
         self._op.join['mode'] = JoinMode.DP
+        self._op.broadcast_input = True
+        return self
+
+    def withValueCols(self, colA, colB=None):
+        """Payload columns joined into the (c0=A, c1=B) output pair."""
+        self._op.join['colA'] = int(colA)
+        self._op.join['colB'] = int(colA if colB is None else colB)
         return self

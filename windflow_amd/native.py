@@ -51,3 +51,10 @@ def count_sink():
 def split_mod(col=0):
     """split branch = x % n_branches on i64 column."""
     return NativeLogic("split", "mod_i64", [], [col])
+
+
+def win_agg(op, col=0):
+    """Compiled incremental window combine for the window builders:
+    op in {'sum','min','max','count','avg'} over payload column `col`."""
+    assert op in ("sum", "min", "max", "count", "avg"), op
+    return NativeLogic("win_agg", op, [], [col])

@@ -20,7 +20,9 @@ class _Node:
         self.parallelism = op.parallelism
         self.device = op.device
         self.n_branches = op.n_branches
-        self.engine_id = None
+        self.engine_id_in = None     # engine op receiving this node's inputs
+        self.engine_id_out = None    # engine op producing this node's outputs
+        self.in_collector = None     # collector override for incoming edges
 
     @property
     def last(self):
@@ -30,9 +32,10 @@ class _Node:
 class MultiPipe:
     """A linear pipeline inside the PipeGraph (reference multipipe.hpp:96)."""
 
-    def __init__(self, graph, tails):
+    def __init__(self, graph, tails, tags=None):
         self.graph = graph
         self.tails = tails           # list of (node_idx, branch)
+        self.tags = tags or [0] * len(tails)   # join stream tag per tail
         self.closed = False
 
     # ---- internal ----
@@ -51,13 +54,18 @@ class MultiPipe:
             raise RuntimeError("MultiPipe already has a sink")
         routing = routing if routing is not None else self._routing_for(op)
         # key extraction runs fused into each upstream node before the shuffle
-        if routing == RoutingMode.KEYBY and op.key_extractor not in (None, 'carried'):
+        # (also for BROADCAST joins/windows — the key column must exist)
+        if (routing in (RoutingMode.KEYBY, RoutingMode.BROADCAST)
+                and op.key_extractor not in (None, 'carried')):
             for (n, _b) in self.tails:
                 g._chain_key_extract(g.nodes[n], op.key_extractor)
         node_idx = g._new_node(op)
-        for (n, b) in self.tails:
-            g.edges.append(dict(src=n, branch=b, dst=node_idx, routing=routing, tag=tag))
+        is_join = op.kind == "interval_join"
+        for i, (n, b) in enumerate(self.tails):
+            t = self.tags[i] if is_join else tag
+            g.edges.append(dict(src=n, branch=b, dst=node_idx, routing=routing, tag=t))
         self.tails = [(node_idx, 0)]
+        self.tags = [0]
         return node_idx
 
     # ---- public API (reference multipipe.hpp) ----
@@ -95,12 +103,16 @@ class MultiPipe:
 
     def merge(self, *others):
         """Merge this MultiPipe with others into one (reference
-        pipegraph.hpp:308-460 merge shapes)."""
+        pipegraph.hpp:308-460 merge shapes).  Tail order tags join
+        streams: this pipe's tails are stream A (tag 0), the first merged
+        pipe's are stream B (tag 1), ..."""
         tails = list(self.tails)
-        for o in others:
+        tags = [0] * len(self.tails)
+        for i, o in enumerate(others):
             tails += o.tails
+            tags += [i + 1] * len(o.tails)
             o.closed = True
-        return MultiPipe(self.graph, tails)
+        return MultiPipe(self.graph, tails, tags)
 
     def split(self, split_logic, n_branches):
         """Split into n branches by user logic (reference pipegraph.hpp:265).
@@ -214,15 +226,17 @@ class PipeGraph:
                       ExecutionMode.DETERMINISTIC: CollectorKind.ORDERING,
                       ExecutionMode.PROBABILISTIC: CollectorKind.KSLACK}[self.mode]
         for node in self.nodes:
-            node.engine_id = self._lower_node(e, node)
+            ids = self._lower_node(e, node)
+            if isinstance(ids, tuple):
+                node.engine_id_in, node.engine_id_out = ids
+            else:
+                node.engine_id_in = node.engine_id_out = ids
         # engine emitter order must follow branch order per source node
         for edge in sorted(self.edges, key=lambda d: (d['src'], d['branch'])):
-            ck = default_ck
-            dst_first = self.nodes[edge['dst']].ops[0]
-            if dst_first.kind == "interval_join":
-                ck = default_ck
-            e.add_edge(self.nodes[edge['src']].engine_id,
-                       self.nodes[edge['dst']].engine_id,
+            dst = self.nodes[edge['dst']]
+            ck = dst.in_collector if dst.in_collector is not None else default_ck
+            e.add_edge(self.nodes[edge['src']].engine_id_out,
+                       dst.engine_id_in,
                        edge['routing'], ck, edge['tag'])
         self.engine = e
         return e
