@@ -46,6 +46,39 @@ def build_ffat_graph(n_tuples, batch, n_keys, win, slide, rank, world, device):
     return g, snk
 
 
+def build_a2a_graph(n_tuples, batch, n_keys, rank, world, device, dist_cfg):
+    """Driver config #4: Map_GPU -> RCCL keyby all-to-all -> Reduce_GPU,
+    one rank per GPU."""
+    import windflow_amd as wf
+    from windflow_amd import native_gpu
+    from windflow_amd.builders_gpu import (Source_GPU_Builder, Map_GPU_Builder,
+                                           KeyBy_Exchange_GPU_Builder,
+                                           Reduce_GPU_Builder, Sink_GPU_Builder)
+    src = (Source_GPU_Builder(
+        native_gpu.gpu_source(n_tuples, n_keys, batch, vdt=2, seed=42 + rank))
+        .withOutputSchema([2]).withOutputBatchSize(batch)
+        .withDevice(device).build())
+    mp_ = (Map_GPU_Builder(native_gpu.gpu_affine_map(0, 2.0, 0.5, dtype=2))
+           .withOutputSchema([2]).withOutputBatchSize(batch)
+           .withDevice(device).build())
+    ex = (KeyBy_Exchange_GPU_Builder(native_gpu.gpu_keyby_exchange())
+          .withOutputSchema([2]).withOutputBatchSize(3 * batch)
+          .withDevice(device).build())
+    rd = (Reduce_GPU_Builder(
+        native_gpu.gpu_keyed_reduce(native_gpu.COMB_SUM, 0, 4 * n_keys))
+        .withOutputSchema([2]).withOutputBatchSize(3 * batch)
+        .withDevice(device).build())
+    snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).withDevice(device).build()
+    g = wf.PipeGraph("bench_a2a")
+    g.set_dist(*dist_cfg)
+    mp = g.add_source(src)
+    mp.chain(mp_)
+    mp.chain(ex)
+    mp.chain(rd)
+    mp.chain_sink(snk)
+    return g, snk
+
+
 def build_cpu_graph(n_tuples, batch):
     import windflow_amd as wf
     from windflow_amd import native
@@ -73,7 +106,7 @@ def main():
                     help="distinct keys per rank")
     ap.add_argument("--win", type=int, default=1000)
     ap.add_argument("--slide", type=int, default=100)
-    ap.add_argument("--config", choices=["ffat", "cpu"], default="ffat")
+    ap.add_argument("--config", choices=["ffat", "a2a", "cpu"], default="ffat")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -81,12 +114,20 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
     dist = None
     torch = None
-    if args.config == "ffat":
+    dist_cfg = None
+    if args.config in ("ffat", "a2a"):
         import torch  # noqa: F811
         if world > 1:
             import torch.distributed as dist  # noqa: F811
             dist.init_process_group("nccl", rank=rank, world_size=world)
             torch.cuda.set_device(local_rank)
+        if args.config == "a2a":
+            if world > 1:
+                from windflow_amd.dist import init_from_torch
+                dist_cfg = init_from_torch()
+            else:
+                from windflow_amd import _core
+                dist_cfg = (0, 1, _core.rccl_unique_id())
 
     def sync():
         if torch is not None and torch.cuda.is_available():
@@ -108,13 +149,17 @@ def main():
         dt = time.time() - t0
         n_gpus = 0
     else:
+        def builder(steps):
+            if args.config == "a2a":
+                return build_a2a_graph(steps * B, B, args.keys, rank, world,
+                                       local_rank, dist_cfg)
+            return build_ffat_graph(steps * B, B, args.keys, args.win,
+                                    args.slide, rank, world, local_rank)
         # warmup engine (also JIT-warms pools/streams/arena)
         if W > 0:
-            gw, _ = build_ffat_graph(W * B, B, args.keys, args.win, args.slide,
-                                     rank, world, local_rank)
+            gw, _ = builder(W)
             gw.run()
-        g, snk = build_ffat_graph(K * B, B, args.keys, args.win, args.slide,
-                                  rank, world, local_rank)
+        g, snk = builder(K)
         g.prepare()   # threads spawned + streams/pools warm, gated
         sync()
         barrier()
@@ -149,8 +194,9 @@ def main():
             "dtype": "bf16" if args.config == "ffat" else "int64",
             "data": "synthetic",
             "config": {
-                "model": "keyed_ffat_cb_window" if args.config == "ffat"
-                         else "cpu_source_map_filter_sink",
+                "model": {"ffat": "keyed_ffat_cb_window",
+                          "a2a": "map_gpu_rccl_keyby_reduce_gpu",
+                          "cpu": "cpu_source_map_filter_sink"}[args.config],
                 "global_batch": B * max(world, 1),
                 "win": args.win,
                 "slide": args.slide,

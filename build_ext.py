@@ -105,10 +105,12 @@ def build(force=False):
         except Exception:
             pass
         if torchlib:
-            hip_link = [f"-L{torchlib}", "-l:libamdhip64.so",
+            # same single-runtime rule applies to RCCL (torch bundles its own)
+            hip_link = [f"-L{torchlib}", "-l:libamdhip64.so", "-l:librccl.so",
                         f"-Wl,-rpath,{torchlib}"]
         else:
-            hip_link = [f"-L{ROCM}/lib", "-lamdhip64", f"-Wl,-rpath,{ROCM}/lib"]
+            hip_link = [f"-L{ROCM}/lib", "-lamdhip64", "-lrccl",
+                        f"-Wl,-rpath,{ROCM}/lib"]
         link = ["g++", "-shared", "-o", out] + objs + hip_link + ["-pthread"]
         subprocess.check_call(link)
     return out

@@ -305,12 +305,23 @@ PYBIND11_MODULE(_core, m) {
         .value("U16", DType::U16)
         .value("U8", DType::U8);
 
+    m.def("rccl_unique_id",
+          [] { return py::bytes(wfa_rccl_unique_id()); },
+          "ncclUniqueId bytes for Engine.set_dist (call on rank 0, broadcast)");
+
     py::class_<Engine>(m, "Engine")
         .def(py::init<>())
         .def_readwrite("mode", &Engine::mode)
         .def_readwrite("time_policy", &Engine::time_policy)
         .def_readwrite("queue_capacity", &Engine::queue_capacity)
         .def_readwrite("pin_threads", &Engine::pin_threads)
+        .def("set_dist",
+             [](Engine& e, int rank, int world, py::bytes id) {
+                 e.dist_rank = rank;
+                 e.dist_world = world;
+                 e.rccl_id = std::string(id);
+             },
+             py::arg("rank"), py::arg("world"), py::arg("rccl_id"))
         .def("add_op",
              [](Engine& e, const std::string& name, int par, const std::string& kind,
                 const std::string& spec, std::vector<double> fp, std::vector<int64_t> ip,

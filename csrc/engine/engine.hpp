@@ -360,6 +360,10 @@ struct Engine {
     std::unordered_map<int, std::atomic<int64_t>> sink_tuples;
     int64_t queue_capacity = 128;   // batches per SPSC queue
     bool pin_threads = false;
+    // distributed (one process per GPU, RCCL over xGMI): set before build()
+    int dist_rank = 0;
+    int dist_world = 1;
+    std::string rccl_id;            // ncclUniqueId bytes (broadcast by rank 0)
 
     Pool* make_pool(const Schema& s, int64_t cap, bool pinned) {
         pools.push_back(std::make_unique<Pool>(s, cap, pinned));
@@ -409,5 +413,9 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
                                         const std::vector<int64_t>& ip, Engine* eng,
                                         int op_id, int device, const Schema& os,
                                         int64_t out_batch);
+
+// RCCL bootstrap: rank 0 generates the id, broadcasts it out-of-band
+// (torch.distributed store), every rank passes it to Engine::rccl_id.
+std::string wfa_rccl_unique_id();
 
 }  // namespace wfa

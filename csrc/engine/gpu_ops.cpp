@@ -18,6 +18,7 @@
 
 #ifdef WFA_WITH_HIP
 #include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
 
 #include "../hip/wfa_kernels.h"
 
@@ -610,6 +611,243 @@ struct GpuCountSink : GpuLogicBase {
     }
 };
 
+// ===== RCCL keyby exchange: the inter-GPU shuffle =====
+// Replaces the reference's KeyBy_Emitter_GPU host-staged re-batching
+// (keyby_emitter_gpu.hpp:399-638) with an MI355X-native design: rows are
+// bucketed on-device by hash(key) % world, gathered into a contiguous
+// per-destination send layout, and exchanged with a grouped RCCL
+// send/recv all-to-allv over xGMI.  Watermarks ride in the metadata
+// allgather and min-fold across ranks (the collector rule).
+// One replica per rank; every rank executes the same collective sequence
+// (EOS rounds with zero payload keep lagging ranks in lockstep).
+#define NCCLCHK(x)                                                              \
+    do {                                                                        \
+        ncclResult_t r_ = (x);                                                  \
+        if (r_ != ncclSuccess)                                                  \
+            throw std::runtime_error(std::string("RCCL error: ") +              \
+                                     ncclGetErrorString(r_) + " at " #x);       \
+    } while (0)
+
+static ncclComm_t get_rccl_comm(const std::string& id, int rank, int world,
+                                int device) {
+    static std::mutex mu;
+    static std::map<std::string, ncclComm_t> cache;
+    std::lock_guard<std::mutex> g(mu);
+    auto key = id + "/" + std::to_string(rank);
+    auto it = cache.find(key);
+    if (it != cache.end()) return it->second;
+    if (id.size() != sizeof(ncclUniqueId))
+        throw std::runtime_error("rccl_id not set (PipeGraph.set_dist) or wrong size");
+    ncclUniqueId uid;
+    memcpy(&uid, id.data(), sizeof(uid));
+    HIPCHK(hipSetDevice(device));
+    ncclComm_t comm;
+    NCCLCHK(ncclCommInitRank(&comm, world, uid, rank));
+    cache[key] = comm;
+    return comm;
+}
+
+std::string wfa_rccl_unique_id() {
+    ncclUniqueId id;
+    NCCLCHK(ncclGetUniqueId(&id));
+    return std::string((char*)&id, sizeof(id));
+}
+
+struct GpuExchangeLogic : GpuLogicBase {
+    Engine* eng;
+    int rank = 0, world = 1;
+    ncclComm_t comm = nullptr;
+    int bits = 1;
+    // scratch
+    uint32_t *dest = nullptr, *idx = nullptr, *dest_t = nullptr, *idx_t = nullptr;
+    uint32_t* hist = nullptr;
+    uint32_t* d_counts = nullptr;
+    int64_t* d_meta = nullptr;      // [world+2] send metadata
+    int64_t* d_meta_all = nullptr;  // [world*(world+2)] allgathered
+    int64_t* h_meta = nullptr;      // pinned mirror of d_meta_all
+    uint32_t* h_counts = nullptr;   // pinned send counts
+    void** d_colptrs = nullptr;
+    int* d_esize = nullptr;
+    std::unique_ptr<Pool> send_pool;
+    int64_t cur_wm = 0;
+
+    GpuExchangeLogic(Engine* e, int dev, Schema os, int64_t cap) : eng(e) {
+        device = dev;
+        out_schema = std::move(os);
+        out_cap = cap;
+        rank = e->dist_rank;
+        world = e->dist_world;
+        bits = 1;
+        while ((1 << bits) < world) ++bits;
+    }
+    void init_device() override {
+        comm = get_rccl_comm(eng->rccl_id, rank, world, device);
+        auto& A = arena(device);
+        dest = (uint32_t*)A.get(4 * out_cap);
+        idx = (uint32_t*)A.get(4 * out_cap);
+        dest_t = (uint32_t*)A.get(4 * out_cap);
+        idx_t = (uint32_t*)A.get(4 * out_cap);
+        hist = (uint32_t*)A.get(4 * (16 * (out_cap / 2048 + 2) + 16));
+        d_counts = (uint32_t*)A.get(4 * world + 64);
+        d_meta = (int64_t*)A.get(8 * (world + 2));
+        d_meta_all = (int64_t*)A.get(8 * world * (world + 2));
+        size_t nc = out_schema.payload.size();
+        d_colptrs = (void**)A.get(16 * (nc + 1));
+        d_esize = (int*)A.get(4 * (nc + 1));
+        std::vector<int> es;
+        for (auto d : out_schema.payload) es.push_back((int)dsize(d));
+        HIPCHK(hipMemcpy(d_esize, es.data(), 4 * nc, hipMemcpyHostToDevice));
+        HIPCHK(hipHostMalloc((void**)&h_meta, 8 * world * (world + 2),
+                             hipHostMallocDefault));
+        HIPCHK(hipHostMalloc((void**)&h_counts, 4 * world, hipHostMallocDefault));
+        send_pool = std::make_unique<Pool>(out_schema, out_cap, false);
+        send_pool->loc = Loc::DEVICE;
+        send_pool->device = device;
+    }
+    ~GpuExchangeLogic() override {
+        if (h_meta) (void)hipHostFree(h_meta);
+        if (h_counts) (void)hipHostFree(h_counts);
+    }
+
+    // one collective round: exchange (counts, wm, done) then rows.
+    // Returns true when every rank reported done.
+    bool round(Batch* send, const int64_t* scnt, int64_t wm, bool done,
+               EmitCtx& out, RuntimeCtx& ctx) {
+        const size_t nc = out_schema.payload.size();
+        // metadata allgather
+        int64_t meta[10 + 2];  // world <= 8 in practice; heap-free fast path
+        std::vector<int64_t> meta_v;
+        int64_t* m = meta;
+        if (world + 2 > 12) {
+            meta_v.resize(world + 2);
+            m = meta_v.data();
+        }
+        for (int p = 0; p < world; ++p) m[p] = scnt ? scnt[p] : 0;
+        m[world] = done ? WM_MAX : wm;
+        m[world + 1] = done ? 1 : 0;
+        HIPCHK(hipMemcpyAsync(d_meta, m, 8 * (world + 2), hipMemcpyHostToDevice,
+                              stream));
+        NCCLCHK(ncclAllGather(d_meta, d_meta_all, world + 2, ncclInt64, comm, stream));
+        HIPCHK(hipMemcpyAsync(h_meta, d_meta_all, 8 * world * (world + 2),
+                              hipMemcpyDeviceToHost, stream));
+        HIPCHK(hipStreamSynchronize(stream));
+        // recv layout: rows from rank p land at roff[p]
+        int64_t roff[9] = {0};
+        int64_t total = 0;
+        bool all_done = true;
+        int64_t min_wm = WM_MAX;
+        for (int p = 0; p < world; ++p) {
+            const int64_t* mp = h_meta + (int64_t)p * (world + 2);
+            roff[p] = total;
+            total += mp[rank];
+            if (mp[world + 1] == 0) all_done = false;
+            min_wm = std::min(min_wm, mp[world]);
+        }
+        if (total > out_cap)
+            throw std::runtime_error("exchange recv overflow: raise out_batch");
+        Batch* rb = nullptr;
+        if (total > 0 || send) {
+            rb = get_dev();
+            // send offsets: prefix of scnt
+            int64_t soff[9] = {0};
+            for (int p = 1; p < world; ++p)
+                soff[p] = soff[p - 1] + (scnt ? scnt[p - 1] : 0);
+            NCCLCHK(ncclGroupStart());
+            for (int p = 0; p < world; ++p) {
+                int64_t sc = scnt ? scnt[p] : 0;
+                int64_t rc = h_meta[(int64_t)p * (world + 2) + rank];
+                if (sc) {
+                    NCCLCHK(ncclSend(send->ts + soff[p], 8 * sc, ncclChar, p, comm, stream));
+                    NCCLCHK(ncclSend(send->key + soff[p], 8 * sc, ncclChar, p, comm, stream));
+                    for (size_t c = 0; c < nc; ++c) {
+                        size_t es = dsize(out_schema.payload[c]);
+                        NCCLCHK(ncclSend((char*)send->cols[c] + es * soff[p], es * sc,
+                                         ncclChar, p, comm, stream));
+                    }
+                }
+                if (rc) {
+                    NCCLCHK(ncclRecv(rb->ts + roff[p], 8 * rc, ncclChar, p, comm, stream));
+                    NCCLCHK(ncclRecv(rb->key + roff[p], 8 * rc, ncclChar, p, comm, stream));
+                    for (size_t c = 0; c < nc; ++c) {
+                        size_t es = dsize(out_schema.payload[c]);
+                        NCCLCHK(ncclRecv((char*)rb->cols[c] + es * roff[p], es * rc,
+                                         ncclChar, p, comm, stream));
+                    }
+                }
+            }
+            NCCLCHK(ncclGroupEnd());
+            HIPCHK(hipStreamSynchronize(stream));
+        }
+        if (ctx.stats) ctx.stats->num_kernels++;
+        int64_t out_wm = min_wm == WM_MAX ? cur_wm : min_wm;
+        if (rb) {
+            if (total > 0) {
+                rb->count = total;
+                rb->watermark = out_wm;
+                record_ready(rb);
+                out.emit(rb);
+            } else {
+                release(rb);
+                for (auto* e : out.emitters) e->punct(out_wm);
+            }
+        } else if (out_wm > cur_wm) {
+            for (auto* e : out.emitters) e->punct(out_wm);
+        }
+        cur_wm = std::max(cur_wm, out_wm);
+        return all_done;
+    }
+
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        Batch* db = input_on_device(in, ctx);
+        int64_t n = db->count;
+        int64_t wm = db->watermark;
+        if (n > out_cap) throw std::runtime_error("exchange input > out_batch");
+        Batch* sb = send_pool->get();
+        if (sb->ready_event)
+            HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)sb->ready_event, 0));
+        // bucket -> stable sort by dest -> contiguous per-dest send layout
+        wfa_bucket_by_key(stream, db->key, n, world, dest);
+        wfa_iota_u32(stream, idx, n);
+        wfa_count_u32(stream, dest, n, d_counts, world);
+        uint32_t *od, *oi;
+        wfa_sort_pairs(stream, dest, idx, dest_t, idx_t, hist, n, bits, &od, &oi);
+        size_t nc = db->cols.size();
+        std::vector<void*> ptrs(2 * nc);
+        for (size_t c = 0; c < nc; ++c) {
+            ptrs[c] = db->cols[c];
+            ptrs[nc + c] = sb->cols[c];
+        }
+        HIPCHK(hipMemcpyAsync(d_colptrs, ptrs.data(), 8 * 2 * nc,
+                              hipMemcpyHostToDevice, stream));
+        wfa_gather_rows(stream, oi, n, db->ts, sb->ts, db->key, sb->key,
+                        (const void* const*)d_colptrs, (void* const*)(d_colptrs + nc),
+                        d_esize, (int)nc);
+        HIPCHK(hipMemcpyAsync(h_counts, d_counts, 4 * world, hipMemcpyDeviceToHost,
+                              stream));
+        HIPCHK(hipStreamSynchronize(stream));
+        release_after_use(db);
+        int64_t scnt[9];
+        for (int p = 0; p < world; ++p) scnt[p] = h_counts[p];
+        if (ctx.stats) ctx.stats->num_kernels += 4;
+        round(sb, scnt, wm, false, out, ctx);
+        // the collective copied out of sb synchronously; safe to recycle
+        release(sb);
+    }
+
+    // Puncts are NOT collective-safe (their count differs per rank and
+    // would desynchronize the round sequence): swallow them — watermarks
+    // cross ranks inside the data/EOS rounds' metadata allgather.
+    bool on_punct(int64_t, EmitCtx&, RuntimeCtx&) override { return true; }
+
+    void on_eos(EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        // EOS rounds: keep matching other ranks' collectives until all done
+        while (!round(nullptr, nullptr, cur_wm, true, out, ctx)) {
+        }
+    }
+};
+
 std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::string& spec,
                                         const std::vector<double>& fp,
                                         const std::vector<int64_t>& ip, Engine* eng,
@@ -634,6 +872,8 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
         // ip: [comb, vcol, win, slide, max_keys, use_tree]
         return std::make_shared<GpuFfatLogic>((int)ip[0], (int)ip[1], ip[2], ip[3],
                                               ip[4], ip[5] != 0, device, os, out_batch);
+    if (kind == "gpu_exchange")
+        return std::make_shared<GpuExchangeLogic>(eng, device, os, out_batch);
     if (kind == "gpu_to_host")
         return std::make_shared<GpuToHostLogic>(device);
     if (kind == "gpu_count_sink") {
@@ -652,6 +892,7 @@ namespace wfa {
 Batch* gpu_alloc_batch(Pool&) { throw std::runtime_error("built without HIP"); }
 void gpu_free_batch(Batch*) {}
 void gpu_resolve_count(Batch*) {}
+std::string wfa_rccl_unique_id() { throw std::runtime_error("built without HIP"); }
 std::shared_ptr<OpLogic> make_gpu_logic(const std::string&, const std::string&,
                                         const std::vector<double>&,
                                         const std::vector<int64_t>&, Engine*, int, int,

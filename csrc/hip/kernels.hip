@@ -342,3 +342,52 @@ void wfa_bucket_by_key(wfa_stream_t s, const uint64_t* key, int64_t n, int world
     hipLaunchKernelGGL(k_bucket, dim3(nblk(n)), dim3(WFA_THREADS), 0, (hipStream_t)s, key, n, world, dest_out);
 }
 }
+
+// ===== small-value histogram (all-to-all dest counts) =====
+__global__ void k_count_u32(const uint32_t* v, int64_t n, uint32_t* counts) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * (int64_t)blockDim.x)
+        atomicAdd(&counts[v[i]], 1u);
+}
+
+// ===== full-row gather by permutation (ts + key + payload columns) =====
+// cols table layout: [in_0..in_{nc-1}, out_0..out_{nc-1}] device pointers.
+__global__ void k_gather_rows(const uint32_t* idx, int64_t n, const int64_t* ts_in,
+                              int64_t* ts_out, const uint64_t* key_in,
+                              uint64_t* key_out, const void* const* cols_in,
+                              void* const* cols_out, const int* esize, int nc) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * (int64_t)blockDim.x) {
+        uint32_t j = idx[i];
+        ts_out[i] = ts_in[j];
+        key_out[i] = key_in[j];
+        for (int c = 0; c < nc; ++c) {
+            int es = esize[c];
+            const char* src = (const char*)cols_in[c] + (int64_t)j * es;
+            char* dst = (char*)cols_out[c] + i * es;
+            switch (es) {
+                case 8: *(uint64_t*)dst = *(const uint64_t*)src; break;
+                case 4: *(uint32_t*)dst = *(const uint32_t*)src; break;
+                case 2: *(uint16_t*)dst = *(const uint16_t*)src; break;
+                default: for (int k = 0; k < es; ++k) dst[k] = src[k];
+            }
+        }
+    }
+}
+
+extern "C" {
+void wfa_count_u32(wfa_stream_t s, const uint32_t* v, int64_t n, uint32_t* counts,
+                   int n_bins) {
+    hipMemsetAsync(counts, 0, 4 * n_bins, (hipStream_t)s);
+    hipLaunchKernelGGL(k_count_u32, dim3(nblk(n)), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, v, n, counts);
+}
+void wfa_gather_rows(wfa_stream_t s, const uint32_t* idx, int64_t n,
+                     const int64_t* ts_in, int64_t* ts_out, const uint64_t* key_in,
+                     uint64_t* key_out, const void* const* cols_in,
+                     void* const* cols_out, const int* esize, int nc) {
+    hipLaunchKernelGGL(k_gather_rows, dim3(nblk(n)), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, idx, n, ts_in, ts_out, key_in, key_out,
+                       cols_in, cols_out, esize, nc);
+}
+}

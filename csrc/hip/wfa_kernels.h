@@ -130,4 +130,14 @@ void wfa_cast(wfa_stream_t s, const void* in, int dt_in, void* out, int dt_out,
 void wfa_bucket_by_key(wfa_stream_t s, const uint64_t* key, int64_t n, int world,
                        uint32_t* dest_out);
 
+// histogram of small u32 values (per-destination row counts)
+void wfa_count_u32(wfa_stream_t s, const uint32_t* v, int64_t n, uint32_t* counts,
+                   int n_bins);
+
+// full-row gather by permutation: ts, key and every payload column
+void wfa_gather_rows(wfa_stream_t s, const uint32_t* idx, int64_t n,
+                     const int64_t* ts_in, int64_t* ts_out, const uint64_t* key_in,
+                     uint64_t* key_out, const void* const* cols_in,
+                     void* const* cols_out, const int* esize, int nc);
+
 }  // extern "C"

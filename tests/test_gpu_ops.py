@@ -197,6 +197,63 @@ def test_gpu_ffat_cb_min_tree_vs_oracle():
             assert abs(a - b) <= 1e-5 * max(1.0, abs(b))
 
 
+def test_gpu_keyby_exchange_world1():
+    """RCCL self-exchange (world=1): every row routes back to rank 0, so the
+    pipeline is value-preserving; exercises the full bucket->sort->gather->
+    allgather->send/recv path on device."""
+    from windflow_amd import _core
+    from windflow_amd.builders_gpu import KeyBy_Exchange_GPU_Builder
+    n, n_keys, b = 100_000, 97, 25_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=0))
+           .withOutputSchema([0]).withOutputBatchSize(b).build())
+    ex = (KeyBy_Exchange_GPU_Builder(native_gpu.gpu_keyby_exchange())
+          .withOutputSchema([0]).withOutputBatchSize(2 * b).build())
+    g = wf.PipeGraph("a2a")
+    g.set_dist(0, 1, _core.rccl_unique_id())
+    p = g.add_source(src)
+    p.chain(ex)
+    snk = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+    snk.out_schema = [0]
+    p.add_sink(snk)
+    g.run()
+    _, _, val = gen_batch(n, 0, 42, n_keys, 0)
+    assert g.sink_sum(snk) == int(val.sum())
+    assert g.sink_count(snk) == n
+
+
+def test_gpu_exchange_then_reduce():
+    """config #4 shape on one rank: map_gpu -> keyby exchange -> reduce_gpu."""
+    from windflow_amd import _core
+    from windflow_amd.builders_gpu import KeyBy_Exchange_GPU_Builder
+    n, n_keys, b = 100_000, 64, 25_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=2))
+           .withOutputSchema([2]).withOutputBatchSize(b).build())
+    mp_ = (Map_GPU_Builder(native_gpu.gpu_affine_map(0, 2.0, 0.0, dtype=2))
+           .withOutputSchema([2]).withOutputBatchSize(b).build())
+    ex = (KeyBy_Exchange_GPU_Builder(native_gpu.gpu_keyby_exchange())
+          .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
+    rd = (Reduce_GPU_Builder(native_gpu.gpu_keyed_reduce(native_gpu.COMB_SUM, 0, 256))
+          .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
+    res = dict(s=0.0)
+
+    def pysink(cols):
+        res['s'] += float(cols['c0'].astype(np.float64).sum())
+
+    g = wf.PipeGraph("a2a_red")
+    g.set_dist(0, 1, _core.rccl_unique_id())
+    p = g.add_source(src)
+    p.chain(mp_)
+    p.chain(ex)
+    p.chain(rd)
+    snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+    snk.out_schema = [2]
+    p.add_sink(snk)
+    g.run()
+    _, _, val = gen_batch(n, 0, 42, n_keys, 2)
+    ref = float((val.astype(np.float64) * 2.0).sum())
+    assert abs(res['s'] - ref) <= 2e-3 * max(1.0, abs(ref))
+
+
 def test_gpu_reduce_keyed_sum():
     n, n_keys, b = 100_000, 64, 25_000
     src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=2))

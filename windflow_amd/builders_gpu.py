@@ -42,6 +42,14 @@ class Sink_GPU_Builder(_GpuBuilder):
     _kind = "gpu_count_sink"
 
 
+class KeyBy_Exchange_GPU_Builder(_GpuBuilder):
+    """MI355X extension: inter-GPU keyby shuffle (RCCL all-to-allv over
+    xGMI).  The reference has no multi-GPU shuffle at all; this is the
+    MI355X-native replacement for KeyBy_Emitter_GPU's host re-batching
+    (reference keyby_emitter_gpu.hpp:594-638) across GPUs."""
+    _kind = "gpu_exchange"
+
+
 class Ffat_Windows_GPU_Builder(_GpuBuilder):
     """reference builders_gpu.hpp:466 (+withNumWinPerBatch :576)."""
     _kind = "gpu_ffat"

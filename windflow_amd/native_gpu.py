@@ -46,6 +46,13 @@ def gpu_ffat_windows(comb=COMB_SUM, col=0, win=1000, slide=100,
                        [comb, col, win, slide, max_keys, 1 if use_tree else 0])
 
 
+def gpu_keyby_exchange():
+    """Inter-GPU keyby shuffle: rows bucketed by hash(key) % world on device,
+    exchanged with a grouped RCCL send/recv all-to-allv over xGMI.  One
+    replica per rank; requires PipeGraph.set_dist(rank, world, rccl_id)."""
+    return NativeLogic("gpu_exchange", "", [], [])
+
+
 def gpu_count_sink():
     return NativeLogic("gpu_count_sink", "", [], [])
 

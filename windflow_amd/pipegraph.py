@@ -147,6 +147,14 @@ class PipeGraph:
         self.engine = None
         self._sink_map = {}          # sink descriptor id -> engine op id
         self._t0 = self._t1 = None
+        self._dist = None            # (rank, world, rccl_id bytes)
+
+    def set_dist(self, rank, world, rccl_id):
+        """Configure the RCCL communicator for gpu_keyby_exchange stages.
+        rccl_id: bytes from _core.rccl_unique_id() on rank 0, broadcast via
+        torch.distributed (see windflow_amd.dist.init_from_torch)."""
+        self._dist = (int(rank), int(world), rccl_id)
+        return self
 
     # ---- construction ----
     def add_source(self, op):
@@ -222,6 +230,8 @@ class PipeGraph:
         e = _core.Engine()
         e.mode = self.mode
         e.time_policy = self.time_policy
+        if self._dist is not None:
+            e.set_dist(*self._dist)
         default_ck = {ExecutionMode.DEFAULT: CollectorKind.WATERMARK,
                       ExecutionMode.DETERMINISTIC: CollectorKind.ORDERING,
                       ExecutionMode.PROBABILISTIC: CollectorKind.KSLACK}[self.mode]
