@@ -184,7 +184,7 @@ def main():
             "metric": "tuples_per_sec",
             "value": value,
             "unit": "tuples/s",
-            "n_gpus": n_gpus if args.config == "ffat" else 0,
+            "n_gpus": 0 if args.config == "cpu" else n_gpus,
             "steps": K,
             "warmup": W,
             "ms_per_step": dt / K * 1000.0,

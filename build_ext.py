@@ -43,6 +43,7 @@ ENGINE_SRCS = [
     "csrc/engine/engine.cpp",
     "csrc/engine/native_logic.cpp",
     "csrc/engine/windows.cpp",
+    "csrc/engine/persist.cpp",
     "csrc/engine/gpu_ops.cpp",
     "csrc/engine/bindings.cpp",
 ]

@@ -220,6 +220,13 @@ static StageSpec make_stage(Engine& e, int id, const std::string& kind, const st
         e.sink_acc_i64[id].store(0);
         e.sink_tuples[id].store(0);
     }
+    if (kind.rfind("p_", 0) == 0) {
+        Engine* ep = &e;
+        st.factory = [kind, spec, fp, ip, ep, id] {
+            return make_persist_logic(kind, spec, fp, ip, ep, id);
+        };
+        return st;
+    }
     if (kind.rfind("win_", 0) == 0 || kind == "interval_join") {
         Engine* ep = &e;
         WindowFn wf = nullptr;
@@ -420,4 +427,31 @@ PYBIND11_MODULE(_core, m) {
         });
 
     m.def("hash_key", [](uint64_t k) { return KeyByEmitter::mix(k); });
+
+    // persistent keyed state handle for Python P_* logic (reference
+    // DBHandle<T>: get/modify/put with user serialize/deserialize)
+    struct StateStore {
+        std::shared_ptr<void> holder;
+        void* kv = nullptr;
+        void* cache = nullptr;
+    };
+    py::class_<StateStore>(m, "StateStore")
+        .def(py::init([](const std::string& path, int64_t cache_cap) {
+                 auto s = new StateStore();
+                 s->holder = open_state_store(path, cache_cap, &s->kv, &s->cache);
+                 return s;
+             }),
+             py::arg("path"), py::arg("cache_capacity") = 1 << 16)
+        .def("get",
+             [](StateStore& s, uint64_t key) -> py::object {
+                 std::string* v = state_cache_get(s.cache, key);
+                 if (!v) return py::none();
+                 return py::bytes(*v);
+             })
+        .def("put",
+             [](StateStore& s, uint64_t key, py::bytes val) {
+                 state_cache_put(s.cache, key, std::string(val));
+             })
+        .def("flush", [](StateStore& s) { state_cache_flush(s.cache); })
+        .def("__len__", [](StateStore& s) { return state_kv_size(s.kv); });
 }

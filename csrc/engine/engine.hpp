@@ -414,6 +414,19 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
                                         int op_id, int device, const Schema& os,
                                         int64_t out_batch);
 
+// persistent state tier (persist.cpp) — kinds: p_reduce
+std::shared_ptr<OpLogic> make_persist_logic(const std::string& kind,
+                                            const std::string& spec,
+                                            const std::vector<double>& fp,
+                                            const std::vector<int64_t>& ip,
+                                            Engine* eng, int op_id);
+std::shared_ptr<void> open_state_store(const std::string& path, int64_t cache_cap,
+                                       void** kv_out, void** cache_out);
+std::string* state_cache_get(void* cache, uint64_t key);
+void state_cache_put(void* cache, uint64_t key, const std::string& v);
+void state_cache_flush(void* cache);
+int64_t state_kv_size(void* kv);
+
 // RCCL bootstrap: rank 0 generates the id, broadcasts it out-of-band
 // (torch.distributed store), every rank passes it to Engine::rccl_id.
 std::string wfa_rccl_unique_id();

@@ -138,10 +138,16 @@ class PipeGraph:
     """reference pipegraph.hpp:74."""
 
     def __init__(self, name="app", mode=ExecutionMode.DEFAULT,
-                 time_policy=TimePolicy.EVENT_TIME):
+                 time_policy=TimePolicy.EVENT_TIME, tracing=None):
         self.name = name
         self.mode = mode
         self.time_policy = time_policy
+        # tracing: None -> honor WF_TRACING env (runtime knob replacing the
+        # reference's compile-time -DWF_TRACING_ENABLED)
+        import os as _os
+        self.tracing = (bool(_os.environ.get("WF_TRACING"))
+                        if tracing is None else bool(tracing))
+        self._monitor = None
         self.nodes = []
         self.edges = []
         self.engine = None
@@ -254,9 +260,19 @@ class PipeGraph:
     # ---- execution (reference pipegraph.hpp:610-739) ----
     def run(self):
         e = self.build_engine()
+        if self.tracing:
+            from .monitoring import MonitoringThread
+            self._monitor = MonitoringThread(self).start()
         self._t0 = time.time()
         e.run()
         self._t1 = time.time()
+        if self._monitor:
+            self._monitor.stop()
+            import os as _os
+            logdir = _os.environ.get("WF_LOG_DIR")
+            if logdir:
+                from .monitoring import dump_stats
+                dump_stats(self, _os.path.join(logdir, f"{self.name}.json"))
         for node in self.nodes:
             for op in node.ops:
                 if op.closing:
