@@ -1,0 +1,141 @@
+"""Kafka integration (reference wf/kafka/: Kafka_Source :125, Kafka_Sink :71,
+builders_kafka.hpp) + the generic external-connector layer it builds on.
+
+The reference consumes via librdkafka's KafkaConsumer::consume(idleTime)
+loop inside the source replica and produces via producer->produce per
+serialized tuple.  Here:
+
+ - Connector_Source_Builder: any Python iterator/callable yielding column
+   dicts becomes a source replica loop — the engine-facing contract the
+   Kafka source plugs into (also usable for files, sockets, ...).
+ - Kafka_Source_Builder / Kafka_Sink_Builder: same fluent surface as the
+   reference (withBrokers/withTopics/withGroupID/withIdleness/withOffsets);
+   they require `confluent_kafka` (librdkafka) at build() time and raise a
+   clear error when the client library is absent (this image has none).
+
+The user deserializer mirrors the reference signature (kafka_source.hpp:271):
+  deser(msg_value: bytes, shipper: dict-appender) -> bool  (False = stop)
+"""
+from .operators import Operator
+from .builders import _BasicBuilder
+
+
+def _have_kafka():
+    try:
+        import confluent_kafka  # noqa: F401
+        return True
+    except ImportError:
+        return False
+
+
+class Connector_Source_Builder(_BasicBuilder):
+    """Source from an external iterator: fn(replica, parallelism) -> column
+    dict or None (end of stream) — the engine's PySourceLogic contract."""
+    _kind = "source"
+
+
+class Kafka_Source_Builder(_BasicBuilder):
+    """reference builders_kafka.hpp:191-258."""
+    _kind = "source"
+
+    def __init__(self, deser):
+        super().__init__(None)
+        self._deser = deser
+        self._cfg = dict(brokers="localhost:9092", topics=[], group="wf",
+                         idle_ms=100, offsets=None, policy="roundrobin")
+
+    def withBrokers(self, brokers):
+        self._cfg["brokers"] = brokers
+        return self
+
+    def withTopics(self, *topics):
+        self._cfg["topics"] = list(topics)
+        return self
+
+    def withGroupID(self, gid):
+        self._cfg["group"] = gid
+        return self
+
+    def withAssignmentPolicy(self, policy):
+        self._cfg["policy"] = policy
+        return self
+
+    def withIdleness(self, ms):
+        self._cfg["idle_ms"] = int(ms)
+        return self
+
+    def withOffsets(self, offsets):
+        """list of (topic, partition, offset) (reference withOffsets)."""
+        self._cfg["offsets"] = list(offsets)
+        return self
+
+    def build(self):
+        if not _have_kafka():
+            raise RuntimeError(
+                "Kafka_Source requires confluent_kafka (librdkafka); not "
+                "installed in this environment")
+        from confluent_kafka import Consumer, TopicPartition
+        cfg, deser = self._cfg, self._deser
+
+        def source_fn(replica, parallelism):
+            if not hasattr(source_fn, "_consumer"):
+                c = Consumer({
+                    "bootstrap.servers": cfg["brokers"],
+                    "group.id": cfg["group"],
+                    "partition.assignment.strategy": cfg["policy"],
+                    "auto.offset.reset": "earliest",
+                })
+                if cfg["offsets"]:
+                    c.assign([TopicPartition(t, p, o)
+                              for (t, p, o) in cfg["offsets"]])
+                else:
+                    c.subscribe(cfg["topics"])
+                source_fn._consumer = c
+            msg = source_fn._consumer.poll(cfg["idle_ms"] / 1000.0)
+            out = {}
+            more = deser(msg.value() if msg is not None and not msg.error()
+                         else None, out)
+            if not more:
+                source_fn._consumer.close()
+                return None
+            return out if out else {}
+
+        op = self._op.clone()
+        op.logic = source_fn
+        return op
+
+
+class Kafka_Sink_Builder(_BasicBuilder):
+    """reference kafka_sink.hpp:71: user serializer returns
+    (topic, partition, payload bytes) per row."""
+    _kind = "sink"
+
+    def __init__(self, ser):
+        super().__init__(None)
+        self._ser = ser
+        self._cfg = dict(brokers="localhost:9092")
+
+    def withBrokers(self, brokers):
+        self._cfg["brokers"] = brokers
+        return self
+
+    def build(self):
+        if not _have_kafka():
+            raise RuntimeError(
+                "Kafka_Sink requires confluent_kafka (librdkafka); not "
+                "installed in this environment")
+        from confluent_kafka import Producer
+        cfg, ser = self._cfg, self._ser
+        prod = Producer({"bootstrap.servers": cfg["brokers"]})
+
+        def sink_fn(cols):
+            n = len(cols["ts"])
+            for i in range(n):
+                topic, part, payload = ser(cols, i)
+                prod.produce(topic, payload, partition=part)
+            prod.poll(0)
+
+        sink_fn.on_eos = prod.flush
+        op = self._op.clone()
+        op.logic = sink_fn
+        return op
