@@ -467,11 +467,14 @@ struct GpuReduceLogic : GpuLogicBase {
     }
 };
 
-// ===== Ffat_Windows_GPU: keyed CB sliding window =====
+// ===== Ffat_Windows_GPU: keyed CB/TB sliding window =====
 struct GpuFfatLogic : GpuLogicBase {
     int comb, vcol;
     int64_t win, slide, max_keys;
     bool use_tree;
+    bool tb = false;          // time-based (event-time) windows
+    int64_t lateness = 0;
+    int pend_log2 = 16;       // TB pending-pane ring size
     int64_t pane_len, P, S;
     int ring_log2;
     KeyedScratch ks;
@@ -483,10 +486,20 @@ struct GpuFfatLogic : GpuLogicBase {
     uint32_t* st_head = nullptr;
     float* st_wsum = nullptr;
     int64_t* d_on = nullptr;
+    // TB state
+    float* tb_pend = nullptr;
+    int64_t* tb_base = nullptr;
+    int64_t* tb_last = nullptr;
+    uint32_t* tb_flags = nullptr;  // [ignored, overflow]
+    uint32_t* tb_nf = nullptr;
+    uint32_t* h_flags = nullptr;   // pinned readback
+    int64_t batches = 0;
 
     GpuFfatLogic(int comb_, int vc, int64_t w, int64_t sl, int64_t mk, bool tree,
-                 int dev, Schema os, int64_t cap) {
+                 int dev, Schema os, int64_t cap, bool tb_, int64_t lat, int plog2) {
         comb = comb_; vcol = vc; win = w; slide = sl; max_keys = mk; use_tree = tree;
+        tb = tb_; lateness = lat;
+        if (plog2 > 0) pend_log2 = plog2;
         device = dev;
         out_schema = os;  // payload [F32]
         out_cap = cap;
@@ -495,6 +508,8 @@ struct GpuFfatLogic : GpuLogicBase {
         S = slide / pane_len;
         ring_log2 = 1;
         while ((1ll << ring_log2) < P + 2) ++ring_log2;
+        if (tb && use_tree)
+            throw std::runtime_error("gpu_ffat: TB windows use the ring path");
     }
     void init_device() override {
         ks.alloc(device, out_cap, max_keys, stream);
@@ -515,13 +530,43 @@ struct GpuFfatLogic : GpuLogicBase {
         wfa_fill_f32(stream, st_acc, ident, max_keys);
         wfa_fill_f32(stream, ring_or_tree, ident,
                      max_keys * (use_tree ? 2 * (1ll << ring_log2) : (1ll << ring_log2)));
+        if (tb) {
+            int64_t Rp = 1ll << pend_log2;
+            tb_pend = (float*)A.get(4 * max_keys * Rp);
+            tb_base = (int64_t*)A.get(8 * max_keys);
+            tb_last = (int64_t*)A.get(8 * max_keys);
+            tb_flags = (uint32_t*)A.get(64);
+            tb_nf = (uint32_t*)A.get(4 * (max_keys + 1));
+            wfa_fill_f32(stream, tb_pend, ident, max_keys * Rp);
+            wfa_fill_u64(stream, (uint64_t*)tb_base, (uint64_t)-1ll, max_keys);
+            wfa_fill_u64(stream, (uint64_t*)tb_last, (uint64_t)-1ll, max_keys);
+            HIPCHK(hipMemsetAsync(tb_flags, 0, 64, stream));
+            HIPCHK(hipHostMalloc((void**)&h_flags, 64, hipHostMallocDefault));
+        }
     }
+    ~GpuFfatLogic() override {
+        if (h_flags) (void)hipHostFree(h_flags);
+    }
+
+    void check_tb_flags() {
+        HIPCHK(hipMemcpyAsync(h_flags, tb_flags, 8, hipMemcpyDeviceToHost, stream));
+        HIPCHK(hipStreamSynchronize(stream));
+        if (h_flags[1])
+            throw std::runtime_error(
+                "gpu_ffat TB pending-pane ring overflow: raise pend_ring_log2 "
+                "or lateness is too large for the configured ring");
+    }
+
     void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
-        int64_t t0 = now_us();
         ensure_init();
         Batch* db = input_on_device(in, ctx);
         int64_t n = db->count;
         ks.group(stream, db, vcol, ctx);
+        if (tb) {
+            tb_round_with_count(db, n, db->watermark, out, ctx);
+            release(db);
+            return;
+        }
         Batch* ob = get_dev();
         // deterministic output offsets (no atomic cursor)
         uint32_t* nf = (uint32_t*)ks.v_sorted;  // reuse freed scratch
@@ -546,6 +591,41 @@ struct GpuFfatLogic : GpuLogicBase {
         release(db);
         record_ready(ob);
         out.emit(ob);
+    }
+
+    void tb_round_with_count(Batch* db, int64_t n, int64_t wm, EmitCtx& out,
+                             RuntimeCtx& ctx) {
+        Batch* ob = get_dev();
+        int64_t limit = (wm - lateness) / pane_len - 1;
+        wfa_ffat_tb_round(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
+                          ks.v_as_f32, ks.idx_sorted, db ? db->ts : nullptr,
+                          pane_len, P, S, comb, ring_log2, pend_log2, limit,
+                          tb_pend, tb_base, tb_last, st_head, st_wsum,
+                          ring_or_tree, ks.d_nslots, ks.slot_to_key, tb_nf,
+                          tb_flags, tb_flags + 1, ob->key, (float*)ob->cols[0],
+                          ob->ts, ob->capacity, d_on);
+        HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
+        ob->count = -1;
+        ob->watermark = wm - lateness;
+        if (ctx.stats) ctx.stats->num_kernels += 4;
+        record_ready(ob);
+        out.emit(ob);
+        if ((++batches & 63) == 0) check_tb_flags();
+    }
+
+    bool on_punct(int64_t wm, EmitCtx& out, RuntimeCtx& ctx) override {
+        if (!tb) return false;
+        ensure_init();
+        tb_round_with_count(nullptr, 0, wm, out, ctx);
+        return false;
+    }
+
+    void on_eos(EmitCtx& out, RuntimeCtx& ctx) override {
+        if (!tb) return;
+        ensure_init();
+        // complete every remaining data pane (wm -> +inf)
+        tb_round_with_count(nullptr, 0, INT64_MAX / 4, out, ctx);
+        check_tb_flags();
     }
 };
 
@@ -869,9 +949,12 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
         return std::make_shared<GpuReduceLogic>((int)ip[0], (int)ip[1], ip[2], device,
                                                 os, out_batch);
     if (kind == "gpu_ffat")
-        // ip: [comb, vcol, win, slide, max_keys, use_tree]
-        return std::make_shared<GpuFfatLogic>((int)ip[0], (int)ip[1], ip[2], ip[3],
-                                              ip[4], ip[5] != 0, device, os, out_batch);
+        // ip: [comb, vcol, win, slide, max_keys, use_tree,
+        //      wintype(0 CB/1 TB), lateness, pend_ring_log2]
+        return std::make_shared<GpuFfatLogic>(
+            (int)ip[0], (int)ip[1], ip[2], ip[3], ip[4], ip[5] != 0, device, os,
+            out_batch, ip.size() > 6 && ip[6] != 0, ip.size() > 7 ? ip[7] : 0,
+            ip.size() > 8 ? (int)ip[8] : 0);
     if (kind == "gpu_exchange")
         return std::make_shared<GpuExchangeLogic>(eng, device, os, out_batch);
     if (kind == "gpu_to_host")

@@ -635,6 +635,196 @@ extern "C" void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                            slot_to_key, fire_base, out_key, out_val, out_ts, out_cap);
 }
 
+// ===== time-based (TB) keyed windows: pane lift + watermark advance =====
+// Event-time redesign of the reference's TB path (ffat_replica_gpu.hpp:
+// 869-1047: lift -> thrust (key,pane) reduce -> PendingPanes_Queue ->
+// per-key FlatFAT): tuples accumulate into a per-slot circular pending-pane
+// buffer (pane id = ts / pane_len, absolute); the watermark completes panes
+// up to limit = (wm - lateness)/pane_len - 1; completed panes (gaps =
+// identity) advance the same ring/wsum window machine as the CB fold.
+// Keys are aligned to the absolute window grid at first touch (pend_base =
+// first window containing the first tuple), matching the CPU FfatCpuLogic.
+
+// pass 1: per segment, accumulate tuples into the pending ring
+__global__ void k_tb_lift(const uint32_t* seg_start, const uint32_t* seg_slot,
+                          const int64_t* d_nseg, int64_t n, const float* v_f32,
+                          const uint32_t* idx_sorted, const int64_t* ts_orig,
+                          int64_t pane_len, int64_t P, int64_t S, int comb,
+                          int pend_log2, float* pend, int64_t* pend_base,
+                          int64_t* last_pane, uint32_t* ignored,
+                          uint32_t* overflow) {
+    const int64_t nseg = *d_nseg;
+    const uint32_t Rp = 1u << pend_log2;
+    const uint32_t Pm = Rp - 1;
+    for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nseg;
+         j += gridDim.x * (int64_t)blockDim.x) {
+        const uint32_t slot = seg_slot[j];
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        float* pd = pend + (size_t)slot * Rp;
+        int64_t base = pend_base[slot];
+        int64_t lastp = last_pane[slot];
+        uint32_t ign = 0;
+        for (int64_t i = seg_start[j]; i < e; ++i) {
+            const int64_t ts = ts_orig[idx_sorted[i]];
+            const int64_t p = ts / pane_len;
+            if (base < 0) {
+                // first tuple of this key: align to the absolute window grid
+                int64_t w0 = ts - P * pane_len + 1;
+                w0 = w0 <= 0 ? 0 : (w0 + S * pane_len - 1) / (S * pane_len);
+                base = w0 * S;
+            }
+            if (p < base) {
+                ++ign;  // late beyond completed panes
+                continue;
+            }
+            if (p - base >= (int64_t)Rp) {
+                atomicAdd(overflow, 1u);
+                continue;
+            }
+            const float x = v_f32[idx_sorted[i]];
+            float* cell = &pd[(uint64_t)p & Pm];
+            *cell = (comb == 0) ? *cell + x
+                                : (comb == 1 ? fminf(*cell, x) : fmaxf(*cell, x));
+            if (p > lastp) lastp = p;
+        }
+        pend_base[slot] = base;
+        last_pane[slot] = lastp;
+        if (ign) atomicAdd(ignored, ign);
+    }
+}
+
+// pass 2: fires per SLOT (watermark advances every key, batch or not)
+__global__ void k_tb_count(const uint32_t* n_slots, int64_t limit_pane,
+                           const int64_t* pend_base, const int64_t* last_pane,
+                           const uint32_t* st_head, int64_t P, int64_t S,
+                           uint32_t* nf) {
+    const int64_t ns = *n_slots;
+    for (int64_t s = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; s < ns;
+         s += gridDim.x * (int64_t)blockDim.x) {
+        int64_t base = pend_base[s];
+        int64_t ncomp = 0;
+        if (base >= 0) {
+            int64_t hi = min(limit_pane, last_pane[s]);
+            ncomp = hi >= base ? hi - base + 1 : 0;
+        }
+        auto F = [&](uint64_t x) {
+            return x < (uint64_t)P ? 0ull : (x - (uint64_t)P) / (uint64_t)S + 1ull;
+        };
+        const uint64_t h0 = st_head[s];
+        nf[s] = (uint32_t)(F(h0 + (uint64_t)ncomp) - F(h0));
+    }
+}
+
+__global__ void k_tb_scan(uint32_t* nf, const uint32_t* n_slots, int64_t* d_out_n) {
+    const int64_t m = *n_slots;
+    __shared__ uint32_t tot[1024];
+    const int64_t chunk = (m + 1023) / 1024;
+    const int64_t b0 = (int64_t)threadIdx.x * chunk;
+    const int64_t b1 = min(m, b0 + chunk);
+    uint32_t s = 0;
+    for (int64_t i = b0; i < b1; ++i) s += nf[i];
+    tot[threadIdx.x] = s;
+    __syncthreads();
+    for (int off = 1; off < 1024; off <<= 1) {
+        uint32_t t = (threadIdx.x >= off) ? tot[threadIdx.x - off] : 0;
+        __syncthreads();
+        tot[threadIdx.x] += t;
+        __syncthreads();
+    }
+    uint32_t run = tot[threadIdx.x] - s;
+    for (int64_t i = b0; i < b1; ++i) {
+        uint32_t v = nf[i];
+        nf[i] = run;
+        run += v;
+    }
+    if (threadIdx.x == 1023 && d_out_n) *d_out_n = tot[1023];
+}
+
+// pass 3: per slot, insert completed panes into the window ring and fire
+__global__ void k_tb_advance(const uint32_t* n_slots, int64_t limit_pane,
+                             int64_t pane_len, int64_t P, int64_t S, int comb,
+                             int ring_log2, int pend_log2, float* pend,
+                             int64_t* pend_base, const int64_t* last_pane,
+                             uint32_t* st_head, float* st_wsum, float* ring,
+                             const uint64_t* slot_to_key, const uint32_t* nf,
+                             uint64_t* out_key, float* out_val, int64_t* out_ts,
+                             int64_t out_cap) {
+    const int64_t ns = *n_slots;
+    const uint32_t R = 1u << ring_log2;
+    const uint32_t Rm = R - 1;
+    const uint32_t Rp = 1u << pend_log2;
+    const uint32_t Pm = Rp - 1;
+    const float ident = (comb == 0) ? 0.f : (comb == 1 ? INFINITY : -INFINITY);
+    for (int64_t s = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; s < ns;
+         s += gridDim.x * (int64_t)blockDim.x) {
+        int64_t base = pend_base[s];
+        if (base < 0) continue;
+        const int64_t hi = min(limit_pane, last_pane[s]);
+        if (hi < base) continue;
+        float* pd = pend + (size_t)s * Rp;
+        float* rg = ring + (size_t)s * R;
+        uint32_t head = st_head[s];
+        float wsum = st_wsum[s];
+        int64_t w = nf[s];
+        for (int64_t q = base; q <= hi; ++q) {
+            float pane = pd[(uint64_t)q & Pm];
+            pd[(uint64_t)q & Pm] = ident;  // consumed
+            rg[head & Rm] = pane;
+            if (comb == 0) {
+                wsum += pane;
+                if (head >= (uint32_t)P) wsum -= rg[(head - (uint32_t)P) & Rm];
+            }
+            ++head;
+            if (head >= (uint32_t)P && ((head - (uint32_t)P) % (uint32_t)S) == 0) {
+                float res;
+                if (comb == 0) {
+                    res = wsum;
+                } else {
+                    res = rg[(head - 1) & Rm];
+                    for (uint32_t q2 = 2; q2 <= (uint32_t)P; ++q2) {
+                        float pv = rg[(head - q2) & Rm];
+                        res = (comb == 1) ? fminf(res, pv) : fmaxf(res, pv);
+                    }
+                }
+                if (w < out_cap) {
+                    out_key[w] = slot_to_key[s];
+                    out_val[w] = res;
+                    out_ts[w] = (q + 1) * pane_len - 1;  // window end - 1
+                }
+                ++w;
+            }
+        }
+        pend_base[s] = hi + 1;
+        st_head[s] = head;
+        st_wsum[s] = wsum;
+    }
+}
+
+extern "C" void wfa_ffat_tb_round(
+    wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
+    const int64_t* d_nseg, int64_t n, const float* v_f32,
+    const uint32_t* idx_sorted, const int64_t* ts_orig, int64_t pane_len,
+    int64_t P, int64_t S, int comb, int ring_log2, int pend_log2,
+    int64_t limit_pane, float* pend, int64_t* pend_base, int64_t* last_pane,
+    uint32_t* st_head, float* st_wsum, float* ring, const uint32_t* n_slots,
+    const uint64_t* slot_to_key, uint32_t* nf, uint32_t* ignored,
+    uint32_t* overflow, uint64_t* out_key, float* out_val, int64_t* out_ts,
+    int64_t out_cap, int64_t* d_out_n) {
+    hipStream_t st = (hipStream_t)s;
+    if (n > 0)
+        hipLaunchKernelGGL(k_tb_lift, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
+                           st, seg_start, seg_slot, d_nseg, n, v_f32, idx_sorted,
+                           ts_orig, pane_len, P, S, comb, pend_log2, pend, pend_base,
+                           last_pane, ignored, overflow);
+    hipLaunchKernelGGL(k_tb_count, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0, st,
+                       n_slots, limit_pane, pend_base, last_pane, st_head, P, S, nf);
+    hipLaunchKernelGGL(k_tb_scan, dim3(1), dim3(1024), 0, st, nf, n_slots, d_out_n);
+    hipLaunchKernelGGL(k_tb_advance, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
+                       st, n_slots, limit_pane, pane_len, P, S, comb, ring_log2,
+                       pend_log2, pend, pend_base, last_pane, st_head, st_wsum, ring,
+                       slot_to_key, nf, out_key, out_val, out_ts, out_cap);
+}
+
 // ===== FlatFAT-tree fold: O(log R) window query for large P =====
 // Per-slot complete binary tree over R = 2^ring_log2 circular pane leaves
 // (tree[slot*2R + node], node 1 = root, leaves at R..2R-1).  On each pane:

@@ -102,6 +102,22 @@ void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                       uint64_t* out_key, float* out_val, int64_t* out_ts,
                       int64_t out_cap);
 
+// ----- time-based keyed windows: pane lift + watermark-driven advance -----
+// One round = lift the batch's tuples into per-slot pending pane partials
+// (pane id = ts / pane_len, absolute; pend ring of 2^pend_log2 panes), then
+// complete every slot's panes up to limit_pane and fire windows.  Call with
+// n == 0 (and null segment args) for a pure watermark/EOS advance.
+void wfa_ffat_tb_round(
+    wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
+    const int64_t* d_nseg, int64_t n, const float* v_f32,
+    const uint32_t* idx_sorted, const int64_t* ts_orig, int64_t pane_len,
+    int64_t P, int64_t S, int comb, int ring_log2, int pend_log2,
+    int64_t limit_pane, float* pend, int64_t* pend_base, int64_t* last_pane,
+    uint32_t* st_head, float* st_wsum, float* ring, const uint32_t* n_slots,
+    const uint64_t* slot_to_key, uint32_t* nf, uint32_t* ignored,
+    uint32_t* overflow, uint64_t* out_key, float* out_val, int64_t* out_ts,
+    int64_t out_cap, int64_t* d_out_n);
+
 // ----- FlatFAT arena path (non-invertible combines over many panes) -----
 // Per-slot complete binary tree over ring of 2^ring_log2 pane leaves,
 // stored slot-major: tree[slot * 2*R + node].  Incremental leaf update +

@@ -197,6 +197,63 @@ def test_gpu_ffat_cb_min_tree_vs_oracle():
             assert abs(a - b) <= 1e-5 * max(1.0, abs(b))
 
 
+def test_gpu_ffat_tb_vs_oracle():
+    """Event-time (TB) GPU windows vs brute-force oracle: windows
+    [w*slide, w*slide+win) on ts, aligned at the first window containing
+    each key's first tuple, fired once their last pane holds data <= wm."""
+    n, n_keys, win, slide = 120_000, 101, 400, 100  # pane = 100
+    b = 17_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=2))
+           .withOutputSchema([2]).withOutputBatchSize(b).build())
+    ff = (Ffat_Windows_GPU_Builder(
+        native_gpu.gpu_ffat_windows(native_gpu.COMB_SUM, 0, win, slide,
+                                    max_keys=1024, tb=True, pend_ring_log2=10))
+          .withOutputSchema([2]).withOutputBatchSize(b).build())
+    res = dict(rows=[])
+
+    def pysink(cols):
+        res['rows'].append((cols['key'].copy(), cols['c0'].copy(),
+                            cols['ts'].copy()))
+
+    g = wf.PipeGraph("ffattb")
+    p = g.add_source(src)
+    p.chain(ff)
+    snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+    snk.out_schema = [2]
+    p.add_sink(snk)
+    g.run()
+    # oracle
+    ts, key, val = gen_batch(n, 0, 42, n_keys, 2)
+    from collections import defaultdict
+    per_t = defaultdict(list)
+    for t, k, v in zip(ts.tolist(), key.tolist(), val.tolist()):
+        per_t[k].append((t, np.float32(v)))
+    pane = int(np.gcd(win, slide))
+    exp = defaultdict(list)
+    for k, rows in per_t.items():
+        tss = [t for t, _ in rows]
+        t0, tmax = min(tss), max(tss)
+        w0 = max(0, -(-(t0 - win + 1) // slide))
+        w = w0
+        while (w * slide + win - 1) // pane <= tmax // pane:
+            s = float(np.sum(np.array(
+                [v for t, v in rows if w * slide <= t < w * slide + win],
+                dtype=np.float32), dtype=np.float64))
+            exp[k].append((w * slide + win - 1, s))
+            w += 1
+    got = defaultdict(list)
+    for k_arr, v_arr, t_arr in res['rows']:
+        for k, v, t in zip(k_arr.tolist(), v_arr.tolist(), t_arr.tolist()):
+            got[k].append((t, v))
+    assert sum(len(v) for v in got.values()) == sum(len(v) for v in exp.values())
+    for k, pairs in got.items():
+        pairs.sort()
+        ref = sorted(exp[k])
+        for (t_g, v_g), (t_r, v_r) in zip(pairs, ref):
+            assert t_g == t_r, f"key {k}: ts {t_g} vs {t_r}"
+            assert abs(v_g - v_r) <= 1e-3 * max(1.0, abs(v_r)), (k, v_g, v_r)
+
+
 def test_gpu_keyby_exchange_world1():
     """RCCL self-exchange (world=1): every row routes back to rank 0, so the
     pipeline is value-preserving; exercises the full bucket->sort->gather->

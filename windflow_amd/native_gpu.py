@@ -38,12 +38,18 @@ def gpu_keyed_reduce(comb=COMB_SUM, col=0, max_keys=1 << 16):
 
 
 def gpu_ffat_windows(comb=COMB_SUM, col=0, win=1000, slide=100,
-                     max_keys=1 << 16, use_tree=False):
-    """Keyed CB sliding window over panes (pane = gcd(win, slide));
-    use_tree selects the FlatFAT-arena path (O(log) combine for large
-    win/slide ratios and non-invertible combines)."""
+                     max_keys=1 << 16, use_tree=False, tb=False, lateness=0,
+                     pend_ring_log2=0):
+    """Keyed sliding window over panes (pane = gcd(win, slide)).
+    CB (default): windows fire every `slide` tuples per key; use_tree
+    selects the FlatFAT-arena path (O(log) combine for large win/slide
+    ratios and non-invertible combines).
+    TB (tb=True): event-time windows [w*slide, w*slide+win) on ts; panes
+    complete at watermark - lateness; pending out-of-order panes live in a
+    2^pend_ring_log2 ring per key (default 2^16)."""
     return NativeLogic("gpu_ffat", "", [],
-                       [comb, col, win, slide, max_keys, 1 if use_tree else 0])
+                       [comb, col, win, slide, max_keys, 1 if use_tree else 0,
+                        1 if tb else 0, lateness, pend_ring_log2])
 
 
 def gpu_keyby_exchange():
