@@ -401,7 +401,8 @@ struct KeyedScratch {
     const float* v_as_f32 = nullptr;
 
     // sorts (slot, idx) pairs and fills segments; values stay unsorted and
-    // are read through idx_sorted (saves the gather round trip)
+    // are read through idx_sorted (saves the gather round trip).
+    // vcol < 0: no value cast (stateful map/filter operate in place)
     void group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx) {
         int64_t n = db->count;
         if (n > cap)
@@ -413,15 +414,120 @@ struct KeyedScratch {
         uint32_t *os, *oi;
         wfa_sort_pairs(s, slot, idx, slot_t, idx_t, hist, n, bits, &os, &oi);
         idx_sorted = oi;
-        int vdt = (int)db->schema.payload[vcol];
-        const void* vsrc = db->cols[vcol];
-        if (vdt != 2) {  // cast to f32 once (i64/bf16 lifted)
-            wfa_cast(s, vsrc, vdt, v_f32, 2, n);
-            vsrc = v_f32;
+        if (vcol >= 0) {
+            int vdt = (int)db->schema.payload[vcol];
+            const void* vsrc = db->cols[vcol];
+            if (vdt != 2) {  // cast to f32 once (i64/bf16 lifted)
+                wfa_cast(s, vsrc, vdt, v_f32, 2, n);
+                vsrc = v_f32;
+            }
+            v_as_f32 = (const float*)vsrc;
         }
-        v_as_f32 = (const float*)vsrc;
         wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg);
         if (ctx.stats) ctx.stats->num_kernels += 5 + 3 * ((bits + 3) / 4);
+    }
+};
+
+// ===== stateful Map_GPU / Filter_GPU: keyed device state =====
+// Reference map_gpu.hpp:106-295 keeps per-key state objects in a shared
+// tbb map with a spinlock serializing replicas; here state is a dense
+// per-slot arena owned by the replica, advanced in key order by
+// wfa_stateful_apply over the batch's segments.
+struct GpuStatefulMapLogic : GpuLogicBase {
+    int spec, col;
+    double a, b;
+    int64_t max_keys;
+    KeyedScratch ks;
+    double* d_state = nullptr;
+    GpuStatefulMapLogic(int sp, int c, double a_, double b_, int64_t mk, int dev,
+                        Schema os, int64_t cap) {
+        spec = sp; col = c; a = a_; b = b_; max_keys = mk;
+        device = dev;
+        out_schema = os;
+        out_cap = cap;
+    }
+    void init_device() override {
+        ks.alloc(device, out_cap, max_keys, stream);
+        d_state = (double*)arena(device).get(8 * max_keys);
+        HIPCHK(hipMemsetAsync(d_state, 0, 8 * max_keys, stream));
+    }
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        Batch* db = input_on_device(in, ctx);
+        ks.group(stream, db, -1, ctx);
+        wfa_stateful_apply(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, db->count,
+                           ks.idx_sorted, db->cols[col],
+                           (int)db->schema.payload[col], spec, 0, a, b, d_state,
+                           nullptr);
+        record_ready(db);
+        if (ctx.stats) ctx.stats->num_kernels += 1;
+        out.emit(db);
+    }
+};
+
+struct GpuStatefulFilterLogic : GpuLogicBase {
+    int spec, col;
+    double a, b;
+    int64_t max_keys;
+    KeyedScratch ks;
+    double* d_state = nullptr;
+    uint32_t* d_flags = nullptr;
+    uint32_t* d_scan = nullptr;
+    int64_t* d_cnt = nullptr;
+    void** d_colptrs = nullptr;
+    int* d_esize = nullptr;
+    GpuStatefulFilterLogic(int sp, int c, double a_, double b_, int64_t mk, int dev,
+                           Schema os, int64_t cap) {
+        spec = sp; col = c; a = a_; b = b_; max_keys = mk;
+        device = dev;
+        out_schema = os;
+        out_cap = cap;
+    }
+    void init_device() override {
+        ks.alloc(device, out_cap, max_keys, stream);
+        auto& A = arena(device);
+        d_state = (double*)A.get(8 * max_keys);
+        // dedup compares x != state: NaN init keeps every key's first tuple
+        HIPCHK(hipMemsetAsync(d_state, spec == 1 ? 0xFF : 0, 8 * max_keys, stream));
+        d_flags = (uint32_t*)A.get(4 * out_cap);
+        d_scan = (uint32_t*)A.get(4 * (out_cap / 2048 + 2));
+        d_cnt = (int64_t*)A.get(64);
+        size_t nc = out_schema.payload.size();
+        d_colptrs = (void**)A.get(16 * (nc + 1));
+        d_esize = (int*)A.get(4 * (nc + 1));
+        std::vector<int> es;
+        for (auto d : out_schema.payload) es.push_back((int)dsize(d));
+        HIPCHK(hipMemcpy(d_esize, es.data(), 4 * nc, hipMemcpyHostToDevice));
+    }
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        Batch* db = input_on_device(in, ctx);
+        int64_t n = db->count;
+        ks.group(stream, db, -1, ctx);
+        wfa_stateful_apply(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
+                           ks.idx_sorted, db->cols[col],
+                           (int)db->schema.payload[col], spec, 1, a, b, d_state,
+                           d_flags);
+        Batch* ob = get_dev();
+        size_t nc = db->cols.size();
+        std::vector<void*> ptrs(2 * nc);
+        for (size_t c = 0; c < nc; ++c) {
+            ptrs[c] = db->cols[c];
+            ptrs[nc + c] = ob->cols[c];
+        }
+        HIPCHK(hipMemcpyAsync(d_colptrs, ptrs.data(), 8 * 2 * nc,
+                              hipMemcpyHostToDevice, stream));
+        wfa_compact(stream, n, d_flags, d_scan, db->ts, ob->ts, db->key, ob->key,
+                    (const void* const*)d_colptrs, (void* const*)(d_colptrs + nc),
+                    d_esize, (int)nc, d_cnt);
+        HIPCHK(hipMemcpyAsync(ob->lazy_count, d_cnt, 8, hipMemcpyDeviceToHost, stream));
+        ob->count = -1;
+        ob->watermark = db->watermark;
+        ob->stream_tag = db->stream_tag;
+        if (ctx.stats) ctx.stats->num_kernels += 4;
+        release_after_use(db);
+        record_ready(ob);
+        out.emit(ob);
     }
 };
 
@@ -944,6 +1050,15 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
     if (kind == "gpu_filter")
         return std::make_shared<GpuFilterLogic>((int)ip[0], (int)ip[1], fp[0], fp[1],
                                                 device, os, out_batch);
+    if (kind == "gpu_map_keyed")
+        // ip: [spec, col, max_keys]; fp: [a, b]
+        return std::make_shared<GpuStatefulMapLogic>((int)ip[0], (int)ip[1], fp[0],
+                                                     fp[1], ip[2], device, os,
+                                                     out_batch);
+    if (kind == "gpu_filter_keyed")
+        return std::make_shared<GpuStatefulFilterLogic>((int)ip[0], (int)ip[1], fp[0],
+                                                        fp[1], ip[2], device, os,
+                                                        out_batch);
     if (kind == "gpu_reduce")
         // ip: [comb, vcol, max_keys]
         return std::make_shared<GpuReduceLogic>((int)ip[0], (int)ip[1], ip[2], device,

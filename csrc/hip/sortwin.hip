@@ -635,6 +635,65 @@ extern "C" void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                            slot_to_key, fire_base, out_key, out_val, out_ts, out_cap);
 }
 
+// ===== stateful map/filter: per-key device state, key-order walk =====
+// Reference Stateful_MAPGPU_Kernel / Stateful_FILTERGPU_Kernel
+// (map_gpu.hpp:80-102, filter_gpu.hpp:92-115): one worker per distinct key
+// walks that key's tuples IN ORDER applying func(tuple, state).  There the
+// keyed states are objects in a shared tbb map guarded by a spinlock; here
+// state is a dense per-slot arena (f64) indexed by the hash-slot machinery,
+// one thread per segment, results written back to original row positions.
+// map specs:    1 running_sum (x=state+=x)  2 ema (state=a*state+(1-a)x, x=state)
+//               3 running_count (x=++state)
+// filter specs: 1 dedup_consecutive (keep if x != state; state=x)
+//               2 every_kth (keep when ++state % k == 0)
+__global__ void k_stateful(const uint32_t* seg_start, const uint32_t* seg_slot,
+                           const int64_t* d_nseg, int64_t n,
+                           const uint32_t* idx_sorted, void* col, int dt,
+                           int spec, int is_filter, double a, double b,
+                           double* state, uint32_t* flags) {
+    const int64_t nseg = *d_nseg;
+    for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nseg;
+         j += gridDim.x * (int64_t)blockDim.x) {
+        const uint32_t slot = seg_slot[j];
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        double st = state[slot];
+        for (int64_t i = seg_start[j]; i < e; ++i) {
+            const uint32_t r = idx_sorted[i];
+            double x = (dt == 0) ? (double)((int64_t*)col)[r]
+                                 : (double)((float*)col)[r];
+            if (!is_filter) {
+                switch (spec) {
+                    case 1: st += x; x = st; break;
+                    case 2: st = a * st + (1.0 - a) * x; x = st; break;
+                    case 3: st += 1.0; x = st; break;
+                }
+                if (dt == 0)
+                    ((int64_t*)col)[r] = (int64_t)x;
+                else
+                    ((float*)col)[r] = (float)x;
+            } else {
+                uint32_t keep = 1;
+                switch (spec) {
+                    case 1: keep = (x != st); st = x; break;
+                    case 2: st += 1.0; keep = ((int64_t)st % (int64_t)a) == 0; break;
+                }
+                flags[r] = keep;
+            }
+        }
+        state[slot] = st;
+    }
+}
+
+extern "C" void wfa_stateful_apply(wfa_stream_t s, const uint32_t* seg_start,
+                                   const uint32_t* seg_slot, const int64_t* d_nseg,
+                                   int64_t n, const uint32_t* idx_sorted, void* col,
+                                   int dt, int spec, int is_filter, double a,
+                                   double b, double* state, uint32_t* flags) {
+    hipLaunchKernelGGL(k_stateful, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, seg_start, seg_slot, d_nseg, n, idx_sorted,
+                       col, dt, spec, is_filter, a, b, state, flags);
+}
+
 // ===== time-based (TB) keyed windows: pane lift + watermark advance =====
 // Event-time redesign of the reference's TB path (ffat_replica_gpu.hpp:
 // 869-1047: lift -> thrust (key,pane) reduce -> PendingPanes_Queue ->

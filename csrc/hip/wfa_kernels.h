@@ -102,6 +102,14 @@ void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                       uint64_t* out_key, float* out_val, int64_t* out_ts,
                       int64_t out_cap);
 
+// stateful map/filter: per-key (slot) f64 state, key-order segment walk;
+// map writes results to original positions in place, filter fills flags
+void wfa_stateful_apply(wfa_stream_t s, const uint32_t* seg_start,
+                        const uint32_t* seg_slot, const int64_t* d_nseg,
+                        int64_t n, const uint32_t* idx_sorted, void* col,
+                        int dt, int spec, int is_filter, double a, double b,
+                        double* state, uint32_t* flags);
+
 // ----- time-based keyed windows: pane lift + watermark-driven advance -----
 // One round = lift the batch's tuples into per-slot pending pane partials
 // (pane id = ts / pane_len, absolute; pend ring of 2^pend_log2 panes), then

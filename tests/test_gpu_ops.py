@@ -197,6 +197,47 @@ def test_gpu_ffat_cb_min_tree_vs_oracle():
             assert abs(a - b) <= 1e-5 * max(1.0, abs(b))
 
 
+def test_gpu_stateful_map_running_sum():
+    """Keyed device state advanced in key order (reference
+    Stateful_MAPGPU_Kernel): per-key running sum across batches."""
+    n, n_keys, b = 100_000, 64, 25_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=0))
+           .withOutputSchema([0]).withOutputBatchSize(b).build())
+    sm = (Map_GPU_Builder(native_gpu.gpu_keyed_running_sum(0, max_keys=256))
+          .withOutputSchema([0]).withOutputBatchSize(b).build())
+    g, snk = gpu_graph(src, sm)
+    g.run()
+    _, key, val = gen_batch(n, 0, 42, n_keys, 0)
+    acc = {}
+    exp = 0
+    for k, v in zip(key.tolist(), val.tolist()):
+        acc[k] = acc.get(k, 0) + v
+        exp += acc[k]
+    assert g.sink_sum(snk) == exp
+    assert g.sink_count(snk) == n
+
+
+def test_gpu_stateful_filter_dedup():
+    """Stateful filter: drop consecutive per-key duplicate values."""
+    n, n_keys, b = 100_000, 8, 25_000  # few keys -> many consecutive dups
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=0))
+           .withOutputSchema([0]).withOutputBatchSize(b).build())
+    fl = (Filter_GPU_Builder(native_gpu.gpu_keyed_dedup(0, max_keys=64))
+          .withOutputSchema([0]).withOutputBatchSize(b).build())
+    g, snk = gpu_graph(src, fl)
+    g.run()
+    _, key, val = gen_batch(n, 0, 42, n_keys, 0)
+    last = {}
+    s = cnt = 0
+    for k, v in zip(key.tolist(), val.tolist()):
+        if last.get(k) != v:
+            s += v
+            cnt += 1
+        last[k] = v
+    assert g.sink_sum(snk) == s
+    assert g.sink_count(snk) == cnt
+
+
 def test_gpu_ffat_tb_vs_oracle():
     """Event-time (TB) GPU windows vs brute-force oracle: windows
     [w*slide, w*slide+win) on ts, aligned at the first window containing

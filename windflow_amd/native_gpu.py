@@ -32,6 +32,30 @@ def gpu_gt_filter(col=0, thr=0.5, dtype=2):
     return NativeLogic("gpu_filter", "", [float(thr), 0.0], [spec, col])
 
 
+def gpu_keyed_running_sum(col=0, max_keys=1 << 16):
+    """stateful map: per-key running sum written in place (key-order walk)."""
+    return NativeLogic("gpu_map_keyed", "", [0.0, 0.0], [1, col, max_keys])
+
+
+def gpu_keyed_ema(col=0, alpha=0.9, max_keys=1 << 16):
+    """stateful map: x = ema(state, x) per key."""
+    return NativeLogic("gpu_map_keyed", "", [float(alpha), 0.0], [2, col, max_keys])
+
+
+def gpu_keyed_running_count(col=0, max_keys=1 << 16):
+    return NativeLogic("gpu_map_keyed", "", [0.0, 0.0], [3, col, max_keys])
+
+
+def gpu_keyed_dedup(col=0, max_keys=1 << 16):
+    """stateful filter: drop consecutive per-key duplicates."""
+    return NativeLogic("gpu_filter_keyed", "", [0.0, 0.0], [1, col, max_keys])
+
+
+def gpu_keyed_every_kth(k=2, col=0, max_keys=1 << 16):
+    """stateful filter: keep every k-th tuple per key."""
+    return NativeLogic("gpu_filter_keyed", "", [float(k), 0.0], [2, col, max_keys])
+
+
 def gpu_keyed_reduce(comb=COMB_SUM, col=0, max_keys=1 << 16):
     """per-batch keyed reduction -> one (key, agg, ts_max) per distinct key."""
     return NativeLogic("gpu_reduce", "", [], [comb, col, max_keys])
