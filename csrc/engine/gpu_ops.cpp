@@ -132,7 +132,12 @@ Batch* gpu_alloc_batch(Pool& pool) {
 void gpu_resolve_count(Batch* b) {
     if (b->count >= 0) return;
     HIPCHK(hipEventSynchronize((hipEvent_t)b->ready_event));
-    b->count = std::min<int64_t>(*b->lazy_count, b->capacity);
+    if (*b->lazy_count > b->capacity)
+        throw std::runtime_error(
+            "GPU output batch overflow (" + std::to_string(*b->lazy_count) +
+            " rows > capacity " + std::to_string(b->capacity) +
+            ") — raise the operator's out_batch");
+    b->count = *b->lazy_count;
 }
 
 void gpu_free_batch(Batch* b) {

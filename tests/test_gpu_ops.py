@@ -249,7 +249,7 @@ def test_gpu_ffat_tb_vs_oracle():
     ff = (Ffat_Windows_GPU_Builder(
         native_gpu.gpu_ffat_windows(native_gpu.COMB_SUM, 0, win, slide,
                                     max_keys=1024, tb=True, pend_ring_log2=10))
-          .withOutputSchema([2]).withOutputBatchSize(b).build())
+          .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
     res = dict(rows=[])
 
     def pysink(cols):
