@@ -901,27 +901,19 @@ struct GpuExchangeLogic : GpuLogicBase {
     }
 
     // one collective round: exchange (counts, wm, done) then rows.
-    // Returns true when every rank reported done.
-    bool round(Batch* send, const int64_t* scnt, int64_t wm, bool done,
-               EmitCtx& out, RuntimeCtx& ctx) {
+    // Returns true when every rank reported done.  The metadata is packed
+    // on-device (no pre-collective host sync); the single rendezvous sync
+    // is the h_meta readback every rank needs to lay out its recvs.
+    bool round(Batch* send, int64_t wm, bool done, EmitCtx& out, RuntimeCtx& ctx) {
         const size_t nc = out_schema.payload.size();
-        // metadata allgather
-        int64_t meta[10 + 2];  // world <= 8 in practice; heap-free fast path
-        std::vector<int64_t> meta_v;
-        int64_t* m = meta;
-        if (world + 2 > 12) {
-            meta_v.resize(world + 2);
-            m = meta_v.data();
-        }
-        for (int p = 0; p < world; ++p) m[p] = scnt ? scnt[p] : 0;
-        m[world] = done ? WM_MAX : wm;
-        m[world + 1] = done ? 1 : 0;
-        HIPCHK(hipMemcpyAsync(d_meta, m, 8 * (world + 2), hipMemcpyHostToDevice,
-                              stream));
+        if (!send) wfa_fill_u32(stream, d_counts, 0, world);
+        wfa_pack_meta(stream, d_counts, world, done ? WM_MAX : wm, done ? 1 : 0,
+                      d_meta);
         NCCLCHK(ncclAllGather(d_meta, d_meta_all, world + 2, ncclInt64, comm, stream));
         HIPCHK(hipMemcpyAsync(h_meta, d_meta_all, 8 * world * (world + 2),
                               hipMemcpyDeviceToHost, stream));
         HIPCHK(hipStreamSynchronize(stream));
+        const int64_t* scnt = send ? h_meta + (int64_t)rank * (world + 2) : nullptr;
         // recv layout: rows from rank p land at roff[p]
         int64_t roff[9] = {0};
         int64_t total = 0;
@@ -943,31 +935,60 @@ struct GpuExchangeLogic : GpuLogicBase {
             int64_t soff[9] = {0};
             for (int p = 1; p < world; ++p)
                 soff[p] = soff[p - 1] + (scnt ? scnt[p - 1] : 0);
-            NCCLCHK(ncclGroupStart());
-            for (int p = 0; p < world; ++p) {
-                int64_t sc = scnt ? scnt[p] : 0;
-                int64_t rc = h_meta[(int64_t)p * (world + 2) + rank];
-                if (sc) {
-                    NCCLCHK(ncclSend(send->ts + soff[p], 8 * sc, ncclChar, p, comm, stream));
-                    NCCLCHK(ncclSend(send->key + soff[p], 8 * sc, ncclChar, p, comm, stream));
-                    for (size_t c = 0; c < nc; ++c) {
-                        size_t es = dsize(out_schema.payload[c]);
-                        NCCLCHK(ncclSend((char*)send->cols[c] + es * soff[p], es * sc,
-                                         ncclChar, p, comm, stream));
-                    }
-                }
-                if (rc) {
-                    NCCLCHK(ncclRecv(rb->ts + roff[p], 8 * rc, ncclChar, p, comm, stream));
-                    NCCLCHK(ncclRecv(rb->key + roff[p], 8 * rc, ncclChar, p, comm, stream));
-                    for (size_t c = 0; c < nc; ++c) {
-                        size_t es = dsize(out_schema.payload[c]);
-                        NCCLCHK(ncclRecv((char*)rb->cols[c] + es * roff[p], es * rc,
-                                         ncclChar, p, comm, stream));
-                    }
+            // self rows never touch RCCL: direct D2D on our stream (at
+            // world=1 the whole exchange is one device copy)
+            int64_t selfc = scnt ? scnt[rank] : 0;
+            if (selfc) {
+                HIPCHK(hipMemcpyAsync(rb->ts + roff[rank], send->ts + soff[rank],
+                                      8 * selfc, hipMemcpyDeviceToDevice, stream));
+                HIPCHK(hipMemcpyAsync(rb->key + roff[rank], send->key + soff[rank],
+                                      8 * selfc, hipMemcpyDeviceToDevice, stream));
+                for (size_t c = 0; c < nc; ++c) {
+                    size_t es = dsize(out_schema.payload[c]);
+                    HIPCHK(hipMemcpyAsync((char*)rb->cols[c] + es * roff[rank],
+                                          (char*)send->cols[c] + es * soff[rank],
+                                          es * selfc, hipMemcpyDeviceToDevice,
+                                          stream));
                 }
             }
-            NCCLCHK(ncclGroupEnd());
-            HIPCHK(hipStreamSynchronize(stream));
+            bool any_p2p = false;
+            for (int p = 0; p < world; ++p) {
+                if (p == rank) continue;
+                if ((scnt && scnt[p]) || h_meta[(int64_t)p * (world + 2) + rank])
+                    any_p2p = true;
+            }
+            if (any_p2p) {
+                NCCLCHK(ncclGroupStart());
+                for (int p = 0; p < world; ++p) {
+                    if (p == rank) continue;
+                    int64_t sc = scnt ? scnt[p] : 0;
+                    int64_t rc = h_meta[(int64_t)p * (world + 2) + rank];
+                    if (sc) {
+                        NCCLCHK(ncclSend(send->ts + soff[p], 8 * sc, ncclChar, p, comm, stream));
+                        NCCLCHK(ncclSend(send->key + soff[p], 8 * sc, ncclChar, p, comm, stream));
+                        for (size_t c = 0; c < nc; ++c) {
+                            size_t es = dsize(out_schema.payload[c]);
+                            NCCLCHK(ncclSend((char*)send->cols[c] + es * soff[p], es * sc,
+                                             ncclChar, p, comm, stream));
+                        }
+                    }
+                    if (rc) {
+                        NCCLCHK(ncclRecv(rb->ts + roff[p], 8 * rc, ncclChar, p, comm, stream));
+                        NCCLCHK(ncclRecv(rb->key + roff[p], 8 * rc, ncclChar, p, comm, stream));
+                        for (size_t c = 0; c < nc; ++c) {
+                            size_t es = dsize(out_schema.payload[c]);
+                            NCCLCHK(ncclRecv((char*)rb->cols[c] + es * roff[p], es * rc,
+                                             ncclChar, p, comm, stream));
+                        }
+                    }
+                }
+                NCCLCHK(ncclGroupEnd());
+            }
+            // hand off via events (no host sync): the consumer waits on
+            // rb's event; sb's recycler waits on the event its releaser
+            // records below
+            if (send && send->ready_event)
+                HIPCHK(hipEventRecord((hipEvent_t)send->ready_event, stream));
         }
         if (ctx.stats) ctx.stats->num_kernels++;
         int64_t out_wm = min_wm == WM_MAX ? cur_wm : min_wm;
@@ -1014,15 +1035,11 @@ struct GpuExchangeLogic : GpuLogicBase {
         wfa_gather_rows(stream, oi, n, db->ts, sb->ts, db->key, sb->key,
                         (const void* const*)d_colptrs, (void* const*)(d_colptrs + nc),
                         d_esize, (int)nc);
-        HIPCHK(hipMemcpyAsync(h_counts, d_counts, 4 * world, hipMemcpyDeviceToHost,
-                              stream));
-        HIPCHK(hipStreamSynchronize(stream));
         release_after_use(db);
-        int64_t scnt[9];
-        for (int p = 0; p < world; ++p) scnt[p] = h_counts[p];
         if (ctx.stats) ctx.stats->num_kernels += 4;
-        round(sb, scnt, wm, false, out, ctx);
-        // the collective copied out of sb synchronously; safe to recycle
+        round(sb, wm, false, out, ctx);
+        // round() recorded sb's event after the collective consumed it; the
+        // pool's next get() waits on that event before reuse
         release(sb);
     }
 
@@ -1034,7 +1051,7 @@ struct GpuExchangeLogic : GpuLogicBase {
     void on_eos(EmitCtx& out, RuntimeCtx& ctx) override {
         ensure_init();
         // EOS rounds: keep matching other ranks' collectives until all done
-        while (!round(nullptr, nullptr, cur_wm, true, out, ctx)) {
+        while (!round(nullptr, cur_wm, true, out, ctx)) {
         }
     }
 };

@@ -375,7 +375,21 @@ __global__ void k_gather_rows(const uint32_t* idx, int64_t n, const int64_t* ts_
     }
 }
 
+// pack [counts..., wm, done] as i64 for the exchange metadata allgather
+__global__ void k_pack_meta(const uint32_t* counts, int world, int64_t wm,
+                            int64_t done, int64_t* meta) {
+    int i = threadIdx.x;
+    if (i < world) meta[i] = (int64_t)counts[i];
+    if (i == world) meta[world] = wm;
+    if (i == world + 1) meta[world + 1] = done;
+}
+
 extern "C" {
+void wfa_pack_meta(wfa_stream_t s, const uint32_t* counts, int world, int64_t wm,
+                   int64_t done, int64_t* meta) {
+    hipLaunchKernelGGL(k_pack_meta, dim3(1), dim3(64), 0, (hipStream_t)s, counts,
+                       world, wm, done, meta);
+}
 void wfa_count_u32(wfa_stream_t s, const uint32_t* v, int64_t n, uint32_t* counts,
                    int n_bins) {
     hipMemsetAsync(counts, 0, 4 * n_bins, (hipStream_t)s);

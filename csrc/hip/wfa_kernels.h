@@ -154,6 +154,10 @@ void wfa_cast(wfa_stream_t s, const void* in, int dt_in, void* out, int dt_out,
 void wfa_bucket_by_key(wfa_stream_t s, const uint64_t* key, int64_t n, int world,
                        uint32_t* dest_out);
 
+// pack [counts..., wm, done] into an i64 metadata vector on device
+void wfa_pack_meta(wfa_stream_t s, const uint32_t* counts, int world, int64_t wm,
+                   int64_t done, int64_t* meta);
+
 // histogram of small u32 values (per-destination row counts)
 void wfa_count_u32(wfa_stream_t s, const uint32_t* v, int64_t n, uint32_t* counts,
                    int n_bins);
