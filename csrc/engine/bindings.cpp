@@ -27,13 +27,18 @@ static py::array col_view(DType dt, void* data, int64_t n) {
     throw std::runtime_error("bad dtype");
 }
 
-static py::dict batch_views(Batch* b) {
+static py::dict batch_views(Batch* b, RuntimeCtx* ctx = nullptr) {
     py::dict d;
     d["ts"] = col_view(DType::I64, b->ts, b->count);
     d["key"] = col_view(DType::U64, b->key, b->count);
     for (size_t c = 0; c < b->schema.payload.size(); ++c)
         d[py::str("c" + std::to_string(c))] = col_view(b->schema.payload[c], b->cols[c], b->count);
     d["watermark"] = b->watermark;
+    if (ctx) {  // RuntimeContext parity (reference wf/context.hpp:53)
+        d["replica"] = ctx->replica;
+        d["parallelism"] = ctx->parallelism;
+        d["stream_tag"] = b->stream_tag;
+    }
     return d;
 }
 
@@ -84,7 +89,7 @@ static void emit_pydict(py::dict out_d, EmitCtx& out, int64_t wm) {
 struct PyMapLogic : OpLogic {  // in-place mutate
     py::function fn;
     explicit PyMapLogic(py::function f) : fn(std::move(f)) {}
-    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& rctx) override {
         if (b->refcnt.load(std::memory_order_acquire) > 1) {
             Batch* c = clone(b, *b->pool);
             release(b);
@@ -92,7 +97,7 @@ struct PyMapLogic : OpLogic {  // in-place mutate
         }
         {
             py::gil_scoped_acquire gil;
-            fn(batch_views(b));
+            fn(batch_views(b, &rctx));
         }
         out.emit(b);
     }
@@ -101,10 +106,10 @@ struct PyMapLogic : OpLogic {  // in-place mutate
 struct PyTransformLogic : OpLogic {  // returns new column dict (map/flatmap)
     py::function fn;
     explicit PyTransformLogic(py::function f) : fn(std::move(f)) {}
-    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& rctx) override {
         int64_t wm = b->watermark;
         py::gil_scoped_acquire gil;
-        py::object r = fn(batch_views(b));
+        py::object r = fn(batch_views(b, &rctx));
         release(b);
         if (!r.is_none()) emit_pydict(r.cast<py::dict>(), out, wm);
     }
@@ -113,7 +118,7 @@ struct PyTransformLogic : OpLogic {  // returns new column dict (map/flatmap)
 struct PyFilterLogic : OpLogic {  // returns bool mask
     py::function fn;
     explicit PyFilterLogic(py::function f) : fn(std::move(f)) {}
-    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& rctx) override {
         if (b->refcnt.load(std::memory_order_acquire) > 1) {
             Batch* c = clone(b, *b->pool);
             release(b);
@@ -122,7 +127,7 @@ struct PyFilterLogic : OpLogic {  // returns bool mask
         std::vector<char> keep;
         {
             py::gil_scoped_acquire gil;
-            py::object r = fn(batch_views(b));
+            py::object r = fn(batch_views(b, &rctx));
             auto mask = r.cast<py::array_t<bool>>();
             keep.assign(mask.data(), mask.data() + mask.shape(0));
         }
@@ -154,10 +159,10 @@ struct PyFilterLogic : OpLogic {  // returns bool mask
 struct PySinkLogic : OpLogic {
     py::function fn;
     explicit PySinkLogic(py::function f) : fn(std::move(f)) {}
-    void process(Batch* b, EmitCtx&, RuntimeCtx&) override {
+    void process(Batch* b, EmitCtx&, RuntimeCtx& rctx) override {
         {
             py::gil_scoped_acquire gil;
-            fn(batch_views(b));
+            fn(batch_views(b, &rctx));
         }
         release(b);
     }

@@ -252,6 +252,15 @@ Batch* KSlackCollector::next() {
                     int64_t d = t_curr - bts;
                     if (d > K) K = d;  // adapt slack
                 }
+                // late beyond the adapted slack behind what was already
+                // released: drop + account (reference kslack_collector.hpp
+                // feeds the PipeGraph's atomic_num_dropped)
+                if (bts + K < last_rel_ts) {
+                    if (dropped)
+                        dropped->fetch_add(b->count, std::memory_order_relaxed);
+                    release(b);
+                    continue;
+                }
                 t_curr = std::max(t_curr, bts);
                 buf.emplace_back(bts, b);
             }
@@ -262,6 +271,7 @@ Batch* KSlackCollector::next() {
                                        [](auto& a, auto& b) { return a.first < b.first; });
             if (n_open == 0 || it->first <= t_curr - K) {
                 Batch* b = it->second;
+                last_rel_ts = std::max(last_rel_ts, it->first);
                 buf.erase(it);
                 b->watermark = std::max(last_fwd_wm, b->watermark);
                 last_fwd_wm = b->watermark;

@@ -108,6 +108,11 @@ struct BroadcastEmitter : Emitter {
     Pool* punct_pool;
     explicit BroadcastEmitter(std::vector<SpscQueue*> d, Pool* pp) : dests(std::move(d)), punct_pool(pp) {}
     void emit(Batch* b) override {
+        if (b->loc == Loc::DEVICE && dests.size() > 1)
+            throw std::runtime_error(
+                "broadcast of device batches is not supported (per-batch "
+                "events are single-consumer): stage through gpu_to_host or "
+                "use split_gpu per branch");
         account(b);
         b->refcnt.fetch_add((int)dests.size() - 1, std::memory_order_relaxed);
         for (auto* q : dests) q->push(b, abort);
@@ -234,6 +239,7 @@ struct OrderingCollector : Collector {
 struct KSlackCollector : Collector {
     int64_t K = 0;
     int64_t t_curr = 0;
+    int64_t last_rel_ts = INT64_MIN;
     std::vector<std::pair<int64_t, Batch*>> buf;
     std::atomic<int64_t>* dropped = nullptr;
     using Collector::Collector;

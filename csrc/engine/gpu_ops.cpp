@@ -1010,12 +1010,14 @@ struct GpuExchangeLogic : GpuLogicBase {
     }
 
     void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        int64_t t0 = now_us();
         ensure_init();
         Batch* db = input_on_device(in, ctx);
         int64_t n = db->count;
         int64_t wm = db->watermark;
         if (n > out_cap) throw std::runtime_error("exchange input > out_batch");
         Batch* sb = send_pool->get();
+        int64_t t1 = now_us();
         if (sb->ready_event)
             HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)sb->ready_event, 0));
         // bucket -> stable sort by dest -> contiguous per-dest send layout
@@ -1037,10 +1039,14 @@ struct GpuExchangeLogic : GpuLogicBase {
                         d_esize, (int)nc);
         release_after_use(db);
         if (ctx.stats) ctx.stats->num_kernels += 4;
+        int64_t t2 = now_us();
         round(sb, wm, false, out, ctx);
         // round() recorded sb's event after the collective consumed it; the
         // pool's next get() waits on that event before reuse
         release(sb);
+        if (wfa_prof())
+            fprintf(stderr, "[prof] xchg n=%ld in+get=%ld us bucket=%ld us round=%ld us\n",
+                    (long)n, (long)(t1 - t0), (long)(t2 - t1), (long)(now_us() - t2));
     }
 
     // Puncts are NOT collective-safe (their count differs per rank and
