@@ -182,6 +182,19 @@ struct PySourceLogic : OpLogic {
         if (r.is_none()) return false;
         py::dict d = r.cast<py::dict>();
         int64_t wm = d.contains("watermark") ? d["watermark"].cast<int64_t>() : 0;
+        if (ctx.engine && ctx.engine->time_policy == TimePolicy::INGRESS_TIME) {
+            // ingress time: stamp arrival clock, ignore user ts/wm
+            int64_t nowus = now_us();
+            auto n = d.contains("c0") ? d["c0"].cast<py::array>().shape(0)
+                                      : (d.contains("ts")
+                                             ? d["ts"].cast<py::array>().shape(0)
+                                             : 0);
+            py::array_t<int64_t> ts({(py::ssize_t)n});
+            for (py::ssize_t i = 0; i < (py::ssize_t)n; ++i)
+                ts.mutable_at(i) = nowus;
+            d["ts"] = ts;
+            wm = nowus;
+        }
         emit_pydict(d, out, wm);
         return true;
     }

@@ -37,15 +37,19 @@ struct SeqSource : OpLogic {
         if (pos >= len) return false;
         Batch* b = out.new_batch();
         int64_t n = std::min<int64_t>(std::min<int64_t>(bsz, b->capacity), len - pos);
+        // INGRESS_TIME: ts = arrival clock (reference source_shipper.hpp:171)
+        const bool ingress =
+            ctx.engine && ctx.engine->time_policy == TimePolicy::INGRESS_TIME;
+        const int64_t nowus = ingress ? now_us() : 0;
         for (int64_t i = 0; i < n; ++i) {
             int64_t v = voff + pos + i + 1;
-            b->ts[i] = pos + i + 1;
+            b->ts[i] = ingress ? nowus : pos + i + 1;
             b->key[i] = (uint64_t)(v % n_keys);
             b->col<int64_t>(0)[i] = v;
         }
         b->count = n;
         pos += n;
-        b->watermark = pos;  // DEFAULT mode: wm = max emitted ts
+        b->watermark = ingress ? nowus : pos;  // wm = max emitted ts
         out.emit(b);
         return pos < len;
     }

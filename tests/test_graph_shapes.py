@@ -118,6 +118,120 @@ def test_broadcast_routing():
     assert g.sink_sum(snk_op) == exp
 
 
+def test_split_then_merge():
+    """Source -> split(v%2) -> [b0 -> Map(+100), b1 -> Map(+200)] -> merge
+    -> Sink (reference pipegraph merge of split children)."""
+    stream_len = 8000
+    g = wf.PipeGraph("sm")
+    src = (wf.Source_Builder(native.seq_source(stream_len, 5, 64))
+           .withParallelism(1).withOutputSchema([0]).build())
+    mp = g.add_source(src)
+    mp.split(native.split_mod(0), 2)
+    b0 = mp.select(0)
+    b1 = mp.select(1)
+    b0.add(wf.Map_Builder(native.affine_map(0, 1, 100)).withParallelism(2)
+           .withOutputSchema([0]).build())
+    b1.add(wf.Map_Builder(native.affine_map(0, 1, 200)).withParallelism(1)
+           .withOutputSchema([0]).build())
+    merged = b0.merge(b1)
+    snk = wf.Sink_Builder(native.sum_sink(0)).withParallelism(2).build()
+    merged.add_sink(snk)
+    g.run()
+    exp = sum(v + 100 for v in range(1, stream_len + 1) if v % 2 == 0) + \
+          sum(v + 200 for v in range(1, stream_len + 1) if v % 2 == 1)
+    assert g.sink_sum(snk) == exp
+
+
+def test_merge_three_pipes():
+    stream_len = 5000
+    g = wf.PipeGraph("m3")
+    pipes = []
+    for i in range(3):
+        s = (wf.Source_Builder(native.seq_source(stream_len, 3, 128,
+                                                 value_offset=i * 100000))
+             .withParallelism(1).withOutputSchema([0]).build())
+        pipes.append(g.add_source(s))
+    mp = pipes[0].merge(pipes[1], pipes[2])
+    snk = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+    mp.add_sink(snk)
+    g.run()
+    exp = sum(sum(range(1 + i * 100000, stream_len + 1 + i * 100000))
+              for i in range(3))
+    assert g.sink_sum(snk) == exp
+
+
+def test_chained_windows():
+    """window -> window: keyed tumbling sums then keyed tumbling sums of
+    sums — exercises watermark propagation through a firing operator."""
+    from windflow_amd.builders import Keyed_Windows_Builder
+    from collections import defaultdict, Counter
+    stream_len, keys = 4000, 5
+    g = wf.PipeGraph("ww")
+    src = (wf.Source_Builder(native.seq_source(stream_len, keys, 128))
+           .withParallelism(1).withOutputSchema([0]).build())
+    mp = g.add_source(src)
+    mp.add(Keyed_Windows_Builder(func=("sum", 0)).withCBWindows(10, 10)
+           .withOutputSchema([0]).build())
+    mp.add(Keyed_Windows_Builder(func=("sum", 0)).withCBWindows(4, 4)
+           .withOutputSchema([0]).build())
+    rows = []
+
+    def sink(cols):
+        for i in range(len(cols['ts'])):
+            rows.append((int(cols['key'][i]), cols['c0'][i].item()))
+
+    mp.add_sink(wf.Sink_Builder(sink).withParallelism(1).build())
+    g.run()
+    # oracle: per key tumbling-10 sums, then tumbling-4 sums over those
+    per = defaultdict(list)
+    for v in range(1, stream_len + 1):
+        per[v % keys].append(v)
+    exp = Counter()
+    for k, vals in per.items():
+        l1 = [sum(vals[i:i + 10]) for i in range(0, len(vals), 10)]
+        for i in range(0, len(l1), 4):
+            exp[(k, sum(l1[i:i + 4]))] += 1
+    assert Counter(rows) == exp
+
+
+def test_rebalancing_routing():
+    """REBALANCING: round-robin batches across replicas regardless of key."""
+    stream_len = 6000
+    g = wf.PipeGraph("rb")
+    src = (wf.Source_Builder(native.seq_source(stream_len, 1, 64))
+           .withParallelism(1).withOutputSchema([0]).build())
+    mp = g.add_source(src)
+    mp.add(wf.Map_Builder(native.affine_map(0, 1, 0)).withParallelism(3)
+           .withRebalancing().withOutputSchema([0]).build())
+    snk = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+    mp.add_sink(snk)
+    g.run()
+    assert g.sink_sum(snk) == sum(range(1, stream_len + 1))
+
+
+def test_ingress_time_policy():
+    """INGRESS_TIME stamps arrival-clock timestamps (monotone, recent)."""
+    import time
+    seen = dict(lo=None, hi=None)
+
+    def sink(cols):
+        ts = cols['ts']
+        if len(ts):
+            seen['lo'] = int(ts.min()) if seen['lo'] is None else min(seen['lo'], int(ts.min()))
+            seen['hi'] = int(ts.max()) if seen['hi'] is None else max(seen['hi'], int(ts.max()))
+
+    t0 = time.monotonic_ns() // 1000
+    g = wf.PipeGraph("it", wf.ExecutionMode.DEFAULT, wf.TimePolicy.INGRESS_TIME)
+    src = (wf.Source_Builder(native.seq_source(5000, 3, 256))
+           .withParallelism(1).withOutputSchema([0]).build())
+    mp = g.add_source(src)
+    mp.add_sink(wf.Sink_Builder(sink).withParallelism(1).build())
+    g.run()
+    t1 = time.monotonic_ns() // 1000
+    assert seen['lo'] is not None
+    assert t0 <= seen['lo'] <= seen['hi'] <= t1
+
+
 def test_stats_records():
     g = wf.PipeGraph("st")
     src = (wf.Source_Builder(native.seq_source(1000, 3, 128))
