@@ -344,10 +344,32 @@ void wfa_bucket_by_key(wfa_stream_t s, const uint64_t* key, int64_t n, int world
 }
 
 // ===== small-value histogram (all-to-all dest counts) =====
-__global__ void k_count_u32(const uint32_t* v, int64_t n, uint32_t* counts) {
+// nb <= 16: thread-private register counts -> LDS block reduce -> one
+// global atomic per (block, bin).  A naive per-element global atomicAdd
+// serializes on the hot bin (measured 45 ms for 4M rows at world=1).
+__global__ void k_count_u32(const uint32_t* v, int64_t n, uint32_t* counts,
+                            int nb) {
+    uint32_t loc[16];
+#pragma unroll
+    for (int b = 0; b < 16; ++b) loc[b] = 0;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
-         i += gridDim.x * (int64_t)blockDim.x)
-        atomicAdd(&counts[v[i]], 1u);
+         i += gridDim.x * (int64_t)blockDim.x) {
+        uint32_t d = v[i];
+        if (d < 16u) loc[d]++;
+        else atomicAdd(&counts[d], 1u);  // generic fallback
+    }
+    __shared__ uint32_t h[16];
+    if (threadIdx.x < 16) h[threadIdx.x] = 0;
+    __syncthreads();
+#pragma unroll
+    for (int b = 0; b < 16; ++b) {
+        uint32_t c = loc[b];
+        for (int off = 32; off; off >>= 1) c += __shfl_down(c, off, 64);
+        if ((threadIdx.x & 63) == 0 && c) atomicAdd(&h[b], c);
+    }
+    __syncthreads();
+    if (threadIdx.x < (uint32_t)min(nb, 16) && h[threadIdx.x])
+        atomicAdd(&counts[threadIdx.x], h[threadIdx.x]);
 }
 
 // ===== full-row gather by permutation (ts + key + payload columns) =====
@@ -393,8 +415,8 @@ void wfa_pack_meta(wfa_stream_t s, const uint32_t* counts, int world, int64_t wm
 void wfa_count_u32(wfa_stream_t s, const uint32_t* v, int64_t n, uint32_t* counts,
                    int n_bins) {
     hipMemsetAsync(counts, 0, 4 * n_bins, (hipStream_t)s);
-    hipLaunchKernelGGL(k_count_u32, dim3(nblk(n)), dim3(WFA_THREADS), 0,
-                       (hipStream_t)s, v, n, counts);
+    hipLaunchKernelGGL(k_count_u32, dim3(nblk(n, 8)), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, v, n, counts, n_bins);
 }
 void wfa_gather_rows(wfa_stream_t s, const uint32_t* idx, int64_t n,
                      const int64_t* ts_in, int64_t* ts_out, const uint64_t* key_in,
