@@ -13,6 +13,7 @@
 #include <cstring>
 #include <deque>
 #include <map>
+#include <tuple>
 
 #include "engine.hpp"
 
@@ -657,6 +658,7 @@ struct GpuFfatLogic : GpuLogicBase {
     }
     ~GpuFfatLogic() override {
         if (h_flags) (void)hipHostFree(h_flags);
+        destroy_graphs();
     }
 
     void check_tb_flags() {
@@ -668,18 +670,20 @@ struct GpuFfatLogic : GpuLogicBase {
                 "or lateness is too large for the configured ring");
     }
 
-    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
-        ensure_init();
-        Batch* db = input_on_device(in, ctx);
-        int64_t n = db->count;
+    // hipGraph capture of the steady-state per-batch chain (slot -> sort ->
+    // segments -> offsets -> fold -> count copy): the pool rotates a small
+    // fixed set of in/out batches, so the (in, out, n) tuple recurs and one
+    // instantiated graph per tuple replays the whole ~25-kernel chain with
+    // a single launch.  Opt out with WFA_NO_HIPGRAPH=1.
+    std::map<std::tuple<const void*, const void*, int64_t>, hipGraphExec_t> graphs;
+    static bool graphs_enabled() {
+        static int v = -1;
+        if (v < 0) v = getenv("WFA_NO_HIPGRAPH") ? 0 : 1;
+        return v;
+    }
+
+    void chain_cb(Batch* db, Batch* ob, int64_t n, RuntimeCtx& ctx) {
         ks.group(stream, db, vcol, ctx);
-        if (tb) {
-            tb_round_with_count(db, n, db->watermark, out, ctx);
-            release(db);
-            return;
-        }
-        Batch* ob = get_dev();
-        // deterministic output offsets (no atomic cursor)
         uint32_t* nf = (uint32_t*)ks.v_sorted;  // reuse freed scratch
         wfa_ffat_fire_offsets(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                               pane_len, P, S, st_fill, st_head, nf, d_on);
@@ -696,12 +700,46 @@ struct GpuFfatLogic : GpuLogicBase {
                              ring_or_tree, st_head, st_wsum, ks.slot_to_key, nf,
                              ob->key, (float*)ob->cols[0], ob->ts, ob->capacity);
         HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
+    }
+
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        Batch* db = input_on_device(in, ctx);
+        int64_t n = db->count;
+        if (tb) {
+            ks.group(stream, db, vcol, ctx);
+            tb_round_with_count(db, n, db->watermark, out, ctx);
+            release(db);
+            return;
+        }
+        Batch* ob = get_dev();
+        auto key = std::make_tuple((const void*)db->ts, (const void*)ob->ts, n);
+        auto it = graphs.find(key);
+        if (it != graphs.end()) {
+            HIPCHK(hipGraphLaunch(it->second, stream));
+            if (ctx.stats) ctx.stats->num_kernels += 1;
+        } else if (graphs_enabled() && graphs.size() < 64) {
+            HIPCHK(hipStreamBeginCapture(stream, hipStreamCaptureModeThreadLocal));
+            chain_cb(db, ob, n, ctx);
+            hipGraph_t graph;
+            HIPCHK(hipStreamEndCapture(stream, &graph));
+            hipGraphExec_t exec;
+            HIPCHK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0));
+            HIPCHK(hipGraphDestroy(graph));
+            graphs.emplace(key, exec);
+            HIPCHK(hipGraphLaunch(exec, stream));
+        } else {
+            chain_cb(db, ob, n, ctx);
+        }
         ob->count = -1;  // resolved by the consumer via gpu_resolve_count
         ob->watermark = db->watermark;
-        if (ctx.stats) ctx.stats->num_kernels += 1;
         release(db);
         record_ready(ob);
         out.emit(ob);
+    }
+    void destroy_graphs() {
+        for (auto& [k, e] : graphs) (void)hipGraphExecDestroy(e);
+        graphs.clear();
     }
 
     void tb_round_with_count(Batch* db, int64_t n, int64_t wm, EmitCtx& out,
