@@ -138,6 +138,7 @@ def main():
             dist.barrier()
 
     K, W, B = args.steps, args.warmup, args.batch
+    p99_us = None
 
     if args.config == "cpu":
         B = min(B, 65536)
@@ -170,6 +171,8 @@ def main():
         barrier()
         dt = t1 - t0
         n_gpus = world
+        lat = sorted(g.engine.sink_latencies(g._sink_map[id(snk)]))
+        p99_us = lat[min(len(lat) - 1, int(0.99 * len(lat)))] if lat else None
 
     # max step time over ranks
     if dist is not None:
@@ -202,6 +205,8 @@ def main():
                 "slide": args.slide,
                 "keys_per_rank": args.keys,
                 "parallelism": f"keyed-dp{max(world,1)}",
+                "p99_batch_latency_us": (p99_us if args.config != "cpu"
+                                         else None),
             },
         }
         print(json.dumps(out))

@@ -422,6 +422,12 @@ PYBIND11_MODULE(_core, m) {
         .def("sink_sum", [](Engine& e, int op) { return e.sink_acc_i64.at(op).load(); })
         .def("sink_count", [](Engine& e, int op) { return e.sink_tuples.at(op).load(); })
         .def("dropped", [](Engine& e) { return e.dropped_tuples.load(); })
+        .def("sink_latencies",
+             [](Engine& e, int op) {
+                 auto it = e.sink_latencies.find(op);
+                 return it == e.sink_latencies.end() ? std::vector<int64_t>{}
+                                                     : it->second;
+             })
         .def("stats", [](Engine& e) {
             py::list out;
             for (auto& r : e.replicas) {

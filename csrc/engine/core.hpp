@@ -75,6 +75,7 @@ struct Batch {
     void* stream = nullptr;     // hipStream_t of producer (device batches)
     void* ready_event = nullptr;// hipEvent_t signalled when contents valid
 
+    int64_t born_us = 0;        // host clock when the source filled it
     int64_t* ts = nullptr;      // [capacity]
     uint64_t* key = nullptr;    // [capacity]
     // Deferred count: device producers may emit before the row count is

@@ -364,6 +364,7 @@ struct Engine {
     std::unordered_map<int, double> sink_acc_f64;
     std::mutex sink_f64_mu;
     std::unordered_map<int, std::atomic<int64_t>> sink_tuples;
+    std::unordered_map<int, std::vector<int64_t>> sink_latencies;  // us, GPU sinks
     int64_t queue_capacity = 128;   // batches per SPSC queue
     bool pin_threads = false;
     // distributed (one process per GPU, RCCL over xGMI): set before build()

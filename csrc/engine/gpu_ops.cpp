@@ -270,6 +270,7 @@ struct GpuSourceLogic : GpuLogicBase {
         int64_t n = std::min<int64_t>(bsz, len - pos);
         wfa_gen_batch(stream, db->ts, db->key, db->cols[0], vdt, n, pos, seed, n_keys);
         db->count = n;
+        db->born_us = now_us();
         pos += n;
         db->watermark = pos - 1;
         record_ready(db);
@@ -572,6 +573,7 @@ struct GpuReduceLogic : GpuLogicBase {
         HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
         ob->count = -1;
         ob->watermark = db->watermark;
+        ob->born_us = db->born_us;
         if (ctx.stats) ctx.stats->num_kernels += 1;
         release_after_use(db);
         record_ready(ob);
@@ -733,6 +735,7 @@ struct GpuFfatLogic : GpuLogicBase {
         }
         ob->count = -1;  // resolved by the consumer via gpu_resolve_count
         ob->watermark = db->watermark;
+        ob->born_us = db->born_us;
         release(db);
         record_ready(ob);
         out.emit(ob);
@@ -818,12 +821,14 @@ struct GpuCountSink : GpuLogicBase {
     std::deque<Batch*> pending;
     GpuCountSink(Engine* e, int id, int dev) : eng(e), op_id(id) { device = dev; }
     int64_t wait_us = 0;
+    std::vector<int64_t> lat_us;  // per-batch source->sink latency
     void drain_one() {
         Batch* b = pending.front();
         pending.pop_front();
         int64_t t0 = now_us();
         gpu_resolve_count(b);
         wait_us += now_us() - t0;
+        if (b->born_us) lat_us.push_back(now_us() - b->born_us);
         tuples += b->count;
         release(b);
     }
@@ -837,6 +842,11 @@ struct GpuCountSink : GpuLogicBase {
         if (wfa_prof())
             fprintf(stderr, "[prof] sink event-wait total: %ld us\n", (long)wait_us);
         eng->sink_tuples[op_id].fetch_add(tuples, std::memory_order_relaxed);
+        {
+            std::lock_guard<std::mutex> g(eng->sink_f64_mu);
+            auto& v = eng->sink_latencies[op_id];
+            v.insert(v.end(), lat_us.begin(), lat_us.end());
+        }
     }
 };
 
