@@ -393,7 +393,7 @@ struct KeyedScratch {
         idx = (uint32_t*)A.get(4 * cap);
         slot_t = (uint32_t*)A.get(4 * cap);
         idx_t = (uint32_t*)A.get(4 * cap);
-        hist = (uint32_t*)A.get(4 * (16 * (cap / 2048 + 2) + 16));
+        hist = (uint32_t*)A.get(4 * wfa_sort_hist_u32(cap));
         seg_start = (uint32_t*)A.get(4 * cap);
         seg_slot = (uint32_t*)A.get(4 * cap);
         d_nseg = (int64_t*)A.get(64);
@@ -926,7 +926,7 @@ struct GpuExchangeLogic : GpuLogicBase {
         idx = (uint32_t*)A.get(4 * out_cap);
         dest_t = (uint32_t*)A.get(4 * out_cap);
         idx_t = (uint32_t*)A.get(4 * out_cap);
-        hist = (uint32_t*)A.get(4 * (16 * (out_cap / 2048 + 2) + 16));
+        hist = (uint32_t*)A.get(4 * wfa_sort_hist_u32(out_cap));
         d_counts = (uint32_t*)A.get(4 * world + 64);
         d_meta = (int64_t*)A.get(8 * (world + 2));
         d_meta_all = (int64_t*)A.get(8 * world * (world + 2));

@@ -39,10 +39,11 @@ __device__ __forceinline__ uint64_t mix64s(uint64_t k) {
 __global__ void k_key_to_slot(const uint64_t* key, int64_t n, uint64_t* tkeys,
                               uint32_t* tslots, uint32_t* n_slots, int64_t cap,
                               uint32_t* slot_out, uint64_t* slot_to_key) {
+    const uint64_t mask = (uint64_t)cap - 1;  // cap is a power of two
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += gridDim.x * (int64_t)blockDim.x) {
         uint64_t k = key[i];
-        uint64_t p = mix64s(k) % (uint64_t)cap;
+        uint64_t p = mix64s(k) & mask;
         for (;;) {
             uint64_t cur = __hip_atomic_load(&tkeys[p], __ATOMIC_RELAXED,
                                              __HIP_MEMORY_SCOPE_AGENT);
@@ -62,7 +63,7 @@ __global__ void k_key_to_slot(const uint64_t* key, int64_t n, uint64_t* tkeys,
                 if (expected == k) break;  // raced: same key inserted by other
                 // raced: different key took the cell — keep probing
             }
-            p = (p + 1) % (uint64_t)cap;
+            p = (p + 1) & mask;
         }
         // wait for the slot id to be published
         uint32_t s;
@@ -86,6 +87,15 @@ extern "C" void wfa_key_to_slot(wfa_stream_t s, const uint64_t* key, int64_t n,
 // ===== stable LSD radix sort, 4-bit digits =====
 extern "C" int64_t wfa_sort_nblocks(int64_t n) {
     return (n + RS_PER_BLOCK - 1) / RS_PER_BLOCK;
+}
+
+// scratch requirement (u32 elements) covering the 4-bit and 8-bit paths
+extern "C" int64_t wfa_sort_hist_u32(int64_t cap) {
+    int64_t nb4 = (cap + RS_PER_BLOCK - 1) / RS_PER_BLOCK;
+    int64_t nb8 = (cap + WFA_THREADS * 16 - 1) / (WFA_THREADS * 16);  // RS8
+    int64_t a = 16 * nb4 + 16;
+    int64_t b = 256 * nb8 + 512;
+    return (a > b ? a : b) + 64;
 }
 
 __global__ void k_rs_hist(const uint32_t* keys, int64_t n, int shift, uint32_t* hist,
@@ -223,14 +233,168 @@ __global__ void k_rs_scatter(const uint32_t* keys, const uint32_t* vals, int64_t
     }
 }
 
+// ===== 8-bit-digit radix pass (halves the pass count for slot sorts) =====
+// Item order inside a block: per-wave contiguous ranges, lane-strided
+// rounds (waveBase + j*64 + lane) — coalesced loads AND a stable rank
+// order (wave w's range precedes wave w+1's; within a wave, round j
+// precedes j+1; within a round, lane order).  Per-item stable rank =
+// LDS per-wave digit counter before this round + lane rank inside the
+// round's same-digit ballot group.
+#define RS8_IPT 16
+#define RS8_PER_WAVE (64 * RS8_IPT)
+#define RS8_PER_BLOCK (WFA_THREADS * RS8_IPT)
+
+extern "C" int64_t wfa_sort8_nblocks(int64_t n) {
+    return (n + RS8_PER_BLOCK - 1) / RS8_PER_BLOCK;
+}
+
+__global__ void k_rs8_hist(const uint32_t* keys, int64_t n, int shift,
+                           uint32_t* hist, int64_t nblocks) {
+    __shared__ uint32_t h[256];
+    for (int d = threadIdx.x; d < 256; d += blockDim.x) h[d] = 0;
+    __syncthreads();
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    int64_t waveBase = (int64_t)blockIdx.x * RS8_PER_BLOCK + wave * RS8_PER_WAVE;
+#pragma unroll
+    for (int j = 0; j < RS8_IPT; ++j) {
+        int64_t i = waveBase + j * 64 + lane;
+        if (i < n) atomicAdd(&h[(keys[i] >> shift) & 255], 1u);
+    }
+    __syncthreads();
+    for (int d = threadIdx.x; d < 256; d += blockDim.x)
+        hist[(int64_t)d * nblocks + blockIdx.x] = h[d];
+}
+
+// per-digit scan over blocks (256 blocks, one digit each) + digit totals
+__global__ void k_rs8_scan(uint32_t* hist, int64_t nblocks, uint32_t* dt) {
+    uint32_t* a = hist + (int64_t)blockIdx.x * nblocks;
+    __shared__ uint32_t tot[256];
+    const int64_t chunk = (nblocks + 255) / 256;
+    const int64_t b0 = (int64_t)threadIdx.x * chunk;
+    const int64_t b1 = min(nblocks, b0 + chunk);
+    uint32_t s = 0;
+    for (int64_t i = b0; i < b1; ++i) s += a[i];
+    tot[threadIdx.x] = s;
+    __syncthreads();
+    for (int off = 1; off < 256; off <<= 1) {
+        uint32_t t = (threadIdx.x >= off) ? tot[threadIdx.x - off] : 0;
+        __syncthreads();
+        tot[threadIdx.x] += t;
+        __syncthreads();
+    }
+    uint32_t run = tot[threadIdx.x] - s;
+    for (int64_t i = b0; i < b1; ++i) {
+        uint32_t v = a[i];
+        a[i] = run;
+        run += v;
+    }
+    if (threadIdx.x == 255) dt[blockIdx.x] = tot[255];
+}
+
+__global__ void k_rs8_dbase(const uint32_t* dt, uint32_t* dbase) {  // 1 block
+    __shared__ uint32_t tot[256];
+    tot[threadIdx.x] = dt[threadIdx.x];
+    __syncthreads();
+    for (int off = 1; off < 256; off <<= 1) {
+        uint32_t t = (threadIdx.x >= off) ? tot[threadIdx.x - off] : 0;
+        __syncthreads();
+        tot[threadIdx.x] += t;
+        __syncthreads();
+    }
+    dbase[threadIdx.x] = tot[threadIdx.x] - dt[threadIdx.x];
+}
+
+__global__ void k_rs8_scatter(const uint32_t* keys, const uint32_t* vals,
+                              int64_t n, int shift, const uint32_t* hist,
+                              int64_t nblocks, const uint32_t* dbase,
+                              uint32_t* keys_out, uint32_t* vals_out) {
+    __shared__ uint32_t gbase[256];                    // digit base, this block
+    __shared__ uint32_t waveCnt[WFA_THREADS / 64][256];
+    __shared__ uint32_t wavePre[WFA_THREADS / 64][256];
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const uint64_t lt = ((uint64_t)1 << lane) - 1;
+    for (int d = threadIdx.x; d < 256; d += blockDim.x) {
+        gbase[d] = hist[(int64_t)d * nblocks + blockIdx.x] + dbase[d];
+        for (int w = 0; w < WFA_THREADS / 64; ++w) waveCnt[w][d] = 0;
+    }
+    __syncthreads();
+    int64_t waveBase = (int64_t)blockIdx.x * RS8_PER_BLOCK + wave * RS8_PER_WAVE;
+    uint32_t mk[RS8_IPT], mv[RS8_IPT], mr[RS8_IPT];  // key, val, (digit<<24|rank)
+    int nit = 0;
+#pragma unroll
+    for (int j = 0; j < RS8_IPT; ++j) {
+        int64_t i = waveBase + j * 64 + lane;
+        bool valid = i < n;
+        uint32_t k = valid ? keys[i] : 0;
+        uint32_t v = valid ? vals[i] : 0;
+        uint32_t d = (k >> shift) & 255;
+        // same-digit ballot group among valid lanes of this round
+        uint64_t mask = __ballot(valid);
+#pragma unroll
+        for (int bit = 0; bit < 8; ++bit) {
+            uint64_t b = __ballot((d >> bit) & 1);
+            mask &= ((d >> bit) & 1) ? b : ~b;
+        }
+        if (valid) {
+            uint32_t pre = waveCnt[wave][d];
+            uint32_t rank = pre + (uint32_t)__popcll(mask & lt);
+            mk[j] = k;
+            mv[j] = v;
+            mr[j] = (d << 24) | (rank & 0xFFFFFF);
+            // group leader bumps the wave's digit counter
+            if ((mask & lt) == 0) waveCnt[wave][d] = pre + (uint32_t)__popcll(mask);
+            nit = j + 1;
+        }
+    }
+    __syncthreads();
+    // prefix of wave digit counts across waves
+    for (int d = threadIdx.x; d < 256; d += blockDim.x) {
+        uint32_t run = 0;
+        for (int w = 0; w < WFA_THREADS / 64; ++w) {
+            wavePre[w][d] = run;
+            run += waveCnt[w][d];
+        }
+    }
+    __syncthreads();
+    for (int j = 0; j < nit; ++j) {
+        uint32_t d = mr[j] >> 24;
+        uint32_t pos = gbase[d] + wavePre[wave][d] + (mr[j] & 0xFFFFFF);
+        keys_out[pos] = mk[j];
+        vals_out[pos] = mv[j];
+    }
+}
+
 extern "C" void wfa_sort_pairs(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                                uint32_t* slot_tmp, uint32_t* idx_tmp, uint32_t* hist,
                                int64_t n, int bits, uint32_t** out_slot,
                                uint32_t** out_idx) {
     hipStream_t st = (hipStream_t)s;
+    uint32_t *ka = slot, *va = idx, *kb = slot_tmp, *vb = idx_tmp;
+    if (bits > 4) {
+        // 8-bit digits: fewer passes; hist is sized for 16*nblocks4 + 16
+        // which covers 256*nblocks8 + 512 (RS8_PER_BLOCK = 2*RS_PER_BLOCK)
+        int64_t nb = wfa_sort8_nblocks(n);
+        uint32_t* dt = hist + 256 * nb;
+        uint32_t* dbase = dt + 256;
+        int passes = (bits + 7) / 8;
+        for (int p = 0; p < passes; ++p) {
+            int shift = 8 * p;
+            hipLaunchKernelGGL(k_rs8_hist, dim3(nb), dim3(WFA_THREADS), 0, st, ka, n,
+                               shift, hist, nb);
+            hipLaunchKernelGGL(k_rs8_scan, dim3(256), dim3(256), 0, st, hist, nb, dt);
+            hipLaunchKernelGGL(k_rs8_dbase, dim3(1), dim3(256), 0, st, dt, dbase);
+            hipLaunchKernelGGL(k_rs8_scatter, dim3(nb), dim3(WFA_THREADS), 0, st, ka,
+                               va, n, shift, hist, nb, dbase, kb, vb);
+            uint32_t* t;
+            t = ka; ka = kb; kb = t;
+            t = va; va = vb; vb = t;
+        }
+        *out_slot = ka;
+        *out_idx = va;
+        return;
+    }
     int64_t nblocks = wfa_sort_nblocks(n);
     uint32_t* dt = hist + 16 * nblocks;  // hist is sized 16*nblocks + 16
-    uint32_t *ka = slot, *va = idx, *kb = slot_tmp, *vb = idx_tmp;
     int passes = (bits + 3) / 4;
     for (int p = 0; p < passes; ++p) {
         int shift = 4 * p;

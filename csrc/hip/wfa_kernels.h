@@ -51,6 +51,7 @@ void wfa_sort_pairs(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                     int64_t n, int bits,
                     uint32_t** out_slot, uint32_t** out_idx);
 int64_t wfa_sort_nblocks(int64_t n);
+int64_t wfa_sort_hist_u32(int64_t cap);  // hist scratch size in u32
 
 // gather rows by permutation idx (values f32/i64 + ts)
 void wfa_gather(wfa_stream_t s, const uint32_t* idx, int64_t n,
