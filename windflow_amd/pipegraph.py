@@ -233,9 +233,16 @@ class PipeGraph:
         return eid
 
     def build_engine(self):
+        import os as _os
         e = _core.Engine()
         e.mode = self.mode
         e.time_policy = self.time_policy
+        # runtime knobs replacing the reference's compile-time macros
+        # (README.md:32-41: FF_BOUNDED_BUFFER capacity, NO_DEFAULT_MAPPING)
+        if _os.environ.get("WFA_QUEUE_CAP"):
+            e.queue_capacity = int(_os.environ["WFA_QUEUE_CAP"])
+        if _os.environ.get("WFA_PIN_THREADS"):
+            e.pin_threads = True
         if self._dist is not None:
             e.set_dist(*self._dist)
         default_ck = {ExecutionMode.DEFAULT: CollectorKind.WATERMARK,
