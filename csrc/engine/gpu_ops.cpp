@@ -369,9 +369,9 @@ struct KeyedScratch {
     uint32_t* hist = nullptr;
     uint32_t *seg_start = nullptr, *seg_slot = nullptr;
     int64_t* d_nseg = nullptr;
-    float* v_sorted = nullptr;
+    float* v_sorted = nullptr;   // sort output (value payload)
     int64_t* ts_sorted = nullptr;
-    float* v_f32 = nullptr;  // cast buffer
+    float* v_f32 = nullptr;  // cast buffer / sort ping-pong
     int64_t table_cap = 0;
     int64_t max_keys = 0;
     int64_t cap = 0;
@@ -419,17 +419,23 @@ struct KeyedScratch {
                         slot_to_key);
         wfa_iota_u32(s, idx, n);
         uint32_t *os, *oi;
-        wfa_sort_pairs(s, slot, idx, slot_t, idx_t, hist, n, bits, &os, &oi);
-        idx_sorted = oi;
         if (vcol >= 0) {
+            // cast value to f32, then let it ride the sort as a second
+            // payload: folds read values coalesced (no random gather)
             int vdt = (int)db->schema.payload[vcol];
-            const void* vsrc = db->cols[vcol];
-            if (vdt != 2) {  // cast to f32 once (i64/bf16 lifted)
-                wfa_cast(s, vsrc, vdt, v_f32, 2, n);
-                vsrc = v_f32;
-            }
-            v_as_f32 = (const float*)vsrc;
+            if (vdt != 2)
+                wfa_cast(s, db->cols[vcol], vdt, v_f32, 2, n);
+            else
+                HIPCHK(hipMemcpyAsync(v_f32, db->cols[vcol], 4 * n,
+                                      hipMemcpyDeviceToDevice, s));
+            uint32_t* ov;
+            wfa_sort_pairs2(s, slot, idx, slot_t, idx_t, (uint32_t*)v_f32,
+                            (uint32_t*)v_sorted, hist, n, bits, &os, &oi, &ov);
+            v_as_f32 = (const float*)ov;
+        } else {
+            wfa_sort_pairs(s, slot, idx, slot_t, idx_t, hist, n, bits, &os, &oi);
         }
+        idx_sorted = oi;
         wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg);
         if (ctx.stats) ctx.stats->num_kernels += 5 + 3 * ((bits + 3) / 4);
     }
@@ -600,6 +606,7 @@ struct GpuFfatLogic : GpuLogicBase {
     uint32_t* st_head = nullptr;
     float* st_wsum = nullptr;
     int64_t* d_on = nullptr;
+    uint32_t* cb_nf = nullptr;  // CB fire-offset scratch
     // TB state
     float* tb_pend = nullptr;
     int64_t* tb_base = nullptr;
@@ -644,6 +651,7 @@ struct GpuFfatLogic : GpuLogicBase {
         wfa_fill_f32(stream, st_acc, ident, max_keys);
         wfa_fill_f32(stream, ring_or_tree, ident,
                      max_keys * (use_tree ? 2 * (1ll << ring_log2) : (1ll << ring_log2)));
+        cb_nf = (uint32_t*)A.get(4 * (out_cap + 1));
         if (tb) {
             int64_t Rp = 1ll << pend_log2;
             tb_pend = (float*)A.get(4 * max_keys * Rp);
@@ -686,7 +694,7 @@ struct GpuFfatLogic : GpuLogicBase {
 
     void chain_cb(Batch* db, Batch* ob, int64_t n, RuntimeCtx& ctx) {
         ks.group(stream, db, vcol, ctx);
-        uint32_t* nf = (uint32_t*)ks.v_sorted;  // reuse freed scratch
+        uint32_t* nf = cb_nf;
         wfa_ffat_fire_offsets(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                               pane_len, P, S, st_fill, st_head, nf, d_on);
         if (use_tree)

@@ -172,7 +172,8 @@ __global__ void k_rs_scan(uint32_t* a, int64_t m) {  // single block, 1024 thr
 __global__ void k_rs_scatter(const uint32_t* keys, const uint32_t* vals, int64_t n,
                              int shift, const uint32_t* hist, int64_t nblocks,
                              const uint32_t* dt, uint32_t* keys_out,
-                             uint32_t* vals_out) {
+                             uint32_t* vals_out, const uint32_t* vals2,
+                             uint32_t* vals2_out) {
     __shared__ uint32_t tc[WFA_THREADS][17];  // [thread][digit] padded
     __shared__ uint32_t gbase[16];
     __shared__ uint32_t dbase[16];
@@ -186,6 +187,7 @@ __global__ void k_rs_scatter(const uint32_t* keys, const uint32_t* vals, int64_t
     int64_t base = (int64_t)blockIdx.x * RS_PER_BLOCK + (int64_t)threadIdx.x * RS_IPT;
     uint32_t k_[RS_IPT];
     uint32_t v_[RS_IPT];
+    uint32_t v2_[RS_IPT];
     int cnt_local[16];
 #pragma unroll
     for (int d = 0; d < 16; ++d) cnt_local[d] = 0;
@@ -196,6 +198,7 @@ __global__ void k_rs_scatter(const uint32_t* keys, const uint32_t* vals, int64_t
         if (i < n) {
             k_[j] = keys[i];
             v_[j] = vals[i];
+            if (vals2) v2_[j] = vals2[i];
             cnt_local[(k_[j] >> shift) & 15]++;
             nit = j + 1;
         }
@@ -230,6 +233,7 @@ __global__ void k_rs_scatter(const uint32_t* keys, const uint32_t* vals, int64_t
         uint32_t pos = gbase[d] + pre[d] + own;
         keys_out[pos] = k_[j];
         vals_out[pos] = v_[j];
+        if (vals2) vals2_out[pos] = v2_[j];
     }
 }
 
@@ -307,7 +311,8 @@ __global__ void k_rs8_dbase(const uint32_t* dt, uint32_t* dbase) {  // 1 block
 __global__ void k_rs8_scatter(const uint32_t* keys, const uint32_t* vals,
                               int64_t n, int shift, const uint32_t* hist,
                               int64_t nblocks, const uint32_t* dbase,
-                              uint32_t* keys_out, uint32_t* vals_out) {
+                              uint32_t* keys_out, uint32_t* vals_out,
+                              const uint32_t* vals2, uint32_t* vals2_out) {
     __shared__ uint32_t gbase[256];                    // digit base, this block
     __shared__ uint32_t waveCnt[WFA_THREADS / 64][256];
     __shared__ uint32_t wavePre[WFA_THREADS / 64][256];
@@ -320,6 +325,7 @@ __global__ void k_rs8_scatter(const uint32_t* keys, const uint32_t* vals,
     __syncthreads();
     int64_t waveBase = (int64_t)blockIdx.x * RS8_PER_BLOCK + wave * RS8_PER_WAVE;
     uint32_t mk[RS8_IPT], mv[RS8_IPT], mr[RS8_IPT];  // key, val, (digit<<24|rank)
+    uint32_t mv2[RS8_IPT];
     int nit = 0;
 #pragma unroll
     for (int j = 0; j < RS8_IPT; ++j) {
@@ -340,6 +346,7 @@ __global__ void k_rs8_scatter(const uint32_t* keys, const uint32_t* vals,
             uint32_t rank = pre + (uint32_t)__popcll(mask & lt);
             mk[j] = k;
             mv[j] = v;
+            if (vals2) mv2[j] = vals2[i];
             mr[j] = (d << 24) | (rank & 0xFFFFFF);
             // group leader bumps the wave's digit counter
             if ((mask & lt) == 0) waveCnt[wave][d] = pre + (uint32_t)__popcll(mask);
@@ -361,15 +368,19 @@ __global__ void k_rs8_scatter(const uint32_t* keys, const uint32_t* vals,
         uint32_t pos = gbase[d] + wavePre[wave][d] + (mr[j] & 0xFFFFFF);
         keys_out[pos] = mk[j];
         vals_out[pos] = mv[j];
+        if (vals2) vals2_out[pos] = mv2[j];
     }
 }
 
-extern "C" void wfa_sort_pairs(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
-                               uint32_t* slot_tmp, uint32_t* idx_tmp, uint32_t* hist,
-                               int64_t n, int bits, uint32_t** out_slot,
-                               uint32_t** out_idx) {
+extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
+                                uint32_t* slot_tmp, uint32_t* idx_tmp,
+                                uint32_t* val2, uint32_t* val2_tmp,
+                                uint32_t* hist, int64_t n, int bits,
+                                uint32_t** out_slot, uint32_t** out_idx,
+                                uint32_t** out_val2) {
     hipStream_t st = (hipStream_t)s;
     uint32_t *ka = slot, *va = idx, *kb = slot_tmp, *vb = idx_tmp;
+    uint32_t *wa = val2, *wb = val2_tmp;
     if (bits > 4) {
         // 8-bit digits: fewer passes; hist is sized for 16*nblocks4 + 16
         // which covers 256*nblocks8 + 512 (RS8_PER_BLOCK = 2*RS_PER_BLOCK)
@@ -384,13 +395,15 @@ extern "C" void wfa_sort_pairs(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
             hipLaunchKernelGGL(k_rs8_scan, dim3(256), dim3(256), 0, st, hist, nb, dt);
             hipLaunchKernelGGL(k_rs8_dbase, dim3(1), dim3(256), 0, st, dt, dbase);
             hipLaunchKernelGGL(k_rs8_scatter, dim3(nb), dim3(WFA_THREADS), 0, st, ka,
-                               va, n, shift, hist, nb, dbase, kb, vb);
+                               va, n, shift, hist, nb, dbase, kb, vb, wa, wb);
             uint32_t* t;
             t = ka; ka = kb; kb = t;
             t = va; va = vb; vb = t;
+            t = wa; wa = wb; wb = t;
         }
         *out_slot = ka;
         *out_idx = va;
+        if (out_val2) *out_val2 = wa;
         return;
     }
     int64_t nblocks = wfa_sort_nblocks(n);
@@ -402,13 +415,23 @@ extern "C" void wfa_sort_pairs(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                            shift, hist, nblocks);
         hipLaunchKernelGGL(k_rs_scan16, dim3(16), dim3(256), 0, st, hist, nblocks, dt);
         hipLaunchKernelGGL(k_rs_scatter, dim3(nblocks), dim3(WFA_THREADS), 0, st, ka,
-                           va, n, shift, hist, nblocks, dt, kb, vb);
+                           va, n, shift, hist, nblocks, dt, kb, vb, wa, wb);
         uint32_t* t;
         t = ka; ka = kb; kb = t;
         t = va; va = vb; vb = t;
+        t = wa; wa = wb; wb = t;
     }
     *out_slot = ka;
     *out_idx = va;
+    if (out_val2) *out_val2 = wa;
+}
+
+extern "C" void wfa_sort_pairs(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
+                               uint32_t* slot_tmp, uint32_t* idx_tmp, uint32_t* hist,
+                               int64_t n, int bits, uint32_t** out_slot,
+                               uint32_t** out_idx) {
+    wfa_sort_pairs2(s, slot, idx, slot_tmp, idx_tmp, nullptr, nullptr, hist, n,
+                    bits, out_slot, out_idx, nullptr);
 }
 
 // ===== segment extraction =====
@@ -503,20 +526,20 @@ __global__ void k_seg_reduce(const uint32_t* seg_start, const uint32_t* seg_slot
         int64_t tmax = ts_orig ? ts_orig[idx_sorted[b]] : 0;
         if (vdt == 0) {
             const int64_t* v = (const int64_t*)v_orig;
-            int64_t acc = (comb == 3) ? 0 : v[idx_sorted[b]];
+            int64_t acc = (comb == 3) ? 0 : v[b];
             if (comb == 3) acc = e - b;
             else
                 for (int64_t i = b + 1; i < e; ++i) {
-                    int64_t x = v[idx_sorted[i]];
+                    int64_t x = v[i];
                     acc = (comb == 0) ? acc + x : (comb == 1 ? min(acc, x) : max(acc, x));
                 }
             ((int64_t*)out_val)[j] = acc;
         } else {
-            const float* v = (const float*)v_orig;
-            float acc = (comb == 3) ? (float)(e - b) : v[idx_sorted[b]];
+            const float* v = (const float*)v_orig;  // sorted with the keys
+            float acc = (comb == 3) ? (float)(e - b) : v[b];
             if (comb != 3)
                 for (int64_t i = b + 1; i < e; ++i) {
-                    float x = v[idx_sorted[i]];
+                    float x = v[i];
                     acc = (comb == 0) ? acc + x : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
                 }
             ((float*)out_val)[j] = acc;
@@ -640,7 +663,7 @@ __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
         int64_t w = fire_base[j];
         float* rg = ring + (size_t)slot * R;
         for (; i < e; ++i) {
-            float x = v_f32[idx_sorted[i]];
+            float x = v_f32[i];  // sorted with the keys (coalesced)
             acc = (comb == 0) ? acc + x : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
             if (++fill == (uint32_t)pane_len) {
                 // pane complete
@@ -718,7 +741,7 @@ __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_sl
         float* rg = ring + (size_t)slot * R;
         for (int64_t pos = i0; pos < e; pos += 64) {
             const uint32_t nchunk = (uint32_t)min((int64_t)64, e - pos);
-            float v = (lane < (int)nchunk) ? v_f32[idx_sorted[pos + lane]] : ident;
+            float v = (lane < (int)nchunk) ? v_f32[pos + lane] : ident;
             uint32_t rel = (lane < (int)nchunk) ? (fill + (uint32_t)lane) / L : ~0u;
             const uint32_t maxrel = (fill + nchunk - 1) / L;
             const uint32_t ncomplete = (fill + nchunk) / L;  // panes finished here
@@ -904,7 +927,7 @@ __global__ void k_tb_lift(const uint32_t* seg_start, const uint32_t* seg_slot,
                 atomicAdd(overflow, 1u);
                 continue;
             }
-            const float x = v_f32[idx_sorted[i]];
+            const float x = v_f32[i];  // sorted with the keys
             float* cell = &pd[(uint64_t)p & Pm];
             *cell = (comb == 0) ? *cell + x
                                 : (comb == 1 ? fminf(*cell, x) : fmaxf(*cell, x));
@@ -1080,7 +1103,7 @@ __global__ void k_ffat_tree(const uint32_t* seg_start, const uint32_t* seg_slot,
         int64_t w = fire_base[j];
         float* tr = tree + (size_t)slot * 2 * R;
         for (; i < e; ++i) {
-            float x = v_f32[idx_sorted[i]];
+            float x = v_f32[i];  // sorted with the keys (coalesced)
             acc = TCOMB(acc, x);
             if (++fill == (uint32_t)pane_len) {
                 uint32_t leaf = (head & Rm) + R;

@@ -50,6 +50,13 @@ void wfa_sort_pairs(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                     uint32_t* slot_tmp, uint32_t* idx_tmp, uint32_t* hist,
                     int64_t n, int bits,
                     uint32_t** out_slot, uint32_t** out_idx);
+// variant carrying a second 32-bit payload (e.g. f32 value bits) through
+// the sort so downstream folds read values coalesced instead of gathering
+void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
+                     uint32_t* slot_tmp, uint32_t* idx_tmp,
+                     uint32_t* val2, uint32_t* val2_tmp, uint32_t* hist,
+                     int64_t n, int bits, uint32_t** out_slot,
+                     uint32_t** out_idx, uint32_t** out_val2);
 int64_t wfa_sort_nblocks(int64_t n);
 int64_t wfa_sort_hist_u32(int64_t cap);  // hist scratch size in u32
 
