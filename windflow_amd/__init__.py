@@ -27,3 +27,7 @@ from .builders import (  # noqa: F401,E402
     MapReduce_Windows_Builder, Ffat_Windows_Builder, Interval_Join_Builder)
 from .pipegraph import PipeGraph, MultiPipe  # noqa: F401,E402
 from . import native  # noqa: F401,E402
+from .persistent import (  # noqa: F401,E402
+    P_Reduce_Builder, P_Map_Builder, P_Filter_Builder, P_Sink_Builder)
+from .kafka import (  # noqa: F401,E402
+    Connector_Source_Builder, Kafka_Source_Builder, Kafka_Sink_Builder)
