@@ -419,23 +419,21 @@ struct KeyedScratch {
                         slot_to_key);
         wfa_iota_u32(s, idx, n);
         uint32_t *os, *oi;
-        if (vcol >= 0) {
-            // cast value to f32, then let it ride the sort as a second
-            // payload: folds read values coalesced (no random gather)
-            int vdt = (int)db->schema.payload[vcol];
-            if (vdt != 2)
-                wfa_cast(s, db->cols[vcol], vdt, v_f32, 2, n);
-            else
-                HIPCHK(hipMemcpyAsync(v_f32, db->cols[vcol], 4 * n,
-                                      hipMemcpyDeviceToDevice, s));
-            uint32_t* ov;
-            wfa_sort_pairs2(s, slot, idx, slot_t, idx_t, (uint32_t*)v_f32,
-                            (uint32_t*)v_sorted, hist, n, bits, &os, &oi, &ov);
-            v_as_f32 = (const float*)ov;
-        } else {
-            wfa_sort_pairs(s, slot, idx, slot_t, idx_t, hist, n, bits, &os, &oi);
-        }
+        wfa_sort_pairs(s, slot, idx, slot_t, idx_t, hist, n, bits, &os, &oi);
         idx_sorted = oi;
+        if (vcol >= 0) {
+            // A/B-measured: reading values THROUGH idx_sorted in the folds
+            // beats carrying them as a sort payload (the wave fold hides the
+            // gather; the extra scatter writes did not pay — 11.5 vs 13.0
+            // B tuples/s).  wfa_sort_pairs2 remains for payload use cases.
+            int vdt = (int)db->schema.payload[vcol];
+            const void* vsrc = db->cols[vcol];
+            if (vdt != 2) {  // cast to f32 once (i64/bf16 lifted)
+                wfa_cast(s, vsrc, vdt, v_f32, 2, n);
+                vsrc = v_f32;
+            }
+            v_as_f32 = (const float*)vsrc;
+        }
         wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg);
         if (ctx.stats) ctx.stats->num_kernels += 5 + 3 * ((bits + 3) / 4);
     }

@@ -526,20 +526,20 @@ __global__ void k_seg_reduce(const uint32_t* seg_start, const uint32_t* seg_slot
         int64_t tmax = ts_orig ? ts_orig[idx_sorted[b]] : 0;
         if (vdt == 0) {
             const int64_t* v = (const int64_t*)v_orig;
-            int64_t acc = (comb == 3) ? 0 : v[b];
+            int64_t acc = (comb == 3) ? 0 : v[idx_sorted[b]];
             if (comb == 3) acc = e - b;
             else
                 for (int64_t i = b + 1; i < e; ++i) {
-                    int64_t x = v[i];
+                    int64_t x = v[idx_sorted[i]];
                     acc = (comb == 0) ? acc + x : (comb == 1 ? min(acc, x) : max(acc, x));
                 }
             ((int64_t*)out_val)[j] = acc;
         } else {
-            const float* v = (const float*)v_orig;  // sorted with the keys
-            float acc = (comb == 3) ? (float)(e - b) : v[b];
+            const float* v = (const float*)v_orig;
+            float acc = (comb == 3) ? (float)(e - b) : v[idx_sorted[b]];
             if (comb != 3)
                 for (int64_t i = b + 1; i < e; ++i) {
-                    float x = v[i];
+                    float x = v[idx_sorted[i]];
                     acc = (comb == 0) ? acc + x : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
                 }
             ((float*)out_val)[j] = acc;
@@ -663,7 +663,7 @@ __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
         int64_t w = fire_base[j];
         float* rg = ring + (size_t)slot * R;
         for (; i < e; ++i) {
-            float x = v_f32[i];  // sorted with the keys (coalesced)
+            float x = v_f32[idx_sorted[i]];
             acc = (comb == 0) ? acc + x : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
             if (++fill == (uint32_t)pane_len) {
                 // pane complete
@@ -741,7 +741,7 @@ __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_sl
         float* rg = ring + (size_t)slot * R;
         for (int64_t pos = i0; pos < e; pos += 64) {
             const uint32_t nchunk = (uint32_t)min((int64_t)64, e - pos);
-            float v = (lane < (int)nchunk) ? v_f32[pos + lane] : ident;
+            float v = (lane < (int)nchunk) ? v_f32[idx_sorted[pos + lane]] : ident;
             uint32_t rel = (lane < (int)nchunk) ? (fill + (uint32_t)lane) / L : ~0u;
             const uint32_t maxrel = (fill + nchunk - 1) / L;
             const uint32_t ncomplete = (fill + nchunk) / L;  // panes finished here
@@ -927,7 +927,7 @@ __global__ void k_tb_lift(const uint32_t* seg_start, const uint32_t* seg_slot,
                 atomicAdd(overflow, 1u);
                 continue;
             }
-            const float x = v_f32[i];  // sorted with the keys
+            const float x = v_f32[idx_sorted[i]];
             float* cell = &pd[(uint64_t)p & Pm];
             *cell = (comb == 0) ? *cell + x
                                 : (comb == 1 ? fminf(*cell, x) : fmaxf(*cell, x));
@@ -1103,7 +1103,7 @@ __global__ void k_ffat_tree(const uint32_t* seg_start, const uint32_t* seg_slot,
         int64_t w = fire_base[j];
         float* tr = tree + (size_t)slot * 2 * R;
         for (; i < e; ++i) {
-            float x = v_f32[i];  // sorted with the keys (coalesced)
+            float x = v_f32[idx_sorted[i]];
             acc = TCOMB(acc, x);
             if (++fill == (uint32_t)pane_len) {
                 uint32_t leaf = (head & Rm) + R;
