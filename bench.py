@@ -79,6 +79,39 @@ def build_a2a_graph(n_tuples, batch, n_keys, rank, world, device, dist_cfg):
     return g, snk
 
 
+def build_ffat_x_graph(n_tuples, batch, n_keys, win, slide, rank, world, device,
+                       dist_cfg):
+    """Keyed FFAT with a true cross-GPU keyby: source -> RCCL all-to-allv
+    exchange -> FFAT window.  Every key's window state lives on exactly one
+    rank (hash(key) % world)."""
+    import windflow_amd as wf
+    from windflow_amd import native_gpu
+    from windflow_amd.builders_gpu import (Source_GPU_Builder,
+                                           KeyBy_Exchange_GPU_Builder,
+                                           Ffat_Windows_GPU_Builder,
+                                           Sink_GPU_Builder)
+    src = (Source_GPU_Builder(
+        native_gpu.gpu_source(n_tuples, n_keys, batch, vdt=5, seed=42 + rank))
+        .withOutputSchema([5]).withOutputBatchSize(batch)
+        .withDevice(device).build())
+    ex = (KeyBy_Exchange_GPU_Builder(native_gpu.gpu_keyby_exchange())
+          .withOutputSchema([5]).withOutputBatchSize(3 * batch)
+          .withDevice(device).build())
+    ff = (Ffat_Windows_GPU_Builder(
+        native_gpu.gpu_ffat_windows(native_gpu.COMB_SUM, 0, win, slide,
+                                    max_keys=2 * n_keys))
+        .withOutputSchema([2]).withOutputBatchSize(3 * batch)
+        .withDevice(device).build())
+    snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).withDevice(device).build()
+    g = wf.PipeGraph("bench_ffat_x")
+    g.set_dist(*dist_cfg)
+    mp = g.add_source(src)
+    mp.chain(ex)
+    mp.chain(ff)
+    mp.chain_sink(snk)
+    return g, snk
+
+
 def build_cpu_graph(n_tuples, batch):
     import windflow_amd as wf
     from windflow_amd import native
@@ -106,7 +139,7 @@ def main():
                     help="distinct keys per rank")
     ap.add_argument("--win", type=int, default=1000)
     ap.add_argument("--slide", type=int, default=100)
-    ap.add_argument("--config", choices=["ffat", "a2a", "cpu"], default="ffat")
+    ap.add_argument("--config", choices=["ffat", "ffat_x", "a2a", "cpu"], default="ffat")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -115,13 +148,13 @@ def main():
     dist = None
     torch = None
     dist_cfg = None
-    if args.config in ("ffat", "a2a"):
+    if args.config != "cpu":
         import torch  # noqa: F811
         if world > 1:
             import torch.distributed as dist  # noqa: F811
             dist.init_process_group("nccl", rank=rank, world_size=world)
             torch.cuda.set_device(local_rank)
-        if args.config == "a2a":
+        if args.config in ("a2a", "ffat_x"):
             if world > 1:
                 from windflow_amd.dist import init_from_torch
                 dist_cfg = init_from_torch()
@@ -154,6 +187,10 @@ def main():
             if args.config == "a2a":
                 return build_a2a_graph(steps * B, B, args.keys, rank, world,
                                        local_rank, dist_cfg)
+            if args.config == "ffat_x":
+                return build_ffat_x_graph(steps * B, B, args.keys, args.win,
+                                          args.slide, rank, world, local_rank,
+                                          dist_cfg)
             return build_ffat_graph(steps * B, B, args.keys, args.win,
                                     args.slide, rank, world, local_rank)
         # warmup engine (also JIT-warms pools/streams/arena)
@@ -198,6 +235,7 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": {"ffat": "keyed_ffat_cb_window",
+                          "ffat_x": "rccl_keyby_ffat_cb_window",
                           "a2a": "map_gpu_rccl_keyby_reduce_gpu",
                           "cpu": "cpu_source_map_filter_sink"}[args.config],
                 "global_batch": B * max(world, 1),
