@@ -452,6 +452,26 @@ PYBIND11_MODULE(_core, m) {
 
     m.def("hash_key", [](uint64_t k) { return KeyByEmitter::mix(k); });
 
+#ifdef WFA_WITH_HIP
+    // debug hooks: round-trip device primitives for isolation tests
+    m.def("debug_sort_pairs",
+          [](py::array_t<uint32_t> keys, int bits) {
+              auto r = debug_sort_pairs_host(keys.data(), keys.shape(0), bits);
+              py::array_t<uint32_t> ks((py::ssize_t)r.first.size());
+              py::array_t<uint32_t> vs((py::ssize_t)r.second.size());
+              memcpy(ks.mutable_data(), r.first.data(), 4 * r.first.size());
+              memcpy(vs.mutable_data(), r.second.data(), 4 * r.second.size());
+              return py::make_tuple(ks, vs);
+          });
+    m.def("debug_key_slots",
+          [](py::array_t<uint64_t> keys, int64_t max_keys) {
+              auto r = debug_key_slots_host(keys.data(), keys.shape(0), max_keys);
+              py::array_t<uint32_t> s((py::ssize_t)r.size());
+              memcpy(s.mutable_data(), r.data(), 4 * r.size());
+              return s;
+          });
+#endif
+
     // persistent keyed state handle for Python P_* logic (reference
     // DBHandle<T>: get/modify/put with user serialize/deserialize)
     struct StateStore {

@@ -434,6 +434,12 @@ void state_cache_put(void* cache, uint64_t key, const std::string& v);
 void state_cache_flush(void* cache);
 int64_t state_kv_size(void* kv);
 
+// debug hooks (gpu_ops.cpp): device sort / hash-slot round trips
+std::pair<std::vector<uint32_t>, std::vector<uint32_t>> debug_sort_pairs_host(
+    const uint32_t* keys, int64_t n, int bits);
+std::vector<uint32_t> debug_key_slots_host(const uint64_t* keys, int64_t n,
+                                           int64_t max_keys);
+
 // RCCL bootstrap: rank 0 generates the id, broadcasts it out-of-band
 // (torch.distributed store), every rank passes it to Engine::rccl_id.
 std::string wfa_rccl_unique_id();

@@ -1116,6 +1116,52 @@ struct GpuExchangeLogic : GpuLogicBase {
     }
 };
 
+// ===== debug round trips (isolation tests on a real device) =====
+std::pair<std::vector<uint32_t>, std::vector<uint32_t>> debug_sort_pairs_host(
+    const uint32_t* keys, int64_t n, int bits) {
+    HIPCHK(hipSetDevice(0));
+    auto& A = arena(0);
+    uint32_t* d_k = (uint32_t*)A.get(4 * n);
+    uint32_t* d_v = (uint32_t*)A.get(4 * n);
+    uint32_t* d_kt = (uint32_t*)A.get(4 * n);
+    uint32_t* d_vt = (uint32_t*)A.get(4 * n);
+    uint32_t* d_h = (uint32_t*)A.get(4 * wfa_sort_hist_u32(n));
+    HIPCHK(hipMemcpy(d_k, keys, 4 * n, hipMemcpyHostToDevice));
+    wfa_iota_u32(nullptr, d_v, n);
+    uint32_t *ok, *ov;
+    wfa_sort_pairs(nullptr, d_k, d_v, d_kt, d_vt, d_h, n, bits, &ok, &ov);
+    std::vector<uint32_t> hk(n), hv(n);
+    HIPCHK(hipMemcpy(hk.data(), ok, 4 * n, hipMemcpyDeviceToHost));
+    HIPCHK(hipMemcpy(hv.data(), ov, 4 * n, hipMemcpyDeviceToHost));
+    A.put(d_k, 4 * n); A.put(d_v, 4 * n); A.put(d_kt, 4 * n); A.put(d_vt, 4 * n);
+    A.put(d_h, 4 * wfa_sort_hist_u32(n));
+    return {std::move(hk), std::move(hv)};
+}
+
+std::vector<uint32_t> debug_key_slots_host(const uint64_t* keys, int64_t n,
+                                           int64_t max_keys) {
+    HIPCHK(hipSetDevice(0));
+    auto& A = arena(0);
+    int64_t cap = 1;
+    while (cap < 2 * max_keys) cap <<= 1;
+    uint64_t* d_key = (uint64_t*)A.get(8 * n);
+    uint64_t* d_tk = (uint64_t*)A.get(8 * cap);
+    uint32_t* d_ts = (uint32_t*)A.get(4 * cap);
+    uint32_t* d_ns = (uint32_t*)A.get(64);
+    uint32_t* d_out = (uint32_t*)A.get(4 * n);
+    uint64_t* d_s2k = (uint64_t*)A.get(8 * max_keys);
+    HIPCHK(hipMemcpy(d_key, keys, 8 * n, hipMemcpyHostToDevice));
+    wfa_fill_u64(nullptr, d_tk, ~0ULL, cap);
+    wfa_fill_u32(nullptr, d_ts, ~0u, cap);
+    wfa_fill_u32(nullptr, d_ns, 0, 1);
+    wfa_key_to_slot(nullptr, d_key, n, d_tk, d_ts, d_ns, cap, d_out, d_s2k);
+    std::vector<uint32_t> out(n);
+    HIPCHK(hipMemcpy(out.data(), d_out, 4 * n, hipMemcpyDeviceToHost));
+    A.put(d_key, 8 * n); A.put(d_tk, 8 * cap); A.put(d_ts, 4 * cap);
+    A.put(d_ns, 64); A.put(d_out, 4 * n); A.put(d_s2k, 8 * max_keys);
+    return out;
+}
+
 std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::string& spec,
                                         const std::vector<double>& fp,
                                         const std::vector<int64_t>& ip, Engine* eng,
@@ -1173,6 +1219,13 @@ Batch* gpu_alloc_batch(Pool&) { throw std::runtime_error("built without HIP"); }
 void gpu_free_batch(Batch*) {}
 void gpu_resolve_count(Batch*) {}
 std::string wfa_rccl_unique_id() { throw std::runtime_error("built without HIP"); }
+std::pair<std::vector<uint32_t>, std::vector<uint32_t>> debug_sort_pairs_host(
+    const uint32_t*, int64_t, int) {
+    throw std::runtime_error("built without HIP");
+}
+std::vector<uint32_t> debug_key_slots_host(const uint64_t*, int64_t, int64_t) {
+    throw std::runtime_error("built without HIP");
+}
 std::shared_ptr<OpLogic> make_gpu_logic(const std::string&, const std::string&,
                                         const std::vector<double>&,
                                         const std::vector<int64_t>&, Engine*, int, int,
