@@ -717,7 +717,7 @@ struct GpuFfatLogic : GpuLogicBase {
         if (tb) {
             ks.group(stream, db, vcol, ctx);
             tb_round_with_count(db, n, db->watermark, out, ctx);
-            release(db);
+            release_after_use(db);
             return;
         }
         Batch* ob = get_dev();
@@ -742,7 +742,9 @@ struct GpuFfatLogic : GpuLogicBase {
         ob->count = -1;  // resolved by the consumer via gpu_resolve_count
         ob->watermark = db->watermark;
         ob->born_us = db->born_us;
-        release(db);
+        // async kernels still read db: record its reuse event on our stream
+        // (a plain release lets the producer refill it mid-fold)
+        release_after_use(db);
         record_ready(ob);
         out.emit(ob);
     }
