@@ -65,25 +65,31 @@ Pool::~Pool() {
     for (Batch* b : free_list) free_batch(b);
 }
 
-Batch* Pool::get() {
-    {
-        std::lock_guard<std::mutex> g(mu);
-        if (!free_list.empty()) {
-            Batch* b = free_list.back();
-            free_list.pop_back();
-            b->count = 0;
-            b->watermark = 0;
-            b->punct = false;
-            b->stream_tag = -1;
-            b->refcnt.store(1, std::memory_order_relaxed);
-            return b;
-        }
-    }
+Batch* Pool::try_pop() {
+    std::lock_guard<std::mutex> g(mu);
+    if (free_list.empty()) return nullptr;
+    Batch* b = free_list.back();
+    free_list.pop_back();
+    b->count = 0;
+    b->watermark = 0;
+    b->punct = false;
+    b->stream_tag = -1;
+    b->refcnt.store(1, std::memory_order_relaxed);
+    return b;
+}
+
+Batch* Pool::make_new() {
     Batch* b = (loc == Loc::DEVICE) ? gpu_alloc_batch(*this)
                                     : alloc_batch(schema, capacity, pinned);
     b->pool = this;
+    b->refcnt.store(1, std::memory_order_relaxed);
     live.fetch_add(1, std::memory_order_relaxed);
     return b;
+}
+
+Batch* Pool::get() {
+    if (Batch* b = try_pop()) return b;
+    return make_new();
 }
 
 void Pool::put(Batch* b) {

@@ -113,6 +113,8 @@ struct Pool {
     ~Pool();
 
     Batch* get();               // pop or allocate
+    Batch* try_pop();           // pop only (null when empty)
+    Batch* make_new();          // always allocate
     void put(Batch* b);         // return to freelist
 };
 

@@ -204,12 +204,36 @@ struct GpuLogicBase : OpLogic {
         release(b);
     }
 
-    // pool get + wait for the previous user's recorded event
+    // pool get, preferring a batch whose reuse event already completed —
+    // otherwise allocate a fresh one up to a bounded depth so producers
+    // never serialize against consumers still reading a recycled batch
+    // (288 GB HBM makes a deep rotation cheap; reference recycling_gpu
+    // instead spin-waits under memory pressure).
     Batch* get_dev() {
-        Batch* b = dev_pool->get();
-        if (b->ready_event)
-            HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)b->ready_event, 0));
-        return b;
+        constexpr int MAX_DEPTH = 8;
+        Batch* chosen = nullptr;
+        Batch* skipped[MAX_DEPTH];
+        int nskip = 0;
+        while (nskip < MAX_DEPTH) {
+            Batch* b = dev_pool->try_pop();
+            if (!b) break;
+            if (!b->ready_event ||
+                hipEventQuery((hipEvent_t)b->ready_event) == hipSuccess) {
+                chosen = b;
+                break;
+            }
+            skipped[nskip++] = b;
+        }
+        for (int i = 0; i < nskip; ++i) dev_pool->put(skipped[i]);
+        if (!chosen) {
+            if (nskip > 0 && dev_pool->live.load(std::memory_order_relaxed) >= MAX_DEPTH)
+                chosen = dev_pool->get();  // bounded: reuse, stream-waits below
+            else
+                chosen = dev_pool->make_new();
+        }
+        if (chosen->ready_event)
+            HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)chosen->ready_event, 0));
+        return chosen;
     }
 
     // H2D staging: host batch -> fresh device batch (reference

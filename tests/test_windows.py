@@ -280,6 +280,62 @@ def test_interval_join_dp():
     assert got == oracle_join(1200, 5, -10, 10)
 
 
+# ---------------- lateness / out-of-order ----------------
+def run_ooo_graph(win_op, stream_len=3000, disorder=30, batch=100, seed=5):
+    """Source emitting bounded-disorder event times: ts values are a
+    permutation of 1..N where |perm[i] - i| <= disorder; the watermark
+    lags max-emitted-ts by `disorder` (an honest source bound)."""
+    import numpy as np
+    rng = np.random.default_rng(seed)
+    ts = np.arange(1, stream_len + 1, dtype=np.int64)
+    for s in range(0, stream_len, disorder):
+        rng.shuffle(ts[s:s + disorder])
+    state = dict(pos=0)
+
+    def src(replica, par):
+        p = state['pos']
+        if p >= stream_len:
+            return None
+        n = min(batch, stream_len - p)
+        t = ts[p:p + n]
+        state['pos'] += n
+        return dict(ts=t, key=(t % 7).astype(np.uint64), c0=t,
+                    watermark=int(ts[:p + n].max()) - disorder)
+
+    g = wf.PipeGraph("ooo")
+    mp = g.add_source(Source_Builder(src).withParallelism(1)
+                      .withOutputSchema([0]).withOutputBatchSize(batch).build())
+    mp.add(win_op)
+    col = Collect()
+    mp.add_sink(Sink_Builder(col).withParallelism(1).build())
+    g.run()
+    return col.rows
+
+
+def test_tb_lateness_absorbs_disorder():
+    """With lateness >= source disorder, out-of-order arrival must not
+    change any window result (firing is gated on watermark - lateness)."""
+    win, slide, disorder = 120, 30, 30
+    rows = run_ooo_graph(Keyed_Windows_Builder(func=("sum", 0))
+                         .withTBWindows(win, slide).withLateness(disorder)
+                         .withOutputSchema([0]).build())
+    per = defaultdict(list)
+    for v in range(1, 3001):
+        per[v % 7].append((v, v))
+    assert got_counter(rows) == oracle_tb(per, win, slide, "sum")
+
+
+def test_tb_ffat_lateness_absorbs_disorder():
+    win, slide, disorder = 120, 30, 30
+    rows = run_ooo_graph(Ffat_Windows_Builder(comb=("sum", 0))
+                         .withTBWindows(win, slide).withLateness(disorder)
+                         .withOutputSchema([0]).build())
+    per = defaultdict(list)
+    for v in range(1, 3001):
+        per[v % 7].append((v, v))
+    assert got_counter(rows) == oracle_tb(per, win, slide, "sum")
+
+
 # ---------------- deterministic mode ----------------
 def test_windows_deterministic_mode():
     win, slide = 40, 10
