@@ -231,7 +231,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if args.config == "ffat" else "int64",
+            "dtype": {"ffat": "bf16", "ffat_x": "bf16", "a2a": "f32",
+                      "cpu": "int64"}[args.config],
             "data": "synthetic",
             "config": {
                 "model": {"ffat": "keyed_ffat_cb_window",

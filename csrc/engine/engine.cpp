@@ -367,6 +367,12 @@ void Replica::run() {
                 release(b);
                 if (!logic->on_punct(wm, ectx, rctx))
                     for (auto& e : emitters) e->punct(wm);
+            } else if (b->loc == Loc::DEVICE && !logic->accepts_device()) {
+                release(b);
+                engine->abort.store(true);
+                throw std::runtime_error(
+                    "CPU operator received a device batch — route through "
+                    "gpu_to_host() or place the operator on the GPU");
             } else {
                 logic->process(b, ectx, rctx);
             }
@@ -505,7 +511,24 @@ void Engine::build() {
 void Engine::start() {
     for (auto& r : replicas) {
         Replica* rp = r.get();
-        rp->th = std::thread([rp] { rp->run(); });
+        rp->th = std::thread([rp] {
+            try {
+                rp->run();
+            } catch (const std::exception& e) {
+                // fail the whole graph loudly but cleanly (the reference
+                // exits the process; we abort the engine and surface the
+                // error through Engine::wait)
+                {
+                    std::lock_guard<std::mutex> g(rp->engine->error_mu);
+                    if (rp->engine->first_error.empty())
+                        rp->engine->first_error =
+                            rp->engine->ops[rp->op_id].name + "[" +
+                            std::to_string(rp->idx) + "]: " + e.what();
+                }
+                fprintf(stderr, "[windflow_amd] replica error: %s\n", e.what());
+                rp->engine->abort.store(true);
+            }
+        });
     }
 }
 
@@ -520,6 +543,12 @@ void Engine::start_gated() {
 void Engine::wait() {
     for (auto& r : replicas)
         if (r->th.joinable()) r->th.join();
+    std::lock_guard<std::mutex> g(error_mu);
+    if (!first_error.empty()) {
+        std::string e = first_error;
+        first_error.clear();
+        throw std::runtime_error("graph failed: " + e);
+    }
 }
 
 }  // namespace wfa

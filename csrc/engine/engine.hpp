@@ -261,6 +261,10 @@ struct OpLogic {
     virtual bool on_punct(int64_t wm, EmitCtx& out, RuntimeCtx& ctx) { return false; }
     virtual void on_eos(EmitCtx& out, RuntimeCtx& ctx) {}
     virtual bool is_source() const { return false; }
+    // CPU logics dereference batch columns on the host; device batches
+    // must cross gpu_to_host first.  GPU logics override to true and the
+    // replica loop turns violations into a clear error, not a segfault.
+    virtual bool accepts_device() const { return false; }
     // Source: fill-and-emit loop; return false when exhausted.
     virtual bool source_step(EmitCtx& out, RuntimeCtx& ctx) { return false; }
 };
@@ -310,6 +314,7 @@ struct ChainLogic : OpLogic {
     std::vector<std::unique_ptr<Emitter>> glue; // stage->next adapters
     void wire(const std::vector<Pool*>& pools, EmitCtx& final_ctx, RuntimeCtx& rctx);
     bool is_source() const override { return stages.front()->is_source(); }
+    bool accepts_device() const override { return stages.front()->accepts_device(); }
     bool source_step(EmitCtx&, RuntimeCtx& ctx) override {
         return stages.front()->source_step(ctxs[0], ctx);
     }
@@ -355,6 +360,8 @@ struct Engine {
     std::vector<std::unique_ptr<SpscQueue>> queues;
     std::atomic<bool> abort{false};
     std::atomic<int64_t> dropped_tuples{0};
+    std::mutex error_mu;
+    std::string first_error;  // first replica failure (rethrown by wait())
     // gated start: threads spawn + logics warm, then block until open_gate()
     std::atomic<int> gate{0};
     std::atomic<int> warmed{0};

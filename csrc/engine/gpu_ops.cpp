@@ -182,6 +182,7 @@ struct GpuLogicBase : OpLogic {
     }
 
     void warm(RuntimeCtx&) override { ensure_init(); }
+    bool accepts_device() const override { return true; }
 
     // make the producing stream's work visible to our stream
     void wait_ready(Batch* b) {

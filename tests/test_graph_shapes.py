@@ -247,3 +247,20 @@ def test_stats_records():
     assert g.sink_count(snk) == 2000
     js = g.stats_json()
     assert 'operators' in js
+
+
+def test_replica_error_propagates():
+    """A failing user logic aborts the graph and surfaces the error via
+    run() instead of killing the process (improvement over the reference,
+    which exits(EXIT_FAILURE))."""
+    import pytest
+
+    def bad_source(replica, par):
+        raise ValueError("boom")
+
+    g = wf.PipeGraph("err")
+    mp = g.add_source(wf.Source_Builder(bad_source).withParallelism(1)
+                      .withOutputSchema([0]).build())
+    mp.add_sink(wf.Sink_Builder(native.count_sink()).withParallelism(1).build())
+    with pytest.raises(RuntimeError, match="boom"):
+        g.run()
