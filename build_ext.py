@@ -107,10 +107,10 @@ def build(force=False):
             pass
         if torchlib:
             # same single-runtime rule applies to RCCL (torch bundles its own)
-            hip_link = [f"-L{torchlib}", "-l:libamdhip64.so", "-l:librccl.so",
+            hip_link = [f"-L{torchlib}", "-l:libamdhip64.so", "-l:librccl.so", "-l:libhiprtc.so",
                         f"-Wl,-rpath,{torchlib}"]
         else:
-            hip_link = [f"-L{ROCM}/lib", "-lamdhip64", "-lrccl",
+            hip_link = [f"-L{ROCM}/lib", "-lamdhip64", "-lrccl", "-lhiprtc",
                         f"-Wl,-rpath,{ROCM}/lib"]
         link = ["g++", "-shared", "-o", out] + objs + hip_link + ["-pthread"]
         subprocess.check_call(link)

@@ -19,6 +19,7 @@
 
 #ifdef WFA_WITH_HIP
 #include <hip/hip_runtime.h>
+#include <hip/hiprtc.h>
 #include <rccl/rccl.h>
 
 #include "../hip/wfa_kernels.h"
@@ -461,6 +462,170 @@ struct KeyedScratch {
         }
         wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg);
         if (ctx.stats) ctx.stats->num_kernels += 5 + 3 * ((bits + 3) / 4);
+    }
+};
+
+// ===== hiprtc JIT: user device logic (reference MAP_GPU/FILTER_GPU
+// accept arbitrary __device__ lambdas; here a C expression over
+// (v, ts, key) is runtime-compiled for gfx950 and cached) =====
+#define RTCCHK(x)                                                              \
+    do {                                                                       \
+        hiprtcResult r_ = (x);                                                 \
+        if (r_ != HIPRTC_SUCCESS)                                              \
+            throw std::runtime_error(std::string("hiprtc error: ") +           \
+                                     hiprtcGetErrorString(r_) + " at " #x);    \
+    } while (0)
+
+struct JitKernel {
+    hipModule_t mod = nullptr;
+    hipFunction_t fn = nullptr;
+};
+
+static JitKernel jit_compile(const std::string& src, int device) {
+    static std::mutex mu;
+    static std::map<std::string, JitKernel> cache;
+    std::lock_guard<std::mutex> g(mu);
+    auto it = cache.find(src);
+    if (it != cache.end()) return it->second;
+    hiprtcProgram prog;
+    RTCCHK(hiprtcCreateProgram(&prog, src.c_str(), "wfa_jit.hip", 0, nullptr,
+                               nullptr));
+    hipDeviceProp_t props;
+    HIPCHK(hipGetDeviceProperties(&props, device));
+    std::string arch = std::string("--offload-arch=") + props.gcnArchName;
+    const char* opts[] = {arch.c_str(), "-O3"};
+    hiprtcResult cr = hiprtcCompileProgram(prog, 2, opts);
+    if (cr != HIPRTC_SUCCESS) {
+        size_t lsz = 0;
+        hiprtcGetProgramLogSize(prog, &lsz);
+        std::string log(lsz, '\0');
+        hiprtcGetProgramLog(prog, log.data());
+        hiprtcDestroyProgram(&prog);
+        throw std::runtime_error("JIT device logic failed to compile:\n" + log);
+    }
+    size_t csz = 0;
+    RTCCHK(hiprtcGetCodeSize(prog, &csz));
+    std::vector<char> code(csz);
+    RTCCHK(hiprtcGetCode(prog, code.data()));
+    hiprtcDestroyProgram(&prog);
+    JitKernel k;
+    HIPCHK(hipModuleLoadData(&k.mod, code.data()));
+    HIPCHK(hipModuleGetFunction(&k.fn, k.mod, "wfa_jit"));
+    cache[src] = k;
+    return k;
+}
+
+static std::string jit_type(DType d) {
+    switch (d) {
+        case DType::I64: return "long long";
+        case DType::F64: return "double";
+        case DType::F32: return "float";
+        case DType::U64: return "unsigned long long";
+        case DType::I32: return "int";
+        default: throw std::runtime_error("jit: unsupported column dtype");
+    }
+}
+
+// expr sees: v (value column, mutable type T), ts (long long), key (u64)
+static std::string jit_source(const std::string& expr, DType dt, bool filter) {
+    std::string T = jit_type(dt);
+    std::string body = filter
+        ? "        flags[i] = (" + expr + ") ? 1u : 0u;\n"
+        : "        v_[i] = (" + expr + ");\n";
+    return "typedef long long i64; typedef unsigned long long u64;\n"
+           "extern \"C\" __global__ void wfa_jit(" + T + "* v_, const i64* ts_,\n"
+           "        const u64* key_, unsigned int* flags, i64 n) {\n"
+           "    for (i64 i = blockIdx.x * (i64)blockDim.x + threadIdx.x; i < n;\n"
+           "         i += gridDim.x * (i64)blockDim.x) {\n"
+           "        " + T + " v = v_[i]; i64 ts = ts_[i]; u64 key = key_[i];\n"
+           "        (void)v; (void)ts; (void)key;\n" + body +
+           "    }\n}\n";
+}
+
+static void jit_launch(const JitKernel& k, hipStream_t stream, void* col,
+                       const int64_t* ts, const uint64_t* key, uint32_t* flags,
+                       int64_t n) {
+    void* args[] = {&col, (void*)&ts, (void*)&key, &flags, &n};
+    unsigned nb = (unsigned)std::min<int64_t>((n + 255) / 256, 2048);
+    HIPCHK(hipModuleLaunchKernel(k.fn, nb, 1, 1, 256, 1, 1, 0, stream, args,
+                                 nullptr));
+}
+
+struct GpuJitMapLogic : GpuLogicBase {
+    std::string expr;
+    int col;
+    JitKernel k;
+    GpuJitMapLogic(std::string e, int c, int dev, Schema os) : expr(std::move(e)), col(c) {
+        device = dev;
+        out_schema = os;
+    }
+    void init_device() override {
+        k = jit_compile(jit_source(expr, out_schema.payload[col], false), device);
+    }
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        Batch* db = input_on_device(in, ctx);
+        jit_launch(k, stream, db->cols[col], db->ts, db->key, nullptr, db->count);
+        record_ready(db);
+        if (ctx.stats) ctx.stats->num_kernels++;
+        out.emit(db);
+    }
+};
+
+struct GpuJitFilterLogic : GpuLogicBase {
+    std::string expr;
+    int col;
+    JitKernel k;
+    uint32_t* d_flags = nullptr;
+    uint32_t* d_scan = nullptr;
+    int64_t* d_cnt = nullptr;
+    void** d_colptrs = nullptr;
+    int* d_esize = nullptr;
+    GpuJitFilterLogic(std::string e, int c, int dev, Schema os, int64_t cap)
+        : expr(std::move(e)), col(c) {
+        device = dev;
+        out_schema = os;
+        out_cap = cap;
+    }
+    void init_device() override {
+        k = jit_compile(jit_source(expr, out_schema.payload[col], true), device);
+        auto& A = arena(device);
+        d_flags = (uint32_t*)A.get(4 * out_cap);
+        d_scan = (uint32_t*)A.get(4 * (out_cap / 2048 + 2));
+        d_cnt = (int64_t*)A.get(64);
+        size_t nc = out_schema.payload.size();
+        d_colptrs = (void**)A.get(16 * (nc + 1));
+        d_esize = (int*)A.get(4 * (nc + 1));
+        std::vector<int> es;
+        for (auto d : out_schema.payload) es.push_back((int)dsize(d));
+        HIPCHK(hipMemcpy(d_esize, es.data(), 4 * nc, hipMemcpyHostToDevice));
+    }
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        Batch* db = input_on_device(in, ctx);
+        int64_t n = db->count;
+        jit_launch(k, stream, db->cols[col], db->ts, db->key, d_flags, n);
+        Batch* ob = get_dev();
+        size_t nc = db->cols.size();
+        std::vector<void*> ptrs(2 * nc);
+        for (size_t c = 0; c < nc; ++c) {
+            ptrs[c] = db->cols[c];
+            ptrs[nc + c] = ob->cols[c];
+        }
+        HIPCHK(hipMemcpyAsync(d_colptrs, ptrs.data(), 8 * 2 * nc,
+                              hipMemcpyHostToDevice, stream));
+        wfa_compact(stream, n, d_flags, d_scan, db->ts, ob->ts, db->key, ob->key,
+                    (const void* const*)d_colptrs, (void* const*)(d_colptrs + nc),
+                    d_esize, (int)nc, d_cnt);
+        HIPCHK(hipMemcpyAsync(ob->lazy_count, d_cnt, 8, hipMemcpyDeviceToHost, stream));
+        ob->count = -1;
+        ob->watermark = db->watermark;
+        ob->stream_tag = db->stream_tag;
+        ob->born_us = db->born_us;
+        if (ctx.stats) ctx.stats->num_kernels += 4;
+        release_after_use(db);
+        record_ready(ob);
+        out.emit(ob);
     }
 };
 
@@ -1205,6 +1370,13 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
     if (kind == "gpu_filter")
         return std::make_shared<GpuFilterLogic>((int)ip[0], (int)ip[1], fp[0], fp[1],
                                                 device, os, out_batch);
+    if (kind == "gpu_jit_map")
+        // spec = C expression over (v, ts, key); ip: [col]
+        return std::make_shared<GpuJitMapLogic>(spec, (int)(ip.empty() ? 0 : ip[0]),
+                                                device, os);
+    if (kind == "gpu_jit_filter")
+        return std::make_shared<GpuJitFilterLogic>(spec, (int)(ip.empty() ? 0 : ip[0]),
+                                                   device, os, out_batch);
     if (kind == "gpu_map_keyed")
         // ip: [spec, col, max_keys]; fp: [a, b]
         return std::make_shared<GpuStatefulMapLogic>((int)ip[0], (int)ip[1], fp[0],

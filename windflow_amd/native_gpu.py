@@ -32,6 +32,21 @@ def gpu_gt_filter(col=0, thr=0.5, dtype=2):
     return NativeLogic("gpu_filter", "", [float(thr), 0.0], [spec, col])
 
 
+def gpu_jit_map(expr, col=0):
+    """Custom device map, runtime-compiled with hiprtc for the local arch
+    (parity with the reference's __device__ lambda MAP_GPU).  `expr` is a C
+    expression over `v` (the column value, mutable type), `ts` (i64) and
+    `key` (u64); the result is written back to the column, e.g.
+    gpu_jit_map("v * v + 1.0f", col=0)."""
+    return NativeLogic("gpu_jit_map", expr, [], [col])
+
+
+def gpu_jit_filter(expr, col=0):
+    """Custom device predicate (reference FILTER_GPU lambda): keep rows
+    where `expr` is true, e.g. gpu_jit_filter("v > 0.5f && key % 2 == 0")."""
+    return NativeLogic("gpu_jit_filter", expr, [], [col])
+
+
 def gpu_keyed_running_sum(col=0, max_keys=1 << 16):
     """stateful map: per-key running sum written in place (key-order walk)."""
     return NativeLogic("gpu_map_keyed", "", [0.0, 0.0], [1, col, max_keys])
