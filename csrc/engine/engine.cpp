@@ -248,18 +248,19 @@ Batch* KSlackCollector::next() {
                     continue;
                 }
                 int64_t bts = b->count ? b->ts[0] : b->watermark;
-                if (bts < t_curr) {
-                    int64_t d = t_curr - bts;
-                    if (d > K) K = d;  // adapt slack
-                }
-                // late beyond the adapted slack behind what was already
+                // late beyond the CURRENT slack behind what was already
                 // released: drop + account (reference kslack_collector.hpp
-                // feeds the PipeGraph's atomic_num_dropped)
+                // checks against the pre-adaptation K and feeds the
+                // PipeGraph's atomic_num_dropped)
                 if (bts + K < last_rel_ts) {
                     if (dropped)
                         dropped->fetch_add(b->count, std::memory_order_relaxed);
                     release(b);
                     continue;
+                }
+                if (bts < t_curr) {
+                    int64_t d = t_curr - bts;
+                    if (d > K) K = d;  // adapt slack
                 }
                 t_curr = std::max(t_curr, bts);
                 buf.emplace_back(bts, b);

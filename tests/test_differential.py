@@ -103,3 +103,39 @@ def test_reduce_keyed_running_sum():
     # single key 0 (v % 1 == 0): running sums are prefix sums of 1..n
     exp = sum(sum(range(1, k + 1)) for k in range(1, n + 1))
     assert g.sink_sum(snk) == exp
+
+
+def test_kslack_drops_extreme_disorder():
+    """PROBABILISTIC mode with disorder far beyond the adapted slack: late
+    tuples are dropped AND accounted (reference KSlack ->
+    PipeGraph::getNumDroppedTuples)."""
+    import numpy as np
+    n, batch = 20000, 100
+    # a long in-order stream (slack stays ~0, releases advance), then a
+    # burst of ancient timestamps: later than any current slack -> dropped
+    ts = np.arange(1, n + 1, dtype=np.int64)
+    ts[-1000:] = np.arange(1, 1001)
+    state = dict(pos=0)
+
+    def src(replica, par):
+        p = state['pos']
+        if p >= n:
+            return None
+        m = min(batch, n - p)
+        t = ts[p:p + m]
+        state['pos'] += m
+        return dict(ts=t, key=np.zeros(m, dtype=np.uint64), c0=t,
+                    watermark=int(t.max()))
+
+    g = wf.PipeGraph("ks", wf.ExecutionMode.PROBABILISTIC)
+    mp = g.add_source(wf.Source_Builder(src).withParallelism(1)
+                      .withOutputSchema([0]).withOutputBatchSize(batch).build())
+    # a shuffle installs the KSlack collector in front of the sink
+    snk = wf.Sink_Builder(native.count_sink()).withParallelism(2).build()
+    snk.key_extractor = 'carried'
+    mp.add_sink(snk)
+    g.run()
+    got = g.sink_count(snk)
+    dropped = g.getNumDroppedTuples()
+    assert got + dropped == n, (got, dropped)
+    assert dropped > 0
