@@ -387,8 +387,7 @@ struct GpuFilterLogic : GpuLogicBase {
 
 // ===== shared keyed front half: slot -> sort -> gather -> segments =====
 struct KeyedScratch {
-    uint64_t* tkeys = nullptr;
-    uint32_t* tslots = nullptr;
+    uint64_t* tab = nullptr;        // packed (key, slot) 16 B entries
     uint32_t* d_nslots = nullptr;
     uint64_t* slot_to_key = nullptr;
     uint32_t *slot = nullptr, *idx = nullptr, *slot_t = nullptr, *idx_t = nullptr;
@@ -411,8 +410,7 @@ struct KeyedScratch {
         bits = 1;
         while ((1ll << bits) < mk + 1) ++bits;
         auto& A = arena(dev);
-        tkeys = (uint64_t*)A.get(8 * table_cap);
-        tslots = (uint32_t*)A.get(4 * table_cap);
+        tab = (uint64_t*)A.get(16 * table_cap);
         d_nslots = (uint32_t*)A.get(64);
         slot_to_key = (uint64_t*)A.get(8 * mk);
         slot = (uint32_t*)A.get(4 * cap);
@@ -425,8 +423,7 @@ struct KeyedScratch {
         d_nseg = (int64_t*)A.get(64);
         v_sorted = (float*)A.get(4 * cap);   // reused as fire-offset scratch
         v_f32 = (float*)A.get(4 * cap);
-        wfa_fill_u64(s, tkeys, ~0ULL, table_cap);
-        wfa_fill_u32(s, tslots, ~0u, table_cap);
+        wfa_fill_u64(s, tab, ~0ULL, 2 * table_cap);
         wfa_fill_u32(s, d_nslots, 0, 1);
     }
 
@@ -441,7 +438,7 @@ struct KeyedScratch {
         if (n > cap)
             throw std::runtime_error("batch larger than keyed scratch capacity — "
                                      "set the GPU op's out_batch >= upstream batch");
-        wfa_key_to_slot(s, db->key, n, tkeys, tslots, d_nslots, table_cap, slot,
+        wfa_key_to_slot(s, db->key, n, tab, d_nslots, table_cap, slot,
                         slot_to_key);
         wfa_iota_u32(s, idx, n);
         uint32_t *os, *oi;
@@ -1337,19 +1334,17 @@ std::vector<uint32_t> debug_key_slots_host(const uint64_t* keys, int64_t n,
     int64_t cap = 1;
     while (cap < 2 * max_keys) cap <<= 1;
     uint64_t* d_key = (uint64_t*)A.get(8 * n);
-    uint64_t* d_tk = (uint64_t*)A.get(8 * cap);
-    uint32_t* d_ts = (uint32_t*)A.get(4 * cap);
+    uint64_t* d_tab = (uint64_t*)A.get(16 * cap);
     uint32_t* d_ns = (uint32_t*)A.get(64);
     uint32_t* d_out = (uint32_t*)A.get(4 * n);
     uint64_t* d_s2k = (uint64_t*)A.get(8 * max_keys);
     HIPCHK(hipMemcpy(d_key, keys, 8 * n, hipMemcpyHostToDevice));
-    wfa_fill_u64(nullptr, d_tk, ~0ULL, cap);
-    wfa_fill_u32(nullptr, d_ts, ~0u, cap);
+    wfa_fill_u64(nullptr, d_tab, ~0ULL, 2 * cap);
     wfa_fill_u32(nullptr, d_ns, 0, 1);
-    wfa_key_to_slot(nullptr, d_key, n, d_tk, d_ts, d_ns, cap, d_out, d_s2k);
+    wfa_key_to_slot(nullptr, d_key, n, d_tab, d_ns, cap, d_out, d_s2k);
     std::vector<uint32_t> out(n);
     HIPCHK(hipMemcpy(out.data(), d_out, 4 * n, hipMemcpyDeviceToHost));
-    A.put(d_key, 8 * n); A.put(d_tk, 8 * cap); A.put(d_ts, 4 * cap);
+    A.put(d_key, 8 * n); A.put(d_tab, 16 * cap);
     A.put(d_ns, 64); A.put(d_out, 4 * n); A.put(d_s2k, 8 * max_keys);
     return out;
 }

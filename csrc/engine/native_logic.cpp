@@ -272,6 +272,8 @@ struct SplitModI64 : OpLogic {
 // round-robin whole batches across branches (split_gpu-style replication-free)
 struct SplitRR : OpLogic {
     size_t rr = 0;
+    // forwards pointers only — safe for device batches (split_gpu)
+    bool accepts_device() const override { return true; }
     void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
         out.emit_to(rr, b);
         rr = (rr + 1) % out.n_branches();

@@ -36,8 +36,11 @@ __device__ __forceinline__ uint64_t mix64s(uint64_t k) {
 // ===== key -> dense slot (open addressing, device-scope atomics) =====
 #define WFA_EMPTY_KEY (~0ULL)
 
-__global__ void k_key_to_slot(const uint64_t* key, int64_t n, uint64_t* tkeys,
-                              uint32_t* tslots, uint32_t* n_slots, int64_t cap,
+// Entries are PACKED 16 B (key at [2p], slot at [2p+1]) so the common
+// first-probe hit touches ONE cache line instead of two (tkeys+tslots
+// halves split across arrays cost ~2x the L2 traffic per lookup).
+__global__ void k_key_to_slot(const uint64_t* key, int64_t n, uint64_t* tab,
+                              uint32_t* n_slots, int64_t cap,
                               uint32_t* slot_out, uint64_t* slot_to_key) {
     const uint64_t mask = (uint64_t)cap - 1;  // cap is a power of two
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
@@ -45,18 +48,19 @@ __global__ void k_key_to_slot(const uint64_t* key, int64_t n, uint64_t* tkeys,
         uint64_t k = key[i];
         uint64_t p = mix64s(k) & mask;
         for (;;) {
-            uint64_t cur = __hip_atomic_load(&tkeys[p], __ATOMIC_RELAXED,
+            uint64_t cur = __hip_atomic_load(&tab[2 * p], __ATOMIC_RELAXED,
                                              __HIP_MEMORY_SCOPE_AGENT);
             if (cur == k) break;
             if (cur == WFA_EMPTY_KEY) {
                 uint64_t expected = WFA_EMPTY_KEY;
                 bool won = __hip_atomic_compare_exchange_strong(
-                    &tkeys[p], &expected, k, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+                    &tab[2 * p], &expected, k, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
                     __HIP_MEMORY_SCOPE_AGENT);
                 if (won) {  // we inserted
                     uint32_t slot = atomicAdd(n_slots, 1u);
                     slot_to_key[slot] = k;
-                    __hip_atomic_store(&tslots[p], slot, __ATOMIC_RELAXED,
+                    __hip_atomic_store(&tab[2 * p + 1], (uint64_t)slot,
+                                       __ATOMIC_RELAXED,
                                        __HIP_MEMORY_SCOPE_AGENT);
                     break;
                 }
@@ -65,22 +69,22 @@ __global__ void k_key_to_slot(const uint64_t* key, int64_t n, uint64_t* tkeys,
             }
             p = (p + 1) & mask;
         }
-        // wait for the slot id to be published
-        uint32_t s;
+        // wait for the slot id to be published (same 16 B line as the key)
+        uint64_t s;
         do {
-            s = __hip_atomic_load(&tslots[p], __ATOMIC_RELAXED,
+            s = __hip_atomic_load(&tab[2 * p + 1], __ATOMIC_RELAXED,
                                   __HIP_MEMORY_SCOPE_AGENT);
-        } while (s == ~0u);
-        slot_out[i] = s;
+        } while (s == ~0ULL);
+        slot_out[i] = (uint32_t)s;
     }
 }
 
 extern "C" void wfa_key_to_slot(wfa_stream_t s, const uint64_t* key, int64_t n,
-                                uint64_t* table_keys, uint32_t* table_slots,
-                                uint32_t* n_slots, int64_t table_cap,
-                                uint32_t* slot_out, uint64_t* slot_to_key) {
+                                uint64_t* table_packed, uint32_t* n_slots,
+                                int64_t table_cap, uint32_t* slot_out,
+                                uint64_t* slot_to_key) {
     hipLaunchKernelGGL(k_key_to_slot, dim3(nblk(n)), dim3(WFA_THREADS), 0,
-                       (hipStream_t)s, key, n, table_keys, table_slots, n_slots,
+                       (hipStream_t)s, key, n, table_packed, n_slots,
                        table_cap, slot_out, slot_to_key);
 }
 

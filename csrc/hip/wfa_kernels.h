@@ -36,11 +36,12 @@ void wfa_compact(wfa_stream_t s, int64_t n, const uint32_t* flags,
                  const int* col_esize, int n_cols, int64_t* d_count);
 
 // ----- key -> dense slot hash table (open addressing, u64 keys) -----
-// table_keys: u64[table_cap] init to EMPTY(~0); table_slots: u32[table_cap];
-// n_slots: device counter of allocated slots.  slot_out[i] = dense id of key[i].
+// table_packed: u64[2*table_cap] of interleaved (key, slot) 16 B entries,
+// both halves init to ~0; n_slots: device counter of allocated slots.
+// slot_out[i] = dense id of key[i].
 void wfa_key_to_slot(wfa_stream_t s, const uint64_t* key, int64_t n,
-                     uint64_t* table_keys, uint32_t* table_slots,
-                     uint32_t* n_slots, int64_t table_cap, uint32_t* slot_out,
+                     uint64_t* table_packed, uint32_t* n_slots,
+                     int64_t table_cap, uint32_t* slot_out,
                      uint64_t* slot_to_key);
 
 // ----- stable LSD radix sort of (slot, iota idx) pairs, 4-bit digits -----

@@ -197,6 +197,57 @@ def test_gpu_ffat_cb_min_tree_vs_oracle():
             assert abs(a - b) <= 1e-5 * max(1.0, abs(b))
 
 
+def test_gpu_merge_two_sources():
+    """Two device sources merged into one GPU map (reference
+    merge_tests_gpu): fan-in of device batches across streams."""
+    n, b = 60_000, 15_000
+    s1 = (Source_GPU_Builder(native_gpu.gpu_source(n, 31, b, vdt=0, seed=42))
+          .withOutputSchema([0]).withOutputBatchSize(b).build())
+    s2 = (Source_GPU_Builder(native_gpu.gpu_source(n, 31, b, vdt=0, seed=43))
+          .withOutputSchema([0]).withOutputBatchSize(b).build())
+    g = wf.PipeGraph("gmerge")
+    p1 = g.add_source(s1)
+    p2 = g.add_source(s2)
+    mp = p1.merge(p2)
+    mp.add(Map_GPU_Builder(native_gpu.gpu_affine_map(0, 2, 1, dtype=0))
+           .withOutputSchema([0]).withOutputBatchSize(b).build())
+    snk = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+    snk.out_schema = [0]
+    mp.add_sink(snk)
+    g.run()
+    exp = 0
+    for seed in (42, 43):
+        _, _, val = gen_batch(n, 0, seed, 31, 0)
+        exp += int((val * 2 + 1).sum())
+    assert g.sink_sum(snk) == exp
+    assert g.sink_count(snk) == 2 * n
+
+
+def test_gpu_split_round_robin():
+    """split_gpu: device batches distributed across two GPU branches
+    (reference split_tests_gpu; rr distribution instead of replication)."""
+    from windflow_amd.operators import Operator
+    from windflow_amd.builders_gpu import Sink_GPU_Builder
+    n, b = 80_000, 10_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, 31, b, vdt=0))
+           .withOutputSchema([0]).withOutputBatchSize(b).build())
+    g = wf.PipeGraph("gsplit")
+    mp = g.add_source(src)
+    mp.split_gpu(2)
+    snks = []
+    for br in range(2):
+        bmp = mp.select(br)
+        bmp.add(Map_GPU_Builder(native_gpu.gpu_affine_map(0, 1, 100 * (br + 1),
+                                                          dtype=0))
+                .withOutputSchema([0]).withOutputBatchSize(b).build())
+        snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).build()
+        bmp.chain_sink(snk)
+        snks.append(snk)
+    g.run()
+    assert g.sink_count(snks[0]) + g.sink_count(snks[1]) == n
+    assert g.sink_count(snks[0]) == n // 2  # rr over 8 batches
+
+
 def test_gpu_jit_map_filter():
     """hiprtc-compiled user device logic (reference __device__ lambda
     parity): custom expression map + predicate filter."""
