@@ -376,6 +376,98 @@ __global__ void k_rs8_scatter(const uint32_t* keys, const uint32_t* vals,
     }
 }
 
+// LDS-staged variant: items are reordered in LDS (grouped by digit) before
+// the global write, so each wave stores contiguous runs per digit instead
+// of 4 B scattered words.  Used when there is no second payload (the hot
+// keyed-operator path).
+__global__ void k_rs8_scatter_lds(const uint32_t* keys, const uint32_t* vals,
+                                  int64_t n, int shift, const uint32_t* hist,
+                                  int64_t nblocks, const uint32_t* dbase,
+                                  uint32_t* keys_out, uint32_t* vals_out) {
+    __shared__ uint32_t gbase[256];
+    __shared__ uint32_t waveCnt[WFA_THREADS / 64][256];
+    __shared__ uint32_t wavePre[WFA_THREADS / 64][256];
+    __shared__ uint32_t localBase[256];
+    __shared__ uint32_t sk[RS8_PER_BLOCK];
+    __shared__ uint32_t sv[RS8_PER_BLOCK];
+    __shared__ uint8_t sd[RS8_PER_BLOCK];
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const uint64_t lt = ((uint64_t)1 << lane) - 1;
+    for (int d = threadIdx.x; d < 256; d += blockDim.x) {
+        gbase[d] = hist[(int64_t)d * nblocks + blockIdx.x] + dbase[d];
+        for (int w = 0; w < WFA_THREADS / 64; ++w) waveCnt[w][d] = 0;
+    }
+    __syncthreads();
+    int64_t blockStart = (int64_t)blockIdx.x * RS8_PER_BLOCK;
+    int64_t waveBase = blockStart + wave * RS8_PER_WAVE;
+    uint32_t mk[RS8_IPT], mv[RS8_IPT], mr[RS8_IPT];
+    int nit = 0;
+#pragma unroll
+    for (int j = 0; j < RS8_IPT; ++j) {
+        int64_t i = waveBase + j * 64 + lane;
+        bool valid = i < n;
+        uint32_t k = valid ? keys[i] : 0;
+        uint32_t v = valid ? vals[i] : 0;
+        uint32_t d = (k >> shift) & 255;
+        uint64_t mask = __ballot(valid);
+#pragma unroll
+        for (int bit = 0; bit < 8; ++bit) {
+            uint64_t b = __ballot((d >> bit) & 1);
+            mask &= ((d >> bit) & 1) ? b : ~b;
+        }
+        if (valid) {
+            uint32_t pre = waveCnt[wave][d];
+            uint32_t rank = pre + (uint32_t)__popcll(mask & lt);
+            mk[j] = k;
+            mv[j] = v;
+            mr[j] = (d << 24) | (rank & 0xFFFFFF);
+            if ((mask & lt) == 0) waveCnt[wave][d] = pre + (uint32_t)__popcll(mask);
+            nit = j + 1;
+        }
+    }
+    __syncthreads();
+    // wave prefix + block-local digit bases (layout grouped by digit)
+    __shared__ uint32_t blockCnt[256];
+    for (int d = threadIdx.x; d < 256; d += blockDim.x) {
+        uint32_t run = 0;
+        for (int w = 0; w < WFA_THREADS / 64; ++w) {
+            wavePre[w][d] = run;
+            run += waveCnt[w][d];
+        }
+        blockCnt[d] = run;
+    }
+    __syncthreads();
+    if (threadIdx.x < 256) {  // exclusive prefix over digits (256 threads)
+        __shared__ uint32_t tot[256];
+        uint32_t c = blockCnt[threadIdx.x];
+        tot[threadIdx.x] = c;
+        __syncthreads();
+        for (int off = 1; off < 256; off <<= 1) {
+            uint32_t t = (threadIdx.x >= off) ? tot[threadIdx.x - off] : 0;
+            __syncthreads();
+            tot[threadIdx.x] += t;
+            __syncthreads();
+        }
+        localBase[threadIdx.x] = tot[threadIdx.x] - c;
+    }
+    __syncthreads();
+    for (int j = 0; j < nit; ++j) {
+        uint32_t d = mr[j] >> 24;
+        uint32_t lpos = localBase[d] + wavePre[wave][d] + (mr[j] & 0xFFFFFF);
+        sk[lpos] = mk[j];
+        sv[lpos] = mv[j];
+        sd[lpos] = (uint8_t)d;
+    }
+    __syncthreads();
+    const int nblk_items = (int)min((int64_t)RS8_PER_BLOCK, n - blockStart);
+    for (int p = threadIdx.x; p < nblk_items; p += blockDim.x) {
+        uint32_t d = sd[p];
+        uint32_t g = gbase[d] + (uint32_t)p - localBase[d];
+        keys_out[g] = sk[p];
+        vals_out[g] = sv[p];
+    }
+}
+
 extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                                 uint32_t* slot_tmp, uint32_t* idx_tmp,
                                 uint32_t* val2, uint32_t* val2_tmp,
@@ -398,8 +490,12 @@ extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                                shift, hist, nb);
             hipLaunchKernelGGL(k_rs8_scan, dim3(256), dim3(256), 0, st, hist, nb, dt);
             hipLaunchKernelGGL(k_rs8_dbase, dim3(1), dim3(256), 0, st, dt, dbase);
-            hipLaunchKernelGGL(k_rs8_scatter, dim3(nb), dim3(WFA_THREADS), 0, st, ka,
-                               va, n, shift, hist, nb, dbase, kb, vb, wa, wb);
+            if (val2)
+                hipLaunchKernelGGL(k_rs8_scatter, dim3(nb), dim3(WFA_THREADS), 0, st,
+                                   ka, va, n, shift, hist, nb, dbase, kb, vb, wa, wb);
+            else
+                hipLaunchKernelGGL(k_rs8_scatter_lds, dim3(nb), dim3(WFA_THREADS), 0,
+                                   st, ka, va, n, shift, hist, nb, dbase, kb, vb);
             uint32_t* t;
             t = ka; ka = kb; kb = t;
             t = va; va = vb; vb = t;
