@@ -185,6 +185,15 @@ def test_paned_windows_tb(par):
     assert got_counter(rows) == oracle_tb(seq_stream(3000, 7), win, slide)
 
 
+@pytest.mark.parametrize("par", [1, 3])
+def test_paned_windows_cb(par):
+    win, slide = 40, 10  # pane = 10 (count-based panes per key)
+    rows = run_graph(Paned_Windows_Builder(plq_func=("sum", 0))
+                     .withCBWindows(win, slide).withParallelism(par)
+                     .withOutputSchema([0]).build())
+    assert got_counter(rows) == oracle_cb(seq_stream(3000, 7), win, slide)
+
+
 def test_mapreduce_windows_cb():
     win, slide = 40, 10
     rows = run_graph(MapReduce_Windows_Builder(map_func=("sum", 0))
