@@ -258,19 +258,25 @@ extern "C" int64_t wfa_sort8_nblocks(int64_t n) {
 
 __global__ void k_rs8_hist(const uint32_t* keys, int64_t n, int shift,
                            uint32_t* hist, int64_t nblocks) {
-    __shared__ uint32_t h[256];
-    for (int d = threadIdx.x; d < 256; d += blockDim.x) h[d] = 0;
-    __syncthreads();
+    // one LDS histogram copy per wave: hot digits contend within 64 lanes
+    // instead of 256 threads
+    __shared__ uint32_t h[WFA_THREADS / 64][256];
     const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    for (int d = threadIdx.x; d < 256; d += blockDim.x)
+        for (int w = 0; w < WFA_THREADS / 64; ++w) h[w][d] = 0;
+    __syncthreads();
     int64_t waveBase = (int64_t)blockIdx.x * RS8_PER_BLOCK + wave * RS8_PER_WAVE;
 #pragma unroll
     for (int j = 0; j < RS8_IPT; ++j) {
         int64_t i = waveBase + j * 64 + lane;
-        if (i < n) atomicAdd(&h[(keys[i] >> shift) & 255], 1u);
+        if (i < n) atomicAdd(&h[wave][(keys[i] >> shift) & 255], 1u);
     }
     __syncthreads();
-    for (int d = threadIdx.x; d < 256; d += blockDim.x)
-        hist[(int64_t)d * nblocks + blockIdx.x] = h[d];
+    for (int d = threadIdx.x; d < 256; d += blockDim.x) {
+        uint32_t t = 0;
+        for (int w = 0; w < WFA_THREADS / 64; ++w) t += h[w][d];
+        hist[(int64_t)d * nblocks + blockIdx.x] = t;
+    }
 }
 
 // per-digit scan over blocks (256 blocks, one digit each) + digit totals
