@@ -112,6 +112,35 @@ def build_ffat_x_graph(n_tuples, batch, n_keys, win, slide, rank, world, device,
     return g, snk
 
 
+def build_join_graph(n_tuples, batch, n_keys, par):
+    """Driver config #5 shape (CPU side): two keyed sources -> interval
+    join (KP) -> paned windows -> sink, watermark collectors throughout."""
+    import windflow_amd as wf
+    from windflow_amd import native
+    from windflow_amd.builders import (Interval_Join_Builder,
+                                       Paned_Windows_Builder)
+    g = wf.PipeGraph("bench_join")
+    sa = (wf.Source_Builder(native.seq_source(n_tuples, n_keys, batch))
+          .withParallelism(1).withOutputSchema([0])
+          .withOutputBatchSize(batch).build())
+    sb = (wf.Source_Builder(native.seq_source(n_tuples, n_keys, batch,
+                                              value_offset=1_000_000_000))
+          .withParallelism(1).withOutputSchema([0])
+          .withOutputBatchSize(batch).build())
+    mpA = g.add_source(sa)
+    mpB = g.add_source(sb)
+    mp = mpA.merge(mpB)
+    mp.add(Interval_Join_Builder().withBoundaries(-2, 2).withKPMode()
+           .withValueCols(0).withParallelism(par)
+           .withOutputSchema([0, 0]).withOutputBatchSize(batch).build())
+    mp.add(Paned_Windows_Builder(plq_func=("sum", 0))
+           .withTBWindows(1000, 100).withParallelism(par)
+           .withOutputSchema([0]).withOutputBatchSize(batch).build())
+    snk = wf.Sink_Builder(native.count_sink()).withParallelism(1).build()
+    mp.add_sink(snk)
+    return g, snk
+
+
 def build_cpu_graph(n_tuples, batch):
     import windflow_amd as wf
     from windflow_amd import native
@@ -139,7 +168,7 @@ def main():
                     help="distinct keys per rank")
     ap.add_argument("--win", type=int, default=1000)
     ap.add_argument("--slide", type=int, default=100)
-    ap.add_argument("--config", choices=["ffat", "ffat_x", "a2a", "cpu"], default="ffat")
+    ap.add_argument("--config", choices=["ffat", "ffat_x", "a2a", "cpu", "join"], default="ffat")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -148,7 +177,7 @@ def main():
     dist = None
     torch = None
     dist_cfg = None
-    if args.config != "cpu":
+    if args.config not in ("cpu", "join"):
         import torch  # noqa: F811
         if world > 1:
             import torch.distributed as dist  # noqa: F811
@@ -173,11 +202,13 @@ def main():
     K, W, B = args.steps, args.warmup, args.batch
     p99_us = None
 
-    if args.config == "cpu":
+    if args.config in ("cpu", "join"):
         B = min(B, 65536)
-        gw, _ = build_cpu_graph(W * B, B)
+        build = (build_cpu_graph if args.config == "cpu"
+                 else lambda n, b: build_join_graph(n, b, args.keys, 4))
+        gw, _ = build(W * B, B)
         gw.run()
-        g, snk = build_cpu_graph(K * B, B)
+        g, snk = build(K * B, B)
         t0 = time.time()
         g.run()
         dt = time.time() - t0
@@ -224,7 +255,7 @@ def main():
             "metric": "tuples_per_sec",
             "value": value,
             "unit": "tuples/s",
-            "n_gpus": 0 if args.config == "cpu" else n_gpus,
+            "n_gpus": 0 if args.config in ("cpu", "join") else n_gpus,
             "steps": K,
             "warmup": W,
             "ms_per_step": dt / K * 1000.0,
@@ -232,13 +263,14 @@ def main():
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": {"ffat": "bf16", "ffat_x": "bf16", "a2a": "f32",
-                      "cpu": "int64"}[args.config],
+                      "cpu": "int64", "join": "int64"}[args.config],
             "data": "synthetic",
             "config": {
                 "model": {"ffat": "keyed_ffat_cb_window",
                           "ffat_x": "rccl_keyby_ffat_cb_window",
                           "a2a": "map_gpu_rccl_keyby_reduce_gpu",
-                          "cpu": "cpu_source_map_filter_sink"}[args.config],
+                          "cpu": "cpu_source_map_filter_sink",
+                          "join": "interval_join_paned_windows_cpu"}[args.config],
                 "global_batch": B * max(world, 1),
                 "win": args.win,
                 "slide": args.slide,

@@ -167,6 +167,9 @@ struct WindowCore : OpLogic {
     bool emit_meta;     // plq/mr_map: rows are (c0=gwid, c1=partial)
     bool emit_start_ts; // plq: result ts = window start (pane start)
     bool subset_rr;     // mr_map: accumulate only arrival idx % par == replica
+    bool sparse = false;  // plq TB: tumbling panes, only touched panes open
+                          // (gap panes would flood sparse keys; the WLQ
+                          // treats missing panes as the combine identity)
     Engine* eng;
 
     struct Open {
@@ -178,6 +181,7 @@ struct WindowCore : OpLogic {
     struct KeyState {
         int64_t idx = 0;        // CB arrival count
         int64_t next_gwid = -1; // next window to open (-1: uninitialized)
+        int64_t first_unfired = 0;  // sparse mode: panes below this fired
         int64_t last_ts = 0;
         std::deque<Open> open;
     };
@@ -236,8 +240,40 @@ struct WindowCore : OpLogic {
         ks.idx++;
     }
 
+    void add_tb_sparse(uint64_t key, KeyState& ks, int64_t ts, const ValU& v,
+                       bool mine, EmitCtx& out) {
+        // tumbling panes (win == slide): exactly one pane contains ts
+        const int64_t g = ts / slide;
+        if (g < ks.first_unfired) {
+            ignored++;
+            return;
+        }
+        // opens are few and mostly increasing: scan from the back
+        Open* w = nullptr;
+        for (auto it = ks.open.rbegin(); it != ks.open.rend(); ++it) {
+            if (it->gwid == g) {
+                w = &*it;
+                break;
+            }
+            if (it->gwid < g) break;
+        }
+        if (!w) {
+            auto pos = ks.open.end();
+            while (pos != ks.open.begin() && std::prev(pos)->gwid > g) --pos;
+            pos = ks.open.insert(pos, {g, g * slide, {}, owned(g)});
+            fire_heap.emplace(g * slide + win + lateness, key);
+            w = &*pos;
+        }
+        if (w->owned && mine) agg.add(w->acc, v);
+        ks.last_ts = std::max(ks.last_ts, ts);
+    }
+
     void add_tb(uint64_t key, KeyState& ks, int64_t ts, const ValU& v, bool mine,
                 EmitCtx& out) {
+        if (sparse) {
+            add_tb_sparse(key, ks, ts, v, mine, out);
+            return;
+        }
         if (ks.next_gwid < 0) {
             // first window containing the key's first tuple
             int64_t w0 = (ts - win + 1);
@@ -271,6 +307,7 @@ struct WindowCore : OpLogic {
                    ks.open.front().start + win + lateness <= cur_wm) {
                 Open& w = ks.open.front();
                 fire(w, key, emit_start_ts ? w.start : w.start + win - 1, out);
+                ks.first_unfired = std::max(ks.first_unfired, w.gwid + 1);
                 ks.open.pop_front();
             }
         }
@@ -324,6 +361,15 @@ struct WindowCore : OpLogic {
 // c1 = partial) rows from win_plq.  A window w covers panes
 // [w*slide_p, w*slide_p + win_p).  CB fires on completeness (every pane of
 // the window received — PLQ emits gap panes); TB fires on the watermark.
+struct KeyGwidHash {
+    size_t operator()(const std::pair<uint64_t, int64_t>& p) const {
+        uint64_t h = p.first * 0x9e3779b97f4a7c15ULL ^ (uint64_t)p.second;
+        h ^= h >> 29;
+        h *= 0xbf58476d1ce4e5b9ULL;
+        return (size_t)(h ^ (h >> 32));
+    }
+};
+
 struct WlqLogic : OpLogic {
     WinType wt;
     int64_t win_p, slide_p, lateness, pane_len;
@@ -337,7 +383,7 @@ struct WlqLogic : OpLogic {
         int64_t last_ts = 0;
     };
     // (key, gwid) -> partial window
-    std::map<std::pair<uint64_t, int64_t>, WinAcc> wins;
+    std::unordered_map<std::pair<uint64_t, int64_t>, WinAcc, KeyGwidHash> wins;
     using HeapEl = std::pair<int64_t, std::pair<uint64_t, int64_t>>;
     std::priority_queue<HeapEl, std::vector<HeapEl>, std::greater<HeapEl>> fire_heap;
     OutBuf ob;
@@ -434,7 +480,7 @@ struct MrReduceLogic : OpLogic {
         int64_t got = 0;
         int64_t last_ts = 0;
     };
-    std::map<std::pair<uint64_t, int64_t>, WinAcc> wins;
+    std::unordered_map<std::pair<uint64_t, int64_t>, WinAcc, KeyGwidHash> wins;
     OutBuf ob;
     int64_t cur_wm = 0;
 
@@ -756,11 +802,27 @@ struct IntervalJoinLogic : OpLogic {
             bool store = mode == 0 || (store_ctr % par) == (int64_t)replica;
             store_ctr++;
             if (store) insert_sorted(tag == 1 ? ks.b : ks.a, e);
+            purge_key(ks);
         }
         cur_wm = std::max(cur_wm, b->watermark);
         release(b);
-        purge();
         ob.flush(out, cur_wm);
+    }
+
+    // purge one key's archives up to the current watermark bounds (called
+    // inline per touched key — a full-keyspace sweep per batch is O(keys))
+    void purge_key(KeyState& ks) {
+        int64_t a_min = cur_wm - upper, b_min = cur_wm + lower;
+        if (!ks.a.empty() && ks.a.front().ts < a_min) {
+            auto pa = std::lower_bound(ks.a.begin(), ks.a.end(), a_min,
+                                       [](const Entry& x, int64_t t) { return x.ts < t; });
+            ks.a.erase(ks.a.begin(), pa);
+        }
+        if (!ks.b.empty() && ks.b.front().ts < b_min) {
+            auto pb = std::lower_bound(ks.b.begin(), ks.b.end(), b_min,
+                                       [](const Entry& x, int64_t t) { return x.ts < t; });
+            ks.b.erase(ks.b.begin(), pb);
+        }
     }
 
     void purge() {
@@ -976,9 +1038,14 @@ std::shared_ptr<OpLogic> make_window_logic(const std::string& kind,
     if (kind == "win_parallel")
         return std::make_shared<WindowCore>(wt, geti(1), geti(2), geti(3), a, col,
                                             (int)geti(6, 1), false, false, false, eng);
-    if (kind == "win_plq")
-        return std::make_shared<WindowCore>(wt, geti(1), geti(2), geti(3), a, col,
-                                            (int)geti(6, 1), true, true, false, eng);
+    if (kind == "win_plq") {
+        auto w = std::make_shared<WindowCore>(wt, geti(1), geti(2), geti(3), a, col,
+                                              (int)geti(6, 1), true, true, false, eng);
+        // TB tumbling panes: materialize only touched panes (the WLQ treats
+        // missing panes as the combine identity); CB panes are dense per key
+        if (wt == WinType::TB) w->sparse = true;
+        return w;
+    }
     if (kind == "win_mr_map")
         return std::make_shared<WindowCore>(wt, geti(1), geti(2), geti(3), a, col,
                                             0, true, false, true, eng);

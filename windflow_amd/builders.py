@@ -135,7 +135,7 @@ class Paned_Windows_Builder(_WindowsBuilder):
     def __init__(self, plq_func=None, wlq_func=None, lift=None, comb=None):
         super().__init__(plq_func, lift, comb)
         self._op.extra['wlq_func'] = wlq_func
-        self._op.broadcast_input = True
+        self._op.key_extractor = 'carried'  # KEYBY into the PLQ stage
 
     def withPLQParallelism(self, p):
         self._op.extra['plq_par'] = int(p)

@@ -90,13 +90,17 @@ def lower_window_node(graph, e, node):
         plq_par = int(op.extra.get("plq_par", node.parallelism))
         wlq_par = int(op.extra.get("wlq_par", node.parallelism))
         mid_schema = [0, out_schema[0]]  # (pane gwid, partial)
+        # Keys are partitioned (KEYBY) into both stages: each replica owns
+        # every pane/window of its keys, so a tuple is processed once, not
+        # once per replica.  The reference's broadcast + gwid%n ownership
+        # (parallelism beyond the key count) remains as Parallel_Windows.
         plq = e.add_op((op.name or kind) + ".plq", plq_par, "win_plq",
-                       iparams=[wt, pane, pane, lat, comb, col, 1, ui],
+                       iparams=[wt, pane, pane, lat, comb, col, 0, ui],
                        out_schema=mid_schema, out_batch=op.out_batch)
         wlq = e.add_op((op.name or kind) + ".wlq", wlq_par, "win_wlq",
-                       iparams=[wt, win_p, slide_p, lat, comb, pane, 1, ui],
+                       iparams=[wt, win_p, slide_p, lat, comb, pane, 0, ui],
                        out_schema=out_schema, out_batch=op.out_batch)
-        e.add_edge(plq, wlq, RoutingMode.BROADCAST, CollectorKind.WATERMARK, -1)
+        e.add_edge(plq, wlq, RoutingMode.KEYBY, CollectorKind.WATERMARK, -1)
         in_id, out_id = plq, wlq
 
     elif kind == "mapreduce_windows":
