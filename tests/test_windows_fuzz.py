@@ -1,0 +1,48 @@
+"""Randomized differential sweep over the window operators: random
+win/slide/key-count/batch-size/parallelism drawn per case, every operator
+form compared against the same brute-force oracle (the reference randomizes
+degrees but has no oracle)."""
+import random
+from collections import defaultdict
+
+import pytest
+
+import windflow_amd as wf
+from windflow_amd import native
+from windflow_amd.builders import (Keyed_Windows_Builder, Parallel_Windows_Builder,
+                                   Paned_Windows_Builder, MapReduce_Windows_Builder,
+                                   Ffat_Windows_Builder)
+
+import sys, os
+sys.path.insert(0, os.path.dirname(__file__))
+from test_windows import (run_graph, seq_stream, oracle_cb, oracle_tb,
+                          got_counter)  # noqa: E402
+
+BUILDERS = {
+    "keyed": lambda f: Keyed_Windows_Builder(func=f),
+    "parallel": lambda f: Parallel_Windows_Builder(func=f),
+    "paned": lambda f: Paned_Windows_Builder(plq_func=f),
+    "mapreduce": lambda f: MapReduce_Windows_Builder(map_func=f),
+    "ffat": lambda f: Ffat_Windows_Builder(comb=f),
+}
+
+
+@pytest.mark.parametrize("case", range(12))
+def test_window_fuzz(case):
+    rng = random.Random(1000 + case)
+    kind = rng.choice(list(BUILDERS))
+    agg = rng.choice(["sum", "max"])
+    wt = rng.choice(["cb", "tb"])
+    slide = rng.choice([5, 10, 25])
+    win = slide * rng.randint(1, 6)
+    n_keys = rng.choice([1, 3, 11])
+    batch = rng.choice([32, 128, 1000])
+    par = rng.randint(1, 3)
+    stream = rng.choice([1500, 3100])
+    b = BUILDERS[kind]((agg, 0))
+    b = (b.withCBWindows(win, slide) if wt == "cb" else b.withTBWindows(win, slide))
+    op = b.withParallelism(par).withOutputSchema([0]).build()
+    rows = run_graph(op, stream_len=stream, n_keys=n_keys, batch=batch)
+    per = seq_stream(stream, n_keys)
+    exp = (oracle_cb if wt == "cb" else oracle_tb)(per, win, slide, agg)
+    assert got_counter(rows) == exp, (kind, agg, wt, win, slide, n_keys, batch, par)
