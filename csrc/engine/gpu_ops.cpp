@@ -758,9 +758,9 @@ struct GpuReduceLogic : GpuLogicBase {
         int64_t n = db->count;
         ks.group(stream, db, vcol, ctx);
         Batch* ob = get_dev();
-        wfa_segment_reduce(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                           ks.v_as_f32, ks.idx_sorted, db->ts, 2, comb,
-                           ks.slot_to_key, ob->key, ob->cols[0], ob->ts, d_on);
+        wfa_segment_reduce_wave(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
+                                ks.v_as_f32, ks.idx_sorted, db->ts, comb,
+                                ks.slot_to_key, ob->key, ob->cols[0], ob->ts, d_on);
         HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
         ob->count = -1;
         ob->watermark = db->watermark;

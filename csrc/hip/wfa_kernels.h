@@ -81,6 +81,14 @@ void wfa_segment_reduce(wfa_stream_t s, const uint32_t* seg_start,
                         int comb, const uint64_t* slot_to_key,
                         uint64_t* out_key, void* out_val, int64_t* out_ts,
                         int64_t* d_out_n);
+// wave-per-segment variant (f32 values; one wave cooperatively reduces
+// one key's segment — use for batches with large segments)
+void wfa_segment_reduce_wave(
+    wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
+    const int64_t* d_nseg, int64_t n, const void* v_orig,
+    const uint32_t* idx_sorted, const int64_t* ts_orig, int comb,
+    const uint64_t* slot_to_key, uint64_t* out_key, void* out_val,
+    int64_t* out_ts, int64_t* d_out_n);
 
 // closed-form per-segment window-fire counts -> exclusive output offsets
 // (nf) + total (*d_out_n); run BEFORE the fold (reads pristine state)
