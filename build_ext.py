@@ -117,6 +117,31 @@ def build(force=False):
     return out
 
 
+def build_stress(tsan=False):
+    """Build the native engine stress harness (csrc/tests/engine_stress.cpp)
+    — python-free, HIP-free; with tsan=True it is instrumented for
+    ThreadSanitizer (race detection the reference lacks)."""
+    os.makedirs(BUILD, exist_ok=True)
+    out = os.path.join(BUILD, "engine_stress" + ("_tsan" if tsan else ""))
+    srcs = [os.path.join(ROOT, p) for p in
+            ["csrc/tests/engine_stress.cpp", "csrc/engine/engine.cpp",
+             "csrc/engine/core.cpp", "csrc/engine/native_logic.cpp",
+             "csrc/engine/windows.cpp", "csrc/engine/persist.cpp",
+             "csrc/engine/gpu_ops.cpp"]]
+    if not newer(out, srcs + headers("csrc")):
+        return out
+    flags = ["-O1" if tsan else "-O2", "-g", "-std=c++20", "-pthread"]
+    if tsan:
+        flags += ["-fsanitize=thread"]
+    cmd = ["g++"] + flags + srcs + ["-o", out]
+    print("  BUILD", out, flush=True)
+    subprocess.check_call(cmd)
+    return out
+
+
 if __name__ == "__main__":
-    build(force="--force" in sys.argv)
+    if "--stress" in sys.argv:
+        build_stress(tsan="--tsan" in sys.argv)
+    else:
+        build(force="--force" in sys.argv)
     print("OK")

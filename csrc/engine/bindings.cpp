@@ -33,11 +33,11 @@ static py::dict batch_views(Batch* b, RuntimeCtx* ctx = nullptr) {
     d["key"] = col_view(DType::U64, b->key, b->count);
     for (size_t c = 0; c < b->schema.payload.size(); ++c)
         d[py::str("c" + std::to_string(c))] = col_view(b->schema.payload[c], b->cols[c], b->count);
-    d["watermark"] = b->watermark;
+    d["watermark"] = ctx ? ctx->current_wm : b->watermark;
     if (ctx) {  // RuntimeContext parity (reference wf/context.hpp:53)
         d["replica"] = ctx->replica;
         d["parallelism"] = ctx->parallelism;
-        d["stream_tag"] = b->stream_tag;
+        d["stream_tag"] = ctx->current_tag;
     }
     return d;
 }
@@ -92,6 +92,7 @@ struct PyMapLogic : OpLogic {  // in-place mutate
     void process(Batch* b, EmitCtx& out, RuntimeCtx& rctx) override {
         if (b->refcnt.load(std::memory_order_acquire) > 1) {
             Batch* c = clone(b, *b->pool);
+            c->watermark = rctx.current_wm;  // folded (shared original is read-only)
             release(b);
             b = c;
         }
@@ -107,7 +108,7 @@ struct PyTransformLogic : OpLogic {  // returns new column dict (map/flatmap)
     py::function fn;
     explicit PyTransformLogic(py::function f) : fn(std::move(f)) {}
     void process(Batch* b, EmitCtx& out, RuntimeCtx& rctx) override {
-        int64_t wm = b->watermark;
+        int64_t wm = rctx.current_wm;
         py::gil_scoped_acquire gil;
         py::object r = fn(batch_views(b, &rctx));
         release(b);
@@ -121,6 +122,7 @@ struct PyFilterLogic : OpLogic {  // returns bool mask
     void process(Batch* b, EmitCtx& out, RuntimeCtx& rctx) override {
         if (b->refcnt.load(std::memory_order_acquire) > 1) {
             Batch* c = clone(b, *b->pool);
+            c->watermark = rctx.current_wm;
             release(b);
             b = c;
         }
@@ -149,7 +151,7 @@ struct PyFilterLogic : OpLogic {  // returns bool mask
         if (w)
             out.emit(b);
         else {
-            int64_t wm = b->watermark;
+            int64_t wm = rctx.current_wm;
             release(b);
             for (auto* e : out.emitters) e->punct(wm);
         }

@@ -125,10 +125,16 @@ Batch* Collector::next() {
                 n_open--;
                 break;  // re-evaluate loop condition
             }
-            if (b->stream_tag < 0) b->stream_tag = chan_tag[c];
             chan_wm[c] = std::max(chan_wm[c], b->watermark);
             int64_t m = min_wm();
-            b->watermark = m;
+            delivered_wm = m;
+            delivered_tag = b->stream_tag >= 0 ? b->stream_tag : chan_tag[c];
+            if (b->refcnt.load(std::memory_order_acquire) == 1) {
+                // sole consumer: safe to rewrite in place (downstream
+                // relays propagate the folded value)
+                b->watermark = m;
+                b->stream_tag = delivered_tag;
+            }
             if (b->punct) {
                 if (m > last_fwd_wm) {
                     last_fwd_wm = m;
@@ -168,7 +174,6 @@ Batch* OrderingCollector::next() {
                     n_open--;
                     break;
                 }
-                if (b->stream_tag < 0) b->stream_tag = chan_tag[c];
                 chan_wm[c] = std::max(chan_wm[c], b->watermark);
                 pend[c].push_back(b);
             }
@@ -194,12 +199,17 @@ Batch* OrderingCollector::next() {
             Batch* b = pend[best].front();
             pend[best].pop_front();
             rel_wm[best] = std::max(rel_wm[best], b->watermark);
-            b->watermark = released_wm();
-            if (b->punct && b->watermark <= last_fwd_wm) {
+            delivered_wm = released_wm();
+            delivered_tag = b->stream_tag >= 0 ? b->stream_tag : chan_tag[best];
+            if (b->refcnt.load(std::memory_order_acquire) == 1) {
+                b->watermark = delivered_wm;
+                b->stream_tag = delivered_tag;
+            }
+            if (b->punct && delivered_wm <= last_fwd_wm) {
                 release(b);
                 continue;
             }
-            last_fwd_wm = std::max(last_fwd_wm, b->watermark);
+            last_fwd_wm = std::max(last_fwd_wm, delivered_wm);
             return b;
         }
         if (n_open == 0) {
@@ -219,7 +229,12 @@ Batch* OrderingCollector::next() {
             Batch* b = pend[best].front();
             pend[best].pop_front();
             rel_wm[best] = std::max(rel_wm[best], b->watermark);
-            b->watermark = released_wm();
+            delivered_wm = released_wm();
+            delivered_tag = b->stream_tag >= 0 ? b->stream_tag : chan_tag[best];
+            if (b->refcnt.load(std::memory_order_acquire) == 1) {
+                b->watermark = delivered_wm;
+                b->stream_tag = delivered_tag;
+            }
             return b;
         }
         SpscQueue::backoff(spins);
@@ -241,7 +256,6 @@ Batch* KSlackCollector::next() {
                     n_open--;
                     break;
                 }
-                if (b->stream_tag < 0) b->stream_tag = chan_tag[c];
                 chan_wm[c] = std::max(chan_wm[c], b->watermark);
                 if (b->punct) {
                     release(b);
@@ -274,8 +288,11 @@ Batch* KSlackCollector::next() {
                 Batch* b = it->second;
                 last_rel_ts = std::max(last_rel_ts, it->first);
                 buf.erase(it);
-                b->watermark = std::max(last_fwd_wm, b->watermark);
-                last_fwd_wm = b->watermark;
+                delivered_wm = std::max(last_fwd_wm, b->watermark);
+                delivered_tag = b->stream_tag;
+                if (b->refcnt.load(std::memory_order_acquire) == 1)
+                    b->watermark = delivered_wm;
+                last_fwd_wm = delivered_wm;
                 return b;
             }
         } else if (n_open == 0) {
@@ -361,10 +378,11 @@ void Replica::run() {
             int64_t t0 = now_us();
             stats.inputs_received++;
             if (b->count > 0) stats.tuples_received += b->count;
-            rctx.current_wm = b->watermark;
+            rctx.current_wm = collector->delivered_wm;
+            rctx.current_tag = collector->delivered_tag;
             if (b->count) rctx.current_ts = b->ts[b->count - 1];
             if (b->punct) {
-                int64_t wm = b->watermark;
+                int64_t wm = collector->delivered_wm;
                 release(b);
                 if (!logic->on_punct(wm, ectx, rctx))
                     for (auto& e : emitters) e->punct(wm);

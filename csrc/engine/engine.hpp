@@ -52,7 +52,8 @@ struct RuntimeCtx {
     int replica = 0;
     int parallelism = 1;
     int64_t current_ts = 0;
-    int64_t current_wm = 0;
+    int64_t current_wm = 0;   // collector-folded watermark of the current batch
+    int current_tag = -1;     // join stream tag of the current batch
     Engine* engine = nullptr;
     int op_id = 0;
     StatsRecord* stats = nullptr;
@@ -192,6 +193,12 @@ struct Collector {
     size_t n_open = 0;
     size_t rr = 0;
     int64_t last_fwd_wm = -1;
+    // the folded watermark/tag of the batch most recently returned by
+    // next(): shared (broadcast) batches are NOT rewritten in place — a
+    // concurrent consumer would race on the same fields (found by the
+    // TSAN harness, csrc/tests/engine_stress.cpp)
+    int64_t delivered_wm = 0;
+    int delivered_tag = -1;
     std::atomic<bool>* abort = nullptr;
 
     explicit Collector(std::vector<SpscQueue*> ch, std::vector<int> tags = {})

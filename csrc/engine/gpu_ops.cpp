@@ -253,8 +253,8 @@ struct GpuLogicBase : OpLogic {
             bytes += es * n;
         }
         db->count = n;
-        db->watermark = hb->watermark;
-        db->stream_tag = hb->stream_tag;
+        db->watermark = ctx.current_wm;   // folded (hb may be a shared batch)
+        db->stream_tag = ctx.current_tag;
         if (ctx.stats) ctx.stats->bytes_h2d += bytes;
         // the host batch may be recycled by a CPU producer the moment we
         // release it — the copies must have landed first

@@ -13,9 +13,13 @@
 namespace wfa {
 
 // ----- helpers -----
-static Batch* ensure_exclusive(Batch* b, EmitCtx& out) {
+// Shared (broadcast) batches are read-only: clone before mutating, and
+// stamp the clone with the collector-folded watermark from the context
+// (the shared original keeps the producer's value).
+static Batch* ensure_exclusive(Batch* b, EmitCtx& out, RuntimeCtx& ctx) {
     if (b->refcnt.load(std::memory_order_acquire) > 1) {
         Batch* c = clone(b, *b->pool);
+        c->watermark = ctx.current_wm;
         release(b);
         return c;
     }
@@ -87,8 +91,8 @@ struct AffineMapI64 : OpLogic {
     int col;
     int64_t a, bb;
     AffineMapI64(int c, int64_t a_, int64_t b_) : col(c), a(a_), bb(b_) {}
-    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
-        b = ensure_exclusive(b, out);
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
+        b = ensure_exclusive(b, out, ctx);
         int64_t* x = b->col<int64_t>(col);
         const int64_t n = b->count;
         for (int64_t i = 0; i < n; ++i) x[i] = a * x[i] + bb;
@@ -102,8 +106,8 @@ struct ModFilterI64 : OpLogic {
     int64_t m, c;
     bool keep_eq;
     ModFilterI64(int col_, int64_t m_, int64_t c_, bool ke) : col(col_), m(m_), c(c_), keep_eq(ke) {}
-    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
-        b = ensure_exclusive(b, out);
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
+        b = ensure_exclusive(b, out, ctx);
         int64_t* x = b->col<int64_t>(col);
         const size_t np = b->schema.payload.size();
         int64_t w = 0;
@@ -125,7 +129,7 @@ struct ModFilterI64 : OpLogic {
             out.emit(b);
         else {
             // all dropped: keep watermarks flowing (reference filter.hpp:151)
-            int64_t wm = b->watermark;
+            int64_t wm = ctx.current_wm;
             release(b);
             for (auto* e : out.emitters) e->punct(wm);
         }
@@ -136,13 +140,13 @@ struct ModFilterI64 : OpLogic {
 struct DupFlatMapI64 : OpLogic {
     int64_t k;
     explicit DupFlatMapI64(int64_t k_) : k(k_) {}
-    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
         Batch* o = out.new_batch();
         const size_t np = b->schema.payload.size();
         for (int64_t i = 0; i < b->count; ++i) {
             for (int64_t j = 0; j < k; ++j) {
                 if (o->count == o->capacity) {
-                    o->watermark = b->watermark;
+                    o->watermark = ctx.current_wm;
                     out.emit(o);
                     o = out.new_batch();
                 }
@@ -155,7 +159,7 @@ struct DupFlatMapI64 : OpLogic {
                 }
             }
         }
-        o->watermark = b->watermark;
+        o->watermark = ctx.current_wm;
         if (o->count)
             out.emit(o);
         else
@@ -170,14 +174,14 @@ struct KeyedSumReduceI64 : OpLogic {
     int col;
     std::unordered_map<uint64_t, int64_t> acc;
     explicit KeyedSumReduceI64(int c) : col(c) { acc.reserve(1 << 12); }
-    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
         Batch* o = out.new_batch();
         int64_t* x = b->col<int64_t>(col);
         for (int64_t i = 0; i < b->count; ++i) {
             int64_t& a = acc[b->key[i]];
             a += x[i];
             if (o->count == o->capacity) {
-                o->watermark = b->watermark;
+                o->watermark = ctx.current_wm;
                 out.emit(o);
                 o = out.new_batch();
             }
@@ -186,7 +190,7 @@ struct KeyedSumReduceI64 : OpLogic {
             o->key[w] = b->key[i];
             o->col<int64_t>(0)[w] = a;
         }
-        o->watermark = b->watermark;
+        o->watermark = ctx.current_wm;
         if (o->count)
             out.emit(o);
         else
@@ -234,7 +238,7 @@ struct CountSink : OpLogic {
 struct SplitModI64 : OpLogic {
     int col;
     explicit SplitModI64(int c) : col(c) {}
-    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
         const size_t nb = out.n_branches();
         const size_t np = b->schema.payload.size();
         int64_t* x = b->col<int64_t>(col);
@@ -245,7 +249,7 @@ struct SplitModI64 : OpLogic {
                 if (d != br) continue;
                 if (!o) o = out.new_batch();
                 if (o->count == o->capacity) {
-                    o->watermark = b->watermark;
+                    o->watermark = ctx.current_wm;
                     out.emit_to(br, o);
                     o = out.new_batch();
                 }
@@ -258,7 +262,7 @@ struct SplitModI64 : OpLogic {
                 }
             }
             if (o) {
-                o->watermark = b->watermark;
+                o->watermark = ctx.current_wm;
                 if (o->count)
                     out.emit_to(br, o);
                 else

@@ -325,7 +325,7 @@ struct WindowCore : OpLogic {
             else
                 add_tb(b->key[i], ks, b->ts[i], v, mine, out);
         }
-        cur_wm = std::max(cur_wm, b->watermark);
+        cur_wm = std::max(cur_wm, ctx.current_wm);
         release(b);
         if (wt == WinType::TB) fire_tb(out);
         ob.flush(out, cur_wm);
@@ -411,7 +411,7 @@ struct WlqLogic : OpLogic {
         write_val(o, 0, i, agg.result(w.acc), agg.use_int);
     }
 
-    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
         const int64_t n = b->count;
         for (int64_t i = 0; i < n; ++i) {
             uint64_t key = b->key[i];
@@ -440,7 +440,7 @@ struct WlqLogic : OpLogic {
                 }
             }
         }
-        cur_wm = std::max(cur_wm, b->watermark);
+        cur_wm = std::max(cur_wm, ctx.current_wm);
         release(b);
         if (wt == WinType::TB) fire_tb(out);
         ob.flush(out, cur_wm);
@@ -494,7 +494,7 @@ struct MrReduceLogic : OpLogic {
         write_val(o, 0, i, agg.result(w.acc), agg.use_int);
     }
 
-    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
         for (int64_t i = 0; i < b->count; ++i) {
             auto kk = std::make_pair(b->key[i], b->col<int64_t>(0)[i]);
             auto& w = wins[kk];
@@ -512,7 +512,7 @@ struct MrReduceLogic : OpLogic {
                 wins.erase(kk);
             }
         }
-        cur_wm = std::max(cur_wm, b->watermark);
+        cur_wm = std::max(cur_wm, ctx.current_wm);
         release(b);
         ob.flush(out, cur_wm);
     }
@@ -623,7 +623,7 @@ struct FfatCpuLogic : OpLogic {
         write_val(o, 0, i, agg.result(r), agg.use_int);
     }
 
-    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
         for (int64_t i = 0; i < b->count; ++i) {
             auto& ks = keys[b->key[i]];
             if (!ks.init) {
@@ -655,7 +655,7 @@ struct FfatCpuLogic : OpLogic {
                 ks.last_ts = std::max(ks.last_ts, b->ts[i]);
             }
         }
-        cur_wm = std::max(cur_wm, b->watermark);
+        cur_wm = std::max(cur_wm, ctx.current_wm);
         release(b);
         if (wt == WinType::TB) complete_panes(out, false);
         ob.flush(out, cur_wm);
@@ -767,8 +767,8 @@ struct IntervalJoinLogic : OpLogic {
         write_val(o, 1, i, eb.v, b_int);
     }
 
-    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
-        int tag = b->stream_tag;
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
+        int tag = ctx.current_tag;
         int vcol = tag == 1 ? colB : colA;
         bool is_int = b->schema.payload[vcol] == DType::I64 ||
                       b->schema.payload[vcol] == DType::I32;
@@ -804,7 +804,7 @@ struct IntervalJoinLogic : OpLogic {
             if (store) insert_sorted(tag == 1 ? ks.b : ks.a, e);
             purge_key(ks);
         }
-        cur_wm = std::max(cur_wm, b->watermark);
+        cur_wm = std::max(cur_wm, ctx.current_wm);
         release(b);
         ob.flush(out, cur_wm);
     }
@@ -924,7 +924,7 @@ struct PyWindowLogic : OpLogic {
         ks.purged += n;
     }
 
-    void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
         if (!have_schema) {
             in_schema = b->schema;
             have_schema = true;
@@ -944,7 +944,7 @@ struct PyWindowLogic : OpLogic {
                 }
             }
         }
-        cur_wm = std::max(cur_wm, b->watermark);
+        cur_wm = std::max(cur_wm, ctx.current_wm);
         release(b);
         if (wt == WinType::TB) fire_tb(out, false);
         ob.flush(out, cur_wm);
