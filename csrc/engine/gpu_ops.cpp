@@ -428,7 +428,8 @@ struct KeyedScratch {
     }
 
     uint32_t* idx_sorted = nullptr;  // valid after group()
-    const float* v_as_f32 = nullptr;
+    const void* v_as_f32 = nullptr;  // f32 or bf16, per v_dt
+    int v_dt = 2;                    // effective dtype of v_as_f32 (2/5)
 
     // sorts (slot, idx) pairs and fills segments; values stay unsorted and
     // are read through idx_sorted (saves the gather round trip).
@@ -440,22 +441,26 @@ struct KeyedScratch {
                                      "set the GPU op's out_batch >= upstream batch");
         wfa_key_to_slot(s, db->key, n, tab, d_nslots, table_cap, slot,
                         slot_to_key);
-        wfa_iota_u32(s, idx, n);
         uint32_t *os, *oi;
-        wfa_sort_pairs(s, slot, idx, slot_t, idx_t, hist, n, bits, &os, &oi);
+        wfa_sort_pairs2(s, slot, idx, slot_t, idx_t, nullptr, nullptr, hist, n,
+                        bits, &os, &oi, nullptr, /*implicit_iota=*/1);
         idx_sorted = oi;
         if (vcol >= 0) {
             // A/B-measured: reading values THROUGH idx_sorted in the folds
             // beats carrying them as a sort payload (the wave fold hides the
             // gather; the extra scatter writes did not pay — 11.5 vs 13.0
-            // B tuples/s).  wfa_sort_pairs2 remains for payload use cases.
+            // B tuples/s).  f32/bf16 columns are read in place (bf16
+            // converts inline in the fold: half the gather bytes, no cast
+            // kernel); other dtypes cast to f32 once.
             int vdt = (int)db->schema.payload[vcol];
-            const void* vsrc = db->cols[vcol];
-            if (vdt != 2) {  // cast to f32 once (i64/bf16 lifted)
-                wfa_cast(s, vsrc, vdt, v_f32, 2, n);
-                vsrc = v_f32;
+            if (vdt == 2 || vdt == 5) {
+                v_as_f32 = db->cols[vcol];
+                v_dt = vdt;
+            } else {
+                wfa_cast(s, db->cols[vcol], vdt, v_f32, 2, n);
+                v_as_f32 = v_f32;
+                v_dt = 2;
             }
-            v_as_f32 = (const float*)vsrc;
         }
         wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg);
         if (ctx.stats) ctx.stats->num_kernels += 5 + 3 * ((bits + 3) / 4);
@@ -759,7 +764,7 @@ struct GpuReduceLogic : GpuLogicBase {
         ks.group(stream, db, vcol, ctx);
         Batch* ob = get_dev();
         wfa_segment_reduce_wave(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                                ks.v_as_f32, ks.idx_sorted, db->ts, comb,
+                                ks.v_as_f32, ks.v_dt, ks.idx_sorted, db->ts, comb,
                                 ks.slot_to_key, ob->key, ob->cols[0], ob->ts, d_on);
         HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
         ob->count = -1;
@@ -884,13 +889,13 @@ struct GpuFfatLogic : GpuLogicBase {
                               pane_len, P, S, st_fill, st_head, nf, d_on);
         if (use_tree)
             wfa_ffat_tree_fold(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                               ks.v_as_f32, ks.idx_sorted, db->ts, pane_len, P, S,
+                               ks.v_as_f32, ks.v_dt, ks.idx_sorted, db->ts, pane_len, P, S,
                                comb, ring_log2, st_count, st_fill, st_acc,
                                ring_or_tree, st_head, ks.slot_to_key, nf, ob->key,
                                (float*)ob->cols[0], ob->ts, ob->capacity);
         else
             wfa_ffat_cb_fold(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                             ks.v_as_f32, ks.idx_sorted, db->ts, pane_len, P, S,
+                             ks.v_as_f32, ks.v_dt, ks.idx_sorted, db->ts, pane_len, P, S,
                              comb, ring_log2, st_count, st_fill, st_acc,
                              ring_or_tree, st_head, st_wsum, ks.slot_to_key, nf,
                              ob->key, (float*)ob->cols[0], ob->ts, ob->capacity);
@@ -945,7 +950,7 @@ struct GpuFfatLogic : GpuLogicBase {
         Batch* ob = get_dev();
         int64_t limit = (wm - lateness) / pane_len - 1;
         wfa_ffat_tb_round(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                          ks.v_as_f32, ks.idx_sorted, db ? db->ts : nullptr,
+                          ks.v_as_f32, ks.v_dt, ks.idx_sorted, db ? db->ts : nullptr,
                           pane_len, P, S, comb, ring_log2, pend_log2, limit,
                           tb_pend, tb_base, tb_last, st_head, st_wsum,
                           ring_or_tree, ks.d_nslots, ks.slot_to_key, tb_nf,
@@ -1265,10 +1270,10 @@ struct GpuExchangeLogic : GpuLogicBase {
             HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)sb->ready_event, 0));
         // bucket -> stable sort by dest -> contiguous per-dest send layout
         wfa_bucket_by_key(stream, db->key, n, world, dest);
-        wfa_iota_u32(stream, idx, n);
         wfa_count_u32(stream, dest, n, d_counts, world);
         uint32_t *od, *oi;
-        wfa_sort_pairs(stream, dest, idx, dest_t, idx_t, hist, n, bits, &od, &oi);
+        wfa_sort_pairs2(stream, dest, idx, dest_t, idx_t, nullptr, nullptr, hist,
+                        n, bits, &od, &oi, nullptr, /*implicit_iota=*/1);
         size_t nc = db->cols.size();
         std::vector<void*> ptrs(2 * nc);
         for (size_t c = 0; c < nc; ++c) {

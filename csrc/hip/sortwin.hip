@@ -26,6 +26,17 @@ static inline int64_t nblk(int64_t n, int64_t per_thread = 1) {
     return b < 1 ? 1 : (b > WFA_MAX_BLOCKS ? WFA_MAX_BLOCKS : b);
 }
 
+__device__ __forceinline__ float wfa_val_at(const void* v, int vdt, int64_t i) {
+    // vdt 2 = f32, 5 = bf16 (u16); i64 inputs are pre-cast to f32 by group()
+    if (vdt == 5) {
+        uint32_t u = ((const uint16_t*)v)[i];
+        union { uint32_t u; float f; } c;
+        c.u = u << 16;
+        return c.f;
+    }
+    return ((const float*)v)[i];
+}
+
 __device__ __forceinline__ uint64_t mix64s(uint64_t k) {
     k += 0x9e3779b97f4a7c15ULL;
     k = (k ^ (k >> 30)) * 0xbf58476d1ce4e5b9ULL;
@@ -201,7 +212,7 @@ __global__ void k_rs_scatter(const uint32_t* keys, const uint32_t* vals, int64_t
         int64_t i = base + j;
         if (i < n) {
             k_[j] = keys[i];
-            v_[j] = vals[i];
+            v_[j] = vals ? vals[i] : (uint32_t)i;  // null = implicit iota
             if (vals2) v2_[j] = vals2[i];
             cnt_local[(k_[j] >> shift) & 15]++;
             nit = j + 1;
@@ -342,7 +353,7 @@ __global__ void k_rs8_scatter(const uint32_t* keys, const uint32_t* vals,
         int64_t i = waveBase + j * 64 + lane;
         bool valid = i < n;
         uint32_t k = valid ? keys[i] : 0;
-        uint32_t v = valid ? vals[i] : 0;
+        uint32_t v = valid ? (vals ? vals[i] : (uint32_t)i) : 0;
         uint32_t d = (k >> shift) & 255;
         // same-digit ballot group among valid lanes of this round
         uint64_t mask = __ballot(valid);
@@ -413,7 +424,7 @@ __global__ void k_rs8_scatter_lds(const uint32_t* keys, const uint32_t* vals,
         int64_t i = waveBase + j * 64 + lane;
         bool valid = i < n;
         uint32_t k = valid ? keys[i] : 0;
-        uint32_t v = valid ? vals[i] : 0;
+        uint32_t v = valid ? (vals ? vals[i] : (uint32_t)i) : 0;
         uint32_t d = (k >> shift) & 255;
         uint64_t mask = __ballot(valid);
 #pragma unroll
@@ -479,10 +490,13 @@ extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                                 uint32_t* val2, uint32_t* val2_tmp,
                                 uint32_t* hist, int64_t n, int bits,
                                 uint32_t** out_slot, uint32_t** out_idx,
-                                uint32_t** out_val2) {
+                                uint32_t** out_val2, int implicit_iota) {
     hipStream_t st = (hipStream_t)s;
     uint32_t *ka = slot, *va = idx, *kb = slot_tmp, *vb = idx_tmp;
     uint32_t *wa = val2, *wb = val2_tmp;
+    // implicit iota: the first pass reads no payload (vals==nullptr means
+    // payload = global index) — saves the iota kernel + one 4n-byte read
+    bool first = true;
     if (bits > 4) {
         // 8-bit digits: fewer passes; hist is sized for 16*nblocks4 + 16
         // which covers 256*nblocks8 + 512 (RS8_PER_BLOCK = 2*RS_PER_BLOCK)
@@ -496,12 +510,14 @@ extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                                shift, hist, nb);
             hipLaunchKernelGGL(k_rs8_scan, dim3(256), dim3(256), 0, st, hist, nb, dt);
             hipLaunchKernelGGL(k_rs8_dbase, dim3(1), dim3(256), 0, st, dt, dbase);
+            uint32_t* va_eff = (first && implicit_iota) ? nullptr : va;
             if (val2)
                 hipLaunchKernelGGL(k_rs8_scatter, dim3(nb), dim3(WFA_THREADS), 0, st,
-                                   ka, va, n, shift, hist, nb, dbase, kb, vb, wa, wb);
+                                   ka, va_eff, n, shift, hist, nb, dbase, kb, vb, wa, wb);
             else
                 hipLaunchKernelGGL(k_rs8_scatter_lds, dim3(nb), dim3(WFA_THREADS), 0,
-                                   st, ka, va, n, shift, hist, nb, dbase, kb, vb);
+                                   st, ka, va_eff, n, shift, hist, nb, dbase, kb, vb);
+            first = false;
             uint32_t* t;
             t = ka; ka = kb; kb = t;
             t = va; va = vb; vb = t;
@@ -521,7 +537,9 @@ extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                            shift, hist, nblocks);
         hipLaunchKernelGGL(k_rs_scan16, dim3(16), dim3(256), 0, st, hist, nblocks, dt);
         hipLaunchKernelGGL(k_rs_scatter, dim3(nblocks), dim3(WFA_THREADS), 0, st, ka,
-                           va, n, shift, hist, nblocks, dt, kb, vb, wa, wb);
+                           (first && implicit_iota) ? nullptr : va, n, shift, hist,
+                           nblocks, dt, kb, vb, wa, wb);
+        first = false;
         uint32_t* t;
         t = ka; ka = kb; kb = t;
         t = va; va = vb; vb = t;
@@ -537,7 +555,7 @@ extern "C" void wfa_sort_pairs(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                                int64_t n, int bits, uint32_t** out_slot,
                                uint32_t** out_idx) {
     wfa_sort_pairs2(s, slot, idx, slot_tmp, idx_tmp, nullptr, nullptr, hist, n,
-                    bits, out_slot, out_idx, nullptr);
+                    bits, out_slot, out_idx, nullptr, 0);
 }
 
 // ===== segment extraction =====
@@ -688,12 +706,11 @@ __global__ void k_seg_reduce_wave(const uint32_t* seg_start, const uint32_t* seg
     for (int64_t j = wid; j < nseg; j += nw) {
         const int64_t b = seg_start[j];
         const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
-        const float* v = (const float*)v_orig;
         float acc = (comb == 1) ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
         int64_t tmax = INT64_MIN;
         for (int64_t i = b + lane; i < e; i += 64) {
             const uint32_t r = idx_sorted[i];
-            float x = v[r];
+            float x = wfa_val_at(v_orig, vdt, r);
             acc = (comb == 0 || comb == 3)
                       ? acc + x
                       : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
@@ -717,13 +734,13 @@ __global__ void k_seg_reduce_wave(const uint32_t* seg_start, const uint32_t* seg
 
 extern "C" void wfa_segment_reduce_wave(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
-    const int64_t* d_nseg, int64_t n, const void* v_orig,
+    const int64_t* d_nseg, int64_t n, const void* v_orig, int vdt,
     const uint32_t* idx_sorted, const int64_t* ts_orig, int comb,
     const uint64_t* slot_to_key, uint64_t* out_key, void* out_val,
     int64_t* out_ts, int64_t* d_out_n) {
     hipLaunchKernelGGL(k_seg_reduce_wave, dim3(WFA_MAX_BLOCKS), dim3(WFA_THREADS), 0,
                        (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_orig,
-                       idx_sorted, ts_orig, 2, comb, slot_to_key, out_key, out_val,
+                       idx_sorted, ts_orig, vdt, comb, slot_to_key, out_key, out_val,
                        out_ts, d_out_n);
 }
 
@@ -799,8 +816,8 @@ extern "C" void wfa_ffat_fire_offsets(wfa_stream_t s, const uint32_t* seg_start,
 // reference's per-key host loop + per-key stream (ffat_replica_gpu.hpp:
 // 829-867) becomes a dense device-side state machine.
 __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
-                          const int64_t* d_nseg, int64_t n, const float* v_f32,
-                          const uint32_t* idx_sorted, const int64_t* ts_orig,
+                          const int64_t* d_nseg, int64_t n, const void* v_f32,
+                          int vdt, const uint32_t* idx_sorted, const int64_t* ts_orig,
                           int64_t pane_len, int64_t P,
                           int64_t S, int comb, int ring_log2, int64_t* st_count,
                           uint32_t* st_fill, float* st_acc, float* ring,
@@ -825,7 +842,7 @@ __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
         int64_t w = fire_base[j];
         float* rg = ring + (size_t)slot * R;
         for (; i < e; ++i) {
-            float x = v_f32[idx_sorted[i]];
+            float x = wfa_val_at(v_f32, vdt, idx_sorted[i]);
             acc = (comb == 0) ? acc + x : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
             if (++fill == (uint32_t)pane_len) {
                 // pane complete
@@ -873,7 +890,8 @@ __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
 // waves on 256 CUs — 1.5 % occupancy, latency-bound).
 __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_slot,
                                const int64_t* d_nseg, int64_t n,
-                               const float* v_f32, const uint32_t* idx_sorted,
+                               const void* v_f32, int vdt,
+                               const uint32_t* idx_sorted,
                                const int64_t* ts_orig,
                                int64_t pane_len, int64_t P, int64_t S, int comb,
                                int ring_log2, int64_t* st_count, uint32_t* st_fill,
@@ -903,7 +921,9 @@ __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_sl
         float* rg = ring + (size_t)slot * R;
         for (int64_t pos = i0; pos < e; pos += 64) {
             const uint32_t nchunk = (uint32_t)min((int64_t)64, e - pos);
-            float v = (lane < (int)nchunk) ? v_f32[idx_sorted[pos + lane]] : ident;
+            float v = (lane < (int)nchunk)
+                          ? wfa_val_at(v_f32, vdt, idx_sorted[pos + lane])
+                          : ident;
             uint32_t rel = (lane < (int)nchunk) ? (fill + (uint32_t)lane) / L : ~0u;
             const uint32_t maxrel = (fill + nchunk - 1) / L;
             const uint32_t ncomplete = (fill + nchunk) / L;  // panes finished here
@@ -961,7 +981,7 @@ __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_sl
 
 extern "C" void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                                  const uint32_t* seg_slot, const int64_t* d_nseg,
-                                 int64_t n, const float* v_f32,
+                                 int64_t n, const void* v_f32, int vdt,
                                  const uint32_t* idx_sorted, const int64_t* ts_orig,
                                  int64_t pane_len, int64_t P,
                                  int64_t S, int comb, int ring_log2, int64_t* st_count,
@@ -972,13 +992,13 @@ extern "C" void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                                  float* out_val, int64_t* out_ts, int64_t out_cap) {
     if (pane_len >= 32)
         hipLaunchKernelGGL(k_ffat_cb_wave, dim3(WFA_MAX_BLOCKS), dim3(WFA_THREADS), 0,
-                           (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_f32,
+                           (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_f32, vdt,
                            idx_sorted, ts_orig, pane_len, P, S, comb, ring_log2,
                            st_count, st_fill, st_acc, ring, st_head, st_wsum,
                            slot_to_key, fire_base, out_key, out_val, out_ts, out_cap);
     else
         hipLaunchKernelGGL(k_ffat_cb, dim3(WFA_MAX_BLOCKS / 4), dim3(WFA_THREADS), 0,
-                           (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_f32,
+                           (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_f32, vdt,
                            idx_sorted, ts_orig, pane_len, P, S, comb, ring_log2,
                            st_count, st_fill, st_acc, ring, st_head, st_wsum,
                            slot_to_key, fire_base, out_key, out_val, out_ts, out_cap);
@@ -1055,8 +1075,8 @@ extern "C" void wfa_stateful_apply(wfa_stream_t s, const uint32_t* seg_start,
 
 // pass 1: per segment, accumulate tuples into the pending ring
 __global__ void k_tb_lift(const uint32_t* seg_start, const uint32_t* seg_slot,
-                          const int64_t* d_nseg, int64_t n, const float* v_f32,
-                          const uint32_t* idx_sorted, const int64_t* ts_orig,
+                          const int64_t* d_nseg, int64_t n, const void* v_f32,
+                          int vdt, const uint32_t* idx_sorted, const int64_t* ts_orig,
                           int64_t pane_len, int64_t P, int64_t S, int comb,
                           int pend_log2, float* pend, int64_t* pend_base,
                           int64_t* last_pane, uint32_t* ignored,
@@ -1089,7 +1109,7 @@ __global__ void k_tb_lift(const uint32_t* seg_start, const uint32_t* seg_slot,
                 atomicAdd(overflow, 1u);
                 continue;
             }
-            const float x = v_f32[idx_sorted[i]];
+            const float x = wfa_val_at(v_f32, vdt, idx_sorted[i]);
             float* cell = &pd[(uint64_t)p & Pm];
             *cell = (comb == 0) ? *cell + x
                                 : (comb == 1 ? fminf(*cell, x) : fmaxf(*cell, x));
@@ -1210,7 +1230,7 @@ __global__ void k_tb_advance(const uint32_t* n_slots, int64_t limit_pane,
 
 extern "C" void wfa_ffat_tb_round(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
-    const int64_t* d_nseg, int64_t n, const float* v_f32,
+    const int64_t* d_nseg, int64_t n, const void* v_f32, int vdt,
     const uint32_t* idx_sorted, const int64_t* ts_orig, int64_t pane_len,
     int64_t P, int64_t S, int comb, int ring_log2, int pend_log2,
     int64_t limit_pane, float* pend, int64_t* pend_base, int64_t* last_pane,
@@ -1221,7 +1241,7 @@ extern "C" void wfa_ffat_tb_round(
     hipStream_t st = (hipStream_t)s;
     if (n > 0)
         hipLaunchKernelGGL(k_tb_lift, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
-                           st, seg_start, seg_slot, d_nseg, n, v_f32, idx_sorted,
+                           st, seg_start, seg_slot, d_nseg, n, v_f32, vdt, idx_sorted,
                            ts_orig, pane_len, P, S, comb, pend_log2, pend, pend_base,
                            last_pane, ignored, overflow);
     hipLaunchKernelGGL(k_tb_count, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0, st,
@@ -1240,8 +1260,8 @@ extern "C" void wfa_ffat_tb_round(
 // query of the last P leaves via the classic two-pointer FlatFAT walk
 // (Tangwongsan VLDB'15 — reference flatfat.hpp:311-337 getResult).
 __global__ void k_ffat_tree(const uint32_t* seg_start, const uint32_t* seg_slot,
-                            const int64_t* d_nseg, int64_t n, const float* v_f32,
-                            const uint32_t* idx_sorted, const int64_t* ts_orig,
+                            const int64_t* d_nseg, int64_t n, const void* v_f32,
+                            int vdt, const uint32_t* idx_sorted, const int64_t* ts_orig,
                             int64_t pane_len, int64_t P,
                             int64_t S, int comb, int ring_log2, int64_t* st_count,
                             uint32_t* st_fill, float* st_acc, float* tree,
@@ -1265,7 +1285,7 @@ __global__ void k_ffat_tree(const uint32_t* seg_start, const uint32_t* seg_slot,
         int64_t w = fire_base[j];
         float* tr = tree + (size_t)slot * 2 * R;
         for (; i < e; ++i) {
-            float x = v_f32[idx_sorted[i]];
+            float x = wfa_val_at(v_f32, vdt, idx_sorted[i]);
             acc = TCOMB(acc, x);
             if (++fill == (uint32_t)pane_len) {
                 uint32_t leaf = (head & Rm) + R;
@@ -1317,14 +1337,14 @@ __global__ void k_ffat_tree(const uint32_t* seg_start, const uint32_t* seg_slot,
 
 extern "C" void wfa_ffat_tree_fold(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
-    const int64_t* d_nseg, int64_t n, const float* v_f32, const uint32_t* idx_sorted,
-    const int64_t* ts_orig,
+    const int64_t* d_nseg, int64_t n, const void* v_f32, int vdt,
+    const uint32_t* idx_sorted, const int64_t* ts_orig,
     int64_t pane_len, int64_t P, int64_t S, int comb, int ring_log2, int64_t* st_count,
     uint32_t* st_fill, float* st_acc, float* tree, uint32_t* st_head,
     const uint64_t* slot_to_key, const uint32_t* fire_base, uint64_t* out_key,
     float* out_val, int64_t* out_ts, int64_t out_cap) {
     hipLaunchKernelGGL(k_ffat_tree, dim3(WFA_MAX_BLOCKS / 4), dim3(WFA_THREADS), 0,
-                       (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_f32,
+                       (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_f32, vdt,
                        idx_sorted, ts_orig, pane_len, P, S, comb, ring_log2, st_count,
                        st_fill, st_acc, tree, st_head, slot_to_key, fire_base,
                        out_key, out_val, out_ts, out_cap);

@@ -57,7 +57,8 @@ void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                      uint32_t* slot_tmp, uint32_t* idx_tmp,
                      uint32_t* val2, uint32_t* val2_tmp, uint32_t* hist,
                      int64_t n, int bits, uint32_t** out_slot,
-                     uint32_t** out_idx, uint32_t** out_val2);
+                     uint32_t** out_idx, uint32_t** out_val2,
+                     int implicit_iota);
 int64_t wfa_sort_nblocks(int64_t n);
 int64_t wfa_sort_hist_u32(int64_t cap);  // hist scratch size in u32
 
@@ -85,7 +86,7 @@ void wfa_segment_reduce(wfa_stream_t s, const uint32_t* seg_start,
 // one key's segment — use for batches with large segments)
 void wfa_segment_reduce_wave(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
-    const int64_t* d_nseg, int64_t n, const void* v_orig,
+    const int64_t* d_nseg, int64_t n, const void* v_orig, int vdt,
     const uint32_t* idx_sorted, const int64_t* ts_orig, int comb,
     const uint64_t* slot_to_key, uint64_t* out_key, void* out_val,
     int64_t* out_ts, int64_t* d_out_n);
@@ -109,7 +110,7 @@ void wfa_ffat_fire_offsets(wfa_stream_t s, const uint32_t* seg_start,
 // Output appended to out_* at atomic cursor d_out_n.
 void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                       const uint32_t* seg_slot, const int64_t* d_nseg, int64_t n,
-                      const float* v_f32, const uint32_t* idx_sorted,
+                      const void* v_f32, int vdt, const uint32_t* idx_sorted,
                       const int64_t* ts_orig,
                       int64_t pane_len, int64_t P, int64_t S, int comb,
                       int ring_log2,
@@ -134,7 +135,7 @@ void wfa_stateful_apply(wfa_stream_t s, const uint32_t* seg_start,
 // n == 0 (and null segment args) for a pure watermark/EOS advance.
 void wfa_ffat_tb_round(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
-    const int64_t* d_nseg, int64_t n, const float* v_f32,
+    const int64_t* d_nseg, int64_t n, const void* v_f32, int vdt,
     const uint32_t* idx_sorted, const int64_t* ts_orig, int64_t pane_len,
     int64_t P, int64_t S, int comb, int ring_log2, int pend_log2,
     int64_t limit_pane, float* pend, int64_t* pend_base, int64_t* last_pane,
@@ -149,7 +150,7 @@ void wfa_ffat_tb_round(
 // O(log R) range query, one thread per segment, batched over all keys.
 void wfa_ffat_tree_fold(wfa_stream_t s, const uint32_t* seg_start,
                         const uint32_t* seg_slot, const int64_t* d_nseg, int64_t n,
-                        const float* v_f32, const uint32_t* idx_sorted,
+                        const void* v_f32, int vdt, const uint32_t* idx_sorted,
                         const int64_t* ts_orig,
                         int64_t pane_len, int64_t P, int64_t S, int comb,
                         int ring_log2,
