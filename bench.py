@@ -112,6 +112,31 @@ def build_ffat_x_graph(n_tuples, batch, n_keys, win, slide, rank, world, device,
     return g, snk
 
 
+def build_mapfilter_graph(n_tuples, batch, n_keys, rank, world, device):
+    """Driver config #2: Map_GPU + Filter_GPU chained, bf16 tuples."""
+    import windflow_amd as wf
+    from windflow_amd import native_gpu
+    from windflow_amd.builders_gpu import (Source_GPU_Builder, Map_GPU_Builder,
+                                           Filter_GPU_Builder, Sink_GPU_Builder)
+    src = (Source_GPU_Builder(
+        native_gpu.gpu_source(n_tuples, n_keys, batch, vdt=5, seed=42 + rank))
+        .withOutputSchema([5]).withOutputBatchSize(batch)
+        .withDevice(device).build())
+    mp_ = (Map_GPU_Builder(native_gpu.gpu_affine_map(0, 1.5, 0.25, dtype=5))
+           .withOutputSchema([5]).withOutputBatchSize(batch)
+           .withDevice(device).build())
+    fl = (Filter_GPU_Builder(native_gpu.gpu_gt_filter(0, 0.5, dtype=5))
+          .withOutputSchema([5]).withOutputBatchSize(batch)
+          .withDevice(device).build())
+    snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).withDevice(device).build()
+    g = wf.PipeGraph("bench_mapfilter")
+    mp = g.add_source(src)
+    mp.chain(mp_)
+    mp.chain(fl)
+    mp.chain_sink(snk)
+    return g, snk
+
+
 def build_join_graph(n_tuples, batch, n_keys, par):
     """Driver config #5 shape (CPU side): two keyed sources -> interval
     join (KP) -> paned windows -> sink, watermark collectors throughout."""
@@ -168,7 +193,7 @@ def main():
                     help="distinct keys per rank")
     ap.add_argument("--win", type=int, default=1000)
     ap.add_argument("--slide", type=int, default=100)
-    ap.add_argument("--config", choices=["ffat", "ffat_x", "a2a", "cpu", "join"], default="ffat")
+    ap.add_argument("--config", choices=["ffat", "ffat_x", "a2a", "mapfilter", "cpu", "join"], default="ffat")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -215,6 +240,9 @@ def main():
         n_gpus = 0
     else:
         def builder(steps):
+            if args.config == "mapfilter":
+                return build_mapfilter_graph(steps * B, B, args.keys, rank,
+                                             world, local_rank)
             if args.config == "a2a":
                 return build_a2a_graph(steps * B, B, args.keys, rank, world,
                                        local_rank, dist_cfg)
@@ -263,12 +291,14 @@ def main():
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": {"ffat": "bf16", "ffat_x": "bf16", "a2a": "f32",
-                      "cpu": "int64", "join": "int64"}[args.config],
+                      "mapfilter": "bf16", "cpu": "int64",
+                      "join": "int64"}[args.config],
             "data": "synthetic",
             "config": {
                 "model": {"ffat": "keyed_ffat_cb_window",
                           "ffat_x": "rccl_keyby_ffat_cb_window",
                           "a2a": "map_gpu_rccl_keyby_reduce_gpu",
+                          "mapfilter": "map_gpu_filter_gpu_chained",
                           "cpu": "cpu_source_map_filter_sink",
                           "join": "interval_join_paned_windows_cpu"}[args.config],
                 "global_batch": B * max(world, 1),
