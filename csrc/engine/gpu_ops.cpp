@@ -378,6 +378,7 @@ struct GpuFilterLogic : GpuLogicBase {
         ob->count = -1;
         ob->watermark = db->watermark;
         ob->stream_tag = db->stream_tag;
+        ob->born_us = db->born_us;
         if (ctx.stats) ctx.stats->num_kernels += 3;
         release_after_use(db);
         record_ready(ob);
@@ -727,6 +728,7 @@ struct GpuStatefulFilterLogic : GpuLogicBase {
         ob->count = -1;
         ob->watermark = db->watermark;
         ob->stream_tag = db->stream_tag;
+        ob->born_us = db->born_us;
         if (ctx.stats) ctx.stats->num_kernels += 4;
         release_after_use(db);
         record_ready(ob);
