@@ -983,6 +983,79 @@ struct GpuFfatLogic : GpuLogicBase {
     }
 };
 
+// ===== MFMA windowed Gram aggregator =====
+// Per-key tumbling windows over 16-dim f32 vectors; the GEMM-shaped
+// combine sum(v * v^T) runs on the matrix cores (see k_gram in
+// sortwin.hip and the guide's MFMA section).  Output: 16 rows per fired
+// window — (key, ts=last tuple, c0=gwid, c1..c16 = one Gram row).
+struct GpuGramLogic : GpuLogicBase {
+    int64_t win, max_keys;
+    KeyedScratch ks;
+    uint32_t* st_fill = nullptr;
+    uint32_t* st_head = nullptr;
+    float* st_acc = nullptr;
+    uint32_t* nf = nullptr;
+    int64_t* d_on = nullptr;
+    const float** d_incols = nullptr;
+    float** d_outcols = nullptr;
+
+    GpuGramLogic(int64_t w, int64_t mk, int dev, Schema os, int64_t cap) {
+        win = w;
+        max_keys = mk;
+        device = dev;
+        out_schema = os;  // [I64 gwid, F32 x16]
+        out_cap = cap;
+        if (os.payload.size() != 17)
+            throw std::runtime_error("gpu_gram: out schema must be [I64, F32 x16]");
+    }
+    void init_device() override {
+        ks.alloc(device, out_cap, max_keys, stream);
+        auto& A = arena(device);
+        st_fill = (uint32_t*)A.get(4 * max_keys);
+        st_head = (uint32_t*)A.get(4 * max_keys);
+        st_acc = (float*)A.get(4 * max_keys * 256);
+        nf = (uint32_t*)A.get(4 * (out_cap + 1));
+        d_on = (int64_t*)A.get(64);
+        d_incols = (const float**)A.get(16 * 16);
+        d_outcols = (float**)A.get(16 * 16);
+        HIPCHK(hipMemsetAsync(st_fill, 0, 4 * max_keys, stream));
+        HIPCHK(hipMemsetAsync(st_head, 0, 4 * max_keys, stream));
+        HIPCHK(hipMemsetAsync(st_acc, 0, 4 * max_keys * 256, stream));
+    }
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        Batch* db = input_on_device(in, ctx);
+        if (db->schema.payload.size() < 16)
+            throw std::runtime_error("gpu_gram: input needs 16 F32 columns");
+        int64_t n = db->count;
+        ks.group(stream, db, -1, ctx);
+        Batch* ob = get_dev();
+        const void* inp[16];
+        void* outp[16];
+        for (int c = 0; c < 16; ++c) {
+            inp[c] = db->cols[c];
+            outp[c] = ob->cols[1 + c];
+        }
+        HIPCHK(hipMemcpyAsync(d_incols, inp, 16 * 8, hipMemcpyHostToDevice, stream));
+        HIPCHK(hipMemcpyAsync(d_outcols, outp, 16 * 8, hipMemcpyHostToDevice, stream));
+        // tumbling windows = panes of `win`, fire every pane (P=1, S=1)
+        wfa_ffat_fire_offsets(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
+                              win, 1, 1, st_fill, st_head, nf, d_on);
+        wfa_gram_fold(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n, d_incols,
+                      ks.idx_sorted, db->ts, win, st_fill, st_acc, st_head,
+                      ks.slot_to_key, nf, ob->key, ob->col<int64_t>(0), d_outcols,
+                      ob->ts, ob->capacity, d_on);
+        HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
+        ob->count = -1;
+        ob->watermark = db->watermark;
+        ob->born_us = db->born_us;
+        if (ctx.stats) ctx.stats->num_kernels += 3;
+        release_after_use(db);
+        record_ready(ob);
+        out.emit(ob);
+    }
+};
+
 // ===== GPU -> host stage (D2H) =====
 struct GpuToHostLogic : GpuLogicBase {
     explicit GpuToHostLogic(int dev) { device = dev; }
@@ -1399,6 +1472,9 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
             (int)ip[0], (int)ip[1], ip[2], ip[3], ip[4], ip[5] != 0, device, os,
             out_batch, ip.size() > 6 && ip[6] != 0, ip.size() > 7 ? ip[7] : 0,
             ip.size() > 8 ? (int)ip[8] : 0);
+    if (kind == "gpu_gram")
+        // ip: [win, max_keys]
+        return std::make_shared<GpuGramLogic>(ip[0], ip[1], device, os, out_batch);
     if (kind == "gpu_exchange")
         return std::make_shared<GpuExchangeLogic>(eng, device, os, out_batch);
     if (kind == "gpu_to_host")

@@ -1004,6 +1004,103 @@ extern "C" void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                            slot_to_key, fire_base, out_key, out_val, out_ts, out_cap);
 }
 
+// ===== MFMA windowed Gram aggregation (matrix-core combine) =====
+// Per-key tumbling windows over 16-dim f32 tuple vectors; the window
+// aggregate Σ v·vᵀ (online covariance / Gram) is GEMM-shaped, so it runs
+// on the matrix cores: C(16x16) += A(16x4)·B(4x16) with A = Vᵀ, B = V via
+// v_mfma_f32_16x16x4_f32 (exact f32, guide §3).  Lane layout (CDNA):
+//   A[i=l&15][k=l>>4], B[k=l>>4][j=l&15]  -> a = b = V[k][l&15]
+//   C/D: col = l&15, row = (l>>4)*4 + reg
+// One wave per key segment; the 4-reg accumulator persists across batches
+// in a per-slot arena laid out in the SAME lane mapping.
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+__global__ void k_gram(const uint32_t* seg_start, const uint32_t* seg_slot,
+                       const int64_t* d_nseg, int64_t n,
+                       const float* const* colp,  // [16] input column ptrs
+                       const uint32_t* idx_sorted, const int64_t* ts_orig,
+                       int64_t win, uint32_t* st_fill, float* st_acc /*256/slot*/,
+                       uint32_t* st_head, const uint64_t* slot_to_key,
+                       const uint32_t* fire_base, uint64_t* out_key,
+                       int64_t* out_gwid, float* const* out_colp /*16*/,
+                       int64_t* out_ts, int64_t out_cap) {
+    const int64_t nseg = *d_nseg;
+    const int lane = threadIdx.x & 63;
+    const int64_t wid = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
+    const int krow = lane >> 4;   // 0..3 (K slice of this step)
+    const int dim = lane & 15;    // matrix column
+    const float* mycol = colp[dim];
+    for (int64_t j = wid; j < nseg; j += nw) {
+        const uint32_t slot = seg_slot[j];
+        const int64_t i0 = seg_start[j];
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        int64_t cnt = st_fill[slot];
+        uint32_t head = st_head[slot];
+        f32x4 acc;
+        float* sa = st_acc + (size_t)slot * 256;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) acc[r] = sa[lane * 4 + r];
+        int64_t w = fire_base[j];
+        int64_t pos = i0;
+        int64_t last_ts = 0;
+        while (pos < e) {
+            int64_t chunk = min((int64_t)4, min(e - pos, win - cnt));
+            float v = 0.f;
+            if (krow < chunk) v = mycol[idx_sorted[pos + krow]];
+            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(v, v, acc, 0, 0, 0);
+            last_ts = ts_orig[idx_sorted[pos + chunk - 1]];
+            pos += chunk;
+            cnt += chunk;
+            if (cnt == win) {
+                // fire: 16 output rows (one Gram row each)
+                const int64_t base = w * 16;
+                if (base + 16 <= out_cap) {
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        int row = (lane >> 4) * 4 + r;
+                        out_colp[dim][base + row] = acc[r];
+                        if (dim == 0) {
+                            out_key[base + row] = slot_to_key[slot];
+                            out_gwid[base + row] = head;
+                            out_ts[base + row] = last_ts;
+                        }
+                    }
+                }
+                ++w;
+                ++head;
+#pragma unroll
+                for (int r = 0; r < 4; ++r) acc[r] = 0.f;
+                cnt = 0;
+            }
+        }
+        st_fill[slot] = (uint32_t)cnt;
+        st_head[slot] = head;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) sa[lane * 4 + r] = acc[r];
+    }
+}
+
+__global__ void k_scale16(int64_t* d_out_n) { *d_out_n *= 16; }
+
+extern "C" void wfa_gram_fold(wfa_stream_t s, const uint32_t* seg_start,
+                              const uint32_t* seg_slot, const int64_t* d_nseg,
+                              int64_t n, const float* const* colp,
+                              const uint32_t* idx_sorted, const int64_t* ts_orig,
+                              int64_t win, uint32_t* st_fill, float* st_acc,
+                              uint32_t* st_head, const uint64_t* slot_to_key,
+                              const uint32_t* fire_base, uint64_t* out_key,
+                              int64_t* out_gwid, float* const* out_colp,
+                              int64_t* out_ts, int64_t out_cap,
+                              int64_t* d_out_n) {
+    hipStream_t st = (hipStream_t)s;
+    hipLaunchKernelGGL(k_gram, dim3(WFA_MAX_BLOCKS / 2), dim3(WFA_THREADS), 0, st,
+                       seg_start, seg_slot, d_nseg, n, colp, idx_sorted, ts_orig,
+                       win, st_fill, st_acc, st_head, slot_to_key, fire_base,
+                       out_key, out_gwid, out_colp, out_ts, out_cap);
+    hipLaunchKernelGGL(k_scale16, dim3(1), dim3(1), 0, st, d_out_n);
+}
+
 // ===== stateful map/filter: per-key device state, key-order walk =====
 // Reference Stateful_MAPGPU_Kernel / Stateful_FILTERGPU_Kernel
 // (map_gpu.hpp:80-102, filter_gpu.hpp:92-115): one worker per distinct key

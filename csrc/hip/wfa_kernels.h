@@ -120,6 +120,21 @@ void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                       uint64_t* out_key, float* out_val, int64_t* out_ts,
                       int64_t out_cap);
 
+// MFMA windowed Gram aggregation: per-key tumbling windows over 16-dim
+// f32 vectors; window combine Σ v·vᵀ on the matrix cores
+// (v_mfma_f32_16x16x4_f32).  Output: 16 rows per fired window
+// (key, gwid, ts, Gram row).  d_out_n (from the fire-offset scan) is
+// scaled to rows (×16) by the wrapper.
+void wfa_gram_fold(wfa_stream_t s, const uint32_t* seg_start,
+                   const uint32_t* seg_slot, const int64_t* d_nseg,
+                   int64_t n, const float* const* colp,
+                   const uint32_t* idx_sorted, const int64_t* ts_orig,
+                   int64_t win, uint32_t* st_fill, float* st_acc,
+                   uint32_t* st_head, const uint64_t* slot_to_key,
+                   const uint32_t* fire_base, uint64_t* out_key,
+                   int64_t* out_gwid, float* const* out_colp,
+                   int64_t* out_ts, int64_t out_cap, int64_t* d_out_n);
+
 // stateful map/filter: per-key (slot) f64 state, key-order segment walk;
 // map writes results to original positions in place, filter fills flags
 void wfa_stateful_apply(wfa_stream_t s, const uint32_t* seg_start,
