@@ -39,10 +39,11 @@ def test_window_fuzz(case):
     batch = rng.choice([32, 128, 1000])
     par = rng.randint(1, 3)
     stream = rng.choice([1500, 3100])
+    mode = rng.choice([wf.ExecutionMode.DEFAULT, wf.ExecutionMode.DETERMINISTIC])
     b = BUILDERS[kind]((agg, 0))
     b = (b.withCBWindows(win, slide) if wt == "cb" else b.withTBWindows(win, slide))
     op = b.withParallelism(par).withOutputSchema([0]).build()
-    rows = run_graph(op, stream_len=stream, n_keys=n_keys, batch=batch)
+    rows = run_graph(op, stream_len=stream, n_keys=n_keys, batch=batch, mode=mode)
     per = seq_stream(stream, n_keys)
     exp = (oracle_cb if wt == "cb" else oracle_tb)(per, win, slide, agg)
     assert got_counter(rows) == exp, (kind, agg, wt, win, slide, n_keys, batch, par)
