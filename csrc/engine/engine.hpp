@@ -147,7 +147,7 @@ struct KeyByEmitter : Emitter {
     int64_t last_cadence_us;
     int64_t cadence_us = 100000;        // WF_DEFAULT_WM_INTERVAL_USEC
     int64_t cur_wm = 0;
-    std::vector<uint32_t> cnt, off;     // scratch
+    std::vector<uint32_t> cnt;          // per-destination count scratch
 
     KeyByEmitter(std::vector<SpscQueue*> d, Pool* op, Pool* pp, int64_t ob)
         : dests(std::move(d)), out_pool(op), punct_pool(pp), out_batch(ob) {

@@ -162,7 +162,6 @@ struct GpuLogicBase : OpLogic {
     std::unique_ptr<Pool> dev_pool;  // device batches this logic emits
     Schema out_schema;
     int64_t out_cap = 1 << 20;
-    int64_t* h_count = nullptr;      // pinned readback
     bool inited = false;
 
     virtual void init_device() {}
@@ -170,7 +169,6 @@ struct GpuLogicBase : OpLogic {
         if (inited) return;
         HIPCHK(hipSetDevice(device));
         HIPCHK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
-        HIPCHK(hipHostMalloc((void**)&h_count, 64, hipHostMallocDefault));
         dev_pool = std::make_unique<Pool>(out_schema, out_cap, false);
         dev_pool->loc = Loc::DEVICE;
         dev_pool->device = device;
@@ -178,7 +176,6 @@ struct GpuLogicBase : OpLogic {
         inited = true;
     }
     ~GpuLogicBase() override {
-        if (h_count) (void)hipHostFree(h_count);
         if (stream) (void)hipStreamDestroy(stream);
     }
 
@@ -395,9 +392,8 @@ struct KeyedScratch {
     uint32_t* hist = nullptr;
     uint32_t *seg_start = nullptr, *seg_slot = nullptr;
     int64_t* d_nseg = nullptr;
-    float* v_sorted = nullptr;   // sort output (value payload)
-    int64_t* ts_sorted = nullptr;
-    float* v_f32 = nullptr;  // cast buffer / sort ping-pong
+    float* v_sorted = nullptr;   // second-payload ping-pong (sort_pairs2)
+    float* v_f32 = nullptr;      // cast buffer / sort ping-pong
     int64_t table_cap = 0;
     int64_t max_keys = 0;
     int64_t cap = 0;
