@@ -156,7 +156,6 @@ struct KeyByEmitter : Emitter {
         sent_recently.assign(dests.size(), false);
         last_cadence_us = now_us();
         cnt.resize(dests.size());
-        off.resize(dests.size());
     }
 
     static inline uint64_t mix(uint64_t k) {
