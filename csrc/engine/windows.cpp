@@ -471,6 +471,125 @@ struct WlqLogic : OpLogic {
     }
 };
 
+// WLQ, keyby-partitioned fast path: a key's panes arrive IN ORDER from its
+// single owning PLQ replica, so the window recombination runs over an
+// ordered per-key deque (front-scan per fire) instead of per-(key,gwid)
+// hash accumulation — no hash ops, cache-friendly, ~5x on dense configs.
+struct WlqKeyedLogic : OpLogic {
+    WinType wt;
+    int64_t win_p, slide_p, lateness, pane_len;
+    AggCfg agg;
+
+    struct KeyState {
+        std::deque<std::pair<int64_t, Acc>> buf;  // (pane id, partial), ordered
+        int64_t next_w = -1;      // next window to fire
+        int64_t last_id = -1;     // newest pane id seen
+        int64_t last_ts = 0;
+        bool armed = false;       // TB: heap entry outstanding for next_w
+    };
+    std::unordered_map<uint64_t, KeyState> keys;
+    using HeapEl = std::pair<int64_t, uint64_t>;
+    std::priority_queue<HeapEl, std::vector<HeapEl>, std::greater<HeapEl>> fire_heap;
+    OutBuf ob;
+    int64_t cur_wm = 0;
+    int64_t ignored = 0;
+
+    WlqKeyedLogic(WinType wt_, int64_t wp, int64_t sp, int64_t lat, AggCfg a,
+                  int64_t pl)
+        : wt(wt_), win_p(wp), slide_p(sp), lateness(lat), pane_len(pl), agg(a) {}
+
+    int64_t fire_at(int64_t w) const {
+        return (w * slide_p + win_p) * pane_len + lateness;
+    }
+
+    void fire(uint64_t key, KeyState& ks, EmitCtx& out) {
+        const int64_t wS = ks.next_w * slide_p, wE = wS + win_p;
+        Acc acc;
+        for (auto& [id, part] : ks.buf) {
+            if (id >= wE) break;
+            agg.merge(acc, part);  // ids < wS were dropped after earlier fires
+        }
+        int64_t i = ob.slot(out, cur_wm);
+        Batch* o = ob.b;
+        o->key[i] = key;
+        o->ts[i] = wt == WinType::CB ? ks.last_ts : wE * pane_len - 1;
+        write_val(o, 0, i, agg.result(acc), agg.use_int);
+        ks.next_w++;
+        const int64_t keep_from = ks.next_w * slide_p;
+        while (!ks.buf.empty() && ks.buf.front().first < keep_from)
+            ks.buf.pop_front();
+    }
+
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
+        const int64_t n = b->count;
+        for (int64_t i = 0; i < n; ++i) {
+            uint64_t key = b->key[i];
+            int64_t p = b->col<int64_t>(0)[i];
+            auto& ks = keys[key];
+            if (ks.next_w < 0) {
+                int64_t w0 = p - win_p + 1;
+                ks.next_w = w0 <= 0 ? 0 : (w0 + slide_p - 1) / slide_p;
+            }
+            if (p < ks.next_w * slide_p) {  // pane for an already-fired window
+                ignored++;
+                continue;
+            }
+            Acc part;
+            ValU v = read_val(b, 1, i);
+            part.i = v.i;
+            part.f = v.f;
+            part.cnt = agg.comb == C_COUNT ? v.i : 1;
+            ks.buf.emplace_back(p, part);
+            ks.last_id = std::max(ks.last_id, p);
+            ks.last_ts = std::max(ks.last_ts, b->ts[i]);
+            if (wt == WinType::CB) {
+                // CB panes are dense per key: window complete when its last
+                // pane (wS + win_p - 1) arrives
+                while (ks.next_w * slide_p + win_p - 1 <= ks.last_id)
+                    fire(key, ks, out);
+            } else if (!ks.armed) {
+                fire_heap.emplace(fire_at(ks.next_w), key);
+                ks.armed = true;
+            }
+        }
+        cur_wm = std::max(cur_wm, ctx.current_wm);
+        release(b);
+        if (wt == WinType::TB) fire_tb(out);
+        ob.flush(out, cur_wm);
+    }
+
+    void fire_tb(EmitCtx& out) {
+        while (!fire_heap.empty() && fire_heap.top().first <= cur_wm) {
+            uint64_t key = fire_heap.top().second;
+            fire_heap.pop();
+            auto& ks = keys[key];
+            ks.armed = false;
+            while (fire_at(ks.next_w) <= cur_wm &&
+                   ks.next_w * slide_p <= ks.last_id)
+                fire(key, ks, out);
+            if (ks.next_w * slide_p <= ks.last_id || !ks.buf.empty()) {
+                fire_heap.emplace(fire_at(ks.next_w), key);
+                ks.armed = true;
+            }
+        }
+    }
+
+    bool on_punct(int64_t wm, EmitCtx& out, RuntimeCtx&) override {
+        cur_wm = std::max(cur_wm, wm);
+        if (wt == WinType::TB) fire_tb(out);
+        ob.flush(out, cur_wm);
+        return false;
+    }
+
+    void on_eos(EmitCtx& out, RuntimeCtx&) override {
+        for (auto& [key, ks] : keys)
+            while (ks.next_w >= 0 && ks.next_w * slide_p <= ks.last_id &&
+                   !ks.buf.empty())
+                fire(key, ks, out);
+        ob.flush(out, cur_wm);
+    }
+};
+
 // ============== MapReduce REDUCE: merge n partials per window ==============
 struct MrReduceLogic : OpLogic {
     int64_t n_partials;
@@ -1022,6 +1141,9 @@ std::shared_ptr<OpLogic> make_window_logic(const std::string& kind,
     }
     if (kind == "win_wlq") {
         AggCfg a{(int)geti(4), geti(7, 1) != 0};
+        if (geti(6) == 0)  // keyby-partitioned: ordered per-key fast path
+            return std::make_shared<WlqKeyedLogic>(wt, geti(1), geti(2), geti(3),
+                                                   a, geti(5, 1));
         return std::make_shared<WlqLogic>(wt, geti(1), geti(2), geti(3), a,
                                           geti(5, 1), (int)geti(6));
     }
