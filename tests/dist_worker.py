@@ -22,6 +22,20 @@ from windflow_amd.dist import cpu_keyby_exchange, _hash_keys  # noqa: E402
 def main():
     td.init_process_group("gloo")
     rank, world = td.get_rank(), td.get_world_size()
+
+    # RCCL bootstrap helper: id bytes must broadcast identically (this is
+    # what wires the 8-GPU exchange in bench.py)
+    from windflow_amd.dist import init_from_torch
+    r, w, rid = init_from_torch()
+    assert (r, w) == (rank, world)
+    ok = [None]
+    if rank == 0:
+        td.broadcast_object_list([rid], src=0)
+        ok[0] = True
+    else:
+        got = [None]
+        td.broadcast_object_list(got, src=0)
+        assert got[0] == rid, "rccl id bytes differ across ranks"
     n, n_keys, batch = 20000, 13, 500  # n % batch == 0: equal rounds per rank
 
     got = {"sum": 0, "rows": 0, "bad_dest": 0}
