@@ -183,6 +183,8 @@ struct WindowCore : OpLogic {
         int64_t next_gwid = -1; // next window to open (-1: uninitialized)
         int64_t first_unfired = 0;  // sparse mode: panes below this fired
         int64_t last_ts = 0;
+        int64_t cum_i = 0;      // CB sum prefix (prefix mode)
+        double cum_f = 0;
         std::deque<Open> open;
     };
     std::unordered_map<uint64_t, KeyState> keys;
@@ -194,11 +196,17 @@ struct WindowCore : OpLogic {
     int64_t rr_ctr = 0;  // mr_map round-robin arrival counter
     int replica = 0, par = 1;
     int64_t ignored = 0;
+    // CB + sum/count: window result = prefix@fire - prefix@open, O(1) per
+    // tuple instead of one add per open window (win/slide adds)
+    bool prefix = false;
 
     WindowCore(WinType wt_, int64_t w, int64_t s, int64_t lat, AggCfg a, int c,
                int own, bool meta, bool start_ts, bool rr, Engine* e)
         : wt(wt_), win(w), slide(s), lateness(lat), agg(a), col(c), own_mode(own),
-          emit_meta(meta), emit_start_ts(start_ts), subset_rr(rr), eng(e) {}
+          emit_meta(meta), emit_start_ts(start_ts), subset_rr(rr), eng(e) {
+        prefix = wt == WinType::CB &&
+                 (agg.comb == C_SUM || agg.comb == C_COUNT);
+    }
 
     void warm(RuntimeCtx& ctx) override {
         replica = ctx.replica;
@@ -227,17 +235,42 @@ struct WindowCore : OpLogic {
                 EmitCtx& out) {
         if (ks.next_gwid < 0) ks.next_gwid = 0;
         while (ks.next_gwid * slide <= ks.idx) {
-            ks.open.push_back({ks.next_gwid, ks.next_gwid * slide, {}, owned(ks.next_gwid)});
+            Open w{ks.next_gwid, ks.next_gwid * slide, {}, owned(ks.next_gwid)};
+            if (prefix) {  // remember the prefix at window open
+                w.acc.i = ks.cum_i;
+                w.acc.f = ks.cum_f;
+            }
+            ks.open.push_back(w);
             ks.next_gwid++;
         }
-        for (auto& w : ks.open)
-            if (w.owned && mine && ks.idx >= w.start) agg.add(w.acc, v);
+        if (prefix) {
+            if (mine) {  // COUNT runs the prefix over 1s (subset-correct)
+                ks.cum_i += agg.comb == C_COUNT ? 1 : v.i;
+                ks.cum_f += agg.comb == C_COUNT ? 1.0 : v.f;
+            }
+        } else {
+            for (auto& w : ks.open)
+                if (w.owned && mine && ks.idx >= w.start) agg.add(w.acc, v);
+        }
         ks.last_ts = ts;
         while (!ks.open.empty() && ks.idx == ks.open.front().start + win - 1) {
-            fire(ks.open.front(), key, ts, out);
+            Open& w = ks.open.front();
+            if (prefix) materialize_prefix(ks, w);
+            fire(w, key, ts, out);
             ks.open.pop_front();
         }
         ks.idx++;
+    }
+
+    // turn (prefix@open stored in acc) into the actual window accumulator
+    void materialize_prefix(KeyState& ks, Open& w) {
+        Acc a;
+        a.i = ks.cum_i - w.acc.i;
+        a.f = ks.cum_f - w.acc.f;
+        a.cnt = agg.comb == C_COUNT
+                    ? a.i  // prefix of 1s (exact under mr_map subsets)
+                    : std::min(ks.idx + 1, w.start + win) - w.start;
+        w.acc = a;
     }
 
     void add_tb_sparse(uint64_t key, KeyState& ks, int64_t ts, const ValU& v,
@@ -344,6 +377,13 @@ struct WindowCore : OpLogic {
         for (auto& [key, ks] : keys) {
             for (auto& w : ks.open) {
                 if (wt == WinType::TB && w.acc.cnt == 0 && !emit_meta) continue;
+                if (prefix) {
+                    Acc a;
+                    a.i = ks.cum_i - w.acc.i;
+                    a.f = ks.cum_f - w.acc.f;
+                    a.cnt = agg.comb == C_COUNT ? a.i : ks.idx - w.start;
+                    w.acc = a;
+                }
                 fire(w, key,
                      wt == WinType::CB ? ks.last_ts
                                        : (emit_start_ts ? w.start
