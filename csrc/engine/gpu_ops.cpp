@@ -1187,6 +1187,9 @@ struct GpuExchangeLogic : GpuLogicBase {
         out_cap = cap;
         rank = e->dist_rank;
         world = e->dist_world;
+        if (world > 8)
+            throw std::runtime_error(
+                "gpu_keyby_exchange supports up to 8 ranks (one xGMI node)");
         bits = 1;
         while ((1 << bits) < world) ++bits;
     }
