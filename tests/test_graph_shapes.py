@@ -264,3 +264,53 @@ def test_replica_error_propagates():
     mp.add_sink(wf.Sink_Builder(native.count_sink()).withParallelism(1).build())
     with pytest.raises(RuntimeError, match="boom"):
         g.run()
+
+
+def test_partial_merge_of_split_branches():
+    """Split 3 ways; merge branches 0+2, keep branch 1 separate (the
+    reference's merge-partial shape, pipegraph.hpp:387)."""
+    n = 9000
+    g = wf.PipeGraph("pm")
+    src = (wf.Source_Builder(native.seq_source(n, 5, 64))
+           .withParallelism(1).withOutputSchema([0]).build())
+    mp = g.add_source(src)
+    mp.split(native.split_mod(0), 3)
+    b0, b1, b2 = mp.select(0), mp.select(1), mp.select(2)
+    merged = b0.merge(b2)
+    s02 = wf.Sink_Builder(native.sum_sink(0)).withParallelism(2).build()
+    merged.add_sink(s02)
+    s1 = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+    b1.add_sink(s1)
+    g.run()
+    exp02 = sum(v for v in range(1, n + 1) if v % 3 in (0, 2))
+    exp1 = sum(v for v in range(1, n + 1) if v % 3 == 1)
+    assert g.sink_sum(s02) == exp02
+    assert g.sink_sum(s1) == exp1
+
+
+def test_nested_split():
+    """Split a split branch again (nested DAG)."""
+    n = 8000
+    g = wf.PipeGraph("ns")
+    src = (wf.Source_Builder(native.seq_source(n, 5, 64))
+           .withParallelism(1).withOutputSchema([0]).build())
+    mp = g.add_source(src)
+    mp.split(native.split_mod(0), 2)          # by v % 2
+    b0, b1 = mp.select(0), mp.select(1)
+    s1 = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+    b1.add_sink(s1)
+    # halve evens again: v/2 parity
+    b0.add(wf.Map_Builder(native.affine_map(0, 1, 0)).withParallelism(1)
+           .withOutputSchema([0]).build())
+    b0.split(native.split_mod(0), 2)          # v%2==0 -> all to branch 0...
+    c0, c1 = b0.select(0), b0.select(1)
+    sc0 = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+    sc1 = wf.Sink_Builder(native.count_sink()).withParallelism(1).build()
+    c0.add_sink(sc0)
+    c1.add_sink(sc1)
+    g.run()
+    evens = [v for v in range(1, n + 1) if v % 2 == 0]
+    odds = sum(v for v in range(1, n + 1) if v % 2 == 1)
+    assert g.sink_sum(s1) == odds
+    assert g.sink_sum(sc0) == sum(evens)   # evens % 2 == 0 -> branch 0
+    assert g.sink_count(sc1) == 0
