@@ -261,12 +261,13 @@ def join_graph(mode_builder, n=2000, keys=5, lower=-10, upper=10, batch=64,
 
 def oracle_join(n, keys, lower, upper):
     """(a from A, b from B) with same key and b.ts - a.ts in [lower, upper].
-    A: v=1..n (ts=v, val=v); B: same ts, val=v+100000."""
+    A: ts=v, val=v, key=v%keys; B: ts=v, val=v+100000,
+    key=(v+100000)%keys (seq_source keys by the OFFSET value)."""
     out = Counter()
     for v in range(1, n + 1):         # a
         k = v % keys
         for bts in range(v + lower, v + upper + 1):
-            if 1 <= bts <= n and bts % keys == k:
+            if 1 <= bts <= n and (bts + 100000) % keys == k:
                 out[(k, v, bts + 100000)] += 1
     return out
 
@@ -353,3 +354,25 @@ def test_windows_deterministic_mode():
                      .withOutputSchema([0]).build(),
                      mode=wf.ExecutionMode.DETERMINISTIC)
     assert got_counter(rows) == oracle_cb(seq_stream(3000, 7), win, slide)
+
+
+def test_py_window_parallel_replicas():
+    """Non-incremental python windows across 3 keyed replicas."""
+    win, slide = 30, 10
+
+    def spread(w):
+        return float(w['c0'].max() - w['c0'].min()) if len(w['c0']) else 0.0
+
+    rows = run_graph(Keyed_Windows_Builder(func=spread).withCBWindows(win, slide)
+                     .withParallelism(3).withOutputSchema([1]).build(),
+                     stream_len=2100)
+    per = seq_stream(2100, 7)
+    exp = Counter()
+    for k, r in per.items():
+        vals = [v for _, v in r]
+        w = 0
+        while w * slide < len(vals):
+            seg = vals[w * slide: w * slide + win]
+            exp[(k, float(max(seg) - min(seg)))] += 1
+            w += 1
+    assert Counter((k, v) for _, k, v in rows) == exp

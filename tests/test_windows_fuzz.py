@@ -47,3 +47,25 @@ def test_window_fuzz(case):
     per = seq_stream(stream, n_keys)
     exp = (oracle_cb if wt == "cb" else oracle_tb)(per, win, slide, agg)
     assert got_counter(rows) == exp, (kind, agg, wt, win, slide, n_keys, batch, par)
+
+
+@pytest.mark.parametrize("case", range(6))
+def test_join_fuzz(case):
+    """Randomized interval joins: bounds/keys/batch/parallelism/mode drawn
+    per case, vs the brute-force pair oracle."""
+    from collections import Counter
+    from test_windows import join_graph, oracle_join
+    rng = random.Random(77 + case)
+    lower = -rng.randint(0, 8)
+    upper = rng.randint(0, 8)
+    keys = rng.choice([2, 5, 9])
+    n = rng.choice([800, 1500])
+    batch = rng.choice([16, 64, 300])
+    par = rng.randint(1, 3)
+    kp = rng.random() < 0.5
+    rows = join_graph(lambda b: (b.withKPMode() if kp else b.withDPMode())
+                      .withParallelism(par),
+                      n=n, keys=keys, lower=lower, upper=upper, batch=batch)
+    got = Counter((k, a, bb) for _, k, a, bb in rows)
+    assert got == oracle_join(n, keys, lower, upper), (lower, upper, keys, n,
+                                                       batch, par, kp)
