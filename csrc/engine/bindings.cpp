@@ -482,12 +482,14 @@ PYBIND11_MODULE(_core, m) {
         void* cache = nullptr;
     };
     py::class_<StateStore>(m, "StateStore")
-        .def(py::init([](const std::string& path, int64_t cache_cap) {
+        .def(py::init([](const std::string& path, int64_t cache_cap, bool fresh) {
                  auto s = new StateStore();
-                 s->holder = open_state_store(path, cache_cap, &s->kv, &s->cache);
+                 s->holder =
+                     open_state_store(path, cache_cap, &s->kv, &s->cache, fresh);
                  return s;
              }),
-             py::arg("path"), py::arg("cache_capacity") = 1 << 16)
+             py::arg("path"), py::arg("cache_capacity") = 1 << 16,
+             py::arg("fresh") = true)
         .def("get",
              [](StateStore& s, uint64_t key) -> py::object {
                  std::string* v = state_cache_get(s.cache, key);

@@ -441,7 +441,8 @@ std::shared_ptr<OpLogic> make_persist_logic(const std::string& kind,
                                             const std::vector<int64_t>& ip,
                                             Engine* eng, int op_id);
 std::shared_ptr<void> open_state_store(const std::string& path, int64_t cache_cap,
-                                       void** kv_out, void** cache_out);
+                                       void** kv_out, void** cache_out,
+                                       bool fresh = true);
 std::string* state_cache_get(void* cache, uint64_t key);
 void state_cache_put(void* cache, uint64_t key, const std::string& v);
 void state_cache_flush(void* cache);

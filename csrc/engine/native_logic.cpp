@@ -218,6 +218,28 @@ struct SumSinkI64 : OpLogic {
     }
 };
 
+// ----- sink: sum over the LAST value seen per key (final accumulators) -----
+struct LastPerKeySinkI64 : OpLogic {
+    Engine* eng;
+    int op_id;
+    int col;
+    std::unordered_map<uint64_t, int64_t> last;
+    int64_t tuples = 0;
+    LastPerKeySinkI64(Engine* e, int id, int c) : eng(e), op_id(id), col(c) {}
+    void process(Batch* b, EmitCtx&, RuntimeCtx&) override {
+        int64_t* x = b->col<int64_t>(col);
+        for (int64_t i = 0; i < b->count; ++i) last[b->key[i]] = x[i];
+        tuples += b->count;
+        release(b);
+    }
+    void on_eos(EmitCtx&, RuntimeCtx&) override {
+        int64_t s = 0;
+        for (auto& [k, v] : last) s += v;
+        eng->sink_acc_i64[op_id].fetch_add(s, std::memory_order_relaxed);
+        eng->sink_tuples[op_id].fetch_add(tuples, std::memory_order_relaxed);
+    }
+};
+
 // ----- sink: count only (throughput benchmarks) -----
 struct CountSink : OpLogic {
     Engine* eng;
@@ -302,6 +324,8 @@ std::shared_ptr<OpLogic> make_native_logic(const std::string& kind, const std::s
         return std::make_shared<KeyedSumReduceI64>((int)ip[0]);
     if (kind == "sink" && spec == "sum_i64")
         return std::make_shared<SumSinkI64>(eng, op_id, (int)ip[0]);
+    if (kind == "sink" && spec == "last_per_key_i64")
+        return std::make_shared<LastPerKeySinkI64>(eng, op_id, (int)ip[0]);
     if (kind == "sink" && spec == "count")
         return std::make_shared<CountSink>(eng, op_id);
     if (kind == "split" && spec == "mod_i64")

@@ -28,7 +28,7 @@ struct VarKV {
     std::string path;
     int fd = -1;
     struct Loc {
-        uint64_t off;
+        uint64_t off;   // offset of the VALUE bytes (header skipped)
         uint32_t len;
     };
     std::unordered_map<uint64_t, Loc> index;
@@ -36,25 +36,52 @@ struct VarKV {
     uint64_t live = 0;      // live bytes
     std::mutex mu;
 
-    explicit VarKV(const std::string& p) : path(p) { open_log(); }
+    // Records are self-describing ([u64 key][u32 len][bytes]) so a log can
+    // be REOPENED and its index rebuilt — checkpoint/resume the reference
+    // entirely lacks (DBs are destroyed on teardown).
+    explicit VarKV(const std::string& p, bool fresh = true) : path(p) {
+        open_log(fresh);
+    }
     ~VarKV() {
         if (fd >= 0) close(fd);
     }
-    void open_log() {
-        fd = ::open(path.c_str(), O_RDWR | O_CREAT | O_TRUNC, 0644);
+    void open_log(bool fresh) {
+        fd = ::open(path.c_str(), O_RDWR | O_CREAT | (fresh ? O_TRUNC : 0), 0644);
         if (fd < 0) throw std::runtime_error("VarKV: cannot open " + path);
         tail = live = 0;
         index.clear();
+        if (!fresh) replay();
+    }
+    void replay() {
+        char hdr[12];
+        uint64_t off = 0;
+        for (;;) {
+            if (pread(fd, hdr, 12, (off_t)off) != 12) break;
+            uint64_t key;
+            uint32_t len;
+            memcpy(&key, hdr, 8);
+            memcpy(&len, hdr + 8, 4);
+            auto it = index.find(key);
+            if (it != index.end()) live -= it->second.len;
+            index[key] = {off + 12, len};
+            live += len;
+            off += 12 + len;
+        }
+        tail = off;
     }
 
     void put(uint64_t key, const void* data, uint32_t len) {
         std::lock_guard<std::mutex> g(mu);
         auto it = index.find(key);
         if (it != index.end()) live -= it->second.len;
-        if (pwrite(fd, data, len, (off_t)tail) != (ssize_t)len)
+        char hdr[12];
+        memcpy(hdr, &key, 8);
+        memcpy(hdr + 8, &len, 4);
+        if (pwrite(fd, hdr, 12, (off_t)tail) != 12 ||
+            pwrite(fd, data, len, (off_t)(tail + 12)) != (ssize_t)len)
             throw std::runtime_error("VarKV: write failed");
-        index[key] = {tail, len};
-        tail += len;
+        index[key] = {tail + 12, len};
+        tail += 12 + len;
         live += len;
         if (tail > (64u << 20) && live * 2 < tail) compact();
     }
@@ -78,26 +105,35 @@ struct VarKV {
     }
     size_t size() const { return index.size(); }
 
-    void compact() {  // called under mu
+    void compact() {  // called under mu; keeps the self-describing format
         std::string tmp = path + ".compact";
         int nfd = ::open(tmp.c_str(), O_RDWR | O_CREAT | O_TRUNC, 0644);
         if (nfd < 0) throw std::runtime_error("VarKV: compact open failed");
         uint64_t ntail = 0;
         std::string buf;
+        uint64_t nlive = 0;
         for (auto& [k, loc] : index) {
             buf.resize(loc.len);
             if (pread(fd, buf.data(), loc.len, (off_t)loc.off) != (ssize_t)loc.len)
                 throw std::runtime_error("VarKV: compact read failed");
-            if (pwrite(nfd, buf.data(), loc.len, (off_t)ntail) != (ssize_t)loc.len)
+            char hdr[12];
+            uint64_t kk = k;
+            uint32_t ln = loc.len;
+            memcpy(hdr, &kk, 8);
+            memcpy(hdr + 8, &ln, 4);
+            if (pwrite(nfd, hdr, 12, (off_t)ntail) != 12 ||
+                pwrite(nfd, buf.data(), ln, (off_t)(ntail + 12)) != (ssize_t)ln)
                 throw std::runtime_error("VarKV: compact write failed");
-            index[k] = {ntail, loc.len};
-            ntail += loc.len;
+            index[k] = {ntail + 12, ln};
+            ntail += 12 + ln;
+            nlive += ln;
         }
         close(fd);
         if (rename(tmp.c_str(), path.c_str()) != 0)
             throw std::runtime_error("VarKV: compact rename failed");
         fd = nfd;
-        tail = live = ntail;
+        tail = ntail;
+        live = nlive;
     }
 };
 
@@ -166,10 +202,23 @@ struct LruCache {
 // ----- P_Reduce: persistent keyed running sum (reference p_reduce.hpp) -----
 struct PReduceLogic : OpLogic {
     int col;
-    VarKV kv;
-    LruCache cache;
-    PReduceLogic(int c, const std::string& path, int64_t cache_cap)
-        : col(c), kv(path), cache(&kv, (size_t)cache_cap) {}
+    std::string dir;
+    int op_id;
+    bool keep;
+    int64_t cache_cap;
+    std::unique_ptr<VarKV> kv;
+    std::unique_ptr<LruCache> cache;
+    PReduceLogic(int c, std::string dir_, int id, bool keep_, int64_t cc)
+        : col(c), dir(std::move(dir_)), op_id(id), keep(keep_), cache_cap(cc) {}
+
+    void warm(RuntimeCtx& ctx) override {
+        // deterministic per-replica log path -> withKeepState() resumes the
+        // accumulators of a previous run
+        std::string path = dir + ".op" + std::to_string(op_id) + ".r" +
+                           std::to_string(ctx.replica) + ".log";
+        kv = std::make_unique<VarKV>(path, /*fresh=*/!keep);
+        cache = std::make_unique<LruCache>(kv.get(), (size_t)cache_cap);
+    }
 
     void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
         Batch* o = out.new_batch();
@@ -177,10 +226,10 @@ struct PReduceLogic : OpLogic {
         for (int64_t i = 0; i < b->count; ++i) {
             uint64_t key = b->key[i];
             int64_t acc = 0;
-            std::string* v = cache.get(key);
+            std::string* v = cache->get(key);
             if (v && v->size() == 8) memcpy(&acc, v->data(), 8);
             acc += x[i];
-            cache.put(key, std::string((char*)&acc, 8));
+            cache->put(key, std::string((char*)&acc, 8));
             if (o->count == o->capacity) {
                 o->watermark = b->watermark;
                 out.emit(o);
@@ -198,7 +247,7 @@ struct PReduceLogic : OpLogic {
             release(o);
         release(b);
     }
-    void on_eos(EmitCtx&, RuntimeCtx&) override { cache.flush(); }
+    void on_eos(EmitCtx&, RuntimeCtx&) override { cache->flush(); }
 };
 
 std::shared_ptr<OpLogic> make_persist_logic(const std::string& kind,
@@ -207,27 +256,25 @@ std::shared_ptr<OpLogic> make_persist_logic(const std::string& kind,
                                             const std::vector<int64_t>& ip,
                                             Engine*, int op_id) {
     if (kind == "p_reduce") {
-        // spec = state dir; ip: [col, cache_capacity].  Every replica gets
-        // its own log (the factory runs once per replica).
-        static std::atomic<int> inst{0};
-        std::string path = spec.empty() ? "/tmp/wfa_preduce" : spec;
-        path += "." + std::to_string(op_id) + "." +
-                std::to_string(inst.fetch_add(1)) + ".log";
-        return std::make_shared<PReduceLogic>((int)(ip.empty() ? 0 : ip[0]), path,
-                                              ip.size() > 1 ? ip[1] : 1 << 16);
+        // spec = state dir; ip: [col, cache_capacity, keep]
+        std::string dir = spec.empty() ? "/tmp/wfa_preduce" : spec;
+        return std::make_shared<PReduceLogic>(
+            (int)(ip.empty() ? 0 : ip[0]), dir, op_id,
+            ip.size() > 2 && ip[2] != 0, ip.size() > 1 ? ip[1] : 1 << 16);
     }
     throw std::runtime_error("unknown persistent logic: " + kind);
 }
 
 // Python-facing store handle (P_Map / P_Filter user logic)
 std::shared_ptr<void> open_state_store(const std::string& path, int64_t cache_cap,
-                                       void** kv_out, void** cache_out) {
+                                       void** kv_out, void** cache_out,
+                                       bool fresh) {
     struct Holder {
         VarKV kv;
         LruCache cache;
-        Holder(const std::string& p, size_t c) : kv(p), cache(&kv, c) {}
+        Holder(const std::string& p, size_t c, bool f) : kv(p, f), cache(&kv, c) {}
     };
-    auto h = std::make_shared<Holder>(path, (size_t)cache_cap);
+    auto h = std::make_shared<Holder>(path, (size_t)cache_cap, fresh);
     *kv_out = &h->kv;
     *cache_out = &h->cache;
     return h;

@@ -44,6 +44,12 @@ def sum_sink(col=0):
     return NativeLogic("sink", "sum_i64", [], [col])
 
 
+def last_per_key_sink(col=0):
+    """Sink that sums the LAST value per key at EOS (read back with
+    g.sink_sum): exposes final per-key accumulators, e.g. after P_Reduce."""
+    return NativeLogic("sink", "last_per_key_i64", [], [col])
+
+
 def count_sink():
     return NativeLogic("sink", "count", [], [])
 

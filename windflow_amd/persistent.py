@@ -54,7 +54,8 @@ class P_Reduce_Builder(_PersistBuilder):
         op = super().build()
         path = op.extra['state_path'] or "/tmp/wfa_state"
         op.logic = NativeLogic("p_reduce", path, [],
-                               [op.logic.iparams[0], op.extra['cache_capacity']])
+                               [op.logic.iparams[0], op.extra['cache_capacity'],
+                                1 if op.extra['keep'] else 0])
         return op
 
 
