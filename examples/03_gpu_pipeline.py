@@ -11,7 +11,7 @@ from windflow_amd import native_gpu
 from windflow_amd.builders_gpu import (Source_GPU_Builder, Map_GPU_Builder,
                                        Ffat_Windows_GPU_Builder, Sink_GPU_Builder)
 
-N, B, KEYS = 40_000_000, 4_000_000, 8192
+N, B, KEYS = 400_000_000, 4_000_000, 8192
 src = (Source_GPU_Builder(native_gpu.gpu_source(N, KEYS, B, vdt=2))
        .withOutputSchema([2]).withOutputBatchSize(B).build())
 jm = (Map_GPU_Builder(native_gpu.gpu_jit_map("v * 0.5f + 1.0f", 0))
@@ -31,4 +31,6 @@ t0 = time.time()
 g.run()
 dt = time.time() - t0
 print(f"{N} tuples -> {g.sink_count(snk)} windows in {dt:.3f}s "
-      f"({N / dt / 1e9:.2f} B tuples/s)")
+      f"({N / dt / 1e9:.2f} B tuples/s, incl. ~0.3s one-time start: "
+      f"engine spin-up + hiprtc JIT; bench.py gates these out and measures "
+      f"the sustained rate)")
