@@ -106,13 +106,25 @@ struct PyMapLogic : OpLogic {  // in-place mutate
 
 struct PyTransformLogic : OpLogic {  // returns new column dict (map/flatmap)
     py::function fn;
+    int64_t last_wm = 0;
     explicit PyTransformLogic(py::function f) : fn(std::move(f)) {}
     void process(Batch* b, EmitCtx& out, RuntimeCtx& rctx) override {
         int64_t wm = rctx.current_wm;
+        last_wm = wm;
         py::gil_scoped_acquire gil;
         py::object r = fn(batch_views(b, &rctx));
         release(b);
         if (!r.is_none()) emit_pydict(r.cast<py::dict>(), out, wm);
+    }
+    void on_eos(EmitCtx& out, RuntimeCtx& ctx) override {
+        // stateful python transforms (e.g. store-backed windows) flush their
+        // pending results at stream end via fn.on_eos(replica) -> dict; the
+        // replica id is passed because the python callable is SHARED across
+        // replicas and must only flush the keys this replica owns
+        py::gil_scoped_acquire gil;
+        if (!py::hasattr(fn, "on_eos")) return;
+        py::object r = fn.attr("on_eos")(ctx.replica);
+        if (!r.is_none()) emit_pydict(r.cast<py::dict>(), out, last_wm);
     }
 };
 

@@ -130,3 +130,79 @@ def test_p_reduce_keep_state_resumes(tmp_path):
         sums.append(g.sink_sum(snk))
     # run 2 ends with every key's accumulator at exactly 2x run 1's
     assert sums[1] == 2 * sums[0] and sums[0] > 0
+
+
+def test_p_flatmap_store(tmp_path):
+    """Persistent flatmap: emit one row per NEW key only (store as seen-set)."""
+    n = 4000
+    def fn(cols, store):
+        keys = cols['key']
+        new = []
+        for i in range(len(keys)):
+            k = int(keys[i])
+            if store.get(k) is None:
+                store.put(k, b"\x01")
+                new.append(i)
+        if not new:
+            return None
+        idx = np.array(new)
+        return {"ts": cols['ts'][idx], "key": keys[idx], "c0": cols['c0'][idx]}
+
+    from windflow_amd.persistent import P_FlatMap_Builder
+    g = wf.PipeGraph("pfm")
+    src = (wf.Source_Builder(native.seq_source(n, 97, 256))
+           .withParallelism(1).withOutputSchema([0]).build())
+    mp = g.add_source(src)
+    mp.add(P_FlatMap_Builder(fn).withStatePath(str(tmp_path / "fm"))
+           .withParallelism(1).withOutputSchema([0]).build())
+    snk = wf.Sink_Builder(native.count_sink()).withParallelism(1).build()
+    mp.add_sink(snk)
+    g.run()
+    assert g.sink_count(snk) == 97  # one row per distinct key
+
+
+def _run_pkw(tmp_path, tb, win, slide, n, keys, par):
+    from windflow_amd.persistent import P_Keyed_Windows_Builder
+    res = []
+
+    def sink_fn(cols):
+        for i in range(len(cols['key'])):
+            res.append((int(cols['key'][i]), float(cols['c0'][i])))
+
+    g = wf.PipeGraph("pkw")
+    src = (wf.Source_Builder(native.seq_source(n, keys, 256))
+           .withParallelism(1).withOutputSchema([0]).build())
+    mp = g.add_source(src)
+    b = (P_Keyed_Windows_Builder(np.sum).withStatePath(str(tmp_path / "kw"))
+         .withParallelism(par).withOutputSchema([1]))  # f64 out
+    b = b.withTBWindows(win, slide) if tb else b.withCBWindows(win, slide)
+    mp.add(b.build())
+    mp.add_sink(wf.Sink_Builder(sink_fn).withParallelism(1).build())
+    g.run()
+    return res
+
+
+def test_p_keyed_windows_cb(tmp_path):
+    from collections import Counter
+    from tests.test_windows import oracle_cb
+    n, keys, win, slide = 5000, 17, 40, 12
+    res = _run_pkw(tmp_path, False, win, slide, n, keys, par=2)
+    per = {}
+    for v in range(1, n + 1):
+        per.setdefault(v % keys, []).append((v, v))
+    exp = oracle_cb(per, win, slide, "sum")
+    got = Counter((k, float(v)) for k, v in res)
+    assert got == Counter({(k, float(v)): c for (k, v), c in exp.items()})
+
+
+def test_p_keyed_windows_tb(tmp_path):
+    from collections import Counter
+    from tests.test_windows import oracle_tb
+    n, keys, win, slide = 4000, 11, 500, 150
+    res = _run_pkw(tmp_path, True, win, slide, n, keys, par=2)
+    per = {}
+    for v in range(1, n + 1):
+        per.setdefault(v % keys, []).append((v, v))
+    exp = oracle_tb(per, win, slide, "sum")
+    got = Counter((k, float(v)) for k, v in res)
+    assert got == Counter({(k, float(v)): c for (k, v), c in exp.items()})

@@ -110,3 +110,173 @@ class P_Sink_Builder(_PersistBuilder):
         op.logic = wrapped
         op.closing = wrapped.store.flush
         return op
+
+
+class P_FlatMap_Builder(_PersistBuilder):
+    """Python persistent flatmap: fn(cols, store) -> dict of output columns
+    (any length) or None (reference p_flatmap.hpp)."""
+    _kind = "flatmap"
+
+    def build(self):
+        op = super().build()
+        path = (op.extra['state_path'] or "/tmp/wfa_state") + f".pfm.{id(op)}.log"
+        wrapped = _PyStateful(op.logic, path, op.extra['cache_capacity'])
+        op.logic = wrapped
+        op.closing = wrapped.store.flush
+        return op
+
+
+class _PKeyedWin:
+    """Store-backed keyed windows (reference p_keyed_windows.hpp): each key's
+    pending tuples live as ONE serialized record in the persistent store
+    ([i64 ts | f64 val] pairs), so window state can exceed host memory and —
+    with keep — survive the graph.  CB fires on count, TB on watermark
+    passage; incomplete windows flush at EOS (same semantics as the in-memory
+    keyed windows, csrc/engine/windows.cpp)."""
+
+    def __init__(self, fn, path, cache_cap, wintype, win, slide):
+        import numpy as _np
+        self.np = _np
+        self.fn = fn
+        self.store = _core.StateStore(path, cache_cap)
+        self.tb = wintype == 1
+        self.win, self.slide = int(win), int(slide)
+        self.keys = {}              # replica -> keys with pending state
+        self.max_ts = {}
+        self.next_q = {}            # TB: first not-yet-fired window per key
+
+    def _load(self, k):
+        blob = self.store.get(k)
+        if not blob:
+            return (self.np.empty(0, self.np.int64),
+                    self.np.empty(0, self.np.float64))
+        a = self.np.frombuffer(blob, self.np.int64)
+        n = len(a) // 2
+        return a[:n].copy(), a[n:].view(self.np.float64).copy()
+
+    def _save(self, k, ts, vals):
+        self.store.put(k, self.np.concatenate(
+            [ts, vals.view(self.np.int64)]).tobytes())
+
+    def _fire_cb(self, k, ts, vals, out, at_eos=False):
+        W, S = self.win, self.slide
+        while len(vals) >= W or (at_eos and len(vals) > 0):
+            w = vals[:W]
+            out.append((ts[min(len(ts), W) - 1], k, float(self.fn(w))))
+            ts, vals = ts[S:], vals[S:]
+            if at_eos and len(vals) == 0:
+                break
+        return ts, vals
+
+    @staticmethod
+    def _first_q(t, W, S):
+        return max(0, (int(t) - W) // S + 1) if int(t) >= W else 0
+
+    def _fire_tb(self, k, ts, vals, wm, out):
+        W, S = self.win, self.slide
+        # fire every absolute-grid window [q*S, q*S+W) with end <= wm, in
+        # order; q advances monotonically (never re-derived backwards) and
+        # jumps over data-free gaps
+        if not len(ts):
+            return ts, vals
+        q = max(self._first_q(ts[0], W, S), self.next_q.get(k, 0))
+        while len(ts):
+            end = q * S + W
+            if end > wm:
+                break
+            m = (ts >= q * S) & (ts < end)
+            # empty in-range windows fire with 0 (same as the in-memory
+            # window ops — oracle_tb in tests/test_windows.py)
+            out.append((end - 1, k, float(self.fn(vals[m])) if m.any() else 0.0))
+            q += 1
+            keep = ts >= q * S
+            ts, vals = ts[keep], vals[keep]
+        self.next_q[k] = max(self.next_q.get(k, 0), q)
+        return ts, vals
+
+    def __call__(self, cols):
+        np = self.np
+        ts_in, key_in, v_in = cols['ts'], cols['key'], cols['c0']
+        wm = cols['watermark']
+        my_keys = self.keys.setdefault(cols.get('replica', 0), set())
+        out = []
+        order = np.argsort(key_in, kind='stable')
+        ks = key_in[order]
+        bounds = np.flatnonzero(np.r_[True, ks[1:] != ks[:-1], True])
+        for i in range(len(bounds) - 1):
+            sel = order[bounds[i]:bounds[i + 1]]
+            k = int(ks[bounds[i]])
+            ts, vals = self._load(k)
+            ts = np.concatenate([ts, ts_in[sel]])
+            vals = np.concatenate([vals, v_in[sel].astype(np.float64)])
+            if self.tb:
+                o = np.argsort(ts, kind='stable')
+                ts, vals = ts[o], vals[o]
+                # drop tuples entirely before the fired horizon (late in
+                # DEFAULT mode, reference window_replica.hpp lateness gate)
+                keep = ts >= self.next_q.get(k, 0) * self.slide
+                ts, vals = ts[keep], vals[keep]
+                if not len(ts):
+                    self._save(k, ts, vals)
+                    my_keys.add(k)
+                    continue
+                self.max_ts[k] = max(self.max_ts.get(k, 0), int(ts[-1]))
+                ts, vals = self._fire_tb(k, ts, vals, wm, out)
+            else:
+                ts, vals = self._fire_cb(k, ts, vals, out)
+            self._save(k, ts, vals)
+            my_keys.add(k)
+        return self._pack(out)
+
+    def _pack(self, out):
+        if not out:
+            return None
+        np = self.np
+        return {"ts": np.array([o[0] for o in out], np.int64),
+                "key": np.array([o[1] for o in out], np.uint64),
+                "c0": np.array([o[2] for o in out], np.float64)}
+
+    def on_eos(self, replica=0):
+        out = []
+        for k in sorted(self.keys.get(replica, ())):
+            ts, vals = self._load(k)
+            if not len(vals):
+                continue
+            if self.tb:
+                # every window containing data has end <= max_ts + win
+                ts, vals = self._fire_tb(k, ts, vals,
+                                         self.max_ts.get(k, 0) + self.win + 1,
+                                         out)
+            else:
+                ts, vals = self._fire_cb(k, ts, vals, out, at_eos=True)
+            self.store.put(k, b"")
+        self.store.flush()
+        return self._pack(out)
+
+
+class P_Keyed_Windows_Builder(_PersistBuilder):
+    """Keyed windows whose per-key archives live in the persistent store
+    (reference p_keyed_windows.hpp over DBHandle).  fn(values) -> scalar."""
+    _kind = "flatmap"
+
+    def __init__(self, func):
+        super().__init__(func)
+        self._op.window = dict(type=0, win=0, slide=0)
+
+    def withCBWindows(self, win, slide):
+        self._op.window = dict(type=0, win=int(win), slide=int(slide))
+        return self
+
+    def withTBWindows(self, win_us, slide_us):
+        self._op.window = dict(type=1, win=int(win_us), slide=int(slide_us))
+        return self
+
+    def build(self):
+        op = super().build()
+        w = op.window
+        op.window = None  # handled here, not by the graph's window lowering
+        path = (op.extra['state_path'] or "/tmp/wfa_state") + f".pkw.{id(op)}.log"
+        wrapped = _PKeyedWin(op.logic, path, op.extra['cache_capacity'],
+                             w['type'], w['win'], w['slide'])
+        op.logic = wrapped
+        return op
