@@ -775,6 +775,55 @@ struct GpuReduceLogic : GpuLogicBase {
     }
 };
 
+// ----- unkeyed full-batch reduce (reference reduce_gpu.hpp:269
+// thrust::reduce): one (value, max_ts) tuple per input batch -----
+struct GpuReduceAllLogic : GpuLogicBase {
+    int comb, vcol;
+    float* d_part = nullptr;
+    int64_t* d_part_ts = nullptr;
+    float* d_cast = nullptr;
+    int64_t cast_cap = 0;
+    GpuReduceAllLogic(int comb_, int vc, int dev, Schema os, int64_t cap) {
+        comb = comb_; vcol = vc;
+        device = dev;
+        out_schema = os;  // payload [F32]
+        out_cap = cap;
+    }
+    void init_device() override {
+        auto& A = arena(device);
+        d_part = (float*)A.get(4 * 512);
+        d_part_ts = (int64_t*)A.get(8 * 512);
+    }
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        Batch* db = input_on_device(in, ctx);
+        int64_t n = db->count;
+        const void* v = db->cols[vcol];
+        int vdt = (int)db->schema.payload[vcol];
+        if (vdt != 2 && vdt != 5) {
+            if (cast_cap < n) {
+                d_cast = (float*)arena(device).get(4 * n);
+                cast_cap = n;
+            }
+            wfa_cast(stream, v, vdt, d_cast, 2, n);
+            v = d_cast;
+            vdt = 2;
+        }
+        Batch* ob = get_dev();
+        wfa_reduce_all(stream, v, vdt, db->ts, n, comb, d_part, d_part_ts,
+                       (float*)ob->cols[0], ob->ts);
+        HIPCHK(hipMemsetAsync(ob->key, 0, 8, stream));
+        *ob->lazy_count = 1;
+        ob->count = 1;
+        ob->watermark = db->watermark;
+        ob->born_us = db->born_us;
+        if (ctx.stats) ctx.stats->num_kernels += 2;
+        release_after_use(db);
+        record_ready(ob);
+        out.emit(ob);
+    }
+};
+
 // ===== Ffat_Windows_GPU: keyed CB/TB sliding window =====
 struct GpuFfatLogic : GpuLogicBase {
     int comb, vcol;
@@ -1460,6 +1509,9 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
         return std::make_shared<GpuStatefulFilterLogic>((int)ip[0], (int)ip[1], fp[0],
                                                         fp[1], ip[2], device, os,
                                                         out_batch);
+    if (kind == "gpu_reduce_all")
+        return std::make_shared<GpuReduceAllLogic>((int)ip[0], (int)ip[1], device,
+                                                   os, out_batch);
     if (kind == "gpu_reduce")
         // ip: [comb, vcol, max_keys]
         return std::make_shared<GpuReduceLogic>((int)ip[0], (int)ip[1], ip[2], device,

@@ -1446,3 +1446,74 @@ extern "C" void wfa_ffat_tree_fold(
                        st_fill, st_acc, tree, st_head, slot_to_key, fire_base,
                        out_key, out_val, out_ts, out_cap);
 }
+
+// ===== unkeyed full-batch reduce (reference reduce_gpu.hpp:269
+// thrust::reduce path): two deterministic stages, no atomics =====
+#define RA_BLOCKS 512
+
+__global__ void k_reduce_all_p1(const void* v, int vdt, const int64_t* ts,
+                                int64_t n, int comb, float* part,
+                                int64_t* part_ts) {
+    __shared__ float ls[WFA_THREADS];
+    __shared__ int64_t lt[WFA_THREADS];
+    float acc = (comb == 1) ? INFINITY : (comb == 2 ? -INFINITY : 0.0f);
+    int64_t tmax = INT64_MIN;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * (int64_t)blockDim.x) {
+        float x = wfa_val_at(v, vdt, i);
+        acc = (comb == 0) ? acc + x : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
+        if (ts) tmax = max(tmax, ts[i]);
+    }
+    ls[threadIdx.x] = acc;
+    lt[threadIdx.x] = tmax;
+    __syncthreads();
+    for (int off = WFA_THREADS / 2; off > 0; off >>= 1) {
+        if (threadIdx.x < off) {
+            float o = ls[threadIdx.x + off];
+            ls[threadIdx.x] = (comb == 0) ? ls[threadIdx.x] + o
+                              : (comb == 1 ? fminf(ls[threadIdx.x], o)
+                                           : fmaxf(ls[threadIdx.x], o));
+            lt[threadIdx.x] = max(lt[threadIdx.x], lt[threadIdx.x + off]);
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        part[blockIdx.x] = ls[0];
+        part_ts[blockIdx.x] = lt[0];
+    }
+}
+
+__global__ void k_reduce_all_p2(const float* part, const int64_t* part_ts,
+                                int64_t n_tuples, int comb, float* out,
+                                int64_t* out_ts) {
+    __shared__ float ls[RA_BLOCKS];
+    __shared__ int64_t lt[RA_BLOCKS];
+    ls[threadIdx.x] = part[threadIdx.x];
+    lt[threadIdx.x] = part_ts[threadIdx.x];
+    __syncthreads();
+    for (int off = RA_BLOCKS / 2; off > 0; off >>= 1) {
+        if (threadIdx.x < off) {
+            float o = ls[threadIdx.x + off];
+            ls[threadIdx.x] = (comb == 0) ? ls[threadIdx.x] + o
+                              : (comb == 1 ? fminf(ls[threadIdx.x], o)
+                                           : fmaxf(ls[threadIdx.x], o));
+            lt[threadIdx.x] = max(lt[threadIdx.x], lt[threadIdx.x + off]);
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        out[0] = (comb == 3) ? (float)n_tuples : ls[0];
+        out_ts[0] = lt[0];
+    }
+}
+
+extern "C" void wfa_reduce_all(wfa_stream_t s, const void* v, int vdt,
+                               const int64_t* ts, int64_t n, int comb,
+                               float* scratch /* >= RA_BLOCKS */,
+                               int64_t* scratch_ts /* >= RA_BLOCKS */,
+                               float* out, int64_t* out_ts) {
+    hipLaunchKernelGGL(k_reduce_all_p1, dim3(RA_BLOCKS), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, v, vdt, ts, n, comb, scratch, scratch_ts);
+    hipLaunchKernelGGL(k_reduce_all_p2, dim3(1), dim3(RA_BLOCKS), 0,
+                       (hipStream_t)s, scratch, scratch_ts, n, comb, out, out_ts);
+}

@@ -201,4 +201,10 @@ void wfa_gather_rows(wfa_stream_t s, const uint32_t* idx, int64_t n,
                      uint64_t* key_out, const void* const* cols_in,
                      void* const* cols_out, const int* esize, int nc);
 
+// unkeyed full-batch reduce: one (value, max_ts) per batch, deterministic
+// two-stage tree (no atomics); scratch arrays hold >= 512 entries
+void wfa_reduce_all(wfa_stream_t s, const void* v, int vdt, const int64_t* ts,
+                    int64_t n, int comb, float* scratch, int64_t* scratch_ts,
+                    float* out, int64_t* out_ts);
+
 }  // extern "C"

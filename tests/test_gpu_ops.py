@@ -478,6 +478,35 @@ def test_gpu_reduce_keyed_sum():
     assert abs(res['s'] - ref) <= 2e-3 * max(1.0, abs(ref))
 
 
+def test_gpu_reduce_all():
+    """Unkeyed full-batch reduce vs numpy: one tuple per batch (sum and max)."""
+    n, b = 1_000_000, 250_000
+    for comb, oracle in ((native_gpu.COMB_SUM, lambda v: v.astype(np.float64).sum()),
+                         (native_gpu.COMB_MAX, lambda v: float(v.max()))):
+        src = (Source_GPU_Builder(native_gpu.gpu_source(n, 64, b, vdt=2))
+               .withOutputSchema([2]).withOutputBatchSize(b).build())
+        rd = (Reduce_GPU_Builder(native_gpu.gpu_reduce_all(comb, 0))
+              .withOutputSchema([2]).withOutputBatchSize(1024).build())
+        got = []
+
+        def pysink(cols):
+            got.extend(float(x) for x in cols['c0'])
+
+        g = wf.PipeGraph("redall")
+        p = g.add_source(src)
+        p.chain(rd)
+        snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+        snk.out_schema = [2]
+        p.add_sink(snk)
+        g.run()
+        assert len(got) == n // b  # exactly one result per batch, in order
+        ts, key, val = gen_batch(n, 0, 42, 64, 2)
+        for i in range(n // b):
+            ref = oracle(val[i * b:(i + 1) * b])
+            tol = 2e-3 * max(1.0, abs(ref)) if comb == native_gpu.COMB_SUM else 1e-6
+            assert abs(got[i] - ref) <= tol
+
+
 def test_gpu_mfma_gram_windows():
     """MFMA Gram windows vs numpy einsum: per-key tumbling windows of 32
     16-dim vectors, window aggregate sum(v v^T) on the matrix cores."""

@@ -76,6 +76,13 @@ def gpu_keyed_reduce(comb=COMB_SUM, col=0, max_keys=1 << 16):
     return NativeLogic("gpu_reduce", "", [], [comb, col, max_keys])
 
 
+def gpu_reduce_all(comb=COMB_SUM, col=0):
+    """per-batch UNKEYED full reduction -> one (value, ts_max) tuple per
+    batch (reference reduce_gpu.hpp:269 thrust::reduce path); deterministic
+    two-stage tree, no sort and no atomics."""
+    return NativeLogic("gpu_reduce_all", "", [], [comb, col])
+
+
 def gpu_ffat_windows(comb=COMB_SUM, col=0, win=1000, slide=100,
                      max_keys=1 << 16, use_tree=False, tb=False, lateness=0,
                      pend_ring_log2=0):
