@@ -1538,6 +1538,96 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
     throw std::runtime_error("unknown gpu logic: " + kind);
 }
 
+// Per-stage hipEvent timing of the keyed FFAT CB chain (slot -> sort ->
+// segments -> fire offsets -> fold) on synthetic data.  The safe substitute
+// for rocprofv3 kernel stats on this pool (profiling runs hang boxes);
+// drives perf work from real measurements, not guesses.
+std::vector<std::pair<std::string, double>> debug_ffat_stage_times(
+    int64_t n, int64_t n_keys, int64_t win, int64_t slide, int iters) {
+    HIPCHK(hipSetDevice(0));
+    hipStream_t s = nullptr;
+    KeyedScratch ks;
+    ks.alloc(0, n, n_keys, s);
+    auto& A = arena(0);
+    uint64_t* d_key = (uint64_t*)A.get(8 * n);
+    int64_t* d_ts = (int64_t*)A.get(8 * n);
+    uint16_t* d_val = (uint16_t*)A.get(2 * n);
+    {
+        std::vector<uint64_t> hk(n);
+        std::vector<int64_t> ht(n);
+        std::vector<uint16_t> hv(n, 0x3f80);  // bf16 1.0
+        uint64_t x = 0x12345;
+        for (int64_t i = 0; i < n; ++i) {
+            x = x * 6364136223846793005ULL + 1442695040888963407ULL;
+            hk[i] = (x >> 33) % (uint64_t)n_keys;
+            ht[i] = i;
+        }
+        HIPCHK(hipMemcpy(d_key, hk.data(), 8 * n, hipMemcpyHostToDevice));
+        HIPCHK(hipMemcpy(d_ts, ht.data(), 8 * n, hipMemcpyHostToDevice));
+        HIPCHK(hipMemcpy(d_val, hv.data(), 2 * n, hipMemcpyHostToDevice));
+    }
+    int64_t pane = std::__gcd(win, slide), P = win / pane, S = slide / pane;
+    int ring_log2 = 1;
+    while ((1ll << ring_log2) < P + 2) ++ring_log2;
+    int64_t R = 1ll << ring_log2;
+    int64_t fires_cap = n / slide + n_keys + 64;
+    int64_t* st_count = (int64_t*)A.get(8 * n_keys);
+    uint32_t* st_fill = (uint32_t*)A.get(4 * n_keys);
+    float* st_acc = (float*)A.get(4 * n_keys);
+    float* ring = (float*)A.get(4 * n_keys * R);
+    uint32_t* st_head = (uint32_t*)A.get(4 * n_keys);
+    float* st_wsum = (float*)A.get(4 * n_keys);
+    uint32_t* nf = (uint32_t*)A.get(4 * (fires_cap + 1));
+    uint64_t* o_key = (uint64_t*)A.get(8 * fires_cap);
+    float* o_val = (float*)A.get(4 * fires_cap);
+    int64_t* o_ts = (int64_t*)A.get(8 * fires_cap);
+    int64_t* d_on = (int64_t*)A.get(64);
+    HIPCHK(hipMemsetAsync(st_count, 0, 8 * n_keys, s));
+    HIPCHK(hipMemsetAsync(st_fill, 0, 4 * n_keys, s));
+    HIPCHK(hipMemsetAsync(st_head, 0, 4 * n_keys, s));
+    HIPCHK(hipMemsetAsync(st_wsum, 0, 4 * n_keys, s));
+    wfa_fill_f32(s, st_acc, 0.f, n_keys);
+    wfa_fill_f32(s, ring, 0.f, n_keys * R);
+
+    constexpr int NS = 5;
+    const char* names[NS] = {"key_to_slot", "radix_sort", "segments",
+                             "fire_offsets", "fold"};
+    hipEvent_t ev[NS + 1];
+    for (auto& e : ev) HIPCHK(hipEventCreate(&e));
+    double acc[NS] = {0};
+    for (int it = -2; it < iters; ++it) {  // 2 warmup rounds
+        HIPCHK(hipEventRecord(ev[0], s));
+        wfa_key_to_slot(s, d_key, n, ks.tab, ks.d_nslots, ks.table_cap, ks.slot,
+                        ks.slot_to_key);
+        HIPCHK(hipEventRecord(ev[1], s));
+        uint32_t *os_, *oi;
+        wfa_sort_pairs2(s, ks.slot, ks.idx, ks.slot_t, ks.idx_t, nullptr, nullptr,
+                        ks.hist, n, ks.bits, &os_, &oi, nullptr, 1);
+        HIPCHK(hipEventRecord(ev[2], s));
+        wfa_segments(s, os_, n, ks.hist, ks.seg_start, ks.seg_slot, ks.d_nseg);
+        HIPCHK(hipEventRecord(ev[3], s));
+        wfa_ffat_fire_offsets(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n, pane, P,
+                              S, st_fill, st_head, nf, d_on);
+        HIPCHK(hipEventRecord(ev[4], s));
+        wfa_ffat_cb_fold(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n, d_val, 5, oi,
+                         d_ts, pane, P, S, 0, ring_log2, st_count, st_fill, st_acc,
+                         ring, st_head, st_wsum, ks.slot_to_key, nf, o_key, o_val,
+                         o_ts, fires_cap);
+        HIPCHK(hipEventRecord(ev[5], s));
+        HIPCHK(hipStreamSynchronize(s));
+        if (it >= 0)
+            for (int k = 0; k < NS; ++k) {
+                float ms = 0;
+                HIPCHK(hipEventElapsedTime(&ms, ev[k], ev[k + 1]));
+                acc[k] += ms * 1000.0;
+            }
+    }
+    for (auto& e : ev) (void)hipEventDestroy(e);
+    std::vector<std::pair<std::string, double>> out;
+    for (int k = 0; k < NS; ++k) out.push_back({names[k], acc[k] / iters});
+    return out;
+}
+
 }  // namespace wfa
 
 #else  // !WFA_WITH_HIP
@@ -1552,6 +1642,11 @@ std::pair<std::vector<uint32_t>, std::vector<uint32_t>> debug_sort_pairs_host(
     throw std::runtime_error("built without HIP");
 }
 std::vector<uint32_t> debug_key_slots_host(const uint64_t*, int64_t, int64_t) {
+    throw std::runtime_error("built without HIP");
+}
+std::vector<std::pair<std::string, double>> debug_ffat_stage_times(int64_t, int64_t,
+                                                                   int64_t, int64_t,
+                                                                   int) {
     throw std::runtime_error("built without HIP");
 }
 std::shared_ptr<OpLogic> make_gpu_logic(const std::string&, const std::string&,

@@ -58,6 +58,9 @@ __global__ void k_key_to_slot(const uint64_t* key, int64_t n, uint64_t* tab,
          i += gridDim.x * (int64_t)blockDim.x) {
         uint64_t k = key[i];
         uint64_t p = mix64s(k) & mask;
+        // (A/B-measured: a plain 16 B vector-load fast path with atomic
+        // fallback changed nothing — the probe is line-fetch/latency bound,
+        // not atomic bound — so the simple atomic protocol stays.)
         for (;;) {
             uint64_t cur = __hip_atomic_load(&tab[2 * p], __ATOMIC_RELAXED,
                                              __HIP_MEMORY_SCOPE_AGENT);
