@@ -478,14 +478,16 @@ PYBIND11_MODULE(_core, m) {
               return py::make_tuple(ks, vs);
           });
     m.def("debug_ffat_stage_times",
-          [](int64_t n, int64_t n_keys, int64_t win, int64_t slide, int iters) {
+          [](int64_t n, int64_t n_keys, int64_t win, int64_t slide, int iters,
+             int vik) {
               py::dict d;
-              for (auto& [k, v] : debug_ffat_stage_times(n, n_keys, win, slide, iters))
+              for (auto& [k, v] :
+                   debug_ffat_stage_times(n, n_keys, win, slide, iters, vik))
                   d[py::str(k)] = v;
               return d;
           },
           py::arg("n"), py::arg("n_keys"), py::arg("win") = 1000,
-          py::arg("slide") = 100, py::arg("iters") = 20);
+          py::arg("slide") = 100, py::arg("iters") = 20, py::arg("vik") = 0);
     m.def("debug_key_slots",
           [](py::array_t<uint64_t> keys, int64_t max_keys) {
               auto r = debug_key_slots_host(keys.data(), keys.shape(0), max_keys);

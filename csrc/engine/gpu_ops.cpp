@@ -429,18 +429,41 @@ struct KeyedScratch {
     int v_dt = 2;                    // effective dtype of v_as_f32 (2/5)
 
     // sorts (slot, idx) pairs and fills segments; values stay unsorted and
-    // are read through idx_sorted (saves the gather round trip).
+    // are read through idx_sorted (saves the gather round trip) — EXCEPT in
+    // value-in-key (VIK) mode: for bf16 value columns with <= 65535 keys the
+    // bf16 bits ride in the low 16 bits of the sort key (radix passes sort
+    // on bits 16.. only, stability keeps per-key order), so the fold reads
+    // values SEQUENTIALLY from the sorted key array instead of gathering
+    // one cache line per tuple.  Numerics are identical (same bf16 bits).
     // vcol < 0: no value cast (stateful map/filter operate in place)
-    void group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx) {
+    void group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
+               bool want_vik = false) {
         int64_t n = db->count;
         if (n > cap)
             throw std::runtime_error("batch larger than keyed scratch capacity — "
                                      "set the GPU op's out_batch >= upstream batch");
+        bool vik = want_vik && vcol >= 0 && max_keys <= 65535 &&
+                   (int)db->schema.payload[vcol] == 5;
+        if (vik) {
+            wfa_key_to_slot_v(s, db->key, n, tab, d_nslots, table_cap, slot,
+                              slot_to_key, (const uint16_t*)db->cols[vcol]);
+            uint32_t *os, *oi;
+            wfa_sort_pairs2(s, slot, idx, slot_t, idx_t, nullptr, nullptr, hist,
+                            n, bits, &os, &oi, nullptr, /*implicit_iota=*/1,
+                            /*base_shift=*/16);
+            idx_sorted = oi;
+            v_as_f32 = os;
+            v_dt = 6;
+            wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg, /*shr=*/16);
+            if (ctx.stats) ctx.stats->num_kernels += 5 + 3 * ((bits + 3) / 4);
+            return;
+        }
         wfa_key_to_slot(s, db->key, n, tab, d_nslots, table_cap, slot,
                         slot_to_key);
         uint32_t *os, *oi;
         wfa_sort_pairs2(s, slot, idx, slot_t, idx_t, nullptr, nullptr, hist, n,
-                        bits, &os, &oi, nullptr, /*implicit_iota=*/1);
+                        bits, &os, &oi, nullptr, /*implicit_iota=*/1,
+                        /*base_shift=*/0);
         idx_sorted = oi;
         if (vcol >= 0) {
             // A/B-measured: reading values THROUGH idx_sorted in the folds
@@ -459,7 +482,7 @@ struct KeyedScratch {
                 v_dt = 2;
             }
         }
-        wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg);
+        wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg, /*shr=*/0);
         if (ctx.stats) ctx.stats->num_kernels += 5 + 3 * ((bits + 3) / 4);
     }
 };
@@ -920,17 +943,23 @@ struct GpuFfatLogic : GpuLogicBase {
     // hipGraph capture of the steady-state per-batch chain (slot -> sort ->
     // segments -> offsets -> fold -> count copy): the pool rotates a small
     // fixed set of in/out batches, so the (in, out, n) tuple recurs and one
-    // instantiated graph per tuple replays the whole ~25-kernel chain with
-    // a single launch.  Opt out with WFA_NO_HIPGRAPH=1.
+    // instantiated graph per tuple replays the whole chain with a single
+    // launch.  A/B on MI355X (4M-tuple batches, value-in-key chain): graphs
+    // 267 us/step vs direct launches 248 us/step — the VIK chain's kernels
+    // are few and large enough that per-launch gaps are cheaper than the
+    // graph launch itself, so graphs are OPT-IN via WFA_HIPGRAPH=1.
     std::map<std::tuple<const void*, const void*, int64_t>, hipGraphExec_t> graphs;
     static bool graphs_enabled() {
         static int v = -1;
-        if (v < 0) v = getenv("WFA_NO_HIPGRAPH") ? 0 : 1;
+        if (v < 0) {
+            const char* e = getenv("WFA_HIPGRAPH");
+            v = (e && e[0] == '1') ? 1 : (getenv("WFA_NO_HIPGRAPH") ? 0 : 0);
+        }
         return v;
     }
 
     void chain_cb(Batch* db, Batch* ob, int64_t n, RuntimeCtx& ctx) {
-        ks.group(stream, db, vcol, ctx);
+        ks.group(stream, db, vcol, ctx, /*want_vik=*/!use_tree);
         uint32_t* nf = cb_nf;
         wfa_ffat_fire_offsets(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                               pane_len, P, S, st_fill, st_head, nf, d_on);
@@ -1396,7 +1425,8 @@ struct GpuExchangeLogic : GpuLogicBase {
         wfa_count_u32(stream, dest, n, d_counts, world);
         uint32_t *od, *oi;
         wfa_sort_pairs2(stream, dest, idx, dest_t, idx_t, nullptr, nullptr, hist,
-                        n, bits, &od, &oi, nullptr, /*implicit_iota=*/1);
+                        n, bits, &od, &oi, nullptr, /*implicit_iota=*/1,
+                        /*base_shift=*/0);
         size_t nc = db->cols.size();
         std::vector<void*> ptrs(2 * nc);
         for (size_t c = 0; c < nc; ++c) {
@@ -1543,7 +1573,7 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
 // for rocprofv3 kernel stats on this pool (profiling runs hang boxes);
 // drives perf work from real measurements, not guesses.
 std::vector<std::pair<std::string, double>> debug_ffat_stage_times(
-    int64_t n, int64_t n_keys, int64_t win, int64_t slide, int iters) {
+    int64_t n, int64_t n_keys, int64_t win, int64_t slide, int iters, int vik) {
     HIPCHK(hipSetDevice(0));
     hipStream_t s = nullptr;
     KeyedScratch ks;
@@ -1597,22 +1627,28 @@ std::vector<std::pair<std::string, double>> debug_ffat_stage_times(
     double acc[NS] = {0};
     for (int it = -2; it < iters; ++it) {  // 2 warmup rounds
         HIPCHK(hipEventRecord(ev[0], s));
-        wfa_key_to_slot(s, d_key, n, ks.tab, ks.d_nslots, ks.table_cap, ks.slot,
-                        ks.slot_to_key);
+        if (vik)
+            wfa_key_to_slot_v(s, d_key, n, ks.tab, ks.d_nslots, ks.table_cap,
+                              ks.slot, ks.slot_to_key, d_val);
+        else
+            wfa_key_to_slot(s, d_key, n, ks.tab, ks.d_nslots, ks.table_cap, ks.slot,
+                            ks.slot_to_key);
         HIPCHK(hipEventRecord(ev[1], s));
         uint32_t *os_, *oi;
         wfa_sort_pairs2(s, ks.slot, ks.idx, ks.slot_t, ks.idx_t, nullptr, nullptr,
-                        ks.hist, n, ks.bits, &os_, &oi, nullptr, 1);
+                        ks.hist, n, ks.bits, &os_, &oi, nullptr, 1, vik ? 16 : 0);
         HIPCHK(hipEventRecord(ev[2], s));
-        wfa_segments(s, os_, n, ks.hist, ks.seg_start, ks.seg_slot, ks.d_nseg);
+        wfa_segments(s, os_, n, ks.hist, ks.seg_start, ks.seg_slot, ks.d_nseg,
+                     vik ? 16 : 0);
         HIPCHK(hipEventRecord(ev[3], s));
         wfa_ffat_fire_offsets(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n, pane, P,
                               S, st_fill, st_head, nf, d_on);
         HIPCHK(hipEventRecord(ev[4], s));
-        wfa_ffat_cb_fold(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n, d_val, 5, oi,
-                         d_ts, pane, P, S, 0, ring_log2, st_count, st_fill, st_acc,
-                         ring, st_head, st_wsum, ks.slot_to_key, nf, o_key, o_val,
-                         o_ts, fires_cap);
+        wfa_ffat_cb_fold(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
+                         vik ? (const void*)os_ : (const void*)d_val, vik ? 6 : 5,
+                         oi, d_ts, pane, P, S, 0, ring_log2, st_count, st_fill,
+                         st_acc, ring, st_head, st_wsum, ks.slot_to_key, nf, o_key,
+                         o_val, o_ts, fires_cap);
         HIPCHK(hipEventRecord(ev[5], s));
         HIPCHK(hipStreamSynchronize(s));
         if (it >= 0)
@@ -1646,7 +1682,7 @@ std::vector<uint32_t> debug_key_slots_host(const uint64_t*, int64_t, int64_t) {
 }
 std::vector<std::pair<std::string, double>> debug_ffat_stage_times(int64_t, int64_t,
                                                                    int64_t, int64_t,
-                                                                   int) {
+                                                                   int, int) {
     throw std::runtime_error("built without HIP");
 }
 std::shared_ptr<OpLogic> make_gpu_logic(const std::string&, const std::string&,

@@ -27,7 +27,15 @@ static inline int64_t nblk(int64_t n, int64_t per_thread = 1) {
 }
 
 __device__ __forceinline__ float wfa_val_at(const void* v, int vdt, int64_t i) {
-    // vdt 2 = f32, 5 = bf16 (u16); i64 inputs are pre-cast to f32 by group()
+    // vdt 2 = f32, 5 = bf16 (u16), 6 = bf16 packed in the low 16 bits of the
+    // SORTED u32 slot array (value-in-key: the fold reads values
+    // sequentially instead of gathering one cache line per tuple);
+    // i64 inputs are pre-cast to f32 by group()
+    if (vdt == 6) {
+        union { uint32_t u; float f; } c;
+        c.u = ((const uint32_t*)v)[i] << 16;
+        return c.f;
+    }
     if (vdt == 5) {
         uint32_t u = ((const uint16_t*)v)[i];
         union { uint32_t u; float f; } c;
@@ -91,6 +99,56 @@ __global__ void k_key_to_slot(const uint64_t* key, int64_t n, uint64_t* tab,
         } while (s == ~0ULL);
         slot_out[i] = (uint32_t)s;
     }
+}
+
+// value-in-key variant: slot_out[i] = slot << 16 | bf16(value).  Only used
+// when the value column is already bf16 (no precision change) and
+// max_keys <= 65535; the radix passes then sort on bits 16.. only.
+__global__ void k_key_to_slot_v(const uint64_t* key, int64_t n, uint64_t* tab,
+                                uint32_t* n_slots, int64_t cap,
+                                uint32_t* slot_out, uint64_t* slot_to_key,
+                                const uint16_t* val) {
+    const uint64_t mask = (uint64_t)cap - 1;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * (int64_t)blockDim.x) {
+        uint64_t k = key[i];
+        uint64_t p = mix64s(k) & mask;
+        for (;;) {
+            uint64_t cur = __hip_atomic_load(&tab[2 * p], __ATOMIC_RELAXED,
+                                             __HIP_MEMORY_SCOPE_AGENT);
+            if (cur == k) break;
+            if (cur == WFA_EMPTY_KEY) {
+                uint64_t expected = WFA_EMPTY_KEY;
+                bool won = __hip_atomic_compare_exchange_strong(
+                    &tab[2 * p], &expected, k, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+                    __HIP_MEMORY_SCOPE_AGENT);
+                if (won) {
+                    uint32_t slot = atomicAdd(n_slots, 1u);
+                    slot_to_key[slot] = k;
+                    __hip_atomic_store(&tab[2 * p + 1], (uint64_t)slot,
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    break;
+                }
+                if (expected == k) break;
+            }
+            p = (p + 1) & mask;
+        }
+        uint64_t sl;
+        do {
+            sl = __hip_atomic_load(&tab[2 * p + 1], __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+        } while (sl == ~0ULL);
+        slot_out[i] = ((uint32_t)sl << 16) | (uint32_t)val[i];
+    }
+}
+
+extern "C" void wfa_key_to_slot_v(wfa_stream_t s, const uint64_t* key, int64_t n,
+                                  uint64_t* table_packed, uint32_t* n_slots,
+                                  int64_t table_cap, uint32_t* slot_out,
+                                  uint64_t* slot_to_key, const uint16_t* val) {
+    hipLaunchKernelGGL(k_key_to_slot_v, dim3(nblk(n)), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, key, n, table_packed, n_slots, table_cap,
+                       slot_out, slot_to_key, val);
 }
 
 extern "C" void wfa_key_to_slot(wfa_stream_t s, const uint64_t* key, int64_t n,
@@ -493,7 +551,8 @@ extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                                 uint32_t* val2, uint32_t* val2_tmp,
                                 uint32_t* hist, int64_t n, int bits,
                                 uint32_t** out_slot, uint32_t** out_idx,
-                                uint32_t** out_val2, int implicit_iota) {
+                                uint32_t** out_val2, int implicit_iota,
+                                int base_shift) {
     hipStream_t st = (hipStream_t)s;
     uint32_t *ka = slot, *va = idx, *kb = slot_tmp, *vb = idx_tmp;
     uint32_t *wa = val2, *wb = val2_tmp;
@@ -508,7 +567,7 @@ extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
         uint32_t* dbase = dt + 256;
         int passes = (bits + 7) / 8;
         for (int p = 0; p < passes; ++p) {
-            int shift = 8 * p;
+            int shift = base_shift + 8 * p;
             hipLaunchKernelGGL(k_rs8_hist, dim3(nb), dim3(WFA_THREADS), 0, st, ka, n,
                                shift, hist, nb);
             hipLaunchKernelGGL(k_rs8_scan, dim3(256), dim3(256), 0, st, hist, nb, dt);
@@ -534,6 +593,7 @@ extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
     int64_t nblocks = wfa_sort_nblocks(n);
     uint32_t* dt = hist + 16 * nblocks;  // hist is sized 16*nblocks + 16
     int passes = (bits + 3) / 4;
+    (void)base_shift;  // 4-bit path is only used for tiny key spaces
     for (int p = 0; p < passes; ++p) {
         int shift = 4 * p;
         hipLaunchKernelGGL(k_rs_hist, dim3(nblocks), dim3(WFA_THREADS), 0, st, ka, n,
@@ -558,17 +618,18 @@ extern "C" void wfa_sort_pairs(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                                int64_t n, int bits, uint32_t** out_slot,
                                uint32_t** out_idx) {
     wfa_sort_pairs2(s, slot, idx, slot_tmp, idx_tmp, nullptr, nullptr, hist, n,
-                    bits, out_slot, out_idx, nullptr, 0);
+                    bits, out_slot, out_idx, nullptr, 0, 0);
 }
 
 // ===== segment extraction =====
-__global__ void k_seg_count(const uint32_t* slot, int64_t n, uint32_t* blk_cnt) {
+__global__ void k_seg_count(const uint32_t* slot, int64_t n, uint32_t* blk_cnt,
+                            int shr) {
     int64_t base = (int64_t)blockIdx.x * RS_PER_BLOCK + (int64_t)threadIdx.x * RS_IPT;
     uint32_t c = 0;
 #pragma unroll
     for (int j = 0; j < RS_IPT; ++j) {
         int64_t i = base + j;
-        if (i < n) c += (i == 0) || (slot[i] != slot[i - 1]);
+        if (i < n) c += (i == 0) || ((slot[i] >> shr) != (slot[i - 1] >> shr));
     }
     __shared__ uint32_t red[WFA_THREADS / 64];
     for (int off = 32; off; off >>= 1) c += __shfl_down(c, off, 64);
@@ -582,13 +643,13 @@ __global__ void k_seg_count(const uint32_t* slot, int64_t n, uint32_t* blk_cnt) 
 }
 
 __global__ void k_seg_scatter(const uint32_t* slot, int64_t n, const uint32_t* blk_off,
-                              uint32_t* seg_start, uint32_t* seg_slot) {
+                              uint32_t* seg_start, uint32_t* seg_slot, int shr) {
     int64_t base = (int64_t)blockIdx.x * RS_PER_BLOCK + (int64_t)threadIdx.x * RS_IPT;
     uint32_t c = 0;
 #pragma unroll
     for (int j = 0; j < RS_IPT; ++j) {
         int64_t i = base + j;
-        if (i < n) c += (i == 0) || (slot[i] != slot[i - 1]);
+        if (i < n) c += (i == 0) || ((slot[i] >> shr) != (slot[i - 1] >> shr));
     }
     __shared__ uint32_t tc[WFA_THREADS];
     tc[threadIdx.x] = c;
@@ -603,22 +664,23 @@ __global__ void k_seg_scatter(const uint32_t* slot, int64_t n, const uint32_t* b
     for (int j = 0; j < RS_IPT; ++j) {
         int64_t i = base + j;
         if (i >= n) continue;
-        if ((i == 0) || (slot[i] != slot[i - 1])) {
+        if ((i == 0) || ((slot[i] >> shr) != (slot[i - 1] >> shr))) {
             seg_start[w] = (uint32_t)i;
-            seg_slot[w] = slot[i];
+            seg_slot[w] = slot[i] >> shr;
             ++w;
         }
     }
 }
 
 __global__ void k_nseg_total(const uint32_t* slot_sorted, int64_t n,
-                             const uint32_t* scan_tmp, int64_t nb, int64_t* d_nseg) {
+                             const uint32_t* scan_tmp, int64_t nb, int64_t* d_nseg,
+                             int shr) {
     // one wave recounts the last block's boundaries in parallel
     if (blockIdx.x == 0 && threadIdx.x < 64) {
         int64_t base = (nb - 1) * (int64_t)RS_PER_BLOCK;
         uint32_t c = 0;
         for (int64_t i = base + threadIdx.x; i < n; i += 64)
-            c += (i == 0) || (slot_sorted[i] != slot_sorted[i - 1]);
+            c += (i == 0) || ((slot_sorted[i] >> shr) != (slot_sorted[i - 1] >> shr));
         for (int off = 32; off; off >>= 1) c += __shfl_down(c, off, 64);
         if (threadIdx.x == 0) *d_nseg = (int64_t)scan_tmp[nb - 1] + c;
     }
@@ -626,16 +688,16 @@ __global__ void k_nseg_total(const uint32_t* slot_sorted, int64_t n,
 
 extern "C" void wfa_segments(wfa_stream_t s, const uint32_t* slot_sorted, int64_t n,
                              uint32_t* scan_tmp, uint32_t* seg_start,
-                             uint32_t* seg_slot, int64_t* d_nseg) {
+                             uint32_t* seg_slot, int64_t* d_nseg, int shr) {
     hipStream_t st = (hipStream_t)s;
     int64_t nb = (n + RS_PER_BLOCK - 1) / RS_PER_BLOCK;
     hipLaunchKernelGGL(k_seg_count, dim3(nb), dim3(WFA_THREADS), 0, st, slot_sorted, n,
-                       scan_tmp);
+                       scan_tmp, shr);
     hipLaunchKernelGGL(k_rs_scan, dim3(1), dim3(1024), 0, st, scan_tmp, nb);
     hipLaunchKernelGGL(k_seg_scatter, dim3(nb), dim3(WFA_THREADS), 0, st, slot_sorted,
-                       n, scan_tmp, seg_start, seg_slot);
+                       n, scan_tmp, seg_start, seg_slot, shr);
     hipLaunchKernelGGL(k_nseg_total, dim3(1), dim3(64), 0, st, slot_sorted, n, scan_tmp,
-                       nb, d_nseg);
+                       nb, d_nseg, shr);
 }
 
 // ===== segmented reduce (Reduce_GPU per-batch semantics) =====
@@ -845,7 +907,7 @@ __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
         int64_t w = fire_base[j];
         float* rg = ring + (size_t)slot * R;
         for (; i < e; ++i) {
-            float x = wfa_val_at(v_f32, vdt, idx_sorted[i]);
+            float x = wfa_val_at(v_f32, vdt, vdt == 6 ? i : idx_sorted[i]);
             acc = (comb == 0) ? acc + x : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
             if (++fill == (uint32_t)pane_len) {
                 // pane complete
@@ -925,7 +987,8 @@ __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_sl
         for (int64_t pos = i0; pos < e; pos += 64) {
             const uint32_t nchunk = (uint32_t)min((int64_t)64, e - pos);
             float v = (lane < (int)nchunk)
-                          ? wfa_val_at(v_f32, vdt, idx_sorted[pos + lane])
+                          ? wfa_val_at(v_f32, vdt,
+                                       vdt == 6 ? pos + lane : idx_sorted[pos + lane])
                           : ident;
             uint32_t rel = (lane < (int)nchunk) ? (fill + (uint32_t)lane) / L : ~0u;
             const uint32_t maxrel = (fill + nchunk - 1) / L;
@@ -1385,7 +1448,7 @@ __global__ void k_ffat_tree(const uint32_t* seg_start, const uint32_t* seg_slot,
         int64_t w = fire_base[j];
         float* tr = tree + (size_t)slot * 2 * R;
         for (; i < e; ++i) {
-            float x = wfa_val_at(v_f32, vdt, idx_sorted[i]);
+            float x = wfa_val_at(v_f32, vdt, vdt == 6 ? i : idx_sorted[i]);
             acc = TCOMB(acc, x);
             if (++fill == (uint32_t)pane_len) {
                 uint32_t leaf = (head & Rm) + R;

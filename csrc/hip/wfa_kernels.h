@@ -39,6 +39,11 @@ void wfa_compact(wfa_stream_t s, int64_t n, const uint32_t* flags,
 // table_packed: u64[2*table_cap] of interleaved (key, slot) 16 B entries,
 // both halves init to ~0; n_slots: device counter of allocated slots.
 // slot_out[i] = dense id of key[i].
+// value-in-key variant: slot_out[i] = slot << 16 | bf16 value (val is u16)
+void wfa_key_to_slot_v(wfa_stream_t s, const uint64_t* key, int64_t n,
+                       uint64_t* table_packed, uint32_t* n_slots,
+                       int64_t table_cap, uint32_t* slot_out,
+                       uint64_t* slot_to_key, const uint16_t* val);
 void wfa_key_to_slot(wfa_stream_t s, const uint64_t* key, int64_t n,
                      uint64_t* table_packed, uint32_t* n_slots,
                      int64_t table_cap, uint32_t* slot_out,
@@ -58,7 +63,7 @@ void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                      uint32_t* val2, uint32_t* val2_tmp, uint32_t* hist,
                      int64_t n, int bits, uint32_t** out_slot,
                      uint32_t** out_idx, uint32_t** out_val2,
-                     int implicit_iota);
+                     int implicit_iota, int base_shift);
 int64_t wfa_sort_nblocks(int64_t n);
 int64_t wfa_sort_hist_u32(int64_t cap);  // hist scratch size in u32
 
@@ -71,7 +76,7 @@ void wfa_gather(wfa_stream_t s, const uint32_t* idx, int64_t n,
 // segment j, seg_slot[j] = its slot; d_nseg = #segments.
 void wfa_segments(wfa_stream_t s, const uint32_t* slot_sorted, int64_t n,
                   uint32_t* scan_tmp, uint32_t* seg_start, uint32_t* seg_slot,
-                  int64_t* d_nseg);
+                  int64_t* d_nseg, int shr);
 
 // ----- per-batch keyed reduction (Reduce_GPU semantics) -----
 // comb: 0 sum 1 min 2 max 3 count ; vdt: dtype of v (F32 or I64 accum f64/i64)
