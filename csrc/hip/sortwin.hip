@@ -835,54 +835,6 @@ __global__ void k_fire_count(const uint32_t* seg_start, const uint32_t* seg_slot
     }
 }
 
-// fused count+scan in ONE launch (single block).  For <= 64K segments the
-// two-kernel version is pure launch/latency overhead (measured 27.6 us for
-// ~16K segments); this runs both phases in ~half that.  The parallel
-// two-kernel path remains for huge segment counts (e.g. 1M distinct keys).
-__global__ void k_fire_offsets_fused(const uint32_t* seg_start,
-                                     const uint32_t* seg_slot,
-                                     const int64_t* d_nseg, int64_t n,
-                                     int64_t pane_len, int64_t P, int64_t S,
-                                     const uint32_t* st_fill,
-                                     const uint32_t* st_head, uint32_t* nf,
-                                     int64_t* d_out_n) {
-    const int64_t m = *d_nseg;
-    __shared__ uint32_t tot[1024];
-    const int64_t chunk = (m + 1023) / 1024;
-    const int64_t b0 = (int64_t)threadIdx.x * chunk;
-    const int64_t b1 = min(m, b0 + chunk);
-    auto F = [&](uint64_t x) {
-        return x < (uint64_t)P ? 0ull : (x - (uint64_t)P) / (uint64_t)S + 1ull;
-    };
-    uint32_t s = 0;
-    for (int64_t j = b0; j < b1; ++j) {
-        const uint32_t slot = seg_slot[j];
-        const int64_t b = seg_start[j];
-        const int64_t e = (j + 1 < m) ? seg_start[j + 1] : n;
-        const uint64_t fill = st_fill[slot];
-        const uint64_t head0 = st_head[slot];
-        const uint64_t ncomp = (fill + (uint64_t)(e - b)) / (uint64_t)pane_len;
-        const uint32_t c = (uint32_t)(F(head0 + ncomp) - F(head0));
-        nf[j] = c;
-        s += c;
-    }
-    tot[threadIdx.x] = s;
-    __syncthreads();
-    for (int off = 1; off < 1024; off <<= 1) {
-        uint32_t t = (threadIdx.x >= off) ? tot[threadIdx.x - off] : 0;
-        __syncthreads();
-        tot[threadIdx.x] += t;
-        __syncthreads();
-    }
-    uint32_t run = tot[threadIdx.x] - s;
-    for (int64_t j = b0; j < b1; ++j) {
-        uint32_t v = nf[j];
-        nf[j] = run;
-        run += v;
-    }
-    if (threadIdx.x == 1023 && d_out_n) *d_out_n = tot[1023];
-}
-
 // exclusive scan of nf[0..*d_nseg) + total -> *d_out_n (single block)
 __global__ void k_fire_scan(uint32_t* nf, const int64_t* d_nseg, int64_t* d_out_n) {
     const int64_t m = *d_nseg;
@@ -914,13 +866,7 @@ extern "C" void wfa_ffat_fire_offsets(wfa_stream_t s, const uint32_t* seg_start,
                                       int64_t n, int64_t pane_len, int64_t P,
                                       int64_t S, const uint32_t* st_fill,
                                       const uint32_t* st_head, uint32_t* nf,
-                                      int64_t* d_out_n, int fused) {
-    if (fused) {
-        hipLaunchKernelGGL(k_fire_offsets_fused, dim3(1), dim3(1024), 0,
-                           (hipStream_t)s, seg_start, seg_slot, d_nseg, n, pane_len,
-                           P, S, st_fill, st_head, nf, d_out_n);
-        return;
-    }
+                                      int64_t* d_out_n) {
     hipStream_t st = (hipStream_t)s;
     hipLaunchKernelGGL(k_fire_count, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
                        st, seg_start, seg_slot, d_nseg, n, pane_len, P, S, st_fill,

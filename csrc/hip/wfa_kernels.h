@@ -102,7 +102,7 @@ void wfa_ffat_fire_offsets(wfa_stream_t s, const uint32_t* seg_start,
                            const uint32_t* seg_slot, const int64_t* d_nseg,
                            int64_t n, int64_t pane_len, int64_t P, int64_t S,
                            const uint32_t* st_fill, const uint32_t* st_head,
-                           uint32_t* nf, int64_t* d_out_n, int fused);
+                           uint32_t* nf, int64_t* d_out_n);
 
 // ----- FFAT/pane sliding-window state machine (CB) -----
 // Batched multi-key redesign of the reference's per-key FlatFAT_GPU
