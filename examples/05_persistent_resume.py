@@ -16,6 +16,10 @@ from windflow_amd import native
 from windflow_amd.persistent import P_Reduce_Builder
 
 state_dir = os.path.join(tempfile.gettempdir(), "wfa_resume_demo")
+# start clean so re-running the demo is deterministic
+import glob
+for f in glob.glob(state_dir + "*"):
+    os.unlink(f)
 
 def run_once(label):
     g = wf.PipeGraph("resume_demo")
