@@ -193,6 +193,12 @@ struct WindowCore : OpLogic {
     std::priority_queue<HeapEl, std::vector<HeapEl>, std::greater<HeapEl>> fire_heap;
     OutBuf ob;
     int64_t cur_wm = 0;
+    // wm stamped on INTERMEDIATE output batches of one processing round:
+    // the pre-round watermark.  Stamping the updated cur_wm on a batch
+    // whose successors (same round, OutBuf overflow) still carry ts <= wm
+    // breaks the watermark contract downstream (found by the 300-config
+    // fuzz: paned TB windows fired before their late panes arrived).
+    int64_t emit_wm = 0;
     int64_t rr_ctr = 0;  // mr_map round-robin arrival counter
     int replica = 0, par = 1;
     int64_t ignored = 0;
@@ -218,7 +224,7 @@ struct WindowCore : OpLogic {
 
     void fire(Open& w, uint64_t key, int64_t res_ts, EmitCtx& out) {
         if (!w.owned) return;
-        int64_t i = ob.slot(out, cur_wm);
+        int64_t i = ob.slot(out, emit_wm);
         Batch* o = ob.b;
         o->ts[i] = res_ts;
         o->key[i] = key;
@@ -226,6 +232,9 @@ struct WindowCore : OpLogic {
         if (emit_meta) {
             o->col<int64_t>(0)[i] = w.gwid;
             write_val(o, 1, i, r, agg.use_int);
+            // tuple count of the slice: empty partials (cnt 0) must not
+            // poison MIN/MAX downstream and AVG needs real weights
+            o->col<int64_t>(2)[i] = w.acc.cnt;
         } else {
             write_val(o, 0, i, r, agg.use_int);
         }
@@ -347,6 +356,7 @@ struct WindowCore : OpLogic {
     }
 
     void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
+        emit_wm = cur_wm;
         const int64_t n = b->count;
         for (int64_t i = 0; i < n; ++i) {
             bool mine = !subset_rr || (rr_ctr % par) == (int64_t)replica;
@@ -365,6 +375,7 @@ struct WindowCore : OpLogic {
     }
 
     bool on_punct(int64_t wm, EmitCtx& out, RuntimeCtx&) override {
+        emit_wm = cur_wm;
         cur_wm = std::max(cur_wm, wm);
         if (wt == WinType::TB) fire_tb(out);
         ob.flush(out, cur_wm);
@@ -372,11 +383,16 @@ struct WindowCore : OpLogic {
     }
 
     void on_eos(EmitCtx& out, RuntimeCtx&) override {
+        emit_wm = cur_wm;
         // flush every remaining open window (reference window_replica
         // eosnotify, :356-408)
         for (auto& [key, ks] : keys) {
             for (auto& w : ks.open) {
-                if (wt == WinType::TB && w.acc.cnt == 0 && !emit_meta) continue;
+                // empty TB windows fire with the default result (0), same as
+                // the steady-state heap path and the reference's FIRED
+                // triggerer — skipping them here made EOS semantics depend
+                // on how far the watermark happened to advance (found by
+                // the 300-config fuzz campaign)
                 if (prefix) {
                     Acc a;
                     a.i = ks.cum_i - w.acc.i;
@@ -428,6 +444,12 @@ struct WlqLogic : OpLogic {
     std::priority_queue<HeapEl, std::vector<HeapEl>, std::greater<HeapEl>> fire_heap;
     OutBuf ob;
     int64_t cur_wm = 0;
+    // wm stamped on INTERMEDIATE output batches of one processing round:
+    // the pre-round watermark.  Stamping the updated cur_wm on a batch
+    // whose successors (same round, OutBuf overflow) still carry ts <= wm
+    // breaks the watermark contract downstream (found by the 300-config
+    // fuzz: paned TB windows fired before their late panes arrived).
+    int64_t emit_wm = 0;
 
     WlqLogic(WinType wt_, int64_t wp, int64_t sp, int64_t lat, AggCfg a,
              int64_t pl, int own)
@@ -443,7 +465,7 @@ struct WlqLogic : OpLogic {
     }
 
     void fire(const std::pair<uint64_t, int64_t>& wk, WinAcc& w, EmitCtx& out) {
-        int64_t i = ob.slot(out, cur_wm);
+        int64_t i = ob.slot(out, emit_wm);
         Batch* o = ob.b;
         o->key[i] = wk.first;
         o->ts[i] = wt == WinType::CB ? w.last_ts
@@ -452,6 +474,7 @@ struct WlqLogic : OpLogic {
     }
 
     void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
+        emit_wm = cur_wm;
         const int64_t n = b->count;
         for (int64_t i = 0; i < n; ++i) {
             uint64_t key = b->key[i];
@@ -460,8 +483,7 @@ struct WlqLogic : OpLogic {
             ValU v = read_val(b, 1, i);
             part.i = v.i;
             part.f = v.f;
-            part.cnt = 1;
-            if (agg.comb == C_COUNT) part.cnt = v.i;  // count partial carries count
+            part.cnt = b->col<int64_t>(2)[i];  // real slice count (0 = empty)
             int64_t w_lo = p - win_p + 1;
             w_lo = w_lo <= 0 ? 0 : (w_lo + slide_p - 1) / slide_p;
             int64_t w_hi = p / slide_p;
@@ -498,6 +520,7 @@ struct WlqLogic : OpLogic {
     }
 
     bool on_punct(int64_t wm, EmitCtx& out, RuntimeCtx&) override {
+        emit_wm = cur_wm;
         cur_wm = std::max(cur_wm, wm);
         if (wt == WinType::TB) fire_tb(out);
         ob.flush(out, cur_wm);
@@ -505,6 +528,7 @@ struct WlqLogic : OpLogic {
     }
 
     void on_eos(EmitCtx& out, RuntimeCtx&) override {
+        emit_wm = cur_wm;
         for (auto& [kk, w] : wins) fire(kk, const_cast<WinAcc&>(w), out);
         wins.clear();
         ob.flush(out, cur_wm);
@@ -532,6 +556,12 @@ struct WlqKeyedLogic : OpLogic {
     std::priority_queue<HeapEl, std::vector<HeapEl>, std::greater<HeapEl>> fire_heap;
     OutBuf ob;
     int64_t cur_wm = 0;
+    // wm stamped on INTERMEDIATE output batches of one processing round:
+    // the pre-round watermark.  Stamping the updated cur_wm on a batch
+    // whose successors (same round, OutBuf overflow) still carry ts <= wm
+    // breaks the watermark contract downstream (found by the 300-config
+    // fuzz: paned TB windows fired before their late panes arrived).
+    int64_t emit_wm = 0;
     int64_t ignored = 0;
 
     WlqKeyedLogic(WinType wt_, int64_t wp, int64_t sp, int64_t lat, AggCfg a,
@@ -549,7 +579,7 @@ struct WlqKeyedLogic : OpLogic {
             if (id >= wE) break;
             agg.merge(acc, part);  // ids < wS were dropped after earlier fires
         }
-        int64_t i = ob.slot(out, cur_wm);
+        int64_t i = ob.slot(out, emit_wm);
         Batch* o = ob.b;
         o->key[i] = key;
         o->ts[i] = wt == WinType::CB ? ks.last_ts : wE * pane_len - 1;
@@ -561,6 +591,7 @@ struct WlqKeyedLogic : OpLogic {
     }
 
     void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
+        emit_wm = cur_wm;
         const int64_t n = b->count;
         for (int64_t i = 0; i < n; ++i) {
             uint64_t key = b->key[i];
@@ -578,7 +609,7 @@ struct WlqKeyedLogic : OpLogic {
             ValU v = read_val(b, 1, i);
             part.i = v.i;
             part.f = v.f;
-            part.cnt = agg.comb == C_COUNT ? v.i : 1;
+            part.cnt = b->col<int64_t>(2)[i];  // real slice count (0 = empty)
             ks.buf.emplace_back(p, part);
             ks.last_id = std::max(ks.last_id, p);
             ks.last_ts = std::max(ks.last_ts, b->ts[i]);
@@ -615,6 +646,7 @@ struct WlqKeyedLogic : OpLogic {
     }
 
     bool on_punct(int64_t wm, EmitCtx& out, RuntimeCtx&) override {
+        emit_wm = cur_wm;
         cur_wm = std::max(cur_wm, wm);
         if (wt == WinType::TB) fire_tb(out);
         ob.flush(out, cur_wm);
@@ -622,6 +654,7 @@ struct WlqKeyedLogic : OpLogic {
     }
 
     void on_eos(EmitCtx& out, RuntimeCtx&) override {
+        emit_wm = cur_wm;
         for (auto& [key, ks] : keys)
             while (ks.next_w >= 0 && ks.next_w * slide_p <= ks.last_id &&
                    !ks.buf.empty())
@@ -642,11 +675,17 @@ struct MrReduceLogic : OpLogic {
     std::unordered_map<std::pair<uint64_t, int64_t>, WinAcc, KeyGwidHash> wins;
     OutBuf ob;
     int64_t cur_wm = 0;
+    // wm stamped on INTERMEDIATE output batches of one processing round:
+    // the pre-round watermark.  Stamping the updated cur_wm on a batch
+    // whose successors (same round, OutBuf overflow) still carry ts <= wm
+    // breaks the watermark contract downstream (found by the 300-config
+    // fuzz: paned TB windows fired before their late panes arrived).
+    int64_t emit_wm = 0;
 
     MrReduceLogic(int64_t n, AggCfg a) : n_partials(n), agg(a) {}
 
     void fire(const std::pair<uint64_t, int64_t>& kk, WinAcc& w, EmitCtx& out) {
-        int64_t i = ob.slot(out, cur_wm);
+        int64_t i = ob.slot(out, emit_wm);
         Batch* o = ob.b;
         o->key[i] = kk.first;
         o->ts[i] = w.last_ts;
@@ -654,6 +693,7 @@ struct MrReduceLogic : OpLogic {
     }
 
     void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
+        emit_wm = cur_wm;
         for (int64_t i = 0; i < b->count; ++i) {
             auto kk = std::make_pair(b->key[i], b->col<int64_t>(0)[i]);
             auto& w = wins[kk];
@@ -661,8 +701,7 @@ struct MrReduceLogic : OpLogic {
             ValU v = read_val(b, 1, i);
             part.i = v.i;
             part.f = v.f;
-            part.cnt = 1;
-            if (agg.comb == C_COUNT) part.cnt = v.i;
+            part.cnt = b->col<int64_t>(2)[i];  // real slice count (0 = empty)
             agg.merge(w.acc, part);
             w.got++;
             w.last_ts = std::max(w.last_ts, b->ts[i]);
@@ -677,6 +716,7 @@ struct MrReduceLogic : OpLogic {
     }
 
     void on_eos(EmitCtx& out, RuntimeCtx&) override {
+        emit_wm = cur_wm;
         for (auto& [kk, w] : wins) fire(kk, const_cast<WinAcc&>(w), out);
         wins.clear();
         ob.flush(out, cur_wm);
@@ -762,6 +802,12 @@ struct FfatCpuLogic : OpLogic {
     std::unordered_map<uint64_t, KeyState> keys;
     OutBuf ob;
     int64_t cur_wm = 0;
+    // wm stamped on INTERMEDIATE output batches of one processing round:
+    // the pre-round watermark.  Stamping the updated cur_wm on a batch
+    // whose successors (same round, OutBuf overflow) still carry ts <= wm
+    // breaks the watermark contract downstream (found by the 300-config
+    // fuzz: paned TB windows fired before their late panes arrived).
+    int64_t emit_wm = 0;
 
     FfatCpuLogic(WinType wt_, int64_t w, int64_t s, int64_t lat, AggCfg a, int c)
         : wt(wt_), win(w), slide(s), lateness(lat), agg(a), col(c) {
@@ -775,7 +821,7 @@ struct FfatCpuLogic : OpLogic {
     void fire(uint64_t key, KeyState& ks, int64_t res_ts, int64_t span,
               EmitCtx& out) {
         Acc r = ks.fat.query_last(span);
-        int64_t i = ob.slot(out, cur_wm);
+        int64_t i = ob.slot(out, emit_wm);
         Batch* o = ob.b;
         o->ts[i] = res_ts;
         o->key[i] = key;
@@ -783,6 +829,7 @@ struct FfatCpuLogic : OpLogic {
     }
 
     void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
+        emit_wm = cur_wm;
         for (int64_t i = 0; i < b->count; ++i) {
             auto& ks = keys[b->key[i]];
             if (!ks.init) {
@@ -845,6 +892,7 @@ struct FfatCpuLogic : OpLogic {
     }
 
     bool on_punct(int64_t wm, EmitCtx& out, RuntimeCtx&) override {
+        emit_wm = cur_wm;
         cur_wm = std::max(cur_wm, wm);
         if (wt == WinType::TB) complete_panes(out, false);
         ob.flush(out, cur_wm);
@@ -852,6 +900,7 @@ struct FfatCpuLogic : OpLogic {
     }
 
     void on_eos(EmitCtx& out, RuntimeCtx&) override {
+        emit_wm = cur_wm;
         if (wt == WinType::TB) {
             complete_panes(out, true);
             // flush partial pane-windows (every slide_p, shrinking span) —
@@ -900,6 +949,12 @@ struct IntervalJoinLogic : OpLogic {
     std::unordered_map<uint64_t, KeyState> keys;
     OutBuf ob;
     int64_t cur_wm = 0;
+    // wm stamped on INTERMEDIATE output batches of one processing round:
+    // the pre-round watermark.  Stamping the updated cur_wm on a batch
+    // whose successors (same round, OutBuf overflow) still carry ts <= wm
+    // breaks the watermark contract downstream (found by the 300-config
+    // fuzz: paned TB windows fired before their late panes arrived).
+    int64_t emit_wm = 0;
     int64_t store_ctr = 0;
     bool a_int = true, b_int = true;
 
@@ -918,7 +973,7 @@ struct IntervalJoinLogic : OpLogic {
     }
 
     void emit_pair(uint64_t key, const Entry& ea, const Entry& eb, EmitCtx& out) {
-        int64_t i = ob.slot(out, cur_wm);
+        int64_t i = ob.slot(out, emit_wm);
         Batch* o = ob.b;
         o->ts[i] = std::max(ea.ts, eb.ts);
         o->key[i] = key;
@@ -927,6 +982,7 @@ struct IntervalJoinLogic : OpLogic {
     }
 
     void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
+        emit_wm = cur_wm;
         int tag = ctx.current_tag;
         int vcol = tag == 1 ? colB : colA;
         bool is_int = b->schema.payload[vcol] == DType::I64 ||
@@ -999,13 +1055,15 @@ struct IntervalJoinLogic : OpLogic {
     }
 
     bool on_punct(int64_t wm, EmitCtx& out, RuntimeCtx&) override {
+        emit_wm = cur_wm;
         cur_wm = std::max(cur_wm, wm);
         purge();
         ob.flush(out, cur_wm);
         return false;
     }
 
-    void on_eos(EmitCtx& out, RuntimeCtx&) override { ob.flush(out, cur_wm); }
+    void on_eos(EmitCtx& out, RuntimeCtx&) override {
+        emit_wm = cur_wm; ob.flush(out, cur_wm); }
 };
 
 // ============== non-incremental (user-function) keyed windows ==============
@@ -1031,6 +1089,12 @@ struct PyWindowLogic : OpLogic {
     bool have_schema = false;
     OutBuf ob;
     int64_t cur_wm = 0;
+    // wm stamped on INTERMEDIATE output batches of one processing round:
+    // the pre-round watermark.  Stamping the updated cur_wm on a batch
+    // whose successors (same round, OutBuf overflow) still carry ts <= wm
+    // breaks the watermark contract downstream (found by the 300-config
+    // fuzz: paned TB windows fired before their late panes arrived).
+    int64_t emit_wm = 0;
 
     PyWindowLogic(WinType wt_, int64_t w, int64_t s, int64_t lat, WindowFn f)
         : wt(wt_), win(w), slide(s), lateness(lat), fn(std::move(f)) {}
@@ -1062,7 +1126,7 @@ struct PyWindowLogic : OpLogic {
         for (size_t c = 0; c < ks.cols.size(); ++c)
             wr.cols.push_back(ks.cols[c].data() + lo * dsize(in_schema.payload[c]));
         double r = fn(wr);
-        int64_t i = ob.slot(out, cur_wm);
+        int64_t i = ob.slot(out, emit_wm);
         Batch* o = ob.b;
         o->ts[i] = res_ts;
         o->key[i] = key;
@@ -1084,6 +1148,7 @@ struct PyWindowLogic : OpLogic {
     }
 
     void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
+        emit_wm = cur_wm;
         if (!have_schema) {
             in_schema = b->schema;
             have_schema = true;
@@ -1140,6 +1205,7 @@ struct PyWindowLogic : OpLogic {
     }
 
     bool on_punct(int64_t wm, EmitCtx& out, RuntimeCtx&) override {
+        emit_wm = cur_wm;
         cur_wm = std::max(cur_wm, wm);
         if (wt == WinType::TB) fire_tb(out, false);
         ob.flush(out, cur_wm);
@@ -1147,6 +1213,7 @@ struct PyWindowLogic : OpLogic {
     }
 
     void on_eos(EmitCtx& out, RuntimeCtx&) override {
+        emit_wm = cur_wm;
         if (wt == WinType::TB) {
             fire_tb(out, true);
         } else {

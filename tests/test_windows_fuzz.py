@@ -27,18 +27,22 @@ BUILDERS = {
 }
 
 
-@pytest.mark.parametrize("case", range(12))
+@pytest.mark.parametrize("case", range(18))
 def test_window_fuzz(case):
-    rng = random.Random(1000 + case)
+    # widened space (min/count aggs, slide=1, 64 keys, batch > stream): the
+    # original narrow draw missed three real bugs — MIN/MAX identity on
+    # empty partials, empty-TB-window skip at EOS, and intermediate-batch
+    # watermark stamping (all found by a 300-config campaign of this loop)
+    rng = random.Random(50000 + case * 17)
     kind = rng.choice(list(BUILDERS))
-    agg = rng.choice(["sum", "max"])
+    agg = rng.choice(["sum", "max", "min", "count"])
     wt = rng.choice(["cb", "tb"])
-    slide = rng.choice([5, 10, 25])
-    win = slide * rng.randint(1, 6)
-    n_keys = rng.choice([1, 3, 11])
-    batch = rng.choice([32, 128, 1000])
-    par = rng.randint(1, 3)
-    stream = rng.choice([1500, 3100])
+    slide = rng.choice([1, 3, 5, 10, 25, 60])
+    win = slide * rng.randint(1, 8)
+    n_keys = rng.choice([1, 2, 3, 11, 64])
+    batch = rng.choice([16, 32, 128, 1000, 4096])
+    par = rng.randint(1, 4)
+    stream = rng.choice([700, 1500, 3100, 7000])
     mode = rng.choice([wf.ExecutionMode.DEFAULT, wf.ExecutionMode.DETERMINISTIC])
     b = BUILDERS[kind]((agg, 0))
     b = (b.withCBWindows(win, slide) if wt == "cb" else b.withTBWindows(win, slide))

@@ -89,7 +89,7 @@ def lower_window_node(graph, e, node):
         ui = _use_int(out_schema, comb)
         plq_par = int(op.extra.get("plq_par", node.parallelism))
         wlq_par = int(op.extra.get("wlq_par", node.parallelism))
-        mid_schema = [0, out_schema[0]]  # (pane gwid, partial)
+        mid_schema = [0, out_schema[0], 0]  # (pane gwid, partial, slice count)
         # Keys are partitioned (KEYBY) into both stages: each replica owns
         # every pane/window of its keys, so a tuple is processed once, not
         # once per replica.  The reference's broadcast + gwid%n ownership
@@ -111,7 +111,7 @@ def lower_window_node(graph, e, node):
         ui = _use_int(out_schema, comb)
         map_par = int(op.extra.get("map_par", node.parallelism))
         red_par = int(op.extra.get("reduce_par", node.parallelism))
-        mid_schema = [0, out_schema[0]]  # (window gwid, partial)
+        mid_schema = [0, out_schema[0], 0]  # (window gwid, partial, slice count)
         mp = e.add_op((op.name or kind) + ".map", map_par, "win_mr_map",
                       iparams=[wt, win, slide, lat, comb, col, 0, ui],
                       out_schema=mid_schema, out_batch=op.out_batch)
