@@ -53,19 +53,19 @@ def test_window_fuzz(case):
     assert got_counter(rows) == exp, (kind, agg, wt, win, slide, n_keys, batch, par)
 
 
-@pytest.mark.parametrize("case", range(6))
+@pytest.mark.parametrize("case", range(8))
 def test_join_fuzz(case):
     """Randomized interval joins: bounds/keys/batch/parallelism/mode drawn
     per case, vs the brute-force pair oracle."""
     from collections import Counter
     from test_windows import join_graph, oracle_join
-    rng = random.Random(77 + case)
-    lower = -rng.randint(0, 8)
-    upper = rng.randint(0, 8)
-    keys = rng.choice([2, 5, 9])
-    n = rng.choice([800, 1500])
-    batch = rng.choice([16, 64, 300])
-    par = rng.randint(1, 3)
+    rng = random.Random(90000 + case * 13)
+    lower = -rng.randint(0, 30)
+    upper = rng.randint(0, 30)
+    keys = rng.choice([1, 2, 5, 9, 33])
+    n = rng.choice([300, 800, 1500, 4000])
+    batch = rng.choice([8, 16, 64, 300, 2048])
+    par = rng.randint(1, 4)
     kp = rng.random() < 0.5
     rows = join_graph(lambda b: (b.withKPMode() if kp else b.withDPMode())
                       .withParallelism(par),
