@@ -186,6 +186,72 @@ struct PySinkLogic : OpLogic {
     }
 };
 
+struct PySplitLogic : OpLogic {  // fn(cols) -> branch ids OR per-branch masks
+    py::function fn;
+    explicit PySplitLogic(py::function f) : fn(std::move(f)) {}
+    void process(Batch* b, EmitCtx& out, RuntimeCtx& rctx) override {
+        const size_t nb = out.n_branches();
+        const size_t np_ = b->schema.payload.size();
+        // per-branch keep masks; a 1-D int array routes each row to ONE
+        // branch, a sequence of bool masks routes rows to ANY subset
+        // (reference splitting_emitter.hpp: integral_t OR vector<integral_t>)
+        std::vector<std::vector<char>> keep(nb, std::vector<char>(b->count, 0));
+        {
+            py::gil_scoped_acquire gil;
+            py::object r = fn(batch_views(b, &rctx));
+            if (py::isinstance<py::array>(r)) {
+                auto ids = r.cast<py::array_t<int32_t>>();
+                for (int64_t i = 0; i < b->count; ++i) {
+                    int32_t d = ids.at(i);
+                    if (d >= 0 && (size_t)d < nb) keep[d][i] = 1;
+                }
+            } else {
+                size_t br = 0;
+                for (auto m : r.cast<py::sequence>()) {
+                    if (br >= nb) break;
+                    auto mask = m.cast<py::array_t<bool>>();
+                    for (int64_t i = 0; i < b->count; ++i)
+                        if (mask.at(i)) keep[br][i] = 1;
+                    ++br;
+                }
+            }
+        }
+        int64_t wm = rctx.current_wm;
+        for (size_t br = 0; br < nb; ++br) {
+            int64_t cnt = 0;
+            for (int64_t i = 0; i < b->count; ++i) cnt += keep[br][i];
+            if (!cnt) {
+                out.emitters[br]->punct(wm);  // keep idle branches alive
+                continue;
+            }
+            Batch* o = out.new_batch();
+            int64_t w = 0;
+            for (int64_t i = 0; i < b->count; ++i) {
+                if (!keep[br][i]) continue;
+                if (w == o->capacity) {
+                    o->count = w;
+                    o->watermark = wm;
+                    out.emit_to(br, o);
+                    o = out.new_batch();
+                    w = 0;
+                }
+                o->ts[w] = b->ts[i];
+                o->key[w] = b->key[i];
+                for (size_t cc = 0; cc < np_; ++cc) {
+                    size_t es = dsize(b->schema.payload[cc]);
+                    memcpy((char*)o->cols[cc] + w * es, (char*)b->cols[cc] + i * es,
+                           es);
+                }
+                ++w;
+            }
+            o->count = w;
+            o->watermark = wm;
+            out.emit_to(br, o);
+        }
+        release(b);
+    }
+};
+
 struct PySourceLogic : OpLogic {
     py::function fn;  // fn(replica, parallelism) -> dict|None
     explicit PySourceLogic(py::function f) : fn(std::move(f)) {}
@@ -293,6 +359,8 @@ static StageSpec make_stage(Engine& e, int id, const std::string& kind, const st
         auto fn = pyfn.cast<py::function>();
         if (kind == "map")
             st.factory = [fn] { return std::make_shared<PyMapLogic>(fn); };
+        else if (kind == "split")
+            st.factory = [fn] { return std::make_shared<PySplitLogic>(fn); };
         else if (kind == "transform" || kind == "flatmap")
             st.factory = [fn] { return std::make_shared<PyTransformLogic>(fn); };
         else if (kind == "filter")

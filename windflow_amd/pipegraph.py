@@ -202,7 +202,8 @@ class PipeGraph:
                     op.out_schema, op.out_batch, None)
         if callable(op.logic):
             k = {"flatmap": "flatmap", "map": "map", "filter": "filter",
-                 "source": "source", "sink": "sink", "transform": "transform"}.get(op.kind)
+                 "source": "source", "sink": "sink", "transform": "transform",
+                 "split": "split"}.get(op.kind)
             if k is None:
                 raise NotImplementedError(f"python logic for kind {op.kind}")
             return (k, "", [], [], op.out_schema, op.out_batch, op.logic)
