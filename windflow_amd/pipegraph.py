@@ -116,7 +116,10 @@ class MultiPipe:
 
     def split(self, split_logic, n_branches):
         """Split into n branches by user logic (reference pipegraph.hpp:265).
-        split_logic: native.split_mod(col) or callable(cols)->int32 branch ids."""
+        split_logic: native.split_mod(col) / native.split_rr(), or a
+        callable(cols) returning int32 branch ids (one branch per tuple) or
+        a sequence of per-branch bool masks (one tuple to ANY subset of
+        branches — reference splitting_emitter.hpp vector<integral_t>)."""
         op = Operator(kind="split", logic=split_logic, name="split",
                       parallelism=self.graph.nodes[self.tails[0][0]].parallelism,
                       n_branches=n_branches,
