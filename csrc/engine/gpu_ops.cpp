@@ -907,7 +907,7 @@ struct GpuFfatLogic : GpuLogicBase {
         HIPCHK(hipMemsetAsync(st_fill, 0, 4 * max_keys, stream));
         HIPCHK(hipMemsetAsync(st_head, 0, 4 * max_keys, stream));
         HIPCHK(hipMemsetAsync(st_wsum, 0, 4 * max_keys, stream));
-        float ident = comb == 0 ? 0.f : (comb == 1 ? INFINITY : -INFINITY);
+        float ident = comb == 1 ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
         wfa_fill_f32(stream, st_acc, ident, max_keys);
         wfa_fill_f32(stream, ring_or_tree, ident,
                      max_keys * (use_tree ? 2 * (1ll << ring_log2) : (1ll << ring_log2)));

@@ -894,7 +894,7 @@ __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
     const int64_t nseg = *d_nseg;
     const uint32_t R = 1u << ring_log2;
     const uint32_t Rm = R - 1;
-    const float ident = (comb == 0) ? 0.f : (comb == 1 ? INFINITY : -INFINITY);
+    const float ident = (comb == 1) ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
     for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nseg;
          j += gridDim.x * (int64_t)blockDim.x) {
         const uint32_t slot = seg_slot[j];
@@ -907,11 +907,13 @@ __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
         int64_t w = fire_base[j];
         float* rg = ring + (size_t)slot * R;
         for (; i < e; ++i) {
-            float x = wfa_val_at(v_f32, vdt, vdt == 6 ? i : idx_sorted[i]);
-            acc = (comb == 0) ? acc + x : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
+            float x = (comb == 3) ? 1.0f
+                                  : wfa_val_at(v_f32, vdt, vdt == 6 ? i : idx_sorted[i]);
+            acc = (comb == 1) ? fminf(acc, x)
+                              : (comb == 2 ? fmaxf(acc, x) : acc + x);
             if (++fill == (uint32_t)pane_len) {
                 // pane complete
-                if (comb == 0) {
+                if (comb == 0 || comb == 3) {
                     wsum += acc;
                     if (head >= (uint32_t)P) wsum -= rg[(head - (uint32_t)P) & Rm];
                 }
@@ -921,7 +923,7 @@ __global__ void k_ffat_cb(const uint32_t* seg_start, const uint32_t* seg_slot,
                 ++head;
                 if (head >= (uint32_t)P && ((head - (uint32_t)P) % (uint32_t)S) == 0) {
                     float res;
-                    if (comb == 0) {
+                    if (comb == 0 || comb == 3) {
                         res = wsum;
                     } else {
                         res = rg[(head - 1) & Rm];
@@ -968,12 +970,12 @@ __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_sl
     const int64_t nseg = *d_nseg;
     const uint32_t R = 1u << ring_log2;
     const uint32_t Rm = R - 1;
-    const float ident = (comb == 0) ? 0.f : (comb == 1 ? INFINITY : -INFINITY);
+    const float ident = (comb == 1) ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
     const int lane = threadIdx.x & 63;
     const int64_t wid = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
     const uint32_t L = (uint32_t)pane_len;
-#define WCOMB(a, b) ((comb == 0) ? (a) + (b) : (comb == 1 ? fminf(a, b) : fmaxf(a, b)))
+#define WCOMB(a, b) ((comb == 1) ? fminf(a, b) : (comb == 2 ? fmaxf(a, b) : (a) + (b)))
     for (int64_t j = wid; j < nseg; j += nw) {
         const uint32_t slot = seg_slot[j];
         const int64_t i0 = seg_start[j];
@@ -987,8 +989,11 @@ __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_sl
         for (int64_t pos = i0; pos < e; pos += 64) {
             const uint32_t nchunk = (uint32_t)min((int64_t)64, e - pos);
             float v = (lane < (int)nchunk)
-                          ? wfa_val_at(v_f32, vdt,
-                                       vdt == 6 ? pos + lane : idx_sorted[pos + lane])
+                          ? ((comb == 3)
+                                 ? 1.0f
+                                 : wfa_val_at(v_f32, vdt,
+                                              vdt == 6 ? pos + lane
+                                                       : idx_sorted[pos + lane]))
                           : ident;
             uint32_t rel = (lane < (int)nchunk) ? (fill + (uint32_t)lane) / L : ~0u;
             const uint32_t maxrel = (fill + nchunk - 1) / L;
@@ -999,7 +1004,7 @@ __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_sl
                 acc = WCOMB(acc, pv);
                 if (r < ncomplete) {  // pane boundary crossed inside this chunk
                     if (lane == 0) rg[head & Rm] = acc;
-                    if (comb == 0) {
+                    if (comb == 0 || comb == 3) {
                         wsum += acc;
                         if (head >= (uint32_t)P) wsum -= rg[(head - (uint32_t)P) & Rm];
                     }
@@ -1009,7 +1014,7 @@ __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_sl
                     if (head >= (uint32_t)P &&
                         ((head - (uint32_t)P) % (uint32_t)S) == 0) {
                         float res;
-                        if (comb == 0) {
+                        if (comb == 0 || comb == 3) {
                             res = wsum;
                         } else {
                             // lane-parallel recombine of the last P panes
@@ -1272,10 +1277,10 @@ __global__ void k_tb_lift(const uint32_t* seg_start, const uint32_t* seg_slot,
                 atomicAdd(overflow, 1u);
                 continue;
             }
-            const float x = wfa_val_at(v_f32, vdt, idx_sorted[i]);
+            const float x = (comb == 3) ? 1.0f : wfa_val_at(v_f32, vdt, idx_sorted[i]);
             float* cell = &pd[(uint64_t)p & Pm];
-            *cell = (comb == 0) ? *cell + x
-                                : (comb == 1 ? fminf(*cell, x) : fmaxf(*cell, x));
+            *cell = (comb == 1) ? fminf(*cell, x)
+                                : (comb == 2 ? fmaxf(*cell, x) : *cell + x);
             if (p > lastp) lastp = p;
         }
         pend_base[slot] = base;
@@ -1345,7 +1350,7 @@ __global__ void k_tb_advance(const uint32_t* n_slots, int64_t limit_pane,
     const uint32_t Rm = R - 1;
     const uint32_t Rp = 1u << pend_log2;
     const uint32_t Pm = Rp - 1;
-    const float ident = (comb == 0) ? 0.f : (comb == 1 ? INFINITY : -INFINITY);
+    const float ident = (comb == 1) ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
     for (int64_t s = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; s < ns;
          s += gridDim.x * (int64_t)blockDim.x) {
         int64_t base = pend_base[s];
@@ -1361,14 +1366,14 @@ __global__ void k_tb_advance(const uint32_t* n_slots, int64_t limit_pane,
             float pane = pd[(uint64_t)q & Pm];
             pd[(uint64_t)q & Pm] = ident;  // consumed
             rg[head & Rm] = pane;
-            if (comb == 0) {
+            if (comb == 0 || comb == 3) {
                 wsum += pane;
                 if (head >= (uint32_t)P) wsum -= rg[(head - (uint32_t)P) & Rm];
             }
             ++head;
             if (head >= (uint32_t)P && ((head - (uint32_t)P) % (uint32_t)S) == 0) {
                 float res;
-                if (comb == 0) {
+                if (comb == 0 || comb == 3) {
                     res = wsum;
                 } else {
                     res = rg[(head - 1) & Rm];
@@ -1435,8 +1440,8 @@ __global__ void k_ffat_tree(const uint32_t* seg_start, const uint32_t* seg_slot,
     const int64_t nseg = *d_nseg;
     const uint32_t R = 1u << ring_log2;
     const uint32_t Rm = R - 1;
-    const float ident = (comb == 0) ? 0.f : (comb == 1 ? INFINITY : -INFINITY);
-#define TCOMB(a, b) ((comb == 0) ? (a) + (b) : (comb == 1 ? fminf(a, b) : fmaxf(a, b)))
+    const float ident = (comb == 1) ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
+#define TCOMB(a, b) ((comb == 1) ? fminf(a, b) : (comb == 2 ? fmaxf(a, b) : (a) + (b)))
     for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nseg;
          j += gridDim.x * (int64_t)blockDim.x) {
         const uint32_t slot = seg_slot[j];
@@ -1448,7 +1453,9 @@ __global__ void k_ffat_tree(const uint32_t* seg_start, const uint32_t* seg_slot,
         int64_t w = fire_base[j];
         float* tr = tree + (size_t)slot * 2 * R;
         for (; i < e; ++i) {
-            float x = wfa_val_at(v_f32, vdt, vdt == 6 ? i : idx_sorted[i]);
+            float x = (comb == 3)
+                          ? 1.0f
+                          : wfa_val_at(v_f32, vdt, vdt == 6 ? i : idx_sorted[i]);
             acc = TCOMB(acc, x);
             if (++fill == (uint32_t)pane_len) {
                 uint32_t leaf = (head & Rm) + R;

@@ -478,6 +478,41 @@ def test_gpu_reduce_keyed_sum():
     assert abs(res['s'] - ref) <= 2e-3 * max(1.0, abs(ref))
 
 
+def test_gpu_ffat_count_comb():
+    """COUNT combiner through the FFAT folds (ring and tree): every full CB
+    window counts exactly `win` tuples; window count per key is exact."""
+    n, n_keys, b, win, slide = 400_000, 101, 100_000, 300, 100
+    for tree in (False, True):
+        src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=2))
+               .withOutputSchema([2]).withOutputBatchSize(b).build())
+        ff = (Ffat_Windows_GPU_Builder(
+            native_gpu.gpu_ffat_windows(native_gpu.COMB_COUNT, 0, win, slide,
+                                        max_keys=256, use_tree=tree))
+            .withOutputSchema([2]).withOutputBatchSize(b).build())
+        got = dict(rows=0, bad=0, per={})
+
+        def pysink(cols):
+            import numpy as _np
+            got['rows'] += len(cols['c0'])
+            got['bad'] += int((_np.abs(cols['c0'] - win) > 1e-3).sum())
+            for k in cols['key'].tolist():
+                got['per'][k] = got['per'].get(k, 0) + 1
+
+        g = wf.PipeGraph("cnt")
+        p = g.add_source(src)
+        p.chain(ff)
+        snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+        snk.out_schema = [2]
+        p.add_sink(snk)
+        g.run()
+        assert got['bad'] == 0
+        ts, key, val = gen_batch(n, 0, 42, n_keys, 2)
+        import collections
+        per = collections.Counter(key.tolist())
+        exp_windows = {k: max(0, (c - win) // slide + 1) for k, c in per.items()}
+        assert got['per'] == {k: v for k, v in exp_windows.items() if v > 0}
+
+
 def test_gpu_reduce_all():
     """Unkeyed full-batch reduce vs numpy: one tuple per batch (sum and max)."""
     n, b = 1_000_000, 250_000
