@@ -69,6 +69,8 @@ def check():
         for k, v in zip(d['key'].tolist(), d['val'].tolist()):
             got[k].append(float(v))
         ts, key, val = gen_batch(n, 0, 42, n_keys, vdt)
+        if vdt == 5:  # decode raw bf16 bit patterns
+            val = (val.astype(np.uint32) << 16).view(np.float32)
         per = defaultdict(list)
         for t, k, v in zip(ts.tolist(), key.tolist(), val.tolist()):
             per[k].append((t, np.float32(v)))
