@@ -83,7 +83,15 @@ struct VarKV {
         index[key] = {tail + 12, len};
         tail += 12 + len;
         live += len;
-        if (tail > (64u << 20) && live * 2 < tail) compact();
+        if (tail > compact_bytes() && live * 2 < tail) compact();
+    }
+    static uint64_t compact_bytes() {
+        // tunable so tests can exercise compaction without writing 64 MB
+        static uint64_t v = [] {
+            const char* e = getenv("WFA_KV_COMPACT_BYTES");
+            return e ? (uint64_t)atoll(e) : (uint64_t)(64u << 20);
+        }();
+        return v;
     }
     bool get(uint64_t key, std::string& out) {
         std::lock_guard<std::mutex> g(mu);
