@@ -376,3 +376,21 @@ def test_py_window_parallel_replicas():
             exp[(k, float(max(seg) - min(seg)))] += 1
             w += 1
     assert Counter((k, v) for _, k, v in rows) == exp
+
+
+def test_py_window_paned_and_mapreduce_fallback():
+    """Python (non-incremental) window functions on the paned and mapreduce
+    forms: executed on the keyed python engine (same windows, same results —
+    the parallel decomposition differs, which is invisible to the user)."""
+    win, slide = 30, 10
+
+    def total(w):
+        return float(w['c0'].sum())
+
+    for B in (Paned_Windows_Builder, MapReduce_Windows_Builder):
+        kw = ("plq_func" if B is Paned_Windows_Builder else "map_func")
+        rows = run_graph(B(**{kw: total}).withCBWindows(win, slide)
+                         .withParallelism(3).withOutputSchema([0]).build(),
+                         stream_len=1800)
+        assert got_counter(rows) == oracle_cb(seq_stream(1800, 7), win, slide,
+                                              "sum"), B.__name__

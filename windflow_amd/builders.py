@@ -152,7 +152,12 @@ class MapReduce_Windows_Builder(_WindowsBuilder):
     def __init__(self, map_func=None, reduce_func=None, lift=None, comb=None):
         super().__init__(map_func, lift, comb)
         self._op.extra['reduce_func'] = reduce_func
-        self._op.broadcast_input = True
+        # python (non-incremental) map functions run on the keyed python
+        # engine (KEYBY); compiled combines use the reference's BROADCAST +
+        # round-robin MAP decomposition
+        self._op.broadcast_input = not callable(map_func)
+        if callable(map_func):
+            self._op.key_extractor = 'carried'   # KEYBY on the key column
 
     def withMAPParallelism(self, p):
         self._op.extra['map_par'] = int(p)

@@ -46,6 +46,16 @@ def _win_params(op):
     return int(w["type"]), int(w["win"]), int(w["slide"]), int(w.get("lateness", 0))
 
 
+def _attach_tail(graph, e, node, out_id):
+    # stages chained after the window op (e.g. chain_sink) attach to out_id
+    for extra_op in node.ops[1:]:
+        a = graph._stage_args(extra_op, node)
+        e.chain_stage(out_id, a[0], a[1], fparams=a[2], iparams=a[3],
+                      out_schema=a[4], out_batch=a[5], pyfn=a[6])
+        if extra_op.kind == "sink":
+            graph._sink_map[id(extra_op)] = out_id
+
+
 def lower_window_node(graph, e, node):
     """Create engine op(s) for one window/join node; returns (in_id, out_id)."""
     op = node.ops[0]
@@ -83,7 +93,16 @@ def lower_window_node(graph, e, node):
         wt, win, slide, lat = _win_params(op)
         comb, col, pyfn = _parse_agg(op.logic)
         if pyfn:
-            raise NotImplementedError("paned windows require a compiled combine")
+            # python pane functions can't be recombined by the native WLQ;
+            # run the same windows on the keyed python engine instead (same
+            # results, key-partitioned rather than pane-partitioned)
+            eid = e.add_op(op.name or kind, node.parallelism, "win_keyed",
+                           iparams=[wt, win, slide, lat, 0, col, 0, 0],
+                           out_schema=out_schema, out_batch=op.out_batch,
+                           pyfn=pyfn)
+            in_id = out_id = eid
+            _attach_tail(graph, e, node, out_id)
+            return in_id, out_id
         pane = math.gcd(win, slide)
         win_p, slide_p = win // pane, slide // pane
         ui = _use_int(out_schema, comb)
@@ -107,7 +126,13 @@ def lower_window_node(graph, e, node):
         wt, win, slide, lat = _win_params(op)
         comb, col, pyfn = _parse_agg(op.logic)
         if pyfn:
-            raise NotImplementedError("mapreduce windows require a compiled combine")
+            eid = e.add_op(op.name or kind, node.parallelism, "win_keyed",
+                           iparams=[wt, win, slide, lat, 0, col, 0, 0],
+                           out_schema=out_schema, out_batch=op.out_batch,
+                           pyfn=pyfn)
+            in_id = out_id = eid
+            _attach_tail(graph, e, node, out_id)
+            return in_id, out_id
         ui = _use_int(out_schema, comb)
         map_par = int(op.extra.get("map_par", node.parallelism))
         red_par = int(op.extra.get("reduce_par", node.parallelism))
@@ -126,11 +151,5 @@ def lower_window_node(graph, e, node):
     else:
         raise NotImplementedError(f"window kind {kind}")
 
-    # stages chained after the window op (e.g. chain_sink) attach to out_id
-    for extra_op in node.ops[1:]:
-        a = graph._stage_args(extra_op, node)
-        e.chain_stage(out_id, a[0], a[1], fparams=a[2], iparams=a[3],
-                      out_schema=a[4], out_batch=a[5], pyfn=a[6])
-        if extra_op.kind == "sink":
-            graph._sink_map[id(extra_op)] = out_id
+    _attach_tail(graph, e, node, out_id)
     return in_id, out_id
