@@ -53,14 +53,20 @@ struct VarKV {
         if (!fresh) replay();
     }
     void replay() {
+        struct stat st {};
+        if (fstat(fd, &st) != 0) throw std::runtime_error("VarKV: fstat failed");
+        const uint64_t fsize = (uint64_t)st.st_size;
         char hdr[12];
         uint64_t off = 0;
         for (;;) {
-            if (pread(fd, hdr, 12, (off_t)off) != 12) break;
+            if (off + 12 > fsize || pread(fd, hdr, 12, (off_t)off) != 12) break;
             uint64_t key;
             uint32_t len;
             memcpy(&key, hdr, 8);
             memcpy(&len, hdr + 8, 4);
+            // torn tail record (crash mid-write): ignore it and let the
+            // next put() overwrite from here
+            if (off + 12 + len > fsize) break;
             auto it = index.find(key);
             if (it != index.end()) live -= it->second.len;
             index[key] = {off + 12, len};
@@ -68,6 +74,8 @@ struct VarKV {
             off += 12 + len;
         }
         tail = off;
+        if (ftruncate(fd, (off_t)tail) != 0)
+            throw std::runtime_error("VarKV: truncate failed");
     }
 
     void put(uint64_t key, const void* data, uint32_t len) {

@@ -245,3 +245,25 @@ print('COMPACT_OK')
     assert "COMPACT_OK" in out.stdout
     # the log actually compacted: far smaller than the ~650KB written
     assert (tmp_path / "c.log").stat().st_size < 300_000
+
+
+def test_statestore_resume_torn_tail(tmp_path):
+    """A crash mid-append leaves a torn final record; replay must ignore it
+    and keep every complete record."""
+    p = str(tmp_path / "torn.log")
+    s = _core.StateStore(p, cache_capacity=2)
+    for k in range(20):
+        s.put(k, struct.pack("<q", k * 11))
+    s.flush()
+    del s
+    with open(p, "ab") as f:      # simulate a crash: header + half a value
+        f.write(struct.pack("<QI", 99, 64) + b"\x01" * 10)
+    r = _core.StateStore(p, cache_capacity=2, fresh=False)
+    assert len(r) == 20 and r.get(99) is None
+    for k in range(20):
+        assert struct.unpack("<q", r.get(k))[0] == k * 11
+    r.put(99, struct.pack("<q", 7))   # tail reclaimed cleanly
+    r.flush()
+    del r
+    r2 = _core.StateStore(p, cache_capacity=2, fresh=False)
+    assert struct.unpack("<q", r2.get(99))[0] == 7
