@@ -320,7 +320,9 @@ __global__ void k_rs_scatter(const uint32_t* keys, const uint32_t* vals, int64_t
 // precedes j+1; within a round, lane order).  Per-item stable rank =
 // LDS per-wave digit counter before this round + lane rank inside the
 // round's same-digit ballot group.
-#define RS8_IPT 16
+#ifndef RS8_IPT
+#define RS8_IPT 16   // items per thread in the 8-bit sort (A/B via -DRS8_IPT=N)
+#endif
 #define RS8_PER_WAVE (64 * RS8_IPT)
 #define RS8_PER_BLOCK (WFA_THREADS * RS8_IPT)
 
