@@ -168,7 +168,7 @@ extern "C" int64_t wfa_sort_nblocks(int64_t n) {
 // scratch requirement (u32 elements) covering the 4-bit and 8-bit paths
 extern "C" int64_t wfa_sort_hist_u32(int64_t cap) {
     int64_t nb4 = (cap + RS_PER_BLOCK - 1) / RS_PER_BLOCK;
-    int64_t nb8 = (cap + WFA_THREADS * 16 - 1) / (WFA_THREADS * 16);  // RS8
+    int64_t nb8 = (cap + RS8_PER_BLOCK - 1) / RS8_PER_BLOCK;
     int64_t a = 16 * nb4 + 16;
     int64_t b = 256 * nb8 + 512;
     return (a > b ? a : b) + 64;
