@@ -20,6 +20,11 @@
 #define WFA_MAX_BLOCKS 2048
 #define RS_IPT 8
 #define RS_PER_BLOCK (WFA_THREADS * RS_IPT)
+#ifndef RS8_IPT
+#define RS8_IPT 16   // items per thread in the 8-bit sort (A/B via -DRS8_IPT=N)
+#endif
+#define RS8_PER_WAVE (64 * RS8_IPT)
+#define RS8_PER_BLOCK (WFA_THREADS * RS8_IPT)
 
 static inline int64_t nblk(int64_t n, int64_t per_thread = 1) {
     int64_t b = (n + WFA_THREADS * per_thread - 1) / (WFA_THREADS * per_thread);
@@ -320,11 +325,6 @@ __global__ void k_rs_scatter(const uint32_t* keys, const uint32_t* vals, int64_t
 // precedes j+1; within a round, lane order).  Per-item stable rank =
 // LDS per-wave digit counter before this round + lane rank inside the
 // round's same-digit ballot group.
-#ifndef RS8_IPT
-#define RS8_IPT 16   // items per thread in the 8-bit sort (A/B via -DRS8_IPT=N)
-#endif
-#define RS8_PER_WAVE (64 * RS8_IPT)
-#define RS8_PER_BLOCK (WFA_THREADS * RS8_IPT)
 
 extern "C" int64_t wfa_sort8_nblocks(int64_t n) {
     return (n + RS8_PER_BLOCK - 1) / RS8_PER_BLOCK;
