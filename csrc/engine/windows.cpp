@@ -1257,6 +1257,9 @@ std::shared_ptr<OpLogic> make_window_logic(const std::string& kind,
     if (userfn && kind == "win_keyed")
         return std::make_shared<PyWindowLogic>(wt, geti(1), geti(2), geti(3),
                                                std::move(userfn));
+    if (userfn)  // never silently drop a user window function
+        throw std::runtime_error("python window function not supported for " +
+                                 kind + " (lowering should map it to win_keyed)");
     AggCfg a{(int)geti(4), geti(7, 1) != 0};
     int col = (int)geti(5);
     if (kind == "win_ffat")

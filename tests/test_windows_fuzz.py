@@ -73,3 +73,56 @@ def test_join_fuzz(case):
     got = Counter((k, a, bb) for _, k, a, bb in rows)
     assert got == oracle_join(n, keys, lower, upper), (lower, upper, keys, n,
                                                        batch, par, kp)
+
+
+@pytest.mark.parametrize("case", range(8))
+def test_py_window_fn_fuzz(case):
+    """Non-incremental python window functions (median/span/sumsq) on every
+    window form vs brute-force oracle (caught the parallel form silently
+    running the native sum combiner instead of the user's function)."""
+    from collections import Counter
+    import numpy as np
+    from test_windows import run_graph, seq_stream
+    from windflow_amd.builders import (Keyed_Windows_Builder,
+                                       Parallel_Windows_Builder,
+                                       Paned_Windows_Builder,
+                                       MapReduce_Windows_Builder)
+    FNS = {"median": lambda a: float(np.median(a)),
+           "span": lambda a: float(max(a) - min(a)),
+           "sumsq": lambda a: float(np.sum(np.asarray(a, np.float64) ** 2))}
+    BUILD = {"keyed": lambda f: Keyed_Windows_Builder(func=f),
+             "parallel": lambda f: Parallel_Windows_Builder(func=f),
+             "paned": lambda f: Paned_Windows_Builder(plq_func=f),
+             "mapreduce": lambda f: MapReduce_Windows_Builder(map_func=f)}
+    rng = random.Random(777 + case * 7)
+    form = rng.choice(list(BUILD))
+    fn = FNS[rng.choice(list(FNS))]
+    wt = rng.choice(["cb", "tb"])
+    slide = rng.choice([5, 10, 20])
+    win = slide * rng.randint(1, 5)
+    keys = rng.choice([1, 3, 7])
+    par = rng.randint(1, 3)
+    stream = rng.choice([600, 1500])
+    pyfn = lambda w, _f=fn: _f(w['c0']) if len(w['c0']) else 0.0
+    b = BUILD[form](pyfn)
+    b = b.withCBWindows(win, slide) if wt == "cb" else b.withTBWindows(win, slide)
+    rows = run_graph(b.withParallelism(par).withOutputSchema([1]).build(),
+                     stream_len=stream, n_keys=keys)
+    per = seq_stream(stream, keys)
+    exp = Counter()
+    for k, r in per.items():
+        vals = [v for _, v in r]
+        tss = [t for t, _ in r]
+        if wt == "cb":
+            w = 0
+            while w * slide < len(vals):
+                exp[(k, round(fn(vals[w*slide:w*slide+win]), 6))] += 1
+                w += 1
+        else:
+            t0, tmax = tss[0], tss[-1]
+            w0 = max(0, -(-(t0 - win + 1) // slide))
+            for w in range(w0, tmax // slide + 1):
+                seg = [v for t, v in zip(tss, vals) if w*slide <= t < w*slide+win]
+                exp[(k, round(fn(seg) if seg else 0.0, 6))] += 1
+    got = Counter((k, round(v, 6)) for _, k, v in rows)
+    assert got == exp, (form, wt, win, slide, keys, par)

@@ -124,7 +124,11 @@ class Parallel_Windows_Builder(_WindowsBuilder):
 
     def __init__(self, func=None, lift=None, comb=None):
         super().__init__(func, lift, comb)
-        self._op.broadcast_input = True
+        # python (non-incremental) functions run on the keyed python engine
+        # (KEYBY); compiled combines use BROADCAST + gwid%n window ownership
+        self._op.broadcast_input = not callable(func)
+        if callable(func):
+            self._op.key_extractor = 'carried'
 
 
 class Paned_Windows_Builder(_WindowsBuilder):

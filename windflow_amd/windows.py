@@ -83,7 +83,9 @@ def lower_window_node(graph, e, node):
         ui = _use_int(out_schema, comb)
         ekind = {"keyed_windows": "win_keyed", "parallel_windows": "win_parallel",
                  "ffat_windows": "win_ffat"}[kind]
-        own = 1 if kind == "parallel_windows" else 0
+        if pyfn:  # python fns execute on the keyed python engine (all forms)
+            ekind = "win_keyed"
+        own = 1 if (kind == "parallel_windows" and not pyfn) else 0
         ip = [wt, win, slide, lat, comb, col, own, ui]
         eid = e.add_op(op.name or kind, node.parallelism, ekind, iparams=ip,
                        out_schema=out_schema, out_batch=op.out_batch, pyfn=pyfn)
