@@ -36,7 +36,10 @@ def main():
         got = [None]
         td.broadcast_object_list(got, src=0)
         assert got[0] == rid, "rccl id bytes differ across ranks"
-    n, n_keys, batch = 20000, 13, 500  # n % batch == 0: equal rounds per rank
+    # config via env (fuzzable); n % batch == 0: equal rounds per rank
+    n = int(os.environ.get("WFZ_N", 20000))
+    n_keys = int(os.environ.get("WFZ_KEYS", 13))
+    batch = int(os.environ.get("WFZ_BATCH", 500))
 
     got = {"sum": 0, "rows": 0, "bad_dest": 0}
 

@@ -11,6 +11,44 @@ import sys
 import pytest
 
 
+def _run_world(world, port, extra_env=None):
+    env = dict(os.environ)
+    env.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+               WORLD_SIZE=str(world), GLOO_SOCKET_IFNAME="lo")
+    if extra_env:
+        env.update(extra_env)
+    worker = os.path.join(os.path.dirname(__file__), "dist_worker.py")
+    procs = []
+    for r in range(world):
+        e = dict(env, RANK=str(r))
+        procs.append(subprocess.Popen([sys.executable, worker], env=e,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=110)
+        outs.append(out.decode())
+    for p, o in zip(procs, outs):
+        assert p.returncode == 0, o
+    assert "DIST_CPU_OK" in outs[0], outs[0]
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("case", range(4))
+def test_cpu_keyby_exchange_fuzz(case):
+    """Exchange correctness across worlds 2-3 and varied key/batch shapes:
+    keys partitioned by the shared splitmix64 hash, nothing lost or
+    duplicated (sum+count conservation across ranks)."""
+    import random
+    rng = random.Random(4000 + case)
+    world = rng.choice([2, 3])
+    batch = rng.choice([100, 500, 2048])
+    n = batch * rng.randint(4, 30)
+    keys = rng.choice([1, 7, 64, 1000])
+    _run_world(world, 29600 + case,
+               dict(WFZ_N=str(n), WFZ_KEYS=str(keys), WFZ_BATCH=str(batch)))
+
+
 @pytest.mark.timeout(120)
 def test_cpu_keyby_exchange_world2():
     env = dict(os.environ)
