@@ -185,6 +185,34 @@ def _run_pkw(tmp_path, tb, win, slide, n, keys, par):
     return res
 
 
+import pytest
+
+
+@pytest.mark.parametrize("case", range(6))
+def test_p_keyed_windows_fuzz(case, tmp_path):
+    """Random CB/TB shapes on the store-backed keyed windows vs the window
+    oracles (caught the TB cursor skipping data-free gaps that must fire
+    as 0)."""
+    import random
+    from collections import Counter
+    from tests.test_windows import oracle_cb, oracle_tb
+    rng = random.Random(6000 + case * 11)
+    tb = rng.random() < 0.5
+    slide = rng.choice([3, 7, 12, 50])
+    win = slide * rng.randint(1, 6)
+    keys = rng.choice([1, 5, 17, 40])
+    n = rng.choice([800, 3000, 6000])
+    par = rng.randint(1, 3)
+    res = _run_pkw(tmp_path, tb, win, slide, n, keys, par=par)
+    per = {}
+    for v in range(1, n + 1):
+        per.setdefault(v % keys, []).append((v, v))
+    exp = (oracle_tb if tb else oracle_cb)(per, win, slide, "sum")
+    got = Counter((k, float(v)) for k, v in res)
+    assert got == Counter({(k, float(v)): c for (k, v), c in exp.items()}), \
+        (tb, win, slide, keys, n, par)
+
+
 def test_p_keyed_windows_cb(tmp_path):
     from collections import Counter
     from tests.test_windows import oracle_cb

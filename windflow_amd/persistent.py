@@ -173,25 +173,32 @@ class _PKeyedWin:
         return max(0, (int(t) - W) // S + 1) if int(t) >= W else 0
 
     def _fire_tb(self, k, ts, vals, wm, out):
+        # fire every absolute-grid window [q*S, q*S+W) with end <= wm whose
+        # start is not beyond the key's newest tuple, in order; q is a
+        # persistent per-key cursor, so data-free gaps fire as 0 (same as
+        # the in-memory engine) and nothing re-fires
         W, S = self.win, self.slide
-        # fire every absolute-grid window [q*S, q*S+W) with end <= wm, in
-        # order; q advances monotonically (never re-derived backwards) and
-        # jumps over data-free gaps
-        if not len(ts):
-            return ts, vals
-        q = max(self._first_q(ts[0], W, S), self.next_q.get(k, 0))
-        while len(ts):
+        mt = self.max_ts.get(k, -1)
+        q = self.next_q.get(k)
+        if q is None:
+            if not len(ts):
+                return ts, vals
+            q = self._first_q(ts[0], W, S)
+        while q * S <= mt:
             end = q * S + W
             if end > wm:
                 break
-            m = (ts >= q * S) & (ts < end)
-            # empty in-range windows fire with 0 (same as the in-memory
-            # window ops — oracle_tb in tests/test_windows.py)
-            out.append((end - 1, k, float(self.fn(vals[m])) if m.any() else 0.0))
+            if len(ts):
+                m = (ts >= q * S) & (ts < end)
+                out.append((end - 1, k,
+                            float(self.fn(vals[m])) if m.any() else 0.0))
+            else:
+                out.append((end - 1, k, 0.0))
             q += 1
-            keep = ts >= q * S
-            ts, vals = ts[keep], vals[keep]
-        self.next_q[k] = max(self.next_q.get(k, 0), q)
+            if len(ts):
+                keep = ts >= q * S
+                ts, vals = ts[keep], vals[keep]
+        self.next_q[k] = q
         return ts, vals
 
     def __call__(self, cols):
@@ -214,7 +221,7 @@ class _PKeyedWin:
                 ts, vals = ts[o], vals[o]
                 # drop tuples entirely before the fired horizon (late in
                 # DEFAULT mode, reference window_replica.hpp lateness gate)
-                keep = ts >= self.next_q.get(k, 0) * self.slide
+                keep = ts >= (self.next_q.get(k) or 0) * self.slide
                 ts, vals = ts[keep], vals[keep]
                 if not len(ts):
                     self._save(k, ts, vals)
