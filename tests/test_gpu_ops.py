@@ -513,6 +513,46 @@ def test_gpu_ffat_count_comb():
         assert got['per'] == {k: v for k, v in exp_windows.items() if v > 0}
 
 
+def test_gpu_jit_expr_fuzz():
+    """Random hiprtc-compiled map expressions vs numpy float32 evaluation."""
+    import random
+    EXPRS = [
+        ("v * 2.5f + 1.0f",            lambda v, t, k: v * np.float32(2.5) + 1),
+        ("fmaxf(v, 100.0f)",           lambda v, t, k: np.maximum(v, 100)),
+        ("fminf(v * v, 9000.0f)",      lambda v, t, k: np.minimum(v * v, 9000)),
+        ("fabsf(v - 500.0f)",          lambda v, t, k: np.abs(v - 500)),
+        ("v + (float)(key & 7)",       lambda v, t, k: v + (k & 7).astype(np.float32)),
+        ("v * 0.5f + (float)(ts % 11)", lambda v, t, k: v * np.float32(0.5)
+                                        + (t % 11).astype(np.float32)),
+    ]
+    n, n_keys, b = 200_000, 64, 50_000
+    rng = random.Random(99)
+    for expr, ref in rng.sample(EXPRS, 4):
+        src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=2))
+               .withOutputSchema([2]).withOutputBatchSize(b).build())
+        jm = (Map_GPU_Builder(native_gpu.gpu_jit_map(expr, 0))
+              .withOutputSchema([2]).withOutputBatchSize(b).build())
+        acc = dict(s=0.0, n=0)
+
+        def pysink(cols):
+            acc['s'] += float(cols['c0'].astype(np.float64).sum())
+            acc['n'] += len(cols['c0'])
+
+        g = wf.PipeGraph("jitf")
+        p = g.add_source(src)
+        p.chain(jm)
+        snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+        snk.out_schema = [2]
+        p.add_sink(snk)
+        g.run()
+        ts, key, val = gen_batch(n, 0, 42, n_keys, 2)
+        exp = float(ref(val.astype(np.float32), ts, key)
+                    .astype(np.float64).sum())
+        assert acc['n'] == n, expr
+        assert abs(acc['s'] - exp) <= 2e-3 * max(1.0, abs(exp)), \
+            (expr, acc['s'], exp)
+
+
 def test_gpu_reduce_all():
     """Unkeyed full-batch reduce vs numpy: one tuple per batch (sum and max)."""
     n, b = 1_000_000, 250_000
