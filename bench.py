@@ -240,6 +240,8 @@ def main():
         g.run()
         dt = time.time() - t0
         n_gpus = 0
+        lat = sorted(g.engine.sink_latencies(g._sink_map[id(snk)]))
+        p99_us = lat[min(len(lat) - 1, int(0.99 * len(lat)))] if lat else None
     else:
         def builder(steps):
             if args.config == "mapfilter":
@@ -308,8 +310,7 @@ def main():
                 "slide": args.slide,
                 "keys_per_rank": args.keys,
                 "parallelism": f"keyed-dp{max(world,1)}",
-                "p99_batch_latency_us": (p99_us if args.config != "cpu"
-                                         else None),
+                "p99_batch_latency_us": p99_us,
             },
         }
         print(json.dumps(out))

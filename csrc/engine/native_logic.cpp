@@ -52,6 +52,7 @@ struct SeqSource : OpLogic {
             b->col<int64_t>(0)[i] = v;
         }
         b->count = n;
+        b->born_us = now_us();   // batch-latency origin (sink_latencies)
         pos += n;
         b->watermark = ingress ? nowus : pos;  // wm = max emitted ts
         out.emit(b);
@@ -205,16 +206,21 @@ struct SumSinkI64 : OpLogic {
     int op_id;
     int col;
     int64_t local = 0, tuples = 0;
+    std::vector<int64_t> lat_us;
     SumSinkI64(Engine* e, int id, int c) : eng(e), op_id(id), col(c) {}
     void process(Batch* b, EmitCtx& out, RuntimeCtx&) override {
         int64_t* x = b->col<int64_t>(col);
         for (int64_t i = 0; i < b->count; ++i) local += x[i];
         tuples += b->count;
+        if (b->born_us) lat_us.push_back(now_us() - b->born_us);
         release(b);
     }
     void on_eos(EmitCtx&, RuntimeCtx&) override {
         eng->sink_acc_i64[op_id].fetch_add(local, std::memory_order_relaxed);
         eng->sink_tuples[op_id].fetch_add(tuples, std::memory_order_relaxed);
+        std::lock_guard<std::mutex> g(eng->sink_f64_mu);
+        auto& v = eng->sink_latencies[op_id];
+        v.insert(v.end(), lat_us.begin(), lat_us.end());
     }
 };
 
@@ -245,13 +251,20 @@ struct CountSink : OpLogic {
     Engine* eng;
     int op_id;
     int64_t tuples = 0;
+    std::vector<int64_t> lat_us;
     CountSink(Engine* e, int id) : eng(e), op_id(id) {}
     void process(Batch* b, EmitCtx&, RuntimeCtx&) override {
         tuples += b->count;
+        if (b->born_us) lat_us.push_back(now_us() - b->born_us);
         release(b);
     }
     void on_eos(EmitCtx&, RuntimeCtx&) override {
         eng->sink_tuples[op_id].fetch_add(tuples, std::memory_order_relaxed);
+        {
+            std::lock_guard<std::mutex> g(eng->sink_f64_mu);
+            auto& v = eng->sink_latencies[op_id];
+            v.insert(v.end(), lat_us.begin(), lat_us.end());
+        }
     }
 };
 
