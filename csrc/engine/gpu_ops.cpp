@@ -875,11 +875,7 @@ struct GpuFfatLogic : GpuLogicBase {
     uint32_t* tb_nf = nullptr;
     uint32_t* h_flags = nullptr;   // pinned readback
     int64_t batches = 0;
-    // two-stage fold scratch (WFA_PANE2=1)
-    uint32_t* p2_base = nullptr;
-    float* p2_vals = nullptr;
-    float* p2_tail = nullptr;
-    int64_t* p2_tmp = nullptr;
+    bool use_pane2 = false;
     // A/B on MI355X (8M-tuple batches): the two-stage fold wins when
     // segments are few (1024 keys: 16.9 vs 13.9 B t/s — the wave fold is
     // occupancy-bound at one wave per key) and loses slightly when the
@@ -931,12 +927,7 @@ struct GpuFfatLogic : GpuLogicBase {
         wfa_fill_f32(stream, ring_or_tree, ident,
                      max_keys * (use_tree ? 2 * (1ll << ring_log2) : (1ll << ring_log2)));
         cb_nf = (uint32_t*)A.get(4 * (out_cap + 1));
-        if (!tb && !use_tree && pane2_enabled() && pane_len >= 32) {
-            p2_base = (uint32_t*)A.get(4 * (max_keys + 1));
-            p2_vals = (float*)A.get(4 * (out_cap / pane_len + max_keys + 64));
-            p2_tail = (float*)A.get(4 * max_keys);
-            p2_tmp = (int64_t*)A.get(64);
-        }
+        use_pane2 = !tb && !use_tree && pane2_enabled() && pane_len >= 32;
         if (tb) {
             int64_t Rp = 1ll << pend_log2;
             tb_pend = (float*)A.get(4 * max_keys * Rp);
@@ -994,13 +985,13 @@ struct GpuFfatLogic : GpuLogicBase {
                                comb, ring_log2, st_count, st_fill, st_acc,
                                ring_or_tree, st_head, ks.slot_to_key, nf, ob->key,
                                (float*)ob->cols[0], ob->ts, ob->capacity);
-        else if (p2_vals)
-            wfa_ffat_cb_fold2(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                              ks.v_as_f32, ks.v_dt, ks.idx_sorted, db->ts, pane_len,
-                              P, S, comb, ring_log2, st_count, st_fill, st_acc,
-                              ring_or_tree, st_head, st_wsum, ks.slot_to_key, nf,
-                              p2_base, p2_vals, p2_tail, p2_tmp, ob->key,
-                              (float*)ob->cols[0], ob->ts, ob->capacity);
+        else if (use_pane2)
+            wfa_ffat_cb_fold_fused(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
+                                   ks.v_as_f32, ks.v_dt, ks.idx_sorted, db->ts,
+                                   pane_len, P, S, comb, ring_log2, st_count,
+                                   st_fill, st_acc, ring_or_tree, st_head, st_wsum,
+                                   ks.slot_to_key, nf, ob->key,
+                                   (float*)ob->cols[0], ob->ts, ob->capacity);
         else
             wfa_ffat_cb_fold(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                              ks.v_as_f32, ks.v_dt, ks.idx_sorted, db->ts, pane_len, P, S,

@@ -1593,212 +1593,141 @@ extern "C" void wfa_reduce_all(wfa_stream_t s, const void* v, int vdt,
                        (hipStream_t)s, scratch, scratch_ts, n, comb, out, out_ts);
 }
 
-// ===== two-stage CB fold (opt-in WFA_PANE2=1) =====
-// Stage 1 parallelizes over PANES: every complete pane of every segment is
-// wave-reduced independently (values read sequentially in VIK mode), plus
-// the head chunk (completing the key's open pane) and the tail chunk (the
-// new open remainder).  Stage 2 runs the per-key window machine over PANE
-// VALUES (~100x fewer items than tuples), fixing the fold's key-count
-// occupancy cliff.  Semantics identical to k_ffat_cb_wave.
+// ===== two-stage CB fold (auto-selected; see gpu_ops.cpp) =====
+// Fused variant: one kernel, block per segment.  Waves cooperatively
+// compute pane partials into LDS in chunks; wave 0 runs the window machine
+// over each chunk.  No global pane_vals traffic, no extra launches.
+#define P2_CHUNK 512
 
-// per-segment pane-slot offsets: pane_base[j] = exclusive scan of the
-// number of COMPLETE panes each segment closes (head-completing pane
-// included); single block, same shape as k_fire_scan
-__global__ void k_pane2_base(const uint32_t* seg_start, const uint32_t* seg_slot,
-                             const int64_t* d_nseg, int64_t n, int64_t pane_len,
-                             const uint32_t* st_fill, uint32_t* pane_base,
-                             int64_t* d_total) {
-    const int64_t m = *d_nseg;
-    __shared__ uint32_t tot[1024];
-    const int64_t chunk = (m + 1023) / 1024;
-    const int64_t b0 = (int64_t)threadIdx.x * chunk;
-    const int64_t b1 = min(m, b0 + chunk);
-    uint32_t s = 0;
-    for (int64_t j = b0; j < b1; ++j) {
-        const int64_t len = ((j + 1 < m) ? seg_start[j + 1] : n) - seg_start[j];
-        const uint32_t c =
-            (uint32_t)(((uint64_t)st_fill[seg_slot[j]] + (uint64_t)len) /
-                       (uint64_t)pane_len);
-        pane_base[j] = c;
-        s += c;
-    }
-    tot[threadIdx.x] = s;
-    __syncthreads();
-    for (int off = 1; off < 1024; off <<= 1) {
-        uint32_t t = (threadIdx.x >= off) ? tot[threadIdx.x - off] : 0;
-        __syncthreads();
-        tot[threadIdx.x] += t;
-        __syncthreads();
-    }
-    uint32_t run = tot[threadIdx.x] - s;
-    for (int64_t j = b0; j < b1; ++j) {
-        uint32_t v = pane_base[j];
-        pane_base[j] = run;
-        run += v;
-    }
-    if (threadIdx.x == 1023 && d_total) *d_total = tot[1023];
-}
-
-// stage 1: one block per segment; waves cooperatively reduce each complete
-// pane (the first may start mid-pane, continuing st_acc) and the tail
-// remainder.  pane_vals[pane_base[j] + i] = i-th closed pane's CHUNK value
-// (the head pane's chunk EXCLUDES the carried st_acc — stage 2 merges it).
-__global__ void k_pane2_partials(const uint32_t* seg_start, const uint32_t* seg_slot,
-                                 const int64_t* d_nseg, int64_t n,
-                                 const void* v_f32, int vdt,
-                                 const uint32_t* idx_sorted, int64_t pane_len,
-                                 int comb, const uint32_t* st_fill,
-                                 const uint32_t* pane_base, float* pane_vals,
-                                 float* tail_vals) {
-    const int64_t nseg = *d_nseg;
-    const float ident = (comb == 1) ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
-    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
-    const int nwaves = blockDim.x >> 6;
-#define PCOMB(a, b) ((comb == 1) ? fminf(a, b) : (comb == 2 ? fmaxf(a, b) : (a) + (b)))
-    for (int64_t j = blockIdx.x; j < nseg; j += gridDim.x) {
-        const int64_t b = seg_start[j];
-        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
-        const uint32_t fill0 = st_fill[seg_slot[j]];
-        const int64_t head_need = (int64_t)pane_len - (int64_t)fill0;  // to close pane 0
-        const int64_t len = e - b;
-        const int64_t ncomplete = ((int64_t)fill0 + len) / pane_len;
-        // pane i covers tuples [b + i*L - fill0, b + (i+1)*L - fill0) ∩ [b, e)
-        for (int64_t i = wave; i < ncomplete; i += nwaves) {
-            const int64_t lo = (i == 0) ? b : b + i * pane_len - (int64_t)fill0;
-            const int64_t hi = b + (i + 1) * pane_len - (int64_t)fill0;
-            float acc = ident;
-            for (int64_t p = lo + lane; p < hi; p += 64) {
-                float x = (comb == 3)
-                              ? 1.0f
-                              : wfa_val_at(v_f32, vdt, vdt == 6 ? p : idx_sorted[p]);
-                acc = PCOMB(acc, x);
-            }
-            for (int o = 32; o; o >>= 1) acc = PCOMB(acc, __shfl_xor(acc, o, 64));
-            if (lane == 0) pane_vals[pane_base[j] + i] = acc;
-        }
-        // tail chunk: the open remainder after the last complete pane
-        if (wave == 0) {
-            const int64_t lo = (ncomplete == 0)
-                                   ? b
-                                   : b + ncomplete * pane_len - (int64_t)fill0;
-            float acc = ident;
-            for (int64_t p = lo + lane; p < e; p += 64) {
-                float x = (comb == 3)
-                              ? 1.0f
-                              : wfa_val_at(v_f32, vdt, vdt == 6 ? p : idx_sorted[p]);
-                acc = PCOMB(acc, x);
-            }
-            for (int o = 32; o; o >>= 1) acc = PCOMB(acc, __shfl_xor(acc, o, 64));
-            if (lane == 0) tail_vals[j] = acc;
-        }
-        (void)head_need;
-    }
-#undef PCOMB
-}
-
-// stage 2: wave per segment over PANE VALUES; identical window machine to
-// k_ffat_cb_wave but the per-pane combine is one value, not pane_len tuples
-__global__ void k_pane2_machine(const uint32_t* seg_start, const uint32_t* seg_slot,
-                                const int64_t* d_nseg, int64_t n,
-                                const int64_t* ts_orig, const uint32_t* idx_sorted,
-                                int64_t pane_len, int64_t P, int64_t S, int comb,
-                                int ring_log2, int64_t* st_count, uint32_t* st_fill,
-                                float* st_acc, float* ring, uint32_t* st_head,
-                                float* st_wsum, const uint64_t* slot_to_key,
-                                const uint32_t* fire_base, const uint32_t* pane_base,
-                                const float* pane_vals, const float* tail_vals,
-                                uint64_t* out_key, float* out_val, int64_t* out_ts,
-                                int64_t out_cap) {
+__global__ void k_pane2_fused(const uint32_t* seg_start, const uint32_t* seg_slot,
+                              const int64_t* d_nseg, int64_t n,
+                              const void* v_f32, int vdt,
+                              const uint32_t* idx_sorted, const int64_t* ts_orig,
+                              int64_t pane_len, int64_t P, int64_t S, int comb,
+                              int ring_log2, int64_t* st_count, uint32_t* st_fill,
+                              float* st_acc, float* ring, uint32_t* st_head,
+                              float* st_wsum, const uint64_t* slot_to_key,
+                              const uint32_t* fire_base,
+                              uint64_t* out_key, float* out_val, int64_t* out_ts,
+                              int64_t out_cap) {
     const int64_t nseg = *d_nseg;
     const uint32_t R = 1u << ring_log2;
     const uint32_t Rm = R - 1;
     const float ident = (comb == 1) ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
-    const int lane = threadIdx.x & 63;
-    const int64_t wid = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-    const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
-#define MCOMB(a, b) ((comb == 1) ? fminf(a, b) : (comb == 2 ? fmaxf(a, b) : (a) + (b)))
-    for (int64_t j = wid; j < nseg; j += nw) {
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int nwaves = blockDim.x >> 6;
+    __shared__ float pv[P2_CHUNK];
+#define FCOMB(a, b) ((comb == 1) ? fminf(a, b) : (comb == 2 ? fmaxf(a, b) : (a) + (b)))
+    for (int64_t j = blockIdx.x; j < nseg; j += gridDim.x) {
         const uint32_t slot = seg_slot[j];
         const int64_t b = seg_start[j];
         const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
-        const uint32_t fill0 = st_fill[slot];
         const int64_t len = e - b;
+        const uint32_t fill0 = st_fill[slot];
         const int64_t ncomplete = ((int64_t)fill0 + len) / pane_len;
+        // wave-0 machine state (lane-uniform registers)
         float acc = st_acc[slot];
         uint32_t head = st_head[slot];
         float wsum = st_wsum[slot];
         int64_t w = fire_base[j];
         float* rg = ring + (size_t)slot * R;
-        for (int64_t i = 0; i < ncomplete; ++i) {
-            // close pane i: carried acc (pane 0 only) + its chunk value
-            float pane = (i == 0) ? MCOMB(acc, pane_vals[pane_base[j]])
-                                  : pane_vals[pane_base[j] + i];
-            if (i == 0) acc = ident;
-            rg[head & Rm] = pane;
-            if (comb == 0 || comb == 3) {
-                wsum += pane;
-                if (head >= (uint32_t)P) wsum -= rg[(head - (uint32_t)P) & Rm];
+        for (int64_t c0 = 0; c0 < ncomplete; c0 += P2_CHUNK) {
+            const int64_t nc = min((int64_t)P2_CHUNK, ncomplete - c0);
+            for (int64_t i = wave; i < nc; i += nwaves) {
+                const int64_t gi = c0 + i;
+                const int64_t lo = (gi == 0) ? b : b + gi * pane_len - (int64_t)fill0;
+                const int64_t hi = b + (gi + 1) * pane_len - (int64_t)fill0;
+                float a = ident;
+                for (int64_t p = lo + lane; p < hi; p += 64) {
+                    float x = (comb == 3)
+                                  ? 1.0f
+                                  : wfa_val_at(v_f32, vdt,
+                                               vdt == 6 ? p : idx_sorted[p]);
+                    a = FCOMB(a, x);
+                }
+                for (int o = 32; o; o >>= 1) a = FCOMB(a, __shfl_xor(a, o, 64));
+                if (lane == 0) pv[i] = a;
             }
-            ++head;
-            if (head >= (uint32_t)P && ((head - (uint32_t)P) % (uint32_t)S) == 0) {
-                float res;
-                if (comb == 0 || comb == 3) {
-                    res = wsum;
-                } else {
-                    float part = ident;
-                    for (uint32_t q = lane; q < (uint32_t)P; q += 64) {
-                        float x = rg[(head - 1 - q) & Rm];
-                        part = MCOMB(part, x);
+            __syncthreads();
+            if (wave == 0) {
+                for (int64_t i = 0; i < nc; ++i) {
+                    const int64_t gi = c0 + i;
+                    float pane = (gi == 0) ? FCOMB(acc, pv[i]) : pv[i];
+                    rg[head & Rm] = pane;
+                    if (comb == 0 || comb == 3) {
+                        wsum += pane;
+                        if (head >= (uint32_t)P)
+                            wsum -= rg[(head - (uint32_t)P) & Rm];
                     }
-                    for (int o = 32; o; o >>= 1)
-                        part = MCOMB(part, __shfl_xor(part, o, 64));
-                    res = part;
+                    ++head;
+                    if (head >= (uint32_t)P &&
+                        ((head - (uint32_t)P) % (uint32_t)S) == 0) {
+                        float res;
+                        if (comb == 0 || comb == 3) {
+                            res = wsum;
+                        } else {
+                            float part = ident;
+                            for (uint32_t q = lane; q < (uint32_t)P; q += 64)
+                                part = FCOMB(part, rg[(head - 1 - q) & Rm]);
+                            for (int o = 32; o; o >>= 1)
+                                part = FCOMB(part, __shfl_xor(part, o, 64));
+                            res = part;
+                        }
+                        if (lane == 0 && w < out_cap) {
+                            out_key[w] = slot_to_key[slot];
+                            out_val[w] = res;
+                            const int64_t last =
+                                b + (gi + 1) * pane_len - (int64_t)fill0 - 1;
+                            out_ts[w] = ts_orig ? ts_orig[idx_sorted[last]] : 0;
+                        }
+                        ++w;
+                    }
                 }
-                if (lane == 0 && w < out_cap) {
-                    out_key[w] = slot_to_key[slot];
-                    out_val[w] = res;
-                    // closing tuple of pane i
-                    const int64_t last = b + (i + 1) * pane_len - (int64_t)fill0 - 1;
-                    out_ts[w] = ts_orig ? ts_orig[idx_sorted[last]] : 0;
+            }
+            __syncthreads();  // pv reused next chunk
+        }
+        // tail remainder -> new open acc
+        {
+            const int64_t lo =
+                (ncomplete == 0) ? b : b + ncomplete * pane_len - (int64_t)fill0;
+            float a = ident;
+            if (wave == 0) {
+                for (int64_t p = lo + lane; p < e; p += 64) {
+                    float x = (comb == 3)
+                                  ? 1.0f
+                                  : wfa_val_at(v_f32, vdt,
+                                               vdt == 6 ? p : idx_sorted[p]);
+                    a = FCOMB(a, x);
                 }
-                ++w;
+                for (int o = 32; o; o >>= 1) a = FCOMB(a, __shfl_xor(a, o, 64));
+                const uint32_t newfill =
+                    (uint32_t)(((int64_t)fill0 + len) - ncomplete * pane_len);
+                if (lane == 0) {
+                    st_fill[slot] = newfill;
+                    st_acc[slot] = (ncomplete == 0) ? FCOMB(acc, a)
+                                                    : (newfill ? a : ident);
+                    st_head[slot] = head;
+                    st_wsum[slot] = wsum;
+                    st_count[slot] += len;
+                }
             }
         }
-        // open remainder
-        const float tail = tail_vals[j];
-        const uint32_t newfill =
-            (uint32_t)(((int64_t)fill0 + len) - ncomplete * pane_len);
-        if (lane == 0) {
-            st_fill[slot] = newfill;
-            st_acc[slot] = (ncomplete == 0) ? MCOMB(acc, tail)
-                                            : (newfill ? tail : ident);
-            st_head[slot] = head;
-            st_wsum[slot] = wsum;
-            st_count[slot] += len;
-        }
+        __syncthreads();  // state published before the block takes segment j+grid
     }
-#undef MCOMB
+#undef FCOMB
 }
 
-extern "C" void wfa_ffat_cb_fold2(
+extern "C" void wfa_ffat_cb_fold_fused(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
     const int64_t* d_nseg, int64_t n, const void* v_f32, int vdt,
     const uint32_t* idx_sorted, const int64_t* ts_orig, int64_t pane_len,
     int64_t P, int64_t S, int comb, int ring_log2, int64_t* st_count,
     uint32_t* st_fill, float* st_acc, float* ring, uint32_t* st_head,
     float* st_wsum, const uint64_t* slot_to_key, const uint32_t* fire_base,
-    uint32_t* pane_base, float* pane_vals, float* tail_vals, int64_t* d_scratch,
     uint64_t* out_key, float* out_val, int64_t* out_ts, int64_t out_cap) {
-    hipStream_t st = (hipStream_t)s;
-    hipLaunchKernelGGL(k_pane2_base, dim3(1), dim3(1024), 0, st, seg_start,
-                       seg_slot, d_nseg, n, pane_len, st_fill, pane_base,
-                       d_scratch);
-    hipLaunchKernelGGL(k_pane2_partials, dim3(WFA_MAX_BLOCKS), dim3(256), 0, st,
-                       seg_start, seg_slot, d_nseg, n, v_f32, vdt, idx_sorted,
-                       pane_len, comb, st_fill, pane_base, pane_vals, tail_vals);
-    hipLaunchKernelGGL(k_pane2_machine, dim3(WFA_MAX_BLOCKS), dim3(WFA_THREADS), 0,
-                       st, seg_start, seg_slot, d_nseg, n, ts_orig, idx_sorted,
-                       pane_len, P, S, comb, ring_log2, st_count, st_fill, st_acc,
-                       ring, st_head, st_wsum, slot_to_key, fire_base, pane_base,
-                       pane_vals, tail_vals, out_key, out_val, out_ts, out_cap);
+    hipLaunchKernelGGL(k_pane2_fused, dim3(WFA_MAX_BLOCKS), dim3(256), 0,
+                       (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_f32, vdt,
+                       idx_sorted, ts_orig, pane_len, P, S, comb, ring_log2,
+                       st_count, st_fill, st_acc, ring, st_head, st_wsum,
+                       slot_to_key, fire_base, out_key, out_val, out_ts, out_cap);
 }

@@ -125,17 +125,15 @@ void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
                       uint64_t* out_key, float* out_val, int64_t* out_ts,
                       int64_t out_cap);
 
-// two-stage CB fold (opt-in): parallel pane partials + per-key pane
-// machine.  pane_base >= (max segments + 1) u32, pane_vals >=
-// (cap/pane_len + max segments + 64) f32, tail_vals >= max segments f32
-void wfa_ffat_cb_fold2(
+// fused two-stage CB fold: waves compute pane partials into LDS, wave 0
+// runs the window machine per chunk — one launch, no global pane traffic
+void wfa_ffat_cb_fold_fused(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
     const int64_t* d_nseg, int64_t n, const void* v_f32, int vdt,
     const uint32_t* idx_sorted, const int64_t* ts_orig, int64_t pane_len,
     int64_t P, int64_t S, int comb, int ring_log2, int64_t* st_count,
     uint32_t* st_fill, float* st_acc, float* ring, uint32_t* st_head,
     float* st_wsum, const uint64_t* slot_to_key, const uint32_t* fire_base,
-    uint32_t* pane_base, float* pane_vals, float* tail_vals, int64_t* d_scratch,
     uint64_t* out_key, float* out_val, int64_t* out_ts, int64_t out_cap);
 
 // MFMA windowed Gram aggregation: per-key tumbling windows over 16-dim
