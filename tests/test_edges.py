@@ -103,3 +103,57 @@ def test_flatmap_zero_and_many():
     g.run()
     exp = 3 * sum(v for v in range(0, 100) if v % 20 == 0)
     assert g.sink_sum(snk) == exp
+
+
+def test_multi_column_mixed_dtypes_through_ops():
+    """3 payload columns (i64, f64, f32) through python source -> map ->
+    filter -> split -> sinks: per-column values stay aligned per row."""
+    import numpy as np
+    n_batches, bsz = 20, 500
+    state = dict(i=0)
+
+    def src(replica, parallelism):
+        if state['i'] >= n_batches:
+            return None
+        base = state['i'] * bsz
+        state['i'] += 1
+        idx = np.arange(base, base + bsz)
+        return {"ts": (idx + 1).astype(np.int64),
+                "key": (idx % 7).astype(np.uint64),
+                "c0": idx.astype(np.int64),
+                "c1": (idx * 0.5).astype(np.float64),
+                "c2": (idx * 2.0).astype(np.float32),
+                "watermark": int(base + bsz)}
+
+    def mapper(cols):
+        cols['c1'][:] = cols['c1'] * 2.0          # f64: now == c0
+        cols['c2'][:] = cols['c2'] + 1.0          # f32: now == 2*c0+1
+
+    def fltr(cols):
+        return (cols['c0'] % 3 == 0)
+
+    rows = []
+
+    def snk(cols):
+        for i in range(len(cols['c0'])):
+            rows.append((int(cols['c0'][i]), float(cols['c1'][i]),
+                         float(cols['c2'][i]), int(cols['key'][i])))
+
+    g = wf.PipeGraph("mc")
+    mp = g.add_source(wf.Source_Builder(src).withParallelism(1)
+                      .withOutputSchema([0, 1, 2]).build())
+    mp.add(wf.Map_Builder(mapper).withParallelism(2)
+           .withOutputSchema([0, 1, 2]).build())
+    mp.add(wf.Filter_Builder(fltr).withParallelism(2)
+           .withOutputSchema([0, 1, 2]).build())
+    br = mp.split(lambda cols: (cols['c0'] % 2).astype(np.int32), 2)
+    s0 = wf.Sink_Builder(snk).withParallelism(1).build()
+    s1 = wf.Sink_Builder(snk).withParallelism(1).build()
+    br.select(0).add_sink(s0)
+    br.select(1).add_sink(s1)
+    g.run()
+
+    total = n_batches * bsz
+    exp = [(v, float(v), float(np.float32(2 * v + 1)), v % 7)
+           for v in range(total) if v % 3 == 0]
+    assert sorted(rows) == sorted(exp)
