@@ -173,13 +173,17 @@ struct DupFlatMapI64 : OpLogic {
 // (reference: wf/reduce.hpp — KEYBY routing, per-key state in replica map)
 struct KeyedSumReduceI64 : OpLogic {
     int col;
+    int64_t init;
     std::unordered_map<uint64_t, int64_t> acc;
-    explicit KeyedSumReduceI64(int c) : col(c) { acc.reserve(1 << 12); }
+    explicit KeyedSumReduceI64(int c, int64_t init_ = 0) : col(c), init(init_) {
+        acc.reserve(1 << 12);
+    }
     void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
         Batch* o = out.new_batch();
         int64_t* x = b->col<int64_t>(col);
         for (int64_t i = 0; i < b->count; ++i) {
-            int64_t& a = acc[b->key[i]];
+            auto [it, fresh] = acc.try_emplace(b->key[i], init);
+            int64_t& a = it->second;
             a += x[i];
             if (o->count == o->capacity) {
                 o->watermark = ctx.current_wm;
@@ -334,7 +338,8 @@ std::shared_ptr<OpLogic> make_native_logic(const std::string& kind, const std::s
     if (kind == "flatmap" && spec == "dup_i64")
         return std::make_shared<DupFlatMapI64>(ip[0]);
     if (kind == "reduce" && spec == "sum_by_key_i64")
-        return std::make_shared<KeyedSumReduceI64>((int)ip[0]);
+        return std::make_shared<KeyedSumReduceI64>((int)ip[0],
+                                                   ip.size() > 1 ? ip[1] : 0);
     if (kind == "sink" && spec == "sum_i64")
         return std::make_shared<SumSinkI64>(eng, op_id, (int)ip[0]);
     if (kind == "sink" && spec == "last_per_key_i64")

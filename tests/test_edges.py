@@ -157,3 +157,34 @@ def test_multi_column_mixed_dtypes_through_ops():
     exp = [(v, float(v), float(np.float32(2 * v + 1)), v % 7)
            for v in range(total) if v % 3 == 0]
     assert sorted(rows) == sorted(exp)
+
+
+def test_reduce_initial_state_and_broadcast():
+    """withInitialState seeds every key's accumulator; withBroadcast routes
+    all inputs to every replica (reference builders.hpp:252/:627)."""
+    n, keys, init = 3000, 5, 1000
+    g = wf.PipeGraph("ribc")
+    mp = g.add_source(wf.Source_Builder(native.seq_source(n, keys, 256))
+                      .withParallelism(1).withOutputSchema([0]).build())
+    red = (wf.Reduce_Builder(native.keyed_sum_reduce(0)).withInitialState(init)
+           .withParallelism(2).withOutputSchema([0]).build())
+    mp.add(red)
+    snk = (wf.Sink_Builder(native.last_per_key_sink(0))
+           .withParallelism(1).build())
+    mp.add_sink(snk)
+    g.run()
+    per = {k: init for k in range(keys)}
+    for v in range(1, n + 1):
+        per[v % keys] += v
+    assert g.sink_sum(snk) == sum(per.values())
+
+    # broadcast: every map replica sees every tuple -> count multiplies
+    g2 = wf.PipeGraph("bc")
+    mp2 = g2.add_source(wf.Source_Builder(native.seq_source(n, keys, 256))
+                        .withParallelism(1).withOutputSchema([0]).build())
+    mp2.add(wf.Map_Builder(native.affine_map(0, 1, 0)).withBroadcast()
+            .withParallelism(3).withOutputSchema([0]).build())
+    snk2 = wf.Sink_Builder(native.count_sink()).withParallelism(1).build()
+    mp2.add_sink(snk2)
+    g2.run()
+    assert g2.sink_count(snk2) == 3 * n

@@ -57,6 +57,12 @@ class _BasicBuilder:
         self._op.rebalancing = True
         return self
 
+    def withBroadcast(self):
+        """BROADCAST routing of inputs (reference builders.hpp:252): every
+        replica of this operator receives every input batch."""
+        self._op.broadcast_input = True
+        return self
+
     def build(self):
         op = self._op.clone()
         if self._needs_key and op.key_extractor is None:
@@ -85,6 +91,18 @@ class Reduce_Builder(_BasicBuilder):
     """Keyed running aggregate — KEYBY routing in (reference reduce.hpp:285)."""
     _kind = "reduce"
     _needs_key = True
+
+    def withInitialState(self, v):
+        """Initial per-key accumulator value (reference builders.hpp:627)."""
+        lg = self._op.logic
+        if not (isinstance(lg, NativeLogic) and lg.kind == "reduce"):
+            raise TypeError("withInitialState applies to native keyed reduces")
+        ip = list(lg.iparams)
+        while len(ip) < 2:
+            ip.append(0)
+        ip[1] = int(v)
+        self._op.logic = NativeLogic(lg.kind, lg.spec, list(lg.fparams), ip)
+        return self
 
 
 class Sink_Builder(_BasicBuilder):
