@@ -185,6 +185,8 @@ struct WindowCore : OpLogic {
         int64_t last_ts = 0;
         int64_t cum_i = 0;      // CB sum prefix (prefix mode)
         double cum_f = 0;
+        int64_t cum_cnt = 0;    // wedge mode: prefix count of (mine) tuples
+        std::deque<std::pair<int64_t, ValU>> wd;  // monotonic wedge (idx, v)
         std::deque<Open> open;
     };
     std::unordered_map<uint64_t, KeyState> keys;
@@ -205,6 +207,9 @@ struct WindowCore : OpLogic {
     // CB + sum/count: window result = prefix@fire - prefix@open, O(1) per
     // tuple instead of one add per open window (win/slide adds)
     bool prefix = false;
+    // CB + min/max: monotonic wedge (sliding-window minimum), O(1)
+    // amortized per tuple; window result = wedge front at fire
+    bool wedge = false;
 
     WindowCore(WinType wt_, int64_t w, int64_t s, int64_t lat, AggCfg a, int c,
                int own, bool meta, bool start_ts, bool rr, Engine* e)
@@ -212,6 +217,7 @@ struct WindowCore : OpLogic {
           emit_meta(meta), emit_start_ts(start_ts), subset_rr(rr), eng(e) {
         prefix = wt == WinType::CB &&
                  (agg.comb == C_SUM || agg.comb == C_COUNT);
+        wedge = wt == WinType::CB && (agg.comb == C_MIN || agg.comb == C_MAX);
     }
 
     void warm(RuntimeCtx& ctx) override {
@@ -249,6 +255,7 @@ struct WindowCore : OpLogic {
                 w.acc.i = ks.cum_i;
                 w.acc.f = ks.cum_f;
             }
+            if (wedge) w.acc.cnt = ks.cum_cnt;  // count@open (empty detection)
             ks.open.push_back(w);
             ks.next_gwid++;
         }
@@ -256,6 +263,23 @@ struct WindowCore : OpLogic {
             if (mine) {  // COUNT runs the prefix over 1s (subset-correct)
                 ks.cum_i += agg.comb == C_COUNT ? 1 : v.i;
                 ks.cum_f += agg.comb == C_COUNT ? 1.0 : v.f;
+            }
+        } else if (wedge) {
+            if (mine) {
+                auto& wd = ks.wd;
+                if (agg.use_int) {
+                    while (!wd.empty() &&
+                           (agg.comb == C_MIN ? wd.back().second.i >= v.i
+                                              : wd.back().second.i <= v.i))
+                        wd.pop_back();
+                } else {
+                    while (!wd.empty() &&
+                           (agg.comb == C_MIN ? wd.back().second.f >= v.f
+                                              : wd.back().second.f <= v.f))
+                        wd.pop_back();
+                }
+                wd.emplace_back(ks.idx, v);
+                ks.cum_cnt++;
             }
         } else {
             for (auto& w : ks.open)
@@ -265,10 +289,22 @@ struct WindowCore : OpLogic {
         while (!ks.open.empty() && ks.idx == ks.open.front().start + win - 1) {
             Open& w = ks.open.front();
             if (prefix) materialize_prefix(ks, w);
+            if (wedge) materialize_wedge(ks, w);
             fire(w, key, ts, out);
             ks.open.pop_front();
         }
         ks.idx++;
+    }
+
+    void materialize_wedge(KeyState& ks, Open& w) {
+        while (!ks.wd.empty() && ks.wd.front().first < w.start) ks.wd.pop_front();
+        Acc a;
+        a.cnt = ks.cum_cnt - w.acc.cnt;
+        if (a.cnt > 0 && !ks.wd.empty()) {
+            a.i = ks.wd.front().second.i;
+            a.f = ks.wd.front().second.f;
+        }
+        w.acc = a;
     }
 
     // turn (prefix@open stored in acc) into the actual window accumulator
@@ -400,6 +436,7 @@ struct WindowCore : OpLogic {
                     a.cnt = agg.comb == C_COUNT ? a.i : ks.idx - w.start;
                     w.acc = a;
                 }
+                if (wedge && wt == WinType::CB) materialize_wedge(ks, w);
                 fire(w, key,
                      wt == WinType::CB ? ks.last_ts
                                        : (emit_start_ts ? w.start
