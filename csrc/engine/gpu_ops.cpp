@@ -874,6 +874,7 @@ struct GpuFfatLogic : GpuLogicBase {
     uint32_t* tb_flags = nullptr;  // [ignored, overflow]
     uint32_t* tb_nf = nullptr;
     uint32_t* h_flags = nullptr;   // pinned readback
+    Engine* eng_ = nullptr;        // central dropped-tuple accounting
     int64_t batches = 0;
     bool use_pane2 = false;
     // A/B on MI355X (8M-tuple batches): the two-stage fold wins when
@@ -947,6 +948,7 @@ struct GpuFfatLogic : GpuLogicBase {
         destroy_graphs();
     }
 
+    uint32_t dropped_seen = 0;
     void check_tb_flags() {
         HIPCHK(hipMemcpyAsync(h_flags, tb_flags, 8, hipMemcpyDeviceToHost, stream));
         HIPCHK(hipStreamSynchronize(stream));
@@ -954,6 +956,13 @@ struct GpuFfatLogic : GpuLogicBase {
             throw std::runtime_error(
                 "gpu_ffat TB pending-pane ring overflow: raise pend_ring_log2 "
                 "or lateness is too large for the configured ring");
+        // surface device-side late-tuple drops in the engine's central
+        // counter (reference PipeGraph::getNumDroppedTuples)
+        if (eng_ && h_flags[0] > dropped_seen) {
+            eng_->dropped_tuples.fetch_add(h_flags[0] - dropped_seen,
+                                           std::memory_order_relaxed);
+            dropped_seen = h_flags[0];
+        }
     }
 
     // hipGraph capture of the steady-state per-batch chain (slot -> sort ->
@@ -1569,13 +1578,16 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
         // ip: [comb, vcol, max_keys]
         return std::make_shared<GpuReduceLogic>((int)ip[0], (int)ip[1], ip[2], device,
                                                 os, out_batch);
-    if (kind == "gpu_ffat")
+    if (kind == "gpu_ffat") {
         // ip: [comb, vcol, win, slide, max_keys, use_tree,
         //      wintype(0 CB/1 TB), lateness, pend_ring_log2]
-        return std::make_shared<GpuFfatLogic>(
+        auto l = std::make_shared<GpuFfatLogic>(
             (int)ip[0], (int)ip[1], ip[2], ip[3], ip[4], ip[5] != 0, device, os,
             out_batch, ip.size() > 6 && ip[6] != 0, ip.size() > 7 ? ip[7] : 0,
             ip.size() > 8 ? (int)ip[8] : 0);
+        l->eng_ = eng;
+        return l;
+    }
     if (kind == "gpu_gram")
         // ip: [win, max_keys]
         return std::make_shared<GpuGramLogic>(ip[0], ip[1], device, os, out_batch);
