@@ -237,6 +237,12 @@ struct WindowCore : OpLogic {
         ValU r = agg.result(w.acc);
         if (emit_meta) {
             o->col<int64_t>(0)[i] = w.gwid;
+            // AVG partials carry the raw SUM (the slice count rides in c2);
+            // emitting the slice MEAN would make the recombiner divide twice
+            if (agg.comb == C_AVG) {
+                r.i = w.acc.i;
+                r.f = w.acc.f;
+            }
             write_val(o, 1, i, r, agg.use_int);
             // tuple count of the slice: empty partials (cnt 0) must not
             // poison MIN/MAX downstream and AVG needs real weights

@@ -394,3 +394,34 @@ def test_py_window_paned_and_mapreduce_fallback():
                          stream_len=1800)
         assert got_counter(rows) == oracle_cb(seq_stream(1800, 7), win, slide,
                                               "sum"), B.__name__
+
+
+def test_avg_through_all_forms():
+    """AVG with real slice weights on every decomposed form: the partials
+    wire carries (value, slice count), so pane/subset recombination weights
+    partial averages correctly (sum-of-sums / sum-of-counts)."""
+    win, slide, n, keys = 40, 10, 3000, 7
+    per = seq_stream(n, keys)
+    exp = {}
+    for k, r in per.items():
+        vals = [v for _, v in r]
+        outs = []
+        w = 0
+        while w * slide < len(vals):
+            seg = vals[w * slide: w * slide + win]
+            outs.append(round(sum(seg) / len(seg), 6))
+            w += 1
+        exp[k] = sorted(outs)
+    for B in (Keyed_Windows_Builder, Parallel_Windows_Builder,
+              Paned_Windows_Builder, MapReduce_Windows_Builder):
+        kw = {"plq_func": ("avg", 0)} if B is Paned_Windows_Builder else (
+             {"map_func": ("avg", 0)} if B is MapReduce_Windows_Builder else
+             {"func": ("avg", 0)})
+        rows = run_graph(B(**kw).withCBWindows(win, slide)
+                         .withParallelism(3).withOutputSchema([1]).build(),
+                         stream_len=n, n_keys=keys)
+        got = {}
+        for _, k, v in rows:
+            got.setdefault(k, []).append(round(float(v), 6))
+        got = {k: sorted(v) for k, v in got.items()}
+        assert got == exp, B.__name__
