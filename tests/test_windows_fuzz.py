@@ -126,3 +126,32 @@ def test_py_window_fn_fuzz(case):
                 exp[(k, round(fn(seg) if seg else 0.0, 6))] += 1
     got = Counter((k, round(v, 6)) for _, k, v in rows)
     assert got == exp, (form, wt, win, slide, keys, par)
+
+
+@pytest.mark.parametrize("case", range(6))
+def test_asymmetric_stage_parallelism_fuzz(case):
+    """Paned/MapReduce with different per-stage parallelism degrees
+    (withPLQ/WLQ/MAP/REDUCEParallelism) vs oracle (80-config campaign ran
+    clean)."""
+    from test_windows import run_graph, seq_stream, oracle_cb, oracle_tb, got_counter
+    rng = random.Random(8800 + case * 13)
+    form = rng.choice(["paned", "mapreduce"])
+    agg = rng.choice(["sum", "max", "min", "count"])
+    wt = rng.choice(["cb", "tb"])
+    slide = rng.choice([5, 10, 25])
+    win = slide * rng.randint(1, 6)
+    keys = rng.choice([1, 3, 11])
+    p1, p2 = rng.randint(1, 4), rng.randint(1, 4)
+    stream = rng.choice([1000, 3000])
+    if form == "paned":
+        b = (Paned_Windows_Builder(plq_func=(agg, 0))
+             .withPLQParallelism(p1).withWLQParallelism(p2))
+    else:
+        b = (MapReduce_Windows_Builder(map_func=(agg, 0))
+             .withMAPParallelism(p1).withREDUCEParallelism(p2))
+    b = b.withCBWindows(win, slide) if wt == "cb" else b.withTBWindows(win, slide)
+    rows = run_graph(b.withOutputSchema([0]).build(), stream_len=stream,
+                     n_keys=keys)
+    per = seq_stream(stream, keys)
+    exp = (oracle_cb if wt == "cb" else oracle_tb)(per, win, slide, agg)
+    assert got_counter(rows) == exp, (form, agg, wt, win, slide, keys, p1, p2)
