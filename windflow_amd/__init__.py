@@ -28,6 +28,7 @@ from .builders import (  # noqa: F401,E402
 from .pipegraph import PipeGraph, MultiPipe  # noqa: F401,E402
 from . import native  # noqa: F401,E402
 from .persistent import (  # noqa: F401,E402
-    P_Reduce_Builder, P_Map_Builder, P_Filter_Builder, P_Sink_Builder)
+    P_Reduce_Builder, P_Map_Builder, P_Filter_Builder, P_Sink_Builder,
+    P_FlatMap_Builder, P_Keyed_Windows_Builder)
 from .kafka import (  # noqa: F401,E402
     Connector_Source_Builder, Kafka_Source_Builder, Kafka_Sink_Builder)
