@@ -43,8 +43,22 @@ static py::dict batch_views(Batch* b, RuntimeCtx* ctx = nullptr) {
 }
 
 // copy a python dict-of-arrays into fresh batches (splitting on capacity)
-static void emit_pydict(py::dict out_d, EmitCtx& out, int64_t wm) {
-    if (out_d.size() == 0) return;
+// Rows grabbed from a python dict UNDER the GIL, emitted WITHOUT it.
+// Holding the GIL across a blocking queue push (or pool get) deadlocks a
+// pipeline of python operators once queues fill: the downstream python
+// stage needs the GIL to drain the very queue the upstream is pushing to
+// (found by the WFA_QUEUE_CAP=2 backpressure campaign).
+struct PyRows {
+    std::vector<int64_t> ts;
+    std::vector<uint64_t> key;
+    std::vector<std::vector<char>> cols;
+    int64_t n = 0;
+    bool has_ts = false, has_key = false;
+};
+
+static PyRows grab_pydict(py::dict out_d) {  // caller holds the GIL
+    PyRows r;
+    if (out_d.size() == 0) return r;
     py::array ts = out_d.contains("ts") ? out_d["ts"].cast<py::array>() : py::array();
     py::array key = out_d.contains("key") ? out_d["key"].cast<py::array>() : py::array();
     std::vector<py::array> cols;
@@ -53,30 +67,45 @@ static void emit_pydict(py::dict out_d, EmitCtx& out, int64_t wm) {
         if (!out_d.contains(nm.c_str())) break;
         cols.push_back(out_d[nm.c_str()].cast<py::array>());
     }
-    int64_t n = 0;
-    if (cols.size()) n = cols[0].shape(0);
-    else if (ts.ndim()) n = ts.shape(0);
+    if (cols.size()) r.n = cols[0].shape(0);
+    else if (ts.ndim()) r.n = ts.shape(0);
+    if (ts.ndim() == 1) {
+        auto a = ts.cast<py::array_t<int64_t>>();
+        r.ts.assign(a.data(), a.data() + r.n);
+        r.has_ts = true;
+    }
+    if (key.ndim() == 1) {
+        auto a = key.cast<py::array_t<uint64_t>>();
+        r.key.assign(a.data(), a.data() + r.n);
+        r.has_key = true;
+    }
+    for (auto& c : cols) {
+        py::buffer_info bi = c.request();
+        r.cols.emplace_back((char*)bi.ptr,
+                            (char*)bi.ptr + (size_t)r.n * bi.itemsize);
+    }
+    return r;
+}
+
+static void emit_rows(const PyRows& r, EmitCtx& out, int64_t wm) {  // no GIL
     int64_t done = 0;
-    while (done < n) {
+    while (done < r.n) {
         Batch* o = out.new_batch();
-        int64_t take = std::min<int64_t>(o->capacity, n - done);
-        if (ts.ndim() == 1) {
-            auto a = ts.cast<py::array_t<int64_t>>();
-            memcpy(o->ts, a.data() + done, take * 8);
-        } else {
+        int64_t take = std::min<int64_t>(o->capacity, r.n - done);
+        if (r.has_ts)
+            memcpy(o->ts, r.ts.data() + done, take * 8);
+        else
             for (int64_t i = 0; i < take; ++i) o->ts[i] = done + i;
-        }
-        if (key.ndim() == 1) {
-            auto a = key.cast<py::array_t<uint64_t>>();
-            memcpy(o->key, a.data() + done, take * 8);
-        } else {
+        if (r.has_key)
+            memcpy(o->key, r.key.data() + done, take * 8);
+        else
             memset(o->key, 0, take * 8);
-        }
-        for (size_t c = 0; c < cols.size() && c < o->schema.payload.size(); ++c) {
+        for (size_t c = 0; c < r.cols.size() && c < o->schema.payload.size(); ++c) {
             size_t es = dsize(o->schema.payload[c]);
-            py::buffer_info bi = cols[c].request();
-            if ((size_t)bi.itemsize != es) throw std::runtime_error("dtype mismatch on c" + std::to_string(c));
-            memcpy(o->cols[c], (char*)bi.ptr + done * es, take * es);
+            size_t have = r.cols[c].size() / (r.n ? r.n : 1);
+            if (have != es)
+                throw std::runtime_error("dtype mismatch on c" + std::to_string(c));
+            memcpy(o->cols[c], r.cols[c].data() + done * es, take * es);
         }
         o->count = take;
         o->watermark = wm;
@@ -111,20 +140,28 @@ struct PyTransformLogic : OpLogic {  // returns new column dict (map/flatmap)
     void process(Batch* b, EmitCtx& out, RuntimeCtx& rctx) override {
         int64_t wm = rctx.current_wm;
         last_wm = wm;
-        py::gil_scoped_acquire gil;
-        py::object r = fn(batch_views(b, &rctx));
-        release(b);
-        if (!r.is_none()) emit_pydict(r.cast<py::dict>(), out, wm);
+        PyRows rows;
+        {
+            py::gil_scoped_acquire gil;
+            py::object r = fn(batch_views(b, &rctx));
+            release(b);
+            if (!r.is_none()) rows = grab_pydict(r.cast<py::dict>());
+        }
+        emit_rows(rows, out, wm);
     }
     void on_eos(EmitCtx& out, RuntimeCtx& ctx) override {
         // stateful python transforms (e.g. store-backed windows) flush their
         // pending results at stream end via fn.on_eos(replica) -> dict; the
         // replica id is passed because the python callable is SHARED across
         // replicas and must only flush the keys this replica owns
-        py::gil_scoped_acquire gil;
-        if (!py::hasattr(fn, "on_eos")) return;
-        py::object r = fn.attr("on_eos")(ctx.replica);
-        if (!r.is_none()) emit_pydict(r.cast<py::dict>(), out, last_wm);
+        PyRows rows;
+        {
+            py::gil_scoped_acquire gil;
+            if (!py::hasattr(fn, "on_eos")) return;
+            py::object r = fn.attr("on_eos")(ctx.replica);
+            if (!r.is_none()) rows = grab_pydict(r.cast<py::dict>());
+        }
+        emit_rows(rows, out, last_wm);
     }
 };
 
@@ -257,25 +294,24 @@ struct PySourceLogic : OpLogic {
     explicit PySourceLogic(py::function f) : fn(std::move(f)) {}
     bool is_source() const override { return true; }
     bool source_step(EmitCtx& out, RuntimeCtx& ctx) override {
-        py::gil_scoped_acquire gil;
-        py::object r = fn(ctx.replica, ctx.parallelism);
-        if (r.is_none()) return false;
-        py::dict d = r.cast<py::dict>();
-        int64_t wm = d.contains("watermark") ? d["watermark"].cast<int64_t>() : 0;
-        if (ctx.engine && ctx.engine->time_policy == TimePolicy::INGRESS_TIME) {
-            // ingress time: stamp arrival clock, ignore user ts/wm
-            int64_t nowus = now_us();
-            auto n = d.contains("c0") ? d["c0"].cast<py::array>().shape(0)
-                                      : (d.contains("ts")
-                                             ? d["ts"].cast<py::array>().shape(0)
-                                             : 0);
-            py::array_t<int64_t> ts({(py::ssize_t)n});
-            for (py::ssize_t i = 0; i < (py::ssize_t)n; ++i)
-                ts.mutable_at(i) = nowus;
-            d["ts"] = ts;
-            wm = nowus;
+        PyRows rows;
+        int64_t wm = 0;
+        {
+            py::gil_scoped_acquire gil;
+            py::object r = fn(ctx.replica, ctx.parallelism);
+            if (r.is_none()) return false;
+            py::dict d = r.cast<py::dict>();
+            wm = d.contains("watermark") ? d["watermark"].cast<int64_t>() : 0;
+            rows = grab_pydict(d);
+            if (ctx.engine && ctx.engine->time_policy == TimePolicy::INGRESS_TIME) {
+                // ingress time: stamp arrival clock, ignore user ts/wm
+                int64_t nowus = now_us();
+                rows.ts.assign((size_t)rows.n, nowus);
+                rows.has_ts = true;
+                wm = nowus;
+            }
         }
-        emit_pydict(d, out, wm);
+        emit_rows(rows, out, wm);
         return true;
     }
 };

@@ -360,3 +360,43 @@ def test_python_split_multi_branch_masks():
     exp0 = sum(v for v in range(1, n + 1) if v % 2 == 0)
     exp1 = sum(v for v in range(1, n + 1) if v % 3 == 0)
     assert [g.sink_sum(s) for s in sinks] == [exp0, exp1]
+
+
+def test_python_pipeline_under_backpressure(monkeypatch):
+    """Queue capacity 2: a chain of python operators must not deadlock.
+    Guards the GIL discipline in the python logic wrappers — holding the
+    GIL across a blocking queue push wedges the pipeline once queues fill
+    (the downstream python stage needs the GIL to drain that queue)."""
+    monkeypatch.setenv("WFA_QUEUE_CAP", "2")
+    stream_len = 4000
+    state = dict(pos=0, total=0)
+
+    def pysource(replica, par):
+        if state['pos'] >= stream_len:
+            return None
+        n = min(256, stream_len - state['pos'])
+        v = np.arange(state['pos'] + 1, state['pos'] + n + 1, dtype=np.int64)
+        state['pos'] += n
+        return dict(c0=v, ts=v, key=(v % 5).astype(np.uint64),
+                    watermark=int(v[-1]))
+
+    def pyflat(cols):
+        v = np.repeat(cols['c0'], 3)
+        return dict(c0=v, ts=np.repeat(cols['ts'], 3),
+                    key=np.repeat(cols['key'], 3))
+
+    def pysink(cols):
+        state['total'] += int(cols['c0'].sum())
+
+    g = wf.PipeGraph("bp")
+    mp = g.add_source(wf.Source_Builder(pysource).withParallelism(1)
+                      .withOutputSchema([0]).withOutputBatchSize(64).build())
+    mp.add(wf.FlatMap_Builder(pyflat).withParallelism(1)
+           .withOutputSchema([0]).withOutputBatchSize(64).build())
+    def ident_map(cols):
+        cols['c0'][:] = cols['c0']
+    mp.add(wf.Map_Builder(ident_map).withParallelism(1)
+           .withOutputSchema([0]).withOutputBatchSize(64).build())
+    mp.add_sink(wf.Sink_Builder(pysink).withParallelism(1).build())
+    g.run()
+    assert state['total'] == 3 * stream_len * (stream_len + 1) // 2
