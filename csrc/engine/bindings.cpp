@@ -617,6 +617,10 @@ PYBIND11_MODULE(_core, m) {
              }),
              py::arg("path"), py::arg("cache_capacity") = 1 << 16,
              py::arg("fresh") = true)
+        .def("erase",
+             [](StateStore& s, uint64_t key) {
+                 return state_kv_erase(s.kv, s.cache, key);
+             })
         .def("get",
              [](StateStore& s, uint64_t key) -> py::object {
                  std::string* v = state_cache_get(s.cache, key);

@@ -447,6 +447,7 @@ std::string* state_cache_get(void* cache, uint64_t key);
 void state_cache_put(void* cache, uint64_t key, const std::string& v);
 void state_cache_flush(void* cache);
 int64_t state_kv_size(void* kv);
+bool state_kv_erase(void* kv, void* cache, uint64_t key);
 
 // debug hooks (gpu_ops.cpp): device sort / hash-slot round trips
 std::pair<std::vector<uint32_t>, std::vector<uint32_t>> debug_sort_pairs_host(

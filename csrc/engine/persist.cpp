@@ -64,6 +64,15 @@ struct VarKV {
             uint32_t len;
             memcpy(&key, hdr, 8);
             memcpy(&len, hdr + 8, 4);
+            if (len == 0xFFFFFFFFu) {  // tombstone: key erased
+                auto it = index.find(key);
+                if (it != index.end()) {
+                    live -= it->second.len;
+                    index.erase(it);
+                }
+                off += 12;
+                continue;
+            }
             // torn tail record (crash mid-write): ignore it and let the
             // next put() overwrite from here
             if (off + 12 + len > fsize) break;
@@ -117,6 +126,15 @@ struct VarKV {
         if (it == index.end()) return false;
         live -= it->second.len;
         index.erase(it);
+        // tombstone record (len = UINT32_MAX): without it a reopened log
+        // would replay the old put and resurrect the key
+        char hdr[12];
+        uint32_t ts = 0xFFFFFFFFu;
+        memcpy(hdr, &key, 8);
+        memcpy(hdr + 8, &ts, 4);
+        if (pwrite(fd, hdr, 12, (off_t)tail) != 12)
+            throw std::runtime_error("VarKV: tombstone write failed");
+        tail += 12;
         return true;
     }
     size_t size() const { return index.size(); }
@@ -205,6 +223,12 @@ struct LruCache {
         e.pos = order.begin();
         e.dirty = dirty;
         return e.val;
+    }
+    void evict(uint64_t key) {
+        auto it = map.find(key);
+        if (it == map.end()) return;
+        order.erase(it->second.pos);
+        map.erase(it);
     }
     void flush() {
         for (auto& [k, e] : map)
@@ -304,5 +328,10 @@ void state_cache_put(void* cache, uint64_t key, const std::string& v) {
 }
 void state_cache_flush(void* cache) { ((LruCache*)cache)->flush(); }
 int64_t state_kv_size(void* kv) { return (int64_t)((VarKV*)kv)->size(); }
+bool state_kv_erase(void* kv, void* cache, uint64_t key) {
+    // drop any cached copy first (a dirty cache entry would re-put it)
+    ((LruCache*)cache)->evict(key);
+    return ((VarKV*)kv)->erase(key);
+}
 
 }  // namespace wfa

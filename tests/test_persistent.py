@@ -295,3 +295,28 @@ def test_statestore_resume_torn_tail(tmp_path):
     del r
     r2 = _core.StateStore(p, cache_capacity=2, fresh=False)
     assert struct.unpack("<q", r2.get(99))[0] == 7
+
+
+def test_statestore_erase_and_resume(tmp_path):
+    """erase() writes a tombstone: the key stays gone after a reopen
+    (without it, replay would resurrect the old put)."""
+    p = str(tmp_path / "er.log")
+    s = _core.StateStore(p, cache_capacity=4)
+    for k in range(10):
+        s.put(k, struct.pack("<q", k))
+    s.flush()
+    assert s.erase(3) and not s.erase(3)
+    assert s.get(3) is None and len(s) == 9
+    s.flush()
+    del s
+    r = _core.StateStore(p, cache_capacity=4, fresh=False)
+    assert len(r) == 9 and r.get(3) is None
+    for k in range(10):
+        if k != 3:
+            assert struct.unpack("<q", r.get(k))[0] == k
+    # re-put after erase works and survives another resume
+    r.put(3, struct.pack("<q", 333))
+    r.flush()
+    del r
+    r2 = _core.StateStore(p, cache_capacity=4, fresh=False)
+    assert struct.unpack("<q", r2.get(3))[0] == 333
