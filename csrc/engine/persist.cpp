@@ -224,11 +224,12 @@ struct LruCache {
         e.dirty = dirty;
         return e.val;
     }
-    void evict(uint64_t key) {
+    bool evict(uint64_t key) {  // true if the key was cached
         auto it = map.find(key);
-        if (it == map.end()) return;
+        if (it == map.end()) return false;
         order.erase(it->second.pos);
         map.erase(it);
+        return true;
     }
     void flush() {
         for (auto& [k, e] : map)
@@ -329,9 +330,12 @@ void state_cache_put(void* cache, uint64_t key, const std::string& v) {
 void state_cache_flush(void* cache) { ((LruCache*)cache)->flush(); }
 int64_t state_kv_size(void* kv) { return (int64_t)((VarKV*)kv)->size(); }
 bool state_kv_erase(void* kv, void* cache, uint64_t key) {
-    // drop any cached copy first (a dirty cache entry would re-put it)
-    ((LruCache*)cache)->evict(key);
-    return ((VarKV*)kv)->erase(key);
+    // drop any cached copy first (a dirty cache entry would re-put it);
+    // the key existed if EITHER tier had it (a fresh put may still live
+    // only in the write-back cache)
+    bool cached = ((LruCache*)cache)->evict(key);
+    bool logged = ((VarKV*)kv)->erase(key);
+    return cached || logged;
 }
 
 }  // namespace wfa

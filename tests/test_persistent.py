@@ -320,3 +320,51 @@ def test_statestore_erase_and_resume(tmp_path):
     del r
     r2 = _core.StateStore(p, cache_capacity=4, fresh=False)
     assert struct.unpack("<q", r2.get(3))[0] == 333
+
+
+@pytest.mark.parametrize("seed", range(3))
+def test_statestore_model_fuzz(seed, tmp_path):
+    """Model-based store check vs a python dict: random put/get/erase/flush
+    with mid-stream reopens and forced compaction (12-seed campaign ran
+    clean; caught erase misreporting for dirty-cache-only keys)."""
+    import os as _os
+    import subprocess
+    import sys as _sys
+    code = f"""
+import random, sys
+sys.path.insert(0, {str(ROOT)!r})
+from windflow_amd import _core
+rng = random.Random({seed})
+model = {{}}
+path = {str(tmp_path / 'm.log')!r}
+s = _core.StateStore(path, cache_capacity=rng.choice([1, 4, 64]))
+for step in range(2000):
+    op = rng.random()
+    k = rng.randint(0, 100)
+    if op < 0.45:
+        v = bytes(rng.getrandbits(8) for _ in range(rng.randint(1, 60)))
+        s.put(k, v); model[k] = v
+    elif op < 0.6:
+        assert s.erase(k) == (k in model), (step, k)
+        model.pop(k, None)
+    elif op < 0.85:
+        assert s.get(k) == model.get(k), (step, k)
+    elif op < 0.93:
+        s.flush()
+    else:
+        s.flush(); del s
+        s = _core.StateStore(path, cache_capacity=rng.choice([1, 4, 64]),
+                             fresh=False)
+        assert len(s) == len(model), (step, len(s), len(model))
+s.flush(); del s
+r = _core.StateStore(path, cache_capacity=4, fresh=False)
+assert len(r) == len(model)
+for k, v in model.items():
+    assert r.get(k) == v
+print("MODEL_OK")
+"""
+    env = dict(_os.environ, WFA_KV_COMPACT_BYTES="15000")
+    out = subprocess.run([_sys.executable, "-c", code], env=env,
+                         capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0 and "MODEL_OK" in out.stdout, \
+        (out.stderr or out.stdout)[-500:]
