@@ -400,3 +400,19 @@ def test_python_pipeline_under_backpressure(monkeypatch):
     mp.add_sink(wf.Sink_Builder(pysink).withParallelism(1).build())
     g.run()
     assert state['total'] == 3 * stream_len * (stream_len + 1) // 2
+
+
+def test_three_way_merge():
+    """Merge of three pipes (reference MultiPipe::merge accepts N pipes)."""
+    g = wf.PipeGraph("m3")
+    mps = []
+    for off in (0, 1000000, 2000000):
+        s = (wf.Source_Builder(native.seq_source(3000, 4, 256, value_offset=off))
+             .withParallelism(1).withOutputSchema([0]).build())
+        mps.append(g.add_source(s))
+    mp = mps[0].merge(mps[1], mps[2])
+    snk = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+    mp.add_sink(snk)
+    g.run()
+    exp = sum(sum(range(off + 1, off + 3001)) for off in (0, 1000000, 2000000))
+    assert g.sink_sum(snk) == exp
