@@ -155,3 +155,71 @@ def test_asymmetric_stage_parallelism_fuzz(case):
     per = seq_stream(stream, keys)
     exp = (oracle_cb if wt == "cb" else oracle_tb)(per, win, slide, agg)
     assert got_counter(rows) == exp, (form, agg, wt, win, slide, keys, p1, p2)
+
+
+@pytest.mark.parametrize("case", range(4))
+def test_irregular_ts_tb_fuzz(case):
+    """TB windows over bursty streams: duplicate timestamps, gaps, negative
+    values, python source (80-config campaign ran clean)."""
+    from collections import Counter
+    import numpy as np
+    import windflow_amd as wf
+    from windflow_amd.builders import Keyed_Windows_Builder, Ffat_Windows_Builder
+    rng = random.Random(940_000 + case * 19)
+    n = rng.choice([500, 2000])
+    keys = rng.choice([1, 3, 9])
+    max_gap = rng.choice([1, 4, 30])
+    slide = rng.choice([5, 10, 50])
+    win = slide * rng.randint(1, 5)
+    agg = rng.choice(["sum", "max", "min", "count"])
+    par = rng.randint(1, 3)
+    batch = rng.choice([32, 256])
+    form = rng.choice(["keyed", "ffat"])
+    ts, t = [], 1
+    for _ in range(n):
+        if rng.random() >= 0.3:
+            t += rng.randint(1, max_gap)
+        ts.append(t)
+    key = [rng.randrange(keys) for _ in range(n)]
+    val = [rng.randint(-50, 50) for _ in range(n)]
+    state = dict(pos=0)
+
+    def src(replica, parallelism):
+        p = state['pos']
+        if p >= len(ts):
+            return None
+        q = min(p + batch, len(ts))
+        state['pos'] = q
+        return dict(ts=np.array(ts[p:q], np.int64),
+                    key=np.array(key[p:q], np.uint64),
+                    c0=np.array(val[p:q], np.int64),
+                    watermark=int(ts[q - 1]))
+
+    rows = []
+
+    def snk(cols):
+        for i in range(len(cols['key'])):
+            rows.append((int(cols['key'][i]), int(cols['c0'][i])))
+
+    B = Keyed_Windows_Builder if form == "keyed" else Ffat_Windows_Builder
+    kw = {"func": (agg, 0)} if form == "keyed" else {"comb": (agg, 0)}
+    g = wf.PipeGraph("irr")
+    mp = g.add_source(wf.Source_Builder(src).withParallelism(1)
+                      .withOutputSchema([0]).withOutputBatchSize(batch).build())
+    mp.add(B(**kw).withTBWindows(win, slide).withParallelism(par)
+           .withOutputSchema([0]).build())
+    mp.add_sink(wf.Sink_Builder(snk).withParallelism(1).build())
+    g.run()
+    per = {}
+    for t2, k, v in zip(ts, key, val):
+        per.setdefault(k, []).append((t2, v))
+    F = {"sum": sum, "max": max, "min": min, "count": len}[agg]
+    exp = Counter()
+    for k, r in per.items():
+        tss = sorted(t2 for t2, _ in r)
+        t0, tmax = tss[0], tss[-1]
+        w0 = max(0, -(-(t0 - win + 1) // slide))
+        for w in range(w0, tmax // slide + 1):
+            seg = [v for t2, v in r if w * slide <= t2 < w * slide + win]
+            exp[(k, F(seg) if seg else 0)] += 1
+    assert Counter(rows) == exp, (form, agg, win, slide, keys, max_gap, par)
