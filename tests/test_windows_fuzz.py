@@ -223,3 +223,69 @@ def test_irregular_ts_tb_fuzz(case):
             seg = [v for t2, v in r if w * slide <= t2 < w * slide + win]
             exp[(k, F(seg) if seg else 0)] += 1
     assert Counter(rows) == exp, (form, agg, win, slide, keys, max_gap, par)
+
+
+@pytest.mark.parametrize("case", range(3))
+def test_lateness_disorder_fuzz(case):
+    """Out-of-order streams with lateness >= jitter: every tuple lands in
+    its windows (no drops).  The first window of a key is anchored at its
+    first ARRIVING tuple (reference window_replica initialization); the
+    oracle reflects that (60-config campaign ran clean)."""
+    from collections import Counter
+    import numpy as np
+    import windflow_amd as wf
+    from windflow_amd.builders import Keyed_Windows_Builder
+    rng = random.Random(950_000 + case * 7)
+    n = rng.choice([800, 2500])
+    keys = rng.choice([1, 3, 9])
+    slide = rng.choice([5, 20])
+    win = slide * rng.randint(1, 5)
+    agg = rng.choice(["sum", "max", "count"])
+    par = rng.randint(1, 3)
+    batch = rng.choice([64, 256])
+    J = rng.choice([3, 10, 40])
+    L = J + rng.randint(0, 20)
+    ts = [max(1, i + rng.randint(-J, J)) for i in range(1, n + 1)]
+    key = [rng.randrange(keys) for _ in range(n)]
+    val = [rng.randint(-30, 30) for _ in range(n)]
+    state = dict(pos=0, wm=0)
+
+    def src(replica, parallelism):
+        p = state['pos']
+        if p >= n:
+            return None
+        q = min(p + batch, n)
+        state['pos'] = q
+        state['wm'] = max(state['wm'], max(ts[p:q]) - J)
+        return dict(ts=np.array(ts[p:q], np.int64),
+                    key=np.array(key[p:q], np.uint64),
+                    c0=np.array(val[p:q], np.int64),
+                    watermark=int(state['wm']))
+
+    rows = []
+
+    def snk(cols):
+        for i in range(len(cols['key'])):
+            rows.append((int(cols['key'][i]), int(cols['c0'][i])))
+
+    g = wf.PipeGraph("late")
+    mp = g.add_source(wf.Source_Builder(src).withParallelism(1)
+                      .withOutputSchema([0]).withOutputBatchSize(batch).build())
+    mp.add(Keyed_Windows_Builder(func=(agg, 0)).withTBWindows(win, slide)
+           .withLateness(L).withParallelism(par).withOutputSchema([0]).build())
+    mp.add_sink(wf.Sink_Builder(snk).withParallelism(1).build())
+    g.run()
+    per, first_arrival = {}, {}
+    for t, k, v in zip(ts, key, val):
+        per.setdefault(k, []).append((t, v))
+        first_arrival.setdefault(k, t)
+    F = {"sum": sum, "max": max, "count": len}[agg]
+    exp = Counter()
+    for k, r in per.items():
+        tmax = max(t for t, _ in r)
+        ta = first_arrival[k]
+        w0 = max(0, -(-(ta - win + 1) // slide))
+        for w in range(w0, tmax // slide + 1):
+            seg = [v for t, v in r if w * slide <= t < w * slide + win]
+            exp[(k, F(seg) if seg else 0)] += 1
+    assert Counter(rows) == exp, (agg, win, slide, keys, J, L, par)
