@@ -66,3 +66,44 @@ def test_topology_fuzz(case):
     g.run()
     assert g.sink_sum(snk) == sum(vals)
     assert g.sink_count(snk) == len(vals)
+
+
+@pytest.mark.parametrize("case", range(4))
+def test_chained_topology_fuzz(case):
+    """Same random op chains as test_topology_fuzz but FUSED into one
+    replica thread via .chain()/.chain_sink() (120-config campaign clean)."""
+    import random
+    import windflow_amd as wf
+    from windflow_amd import native
+    rng = random.Random(1_100_000 + case * 31)
+    n = rng.choice([2000, 10000])
+    keys = rng.choice([1, 4, 13])
+    batch = rng.choice([64, 512, 2048])
+    vals = list(range(1, n + 1))
+    g = wf.PipeGraph("chain")
+    mp = g.add_source(wf.Source_Builder(native.seq_source(n, keys, batch))
+                      .withParallelism(1).withOutputSchema([0]).build())
+    for _ in range(rng.randint(1, 6)):
+        k = rng.choice(["map", "filter", "flatmap"])
+        if k == "map":
+            a, c = rng.choice([1, 2, 3]), rng.randint(0, 5)
+            mp.chain(wf.Map_Builder(native.affine_map(0, a, c))
+                     .withOutputSchema([0]).build())
+            vals = [a * v + c for v in vals]
+        elif k == "filter":
+            m = rng.choice([2, 3, 5])
+            c = rng.randint(0, m - 1)
+            ke = rng.random() < 0.5
+            mp.chain(wf.Filter_Builder(native.mod_filter(0, m, c, ke))
+                     .withOutputSchema([0]).build())
+            vals = [v for v in vals if ((v % m == c) == ke)]
+        else:
+            kk = rng.randint(2, 3)
+            mp.chain(wf.FlatMap_Builder(native.dup_flatmap(kk))
+                     .withOutputSchema([0]).build())
+            vals = [v for v in vals for _ in range(kk)]
+    snk = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+    mp.chain_sink(snk)
+    g.run()
+    assert g.sink_sum(snk) == sum(vals)
+    assert g.sink_count(snk) == len(vals)
