@@ -416,3 +416,43 @@ def test_three_way_merge():
     g.run()
     exp = sum(sum(range(off + 1, off + 3001)) for off in (0, 1000000, 2000000))
     assert g.sink_sum(snk) == exp
+
+
+def test_windows_inside_split_branches():
+    """split -> keyed windows per branch -> merge -> sink (reference
+    graph_tests compose windows inside MultiPipe branches)."""
+    from collections import Counter
+    from windflow_amd.builders import Keyed_Windows_Builder
+    import sys as _s, os as _o
+    _s.path.insert(0, _o.path.dirname(__file__))
+    from test_windows import seq_stream, oracle_cb
+    n, keys, win, slide = 4000, 6, 30, 10
+    g = wf.PipeGraph("ws")
+    mp = g.add_source(wf.Source_Builder(native.seq_source(n, keys, 256))
+                      .withParallelism(1).withOutputSchema([0]).build())
+    br = mp.split(native.split_mod(0), 2)   # by value parity
+    rows = []
+
+    def snk(cols):
+        for i in range(len(cols['key'])):
+            rows.append((int(cols['key'][i]), int(cols['c0'][i])))
+
+    outs = []
+    for i in range(2):
+        b = br.select(i)
+        b.add(Keyed_Windows_Builder(func=("sum", 0)).withCBWindows(win, slide)
+              .withParallelism(2).withOutputSchema([0]).build())
+        outs.append(b)
+    merged = outs[0].merge(outs[1])
+    merged.add_sink(wf.Sink_Builder(snk).withParallelism(1).build())
+    g.run()
+    # oracle: each branch windows its parity sub-stream independently
+    exp = Counter()
+    for parity in (0, 1):
+        per = {}
+        for v in range(1, n + 1):
+            if v % 2 == parity:
+                per.setdefault(v % keys, []).append((v, v))
+        for (k, s), c in oracle_cb(per, win, slide, "sum").items():
+            exp[(k, s)] += c
+    assert Counter(rows) == exp
