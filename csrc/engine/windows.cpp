@@ -1285,6 +1285,9 @@ std::shared_ptr<OpLogic> make_window_logic(const std::string& kind,
         return std::make_shared<IntervalJoinLogic>((int)geti(0), geti(1), geti(2),
                                                    (int)geti(3), (int)geti(4, geti(3)), eng);
     WinType wt = (WinType)geti(0);
+    // defense in depth: win=0/slide=0 would loop forever opening windows
+    if (kind != "win_mr_reduce" && (geti(1) < 1 || geti(2) < 1))
+        throw std::runtime_error(kind + ": window length and slide must be >= 1");
     if (kind == "win_mr_reduce") {
         AggCfg a{(int)geti(1), geti(2, 1) != 0};
         return std::make_shared<MrReduceLogic>(geti(0, 1), a);

@@ -188,3 +188,21 @@ def test_reduce_initial_state_and_broadcast():
     mp2.add_sink(snk2)
     g2.run()
     assert g2.sink_count(snk2) == 3 * n
+
+
+def test_invalid_window_extents_rejected():
+    """win=0/slide=0 used to HANG the engine (infinite window-open loop);
+    all window builders now reject non-positive extents up front."""
+    import pytest
+    from windflow_amd.builders import Keyed_Windows_Builder, Ffat_Windows_Builder
+    from windflow_amd.persistent import P_Keyed_Windows_Builder
+    from windflow_amd import native_gpu
+    for bad in ((0, 0), (10, 0), (0, 10), (-5, 2), (5, -2)):
+        with pytest.raises(ValueError):
+            Keyed_Windows_Builder(func=("sum", 0)).withCBWindows(*bad)
+        with pytest.raises(ValueError):
+            Ffat_Windows_Builder(comb=("sum", 0)).withTBWindows(*bad)
+        with pytest.raises(ValueError):
+            P_Keyed_Windows_Builder(sum).withCBWindows(*bad)
+        with pytest.raises(ValueError):
+            native_gpu.gpu_ffat_windows(0, 0, bad[0], bad[1])

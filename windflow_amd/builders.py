@@ -117,11 +117,20 @@ class _WindowsBuilder(_BasicBuilder):
         self._op.window = dict(type=WinType.CB, win=0, slide=0, lateness=0,
                                lift=lift, comb=comb)
 
+    @staticmethod
+    def _check_extent(win, slide):
+        if win < 1 or slide < 1:
+            raise ValueError(
+                f"window length and slide must be >= 1 (got win={win}, "
+                f"slide={slide})")
+
     def withCBWindows(self, win_len, slide_len):
+        self._check_extent(int(win_len), int(slide_len))
         self._op.window.update(type=WinType.CB, win=int(win_len), slide=int(slide_len))
         return self
 
     def withTBWindows(self, win_us, slide_us):
+        self._check_extent(int(win_us), int(slide_us))
         self._op.window.update(type=WinType.TB, win=int(win_us), slide=int(slide_us))
         return self
 

@@ -93,6 +93,8 @@ def gpu_ffat_windows(comb=COMB_SUM, col=0, win=1000, slide=100,
     TB (tb=True): event-time windows [w*slide, w*slide+win) on ts; panes
     complete at watermark - lateness; pending out-of-order panes live in a
     2^pend_ring_log2 ring per key (default 2^16)."""
+    if win < 1 or slide < 1:
+        raise ValueError("window length and slide must be >= 1")
     return NativeLogic("gpu_ffat", "", [],
                        [comb, col, win, slide, max_keys, 1 if use_tree else 0,
                         1 if tb else 0, lateness, pend_ring_log2])

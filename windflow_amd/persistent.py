@@ -271,10 +271,14 @@ class P_Keyed_Windows_Builder(_PersistBuilder):
         self._op.window = dict(type=0, win=0, slide=0)
 
     def withCBWindows(self, win, slide):
+        if int(win) < 1 or int(slide) < 1:
+            raise ValueError("window length and slide must be >= 1")
         self._op.window = dict(type=0, win=int(win), slide=int(slide))
         return self
 
     def withTBWindows(self, win_us, slide_us):
+        if int(win_us) < 1 or int(slide_us) < 1:
+            raise ValueError("window length and slide must be >= 1")
         self._op.window = dict(type=1, win=int(win_us), slide=int(slide_us))
         return self
 
