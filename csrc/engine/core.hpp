@@ -89,7 +89,17 @@ struct Batch {
     std::atomic<int> refcnt{1}; // broadcast multicast refcount
     Pool* pool = nullptr;       // recycling home
 
-    template <typename T> T* col(size_t i) { return reinterpret_cast<T*>(cols[i]); }
+    template <typename T> T* col(size_t i) {
+        // bounds-checked: an out-of-range column index (user config error)
+        // must be a clear exception, not a segfault.  Fetched once per
+        // BATCH in every hot loop, so the branch costs nothing.
+        if (i >= cols.size())
+            throw std::runtime_error("column c" + std::to_string(i) +
+                                     " out of range (batch has " +
+                                     std::to_string(cols.size()) +
+                                     " payload columns)");
+        return reinterpret_cast<T*>(cols[i]);
+    }
     size_t n_payload() const { return schema.payload.size(); }
 };
 

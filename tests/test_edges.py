@@ -206,3 +206,19 @@ def test_invalid_window_extents_rejected():
             P_Keyed_Windows_Builder(sum).withCBWindows(*bad)
         with pytest.raises(ValueError):
             native_gpu.gpu_ffat_windows(0, 0, bad[0], bad[1])
+
+
+def test_invalid_configs_rejected_cleanly():
+    """parallelism < 1 is rejected at build; an out-of-range column index
+    is a clean engine error, not a segfault."""
+    import pytest
+    with pytest.raises(ValueError):
+        wf.Source_Builder(native.seq_source(10, 1, 8)).withParallelism(0)
+    g = wf.PipeGraph("bad")
+    mp = g.add_source(wf.Source_Builder(native.seq_source(100, 3, 32))
+                      .withParallelism(1).withOutputSchema([0]).build())
+    mp.add(wf.Map_Builder(native.affine_map(5, 1, 0))
+           .withOutputSchema([0]).build())
+    mp.add_sink(wf.Sink_Builder(native.count_sink()).build())
+    with pytest.raises(RuntimeError, match="out of range"):
+        g.run()

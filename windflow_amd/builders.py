@@ -30,6 +30,8 @@ class _BasicBuilder:
         return self
 
     def withParallelism(self, p):
+        if int(p) < 1:
+            raise ValueError(f"parallelism must be >= 1 (got {p})")
         self._op.parallelism = int(p)
         return self
 
