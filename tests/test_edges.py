@@ -222,3 +222,20 @@ def test_invalid_configs_rejected_cleanly():
     mp.add_sink(wf.Sink_Builder(native.count_sink()).build())
     with pytest.raises(RuntimeError, match="out of range"):
         g.run()
+
+
+def test_merge_schema_and_select_bounds_validated():
+    import pytest
+    g = wf.PipeGraph("v")
+    a = g.add_source(wf.Source_Builder(native.seq_source(100, 3, 32))
+                     .withParallelism(1).withOutputSchema([0]).build())
+    b = g.add_source(wf.Source_Builder(native.seq_source(100, 3, 32))
+                     .withParallelism(1).withOutputSchema([1, 1]).build())
+    with pytest.raises(TypeError, match="identical output schemas"):
+        a.merge(b)
+    g2 = wf.PipeGraph("v2")
+    mp = g2.add_source(wf.Source_Builder(native.seq_source(100, 3, 32))
+                       .withParallelism(1).withOutputSchema([0]).build())
+    br = mp.split(native.split_mod(0), 2)
+    with pytest.raises(IndexError):
+        br.select(5)

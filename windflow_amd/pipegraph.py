@@ -106,6 +106,18 @@ class MultiPipe:
         pipegraph.hpp:308-460 merge shapes).  Tail order tags join
         streams: this pipe's tails are stream A (tag 0), the first merged
         pipe's are stream B (tag 1), ..."""
+        def tail_schema(mp):
+            n, _ = mp.tails[0]
+            return tuple(self.graph.nodes[n].ops[-1].out_schema or [0])
+
+        # the reference enforces identical tuple types at compile time;
+        # merging different schemas here would silently reinterpret columns
+        base = tail_schema(self)
+        for o in others:
+            if tail_schema(o) != base:
+                raise TypeError(
+                    f"merge requires identical output schemas "
+                    f"(got {base} vs {tail_schema(o)})")
         tails = list(self.tails)
         tags = [0] * len(self.tails)
         for i, o in enumerate(others):
@@ -131,6 +143,9 @@ class MultiPipe:
     def select(self, branch):
         """MultiPipe of one split branch (reference multipipe.hpp select)."""
         n, _ = self.tails[0]
+        nb = self.graph.nodes[n].n_branches
+        if not (0 <= int(branch) < nb):
+            raise IndexError(f"select({branch}): split has {nb} branches")
         return MultiPipe(self.graph, [(n, branch)])
 
     def split_gpu(self, n_branches):
