@@ -239,3 +239,13 @@ def test_merge_schema_and_select_bounds_validated():
     br = mp.split(native.split_mod(0), 2)
     with pytest.raises(IndexError):
         br.select(5)
+
+
+def test_unterminated_pipe_rejected():
+    import pytest
+    g = wf.PipeGraph("nt")
+    mp = g.add_source(wf.Source_Builder(native.seq_source(100, 3, 32))
+                      .withParallelism(1).withOutputSchema([0]).build())
+    mp.add(wf.Map_Builder(native.affine_map(0, 1, 0)).withOutputSchema([0]).build())
+    with pytest.raises(RuntimeError, match="no consumers"):
+        g.run()

@@ -251,8 +251,22 @@ class PipeGraph:
                 self._sink_map[id(op)] = eid
         return eid
 
+    def _check_terminated(self):
+        # every leaf must end in a sink (reference PipeGraph::run errors on
+        # un-terminated MultiPipes); otherwise results silently vanish
+        has_out = set(d["src"] for d in self.edges)
+        for i, node in enumerate(self.nodes):
+            if i in has_out or node.n_branches:
+                continue
+            if not any(op.kind == "sink" for op in node.ops):
+                name = node.ops[-1].name or node.ops[-1].kind
+                raise RuntimeError(
+                    f"operator '{name}' has no consumers and is not a sink — "
+                    f"terminate the MultiPipe with add_sink()/chain_sink()")
+
     def build_engine(self):
         import os as _os
+        self._check_terminated()
         e = _core.Engine()
         e.mode = self.mode
         e.time_policy = self.time_policy
