@@ -258,7 +258,7 @@ class PipeGraph:
         for i, node in enumerate(self.nodes):
             if i in has_out or node.n_branches:
                 continue
-            if not any(op.kind == "sink" for op in node.ops):
+            if not any("sink" in op.kind for op in node.ops):
                 name = node.ops[-1].name or node.ops[-1].kind
                 raise RuntimeError(
                     f"operator '{name}' has no consumers and is not a sink — "
