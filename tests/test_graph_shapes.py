@@ -456,3 +456,26 @@ def test_windows_inside_split_branches():
         for (k, s), c in oracle_cb(per, win, slide, "sum").items():
             exp[(k, s)] += c
     assert Counter(rows) == exp
+
+
+def test_gpu_graph_lowering_on_cpu():
+    """All four GPU bench graphs construct and lower (build_engine) on a
+    GPU-less box: validation, window lowering and spec generation are
+    python/host-side and must not regress where there is no device."""
+    import bench
+    from windflow_amd import _core
+    for name, build in (
+            ("ffat", lambda: bench.build_ffat_graph(1_000_000, 250_000, 8192,
+                                                    1000, 100, 0, 1, 0)),
+            ("mapfilter", lambda: bench.build_mapfilter_graph(
+                1_000_000, 250_000, 8192, 0, 1, 0)),
+            ("a2a", lambda: bench.build_a2a_graph(
+                1_000_000, 250_000, 8192, 0, 1, 0,
+                (0, 1, _core.rccl_unique_id()))),
+            ("ffat_x", lambda: bench.build_ffat_x_graph(
+                1_000_000, 250_000, 8192, 1000, 100, 0, 1, 0,
+                (0, 1, _core.rccl_unique_id()))),
+    ):
+        g, snk = build()
+        e = g.build_engine()
+        assert e is not None, name
