@@ -479,3 +479,57 @@ def test_gpu_graph_lowering_on_cpu():
         g, snk = build()
         e = g.build_engine()
         assert e is not None, name
+
+
+def test_composite_split_windows_merge_fuzz():
+    """Randomized composite DAGs: split -> per-branch windows (independent
+    configs, keyed and FFAT forms) -> merge (40-config campaign clean)."""
+    import random
+    from collections import Counter
+    from windflow_amd.builders import Keyed_Windows_Builder, Ffat_Windows_Builder
+    import sys as _s, os as _o
+    _s.path.insert(0, _o.path.dirname(__file__))
+    from test_windows import oracle_cb
+    for case in (3, 17):
+        rng = random.Random(12_000_000 + case)
+        n = rng.choice([2000, 6000])
+        keys = rng.choice([4, 9])
+        nb = rng.choice([2, 3])
+        cfgs = []
+        for i in range(nb):
+            slide = rng.choice([5, 10])
+            win = slide * rng.randint(1, 4)
+            agg = rng.choice(["sum", "max", "count"])
+            par = rng.randint(1, 3)
+            form = rng.choice([Keyed_Windows_Builder, Ffat_Windows_Builder])
+            cfgs.append((win, slide, agg, par, form))
+        g = wf.PipeGraph("comp")
+        mp = g.add_source(wf.Source_Builder(native.seq_source(n, keys, 256))
+                          .withParallelism(1).withOutputSchema([0]).build())
+        br = mp.split(native.split_mod(0), nb)
+        rows = []
+
+        def snk(cols):
+            for i in range(len(cols['key'])):
+                rows.append((int(cols['key'][i]), int(cols['c0'][i])))
+
+        outs = []
+        for i, (win, slide, agg, par, B) in enumerate(cfgs):
+            bsel = br.select(i)
+            kw = ({"func": (agg, 0)} if B is Keyed_Windows_Builder
+                  else {"comb": (agg, 0)})
+            bsel.add(B(**kw).withCBWindows(win, slide).withParallelism(par)
+                     .withOutputSchema([0]).build())
+            outs.append(bsel)
+        merged = outs[0].merge(*outs[1:])
+        merged.add_sink(wf.Sink_Builder(snk).withParallelism(1).build())
+        g.run()
+        exp = Counter()
+        for i, (win, slide, agg, par, B) in enumerate(cfgs):
+            per = {}
+            for v in range(1, n + 1):
+                if v % nb == i:
+                    per.setdefault(v % keys, []).append((v, v))
+            for (k, s), c in oracle_cb(per, win, slide, agg).items():
+                exp[(k, s)] += c
+        assert Counter(rows) == exp, case
