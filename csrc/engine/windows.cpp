@@ -1230,19 +1230,20 @@ struct PyWindowLogic : OpLogic {
             for (;;) {
                 int64_t start = ks.next_gwid * slide, end = start + win;
                 if (start > ks.max_ts) break;  // never open past the key's data
-                bool ready = eos ? !ks.ts.empty() && ks.ts.front() < end
-                                 : end + lateness <= cur_wm;
+                // at EOS every remaining in-range window fires, INCLUDING
+                // empty ones and windows past the front tuple (firing only
+                // windows containing the front lost trailing windows —
+                // found by the python-fn TB fuzz)
+                bool ready = eos || end + lateness <= cur_wm;
                 if (!ready) break;
                 auto lo = std::lower_bound(ks.ts.begin(), ks.ts.end(), start) - ks.ts.begin();
                 auto hi = std::lower_bound(ks.ts.begin(), ks.ts.end(), end) - ks.ts.begin();
-                if (hi > lo || !eos)
-                    fire(key, ks, lo, hi, ks.next_gwid, end - 1, out);
+                fire(key, ks, lo, hi, ks.next_gwid, end - 1, out);
                 ks.next_gwid++;
                 // purge rows older than the next window's start
                 auto nxt = std::lower_bound(ks.ts.begin(), ks.ts.end(),
                                             ks.next_gwid * slide) - ks.ts.begin();
                 purge_front(ks, nxt);
-                if (eos && ks.ts.empty()) break;
             }
         }
     }

@@ -75,7 +75,7 @@ def test_join_fuzz(case):
                                                        batch, par, kp)
 
 
-@pytest.mark.parametrize("case", range(8))
+@pytest.mark.parametrize("case", range(10))
 def test_py_window_fn_fuzz(case):
     """Non-incremental python window functions (median/span/sumsq) on every
     window form vs brute-force oracle (caught the parallel form silently
@@ -95,10 +95,14 @@ def test_py_window_fn_fuzz(case):
              "paned": lambda f: Paned_Windows_Builder(plq_func=f),
              "mapreduce": lambda f: MapReduce_Windows_Builder(map_func=f)}
     rng = random.Random(777 + case * 7)
+    if case >= 8:
+        # the regime that exposed the PyWindowLogic TB EOS bug: key stride
+        # wider than the window -> trailing/empty windows flushed at EOS
+        rng = random.Random(26_000_055 + (case - 8) * 26)
     form = rng.choice(list(BUILD))
     fn = FNS[rng.choice(list(FNS))]
-    wt = rng.choice(["cb", "tb"])
-    slide = rng.choice([5, 10, 20])
+    wt = "tb" if case >= 8 else rng.choice(["cb", "tb"])
+    slide = rng.choice([3, 5, 10, 20] if case >= 8 else [5, 10, 20])
     win = slide * rng.randint(1, 5)
     keys = rng.choice([1, 3, 7])
     par = rng.randint(1, 3)
