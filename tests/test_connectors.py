@@ -97,8 +97,10 @@ def test_kafka_end_to_end_with_mock_client(monkeypatch):
         def __init__(self, conf):
             self.conf = conf
 
-        def produce(self, topic, payload, partition=0):
+        def produce(self, topic, payload, partition=0, on_delivery=None):
             produced.append((topic, partition, payload))
+            if on_delivery is not None:
+                on_delivery(None, None)  # delivered OK
 
         def poll(self, t):
             pass

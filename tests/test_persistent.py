@@ -368,3 +368,48 @@ print("MODEL_OK")
                          capture_output=True, text=True, timeout=120)
     assert out.returncode == 0 and "MODEL_OK" in out.stdout, \
         (out.stderr or out.stdout)[-500:]
+
+
+def test_p_map_keep_state_resumes_and_cleanup(tmp_path):
+    """Python persistent ops honor withKeepState(): a NAMED P_Map's state
+    path is deterministic (user path + op name), keep=True reopens the same
+    log with fresh=False (state survives runs), and without keep the log
+    file is deleted when the graph closes the operator (reference
+    db_handle.hpp deleteDb)."""
+    import glob
+
+    def counting(cols, store):
+        for i in range(len(cols['key'])):
+            k = int(cols['key'][i])
+            prev = store.get(k)
+            c = (struct.unpack("<q", prev)[0] if prev else 0) + 1
+            store.put(k, struct.pack("<q", c))
+            cols['c0'][i] = c
+
+    def run_once(keep):
+        g = wf.PipeGraph("pmk")
+        src = (wf.Source_Builder(native.seq_source(100, 4, 32))
+               .withParallelism(1).withOutputSchema([0]).build())
+        mp = g.add_source(src)
+        b = (P_Map_Builder(counting).withName("counter")
+             .withStatePath(str(tmp_path / "pm")))
+        if keep:
+            b = b.withKeepState()
+        mp.add(b.withParallelism(1).withOutputSchema([0]).build())
+        snk = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+        mp.add_sink(snk)
+        g.run()
+        return g.sink_sum(snk)
+
+    # run 1 with keep: counts 1..25 per key (100 tuples over 4 keys)
+    s1 = run_once(keep=True)
+    assert s1 == 4 * (25 * 26 // 2)
+    logs = glob.glob(str(tmp_path / "pm.pmap.counter.log"))
+    assert logs, "deterministic state path (user path + op name) not found"
+    # run 2 with keep: state RESUMED -> counts 26..50 per key
+    s2 = run_once(keep=True)
+    assert s2 == 4 * (50 * 51 // 2 - 25 * 26 // 2)
+    # run 3 without keep: fresh store (counts restart), log deleted at close
+    s3 = run_once(keep=False)
+    assert s3 == s1
+    assert not glob.glob(str(tmp_path / "pm.pmap.counter.log"))

@@ -157,6 +157,13 @@ class Parallel_Windows_Builder(_WindowsBuilder):
         # (KEYBY); compiled combines use BROADCAST + gwid%n window ownership
         self._op.broadcast_input = not callable(func)
         if callable(func):
+            import warnings
+            warnings.warn(
+                "Parallel_Windows with a Python callable runs on the keyed "
+                "python engine (per-key KEYBY partitioning), not the "
+                "reference's whole-stream BROADCAST decomposition; same "
+                "windows per key, different replica ownership",
+                stacklevel=2)
             self._op.key_extractor = 'carried'
 
 
@@ -168,6 +175,13 @@ class Paned_Windows_Builder(_WindowsBuilder):
     def __init__(self, plq_func=None, wlq_func=None, lift=None, comb=None):
         super().__init__(plq_func, lift, comb)
         self._op.extra['wlq_func'] = wlq_func
+        if callable(plq_func):
+            import warnings
+            warnings.warn(
+                "Paned_Windows with a Python callable runs the full windows "
+                "on the keyed python engine (no PLQ/WLQ pane decomposition): "
+                "same results, key-partitioned instead of pane-partitioned",
+                stacklevel=2)
         self._op.key_extractor = 'carried'  # KEYBY into the PLQ stage
 
     def withPLQParallelism(self, p):
@@ -190,6 +204,13 @@ class MapReduce_Windows_Builder(_WindowsBuilder):
         # round-robin MAP decomposition
         self._op.broadcast_input = not callable(map_func)
         if callable(map_func):
+            import warnings
+            warnings.warn(
+                "MapReduce_Windows with a Python callable runs on the keyed "
+                "python engine (per-key KEYBY partitioning), not the "
+                "reference's round-robin MAP decomposition; same windows per "
+                "key, different replica ownership",
+                stacklevel=2)
             self._op.key_extractor = 'carried'   # KEYBY on the key column
 
     def withMAPParallelism(self, p):

@@ -43,7 +43,21 @@ def cpu_keyby_exchange(n_cols=1):
         keys = np.asarray(cols['key'], dtype=np.uint64)
         dest = (_hash_keys(keys) % np.uint64(world)).astype(np.int64)
         arrays = [np.asarray(cols['ts'], dtype=np.int64), keys.view(np.int64)]
-        arrays += [np.asarray(cols[f'c{i}']) for i in range(n_cols)]
+        # Rows travel as i64 words; each payload column is widened to an
+        # exact 8-byte carrier (f32->f64, i32/u32/u16/u8->i64, f64/u64 bit
+        # views) and restored to its ORIGINAL dtype on receive — round 1
+        # left float payloads as int64 bit patterns (silent corruption).
+        col_dtypes = []
+        for i in range(n_cols):
+            a = np.asarray(cols[f'c{i}'])
+            col_dtypes.append(a.dtype)
+            if a.dtype.itemsize == 8:
+                a = a.view(np.int64)
+            elif np.issubdtype(a.dtype, np.floating):
+                a = a.astype(np.float64).view(np.int64)
+            else:
+                a = a.astype(np.int64)
+            arrays.append(a)
         counts = np.bincount(dest, minlength=world)
         # gloo has no all_to_all: allgather the count matrix, then paired
         # isend/irecv per peer (the RCCL path uses grouped send/recv too)
@@ -76,7 +90,14 @@ def cpu_keyby_exchange(n_cols=1):
         out = {'ts': np.concatenate(out_ts),
                'key': np.concatenate(out_key)}
         for c in range(n_cols):
-            out[f'c{c}'] = np.concatenate(out_cols[c])
+            raw = np.ascontiguousarray(np.concatenate(out_cols[c]))
+            dt = col_dtypes[c]
+            if dt.itemsize == 8:
+                out[f'c{c}'] = raw.view(dt)
+            elif np.issubdtype(dt, np.floating):
+                out[f'c{c}'] = raw.view(np.float64).astype(dt)
+            else:
+                out[f'c{c}'] = raw.astype(dt)
         return out
 
     return exchange

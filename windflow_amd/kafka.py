@@ -77,8 +77,15 @@ class Kafka_Source_Builder(_BasicBuilder):
         from confluent_kafka import Consumer, TopicPartition
         cfg, deser = self._cfg, self._deser
 
-        def source_fn(replica, parallelism):
-            if not hasattr(source_fn, "_consumer"):
+        # One Consumer per source REPLICA (keyed by replica index): librdkafka
+        # consumers are not thread-safe, and a replica whose deser stops must
+        # close only its own consumer (reference kafka_source.hpp creates one
+        # consumer per replica in svc_init).
+        consumers = {}
+
+        def _consumer_for(replica):
+            c = consumers.get(replica)
+            if c is None:
                 c = Consumer({
                     "bootstrap.servers": cfg["brokers"],
                     "group.id": cfg["group"],
@@ -90,13 +97,18 @@ class Kafka_Source_Builder(_BasicBuilder):
                               for (t, p, o) in cfg["offsets"]])
                 else:
                     c.subscribe(cfg["topics"])
-                source_fn._consumer = c
-            msg = source_fn._consumer.poll(cfg["idle_ms"] / 1000.0)
+                consumers[replica] = c
+            return c
+
+        def source_fn(replica, parallelism):
+            c = _consumer_for(replica)
+            msg = c.poll(cfg["idle_ms"] / 1000.0)
             out = {}
             more = deser(msg.value() if msg is not None and not msg.error()
                          else None, out)
             if not more:
-                source_fn._consumer.close()
+                c.close()
+                consumers.pop(replica, None)
                 return None
             return out if out else {}
 
@@ -127,15 +139,35 @@ class Kafka_Sink_Builder(_BasicBuilder):
         from confluent_kafka import Producer
         cfg, ser = self._cfg, self._ser
         prod = Producer({"bootstrap.servers": cfg["brokers"]})
+        failures = []
+
+        def _on_delivery(err, msg):
+            if err is not None:
+                failures.append(err)
 
         def sink_fn(cols):
             n = len(cols["ts"])
             for i in range(n):
                 topic, part, payload = ser(cols, i)
-                prod.produce(topic, payload, partition=part)
+                # BufferError = librdkafka's local queue is full: drain
+                # delivery callbacks and retry instead of losing the row
+                while True:
+                    try:
+                        prod.produce(topic, payload, partition=part,
+                                     on_delivery=_on_delivery)
+                        break
+                    except BufferError:
+                        prod.poll(0.1)
             prod.poll(0)
+            if failures:
+                raise RuntimeError(f"Kafka delivery failed: {failures[0]}")
 
-        sink_fn.on_eos = prod.flush
+        def _flush():
+            prod.flush()
+            if failures:
+                raise RuntimeError(f"Kafka delivery failed: {failures[0]}")
+
+        sink_fn.on_eos = _flush
         op = self._op.clone()
         op.logic = sink_fn
         return op

@@ -13,11 +13,23 @@ Usage:
     P_Map_Builder(fn).withStatePath(...)                          # python:
         fn(cols, store) -> None, with store.get(key)/store.put(key, bytes)
 """
+import itertools
 import os
 
 from . import _core
 from .operators import Operator, NativeLogic
 from .builders import _BasicBuilder
+
+# Deterministic state-file naming: user path + tag + operator name (or a
+# construction-order index when unnamed).  id(op) (used in round 1) changed
+# every run, so withKeepState() could never find its previous log.
+_unnamed_seq = itertools.count()
+
+
+def _state_path(op, tag):
+    base = op.extra['state_path'] or "/tmp/wfa_state"
+    name = op.name if op.name else f"op{next(_unnamed_seq)}"
+    return f"{base}.{tag}.{name}.log"
 
 
 class _PersistBuilder(_BasicBuilder):
@@ -61,15 +73,26 @@ class P_Reduce_Builder(_PersistBuilder):
 
 class _PyStateful:
     """Wraps a user fn(cols, store) into a per-batch map callable holding a
-    StateStore; flushed when the graph closes the operator."""
+    StateStore; flushed (and, unless keep, deleted — reference db_handle.hpp
+    deleteDb semantics) when the graph closes the operator."""
 
-    def __init__(self, fn, path, cache_cap, filter_mode=False):
+    def __init__(self, fn, path, cache_cap, filter_mode=False, keep=False):
         self.fn = fn
-        self.store = _core.StateStore(path, cache_cap)
+        self.path = path
+        self.keep = keep
+        self.store = _core.StateStore(path, cache_cap, fresh=not keep)
         self.filter_mode = filter_mode
 
     def __call__(self, cols):
         return self.fn(cols, self.store)
+
+    def close(self):
+        self.store.flush()
+        if not self.keep:
+            try:
+                os.unlink(self.path)
+            except OSError:
+                pass
 
 
 class P_Map_Builder(_PersistBuilder):
@@ -79,10 +102,11 @@ class P_Map_Builder(_PersistBuilder):
 
     def build(self):
         op = super().build()
-        path = (op.extra['state_path'] or "/tmp/wfa_state") + f".pmap.{id(op)}.log"
-        wrapped = _PyStateful(op.logic, path, op.extra['cache_capacity'])
+        path = _state_path(op, "pmap")
+        wrapped = _PyStateful(op.logic, path, op.extra['cache_capacity'],
+                              keep=op.extra['keep'])
         op.logic = wrapped
-        op.closing = wrapped.store.flush
+        op.closing = wrapped.close
         return op
 
 
@@ -92,10 +116,11 @@ class P_Filter_Builder(_PersistBuilder):
 
     def build(self):
         op = super().build()
-        path = (op.extra['state_path'] or "/tmp/wfa_state") + f".pfil.{id(op)}.log"
-        wrapped = _PyStateful(op.logic, path, op.extra['cache_capacity'], True)
+        path = _state_path(op, "pfil")
+        wrapped = _PyStateful(op.logic, path, op.extra['cache_capacity'], True,
+                              keep=op.extra['keep'])
         op.logic = wrapped
-        op.closing = wrapped.store.flush
+        op.closing = wrapped.close
         return op
 
 
@@ -105,10 +130,11 @@ class P_Sink_Builder(_PersistBuilder):
 
     def build(self):
         op = super().build()
-        path = (op.extra['state_path'] or "/tmp/wfa_state") + f".psnk.{id(op)}.log"
-        wrapped = _PyStateful(op.logic, path, op.extra['cache_capacity'])
+        path = _state_path(op, "psnk")
+        wrapped = _PyStateful(op.logic, path, op.extra['cache_capacity'],
+                              keep=op.extra['keep'])
         op.logic = wrapped
-        op.closing = wrapped.store.flush
+        op.closing = wrapped.close
         return op
 
 
@@ -119,10 +145,11 @@ class P_FlatMap_Builder(_PersistBuilder):
 
     def build(self):
         op = super().build()
-        path = (op.extra['state_path'] or "/tmp/wfa_state") + f".pfm.{id(op)}.log"
-        wrapped = _PyStateful(op.logic, path, op.extra['cache_capacity'])
+        path = _state_path(op, "pfm")
+        wrapped = _PyStateful(op.logic, path, op.extra['cache_capacity'],
+                              keep=op.extra['keep'])
         op.logic = wrapped
-        op.closing = wrapped.store.flush
+        op.closing = wrapped.close
         return op
 
 
@@ -134,11 +161,13 @@ class _PKeyedWin:
     passage; incomplete windows flush at EOS (same semantics as the in-memory
     keyed windows, csrc/engine/windows.cpp)."""
 
-    def __init__(self, fn, path, cache_cap, wintype, win, slide):
+    def __init__(self, fn, path, cache_cap, wintype, win, slide, keep=False):
         import numpy as _np
         self.np = _np
         self.fn = fn
-        self.store = _core.StateStore(path, cache_cap)
+        self.path = path
+        self.keep = keep
+        self.store = _core.StateStore(path, cache_cap, fresh=not keep)
         self.tb = wintype == 1
         self.win, self.slide = int(win), int(slide)
         self.keys = {}              # replica -> keys with pending state
@@ -260,6 +289,14 @@ class _PKeyedWin:
         self.store.flush()
         return self._pack(out)
 
+    def close(self):
+        self.store.flush()
+        if not self.keep:
+            try:
+                os.unlink(self.path)
+            except OSError:
+                pass
+
 
 class P_Keyed_Windows_Builder(_PersistBuilder):
     """Keyed windows whose per-key archives live in the persistent store
@@ -286,8 +323,10 @@ class P_Keyed_Windows_Builder(_PersistBuilder):
         op = super().build()
         w = op.window
         op.window = None  # handled here, not by the graph's window lowering
-        path = (op.extra['state_path'] or "/tmp/wfa_state") + f".pkw.{id(op)}.log"
+        path = _state_path(op, "pkw")
         wrapped = _PKeyedWin(op.logic, path, op.extra['cache_capacity'],
-                             w['type'], w['win'], w['slide'])
+                             w['type'], w['win'], w['slide'],
+                             keep=op.extra['keep'])
         op.logic = wrapped
+        op.closing = wrapped.close
         return op
