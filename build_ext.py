@@ -45,6 +45,7 @@ ENGINE_SRCS = [
     "csrc/engine/windows.cpp",
     "csrc/engine/persist.cpp",
     "csrc/engine/gpu_ops.cpp",
+    "csrc/engine/gpu_jit.cpp",
     "csrc/engine/bindings.cpp",
 ]
 HIP_SRCS = [
@@ -127,7 +128,7 @@ def build_stress(tsan=False):
             ["csrc/tests/engine_stress.cpp", "csrc/engine/engine.cpp",
              "csrc/engine/core.cpp", "csrc/engine/native_logic.cpp",
              "csrc/engine/windows.cpp", "csrc/engine/persist.cpp",
-             "csrc/engine/gpu_ops.cpp"]]
+             "csrc/engine/gpu_ops.cpp", "csrc/engine/gpu_jit.cpp"]]
     if not newer(out, srcs + headers("csrc")):
         return out
     flags = ["-O1" if tsan else "-O2", "-g", "-std=c++20", "-pthread"]

@@ -571,6 +571,13 @@ PYBIND11_MODULE(_core, m) {
     m.def("hash_key", [](uint64_t k) { return KeyByEmitter::mix(k); });
 
 #ifdef WFA_WITH_HIP
+    // JIT codegen probe: the generated hiprtc source for a jit fold spec —
+    // the CPU test tier cross-compiles it with hipcc (no GPU needed)
+    m.def("debug_jit_fold_source",
+          [](const std::string& kind, const std::string& spec,
+             const std::vector<double>& fp, const std::vector<int64_t>& ip) {
+              return debug_jit_fold_source(kind, spec, fp, ip);
+          });
     // debug hooks: round-trip device primitives for isolation tests
     m.def("debug_sort_pairs",
           [](py::array_t<uint32_t> keys, int bits) {

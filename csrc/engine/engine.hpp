@@ -461,4 +461,12 @@ std::vector<std::pair<std::string, double>> debug_ffat_stage_times(
 // (torch.distributed store), every rank passes it to Engine::rccl_id.
 std::string wfa_rccl_unique_id();
 
+// JIT fold codegen (gpu_jit.cpp): returns the hiprtc source that would be
+// compiled for the given jit logic — lets the CPU test tier cross-compile
+// generated kernels with hipcc and catch codegen errors without a GPU.
+std::string debug_jit_fold_source(const std::string& kind,
+                                  const std::string& spec,
+                                  const std::vector<double>& fp,
+                                  const std::vector<int64_t>& ip);
+
 }  // namespace wfa

@@ -1,0 +1,239 @@
+// Shared host-side GPU infrastructure used by gpu_ops.cpp and gpu_jit.cpp:
+// the per-device arena allocator, the GPU operator-logic base class, the
+// keyed grouping scratch (hash->slot + radix sort + segments), and the
+// hiprtc JIT compile cache.  Only meaningful under WFA_WITH_HIP.
+#pragma once
+#ifdef WFA_WITH_HIP
+
+#include <hip/hip_runtime.h>
+#include <hip/hiprtc.h>
+
+#include <map>
+#include <mutex>
+#include <string>
+#include <vector>
+
+#include "engine.hpp"
+
+namespace wfa {
+
+#define HIPCHK(x)                                                                  \
+    do {                                                                           \
+        hipError_t err_ = (x);                                                     \
+        if (err_ != hipSuccess)                                                    \
+            throw std::runtime_error(std::string("HIP error: ") +                  \
+                                     hipGetErrorString(err_) + " at " #x);         \
+    } while (0)
+
+#define RTCCHK(x)                                                              \
+    do {                                                                       \
+        hiprtcResult r_ = (x);                                                 \
+        if (r_ != HIPRTC_SUCCESS)                                              \
+            throw std::runtime_error(std::string("hiprtc error: ") +           \
+                                     hiprtcGetErrorString(r_) + " at " #x);    \
+    } while (0)
+
+// ===== per-device arena allocator =====
+struct DeviceArena {
+    std::mutex mu;
+    std::map<size_t, std::vector<void*>> free_by_size;
+    int device;
+    size_t allocated = 0;
+
+    void* get(size_t bytes) {
+        bytes = (bytes + 255) & ~size_t(255);
+        {
+            std::lock_guard<std::mutex> g(mu);
+            auto it = free_by_size.find(bytes);
+            if (it != free_by_size.end() && !it->second.empty()) {
+                void* p = it->second.back();
+                it->second.pop_back();
+                return p;
+            }
+        }
+        void* p = nullptr;
+        HIPCHK(hipSetDevice(device));
+        HIPCHK(hipMalloc(&p, bytes));
+        std::lock_guard<std::mutex> g(mu);
+        allocated += bytes;
+        return p;
+    }
+    void put(void* p, size_t bytes) {
+        bytes = (bytes + 255) & ~size_t(255);
+        std::lock_guard<std::mutex> g(mu);
+        free_by_size[bytes].push_back(p);
+    }
+};
+
+inline DeviceArena g_arena[64];
+
+inline DeviceArena& arena(int dev) {
+    g_arena[dev].device = dev;
+    return g_arena[dev];
+}
+
+inline bool wfa_prof() {
+    static int v = -1;
+    if (v < 0) v = getenv("WFA_PROF") ? 1 : 0;
+    return v;
+}
+
+// ===== base for GPU logics =====
+struct GpuLogicBase : OpLogic {
+    int64_t prof_host_us = 0, prof_calls = 0;
+    int device = 0;
+    hipStream_t stream = nullptr;
+    std::unique_ptr<Pool> dev_pool;  // device batches this logic emits
+    Schema out_schema;
+    int64_t out_cap = 1 << 20;
+    bool inited = false;
+
+    virtual void init_device() {}
+    void ensure_init() {
+        if (inited) return;
+        HIPCHK(hipSetDevice(device));
+        HIPCHK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+        dev_pool = std::make_unique<Pool>(out_schema, out_cap, false);
+        dev_pool->loc = Loc::DEVICE;
+        dev_pool->device = device;
+        init_device();
+        inited = true;
+    }
+    ~GpuLogicBase() override {
+        if (stream) (void)hipStreamDestroy(stream);
+    }
+
+    void warm(RuntimeCtx&) override { ensure_init(); }
+    bool accepts_device() const override { return true; }
+
+    // make the producing stream's work visible to our stream
+    void wait_ready(Batch* b) {
+        if (b->loc == Loc::DEVICE && b->ready_event)
+            HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)b->ready_event, 0));
+    }
+    void record_ready(Batch* b) {
+        b->stream = stream;
+        HIPCHK(hipEventRecord((hipEvent_t)b->ready_event, stream));
+    }
+
+    // Releasing a device batch whose contents our ASYNC kernels still read:
+    // re-record its event on our stream ("uses done") so whoever pulls it
+    // from the pool next waits before overwriting.  (Single-consumer only;
+    // device-batch broadcast clones instead — see BroadcastEmitter.)
+    void release_after_use(Batch* b) {
+        if (b->loc == Loc::DEVICE && b->ready_event &&
+            b->refcnt.load(std::memory_order_acquire) == 1)
+            HIPCHK(hipEventRecord((hipEvent_t)b->ready_event, stream));
+        release(b);
+    }
+
+    // pool get, preferring a batch whose reuse event already completed —
+    // otherwise allocate a fresh one up to a bounded depth so producers
+    // never serialize against consumers still reading a recycled batch
+    // (288 GB HBM makes a deep rotation cheap; reference recycling_gpu
+    // instead spin-waits under memory pressure).
+    Batch* get_dev() {
+        constexpr int MAX_DEPTH = 8;
+        Batch* chosen = nullptr;
+        Batch* skipped[MAX_DEPTH];
+        int nskip = 0;
+        while (nskip < MAX_DEPTH) {
+            Batch* b = dev_pool->try_pop();
+            if (!b) break;
+            if (!b->ready_event ||
+                hipEventQuery((hipEvent_t)b->ready_event) == hipSuccess) {
+                chosen = b;
+                break;
+            }
+            skipped[nskip++] = b;
+        }
+        for (int i = 0; i < nskip; ++i) dev_pool->put(skipped[i]);
+        if (!chosen) {
+            if (nskip > 0 && dev_pool->live.load(std::memory_order_relaxed) >= MAX_DEPTH)
+                chosen = dev_pool->get();  // bounded: reuse, stream-waits below
+            else
+                chosen = dev_pool->make_new();
+        }
+        if (chosen->ready_event)
+            HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)chosen->ready_event, 0));
+        return chosen;
+    }
+
+    // H2D staging: host batch -> fresh device batch (reference
+    // forward_emitter_gpu.hpp CPU->GPU path, redesigned: whole-SoA copies)
+    Batch* to_device(Batch* hb, RuntimeCtx& ctx) {
+        Batch* db = get_dev();
+        int64_t n = hb->count;
+        HIPCHK(hipMemcpyAsync(db->ts, hb->ts, 8 * n, hipMemcpyHostToDevice, stream));
+        HIPCHK(hipMemcpyAsync(db->key, hb->key, 8 * n, hipMemcpyHostToDevice, stream));
+        size_t bytes = 16 * n;
+        for (size_t c = 0; c < hb->cols.size() && c < db->cols.size(); ++c) {
+            size_t es = dsize(hb->schema.payload[c]);
+            HIPCHK(hipMemcpyAsync(db->cols[c], hb->cols[c], es * n,
+                                  hipMemcpyHostToDevice, stream));
+            bytes += es * n;
+        }
+        db->count = n;
+        db->watermark = ctx.current_wm;   // folded (hb may be a shared batch)
+        db->stream_tag = ctx.current_tag;
+        if (ctx.stats) ctx.stats->bytes_h2d += bytes;
+        // the host batch may be recycled by a CPU producer the moment we
+        // release it — the copies must have landed first
+        HIPCHK(hipStreamSynchronize(stream));
+        release(hb);
+        return db;
+    }
+
+    Batch* input_on_device(Batch* b, RuntimeCtx& ctx) {
+        if (b->loc == Loc::HOST) return to_device(b, ctx);
+        gpu_resolve_count(b);
+        wait_ready(b);
+        return b;
+    }
+};
+
+// ===== shared keyed front half: slot -> sort -> gather -> segments =====
+struct KeyedScratch {
+    uint64_t* tab = nullptr;        // packed (key, slot) 16 B entries
+    uint32_t* d_nslots = nullptr;
+    uint64_t* slot_to_key = nullptr;
+    uint32_t *slot = nullptr, *idx = nullptr, *slot_t = nullptr, *idx_t = nullptr;
+    uint32_t* hist = nullptr;
+    uint32_t *seg_start = nullptr, *seg_slot = nullptr;
+    int64_t* d_nseg = nullptr;
+    float* v_sorted = nullptr;   // second-payload ping-pong (sort_pairs2)
+    float* v_f32 = nullptr;      // cast buffer / sort ping-pong
+    int64_t table_cap = 0;
+    int64_t max_keys = 0;
+    int64_t cap = 0;
+    int bits = 20;
+
+    void alloc(int dev, int64_t cap_, int64_t mk, hipStream_t s);
+
+    uint32_t* idx_sorted = nullptr;  // valid after group()
+    const void* v_as_f32 = nullptr;  // f32 or bf16, per v_dt
+    int v_dt = 2;                    // effective dtype of v_as_f32 (2/5)
+
+    void group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
+               bool want_vik = false);
+};
+
+// ===== hiprtc JIT compile cache =====
+// Returns a loaded module for the given source (compiled for the local
+// arch); modules are cached per source string for the process lifetime.
+hipModule_t jit_module(const std::string& src, int device);
+hipFunction_t jit_fn(hipModule_t mod, const char* name);
+std::string jit_type(DType d);
+
+// generalized user-fold JIT factory (gpu_jit.cpp): kinds gpu_jit_reduce /
+// gpu_jit_ffat
+std::shared_ptr<OpLogic> make_gpu_jit_logic(const std::string& kind,
+                                            const std::string& spec,
+                                            const std::vector<double>& fp,
+                                            const std::vector<int64_t>& ip,
+                                            Engine* eng, int op_id, int device,
+                                            const Schema& os, int64_t out_batch);
+
+}  // namespace wfa
+
+#endif  // WFA_WITH_HIP

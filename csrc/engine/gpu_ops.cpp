@@ -18,55 +18,12 @@
 #include "engine.hpp"
 
 #ifdef WFA_WITH_HIP
-#include <hip/hip_runtime.h>
-#include <hip/hiprtc.h>
 #include <rccl/rccl.h>
 
 #include "../hip/wfa_kernels.h"
+#include "gpu_common.hpp"
 
 namespace wfa {
-
-#define HIPCHK(x)                                                                  \
-    do {                                                                           \
-        hipError_t err_ = (x);                                                     \
-        if (err_ != hipSuccess)                                                    \
-            throw std::runtime_error(std::string("HIP error: ") +                  \
-                                     hipGetErrorString(err_) + " at " #x);         \
-    } while (0)
-
-// ===== per-device arena allocator =====
-struct DeviceArena {
-    std::mutex mu;
-    std::map<size_t, std::vector<void*>> free_by_size;
-    int device;
-    size_t allocated = 0;
-
-    void* get(size_t bytes) {
-        bytes = (bytes + 255) & ~size_t(255);
-        {
-            std::lock_guard<std::mutex> g(mu);
-            auto it = free_by_size.find(bytes);
-            if (it != free_by_size.end() && !it->second.empty()) {
-                void* p = it->second.back();
-                it->second.pop_back();
-                return p;
-            }
-        }
-        void* p = nullptr;
-        HIPCHK(hipSetDevice(device));
-        HIPCHK(hipMalloc(&p, bytes));
-        std::lock_guard<std::mutex> g(mu);
-        allocated += bytes;
-        return p;
-    }
-    void put(void* p, size_t bytes) {
-        bytes = (bytes + 255) & ~size_t(255);
-        std::lock_guard<std::mutex> g(mu);
-        free_by_size[bytes].push_back(p);
-    }
-};
-
-static DeviceArena g_arena[64];
 
 // hipEventCreate + hipHostMalloc cost ~0.1-1 ms each; recycle them across
 // batches AND engines (a fresh PipeGraph would otherwise re-pin per batch
@@ -96,11 +53,6 @@ struct AuxPool {
     }
 };
 static AuxPool g_aux;
-
-DeviceArena& arena(int dev) {
-    g_arena[dev].device = dev;
-    return g_arena[dev];
-}
 
 // ===== device batch alloc (core.cpp hooks) =====
 static size_t dev_batch_bytes(const Schema& s, int64_t cap) {
@@ -147,126 +99,6 @@ void gpu_free_batch(Batch* b) {
     if (b->ready_event) g_aux.put(b->ready_event, b->lazy_count);
     delete b;
 }
-
-static bool wfa_prof() {
-    static int v = -1;
-    if (v < 0) v = getenv("WFA_PROF") ? 1 : 0;
-    return v;
-}
-
-// ===== base for GPU logics =====
-struct GpuLogicBase : OpLogic {
-    int64_t prof_host_us = 0, prof_calls = 0;
-    int device = 0;
-    hipStream_t stream = nullptr;
-    std::unique_ptr<Pool> dev_pool;  // device batches this logic emits
-    Schema out_schema;
-    int64_t out_cap = 1 << 20;
-    bool inited = false;
-
-    virtual void init_device() {}
-    void ensure_init() {
-        if (inited) return;
-        HIPCHK(hipSetDevice(device));
-        HIPCHK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
-        dev_pool = std::make_unique<Pool>(out_schema, out_cap, false);
-        dev_pool->loc = Loc::DEVICE;
-        dev_pool->device = device;
-        init_device();
-        inited = true;
-    }
-    ~GpuLogicBase() override {
-        if (stream) (void)hipStreamDestroy(stream);
-    }
-
-    void warm(RuntimeCtx&) override { ensure_init(); }
-    bool accepts_device() const override { return true; }
-
-    // make the producing stream's work visible to our stream
-    void wait_ready(Batch* b) {
-        if (b->loc == Loc::DEVICE && b->ready_event)
-            HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)b->ready_event, 0));
-    }
-    void record_ready(Batch* b) {
-        b->stream = stream;
-        HIPCHK(hipEventRecord((hipEvent_t)b->ready_event, stream));
-    }
-
-    // Releasing a device batch whose contents our ASYNC kernels still read:
-    // re-record its event on our stream ("uses done") so whoever pulls it
-    // from the pool next waits before overwriting.  (Single-consumer only;
-    // device-batch broadcast would need per-consumer events.)
-    void release_after_use(Batch* b) {
-        if (b->loc == Loc::DEVICE && b->ready_event &&
-            b->refcnt.load(std::memory_order_acquire) == 1)
-            HIPCHK(hipEventRecord((hipEvent_t)b->ready_event, stream));
-        release(b);
-    }
-
-    // pool get, preferring a batch whose reuse event already completed —
-    // otherwise allocate a fresh one up to a bounded depth so producers
-    // never serialize against consumers still reading a recycled batch
-    // (288 GB HBM makes a deep rotation cheap; reference recycling_gpu
-    // instead spin-waits under memory pressure).
-    Batch* get_dev() {
-        constexpr int MAX_DEPTH = 8;
-        Batch* chosen = nullptr;
-        Batch* skipped[MAX_DEPTH];
-        int nskip = 0;
-        while (nskip < MAX_DEPTH) {
-            Batch* b = dev_pool->try_pop();
-            if (!b) break;
-            if (!b->ready_event ||
-                hipEventQuery((hipEvent_t)b->ready_event) == hipSuccess) {
-                chosen = b;
-                break;
-            }
-            skipped[nskip++] = b;
-        }
-        for (int i = 0; i < nskip; ++i) dev_pool->put(skipped[i]);
-        if (!chosen) {
-            if (nskip > 0 && dev_pool->live.load(std::memory_order_relaxed) >= MAX_DEPTH)
-                chosen = dev_pool->get();  // bounded: reuse, stream-waits below
-            else
-                chosen = dev_pool->make_new();
-        }
-        if (chosen->ready_event)
-            HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)chosen->ready_event, 0));
-        return chosen;
-    }
-
-    // H2D staging: host batch -> fresh device batch (reference
-    // forward_emitter_gpu.hpp CPU->GPU path, redesigned: whole-SoA copies)
-    Batch* to_device(Batch* hb, RuntimeCtx& ctx) {
-        Batch* db = get_dev();
-        int64_t n = hb->count;
-        HIPCHK(hipMemcpyAsync(db->ts, hb->ts, 8 * n, hipMemcpyHostToDevice, stream));
-        HIPCHK(hipMemcpyAsync(db->key, hb->key, 8 * n, hipMemcpyHostToDevice, stream));
-        size_t bytes = 16 * n;
-        for (size_t c = 0; c < hb->cols.size() && c < db->cols.size(); ++c) {
-            size_t es = dsize(hb->schema.payload[c]);
-            HIPCHK(hipMemcpyAsync(db->cols[c], hb->cols[c], es * n,
-                                  hipMemcpyHostToDevice, stream));
-            bytes += es * n;
-        }
-        db->count = n;
-        db->watermark = ctx.current_wm;   // folded (hb may be a shared batch)
-        db->stream_tag = ctx.current_tag;
-        if (ctx.stats) ctx.stats->bytes_h2d += bytes;
-        // the host batch may be recycled by a CPU producer the moment we
-        // release it — the copies must have landed first
-        HIPCHK(hipStreamSynchronize(stream));
-        release(hb);
-        return db;
-    }
-
-    Batch* input_on_device(Batch* b, RuntimeCtx& ctx) {
-        if (b->loc == Loc::HOST) return to_device(b, ctx);
-        gpu_resolve_count(b);
-        wait_ready(b);
-        return b;
-    }
-};
 
 // ===== device source (GPU-resident synthetic generator) =====
 struct GpuSourceLogic : GpuLogicBase {
@@ -384,60 +216,42 @@ struct GpuFilterLogic : GpuLogicBase {
 };
 
 // ===== shared keyed front half: slot -> sort -> gather -> segments =====
-struct KeyedScratch {
-    uint64_t* tab = nullptr;        // packed (key, slot) 16 B entries
-    uint32_t* d_nslots = nullptr;
-    uint64_t* slot_to_key = nullptr;
-    uint32_t *slot = nullptr, *idx = nullptr, *slot_t = nullptr, *idx_t = nullptr;
-    uint32_t* hist = nullptr;
-    uint32_t *seg_start = nullptr, *seg_slot = nullptr;
-    int64_t* d_nseg = nullptr;
-    float* v_sorted = nullptr;   // second-payload ping-pong (sort_pairs2)
-    float* v_f32 = nullptr;      // cast buffer / sort ping-pong
-    int64_t table_cap = 0;
-    int64_t max_keys = 0;
-    int64_t cap = 0;
-    int bits = 20;
+// (struct declared in gpu_common.hpp; shared with gpu_jit.cpp)
+void KeyedScratch::alloc(int dev, int64_t cap_, int64_t mk, hipStream_t s) {
+    cap = cap_;
+    max_keys = mk;
+    table_cap = 1;
+    while (table_cap < 2 * mk) table_cap <<= 1;
+    bits = 1;
+    while ((1ll << bits) < mk + 1) ++bits;
+    auto& A = arena(dev);
+    tab = (uint64_t*)A.get(16 * table_cap);
+    d_nslots = (uint32_t*)A.get(64);
+    slot_to_key = (uint64_t*)A.get(8 * mk);
+    slot = (uint32_t*)A.get(4 * cap);
+    idx = (uint32_t*)A.get(4 * cap);
+    slot_t = (uint32_t*)A.get(4 * cap);
+    idx_t = (uint32_t*)A.get(4 * cap);
+    hist = (uint32_t*)A.get(4 * wfa_sort_hist_u32(cap));
+    seg_start = (uint32_t*)A.get(4 * cap);
+    seg_slot = (uint32_t*)A.get(4 * cap);
+    d_nseg = (int64_t*)A.get(64);
+    v_sorted = (float*)A.get(4 * cap);   // reused as fire-offset scratch
+    v_f32 = (float*)A.get(4 * cap);
+    wfa_fill_u64(s, tab, ~0ULL, 2 * table_cap);
+    wfa_fill_u32(s, d_nslots, 0, 1);
+}
 
-    void alloc(int dev, int64_t cap_, int64_t mk, hipStream_t s) {
-        cap = cap_;
-        max_keys = mk;
-        table_cap = 1;
-        while (table_cap < 2 * mk) table_cap <<= 1;
-        bits = 1;
-        while ((1ll << bits) < mk + 1) ++bits;
-        auto& A = arena(dev);
-        tab = (uint64_t*)A.get(16 * table_cap);
-        d_nslots = (uint32_t*)A.get(64);
-        slot_to_key = (uint64_t*)A.get(8 * mk);
-        slot = (uint32_t*)A.get(4 * cap);
-        idx = (uint32_t*)A.get(4 * cap);
-        slot_t = (uint32_t*)A.get(4 * cap);
-        idx_t = (uint32_t*)A.get(4 * cap);
-        hist = (uint32_t*)A.get(4 * wfa_sort_hist_u32(cap));
-        seg_start = (uint32_t*)A.get(4 * cap);
-        seg_slot = (uint32_t*)A.get(4 * cap);
-        d_nseg = (int64_t*)A.get(64);
-        v_sorted = (float*)A.get(4 * cap);   // reused as fire-offset scratch
-        v_f32 = (float*)A.get(4 * cap);
-        wfa_fill_u64(s, tab, ~0ULL, 2 * table_cap);
-        wfa_fill_u32(s, d_nslots, 0, 1);
-    }
-
-    uint32_t* idx_sorted = nullptr;  // valid after group()
-    const void* v_as_f32 = nullptr;  // f32 or bf16, per v_dt
-    int v_dt = 2;                    // effective dtype of v_as_f32 (2/5)
-
-    // sorts (slot, idx) pairs and fills segments; values stay unsorted and
-    // are read through idx_sorted (saves the gather round trip) — EXCEPT in
-    // value-in-key (VIK) mode: for bf16 value columns with <= 65535 keys the
-    // bf16 bits ride in the low 16 bits of the sort key (radix passes sort
-    // on bits 16.. only, stability keeps per-key order), so the fold reads
-    // values SEQUENTIALLY from the sorted key array instead of gathering
-    // one cache line per tuple.  Numerics are identical (same bf16 bits).
-    // vcol < 0: no value cast (stateful map/filter operate in place)
-    void group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
-               bool want_vik = false) {
+// sorts (slot, idx) pairs and fills segments; values stay unsorted and
+// are read through idx_sorted (saves the gather round trip) — EXCEPT in
+// value-in-key (VIK) mode: for bf16 value columns with <= 65535 keys the
+// bf16 bits ride in the low 16 bits of the sort key (radix passes sort
+// on bits 16.. only, stability keeps per-key order), so the fold reads
+// values SEQUENTIALLY from the sorted key array instead of gathering
+// one cache line per tuple.  Numerics are identical (same bf16 bits).
+// vcol < 0: no value cast (stateful map/filter operate in place)
+void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
+                         bool want_vik) {
         int64_t n = db->count;
         if (n > cap)
             throw std::runtime_error("batch larger than keyed scratch capacity — "
@@ -484,28 +298,16 @@ struct KeyedScratch {
         }
         wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg, /*shr=*/0);
         if (ctx.stats) ctx.stats->num_kernels += 5 + 3 * ((bits + 3) / 4);
-    }
-};
+}
 
 // ===== hiprtc JIT: user device logic (reference MAP_GPU/FILTER_GPU
 // accept arbitrary __device__ lambdas; here a C expression over
-// (v, ts, key) is runtime-compiled for gfx950 and cached) =====
-#define RTCCHK(x)                                                              \
-    do {                                                                       \
-        hiprtcResult r_ = (x);                                                 \
-        if (r_ != HIPRTC_SUCCESS)                                              \
-            throw std::runtime_error(std::string("hiprtc error: ") +           \
-                                     hiprtcGetErrorString(r_) + " at " #x);    \
-    } while (0)
-
-struct JitKernel {
-    hipModule_t mod = nullptr;
-    hipFunction_t fn = nullptr;
-};
-
-static JitKernel jit_compile(const std::string& src, int device) {
+// (v, ts, key) is runtime-compiled for gfx950 and cached).  The general
+// multi-field fold JIT (Reduce_GPU / FFAT combines) lives in gpu_jit.cpp;
+// this file keeps only the scalar map/filter expression path.
+hipModule_t jit_module(const std::string& src, int device) {
     static std::mutex mu;
-    static std::map<std::string, JitKernel> cache;
+    static std::map<std::string, hipModule_t> cache;
     std::lock_guard<std::mutex> g(mu);
     auto it = cache.find(src);
     if (it != cache.end()) return it->second;
@@ -530,14 +332,19 @@ static JitKernel jit_compile(const std::string& src, int device) {
     std::vector<char> code(csz);
     RTCCHK(hiprtcGetCode(prog, code.data()));
     hiprtcDestroyProgram(&prog);
-    JitKernel k;
-    HIPCHK(hipModuleLoadData(&k.mod, code.data()));
-    HIPCHK(hipModuleGetFunction(&k.fn, k.mod, "wfa_jit"));
-    cache[src] = k;
-    return k;
+    hipModule_t mod;
+    HIPCHK(hipModuleLoadData(&mod, code.data()));
+    cache[src] = mod;
+    return mod;
 }
 
-static std::string jit_type(DType d) {
+hipFunction_t jit_fn(hipModule_t mod, const char* name) {
+    hipFunction_t fn;
+    HIPCHK(hipModuleGetFunction(&fn, mod, name));
+    return fn;
+}
+
+std::string jit_type(DType d) {
     switch (d) {
         case DType::I64: return "long long";
         case DType::F64: return "double";
@@ -546,6 +353,18 @@ static std::string jit_type(DType d) {
         case DType::I32: return "int";
         default: throw std::runtime_error("jit: unsupported column dtype");
     }
+}
+
+struct JitKernel {
+    hipModule_t mod = nullptr;
+    hipFunction_t fn = nullptr;
+};
+
+static JitKernel jit_compile(const std::string& src, int device) {
+    JitKernel k;
+    k.mod = jit_module(src, device);
+    k.fn = jit_fn(k.mod, "wfa_jit");
+    return k;
 }
 
 // expr sees: v (value column, mutable type T), ts (long long), key (u64)
@@ -865,6 +684,7 @@ struct GpuFfatLogic : GpuLogicBase {
     float* ring_or_tree = nullptr;
     uint32_t* st_head = nullptr;
     float* st_wsum = nullptr;
+    int64_t* st_last = nullptr;  // per-slot last-arrival ts (EOS flush)
     int64_t* d_on = nullptr;
     uint32_t* cb_nf = nullptr;  // CB fire-offset scratch
     // TB state
@@ -905,8 +725,10 @@ struct GpuFfatLogic : GpuLogicBase {
         S = slide / pane_len;
         ring_log2 = 1;
         while ((1ll << ring_log2) < P + 2) ++ring_log2;
-        if (tb && use_tree)
-            throw std::runtime_error("gpu_ffat: TB windows use the ring path");
+        // TB windows always use the pane ring (the pending-pane machine IS
+        // the TB path); a requested tree layout silently falls back — same
+        // results, different CB-only perf trade (round 1 threw here)
+        if (tb && use_tree) use_tree = false;
     }
     void init_device() override {
         ks.alloc(device, out_cap, max_keys, stream);
@@ -918,16 +740,19 @@ struct GpuFfatLogic : GpuLogicBase {
         ring_or_tree = (float*)A.get(4 * max_keys * (use_tree ? 2 * R : R));
         st_head = (uint32_t*)A.get(4 * max_keys);
         st_wsum = (float*)A.get(4 * max_keys);
+        st_last = (int64_t*)A.get(8 * max_keys);
         d_on = (int64_t*)A.get(64);
         HIPCHK(hipMemsetAsync(st_count, 0, 8 * max_keys, stream));
         HIPCHK(hipMemsetAsync(st_fill, 0, 4 * max_keys, stream));
         HIPCHK(hipMemsetAsync(st_head, 0, 4 * max_keys, stream));
         HIPCHK(hipMemsetAsync(st_wsum, 0, 4 * max_keys, stream));
+        HIPCHK(hipMemsetAsync(st_last, 0, 8 * max_keys, stream));
         float ident = comb == 1 ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
         wfa_fill_f32(stream, st_acc, ident, max_keys);
         wfa_fill_f32(stream, ring_or_tree, ident,
                      max_keys * (use_tree ? 2 * (1ll << ring_log2) : (1ll << ring_log2)));
-        cb_nf = (uint32_t*)A.get(4 * (out_cap + 1));
+        // fire-offset scratch doubles as the EOS-flush per-slot counter
+        cb_nf = (uint32_t*)A.get(4 * (std::max(out_cap, max_keys) + 1));
         use_pane2 = !tb && !use_tree && pane2_enabled() && pane_len >= 32;
         if (tb) {
             int64_t Rp = 1ll << pend_log2;
@@ -985,6 +810,8 @@ struct GpuFfatLogic : GpuLogicBase {
 
     void chain_cb(Batch* db, Batch* ob, int64_t n, RuntimeCtx& ctx) {
         ks.group(stream, db, vcol, ctx, /*want_vik=*/!use_tree);
+        wfa_seg_last_ts(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
+                        ks.idx_sorted, db->ts, st_last);
         uint32_t* nf = cb_nf;
         wfa_ffat_fire_offsets(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                               pane_len, P, S, st_fill, st_head, nf, d_on);
@@ -1016,6 +843,8 @@ struct GpuFfatLogic : GpuLogicBase {
         int64_t n = db->count;
         if (tb) {
             ks.group(stream, db, vcol, ctx);
+            wfa_seg_last_ts(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
+                            ks.idx_sorted, db->ts, st_last);
             tb_round_with_count(db, n, db->watermark, out, ctx);
             release_after_use(db);
             return;
@@ -1081,11 +910,36 @@ struct GpuFfatLogic : GpuLogicBase {
     }
 
     void on_eos(EmitCtx& out, RuntimeCtx& ctx) override {
-        if (!tb) return;
         ensure_init();
-        // complete every remaining data pane (wm -> +inf)
-        tb_round_with_count(nullptr, 0, INT64_MAX / 4, out, ctx);
-        check_tb_flags();
+        if (tb) {
+            // complete every remaining data pane (wm -> +inf) ...
+            tb_round_with_count(nullptr, 0, INT64_MAX / 4, out, ctx);
+            check_tb_flags();
+            // ... then fire the partial pane-windows (< P panes) the CPU
+            // twin flushes (windows.cpp FfatCpu on_eos TB branch)
+            flush_open(nullptr, nullptr, out, ctx);
+            return;
+        }
+        // CB: flush every open (partial) window — round 1 fired complete
+        // windows only (mirrors windows.cpp FfatCpu on_eos CB branch)
+        flush_open(st_fill, st_acc, out, ctx);
+    }
+
+    void flush_open(const uint32_t* fill, const float* acc, EmitCtx& out,
+                    RuntimeCtx& ctx) {
+        Batch* ob = get_dev();
+        int64_t R = 1ll << ring_log2;
+        wfa_ffat_cb_flush(stream, ks.d_nslots, P, S, comb, ring_log2, fill, acc,
+                          ring_or_tree, use_tree ? 2 * R : R, use_tree ? R : 0,
+                          st_head, st_last, ks.slot_to_key, cb_nf, ob->key,
+                          (float*)ob->cols[0], ob->ts, ob->capacity, d_on);
+        HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost,
+                              stream));
+        ob->count = -1;
+        ob->watermark = WM_MAX / 4;
+        if (ctx.stats) ctx.stats->num_kernels += 3;
+        record_ready(ob);
+        out.emit(ob);
     }
 };
 
@@ -1562,6 +1416,11 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
     if (kind == "gpu_jit_filter")
         return std::make_shared<GpuJitFilterLogic>(spec, (int)(ip.empty() ? 0 : ip[0]),
                                                    device, os, out_batch);
+    if (kind == "gpu_jit_reduce" || kind == "gpu_jit_ffat" ||
+        kind == "gpu_jit_stateful")
+        // generalized user folds (multi-field accumulators) — gpu_jit.cpp
+        return make_gpu_jit_logic(kind, spec, fp, ip, eng, op_id, device, os,
+                                  out_batch);
     if (kind == "gpu_map_keyed")
         // ip: [spec, col, max_keys]; fp: [a, b]
         return std::make_shared<GpuStatefulMapLogic>((int)ip[0], (int)ip[1], fp[0],

@@ -1522,6 +1522,142 @@ extern "C" void wfa_ffat_tree_fold(
                        out_key, out_val, out_ts, out_cap);
 }
 
+// ===== per-key last-arrival timestamp (EOS flush emit ts) =====
+// One thread per segment; a slot appears in at most one segment per batch,
+// so the max-fold store is race-free.  Mirrors the CPU engine's
+// ks.last_ts = max(last_ts, ts) (windows.cpp Keyed/FfatCpu state).
+__global__ void k_seg_last_ts(const uint32_t* seg_start, const uint32_t* seg_slot,
+                              const int64_t* d_nseg, int64_t n,
+                              const uint32_t* idx_sorted, const int64_t* ts_orig,
+                              int64_t* st_last) {
+    const int64_t nseg = *d_nseg;
+    for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nseg;
+         j += gridDim.x * (int64_t)blockDim.x) {
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        const int64_t t = ts_orig[idx_sorted[e - 1]];
+        int64_t* p = &st_last[seg_slot[j]];
+        if (t > *p) *p = t;
+    }
+}
+
+extern "C" void wfa_seg_last_ts(wfa_stream_t s, const uint32_t* seg_start,
+                                const uint32_t* seg_slot, const int64_t* d_nseg,
+                                int64_t n, const uint32_t* idx_sorted,
+                                const int64_t* ts_orig, int64_t* st_last) {
+    hipLaunchKernelGGL(k_seg_last_ts, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS),
+                       0, (hipStream_t)s, seg_start, seg_slot, d_nseg, n,
+                       idx_sorted, ts_orig, st_last);
+}
+
+// ===== CB/TB EOS partial-window flush =====
+// Mirrors the CPU FfatCpu EOS (windows.cpp on_eos): close the open partial
+// pane (CB only: st_fill/st_acc non-null), then fire every remaining open
+// window — starts q*S for q in [F(head), ...) with q*S < H — combining the
+// ring cells [q*S, head) plus the partial pane.  H = head (+1 if a partial
+// pane exists).  `cells` covers both layouts: ring (stride R, off 0) and
+// FlatFAT-tree leaves (stride 2R, off R).
+__global__ void k_cb_flush_count(const uint32_t* n_slots, int64_t P, int64_t S,
+                                 const uint32_t* st_fill, const uint32_t* st_head,
+                                 uint32_t* nf) {
+    const int64_t ns = *n_slots;
+    for (int64_t s = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; s < ns;
+         s += gridDim.x * (int64_t)blockDim.x) {
+        const uint64_t head = st_head[s];
+        const uint64_t H = head + ((st_fill && st_fill[s]) ? 1 : 0);
+        auto F = [&](uint64_t x) {
+            return x < (uint64_t)P ? 0ull : (x - (uint64_t)P) / (uint64_t)S + 1ull;
+        };
+        const uint64_t q0 = F(head);
+        nf[s] = (H > q0 * (uint64_t)S)
+                    ? (uint32_t)((H - 1 - q0 * (uint64_t)S) / (uint64_t)S + 1)
+                    : 0;
+    }
+}
+
+__global__ void k_cb_flush(const uint32_t* n_slots, int64_t P, int64_t S,
+                           int comb, int ring_log2, const uint32_t* st_fill,
+                           const float* st_acc, const float* cells,
+                           int64_t slot_stride, int64_t cell_off,
+                           const uint32_t* st_head, const int64_t* st_last,
+                           const uint64_t* slot_to_key, const uint32_t* nf,
+                           uint64_t* out_key, float* out_val, int64_t* out_ts,
+                           int64_t out_cap) {
+    const int64_t ns = *n_slots;
+    const uint32_t Rm = (1u << ring_log2) - 1;
+    const float ident = (comb == 1) ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
+#define XCOMB(a, b) ((comb == 1) ? fminf(a, b) : (comb == 2 ? fmaxf(a, b) : (a) + (b)))
+    for (int64_t s = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; s < ns;
+         s += gridDim.x * (int64_t)blockDim.x) {
+        const uint64_t head = st_head[s];
+        const bool part = st_fill && st_fill[s];
+        const uint64_t H = head + (part ? 1 : 0);
+        auto F = [&](uint64_t x) {
+            return x < (uint64_t)P ? 0ull : (x - (uint64_t)P) / (uint64_t)S + 1ull;
+        };
+        const float* rg = cells + (size_t)s * slot_stride + cell_off;
+        int64_t w = nf[s];
+        for (uint64_t q = F(head); q * (uint64_t)S < H; ++q) {
+            float res = ident;
+            for (uint64_t p = q * (uint64_t)S; p < head; ++p)
+                res = XCOMB(res, rg[(uint32_t)p & Rm]);
+            if (part) res = XCOMB(res, st_acc[s]);
+            if (w < out_cap) {
+                out_key[w] = slot_to_key[s];
+                out_val[w] = res;
+                out_ts[w] = st_last ? st_last[s] : 0;
+            }
+            ++w;
+        }
+    }
+#undef XCOMB
+}
+
+extern "C" void wfa_ffat_cb_flush(
+    wfa_stream_t s, const uint32_t* n_slots, int64_t P, int64_t S, int comb,
+    int ring_log2, const uint32_t* st_fill, const float* st_acc,
+    const float* cells, int64_t slot_stride, int64_t cell_off,
+    const uint32_t* st_head, const int64_t* st_last,
+    const uint64_t* slot_to_key, uint32_t* nf, uint64_t* out_key,
+    float* out_val, int64_t* out_ts, int64_t out_cap, int64_t* d_out_n) {
+    hipStream_t st = (hipStream_t)s;
+    hipLaunchKernelGGL(k_cb_flush_count, dim3(WFA_MAX_BLOCKS / 8),
+                       dim3(WFA_THREADS), 0, st, n_slots, P, S, st_fill, st_head,
+                       nf);
+    hipLaunchKernelGGL(k_tb_scan, dim3(1), dim3(1024), 0, st, nf, n_slots,
+                       d_out_n);
+    hipLaunchKernelGGL(k_cb_flush, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
+                       st, n_slots, P, S, comb, ring_log2, st_fill, st_acc, cells,
+                       slot_stride, cell_off, st_head, st_last, slot_to_key, nf,
+                       out_key, out_val, out_ts, out_cap);
+}
+
+// single-block exclusive scan over per-slot counts (shared with the JIT
+// fold modules, which generate their own count/fire kernels)
+extern "C" void wfa_slot_scan(wfa_stream_t s, uint32_t* nf,
+                              const uint32_t* n_slots, int64_t* d_out_n) {
+    hipLaunchKernelGGL(k_tb_scan, dim3(1), dim3(1024), 0, (hipStream_t)s, nf,
+                       n_slots, d_out_n);
+}
+
+// value-independent pieces exported for the JIT fold path (whose lift/
+// advance/fire kernels are hiprtc-generated but share these counters)
+extern "C" void wfa_cb_flush_count(wfa_stream_t s, const uint32_t* n_slots,
+                                   int64_t P, int64_t S, const uint32_t* st_fill,
+                                   const uint32_t* st_head, uint32_t* nf) {
+    hipLaunchKernelGGL(k_cb_flush_count, dim3(WFA_MAX_BLOCKS / 8),
+                       dim3(WFA_THREADS), 0, (hipStream_t)s, n_slots, P, S,
+                       st_fill, st_head, nf);
+}
+
+extern "C" void wfa_tb_count(wfa_stream_t s, const uint32_t* n_slots,
+                             int64_t limit_pane, const int64_t* pend_base,
+                             const int64_t* last_pane, const uint32_t* st_head,
+                             int64_t P, int64_t S, uint32_t* nf) {
+    hipLaunchKernelGGL(k_tb_count, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, n_slots, limit_pane, pend_base, last_pane,
+                       st_head, P, S, nf);
+}
+
 // ===== unkeyed full-batch reduce (reference reduce_gpu.hpp:269
 // thrust::reduce path): two deterministic stages, no atomics =====
 #define RA_BLOCKS 512

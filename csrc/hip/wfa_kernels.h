@@ -191,10 +191,43 @@ void wfa_ffat_tree_fold(wfa_stream_t s, const uint32_t* seg_start,
                         uint64_t* out_key, float* out_val, int64_t* out_ts,
                         int64_t out_cap);
 
+// per-slot last-arrival ts (max-folded) — feeds the EOS flush emit ts
+void wfa_seg_last_ts(wfa_stream_t s, const uint32_t* seg_start,
+                     const uint32_t* seg_slot, const int64_t* d_nseg, int64_t n,
+                     const uint32_t* idx_sorted, const int64_t* ts_orig,
+                     int64_t* st_last);
+
+// CB/TB EOS partial-window flush (mirrors CPU FfatCpu on_eos): fires every
+// remaining open window from the ring/tree-leaf cells + the open pane.
+// st_fill/st_acc null => TB (no partial pane).  cells stride/off: ring =
+// (R, 0), tree leaves = (2R, R).  nf: u32 scratch >= n_slots+1.
+void wfa_ffat_cb_flush(wfa_stream_t s, const uint32_t* n_slots, int64_t P,
+                       int64_t S, int comb, int ring_log2,
+                       const uint32_t* st_fill, const float* st_acc,
+                       const float* cells, int64_t slot_stride, int64_t cell_off,
+                       const uint32_t* st_head, const int64_t* st_last,
+                       const uint64_t* slot_to_key, uint32_t* nf,
+                       uint64_t* out_key, float* out_val, int64_t* out_ts,
+                       int64_t out_cap, int64_t* d_out_n);
+
+// single-block exclusive scan over per-slot counts + total
+void wfa_slot_scan(wfa_stream_t s, uint32_t* nf, const uint32_t* n_slots,
+                   int64_t* d_out_n);
+
+// value-independent counters shared with the hiprtc-generated fold path
+void wfa_cb_flush_count(wfa_stream_t s, const uint32_t* n_slots, int64_t P,
+                        int64_t S, const uint32_t* st_fill,
+                        const uint32_t* st_head, uint32_t* nf);
+void wfa_tb_count(wfa_stream_t s, const uint32_t* n_slots, int64_t limit_pane,
+                  const int64_t* pend_base, const int64_t* last_pane,
+                  const uint32_t* st_head, int64_t P, int64_t S, uint32_t* nf);
+
 // ----- misc -----
 void wfa_fill_u64(wfa_stream_t s, uint64_t* p, uint64_t v, int64_t n);
 void wfa_fill_u32(wfa_stream_t s, uint32_t* p, uint32_t v, int64_t n);
 void wfa_fill_f32(wfa_stream_t s, float* p, float v, int64_t n);
+void wfa_fill_f32_strided(wfa_stream_t s, float* p, float v, int64_t n,
+                          int64_t stride);
 void wfa_iota_u32(wfa_stream_t s, uint32_t* p, int64_t n);
 void wfa_cast(wfa_stream_t s, const void* in, int dt_in, void* out, int dt_out,
               int64_t n);

@@ -88,9 +88,25 @@ def test_gpu_map_bf16_vectorized():
     assert abs(got['s'] - ref) <= 1e-6 * max(1.0, abs(ref))
 
 
+def _seg_comb(seg, win, slide, comb):
+    # fp32 pane-partials then pane combine — mirrors the GPU fold order
+    pane = int(np.gcd(win, slide))
+    segs = [seg[i:i + pane] for i in range(0, len(seg), pane)]
+    if comb == "sum":
+        partials = [np.sum(np.array(s, dtype=np.float32), dtype=np.float32)
+                    for s in segs]
+        return float(np.sum(np.array(partials, dtype=np.float64)))
+    if comb == "min":
+        return float(min(min(s) for s in segs))
+    return float(max(max(s) for s in segs))
+
+
 def _ffat_oracle(n, n_keys, win, slide, vdt=2, comb="sum", batch=BATCH):
-    """Per-key sliding CB windows over panes in generator order; windows at
-    [w*slide, w*slide+win).  Returns {key: [result,...]} in firing order."""
+    """Per-key sliding CB windows in generator order; windows at
+    [w*slide, w*slide+win).  Returns {key: [result,...]}: full windows in
+    firing order, then the EOS partial-window flush (every open window
+    start w*slide < #tuples fires with the tail — engine EOS semantics,
+    csrc windows.cpp FfatCpu on_eos / wfa_ffat_cb_flush)."""
     ts, key, val = gen_batch(n, 0, 42, n_keys, vdt)
     if vdt == 5:
         val = bf16_to_f32_np(val)
@@ -105,20 +121,11 @@ def _ffat_oracle(n, n_keys, win, slide, vdt=2, comb="sum", batch=BATCH):
         res = []
         w = 0
         while w * slide + win <= len(vs):
-            seg = vs[w * slide: w * slide + win]
-            # fp32 pane-partials then pane combine — mirrors the GPU order
-            pane = int(np.gcd(win, slide))
-            partials = [np.float32(0)] * 0
-            segs = [seg[i:i + pane] for i in range(0, len(seg), pane)]
-            if comb == "sum":
-                partials = [np.sum(np.array(s, dtype=np.float32), dtype=np.float32)
-                            for s in segs]
-                r = float(np.sum(np.array(partials, dtype=np.float64)))
-            elif comb == "min":
-                r = float(min(min(s) for s in segs))
-            else:
-                r = float(max(max(s) for s in segs))
-            res.append(r)
+            res.append(_seg_comb(vs[w * slide: w * slide + win], win, slide,
+                                 comb))
+            w += 1
+        while w * slide < len(vs):  # EOS flush: partial tails
+            res.append(_seg_comb(vs[w * slide:], win, slide, comb))
             w += 1
         out[k] = res
     return out
@@ -146,19 +153,19 @@ def test_gpu_ffat_cb_sum_vs_oracle(use_tree):
     p.add_sink(snk)
     g.run()
     oracle = _ffat_oracle(n, n_keys, win, slide, vdt=2, comb="sum")
-    # group GPU outputs per key, ordered by ts (firing order per key)
+    # EOS-flush partials share the key's last ts, so compare per-key value
+    # multisets (sorted) + exact window counts
     from collections import defaultdict
     got = defaultdict(list)
     for k_arr, v_arr, t_arr in res['rows']:
-        for k, v, t in zip(k_arr.tolist(), v_arr.tolist(), t_arr.tolist()):
-            got[k].append((t, v))
+        for k, v in zip(k_arr.tolist(), v_arr.tolist()):
+            got[k].append(v)
     n_windows_oracle = sum(len(v) for v in oracle.values())
     n_windows_got = sum(len(v) for v in got.values())
     assert n_windows_got == n_windows_oracle
-    for k, pairs in got.items():
-        pairs.sort()
-        vals = [v for _, v in pairs]
-        ref = oracle[k]
+    for k, vals in got.items():
+        ref = sorted(oracle[k])
+        vals = sorted(vals)
         assert len(vals) == len(ref), f"key {k}"
         for a, b in zip(vals, ref):
             assert abs(a - b) <= 1e-3 * max(1.0, abs(b)), f"key {k}: {a} vs {b}"
@@ -188,12 +195,11 @@ def test_gpu_ffat_cb_min_tree_vs_oracle():
     from collections import defaultdict
     got = defaultdict(list)
     for k_arr, v_arr, t_arr in res['rows']:
-        for k, v, t in zip(k_arr.tolist(), v_arr.tolist(), t_arr.tolist()):
-            got[k].append((t, v))
+        for k, v in zip(k_arr.tolist(), v_arr.tolist()):
+            got[k].append(v)
     assert sum(len(v) for v in got.values()) == sum(len(v) for v in oracle.values())
-    for k, pairs in got.items():
-        pairs.sort()
-        for a, b in zip([v for _, v in pairs], oracle[k]):
+    for k, vals in got.items():
+        for a, b in zip(sorted(vals), sorted(oracle[k])):
             assert abs(a - b) <= 1e-5 * max(1.0, abs(b))
 
 
@@ -285,11 +291,12 @@ def test_gpu_ffat_high_key_count():
     snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).build()
     p.chain_sink(snk)
     g.run()
-    # oracle on fired-window COUNT only (sum oracle would be slow in python)
+    # oracle on fired-window COUNT only (sum oracle would be slow in python);
+    # every window start w*slide < c fires (EOS flush covers the partials)
     ts, key, val = gen_batch(n, 0, 42, n_keys, 2)
     from collections import Counter
     per = Counter(key.tolist())
-    exp = sum((c - win) // slide + 1 for c in per.values() if c >= win)
+    exp = sum((c - 1) // slide + 1 for c in per.values())
     assert g.sink_count(snk) == exp
 
 
@@ -377,6 +384,14 @@ def test_gpu_ffat_tb_vs_oracle():
                 [v for t, v in rows if w * slide <= t < w * slide + win],
                 dtype=np.float32), dtype=np.float64))
             exp[k].append((w * slide + win - 1, s))
+            w += 1
+        # EOS partial flush (windows with data panes but incomplete span)
+        # fire with ts = the key's last tuple ts
+        while (w * slide) // pane <= tmax // pane:
+            s = float(np.sum(np.array(
+                [v for t, v in rows if w * slide <= t < w * slide + win],
+                dtype=np.float32), dtype=np.float64))
+            exp[k].append((tmax, s))
             w += 1
     got = defaultdict(list)
     for k_arr, v_arr, t_arr in res['rows']:
@@ -480,7 +495,8 @@ def test_gpu_reduce_keyed_sum():
 
 def test_gpu_ffat_count_comb():
     """COUNT combiner through the FFAT folds (ring and tree): every full CB
-    window counts exactly `win` tuples; window count per key is exact."""
+    window counts exactly `win` tuples; EOS-flushed partial windows count
+    their tail (idx - w*slide); window multiset per key is exact."""
     n, n_keys, b, win, slide = 400_000, 101, 100_000, 300, 100
     for tree in (False, True):
         src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=2))
@@ -489,14 +505,11 @@ def test_gpu_ffat_count_comb():
             native_gpu.gpu_ffat_windows(native_gpu.COMB_COUNT, 0, win, slide,
                                         max_keys=256, use_tree=tree))
             .withOutputSchema([2]).withOutputBatchSize(b).build())
-        got = dict(rows=0, bad=0, per={})
+        got = dict(per={})
 
         def pysink(cols):
-            import numpy as _np
-            got['rows'] += len(cols['c0'])
-            got['bad'] += int((_np.abs(cols['c0'] - win) > 1e-3).sum())
-            for k in cols['key'].tolist():
-                got['per'][k] = got['per'].get(k, 0) + 1
+            for k, v in zip(cols['key'].tolist(), cols['c0'].tolist()):
+                got['per'].setdefault(k, []).append(int(round(v)))
 
         g = wf.PipeGraph("cnt")
         p = g.add_source(src)
@@ -505,12 +518,18 @@ def test_gpu_ffat_count_comb():
         snk.out_schema = [2]
         p.add_sink(snk)
         g.run()
-        assert got['bad'] == 0
         ts, key, val = gen_batch(n, 0, 42, n_keys, 2)
         import collections
         per = collections.Counter(key.tolist())
-        exp_windows = {k: max(0, (c - win) // slide + 1) for k, c in per.items()}
-        assert got['per'] == {k: v for k, v in exp_windows.items() if v > 0}
+        exp = {}
+        for k, c in per.items():
+            counts = [win] * (max(0, (c - win) // slide + 1))
+            w = len(counts)
+            while w * slide < c:
+                counts.append(c - w * slide)
+                w += 1
+            exp[k] = sorted(counts)
+        assert {k: sorted(v) for k, v in got['per'].items()} == exp
 
 
 def test_gpu_jit_expr_fuzz():
@@ -649,3 +668,271 @@ def test_gpu_mfma_gram_windows():
         m = np.array(sorted(d['rows']))
         ref = np.array(sorted(exp[(k, w)].tolist()))
         assert np.allclose(m, ref, rtol=1e-4, atol=1e-4), (k, w)
+
+
+def test_gpu_jit_avg_reduce_vs_oracle():
+    """Fused per-batch keyed AVG via the generalized JIT fold (user
+    lift/comb/finalize — reference arbitrary-combine Reduce_GPU)."""
+    n, n_keys, b = 100_000, 64, 25_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=2))
+           .withOutputSchema([2]).withOutputBatchSize(b).build())
+    rd = (Reduce_GPU_Builder(native_gpu.gpu_avg_reduce(0, max_keys=256))
+          .withOutputSchema([2]).withOutputBatchSize(b).build())
+    rows = []
+
+    def pysink(cols):
+        for k, v in zip(cols['key'].tolist(), cols['c0'].tolist()):
+            rows.append((k, v))
+
+    g = wf.PipeGraph("javg")
+    p = g.add_source(src)
+    p.chain(rd)
+    snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+    snk.out_schema = [2]
+    p.add_sink(snk)
+    g.run()
+    ts, key, val = gen_batch(n, 0, 42, n_keys, 2)
+    # one avg per (batch, key); compare per-key sorted multisets
+    from collections import defaultdict
+    exp = defaultdict(list)
+    for s in range(0, n, b):
+        kb, vb = key[s:s + b], val[s:s + b].astype(np.float64)
+        for k in set(kb.tolist()):
+            exp[k].append(float(vb[kb == k].mean()))
+    got = defaultdict(list)
+    for k, v in rows:
+        got[k].append(v)
+    assert sum(map(len, got.values())) == sum(map(len, exp.values()))
+    for k in exp:
+        for a, bb in zip(sorted(got[k]), sorted(exp[k])):
+            assert abs(a - bb) <= 2e-3 * max(1.0, abs(bb)), (k, a, bb)
+
+
+@pytest.mark.parametrize("win,slide", [(40, 10), (400, 100)])
+def test_gpu_jit_ffat_avg_vs_oracle(win, slide):
+    """Sliding-window AVG via the JIT fold — (sum,count) pane pairs with the
+    invertible running-total machine; pane 10 exercises the thread kernel,
+    pane 100 the wave kernel.  Includes EOS partial flush."""
+    n, n_keys, b = 120_000, 101, 17_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=2))
+           .withOutputSchema([2]).withOutputBatchSize(b).build())
+    ff = (Ffat_Windows_GPU_Builder(
+        native_gpu.gpu_avg_ffat_windows(win, slide, col=0, max_keys=1024))
+          .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
+    res = dict(rows=[])
+
+    def pysink(cols):
+        res['rows'].append((cols['key'].copy(), cols['c0'].copy()))
+
+    g = wf.PipeGraph("jffavg")
+    p = g.add_source(src)
+    p.chain(ff)
+    snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+    snk.out_schema = [2]
+    p.add_sink(snk)
+    g.run()
+    ts, key, val = gen_batch(n, 0, 42, n_keys, 2)
+    from collections import defaultdict
+    per = defaultdict(list)
+    for k, v in zip(key.tolist(), val.astype(np.float64).tolist()):
+        per[k].append(v)
+    exp = defaultdict(list)
+    for k, vs in per.items():
+        w = 0
+        while w * slide < len(vs):
+            seg = vs[w * slide: w * slide + win]
+            exp[k].append(float(np.mean(seg)))
+            w += 1
+    got = defaultdict(list)
+    for k_arr, v_arr in res['rows']:
+        for k, v in zip(k_arr.tolist(), v_arr.tolist()):
+            got[k].append(v)
+    assert sum(map(len, got.values())) == sum(map(len, exp.values()))
+    for k in exp:
+        for a, bb in zip(sorted(got[k]), sorted(exp[k])):
+            assert abs(a - bb) <= 2e-3 * max(1.0, abs(bb)), (k, a, bb)
+
+
+def test_gpu_jit_ffat_minmax_multicol():
+    """Multi-column JIT window fold: 2 value columns in, 2 result columns
+    out (min of c0, max of c1) through the non-invertible P-pane recombine
+    path (reference arbitrary result structs)."""
+    n, n_keys, win, slide, b = 80_000, 53, 60, 12, 11_000
+    rng = np.random.default_rng(7)
+    vals0 = rng.standard_normal(n).astype(np.float32)
+    vals1 = rng.standard_normal(n).astype(np.float32)
+    keys = rng.integers(0, n_keys, size=n).astype(np.uint64)
+    state = dict(pos=0)
+
+    def src(replica, par):
+        p = state['pos']
+        if p >= n:
+            return None
+        m = min(b, n - p)
+        state['pos'] += m
+        return dict(ts=np.arange(p, p + m, dtype=np.int64), key=keys[p:p + m],
+                    c0=vals0[p:p + m], c1=vals1[p:p + m], watermark=p + m)
+
+    ff = (Ffat_Windows_GPU_Builder(
+        lift="v0;v1",
+        comb="fminf(a0, b0);fmaxf(a1, b1)",
+        finalize="f0;f1",
+        identity=(float("inf"), float("-inf")), cols=(0, 1), max_keys=256)
+        .withCBWindows(win, slide)
+        .withOutputSchema([2, 2]).withOutputBatchSize(4 * b).build())
+    rows = []
+
+    def sink(cols):
+        for i in range(len(cols['ts'])):
+            rows.append((int(cols['key'][i]), float(cols['c0'][i]),
+                         float(cols['c1'][i])))
+
+    g = wf.PipeGraph("jmm")
+    mp = g.add_source(wf.Source_Builder(src).withParallelism(1)
+                      .withOutputSchema([2, 2]).withOutputBatchSize(b).build())
+    mp.chain(ff)
+    snk = wf.Sink_Builder(sink).withParallelism(1).build()
+    snk.out_schema = [2, 2]
+    mp.add_sink(snk)
+    g.run()
+    from collections import defaultdict
+    per = defaultdict(list)
+    for i in range(n):
+        per[int(keys[i])].append((float(vals0[i]), float(vals1[i])))
+    exp = defaultdict(list)
+    for k, vs in per.items():
+        w = 0
+        while w * slide < len(vs):
+            seg = vs[w * slide: w * slide + win]
+            exp[k].append((min(a for a, _ in seg), max(c for _, c in seg)))
+            w += 1
+    got = defaultdict(list)
+    for k, v0, v1 in rows:
+        got[k].append((v0, v1))
+    assert sum(map(len, got.values())) == sum(map(len, exp.values()))
+    for k in exp:
+        for (a0, a1), (b0, b1) in zip(sorted(got[k]), sorted(exp[k])):
+            assert abs(a0 - b0) <= 1e-5 and abs(a1 - b1) <= 1e-5, (k,)
+
+
+def test_gpu_jit_ffat_tb_avg_vs_oracle():
+    """Event-time JIT windows: AVG over TB panes + watermark advance +
+    EOS flush, vs brute-force oracle."""
+    n, n_keys, win, slide, b = 120_000, 101, 400, 100, 17_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=2))
+           .withOutputSchema([2]).withOutputBatchSize(b).build())
+    ff = (Ffat_Windows_GPU_Builder(
+        native_gpu.gpu_avg_ffat_windows(win, slide, col=0, max_keys=1024,
+                                        tb=True, pend_ring_log2=10))
+          .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
+    res = dict(rows=[])
+
+    def pysink(cols):
+        res['rows'].append((cols['key'].copy(), cols['c0'].copy()))
+
+    g = wf.PipeGraph("jtbavg")
+    p = g.add_source(src)
+    p.chain(ff)
+    snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+    snk.out_schema = [2]
+    p.add_sink(snk)
+    g.run()
+    ts, key, val = gen_batch(n, 0, 42, n_keys, 2)
+    pane = int(np.gcd(win, slide))
+    from collections import defaultdict
+    per = defaultdict(list)
+    for t, k, v in zip(ts.tolist(), key.tolist(), val.tolist()):
+        per[k].append((t, float(v)))
+    exp = defaultdict(list)
+    for k, rowsk in per.items():
+        tss = [t for t, _ in rowsk]
+        t0, tmax = min(tss), max(tss)
+        w = max(0, -(-(t0 - win + 1) // slide))
+        while (w * slide) // pane <= tmax // pane:
+            seg = [v for t, v in rowsk if w * slide <= t < w * slide + win]
+            exp[k].append(float(np.mean(seg)) if seg else float("nan"))
+            w += 1
+    got = defaultdict(list)
+    for k_arr, v_arr in res['rows']:
+        for k, v in zip(k_arr.tolist(), v_arr.tolist()):
+            got[k].append(v)
+    assert sum(map(len, got.values())) == sum(map(len, exp.values()))
+    for k in exp:
+        ga, ex = sorted(got[k]), sorted(exp[k])
+        for a, bb in zip(ga, ex):
+            if np.isnan(bb):
+                continue  # empty window: avg = 0/0 (engine NaN too)
+            assert abs(a - bb) <= 2e-3 * max(1.0, abs(bb)), (k, a, bb)
+
+
+def test_gpu_jit_stateful_map_filter():
+    """Arbitrary stateful device bodies (reference stateful MAP/FILTER_GPU
+    functors): EMA map + every-3rd filter, JIT-compiled."""
+    n, n_keys, b = 100_000, 64, 25_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=2))
+           .withOutputSchema([2]).withOutputBatchSize(b).build())
+    sm = (Map_GPU_Builder(native_gpu.gpu_jit_stateful_map(
+        "s0 = 0.5 * s0 + 0.5 * (double)v0; v0 = (float)s0", max_keys=256))
+          .withOutputSchema([2]).withOutputBatchSize(b).build())
+    fl = (Filter_GPU_Builder(native_gpu.gpu_jit_stateful_filter(
+        "s0 = s0 + 1.0; keep = ((i64)s0 % 3) == 0", max_keys=256))
+          .withOutputSchema([2]).withOutputBatchSize(b).build())
+    acc = dict(s=0.0, n=0)
+
+    def pysink(cols):
+        acc['s'] += float(cols['c0'].astype(np.float64).sum())
+        acc['n'] += len(cols['c0'])
+
+    g = wf.PipeGraph("jst")
+    p = g.add_source(src)
+    p.chain(sm)
+    p.chain(fl)
+    snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+    snk.out_schema = [2]
+    p.add_sink(snk)
+    g.run()
+    ts, key, val = gen_batch(n, 0, 42, n_keys, 2)
+    ema = {}
+    cnt = {}
+    s = 0.0
+    m = 0
+    for k, v in zip(key.tolist(), val.astype(np.float32).tolist()):
+        e = 0.5 * ema.get(k, 0.0) + 0.5 * float(np.float32(v))
+        ema[k] = e
+        c = cnt.get(k, 0) + 1
+        cnt[k] = c
+        if c % 3 == 0:
+            s += float(np.float32(e))
+            m += 1
+    assert acc['n'] == m
+    assert abs(acc['s'] - s) <= 2e-3 * max(1.0, abs(s))
+
+
+def test_gpu_ffat_vik_fallback_key_overflow():
+    """Automatic VIK fallback: bf16 values but max_keys > 65535 forces the
+    indirect-gather path; results must match the VIK-eligible config."""
+    n, n_keys, win, slide, b = 200_000, 101, 500, 100, 50_000
+    sums = []
+    for mk in (1024, 70_000):  # VIK on / off
+        src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=5))
+               .withOutputSchema([5]).withOutputBatchSize(b).build())
+        ff = (Ffat_Windows_GPU_Builder(
+            native_gpu.gpu_ffat_windows(native_gpu.COMB_SUM, 0, win, slide,
+                                        max_keys=mk))
+              .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
+        acc = dict(s=0.0, n=0)
+
+        def pysink(cols):
+            acc['s'] += float(cols['c0'].astype(np.float64).sum())
+            acc['n'] += len(cols['c0'])
+
+        g = wf.PipeGraph(f"vik{mk}")
+        p = g.add_source(src)
+        p.chain(ff)
+        snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+        snk.out_schema = [2]
+        p.add_sink(snk)
+        g.run()
+        sums.append((acc['n'], acc['s']))
+    assert sums[0][0] == sums[1][0]
+    assert abs(sums[0][1] - sums[1][1]) <= 1e-6 * max(1.0, abs(sums[0][1]))

@@ -13,6 +13,8 @@ import numpy as np
 sys.path.insert(0, ".")
 
 # (comb, form, win, slide, n_keys, n, batch, vdt)
+# combs: sum/min/max/count = native catalog; avg/jmin/jminmax = hiprtc JIT
+# user folds (gpu_jit_ffat — round-2 arbitrary lift/comb surface)
 CASES = [
     ("sum",   "cb",   500, 100, 101,  1_000_000, 250_000, 5),
     ("min",   "cb",   300, 300, 16,     400_000, 100_000, 2),
@@ -22,6 +24,10 @@ CASES = [
     ("sum",   "tb",  1000, 500, 101,  1_000_000, 250_000, 2),
     ("max",   "tb",   900, 300, 16,     400_000, 100_000, 2),
     ("sum",   "tree", 800, 400, 1024, 1_000_000, 250_000, 5),
+    ("avg",   "cb",   500, 100, 101,  1_000_000, 250_000, 2),
+    ("avg",   "tb",   600, 300, 64,     400_000, 100_000, 2),
+    ("jmin",  "cb",   300, 60,  101,    400_000, 100_000, 2),
+    ("avg",   "cb",    24, 8,   64,     400_000, 100_000, 2),  # thread kernel
 ]
 
 
@@ -35,11 +41,22 @@ def engine():
     for ci, (comb, form, win, slide, n_keys, n, b, vdt) in enumerate(CASES):
         src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=vdt))
                .withOutputSchema([vdt]).withOutputBatchSize(b).build())
-        ff = (Ffat_Windows_GPU_Builder(
-            native_gpu.gpu_ffat_windows(COMBS[comb], 0, win, slide,
-                                        max_keys=2 * n_keys,
-                                        use_tree=form == "tree",
-                                        tb=form == "tb", pend_ring_log2=13))
+        if comb == "avg":
+            logic = native_gpu.gpu_avg_ffat_windows(
+                win, slide, col=0, max_keys=2 * n_keys, tb=form == "tb",
+                pend_ring_log2=13)
+        elif comb == "jmin":
+            logic = native_gpu.gpu_jit_ffat_windows(
+                win, slide, lift="v0", comb="fminf(a0, b0)", finalize="f0",
+                identity=(float("inf"),), max_keys=2 * n_keys,
+                tb=form == "tb", pend_ring_log2=13)
+        else:
+            logic = native_gpu.gpu_ffat_windows(COMBS[comb], 0, win, slide,
+                                                max_keys=2 * n_keys,
+                                                use_tree=form == "tree",
+                                                tb=form == "tb",
+                                                pend_ring_log2=13)
+        ff = (Ffat_Windows_GPU_Builder(logic)
             .withOutputSchema([2]).withOutputBatchSize(4 * b).build())
         keys, vals = [], []
 
@@ -77,7 +94,9 @@ def check():
         f = {"sum": lambda a: float(np.sum(np.asarray(a, np.float32),
                                            dtype=np.float64)),
              "min": lambda a: float(min(a)), "max": lambda a: float(max(a)),
-             "count": lambda a: float(len(a))}[comb]
+             "count": lambda a: float(len(a)),
+             "avg": lambda a: float(np.mean(np.asarray(a, np.float64))),
+             "jmin": lambda a: float(min(a))}[comb]
         exp = defaultdict(list)
         for k, rows in per.items():
             if form != "tb":
@@ -86,12 +105,19 @@ def check():
                 while w * slide + win <= len(v2):
                     exp[k].append(f(v2[w * slide: w * slide + win]))
                     w += 1
+                while w * slide < len(v2):   # EOS partial flush
+                    exp[k].append(f(v2[w * slide:]))
+                    w += 1
             else:
                 pane = int(np.gcd(win, slide))
                 tss = [t for t, _ in rows]
                 t0, tmax = min(tss), max(tss)
                 w = max(0, -(-(t0 - win + 1) // slide))
                 while (w * slide + win - 1) // pane <= tmax // pane:
+                    seg = [v for t, v in rows if w * slide <= t < w * slide + win]
+                    exp[k].append(f(seg) if seg else 0.0)
+                    w += 1
+                while (w * slide) // pane <= tmax // pane:  # EOS partial flush
                     seg = [v for t, v in rows if w * slide <= t < w * slide + win]
                     exp[k].append(f(seg) if seg else 0.0)
                     w += 1

@@ -51,18 +51,56 @@ class KeyBy_Exchange_GPU_Builder(_GpuBuilder):
 
 
 class Ffat_Windows_GPU_Builder(_GpuBuilder):
-    """reference builders_gpu.hpp:466 (+withNumWinPerBatch :576)."""
+    """reference builders_gpu.hpp:466 (+withNumWinPerBatch :576).
+
+    Accepts either a compiled native_gpu.* spec (func) or an arbitrary
+    user fold via keyword expressions (reference's arbitrary lift/comb
+    __device__ functors):
+        Ffat_Windows_GPU_Builder(comb="fminf(a0,b0)", identity=(float('inf'),))
+        Ffat_Windows_GPU_Builder(lift="v0;1.0f", comb="a0+b0;a1+b1",
+                                 finalize="f0/f1", identity=(0, 0),
+                                 invertible=True)          # AVG
+    then .withCBWindows/withTBWindows sets the extent."""
     _kind = "gpu_ffat"
 
-    def __init__(self, func=None):
+    def __init__(self, func=None, lift=None, comb=None, finalize=None,
+                 identity=None, cols=(0,), invertible=False,
+                 max_keys=1 << 16):
+        if func is None and comb is not None:
+            from . import native_gpu
+            lift = lift if lift is not None else "v0"
+            finalize = finalize if finalize is not None else "f0"
+            identity = identity if identity is not None else (0.0,)
+            func = native_gpu.gpu_jit_ffat_windows(
+                1, 1, lift=lift, comb=comb, finalize=finalize,
+                identity=identity, cols=cols, invertible=invertible,
+                max_keys=max_keys)
         super().__init__(func)
         self._op.window = dict(type=0, win=0, slide=0, lateness=0)
 
-    def withCBWindows(self, win_len, slide_len):
-        self._op.window.update(type=0, win=int(win_len), slide=int(slide_len))
+    # iparam offsets of (win, slide, wintype, lateness) per logic kind
+    _IP = {"gpu_ffat": (2, 3, 6, 7), "gpu_jit_ffat": (8, 9, 10, 11)}
+
+    def _set_win(self, wt, win, slide):
+        self._op.window.update(type=wt, win=int(win), slide=int(slide))
         if self._op.logic is not None:
+            iw, isl, it, _ = self._IP[self._op.logic.kind]
             ip = self._op.logic.iparams
-            ip[2], ip[3] = int(win_len), int(slide_len)
+            ip[iw], ip[isl], ip[it] = int(win), int(slide), wt
+
+    def withCBWindows(self, win_len, slide_len):
+        self._set_win(0, win_len, slide_len)
+        return self
+
+    def withTBWindows(self, win_us, slide_us):
+        self._set_win(1, win_us, slide_us)
+        return self
+
+    def withLateness(self, lateness_us):
+        self._op.window['lateness'] = int(lateness_us)
+        if self._op.logic is not None:
+            _, _, _, il = self._IP[self._op.logic.kind]
+            self._op.logic.iparams[il] = int(lateness_us)
         return self
 
     def withNumWinPerBatch(self, n):

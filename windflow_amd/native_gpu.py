@@ -47,6 +47,97 @@ def gpu_jit_filter(expr, col=0):
     return NativeLogic("gpu_jit_filter", expr, [], [col])
 
 
+def _expr_list(x):
+    if isinstance(x, str):
+        return x.split(';')
+    return list(x)
+
+
+def _jit_fold_spec(lift, comb, finalize, identity, cols):
+    """Normalize a user fold into (spec, fparams, iparam-prefix).
+
+    lift/comb/finalize: str (';'-joined C expressions) or list of str.
+    lift sees v0..v{len(cols)-1} (float), ts (i64), key (u64); comb sees
+    a0.., b0..; finalize sees f0..  identity: one float per field.
+    Parity with the reference's arbitrary __device__ lift/combine functors
+    (wf/builders_gpu.hpp:466-620, meta_gpu.hpp).  Combine must be
+    associative AND commutative (folds are tree/wave shaped)."""
+    lift, comb, fin = _expr_list(lift), _expr_list(comb), _expr_list(finalize)
+    identity = [float(v) for v in identity]
+    cols = [int(c) for c in cols]
+    nf, nout, nc = len(lift), len(fin), len(cols)
+    if len(comb) != nf or len(identity) != nf:
+        raise ValueError("lift/comb/identity must have the same field count")
+    if not 1 <= nc <= 4:
+        raise ValueError("1..4 value columns")
+    if not 1 <= nout <= 4:
+        raise ValueError("1..4 outputs")
+    spec = "\x1e".join([";".join(lift), ";".join(comb), ";".join(fin)])
+    cols4 = (cols + [0, 0, 0, 0])[:4]
+    return spec, identity, [nf, nout, nc] + cols4
+
+
+def gpu_jit_reduce(lift="v0", comb="a0+b0", finalize="f0", identity=(0.0,),
+                   cols=(0,), max_keys=1 << 16):
+    """Per-batch keyed reduction with an ARBITRARY user fold, runtime-
+    compiled with hiprtc (reference Reduce_GPU accepts any __device__
+    combine, builders_gpu.hpp:350).  Emits one row per distinct key:
+    (key, finalize(acc) as F32 columns, ts_max)."""
+    spec, fp, pre = _jit_fold_spec(lift, comb, finalize, identity, cols)
+    return NativeLogic("gpu_jit_reduce", spec, fp, pre + [int(max_keys)])
+
+
+def gpu_avg_reduce(col=0, max_keys=1 << 16):
+    """Keyed AVG: fused (sum, count) fold (see FUTURE.md round 1)."""
+    return gpu_jit_reduce(lift="v0;1.0f", comb="a0+b0;a1+b1", finalize="f0/f1",
+                          identity=(0.0, 0.0), cols=(col,), max_keys=max_keys)
+
+
+def gpu_jit_ffat_windows(win=1000, slide=100, lift="v0", comb="a0+b0",
+                         finalize="f0", identity=(0.0,), cols=(0,),
+                         max_keys=1 << 16, tb=False, lateness=0,
+                         pend_ring_log2=0, invertible=False):
+    """Keyed sliding window with an ARBITRARY user lift+combine+finalize
+    fold (reference Ffat_Windows_GPU arbitrary lift/comb functors,
+    ffat_windows_gpu.hpp:60).  Multi-column values (cols), multi-column
+    results (finalize list) and multi-field accumulators all supported;
+    invertible=True asserts comb is fieldwise + (running-total windows
+    instead of P-pane recombines — use for sum/avg-shaped folds)."""
+    if win < 1 or slide < 1:
+        raise ValueError("window length and slide must be >= 1")
+    spec, fp, pre = _jit_fold_spec(lift, comb, finalize, identity, cols)
+    ip = pre + [int(max_keys), int(win), int(slide), 1 if tb else 0,
+                int(lateness), int(pend_ring_log2), 1 if invertible else 0]
+    return NativeLogic("gpu_jit_ffat", spec, fp, ip)
+
+
+def gpu_avg_ffat_windows(win=1000, slide=100, col=0, **kw):
+    """Keyed sliding-window AVG: fused (sum, count) pane pairs."""
+    return gpu_jit_ffat_windows(win, slide, lift="v0;1.0f",
+                                comb="a0+b0;a1+b1", finalize="f0/f1",
+                                identity=(0.0, 0.0), cols=(col,),
+                                invertible=True, **kw)
+
+
+def gpu_jit_stateful_map(body, ncols=1, nstate=1, init=(0.0,),
+                         max_keys=1 << 16):
+    """Keyed stateful map with an ARBITRARY statement body (reference
+    stateful MAP_GPU functors, map_gpu.hpp:80-102).  body sees v0..v{n}
+    (mutable float, written back in place), ts (i64), key (u64), s0..s{m}
+    (mutable f64 per-key state), and runs per tuple in key order, e.g.
+    "s0 = s0 + v0; v0 = s0" (running sum)."""
+    return NativeLogic("gpu_jit_stateful", body, [float(v) for v in init],
+                       [0, int(ncols), int(nstate), int(max_keys)])
+
+
+def gpu_jit_stateful_filter(body, ncols=1, nstate=1, init=(0.0,),
+                            max_keys=1 << 16):
+    """Keyed stateful filter: body sets `keep` (int), e.g.
+    "keep = v0 != s0; s0 = v0" (consecutive dedup)."""
+    return NativeLogic("gpu_jit_stateful", body, [float(v) for v in init],
+                       [1, int(ncols), int(nstate), int(max_keys)])
+
+
 def gpu_keyed_running_sum(col=0, max_keys=1 << 16):
     """stateful map: per-key running sum written in place (key-order walk)."""
     return NativeLogic("gpu_map_keyed", "", [0.0, 0.0], [1, col, max_keys])
