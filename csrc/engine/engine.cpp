@@ -158,9 +158,28 @@ int64_t OrderingCollector::released_wm() const {
     return m == WM_MAX ? chan_max_wm() : m;
 }
 
+// carve rows [o, o+m) of h into a fresh batch from out_pool
+Batch* OrderingCollector::carve(Batch* h, int64_t o, int64_t m) {
+    Batch* sb = out_pool->get();
+    if (m > sb->capacity)
+        throw std::runtime_error("ordering collector carve > pool capacity");
+    memcpy(sb->ts, h->ts + o, 8 * m);
+    memcpy(sb->key, h->key + o, 8 * m);
+    const size_t np = h->schema.payload.size();
+    for (size_t c = 0; c < np && c < sb->cols.size(); ++c) {
+        size_t es = dsize(h->schema.payload[c]);
+        memcpy((char*)sb->cols[c] + 0, (char*)h->cols[c] + o * es, es * m);
+    }
+    sb->count = m;
+    return sb;
+}
+
 Batch* OrderingCollector::next() {
-    if (pend.empty()) pend.resize(chans.size());
-    if (rel_wm.empty()) rel_wm.assign(chans.size(), 0);
+    if (pend.empty()) {
+        pend.resize(chans.size());
+        off.assign(chans.size(), 0);
+        rel_wm.assign(chans.size(), 0);
+    }
     int spins = 0;
     for (;;) {
         if (abort && abort->load(std::memory_order_relaxed)) return nullptr;
@@ -178,72 +197,92 @@ Batch* OrderingCollector::next() {
                 pend[c].push_back(b);
             }
         }
-        // release the batch with the smallest head timestamp, but only when
-        // every open channel has something pending (or is closed)
+        // pick the channel with the smallest head-ROW timestamp; release
+        // only when every open channel has something pending (or at EOS)
         bool all_ready = true;
         int best = -1;
         int64_t best_ts = WM_MAX;
+        auto head_ts = [&](size_t c) {
+            Batch* h = pend[c].front();
+            if (h->punct || h->count == 0) return h->watermark;
+            return h->ts[off[c]];
+        };
         for (size_t c = 0; c < chans.size(); ++c) {
             if (pend[c].empty()) {
                 if (open[c]) all_ready = false;
                 continue;
             }
-            Batch* h = pend[c].front();
-            int64_t hts = h->punct ? h->watermark : (h->count ? h->ts[0] : h->watermark);
+            int64_t hts = head_ts(c);
             if (hts < best_ts) {
                 best_ts = hts;
                 best = (int)c;
             }
         }
-        if (best >= 0 && all_ready) {
-            Batch* b = pend[best].front();
-            pend[best].pop_front();
-            rel_wm[best] = std::max(rel_wm[best], b->watermark);
-            delivered_wm = released_wm();
-            delivered_tag = b->stream_tag >= 0 ? b->stream_tag : chan_tag[best];
-            if (b->refcnt.load(std::memory_order_acquire) == 1) {
-                b->watermark = delivered_wm;
-                b->stream_tag = delivered_tag;
-            }
-            if (b->punct && delivered_wm <= last_fwd_wm) {
-                release(b);
-                continue;
-            }
-            last_fwd_wm = std::max(last_fwd_wm, delivered_wm);
-            return b;
-        }
-        if (n_open == 0) {
-            // flush remaining in ts order
-            best = -1;
-            best_ts = WM_MAX;
-            for (size_t c = 0; c < chans.size(); ++c) {
-                if (pend[c].empty()) continue;
-                Batch* h = pend[c].front();
-                int64_t hts = h->punct ? h->watermark : (h->count ? h->ts[0] : h->watermark);
-                if (hts < best_ts) {
-                    best_ts = hts;
-                    best = (int)c;
+        const bool flush = (n_open == 0);
+        if (best >= 0 && (all_ready || flush)) {
+            // merge frontier: rows of `best` may be released while their ts
+            // does not exceed any other channel's next available row
+            int64_t frontier = WM_MAX;
+            for (size_t c = 0; c < chans.size(); ++c)
+                if ((int)c != best && !pend[c].empty())
+                    frontier = std::min(frontier, head_ts(c));
+            Batch* h = pend[best].front();
+            int64_t o = off[best];
+            int64_t m = 0;
+            if (!h->punct)
+                while (o + m < h->count && h->ts[o + m] <= frontier) ++m;
+            const bool whole = h->punct || (o == 0 && o + m == h->count);
+            // device batches can't be carved on host (their producers are
+            // DEFAULT-mode, so this path is CPU in practice); schema-
+            // mismatched merges release whole batches too
+            if (whole || !out_pool || h->loc == Loc::DEVICE ||
+                !(h->schema == out_pool->schema)) {
+                pend[best].pop_front();
+                off[best] = 0;
+                rel_wm[best] = std::max(rel_wm[best], h->watermark);
+                delivered_wm = released_wm();
+                delivered_tag = h->stream_tag >= 0 ? h->stream_tag : chan_tag[best];
+                if (h->refcnt.load(std::memory_order_acquire) == 1) {
+                    h->watermark = delivered_wm;
+                    h->stream_tag = delivered_tag;
                 }
+                if (h->punct && delivered_wm <= last_fwd_wm) {
+                    release(h);
+                    continue;
+                }
+                last_fwd_wm = std::max(last_fwd_wm, delivered_wm);
+                return h;
             }
-            if (best < 0) return nullptr;
-            Batch* b = pend[best].front();
-            pend[best].pop_front();
-            rel_wm[best] = std::max(rel_wm[best], b->watermark);
+            // partial: carve the safe prefix, keep the rest pending
+            Batch* sb = carve(h, o, m);
+            const int h_tag = h->stream_tag;   // read before release(h)
+            off[best] += m;
+            if (off[best] == h->count) {
+                pend[best].pop_front();
+                off[best] = 0;
+                rel_wm[best] = std::max(rel_wm[best], h->watermark);
+                release(h);
+            } else {
+                // rows beyond the frontier stay pending: cap the channel's
+                // released-wm at the frontier
+                rel_wm[best] = std::max(rel_wm[best],
+                                        std::min(h->watermark, frontier));
+            }
             delivered_wm = released_wm();
-            delivered_tag = b->stream_tag >= 0 ? b->stream_tag : chan_tag[best];
-            if (b->refcnt.load(std::memory_order_acquire) == 1) {
-                b->watermark = delivered_wm;
-                b->stream_tag = delivered_tag;
-            }
-            return b;
+            delivered_tag = h_tag >= 0 ? h_tag : chan_tag[best];
+            sb->watermark = delivered_wm;
+            sb->stream_tag = delivered_tag;
+            last_fwd_wm = std::max(last_fwd_wm, delivered_wm);
+            return sb;
         }
+        if (flush && best < 0) return nullptr;
         SpscQueue::backoff(spins);
     }
 }
 
-Batch* KSlackCollector::next() {
-    // Batched K-slack: buffer (first_ts, batch); release when
-    // first_ts <= t_curr - K.  K adapts to max observed disorder.
+// Batch-granularity fallback (device batches / no pool): buffer
+// (first_ts, batch); release when first_ts <= t_curr - K.
+Batch* KSlackCollector::next_batchwise() {
     int spins = 0;
     for (;;) {
         if (abort && abort->load(std::memory_order_relaxed)) return nullptr;
@@ -262,10 +301,6 @@ Batch* KSlackCollector::next() {
                     continue;
                 }
                 int64_t bts = b->count ? b->ts[0] : b->watermark;
-                // late beyond the CURRENT slack behind what was already
-                // released: drop + account (reference kslack_collector.hpp
-                // checks against the pre-adaptation K and feeds the
-                // PipeGraph's atomic_num_dropped)
                 if (bts + K < last_rel_ts) {
                     if (dropped)
                         dropped->fetch_add(b->count, std::memory_order_relaxed);
@@ -281,7 +316,6 @@ Batch* KSlackCollector::next() {
             }
         }
         if (!buf.empty()) {
-            // release oldest if past slack horizon (or everything at EOS)
             auto it = std::min_element(buf.begin(), buf.end(),
                                        [](auto& a, auto& b) { return a.first < b.first; });
             if (n_open == 0 || it->first <= t_curr - K) {
@@ -298,6 +332,112 @@ Batch* KSlackCollector::next() {
         } else if (n_open == 0) {
             return nullptr;
         }
+        SpscQueue::backoff(spins);
+    }
+}
+
+// Per-tuple K-slack (reference kslack_collector.hpp:52): rows are buffered
+// individually in a ts min-heap, released in sorted order once past the
+// slack horizon t_curr - K, and rebuilt into fresh batches.  Rows behind
+// the last released ts are dropped and counted per tuple.
+Batch* KSlackCollector::next() {
+    if (!out_pool) return next_batchwise();
+    int spins = 0;
+    for (;;) {
+        if (abort && abort->load(std::memory_order_relaxed)) return nullptr;
+        bool saw_device = false;
+        for (size_t c = 0; c < chans.size(); ++c) {
+            if (!open[c]) continue;
+            Batch* b;
+            while ((b = chans[c]->try_pop()) != nullptr) {
+                if (b == EOS_TAG) {
+                    open[c] = false;
+                    n_open--;
+                    break;
+                }
+                chan_wm[c] = std::max(chan_wm[c], b->watermark);
+                if (b->punct) {
+                    release(b);
+                    continue;
+                }
+                if (b->loc == Loc::DEVICE || !(b->schema == out_pool->schema)) {
+                    // device rows are not host-addressable (and schema-
+                    // mismatched merges can't share one rebuild pool):
+                    // hand such batches through as one unit
+                    saw_device = true;
+                    int64_t bts = b->count ? 0 : b->watermark;
+                    buf.emplace_back(bts, b);
+                    continue;
+                }
+                int64_t kept = 0;
+                for (int64_t i = 0; i < b->count; ++i) {
+                    int64_t ts = b->ts[i];
+                    if (ts < last_rel_ts) {
+                        // late beyond what was already released
+                        if (dropped)
+                            dropped->fetch_add(1, std::memory_order_relaxed);
+                        continue;
+                    }
+                    if (ts < t_curr && t_curr - ts > K) K = t_curr - ts;
+                    t_curr = std::max(t_curr, ts);
+                    heap.push_back({ts, seq++, b, i});
+                    std::push_heap(heap.begin(), heap.end(), std::greater<>());
+                    ++kept;
+                }
+                if (kept == 0)
+                    release(b);
+                else
+                    remaining[b] = kept;
+            }
+        }
+        (void)saw_device;
+        if (!buf.empty()) {
+            // pass-through batches (device / foreign schema), FIFO
+            Batch* b = buf.front().second;
+            buf.erase(buf.begin());
+            delivered_wm = std::max(last_fwd_wm, b->watermark);
+            delivered_tag = b->stream_tag;
+            last_fwd_wm = delivered_wm;
+            return b;
+        }
+        const int64_t horizon = (n_open == 0) ? WM_MAX : t_curr - K;
+        while (!heap.empty() && heap.front().ts <= horizon) {
+            if (!open_out) open_out = out_pool->get();
+            Row r = heap.front();
+            std::pop_heap(heap.begin(), heap.end(), std::greater<>());
+            heap.pop_back();
+            Batch* src = r.b;
+            int64_t w = open_out->count;
+            open_out->ts[w] = src->ts[r.i];
+            open_out->key[w] = src->key[r.i];
+            const size_t np = src->schema.payload.size();
+            for (size_t cc = 0; cc < np && cc < open_out->cols.size(); ++cc) {
+                size_t es = dsize(src->schema.payload[cc]);
+                memcpy((char*)open_out->cols[cc] + w * es,
+                       (char*)src->cols[cc] + r.i * es, es);
+            }
+            open_out->count = w + 1;
+            last_rel_ts = std::max(last_rel_ts, r.ts);
+            auto it = remaining.find(src);
+            if (--it->second == 0) {
+                remaining.erase(it);
+                release(src);
+            }
+            if (open_out->count == open_out->capacity) break;
+        }
+        if (open_out && open_out->count > 0 &&
+            (heap.empty() || heap.front().ts > horizon ||
+             open_out->count == open_out->capacity)) {
+            Batch* b = open_out;
+            open_out = nullptr;
+            b->watermark = last_rel_ts;
+            delivered_wm = std::max(last_fwd_wm, b->watermark);
+            delivered_tag = b->stream_tag;
+            b->watermark = delivered_wm;
+            last_fwd_wm = delivered_wm;
+            return b;
+        }
+        if (n_open == 0 && heap.empty() && !open_out) return nullptr;
         SpscQueue::backoff(spins);
     }
 }
@@ -506,16 +646,24 @@ void Engine::build() {
             ch->wire(op_pools[rep->op_id], rep->ectx, rep->rctx);
         if (!rep->logic->is_source()) {
             CollectorKind ck = CollectorKind::WATERMARK;
+            Pool* in_pool = nullptr;  // producer-side pool: input-shaped
             for (auto& es : edges)
-                if (es.to == rep->op_id) ck = es.collector;
+                if (es.to == rep->op_id) {
+                    ck = es.collector;
+                    if (!in_pool) in_pool = op_pool[es.from];
+                }
             int g = rep_gid(rep->op_id, rep->idx);
             switch (ck) {
-                case CollectorKind::ORDERING:
-                    rep->collector = std::make_unique<OrderingCollector>(in_chans[g], in_tags[g]);
+                case CollectorKind::ORDERING: {
+                    auto oc = std::make_unique<OrderingCollector>(in_chans[g], in_tags[g]);
+                    oc->out_pool = in_pool;  // per-tuple carve buffers
+                    rep->collector = std::move(oc);
                     break;
+                }
                 case CollectorKind::KSLACK: {
                     auto ks = std::make_unique<KSlackCollector>(in_chans[g], in_tags[g]);
                     ks->dropped = &dropped_tuples;
+                    ks->out_pool = in_pool;  // per-tuple rebuild buffers
                     rep->collector = std::move(ks);
                     break;
                 }

@@ -225,31 +225,55 @@ struct Collector {
     virtual Batch* next();
 };
 
-// ORDERING (DETERMINISTIC mode): releases batches in global timestamp order
-// across channels — a k-way merge; a channel must be non-empty (or closed)
-// before anything is released (reference: wf/ordering_collector.hpp).
+// ORDERING (DETERMINISTIC mode): releases rows in global timestamp order
+// across channels — a TUPLE-granular k-way merge (reference
+// wf/ordering_collector.hpp:51 orders per-tuple messages).  Batches whose
+// rows all precede every other channel's head pass through untouched (the
+// common ts-contiguous case, zero copy); otherwise the head batch is SPLIT
+// at the merge boundary and the safe prefix is carved into a fresh batch.
+// A channel must be non-empty (or closed) before anything is released.
 struct OrderingCollector : Collector {
     std::vector<std::deque<Batch*>> pend;
-    // wm of the last *released* batch per channel: a released batch must not
-    // carry watermark knowledge from batches still buffered in pend (that
+    std::vector<int64_t> off;   // consumed-row offset into each head batch
+    // wm of the last *released* rows per channel: a released batch must not
+    // carry watermark knowledge from rows still buffered in pend (that
     // would fire downstream windows before their content arrives)
     std::vector<int64_t> rel_wm;
+    Pool* out_pool = nullptr;   // sub-batch carving; null => batch granularity
     using Collector::Collector;
     int64_t released_wm() const;
+    Batch* carve(Batch* h, int64_t o, int64_t m);
     Batch* next() override;
 };
 
-// KSLACK (PROBABILISTIC mode): release buffered batches up to wm estimate
-// t_curr - K, adapting K to observed disorder; late batches are dropped and
-// counted (reference: wf/kslack_collector.hpp).
+// KSLACK (PROBABILISTIC mode): buffer TUPLES, release them in ts order up
+// to the slack horizon t_curr - K, adapting K to observed disorder; late
+// tuples (behind the last released ts) are dropped and counted per tuple
+// (reference: wf/kslack_collector.hpp:52 buffers/drops per tuple).
 struct KSlackCollector : Collector {
     int64_t K = 0;
     int64_t t_curr = 0;
     int64_t last_rel_ts = INT64_MIN;
-    std::vector<std::pair<int64_t, Batch*>> buf;
+    // min-heap of buffered rows (ts, seq for stable order, batch, row idx)
+    struct Row {
+        int64_t ts;
+        uint64_t seq;
+        Batch* b;
+        int64_t i;
+        bool operator>(const Row& o) const {
+            return ts != o.ts ? ts > o.ts : seq > o.seq;
+        }
+    };
+    std::vector<Row> heap;                        // std::push/pop_heap, min
+    std::unordered_map<Batch*, int64_t> remaining;  // unreleased rows per batch
+    uint64_t seq = 0;
+    Pool* out_pool = nullptr;   // rebuilt output batches; null => batch mode
+    Batch* open_out = nullptr;
+    std::vector<std::pair<int64_t, Batch*>> buf;  // batch-granularity fallback
     std::atomic<int64_t>* dropped = nullptr;
     using Collector::Collector;
     Batch* next() override;
+    Batch* next_batchwise();    // fallback for device batches / no pool
 };
 
 // ===== operator logic =====

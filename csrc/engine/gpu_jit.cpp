@@ -382,7 +382,7 @@ static std::string gen_cb_flush(const JitFoldSpec&) {
         "    const u32* n_slots, i64 P, i64 S, int ring_log2,\n"
         "    const u32* st_fill, const float* st_acc, const float* ring,\n"
         "    const u32* st_head, const i64* st_last, const u64* slot_to_key,\n"
-        "    const u32* nf, u64* out_key,\n"
+        "    const u32* nf, i64 out_base, u64* out_key,\n"
         "    float* o0, float* o1, float* o2, float* o3, i64* out_ts,\n"
         "    i64 out_cap) {\n"
         "    const i64 ns = *n_slots;\n"
@@ -402,7 +402,8 @@ static std::string gen_cb_flush(const JitFoldSpec&) {
         "            for (u64 p = q * (u64)S; p < head; ++p)\n"
         "                res = jcomb(res, acc_load(rg + (usz)((u32)p & Rm) * NF));\n"
         "            if (part) res = jcomb(res, acc_load(st_acc + (usz)s * NF));\n"
-        "            if (w < out_cap) JEMIT(w, res, st_last ? st_last[s] : 0);\n"
+        "            if (w >= out_base && w - out_base < out_cap)\n"
+        "                JEMIT(w - out_base, res, st_last ? st_last[s] : 0);\n"
         "            ++w;\n"
         "        }\n"
         "    }\n"
@@ -919,31 +920,40 @@ struct GpuJitFfatLogic : GpuLogicBase {
     }
 
     void flush_open(bool with_open_pane, EmitCtx& out, RuntimeCtx& ctx) {
-        Batch* ob = get_dev();
         const uint32_t* fill = with_open_pane ? st_fill : nullptr;
         const float* acc = with_open_pane ? st_acc : nullptr;
         wfa_cb_flush_count(stream, ks.d_nslots, P, S, fill, st_head, nf);
         wfa_slot_scan(stream, nf, ks.d_nslots, d_on);
-        float* o[4];
-        for (int m = 0; m < 4; ++m)
-            o[m] = (float*)ob->cols[std::min<size_t>(m, ob->cols.size() - 1)];
-        int rl = ring_log2;
-        ArgPack a;
-        a.add(ks.d_nslots); a.add(P); a.add(S); a.add(rl);
-        a.add(fill); a.add(acc); a.add(ring); a.add(st_head); a.add(st_last);
-        a.add(ks.slot_to_key); a.add(nf);
-        a.add(ob->key); a.add(o[0]); a.add(o[1]); a.add(o[2]); a.add(o[3]);
-        a.add(ob->ts);
-        int64_t cap = ob->capacity;
-        a.add(cap);
-        launch(f_flush, stream, 256, a);
-        HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost,
-                              stream));
-        ob->count = -1;
-        ob->watermark = WM_MAX / 4;
-        if (ctx.stats) ctx.stats->num_kernels += 3;
-        record_ready(ob);
-        out.emit(ob);
+        // EOS only: one sync for the total, then page the fires (any window
+        // count fits the output batch capacity)
+        int64_t total = 0;
+        HIPCHK(hipStreamSynchronize(stream));
+        HIPCHK(hipMemcpy(&total, d_on, 8, hipMemcpyDeviceToHost));
+        for (int64_t base = 0; base < total || base == 0; ) {
+            Batch* ob = get_dev();
+            float* o[4];
+            for (int m = 0; m < 4; ++m)
+                o[m] = (float*)ob->cols[std::min<size_t>(m, ob->cols.size() - 1)];
+            int rl = ring_log2;
+            ArgPack a;
+            a.add(ks.d_nslots); a.add(P); a.add(S); a.add(rl);
+            a.add(fill); a.add(acc); a.add(ring); a.add(st_head); a.add(st_last);
+            a.add(ks.slot_to_key); a.add(nf); a.add(base);
+            a.add(ob->key); a.add(o[0]); a.add(o[1]); a.add(o[2]); a.add(o[3]);
+            a.add(ob->ts);
+            int64_t cap = ob->capacity;
+            a.add(cap);
+            launch(f_flush, stream, 256, a);
+            int64_t page = std::min(total - base, cap);
+            ob->count = page < 0 ? 0 : page;
+            ob->watermark = WM_MAX / 4;
+            if (ctx.stats) ctx.stats->num_kernels += 1;
+            record_ready(ob);
+            out.emit(ob);
+            base += cap;
+            if (total == 0) break;
+        }
+        if (ctx.stats) ctx.stats->num_kernels += 2;
     }
 
     void on_eos(EmitCtx& out, RuntimeCtx& ctx) override {

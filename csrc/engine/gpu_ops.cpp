@@ -927,19 +927,32 @@ struct GpuFfatLogic : GpuLogicBase {
 
     void flush_open(const uint32_t* fill, const float* acc, EmitCtx& out,
                     RuntimeCtx& ctx) {
-        Batch* ob = get_dev();
         int64_t R = 1ll << ring_log2;
-        wfa_ffat_cb_flush(stream, ks.d_nslots, P, S, comb, ring_log2, fill, acc,
-                          ring_or_tree, use_tree ? 2 * R : R, use_tree ? R : 0,
-                          st_head, st_last, ks.slot_to_key, cb_nf, ob->key,
-                          (float*)ob->cols[0], ob->ts, ob->capacity, d_on);
-        HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost,
-                              stream));
-        ob->count = -1;
-        ob->watermark = WM_MAX / 4;
-        if (ctx.stats) ctx.stats->num_kernels += 3;
-        record_ready(ob);
-        out.emit(ob);
+        wfa_cb_flush_count(stream, ks.d_nslots, P, S, fill, st_head, cb_nf);
+        wfa_slot_scan(stream, cb_nf, ks.d_nslots, d_on);
+        // EOS only: one sync to learn the total, then page the fires so any
+        // window count fits the output batch capacity
+        int64_t total = 0;
+        HIPCHK(hipStreamSynchronize(stream));
+        HIPCHK(hipMemcpy(&total, d_on, 8, hipMemcpyDeviceToHost));
+        for (int64_t base = 0; base < total || base == 0; ) {
+            Batch* ob = get_dev();
+            int64_t page = std::min(total - base, ob->capacity);
+            wfa_ffat_cb_flush_fire(stream, ks.d_nslots, P, S, comb, ring_log2,
+                                   fill, acc, ring_or_tree,
+                                   use_tree ? 2 * R : R, use_tree ? R : 0,
+                                   st_head, st_last, ks.slot_to_key, cb_nf,
+                                   base, ob->key, (float*)ob->cols[0], ob->ts,
+                                   ob->capacity);
+            ob->count = page < 0 ? 0 : page;
+            ob->watermark = WM_MAX / 4;
+            if (ctx.stats) ctx.stats->num_kernels += 1;
+            record_ready(ob);
+            out.emit(ob);
+            base += ob->capacity;
+            if (total == 0) break;
+        }
+        if (ctx.stats) ctx.stats->num_kernels += 2;
     }
 };
 

@@ -1580,8 +1580,8 @@ __global__ void k_cb_flush(const uint32_t* n_slots, int64_t P, int64_t S,
                            int64_t slot_stride, int64_t cell_off,
                            const uint32_t* st_head, const int64_t* st_last,
                            const uint64_t* slot_to_key, const uint32_t* nf,
-                           uint64_t* out_key, float* out_val, int64_t* out_ts,
-                           int64_t out_cap) {
+                           int64_t out_base, uint64_t* out_key, float* out_val,
+                           int64_t* out_ts, int64_t out_cap) {
     const int64_t ns = *n_slots;
     const uint32_t Rm = (1u << ring_log2) - 1;
     const float ident = (comb == 1) ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
@@ -1601,10 +1601,12 @@ __global__ void k_cb_flush(const uint32_t* n_slots, int64_t P, int64_t S,
             for (uint64_t p = q * (uint64_t)S; p < head; ++p)
                 res = XCOMB(res, rg[(uint32_t)p & Rm]);
             if (part) res = XCOMB(res, st_acc[s]);
-            if (w < out_cap) {
-                out_key[w] = slot_to_key[s];
-                out_val[w] = res;
-                out_ts[w] = st_last ? st_last[s] : 0;
+            // window slice [out_base, out_base + out_cap): totals beyond one
+            // output batch are flushed in host-driven pages
+            if (w >= out_base && w - out_base < out_cap) {
+                out_key[w - out_base] = slot_to_key[s];
+                out_val[w - out_base] = res;
+                out_ts[w - out_base] = st_last ? st_last[s] : 0;
             }
             ++w;
         }
@@ -1612,23 +1614,20 @@ __global__ void k_cb_flush(const uint32_t* n_slots, int64_t P, int64_t S,
 #undef XCOMB
 }
 
-extern "C" void wfa_ffat_cb_flush(
+// fire one page [out_base, out_base+out_cap) of the flush windows; run
+// wfa_cb_flush_count + wfa_slot_scan first (nf = per-slot offsets)
+extern "C" void wfa_ffat_cb_flush_fire(
     wfa_stream_t s, const uint32_t* n_slots, int64_t P, int64_t S, int comb,
     int ring_log2, const uint32_t* st_fill, const float* st_acc,
     const float* cells, int64_t slot_stride, int64_t cell_off,
     const uint32_t* st_head, const int64_t* st_last,
-    const uint64_t* slot_to_key, uint32_t* nf, uint64_t* out_key,
-    float* out_val, int64_t* out_ts, int64_t out_cap, int64_t* d_out_n) {
-    hipStream_t st = (hipStream_t)s;
-    hipLaunchKernelGGL(k_cb_flush_count, dim3(WFA_MAX_BLOCKS / 8),
-                       dim3(WFA_THREADS), 0, st, n_slots, P, S, st_fill, st_head,
-                       nf);
-    hipLaunchKernelGGL(k_tb_scan, dim3(1), dim3(1024), 0, st, nf, n_slots,
-                       d_out_n);
+    const uint64_t* slot_to_key, const uint32_t* nf, int64_t out_base,
+    uint64_t* out_key, float* out_val, int64_t* out_ts, int64_t out_cap) {
     hipLaunchKernelGGL(k_cb_flush, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
-                       st, n_slots, P, S, comb, ring_log2, st_fill, st_acc, cells,
-                       slot_stride, cell_off, st_head, st_last, slot_to_key, nf,
-                       out_key, out_val, out_ts, out_cap);
+                       (hipStream_t)s, n_slots, P, S, comb, ring_log2, st_fill,
+                       st_acc, cells, slot_stride, cell_off, st_head, st_last,
+                       slot_to_key, nf, out_base, out_key, out_val, out_ts,
+                       out_cap);
 }
 
 // single-block exclusive scan over per-slot counts (shared with the JIT

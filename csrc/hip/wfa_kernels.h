@@ -200,15 +200,17 @@ void wfa_seg_last_ts(wfa_stream_t s, const uint32_t* seg_start,
 // CB/TB EOS partial-window flush (mirrors CPU FfatCpu on_eos): fires every
 // remaining open window from the ring/tree-leaf cells + the open pane.
 // st_fill/st_acc null => TB (no partial pane).  cells stride/off: ring =
-// (R, 0), tree leaves = (2R, R).  nf: u32 scratch >= n_slots+1.
-void wfa_ffat_cb_flush(wfa_stream_t s, const uint32_t* n_slots, int64_t P,
-                       int64_t S, int comb, int ring_log2,
-                       const uint32_t* st_fill, const float* st_acc,
-                       const float* cells, int64_t slot_stride, int64_t cell_off,
-                       const uint32_t* st_head, const int64_t* st_last,
-                       const uint64_t* slot_to_key, uint32_t* nf,
-                       uint64_t* out_key, float* out_val, int64_t* out_ts,
-                       int64_t out_cap, int64_t* d_out_n);
+// (R, 0), tree leaves = (2R, R).  nf: u32 scratch >= n_slots+1.  Drive as
+// count -> scan (total) -> one fire call per out_cap-sized page.
+void wfa_ffat_cb_flush_fire(wfa_stream_t s, const uint32_t* n_slots, int64_t P,
+                            int64_t S, int comb, int ring_log2,
+                            const uint32_t* st_fill, const float* st_acc,
+                            const float* cells, int64_t slot_stride,
+                            int64_t cell_off, const uint32_t* st_head,
+                            const int64_t* st_last,
+                            const uint64_t* slot_to_key, const uint32_t* nf,
+                            int64_t out_base, uint64_t* out_key,
+                            float* out_val, int64_t* out_ts, int64_t out_cap);
 
 // single-block exclusive scan over per-slot counts + total
 void wfa_slot_scan(wfa_stream_t s, uint32_t* nf, const uint32_t* n_slots,

@@ -89,7 +89,8 @@ def gpu_jit_reduce(lift="v0", comb="a0+b0", finalize="f0", identity=(0.0,),
 
 def gpu_avg_reduce(col=0, max_keys=1 << 16):
     """Keyed AVG: fused (sum, count) fold (see FUTURE.md round 1)."""
-    return gpu_jit_reduce(lift="v0;1.0f", comb="a0+b0;a1+b1", finalize="f0/f1",
+    return gpu_jit_reduce(lift="v0;1.0f", comb="a0+b0;a1+b1",
+                          finalize="(f1 > 0.0f) ? (f0 / f1) : 0.0f",
                           identity=(0.0, 0.0), cols=(col,), max_keys=max_keys)
 
 
@@ -113,8 +114,11 @@ def gpu_jit_ffat_windows(win=1000, slide=100, lift="v0", comb="a0+b0",
 
 def gpu_avg_ffat_windows(win=1000, slide=100, col=0, **kw):
     """Keyed sliding-window AVG: fused (sum, count) pane pairs."""
+    # empty (gap) windows fire the reference's default result 0, not 0/0:
+    # the count field gates the division
     return gpu_jit_ffat_windows(win, slide, lift="v0;1.0f",
-                                comb="a0+b0;a1+b1", finalize="f0/f1",
+                                comb="a0+b0;a1+b1",
+                                finalize="(f1 > 0.0f) ? (f0 / f1) : 0.0f",
                                 identity=(0.0, 0.0), cols=(col,),
                                 invertible=True, **kw)
 
