@@ -186,8 +186,10 @@ def build_cpu_graph(n_tuples, batch):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
-    ap.add_argument("--warmup", type=int, default=5)
+    # defaults sized so the driver's own record carries a sustained claim
+    # (~1.7 B tuples, ~100 ms timed region) while finishing in seconds
+    ap.add_argument("--steps", type=int, default=200)
+    ap.add_argument("--warmup", type=int, default=20)
     # default micro-batch = measured throughput plateau (BASELINE.md sweep:
     # 16.2 B t/s @4M, 19.8 @8M, 19.8 @16M; p99 2.2 ms at 8M)
     ap.add_argument("--batch", type=int, default=8_388_608)
