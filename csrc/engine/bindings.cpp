@@ -364,7 +364,54 @@ static StageSpec make_stage(Engine& e, int id, const std::string& kind, const st
     if (kind.rfind("win_", 0) == 0 || kind == "interval_join") {
         Engine* ep = &e;
         WindowFn wf = nullptr;
-        if (!pyfn.is_none()) {
+        JoinFn jf = nullptr;
+        if (!pyfn.is_none() && kind == "interval_join") {
+            // vectorized user predicate/result (reference
+            // interval_join.hpp:279-307): fn(pairs_dict) -> None (drop all)
+            // | values array (keep all) | (mask, values) tuple
+            auto f = pyfn.cast<py::function>();
+            jf = [f](const JoinPairs& p, uint8_t* keep, double* out) {
+                py::gil_scoped_acquire gil;
+                py::dict d;
+                d["key"] = py::array_t<uint64_t>({p.n}, {(int64_t)8}, p.key,
+                                                 py::none());
+                d["ts_a"] = py::array_t<int64_t>({p.n}, {(int64_t)8}, p.ts_a,
+                                                 py::none());
+                d["ts_b"] = py::array_t<int64_t>({p.n}, {(int64_t)8}, p.ts_b,
+                                                 py::none());
+                d["a"] = py::array_t<double>({p.n}, {(int64_t)8}, p.va,
+                                             py::none());
+                d["b"] = py::array_t<double>({p.n}, {(int64_t)8}, p.vb,
+                                             py::none());
+                py::object r = f(d);
+                if (r.is_none()) return;  // keep[] stays all-zero
+                py::object mask = py::none(), vals;
+                if (py::isinstance<py::tuple>(r)) {
+                    auto t = r.cast<py::tuple>();
+                    mask = t[0];
+                    vals = t[1];
+                } else {
+                    vals = r;
+                }
+                auto va = py::array_t<double, py::array::c_style |
+                                                  py::array::forcecast>(vals);
+                if (va.shape(0) != p.n)
+                    throw std::runtime_error(
+                        "join function must return one value per pair");
+                memcpy(out, va.data(), 8 * p.n);
+                if (mask.is_none()) {
+                    memset(keep, 1, p.n);
+                } else {
+                    auto ma = py::array_t<bool, py::array::c_style |
+                                                    py::array::forcecast>(mask);
+                    if (ma.shape(0) != p.n)
+                        throw std::runtime_error(
+                            "join mask must have one entry per pair");
+                    for (int64_t i = 0; i < p.n; ++i)
+                        keep[i] = ma.data()[i] ? 1 : 0;
+                }
+            };
+        } else if (!pyfn.is_none()) {
             auto f = pyfn.cast<py::function>();
             wf = [f](const WinRows& wr) -> double {
                 py::gil_scoped_acquire gil;
@@ -378,8 +425,8 @@ static StageSpec make_stage(Engine& e, int id, const std::string& kind, const st
                 return f(d).cast<double>();
             };
         }
-        st.factory = [kind, fp, ip, ep, id, wf] {
-            return make_window_logic(kind, fp, ip, ep, id, wf);
+        st.factory = [kind, fp, ip, ep, id, wf, jf] {
+            return make_window_logic(kind, fp, ip, ep, id, wf, jf);
         };
         return st;
     }
@@ -538,6 +585,12 @@ PYBIND11_MODULE(_core, m) {
              })
         .def("abort_now", [](Engine& e) { e.abort.store(true); })
         .def("sink_sum", [](Engine& e, int op) { return e.sink_acc_i64.at(op).load(); })
+        .def("sink_sum_f",
+             [](Engine& e, int op) {
+                 std::lock_guard<std::mutex> g(e.sink_f64_mu);
+                 auto it = e.sink_acc_f64.find(op);
+                 return it == e.sink_acc_f64.end() ? 0.0 : it->second;
+             })
         .def("sink_count", [](Engine& e, int op) { return e.sink_tuples.at(op).load(); })
         .def("dropped", [](Engine& e) { return e.dropped_tuples.load(); })
         .def("sink_latencies",

@@ -442,13 +442,28 @@ struct WinRows {
 };
 using WindowFn = std::function<double(const WinRows&)>;
 
+// Vectorized user join function (reference interval_join.hpp:279-307
+// arbitrary predicate/result): called once per batch of candidate pairs
+// with SoA views; fills keep[] (0/1) and out[] (one result per kept pair).
+struct JoinPairs {
+    int64_t n = 0;
+    const uint64_t* key = nullptr;
+    const int64_t* ts_a = nullptr;
+    const int64_t* ts_b = nullptr;
+    const double* va = nullptr;
+    const double* vb = nullptr;
+};
+using JoinFn =
+    std::function<void(const JoinPairs&, uint8_t* keep, double* out)>;
+
 // kinds: win_keyed, win_parallel, win_plq, win_wlq, win_mr_map,
 // win_mr_reduce, win_ffat, interval_join  (iparam layouts in windows.cpp)
 std::shared_ptr<OpLogic> make_window_logic(const std::string& kind,
                                            const std::vector<double>& fp,
                                            const std::vector<int64_t>& ip,
                                            Engine* eng, int op_id,
-                                           WindowFn userfn = nullptr);
+                                           WindowFn userfn = nullptr,
+                                           JoinFn joinfn = nullptr);
 
 // GPU logic factory (gpu_ops.cpp) — kinds: gpu_source/gpu_map/gpu_filter/
 // gpu_reduce/gpu_ffat/gpu_to_host/gpu_count_sink

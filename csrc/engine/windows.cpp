@@ -1000,6 +1000,13 @@ struct IntervalJoinLogic : OpLogic {
     int64_t emit_wm = 0;
     int64_t store_ctr = 0;
     bool a_int = true, b_int = true;
+    // user predicate/result (reference interval_join.hpp:279-307): when
+    // set, matched pairs are STAGED and the vectorized callback decides
+    // keep + result value per pair (one call per processed batch)
+    JoinFn pairfn = nullptr;
+    std::vector<uint64_t> p_key;
+    std::vector<int64_t> p_tsa, p_tsb;
+    std::vector<double> p_va, p_vb;
 
     IntervalJoinLogic(int m, int64_t lo, int64_t up, int ca, int cb, Engine* e)
         : mode(m), lower(lo), upper(up), colA(ca), colB(cb), eng(e) {}
@@ -1016,12 +1023,49 @@ struct IntervalJoinLogic : OpLogic {
     }
 
     void emit_pair(uint64_t key, const Entry& ea, const Entry& eb, EmitCtx& out) {
+        if (pairfn) {
+            p_key.push_back(key);
+            p_tsa.push_back(ea.ts);
+            p_tsb.push_back(eb.ts);
+            p_va.push_back(a_int ? (double)ea.v.i : ea.v.f);
+            p_vb.push_back(b_int ? (double)eb.v.i : eb.v.f);
+            return;
+        }
         int64_t i = ob.slot(out, emit_wm);
         Batch* o = ob.b;
         o->ts[i] = std::max(ea.ts, eb.ts);
         o->key[i] = key;
         write_val(o, 0, i, ea.v, a_int);
         write_val(o, 1, i, eb.v, b_int);
+    }
+
+    void flush_pairs(EmitCtx& out) {
+        if (!pairfn || p_key.empty()) return;
+        const int64_t n = (int64_t)p_key.size();
+        std::vector<uint8_t> keep(n, 0);
+        std::vector<double> res(n, 0.0);
+        JoinPairs jp{n, p_key.data(), p_tsa.data(), p_tsb.data(), p_va.data(),
+                     p_vb.data()};
+        pairfn(jp, keep.data(), res.data());
+        for (int64_t i = 0; i < n; ++i) {
+            if (!keep[i]) continue;
+            int64_t w = ob.slot(out, emit_wm);
+            Batch* o = ob.b;
+            o->ts[w] = std::max(p_tsa[i], p_tsb[i]);
+            o->key[w] = p_key[i];
+            ValU v;
+            v.f = res[i];
+            v.i = (int64_t)res[i];
+            bool int_out = o->schema.payload.size() > 0 &&
+                           (o->schema.payload[0] == DType::I64 ||
+                            o->schema.payload[0] == DType::I32);
+            write_val(o, 0, w, v, int_out);
+        }
+        p_key.clear();
+        p_tsa.clear();
+        p_tsb.clear();
+        p_va.clear();
+        p_vb.clear();
     }
 
     void process(Batch* b, EmitCtx& out, RuntimeCtx& ctx) override {
@@ -1064,6 +1108,7 @@ struct IntervalJoinLogic : OpLogic {
         }
         cur_wm = std::max(cur_wm, ctx.current_wm);
         release(b);
+        flush_pairs(out);
         ob.flush(out, cur_wm);
     }
 
@@ -1101,12 +1146,16 @@ struct IntervalJoinLogic : OpLogic {
         emit_wm = cur_wm;
         cur_wm = std::max(cur_wm, wm);
         purge();
+        flush_pairs(out);
         ob.flush(out, cur_wm);
         return false;
     }
 
     void on_eos(EmitCtx& out, RuntimeCtx&) override {
-        emit_wm = cur_wm; ob.flush(out, cur_wm); }
+        emit_wm = cur_wm;
+        flush_pairs(out);
+        ob.flush(out, cur_wm);
+    }
 };
 
 // ============== non-incremental (user-function) keyed windows ==============
@@ -1280,11 +1329,14 @@ std::shared_ptr<OpLogic> make_window_logic(const std::string& kind,
                                            const std::vector<double>& fp,
                                            const std::vector<int64_t>& ip,
                                            Engine* eng, int op_id,
-                                           WindowFn userfn) {
+                                           WindowFn userfn, JoinFn joinfn) {
     auto geti = [&](size_t i, int64_t dflt = 0) { return i < ip.size() ? ip[i] : dflt; };
-    if (kind == "interval_join")
-        return std::make_shared<IntervalJoinLogic>((int)geti(0), geti(1), geti(2),
-                                                   (int)geti(3), (int)geti(4, geti(3)), eng);
+    if (kind == "interval_join") {
+        auto j = std::make_shared<IntervalJoinLogic>((int)geti(0), geti(1), geti(2),
+                                                     (int)geti(3), (int)geti(4, geti(3)), eng);
+        j->pairfn = joinfn;
+        return j;
+    }
     WinType wt = (WinType)geti(0);
     // defense in depth: win=0/slide=0 would loop forever opening windows
     if (kind != "win_mr_reduce" && (geti(1) < 1 || geti(2) < 1))

@@ -24,6 +24,35 @@ def affine_map(col=0, a=1, b=0):
     return NativeLogic("map", "affine_i64", [], [col, a, b])
 
 
+def rand_source_f(stream_len, n_keys=1, batch=1024, seed=42, dtype="f32"):
+    """Random keyed FLOAT stream: value ~ U[0, 1000) in an f32/f64 column
+    (round-2 float-path catalog — no GIL on float payloads)."""
+    return NativeLogic("source", f"rand_{dtype}", [], [stream_len, n_keys, batch, seed])
+
+
+def affine_map_f(col=0, a=1.0, b=0.0, dtype="f32"):
+    """x = a*x + b on an f32/f64 column (in place, native)."""
+    return NativeLogic("map", f"affine_{dtype}", [float(a), float(b)], [col])
+
+
+def gt_filter_f(col=0, thr=0.0, keep_gt=True, dtype="f32"):
+    """keep rows where x > thr (keep_gt=False keeps x <= thr), native."""
+    return NativeLogic("filter", f"gt_{dtype}", [float(thr)],
+                       [col, 1 if keep_gt else 0])
+
+
+def keyed_reduce_f(col=0, comb="sum", init=0.0, dtype="f32"):
+    """per-key running sum/min/max over a float column (f64 state); emits
+    the updated (key, acc) per input (reference reduce.hpp + withInitialState)."""
+    ci = {"sum": 0, "min": 1, "max": 2}[comb]
+    return NativeLogic("reduce", f"comb_by_key_{dtype}", [float(init)], [col, ci])
+
+
+def sum_sink_f(col=0, dtype="f32"):
+    """accumulate f64 sum of a float column (read back with g.sink_sum_f)."""
+    return NativeLogic("sink", f"sum_{dtype}", [], [col])
+
+
 def mod_filter(col=0, m=2, c=0, keep_eq=False):
     """keep rows where (x % m != c); keep_eq flips to ==."""
     return NativeLogic("filter", "mod_i64", [], [col, m, c, 1 if keep_eq else 0])

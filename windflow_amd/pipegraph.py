@@ -350,6 +350,10 @@ class PipeGraph:
     def sink_sum(self, sink_op):
         return self.engine.sink_sum(self._sink_map[id(sink_op)])
 
+    def sink_sum_f(self, sink_op):
+        """f64 accumulator of a float sum sink (native.sum_sink_f)."""
+        return self.engine.sink_sum_f(self._sink_map[id(sink_op)])
+
     def sink_count(self, sink_op):
         return self.engine.sink_count(self._sink_map[id(sink_op)])
 

@@ -68,8 +68,11 @@ def lower_window_node(graph, e, node):
         ca = int(j.get("colA", 0))
         cb = int(j.get("colB", ca))
         ip = [mode, int(j["lower"]), int(j["upper"]), ca, cb]
+        # user predicate/result callable (vectorized over matched pairs)
+        pyfn = op.logic if callable(op.logic) else None
         eid = e.add_op(op.name or kind, node.parallelism, "interval_join",
-                       iparams=ip, out_schema=out_schema, out_batch=op.out_batch)
+                       iparams=ip, out_schema=out_schema, out_batch=op.out_batch,
+                       pyfn=pyfn)
         if mode == JoinMode.DP:
             node.in_collector = CollectorKind.ORDERING
         in_id = out_id = eid
