@@ -210,7 +210,12 @@ def main():
         import torch  # noqa: F811
         if world > 1:
             import torch.distributed as dist  # noqa: F811
-            dist.init_process_group("nccl", rank=rank, world_size=world)
+            # gloo for the out-of-band bootstrap (rccl id broadcast +
+            # barriers): torch's own nccl group would refuse multiple ranks
+            # per device, but the ENGINE's RCCL communicator is the thing
+            # under test; device collectives all run through it.
+            backend = os.environ.get("WFA_DIST_BACKEND", "gloo")
+            dist.init_process_group(backend, rank=rank, world_size=world)
             torch.cuda.set_device(local_rank)
         if args.config in ("a2a", "ffat_x"):
             if world > 1:
@@ -276,9 +281,10 @@ def main():
         lat = sorted(g.engine.sink_latencies(g._sink_map[id(snk)]))
         p99_us = lat[min(len(lat) - 1, int(0.99 * len(lat)))] if lat else None
 
-    # max step time over ranks
+    # max step time over ranks (tensor device must match the backend)
     if dist is not None:
-        t = torch.tensor([dt], dtype=torch.float64, device="cuda")
+        dev = "cuda" if dist.get_backend() == "nccl" else "cpu"
+        t = torch.tensor([dt], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         dt = float(t.item())
 

@@ -37,7 +37,10 @@ from windflow_amd.synth import gen_batch       # noqa: E402
 
 
 def main():
-    td.init_process_group("nccl")
+    # gloo bootstrap: torch's own nccl group refuses several ranks on one
+    # device; the engine's RCCL communicator (the thing under test) is
+    # initialized separately from the broadcast id
+    td.init_process_group("gloo")
     rank, world, rid = init_from_torch()
     device = rank % max(1, torch.cuda.device_count())
     torch.cuda.set_device(device)
