@@ -140,4 +140,9 @@ void gpu_resolve_count(Batch* b);  // blocks on ready_event if count == -1
 // Deep-copy b (same pool class) — used for copy-on-write under broadcast.
 Batch* clone(Batch* b, Pool& pool);
 
+// Device-batch clone for broadcast fan-out (reference
+// splitting_emitter_gpu.hpp:186-200 D2D per-branch replication): D2D copy
+// on the producing stream into a fresh batch from b's own pool.
+Batch* gpu_clone_batch(Batch* b);
+
 }  // namespace wfa

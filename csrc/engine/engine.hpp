@@ -109,12 +109,16 @@ struct BroadcastEmitter : Emitter {
     Pool* punct_pool;
     explicit BroadcastEmitter(std::vector<SpscQueue*> d, Pool* pp) : dests(std::move(d)), punct_pool(pp) {}
     void emit(Batch* b) override {
-        if (b->loc == Loc::DEVICE && dests.size() > 1)
-            throw std::runtime_error(
-                "broadcast of device batches is not supported (per-batch "
-                "events are single-consumer): stage through gpu_to_host or "
-                "use split_gpu per branch");
         account(b);
+        if (b->loc == Loc::DEVICE && dests.size() > 1) {
+            // per-batch events are single-consumer: broadcast device
+            // batches as D2D clones (reference splitting_emitter_gpu
+            // replicates per branch; round 1 rejected this case)
+            for (size_t i = 1; i < dests.size(); ++i)
+                dests[i]->push(gpu_clone_batch(b), abort);
+            dests[0]->push(b, abort);
+            return;
+        }
         b->refcnt.fetch_add((int)dests.size() - 1, std::memory_order_relaxed);
         for (auto* q : dests) q->push(b, abort);
     }

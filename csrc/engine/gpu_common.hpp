@@ -10,6 +10,7 @@
 
 #include <map>
 #include <mutex>
+#include <set>
 #include <string>
 #include <vector>
 
@@ -187,8 +188,73 @@ struct GpuLogicBase : OpLogic {
     Batch* input_on_device(Batch* b, RuntimeCtx& ctx) {
         if (b->loc == Loc::HOST) return to_device(b, ctx);
         gpu_resolve_count(b);
+        if (b->device != device || force_peer_copy())
+            return peer_copy(b, ctx);
         wait_ready(b);
         return b;
+    }
+
+    // WFA_FORCE_PEER_COPY=1 exercises the cross-device path on a 1-GPU box
+    // (device 0 -> 0 through hipMemcpyPeerAsync)
+    static bool force_peer_copy() {
+        static int v = -1;
+        if (v < 0) {
+            const char* e = getenv("WFA_FORCE_PEER_COPY");
+            v = (e && e[0] == '1') ? 1 : 0;
+        }
+        return v;
+    }
+
+    // Cross-GPU forward: producer replica on another device handed us its
+    // batch pointer (withDevice(0) -> withDevice(1) chains move
+    // over xGMI).  hipMemcpyPeerAsync uses the direct link when peer access
+    // is up (enabled once per pair below) and stages through the host
+    // otherwise — the reference has no P2P path at all
+    // (wf/forward_emitter_gpu.hpp:296-328 always stages via pinned host).
+    Batch* peer_copy(Batch* b, RuntimeCtx& ctx) {
+        enable_peer(device, b->device);
+        // producer-side contents must be valid before the engine reads them
+        // from the consumer device: wait on the producing event (legal
+        // cross-device), then copy on OUR stream
+        wait_ready(b);
+        Batch* db = get_dev();
+        const int64_t n = b->count;
+        HIPCHK(hipMemcpyPeerAsync(db->ts, device, b->ts, b->device, 8 * n,
+                                  stream));
+        HIPCHK(hipMemcpyPeerAsync(db->key, device, b->key, b->device, 8 * n,
+                                  stream));
+        size_t bytes = 16 * n;
+        for (size_t c = 0; c < b->cols.size() && c < db->cols.size(); ++c) {
+            size_t es = dsize(b->schema.payload[c]);
+            HIPCHK(hipMemcpyPeerAsync(db->cols[c], device, b->cols[c],
+                                      b->device, es * n, stream));
+            bytes += es * n;
+        }
+        db->count = n;
+        db->watermark = b->watermark;
+        db->stream_tag = b->stream_tag;
+        db->born_us = b->born_us;
+        if (ctx.stats) ctx.stats->bytes_d2h += bytes;  // inter-device traffic
+        // the source batch may be recycled by its (other-device) producer
+        // the moment we release: the copies must have read it first
+        HIPCHK(hipStreamSynchronize(stream));
+        release(b);
+        return db;
+    }
+
+    static void enable_peer(int dst, int src) {
+        if (dst == src) return;
+        static std::mutex mu;
+        static std::set<std::pair<int, int>> done;
+        std::lock_guard<std::mutex> g(mu);
+        if (!done.insert({dst, src}).second) return;
+        int can = 0;
+        if (hipDeviceCanAccessPeer(&can, dst, src) == hipSuccess && can) {
+            HIPCHK(hipSetDevice(dst));
+            hipError_t e = hipDeviceEnablePeerAccess(src, 0);
+            if (e != hipSuccess && e != hipErrorPeerAccessAlreadyEnabled)
+                (void)hipGetLastError();  // fall back to staged copies
+        }
     }
 };
 

@@ -1073,6 +1073,134 @@ struct GpuJitStatefulLogic : GpuLogicBase {
     }
 };
 
+// ===== per-tuple device split =====
+// Reference Splitting_Emitter_GPU (splitting_emitter_gpu.hpp:186-200)
+// replicates the whole batch per branch; here the user BRANCH EXPRESSION
+// over (v0.., ts, key) is JIT-compiled, evaluated per row, and each branch
+// receives a compacted batch of exactly its rows (round 1 offered only
+// whole-batch round-robin).  Rows whose branch id falls outside [0, n)
+// are dropped.
+static std::string gen_split_source(const std::string& expr, int ncols) {
+    std::string s;
+    s += "typedef long long i64; typedef unsigned long long u64;\n"
+         "typedef unsigned int u32;\n";
+    s += "__device__ __forceinline__ float jld(const void* p, int dt, i64 i) {\n"
+         "    switch (dt) {\n"
+         "        case 0: return (float)((const i64*)p)[i];\n"
+         "        case 1: return (float)((const double*)p)[i];\n"
+         "        case 2: return ((const float*)p)[i];\n"
+         "        case 3: return (float)((const u64*)p)[i];\n"
+         "        case 4: return (float)((const int*)p)[i];\n"
+         "        default: {\n"
+         "            union { u32 u; float f; } c;\n"
+         "            c.u = (u32)((const unsigned short*)p)[i] << 16;\n"
+         "            return c.f;\n"
+         "        }\n"
+         "    }\n"
+         "}\n";
+    s += "extern \"C\" __global__ void jit_split(i64 n,\n"
+         "    const void* c0, const void* c1, const void* c2, const void* c3,\n"
+         "    int dt0, int dt1, int dt2, int dt3,\n"
+         "    const i64* ts_arr, const u64* key_arr, u32* branch) {\n"
+         "    for (i64 i = blockIdx.x * (i64)blockDim.x + threadIdx.x; i < n;\n"
+         "         i += gridDim.x * (i64)blockDim.x) {\n"
+         "        const i64 ts = ts_arr[i]; (void)ts;\n"
+         "        const u64 key = key_arr[i]; (void)key;\n";
+    for (int c = 0; c < ncols; ++c) {
+        auto ci = std::to_string(c);
+        s += "        const float v" + ci + " = jld(c" + ci + ", dt" + ci +
+             ", i); (void)v" + ci + ";\n";
+    }
+    s += "        branch[i] = (u32)(" + expr + ");\n"
+         "    }\n}\n";
+    return s;
+}
+
+struct GpuJitSplitLogic : GpuLogicBase {
+    std::string expr;
+    int ncols;
+    int cols[4] = {0, 1, 2, 3};
+    hipFunction_t f_split = nullptr;
+    uint32_t* d_branch = nullptr;
+    uint32_t* d_flags = nullptr;
+    uint32_t* d_scan = nullptr;
+    int64_t* d_cnt = nullptr;
+    void** d_colptrs = nullptr;
+    int* d_esize = nullptr;
+    std::string src;
+
+    GpuJitSplitLogic(std::string e, int nc, int dev, Schema os, int64_t cap)
+        : expr(std::move(e)), ncols(nc) {
+        device = dev;
+        out_schema = os;
+        out_cap = cap;
+        src = gen_split_source(expr, ncols);
+    }
+    void init_device() override {
+        f_split = jit_fn(jit_module(src, device), "jit_split");
+        auto& A = arena(device);
+        d_branch = (uint32_t*)A.get(4 * out_cap);
+        d_flags = (uint32_t*)A.get(4 * out_cap);
+        d_scan = (uint32_t*)A.get(4 * (out_cap / 2048 + 2));
+        d_cnt = (int64_t*)A.get(64);
+        size_t nc = out_schema.payload.size();
+        d_colptrs = (void**)A.get(16 * (nc + 1));
+        d_esize = (int*)A.get(4 * (nc + 1));
+        std::vector<int> es;
+        for (auto d : out_schema.payload) es.push_back((int)dsize(d));
+        HIPCHK(hipMemcpy(d_esize, es.data(), 4 * nc, hipMemcpyHostToDevice));
+    }
+    void process(Batch* in, EmitCtx& out, RuntimeCtx& ctx) override {
+        ensure_init();
+        Batch* db = input_on_device(in, ctx);
+        int64_t n = db->count;
+        if (n > out_cap)
+            throw std::runtime_error("gpu split input > out_batch capacity");
+        JitFoldSpec tmp;
+        tmp.ncols = ncols;
+        for (int c = 0; c < 4; ++c) tmp.cols[c] = cols[c];
+        ColArgs ca;
+        ca.fill(db, tmp);
+        ArgPack a;
+        a.add(n);
+        a.add(ca.c[0]); a.add(ca.c[1]); a.add(ca.c[2]); a.add(ca.c[3]);
+        a.add(ca.dt[0]); a.add(ca.dt[1]); a.add(ca.dt[2]); a.add(ca.dt[3]);
+        a.add(db->ts); a.add(db->key); a.add(d_branch);
+        launch(f_split, stream, 2048, a);
+        const size_t np = db->cols.size();
+        const size_t nb = out.n_branches();
+        for (size_t br = 0; br < nb; ++br) {
+            wfa_flags_eq_u32(stream, d_branch, n, (uint32_t)br, d_flags);
+            Batch* ob = get_dev();
+            std::vector<void*> ptrs(2 * np);
+            for (size_t c = 0; c < np; ++c) {
+                ptrs[c] = db->cols[c];
+                ptrs[np + c] = ob->cols[c];
+            }
+            HIPCHK(hipMemcpyAsync(d_colptrs, ptrs.data(), 8 * 2 * np,
+                                  hipMemcpyHostToDevice, stream));
+            wfa_compact(stream, n, d_flags, d_scan, db->ts, ob->ts, db->key,
+                        ob->key, (const void* const*)d_colptrs,
+                        (void* const*)(d_colptrs + np), d_esize, (int)np,
+                        d_cnt);
+            HIPCHK(hipMemcpyAsync(ob->lazy_count, d_cnt, 8,
+                                  hipMemcpyDeviceToHost, stream));
+            // branch batches overlap in flight: each needs the count landed
+            // before its event fires; one event per batch handles it, but
+            // d_cnt is shared — synchronize the copy before the next branch
+            HIPCHK(hipStreamSynchronize(stream));
+            ob->count = *ob->lazy_count;
+            ob->watermark = db->watermark;
+            ob->stream_tag = db->stream_tag;
+            ob->born_us = db->born_us;
+            record_ready(ob);
+            out.emit_to(br, ob);
+        }
+        if (ctx.stats) ctx.stats->num_kernels += 1 + 3 * (int)nb;
+        release_after_use(db);
+    }
+};
+
 std::shared_ptr<OpLogic> make_gpu_jit_logic(const std::string& kind,
                                             const std::string& spec,
                                             const std::vector<double>& fp,
@@ -1085,6 +1213,10 @@ std::shared_ptr<OpLogic> make_gpu_jit_logic(const std::string& kind,
             spec, (int)ip.at(1), (int)ip.at(2), ip.at(0) != 0, fp, ip.at(3),
             device, os, out_batch);
     }
+    if (kind == "gpu_split")
+        // spec = branch expression; ip: [ncols]
+        return std::make_shared<GpuJitSplitLogic>(
+            spec, ip.empty() ? 1 : (int)ip.at(0), device, os, out_batch);
     JitFoldSpec fs = parse_fold_spec(spec, fp, ip);
     if (kind == "gpu_jit_reduce") {
         // ip: [nf, nout, ncols, col0..3, max_keys]
@@ -1114,6 +1246,8 @@ std::string debug_jit_fold_source(const std::string& kind,
     if (kind == "gpu_jit_stateful")
         return gen_stateful_source(spec, (int)ip.at(1), (int)ip.at(2),
                                    ip.at(0) != 0);
+    if (kind == "gpu_split")
+        return gen_split_source(spec, ip.empty() ? 1 : (int)ip.at(0));
     JitFoldSpec fs = parse_fold_spec(spec, fp, ip);
     if (kind == "gpu_jit_ffat") fs.invertible = ip.size() > 13 && ip[13] != 0;
     return gen_fold_source(fs, kind == "gpu_jit_ffat");

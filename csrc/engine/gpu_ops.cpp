@@ -100,6 +100,35 @@ void gpu_free_batch(Batch* b) {
     delete b;
 }
 
+// D2D clone on the producing stream (broadcast fan-out of device batches —
+// reference broadcast_emitter_gpu.hpp / splitting_emitter_gpu.hpp:186-200
+// replicate per consumer; per-batch events are single-consumer here, so a
+// clone per extra destination keeps the recycling protocol intact).
+Batch* gpu_clone_batch(Batch* b) {
+    gpu_resolve_count(b);  // count on host; contents valid once event fired
+    Pool* p = b->pool;
+    if (!p) throw std::runtime_error("gpu_clone_batch: batch has no pool");
+    Batch* c = p->get();
+    hipStream_t s = (hipStream_t)b->stream;
+    if (c->ready_event)
+        HIPCHK(hipStreamWaitEvent(s, (hipEvent_t)c->ready_event, 0));
+    const int64_t n = b->count;
+    HIPCHK(hipMemcpyAsync(c->ts, b->ts, 8 * n, hipMemcpyDeviceToDevice, s));
+    HIPCHK(hipMemcpyAsync(c->key, b->key, 8 * n, hipMemcpyDeviceToDevice, s));
+    for (size_t i = 0; i < b->cols.size(); ++i) {
+        size_t es = dsize(b->schema.payload[i]);
+        HIPCHK(hipMemcpyAsync(c->cols[i], b->cols[i], es * n,
+                              hipMemcpyDeviceToDevice, s));
+    }
+    c->count = n;
+    c->watermark = b->watermark;
+    c->stream_tag = b->stream_tag;
+    c->born_us = b->born_us;
+    c->stream = s;
+    HIPCHK(hipEventRecord((hipEvent_t)c->ready_event, s));
+    return c;
+}
+
 // ===== device source (GPU-resident synthetic generator) =====
 struct GpuSourceLogic : GpuLogicBase {
     int64_t len, n_keys, bsz;
@@ -1430,7 +1459,7 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
         return std::make_shared<GpuJitFilterLogic>(spec, (int)(ip.empty() ? 0 : ip[0]),
                                                    device, os, out_batch);
     if (kind == "gpu_jit_reduce" || kind == "gpu_jit_ffat" ||
-        kind == "gpu_jit_stateful")
+        kind == "gpu_jit_stateful" || kind == "gpu_split")
         // generalized user folds (multi-field accumulators) — gpu_jit.cpp
         return make_gpu_jit_logic(kind, spec, fp, ip, eng, op_id, device, os,
                                   out_batch);
@@ -1579,6 +1608,7 @@ namespace wfa {
 Batch* gpu_alloc_batch(Pool&) { throw std::runtime_error("built without HIP"); }
 void gpu_free_batch(Batch*) {}
 void gpu_resolve_count(Batch*) {}
+Batch* gpu_clone_batch(Batch*) { throw std::runtime_error("built without HIP"); }
 std::string wfa_rccl_unique_id() { throw std::runtime_error("built without HIP"); }
 std::pair<std::vector<uint32_t>, std::vector<uint32_t>> debug_sort_pairs_host(
     const uint32_t*, int64_t, int) {

@@ -316,6 +316,12 @@ __global__ void k_cast(const void* in, int di, void* out, int d_o, int64_t n) {
         }
     }
 }
+__global__ void k_flags_eq_u32(const uint32_t* v, int64_t n, uint32_t b,
+                               uint32_t* flags) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * (int64_t)blockDim.x)
+        flags[i] = v[i] == b ? 1u : 0u;
+}
 __global__ void k_fill_f32_strided(float* p, float v, int64_t n, int64_t stride) {
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += gridDim.x * (int64_t)blockDim.x)
@@ -336,6 +342,9 @@ void wfa_fill_u32(wfa_stream_t s, uint32_t* p, uint32_t v, int64_t n) {
 }
 void wfa_fill_f32(wfa_stream_t s, float* p, float v, int64_t n) {
     hipLaunchKernelGGL(k_fill_f32, dim3(nblk(n)), dim3(WFA_THREADS), 0, (hipStream_t)s, p, v, n);
+}
+void wfa_flags_eq_u32(wfa_stream_t s, const uint32_t* v, int64_t n, uint32_t b, uint32_t* flags) {
+    hipLaunchKernelGGL(k_flags_eq_u32, dim3(nblk(n)), dim3(WFA_THREADS), 0, (hipStream_t)s, v, n, b, flags);
 }
 void wfa_fill_f32_strided(wfa_stream_t s, float* p, float v, int64_t n, int64_t stride) {
     hipLaunchKernelGGL(k_fill_f32_strided, dim3(nblk(n)), dim3(WFA_THREADS), 0, (hipStream_t)s, p, v, n, stride);

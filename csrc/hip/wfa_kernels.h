@@ -230,6 +230,9 @@ void wfa_fill_u32(wfa_stream_t s, uint32_t* p, uint32_t v, int64_t n);
 void wfa_fill_f32(wfa_stream_t s, float* p, float v, int64_t n);
 void wfa_fill_f32_strided(wfa_stream_t s, float* p, float v, int64_t n,
                           int64_t stride);
+// per-branch split flags: flags[i] = (v[i] == b)
+void wfa_flags_eq_u32(wfa_stream_t s, const uint32_t* v, int64_t n, uint32_t b,
+                      uint32_t* flags);
 void wfa_iota_u32(wfa_stream_t s, uint32_t* p, int64_t n);
 void wfa_cast(wfa_stream_t s, const void* in, int dt_in, void* out, int dt_out,
               int64_t n);

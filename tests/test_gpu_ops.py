@@ -937,3 +937,113 @@ def test_gpu_ffat_vik_fallback_key_overflow():
         sums.append((acc['n'], acc['s']))
     assert sums[0][0] == sums[1][0]
     assert abs(sums[0][1] - sums[1][1]) <= 1e-6 * max(1.0, abs(sums[0][1]))
+
+
+def test_gpu_split_per_tuple():
+    """Per-tuple device split (reference splitting_emitter_gpu per-branch
+    batches, upgraded from replication to true routing): a JIT branch
+    expression routes each row; per-branch compaction on device."""
+    n, n_keys, b = 120_000, 97, 20_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=0))
+           .withOutputSchema([0]).withOutputBatchSize(b).build())
+    g = wf.PipeGraph("gsplit2")
+    mp = g.add_source(src)
+    mp.split_gpu(3, expr="key % 3")
+    snks = []
+    for br in range(3):
+        bmp = mp.select(br)
+        bmp.add(Map_GPU_Builder(native_gpu.gpu_affine_map(0, 1, 0, dtype=0))
+                .withOutputSchema([0]).withOutputBatchSize(b).build())
+        snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).build()
+        bmp.chain_sink(snk)
+        snks.append(snk)
+    g.run()
+    ts, key, val = gen_batch(n, 0, 42, n_keys, 0)
+    for br in range(3):
+        assert g.sink_count(snks[br]) == int((key % 3 == br).sum()), br
+
+
+def test_gpu_split_per_tuple_value_expr():
+    """Branch expression over the VALUE column with drop semantics: rows
+    whose expression falls outside [0, n_branches) disappear."""
+    n, b = 80_000, 20_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, 31, b, vdt=2))
+           .withOutputSchema([2]).withOutputBatchSize(b).build())
+    g = wf.PipeGraph("gsplitv")
+    mp = g.add_source(src)
+    # v0 in [0,1): branch 0 for v<0.25, 1 for v<0.5, else 7 (dropped)
+    mp.split_gpu(2, expr="v0 < 0.25f ? 0 : (v0 < 0.5f ? 1 : 7)")
+    snks = []
+    for br in range(2):
+        bmp = mp.select(br)
+        snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).build()
+        bmp.chain_sink(snk)
+        snks.append(snk)
+    g.run()
+    ts, key, val = gen_batch(n, 0, 42, 31, 2)
+    assert g.sink_count(snks[0]) == int((val < 0.25).sum())
+    assert g.sink_count(snks[1]) == int(((val >= 0.25) & (val < 0.5)).sum())
+
+
+def test_gpu_device_broadcast():
+    """Device-batch broadcast (round 1 raised an error): every replica of a
+    broadcast-input GPU sink sees ALL batches via D2D clones."""
+    n, b = 60_000, 15_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, 31, b, vdt=0))
+           .withOutputSchema([0]).withOutputBatchSize(b).build())
+    g = wf.PipeGraph("gbcast")
+    mp = g.add_source(src)
+    snk = (Sink_GPU_Builder(native_gpu.gpu_count_sink())
+           .withParallelism(3).withBroadcast().build())
+    mp.add_sink(snk)
+    g.run()
+    assert g.sink_count(snk) == 3 * n  # each of the 3 replicas saw all rows
+
+
+def test_gpu_peer_copy_path_forced():
+    """Cross-device forward path (hipMemcpyPeerAsync): a 1-GPU box can't
+    place ops on two devices, so WFA_FORCE_PEER_COPY=1 routes every
+    device->device hand-off through the peer-copy machinery (device 0->0)
+    — same code path, same events/sync protocol.  Results must be
+    identical to the pointer-passing run."""
+    import os
+    import subprocess
+    import sys
+    script = r"""
+import sys
+sys.path.insert(0, ".")
+import windflow_amd as wf
+from windflow_amd import native, native_gpu
+from windflow_amd.builders_gpu import (Source_GPU_Builder, Map_GPU_Builder,
+                                       Filter_GPU_Builder)
+from windflow_amd.synth import gen_batch
+n, n_keys, b = 100_000, 97, 25_000
+src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=0))
+       .withOutputSchema([0]).withOutputBatchSize(b).build())
+mp_ = (Map_GPU_Builder(native_gpu.gpu_affine_map(0, 3, 1, dtype=0))
+       .withOutputSchema([0]).withOutputBatchSize(b)
+       .withParallelism(1).build())
+fl = (Filter_GPU_Builder(native_gpu.gpu_mod_filter(0, 5, 0))
+      .withOutputSchema([0]).withOutputBatchSize(b)
+      .withParallelism(1).build())
+g = wf.PipeGraph("peer")
+p = g.add_source(src)
+p.add(mp_)   # add (not chain): a real queue hop -> peer copy on receive
+p.add(fl)
+snk = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+snk.out_schema = [0]
+p.add_sink(snk)
+g.run()
+_, _, val = gen_batch(n, 0, 42, n_keys, 0)
+v = val * 3 + 1
+keep = v[v % 5 != 0]
+assert g.sink_sum(snk) == int(keep.sum()), (g.sink_sum(snk), int(keep.sum()))
+assert g.sink_count(snk) == len(keep)
+print("PEER_OK")
+"""
+    env = dict(os.environ, WFA_FORCE_PEER_COPY="1")
+    r = subprocess.run([sys.executable, "-c", script], env=env,
+                       capture_output=True, text=True, timeout=300,
+                       cwd=os.path.dirname(os.path.dirname(
+                           os.path.abspath(__file__))))
+    assert r.returncode == 0 and "PEER_OK" in r.stdout, r.stderr[-2000:]

@@ -533,3 +533,31 @@ def test_composite_split_windows_merge_fuzz():
             for (k, s), c in oracle_cb(per, win, slide, agg).items():
                 exp[(k, s)] += c
         assert Counter(rows) == exp, case
+
+
+def test_cross_device_chain_lowering():
+    """withDevice(0) -> withDevice(1) chains lower to SEPARATE nodes with
+    the right device per op (cross-GPU forward path): chain()
+    must not fuse across devices, and the lowered graph keeps the edge."""
+    from windflow_amd import native_gpu
+    from windflow_amd.builders_gpu import (Source_GPU_Builder, Map_GPU_Builder,
+                                           Sink_GPU_Builder)
+    src = (Source_GPU_Builder(native_gpu.gpu_source(1000, 7, 100, vdt=0))
+           .withOutputSchema([0]).withOutputBatchSize(100)
+           .withDevice(0).build())
+    m0 = (Map_GPU_Builder(native_gpu.gpu_affine_map(0, 2, 0, dtype=0))
+          .withOutputSchema([0]).withOutputBatchSize(100)
+          .withDevice(0).build())
+    m1 = (Map_GPU_Builder(native_gpu.gpu_affine_map(0, 3, 0, dtype=0))
+          .withOutputSchema([0]).withOutputBatchSize(100)
+          .withDevice(1).build())
+    snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).withDevice(1).build()
+    g = wf.PipeGraph("xdev")
+    mp = g.add_source(src)
+    mp.chain(m0)     # same device: fuses
+    mp.chain(m1)     # device 0 -> 1: must NOT fuse
+    mp.chain_sink(snk)
+    assert len(g.nodes) == 2, "cross-device chain must break the fusion"
+    assert g.nodes[0].device == 0 and g.nodes[1].device == 1
+    assert len(g.nodes[0].ops) == 2 and len(g.nodes[1].ops) == 2
+    assert g.edges[0]['src'] == 0 and g.edges[0]['dst'] == 1

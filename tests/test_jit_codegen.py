@@ -88,3 +88,10 @@ def test_builder_comb_expr_sets_extent():
     op2 = b2.build()
     ip2 = op2.logic.iparams
     assert ip2[8] == 1000 and ip2[9] == 100 and ip2[10] == 1 and ip2[11] == 30
+
+
+def test_split_source_compiles():
+    logic = native_gpu.gpu_jit_split("key % 3", ncols=1)
+    _hipcc_compiles(_src_for(logic))
+    logic2 = native_gpu.gpu_jit_split("v0 > 0.5f ? 1 : 0", ncols=2)
+    _hipcc_compiles(_src_for(logic2))

@@ -36,6 +36,13 @@ def lower_gpu_node(graph, e, node):
     consumers = [graph.nodes[ed['dst']] for ed in graph.edges
                  if ed['src'] == node_idx]
     if any(c.device < 0 and not c.ops[0].gpu for c in consumers):
+        if node.n_branches:
+            # a chained D2H stage would lose per-branch routing: branches
+            # must stay on the GPU (or add an explicit gpu_to_host per
+            # branch after select())
+            raise NotImplementedError(
+                "per-tuple gpu split branches must feed GPU operators; "
+                "add a gpu_to_host() stage on each branch for CPU consumers")
         last = node.ops[-1]
         e.chain_stage(eid, "gpu_to_host", "", out_schema=last.out_schema,
                       out_batch=last.out_batch)

@@ -142,6 +142,15 @@ def gpu_jit_stateful_filter(body, ncols=1, nstate=1, init=(0.0,),
                        [1, int(ncols), int(nstate), int(max_keys)])
 
 
+def gpu_jit_split(expr, ncols=1):
+    """Per-tuple device split: a JIT-compiled branch expression over
+    (v0..v{ncols-1}, ts, key) routes each ROW to one branch, compacted on
+    device per branch (reference splitting_emitter_gpu replicates whole
+    batches; per-tuple routing is the CPU Splitting_Emitter semantics).
+    Rows whose branch id is outside [0, n_branches) are dropped."""
+    return NativeLogic("gpu_split", expr, [], [int(ncols)])
+
+
 def gpu_keyed_running_sum(col=0, max_keys=1 << 16):
     """stateful map: per-key running sum written in place (key-order walk)."""
     return NativeLogic("gpu_map_keyed", "", [0.0, 0.0], [1, col, max_keys])

@@ -148,8 +148,21 @@ class MultiPipe:
             raise IndexError(f"select({branch}): split has {nb} branches")
         return MultiPipe(self.graph, [(n, branch)])
 
-    def split_gpu(self, n_branches):
-        return self.split(NativeLogic("split", "rr", [], []), n_branches)
+    def split_gpu(self, n_branches, expr=None, ncols=1):
+        """Device split.  Without `expr`: whole-batch round-robin (round-1
+        behavior).  With `expr`: PER-TUPLE routing by a JIT branch
+        expression over (v0.., ts, key) with on-device compaction per
+        branch (reference splitting_emitter_gpu.hpp per-branch batches)."""
+        if expr is None:
+            return self.split(NativeLogic("split", "rr", [], []), n_branches)
+        from .native_gpu import gpu_jit_split
+        updev = self.graph.nodes[self.tails[0][0]].device  # upstream device
+        mp = self.split(gpu_jit_split(expr, ncols), n_branches)
+        node = self.graph.nodes[self.tails[0][0]]          # the split node
+        node.ops[0].gpu = True
+        node.device = updev if updev >= 0 else 0
+        node.ops[0].device = node.device
+        return mp
 
 
 class PipeGraph:
