@@ -1186,6 +1186,13 @@ struct GpuExchangeLogic : GpuLogicBase {
     int* d_esize = nullptr;
     std::unique_ptr<Pool> send_pool;
     int64_t cur_wm = 0;
+    // metadata rendezvous runs on its own stream, OVERLAPPED with the
+    // sort/gather of the same batch (round 1 synced the whole chain per
+    // batch: the ~800 us "round" was really the pipeline drain — the
+    // k_count_u32 anomaly in profiles/a2a_kernel_stats_r01_rs8.csv was this
+    // serialization attributed to the first kernel after the sync)
+    hipStream_t meta_stream = nullptr;
+    hipEvent_t ev_counts = nullptr;
 
     GpuExchangeLogic(Engine* e, int dev, Schema os, int64_t cap) : eng(e) {
         device = dev;
@@ -1222,25 +1229,41 @@ struct GpuExchangeLogic : GpuLogicBase {
         send_pool = std::make_unique<Pool>(out_schema, out_cap, false);
         send_pool->loc = Loc::DEVICE;
         send_pool->device = device;
+        HIPCHK(hipStreamCreateWithFlags(&meta_stream, hipStreamNonBlocking));
+        HIPCHK(hipEventCreateWithFlags(&ev_counts, hipEventDisableTiming));
     }
     ~GpuExchangeLogic() override {
         if (h_meta) (void)hipHostFree(h_meta);
         if (h_counts) (void)hipHostFree(h_counts);
+        if (meta_stream) (void)hipStreamDestroy(meta_stream);
+        if (ev_counts) (void)hipEventDestroy(ev_counts);
     }
 
-    // one collective round: exchange (counts, wm, done) then rows.
-    // Returns true when every rank reported done.  The metadata is packed
-    // on-device (no pre-collective host sync); the single rendezvous sync
-    // is the h_meta readback every rank needs to lay out its recvs.
-    bool round(Batch* send, int64_t wm, bool done, EmitCtx& out, RuntimeCtx& ctx) {
-        const size_t nc = out_schema.payload.size();
-        if (!send) wfa_fill_u32(stream, d_counts, 0, world);
-        wfa_pack_meta(stream, d_counts, world, done ? WM_MAX : wm, done ? 1 : 0,
-                      d_meta);
-        NCCLCHK(ncclAllGather(d_meta, d_meta_all, world + 2, ncclInt64, comm, stream));
+    // kick the metadata rendezvous on meta_stream: pack counts -> allgather
+    // -> D2H.  d_counts must be ready on `stream` (ev_counts recorded);
+    // meta_stream waits on the event, so sort/gather on `stream` overlap
+    // the collective instead of serializing behind it.
+    void start_meta(bool have_counts, int64_t wm, bool done) {
+        if (!have_counts) wfa_fill_u32(meta_stream, d_counts, 0, world);
+        else HIPCHK(hipStreamWaitEvent(meta_stream, ev_counts, 0));
+        wfa_pack_meta(meta_stream, d_counts, world, done ? WM_MAX : wm,
+                      done ? 1 : 0, d_meta);
+        if (world > 1)
+            NCCLCHK(ncclAllGather(d_meta, d_meta_all, world + 2, ncclInt64,
+                                  comm, meta_stream));
+        else
+            HIPCHK(hipMemcpyAsync(d_meta_all, d_meta, 8 * (world + 2),
+                                  hipMemcpyDeviceToDevice, meta_stream));
         HIPCHK(hipMemcpyAsync(h_meta, d_meta_all, 8 * world * (world + 2),
-                              hipMemcpyDeviceToHost, stream));
-        HIPCHK(hipStreamSynchronize(stream));
+                              hipMemcpyDeviceToHost, meta_stream));
+    }
+
+    // one collective round: metadata already in flight on meta_stream
+    // (start_meta); sync META ONLY, lay out recvs, then move rows.
+    // Returns true when every rank reported done.
+    bool round(Batch* send, EmitCtx& out, RuntimeCtx& ctx) {
+        const size_t nc = out_schema.payload.size();
+        HIPCHK(hipStreamSynchronize(meta_stream));
         const int64_t* scnt = send ? h_meta + (int64_t)rank * (world + 2) : nullptr;
         // recv layout: rows from rank p land at roff[p]
         int64_t roff[9] = {0};
@@ -1348,9 +1371,13 @@ struct GpuExchangeLogic : GpuLogicBase {
         int64_t t1 = now_us();
         if (sb->ready_event)
             HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)sb->ready_event, 0));
-        // bucket -> stable sort by dest -> contiguous per-dest send layout
+        // bucket -> counts first: the metadata rendezvous departs on its own
+        // stream while the sort/gather still run on `stream`
         wfa_bucket_by_key(stream, db->key, n, world, dest);
         wfa_count_u32(stream, dest, n, d_counts, world);
+        HIPCHK(hipEventRecord(ev_counts, stream));
+        start_meta(true, wm, false);
+        // stable sort by dest -> contiguous per-dest send layout
         uint32_t *od, *oi;
         wfa_sort_pairs2(stream, dest, idx, dest_t, idx_t, nullptr, nullptr, hist,
                         n, bits, &od, &oi, nullptr, /*implicit_iota=*/1,
@@ -1369,7 +1396,7 @@ struct GpuExchangeLogic : GpuLogicBase {
         release_after_use(db);
         if (ctx.stats) ctx.stats->num_kernels += 4;
         int64_t t2 = now_us();
-        round(sb, wm, false, out, ctx);
+        round(sb, out, ctx);
         // round() recorded sb's event after the collective consumed it; the
         // pool's next get() waits on that event before reuse
         release(sb);
@@ -1386,7 +1413,9 @@ struct GpuExchangeLogic : GpuLogicBase {
     void on_eos(EmitCtx& out, RuntimeCtx& ctx) override {
         ensure_init();
         // EOS rounds: keep matching other ranks' collectives until all done
-        while (!round(nullptr, cur_wm, true, out, ctx)) {
+        for (;;) {
+            start_meta(false, cur_wm, true);
+            if (round(nullptr, out, ctx)) break;
         }
     }
 };

@@ -74,7 +74,17 @@ class MonitoringThread:
         try:
             self.sock = socket.create_connection((self.host, self.port), timeout=2)
             self._send(dict(type="diagram", graph=self.pg.name,
-                            dot=graph_dot(self.pg)))
+                            dot=graph_dot(self.pg),
+                            nodes=[dict(id=i,
+                                        label="|".join(op.name or op.kind
+                                                       for op in nd.ops),
+                                        parallelism=nd.parallelism,
+                                        gpu=bool(nd.ops[0].gpu),
+                                        device=nd.ops[0].device)
+                                   for i, nd in enumerate(self.pg.nodes)],
+                            edges=[dict(src=e["src"], dst=e["dst"],
+                                        routing=int(e["routing"]))
+                                   for e in self.pg.edges]))
             while not self._stop.wait(self.interval):
                 self._send(self._report())
             self._send(self._report())
