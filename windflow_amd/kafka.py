@@ -28,6 +28,18 @@ def _have_kafka():
         return False
 
 
+def _client():
+    """confluent_kafka when installed (production), else the in-repo
+    socket client (windflow_amd.kafka_client) speaking the same API —
+    the connector path runs end-to-end either way."""
+    try:
+        import confluent_kafka
+        return confluent_kafka
+    except ImportError:
+        from . import kafka_client
+        return kafka_client
+
+
 class Connector_Source_Builder(_BasicBuilder):
     """Source from an external iterator: fn(replica, parallelism) -> column
     dict or None (end of stream) — the engine's PySourceLogic contract."""
@@ -70,11 +82,8 @@ class Kafka_Source_Builder(_BasicBuilder):
         return self
 
     def build(self):
-        if not _have_kafka():
-            raise RuntimeError(
-                "Kafka_Source requires confluent_kafka (librdkafka); not "
-                "installed in this environment")
-        from confluent_kafka import Consumer, TopicPartition
+        mod = _client()
+        Consumer, TopicPartition = mod.Consumer, mod.TopicPartition
         cfg, deser = self._cfg, self._deser
 
         # One Consumer per source REPLICA (keyed by replica index): librdkafka
@@ -132,11 +141,7 @@ class Kafka_Sink_Builder(_BasicBuilder):
         return self
 
     def build(self):
-        if not _have_kafka():
-            raise RuntimeError(
-                "Kafka_Sink requires confluent_kafka (librdkafka); not "
-                "installed in this environment")
-        from confluent_kafka import Producer
+        Producer = _client().Producer
         cfg, ser = self._cfg, self._ser
         prod = Producer({"bootstrap.servers": cfg["brokers"]})
         failures = []
