@@ -249,3 +249,33 @@ def test_unterminated_pipe_rejected():
     mp.add(wf.Map_Builder(native.affine_map(0, 1, 0)).withOutputSchema([0]).build())
     with pytest.raises(RuntimeError, match="no consumers"):
         g.run()
+
+
+def test_python_user_reduce():
+    """Arbitrary user reduce fn(acc, v) -> acc' with per-key state and
+    KEYBY routing (reference reduce.hpp arbitrary functor)."""
+    n, n_keys, batch = 8000, 5, 256
+    g = wf.PipeGraph("pyred")
+    src = (wf.Source_Builder(native.seq_source(n, n_keys, batch))
+           .withParallelism(1).withOutputSchema([0])
+           .withOutputBatchSize(batch).build())
+    mp = g.add_source(src)
+    mp.add(wf.Reduce_Builder(lambda acc, v: max(acc, int(v)))
+           .withInitialState(-1)
+           .withParallelism(2).withOutputSchema([0])
+           .withOutputBatchSize(batch).withKeyBy('carried').build())
+    rows = dict(last={})
+
+    def sink(cols):
+        for k, v in zip(cols['key'].tolist(), cols['c0'].tolist()):
+            rows['last'][k] = v
+
+    snk = wf.Sink_Builder(sink).withParallelism(1).build()
+    snk.out_schema = [0]
+    mp.add_sink(snk)
+    g.run()
+    # final per-key running max == max v per key (seq: v=1..n, key=v%5)
+    exp = {}
+    for v in range(1, n + 1):
+        exp[v % n_keys] = max(exp.get(v % n_keys, -1), v)
+    assert rows['last'] == exp

@@ -90,21 +90,56 @@ class FlatMap_Builder(_BasicBuilder):
 
 
 class Reduce_Builder(_BasicBuilder):
-    """Keyed running aggregate — KEYBY routing in (reference reduce.hpp:285)."""
+    """Keyed running aggregate — KEYBY routing in (reference reduce.hpp:285).
+
+    Accepts a native spec (native.keyed_sum_reduce / keyed_reduce_f) or an
+    ARBITRARY user function fn(acc, value) -> new_acc applied per tuple in
+    key order, per-key state in the replica (the reference's
+    void(const tuple_t&, result_t&) signature): emits the updated
+    (key, acc) per input tuple."""
     _kind = "reduce"
     _needs_key = True
 
     def withInitialState(self, v):
         """Initial per-key accumulator value (reference builders.hpp:627)."""
         lg = self._op.logic
+        if callable(lg) and not isinstance(lg, NativeLogic):
+            self._op.extra['reduce_init'] = v
+            return self
         if not (isinstance(lg, NativeLogic) and lg.kind == "reduce"):
-            raise TypeError("withInitialState applies to native keyed reduces")
+            raise TypeError("withInitialState applies to keyed reduces")
         ip = list(lg.iparams)
         while len(ip) < 2:
             ip.append(0)
         ip[1] = int(v)
         self._op.logic = NativeLogic(lg.kind, lg.spec, list(lg.fparams), ip)
         return self
+
+    def build(self):
+        op = super().build()
+        fn = op.logic
+        if callable(fn) and not isinstance(fn, NativeLogic):
+            import numpy as np
+            state = {}
+            init = op.extra.get('reduce_init', 0)
+            out_dt = {0: np.int64, 1: np.float64, 2: np.float32,
+                      3: np.uint64, 4: np.int32}.get(
+                          (op.out_schema or [0])[0], np.float64)
+
+            def reduce_batch(cols, _fn=fn):
+                keys = cols['key']
+                v = cols['c0']
+                out = np.empty(len(keys), dtype=out_dt)
+                for i in range(len(keys)):
+                    k = int(keys[i])
+                    a = _fn(state.get(k, init), v[i])
+                    state[k] = a
+                    out[i] = a
+                return {"ts": cols['ts'], "key": keys, "c0": out}
+
+            op.kind = "flatmap"
+            op.logic = reduce_batch
+        return op
 
 
 class Sink_Builder(_BasicBuilder):
