@@ -166,21 +166,33 @@ def build_join_graph(n_tuples, batch, n_keys, par):
     return g, snk
 
 
-def build_cpu_graph(n_tuples, batch):
+def build_cpu_graph(n_tuples, batch, par=1):
+    """Config #1: Source -> Map -> Filter -> Sink.  par=1 chains all four
+    into ONE thread (round-1 shape); par>1 runs `par` source replicas
+    each chained with its own map+filter+sink replica (rebalanced
+    stream shards, zero cross-thread shuffles — operator replication,
+    SURVEY §2.10)."""
     import windflow_amd as wf
     from windflow_amd import native
-    src = (wf.Source_Builder(native.seq_source(n_tuples, 1000, batch))
-           .withParallelism(1).withOutputSchema([0])
-           .withOutputBatchSize(batch).build())
     g = wf.PipeGraph("bench_cpu")
-    mp = g.add_source(src)
-    mp.chain(wf.Map_Builder(native.affine_map(0, 3, 1)).withParallelism(1)
-             .withOutputSchema([0]).withOutputBatchSize(batch).build())
-    mp.chain(wf.Filter_Builder(native.mod_filter(0, 5, 0)).withParallelism(1)
-             .withOutputSchema([0]).withOutputBatchSize(batch).build())
-    snk = wf.Sink_Builder(native.count_sink()).withParallelism(1).build()
-    mp.add_sink(snk)
-    return g, snk
+    per = n_tuples // par
+    pipes = []
+    snks = []
+    for r in range(par):
+        src = (wf.Source_Builder(native.seq_source(per, 1000, batch,
+                                                   value_offset=r * per))
+               .withParallelism(1).withOutputSchema([0])
+               .withOutputBatchSize(batch).build())
+        mp = g.add_source(src)
+        mp.chain(wf.Map_Builder(native.affine_map(0, 3, 1)).withParallelism(1)
+                 .withOutputSchema([0]).withOutputBatchSize(batch).build())
+        mp.chain(wf.Filter_Builder(native.mod_filter(0, 5, 0)).withParallelism(1)
+                 .withOutputSchema([0]).withOutputBatchSize(batch).build())
+        snk = wf.Sink_Builder(native.count_sink()).withParallelism(1).build()
+        mp.chain_sink(snk)
+        snks.append(snk)
+        pipes.append(mp)
+    return g, snks[0]
 
 
 def main():
@@ -195,6 +207,8 @@ def main():
     ap.add_argument("--batch", type=int, default=8_388_608)
     ap.add_argument("--keys", type=int, default=8192,
                     help="distinct keys per rank")
+    ap.add_argument("--par", type=int, default=1,
+                    help="CPU configs: operator parallelism")
     ap.add_argument("--win", type=int, default=1000)
     ap.add_argument("--slide", type=int, default=100)
     ap.add_argument("--config", choices=["ffat", "ffat_x", "a2a", "mapfilter", "cpu", "join"], default="ffat")
@@ -238,8 +252,10 @@ def main():
 
     if args.config in ("cpu", "join"):
         B = min(B, 65536)
-        build = (build_cpu_graph if args.config == "cpu"
-                 else lambda n, b: build_join_graph(n, b, args.keys, 4))
+        build = ((lambda n, b: build_cpu_graph(n, b, args.par))
+                 if args.config == "cpu"
+                 else lambda n, b: build_join_graph(n, b, args.keys,
+                                                    args.par if args.par > 1 else 4))
         gw, _ = build(W * B, B)
         gw.run()
         g, snk = build(K * B, B)
