@@ -652,6 +652,14 @@ PYBIND11_MODULE(_core, m) {
           },
           py::arg("n"), py::arg("n_keys"), py::arg("win") = 1000,
           py::arg("slide") = 100, py::arg("iters") = 20, py::arg("vik") = 0);
+    m.def("debug_a2a_stage_times",
+          [](int64_t n, int64_t n_keys, int iters) {
+              py::list out;
+              for (auto& [k, v] : debug_a2a_stage_times(n, n_keys, iters))
+                  out.append(py::make_tuple(k, v));
+              return out;
+          },
+          py::arg("n"), py::arg("n_keys") = 8192, py::arg("iters") = 20);
     m.def("debug_key_slots",
           [](py::array_t<uint64_t> keys, int64_t max_keys) {
               auto r = debug_key_slots_host(keys.data(), keys.shape(0), max_keys);

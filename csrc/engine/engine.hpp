@@ -499,6 +499,8 @@ std::vector<uint32_t> debug_key_slots_host(const uint64_t* keys, int64_t n,
                                            int64_t max_keys);
 std::vector<std::pair<std::string, double>> debug_ffat_stage_times(
     int64_t n, int64_t n_keys, int64_t win, int64_t slide, int iters, int vik);
+std::vector<std::pair<std::string, double>> debug_a2a_stage_times(
+    int64_t n, int64_t n_keys, int iters);
 
 // RCCL bootstrap: rank 0 generates the id, broadcasts it out-of-band
 // (torch.distributed store), every rank passes it to Engine::rccl_id.

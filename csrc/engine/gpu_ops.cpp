@@ -1174,8 +1174,11 @@ struct GpuExchangeLogic : GpuLogicBase {
     int rank = 0, world = 1;
     ncclComm_t comm = nullptr;
     int bits = 1;
-    // scratch
-    uint32_t *dest = nullptr, *idx = nullptr, *dest_t = nullptr, *idx_t = nullptr;
+    // scratch (dest ping-pongs: batch N's bucket runs on meta_stream while
+    // batch N-1's sort still READS its dest array on the main stream)
+    uint32_t* dest2[2] = {nullptr, nullptr};
+    uint32_t *idx = nullptr, *dest_t = nullptr, *idx_t = nullptr;
+    int dest_flip = 0;
     uint32_t* hist = nullptr;
     uint32_t* d_counts = nullptr;
     int64_t* d_meta = nullptr;      // [world+2] send metadata
@@ -1209,7 +1212,8 @@ struct GpuExchangeLogic : GpuLogicBase {
     void init_device() override {
         comm = get_rccl_comm(eng->rccl_id, rank, world, device);
         auto& A = arena(device);
-        dest = (uint32_t*)A.get(4 * out_cap);
+        dest2[0] = (uint32_t*)A.get(4 * out_cap);
+        dest2[1] = (uint32_t*)A.get(4 * out_cap);
         idx = (uint32_t*)A.get(4 * out_cap);
         dest_t = (uint32_t*)A.get(4 * out_cap);
         idx_t = (uint32_t*)A.get(4 * out_cap);
@@ -1245,7 +1249,6 @@ struct GpuExchangeLogic : GpuLogicBase {
     // the collective instead of serializing behind it.
     void start_meta(bool have_counts, int64_t wm, bool done) {
         if (!have_counts) wfa_fill_u32(meta_stream, d_counts, 0, world);
-        else HIPCHK(hipStreamWaitEvent(meta_stream, ev_counts, 0));
         wfa_pack_meta(meta_stream, d_counts, world, done ? WM_MAX : wm,
                       done ? 1 : 0, d_meta);
         if (world > 1)
@@ -1371,13 +1374,21 @@ struct GpuExchangeLogic : GpuLogicBase {
         int64_t t1 = now_us();
         if (sb->ready_event)
             HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)sb->ready_event, 0));
-        // bucket -> counts first: the metadata rendezvous departs on its own
-        // stream while the sort/gather still run on `stream`
-        wfa_bucket_by_key(stream, db->key, n, world, dest);
-        wfa_count_u32(stream, dest, n, d_counts, world);
-        HIPCHK(hipEventRecord(ev_counts, stream));
+        // bucket + count + metadata rendezvous all run on meta_stream,
+        // fully decoupled from the main stream's backlog (the previous
+        // batch's sort/gather/sends): the per-batch host sync then waits
+        // only for THIS batch's input + tiny kernels, not the pipeline
+        uint32_t* dest = dest2[dest_flip];
+        dest_flip ^= 1;
+        if (db->ready_event)
+            HIPCHK(hipStreamWaitEvent(meta_stream, (hipEvent_t)db->ready_event, 0));
+        wfa_bucket_by_key(meta_stream, db->key, n, world, dest);
+        wfa_count_u32(meta_stream, dest, n, d_counts, world);
+        HIPCHK(hipEventRecord(ev_counts, meta_stream));
         start_meta(true, wm, false);
-        // stable sort by dest -> contiguous per-dest send layout
+        // stable sort by dest -> contiguous per-dest send layout (main
+        // stream; waits for this batch's bucket output)
+        HIPCHK(hipStreamWaitEvent(stream, ev_counts, 0));
         uint32_t *od, *oi;
         wfa_sort_pairs2(stream, dest, idx, dest_t, idx_t, nullptr, nullptr, hist,
                         n, bits, &od, &oi, nullptr, /*implicit_iota=*/1,
@@ -1533,6 +1544,116 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
     throw std::runtime_error("unknown gpu logic: " + kind);
 }
 
+// Per-stage hipEvent timing of the a2a exchange + reduce chain on
+// synthetic data (explain the a2a cost).  Stages:
+// gen, map, bucket, count, sort(dest), gather_rows, selfD2D, meta(D2H),
+// then the downstream reduce chain: key_to_slot, sort(slot), segments,
+// seg_reduce.  All on one stream, world=1 semantics.
+std::vector<std::pair<std::string, double>> debug_a2a_stage_times(
+    int64_t n, int64_t n_keys, int iters) {
+    HIPCHK(hipSetDevice(0));
+    hipStream_t s = nullptr;
+    auto& A = arena(0);
+    int world = 1;
+    Schema sch;
+    sch.payload.push_back(DType::F32);
+    int64_t* d_ts = (int64_t*)A.get(8 * n);
+    uint64_t* d_key = (uint64_t*)A.get(8 * n);
+    float* d_val = (float*)A.get(4 * n);
+    int64_t* o_ts = (int64_t*)A.get(8 * n);
+    uint64_t* o_key = (uint64_t*)A.get(8 * n);
+    float* o_val = (float*)A.get(4 * n);
+    int64_t* r_ts = (int64_t*)A.get(8 * n);
+    uint64_t* r_key = (uint64_t*)A.get(8 * n);
+    float* r_val = (float*)A.get(4 * n);
+    uint32_t* dest = (uint32_t*)A.get(4 * n);
+    uint32_t* idx = (uint32_t*)A.get(4 * n);
+    uint32_t* dest_t = (uint32_t*)A.get(4 * n);
+    uint32_t* idx_t = (uint32_t*)A.get(4 * n);
+    uint32_t* hist = (uint32_t*)A.get(4 * wfa_sort_hist_u32(n));
+    uint32_t* d_counts = (uint32_t*)A.get(64);
+    int64_t* d_meta = (int64_t*)A.get(8 * 3);
+    int64_t* h_meta = nullptr;
+    HIPCHK(hipHostMalloc((void**)&h_meta, 64, hipHostMallocDefault));
+    void** d_colptrs = (void**)A.get(64);
+    int* d_esize = (int*)A.get(64);
+    {
+        int es = 4;
+        HIPCHK(hipMemcpy(d_esize, &es, 4, hipMemcpyHostToDevice));
+        void* ptrs[2] = {d_val, o_val};
+        HIPCHK(hipMemcpy(d_colptrs, ptrs, 16, hipMemcpyHostToDevice));
+    }
+    KeyedScratch ks;
+    ks.alloc(0, n, 4 * n_keys, s);
+    uint64_t* rk = (uint64_t*)A.get(8 * n);
+    float* rv = (float*)A.get(4 * n);
+    int64_t* rt = (int64_t*)A.get(8 * n);
+    int64_t* d_on = (int64_t*)A.get(64);
+    wfa_gen_batch(s, d_ts, d_key, d_val, 2, n, 0, 42, n_keys);
+
+    constexpr int NS = 11;
+    const char* names[NS] = {"gen",        "map",    "bucket", "count",
+                             "sort_dest",  "gather", "selfD2D", "meta_d2h",
+                             "key_to_slot", "sort_slot+seg", "seg_reduce"};
+    hipEvent_t ev[NS + 1];
+    for (auto& e : ev) HIPCHK(hipEventCreate(&e));
+    double acc[NS] = {0};
+    for (int it = -2; it < iters; ++it) {
+        HIPCHK(hipEventRecord(ev[0], s));
+        wfa_gen_batch(s, d_ts, d_key, d_val, 2, n, (int64_t)it * n, 42, n_keys);
+        HIPCHK(hipEventRecord(ev[1], s));
+        wfa_map_apply(s, 2, d_val, 2, n, 2.0, 0.5);
+        HIPCHK(hipEventRecord(ev[2], s));
+        wfa_bucket_by_key(s, d_key, n, world, dest);
+        HIPCHK(hipEventRecord(ev[3], s));
+        wfa_count_u32(s, dest, n, d_counts, world);
+        HIPCHK(hipEventRecord(ev[4], s));
+        uint32_t *od, *oi;
+        wfa_sort_pairs2(s, dest, idx, dest_t, idx_t, nullptr, nullptr, hist, n,
+                        1, &od, &oi, nullptr, 1, 0);
+        HIPCHK(hipEventRecord(ev[5], s));
+        const void* ci[1] = {d_val};
+        void* co[1] = {o_val};
+        (void)ci; (void)co;
+        wfa_gather_rows(s, oi, n, d_ts, o_ts, d_key, o_key,
+                        (const void* const*)d_colptrs,
+                        (void* const*)(d_colptrs + 1), d_esize, 1);
+        HIPCHK(hipEventRecord(ev[6], s));
+        HIPCHK(hipMemcpyAsync(r_ts, o_ts, 8 * n, hipMemcpyDeviceToDevice, s));
+        HIPCHK(hipMemcpyAsync(r_key, o_key, 8 * n, hipMemcpyDeviceToDevice, s));
+        HIPCHK(hipMemcpyAsync(r_val, o_val, 4 * n, hipMemcpyDeviceToDevice, s));
+        HIPCHK(hipEventRecord(ev[7], s));
+        wfa_pack_meta(s, d_counts, world, 123, 0, d_meta);
+        HIPCHK(hipMemcpyAsync(h_meta, d_meta, 24, hipMemcpyDeviceToHost, s));
+        HIPCHK(hipEventRecord(ev[8], s));
+        // downstream reduce chain over the received batch
+        wfa_key_to_slot(s, r_key, n, ks.tab, ks.d_nslots, ks.table_cap, ks.slot,
+                        ks.slot_to_key);
+        HIPCHK(hipEventRecord(ev[9], s));
+        uint32_t *os2, *oi2;
+        wfa_sort_pairs2(s, ks.slot, ks.idx, ks.slot_t, ks.idx_t, nullptr,
+                        nullptr, ks.hist, n, ks.bits, &os2, &oi2, nullptr, 1, 0);
+        wfa_segments(s, os2, n, ks.hist, ks.seg_start, ks.seg_slot, ks.d_nseg, 0);
+        HIPCHK(hipEventRecord(ev[10], s));
+        wfa_segment_reduce_wave(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
+                                r_val, 2, oi2, r_ts, 0, ks.slot_to_key, rk, rv,
+                                rt, d_on);
+        HIPCHK(hipEventRecord(ev[11], s));
+        HIPCHK(hipStreamSynchronize(s));
+        if (it >= 0)
+            for (int k = 0; k < NS; ++k) {
+                float ms = 0;
+                HIPCHK(hipEventElapsedTime(&ms, ev[k], ev[k + 1]));
+                acc[k] += ms * 1000.0;
+            }
+    }
+    for (auto& e : ev) (void)hipEventDestroy(e);
+    (void)hipHostFree(h_meta);
+    std::vector<std::pair<std::string, double>> out;
+    for (int k = 0; k < NS; ++k) out.push_back({names[k], acc[k] / iters});
+    return out;
+}
+
 // Per-stage hipEvent timing of the keyed FFAT CB chain (slot -> sort ->
 // segments -> fire offsets -> fold) on synthetic data.  The safe substitute
 // for rocprofv3 kernel stats on this pool (profiling runs hang boxes);
@@ -1649,6 +1770,10 @@ std::vector<uint32_t> debug_key_slots_host(const uint64_t*, int64_t, int64_t) {
 std::vector<std::pair<std::string, double>> debug_ffat_stage_times(int64_t, int64_t,
                                                                    int64_t, int64_t,
                                                                    int, int) {
+    throw std::runtime_error("built without HIP");
+}
+std::vector<std::pair<std::string, double>> debug_a2a_stage_times(int64_t, int64_t,
+                                                                  int) {
     throw std::runtime_error("built without HIP");
 }
 std::shared_ptr<OpLogic> make_gpu_logic(const std::string&, const std::string&,

@@ -132,10 +132,12 @@ class MultiPipe:
         callable(cols) returning int32 branch ids (one branch per tuple) or
         a sequence of per-branch bool masks (one tuple to ANY subset of
         branches — reference splitting_emitter.hpp vector<integral_t>)."""
+        up = self.graph.nodes[self.tails[0][0]]
         op = Operator(kind="split", logic=split_logic, name="split",
-                      parallelism=self.graph.nodes[self.tails[0][0]].parallelism,
+                      parallelism=up.parallelism,
                       n_branches=n_branches,
-                      out_schema=self.graph.nodes[self.tails[0][0]].last.out_schema)
+                      out_schema=up.last.out_schema,
+                      out_batch=up.last.out_batch)
         node_idx = self._attach(op)
         self.closed = True
         return self
