@@ -281,7 +281,7 @@ struct KeyedScratch {
     int v_dt = 2;                    // effective dtype of v_as_f32 (2/5)
 
     void group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
-               bool want_vik = false);
+               bool want_vik = false, bool want_carry = false);
 };
 
 // ===== hiprtc JIT compile cache =====

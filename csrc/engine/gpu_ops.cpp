@@ -280,7 +280,7 @@ void KeyedScratch::alloc(int dev, int64_t cap_, int64_t mk, hipStream_t s) {
 // one cache line per tuple.  Numerics are identical (same bf16 bits).
 // vcol < 0: no value cast (stateful map/filter operate in place)
 void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
-                         bool want_vik) {
+                         bool want_vik, bool want_carry) {
         int64_t n = db->count;
         if (n > cap)
             throw std::runtime_error("batch larger than keyed scratch capacity — "
@@ -303,6 +303,24 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
         }
         wfa_key_to_slot(s, db->key, n, tab, d_nslots, table_cap, slot,
                         slot_to_key);
+        if (want_carry && vcol >= 0 && (int)db->schema.payload[vcol] == 2) {
+            // carry the f32 value bits as a second sort payload: the fold
+            // then reads values SEQUENTIALLY at the sorted position instead
+            // of one random cache line per tuple (A/B: wins for the
+            // latency-bound segmented reduce; the FFAT folds hide the
+            // gather and keep the cheaper LDS scatter instead)
+            wfa_cast(s, db->cols[vcol], 2, v_f32, 2, n);  // sort ping buffer
+            uint32_t *os, *oi, *ov;
+            wfa_sort_pairs2(s, slot, idx, slot_t, idx_t, (uint32_t*)v_f32,
+                            (uint32_t*)v_sorted, hist, n, bits, &os, &oi, &ov,
+                            /*implicit_iota=*/1, /*base_shift=*/0);
+            idx_sorted = oi;
+            v_as_f32 = ov;
+            v_dt = 7;
+            wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg, 0);
+            if (ctx.stats) ctx.stats->num_kernels += 6 + 3 * ((bits + 3) / 4);
+            return;
+        }
         uint32_t *os, *oi;
         wfa_sort_pairs2(s, slot, idx, slot_t, idx_t, nullptr, nullptr, hist, n,
                         bits, &os, &oi, nullptr, /*implicit_iota=*/1,
@@ -630,7 +648,7 @@ struct GpuReduceLogic : GpuLogicBase {
         ensure_init();
         Batch* db = input_on_device(in, ctx);
         int64_t n = db->count;
-        ks.group(stream, db, vcol, ctx);
+        ks.group(stream, db, vcol, ctx, /*want_vik=*/false, /*want_carry=*/true);
         Batch* ob = get_dev();
         wfa_segment_reduce_wave(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                                 ks.v_as_f32, ks.v_dt, ks.idx_sorted, db->ts, comb,
@@ -1261,6 +1279,33 @@ struct GpuExchangeLogic : GpuLogicBase {
                               hipMemcpyDeviceToHost, meta_stream));
     }
 
+    // analyzed metadata of the current round (valid after sync_meta)
+    struct Meta {
+        int64_t roff[9] = {0};
+        int64_t total = 0;
+        int64_t remote_in = 0;   // rows arriving from other ranks
+        int64_t remote_out = 0;  // rows leaving to other ranks
+        bool all_done = true;
+        int64_t min_wm = WM_MAX;
+    };
+
+    Meta sync_meta(bool sending) {
+        HIPCHK(hipStreamSynchronize(meta_stream));
+        Meta m;
+        const int64_t* scnt = sending ? h_meta + (int64_t)rank * (world + 2)
+                                      : nullptr;
+        for (int p = 0; p < world; ++p) {
+            const int64_t* mp = h_meta + (int64_t)p * (world + 2);
+            m.roff[p] = m.total;
+            m.total += mp[rank];
+            if (p != rank) m.remote_in += mp[rank];
+            if (scnt && p != rank) m.remote_out += scnt[p];
+            if (mp[world + 1] == 0) m.all_done = false;
+            m.min_wm = std::min(m.min_wm, mp[world]);
+        }
+        return m;
+    }
+
     // one collective round: metadata already in flight on meta_stream
     // (start_meta); sync META ONLY, lay out recvs, then move rows.
     // Returns true when every rank reported done.
@@ -1386,6 +1431,28 @@ struct GpuExchangeLogic : GpuLogicBase {
         wfa_count_u32(meta_stream, dest, n, d_counts, world);
         HIPCHK(hipEventRecord(ev_counts, meta_stream));
         start_meta(true, wm, false);
+        Meta m = sync_meta(true);
+        // WFA_XCHG_NO_SELFPASS=1 forces the full sort/gather/transfer path
+        // (keeps the world=1 device tests exercising the real machinery)
+        if (m.remote_in == 0 && m.remote_out == 0 &&
+            !getenv("WFA_XCHG_NO_SELFPASS")) {
+            // pure-self round: every row of every rank stays home this
+            // round — forward the batch untouched (no dest sort, no
+            // gather, no self copy).  At world=1 EVERY round is pure-self
+            // (the round-1 path burned ~280 us/batch re-materializing an
+            // identity permutation); at world>1 this engages whenever the
+            // allgathered count matrix is diagonal.
+            release(sb);  // staged send batch unused
+            int64_t out_wm = m.min_wm == WM_MAX ? cur_wm : m.min_wm;
+            cur_wm = std::max(cur_wm, out_wm);
+            db->watermark = out_wm;
+            if (ctx.stats) ctx.stats->num_kernels += 3;
+            out.emit(db);
+            if (wfa_prof())
+                fprintf(stderr, "[prof] xchg n=%ld self-forward meta=%ld us\n",
+                        (long)n, (long)(now_us() - t1));
+            return;
+        }
         // stable sort by dest -> contiguous per-dest send layout (main
         // stream; waits for this batch's bucket output)
         HIPCHK(hipStreamWaitEvent(stream, ev_counts, 0));

@@ -35,7 +35,9 @@ __device__ __forceinline__ float wfa_val_at(const void* v, int vdt, int64_t i) {
     // vdt 2 = f32, 5 = bf16 (u16), 6 = bf16 packed in the low 16 bits of the
     // SORTED u32 slot array (value-in-key: the fold reads values
     // sequentially instead of gathering one cache line per tuple);
-    // i64 inputs are pre-cast to f32 by group()
+    // 7 = f32 carried through the sort as a second payload (also read
+    // sequentially at the SORTED position); i64 inputs pre-cast by group()
+    if (vdt == 7) return ((const float*)v)[i];
     if (vdt == 6) {
         union { uint32_t u; float f; } c;
         c.u = ((const uint32_t*)v)[i] << 16;
@@ -777,7 +779,9 @@ __global__ void k_seg_reduce_wave(const uint32_t* seg_start, const uint32_t* seg
         int64_t tmax = INT64_MIN;
         for (int64_t i = b + lane; i < e; i += 64) {
             const uint32_t r = idx_sorted[i];
-            float x = wfa_val_at(v_orig, vdt, r);
+            // vdt >= 6: value rides the sort (VIK / carried payload) and is
+            // read at the SORTED position — sequential, not gathered
+            float x = wfa_val_at(v_orig, vdt, vdt >= 6 ? i : r);
             acc = (comb == 0 || comb == 3)
                       ? acc + x
                       : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));

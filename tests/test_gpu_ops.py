@@ -406,10 +406,13 @@ def test_gpu_ffat_tb_vs_oracle():
             assert abs(v_g - v_r) <= 1e-3 * max(1.0, abs(v_r)), (k, v_g, v_r)
 
 
-def test_gpu_keyby_exchange_world1():
+def test_gpu_keyby_exchange_world1(monkeypatch):
     """RCCL self-exchange (world=1): every row routes back to rank 0, so the
-    pipeline is value-preserving; exercises the full bucket->sort->gather->
-    allgather->send/recv path on device."""
+    pipeline is value-preserving; WFA_XCHG_NO_SELFPASS forces the full
+    bucket->sort->gather->allgather->transfer machinery (the production
+    world=1 path self-forwards without copying)."""
+    import os
+    monkeypatch.setenv("WFA_XCHG_NO_SELFPASS", "1")
     from windflow_amd import _core
     from windflow_amd.builders_gpu import KeyBy_Exchange_GPU_Builder
     n, n_keys, b = 100_000, 97, 25_000
