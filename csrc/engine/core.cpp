@@ -74,6 +74,7 @@ Batch* Pool::try_pop() {
     b->watermark = 0;
     b->punct = false;
     b->stream_tag = -1;
+    b->ts_mono = false;
     b->refcnt.store(1, std::memory_order_relaxed);
     return b;
 }
@@ -113,6 +114,7 @@ Batch* clone(Batch* b, Pool& pool) {
     c->watermark = b->watermark;
     c->punct = b->punct;
     c->stream_tag = b->stream_tag;
+    c->ts_mono = b->ts_mono;
     memcpy(c->ts, b->ts, b->count * 8);
     memcpy(c->key, b->key, b->count * 8);
     for (size_t i = 0; i < b->cols.size(); ++i)

@@ -76,6 +76,10 @@ struct Batch {
     void* ready_event = nullptr;// hipEvent_t signalled when contents valid
 
     int64_t born_us = 0;        // host clock when the source filled it
+    // ts column is nondecreasing in row order (set by monotonic sources,
+    // preserved by order-preserving ops): lets keyed reductions take the
+    // per-segment LAST row as the ts max instead of a per-row gather
+    bool ts_mono = false;
     int64_t* ts = nullptr;      // [capacity]
     uint64_t* key = nullptr;    // [capacity]
     // Deferred count: device producers may emit before the row count is

@@ -124,6 +124,7 @@ Batch* gpu_clone_batch(Batch* b) {
     c->watermark = b->watermark;
     c->stream_tag = b->stream_tag;
     c->born_us = b->born_us;
+    c->ts_mono = b->ts_mono;
     c->stream = s;
     HIPCHK(hipEventRecord((hipEvent_t)c->ready_event, s));
     return c;
@@ -154,6 +155,7 @@ struct GpuSourceLogic : GpuLogicBase {
         int64_t n = std::min<int64_t>(bsz, len - pos);
         wfa_gen_batch(stream, db->ts, db->key, db->cols[0], vdt, n, pos, seed, n_keys);
         db->count = n;
+        db->ts_mono = true;  // ts = start + i
         db->born_us = now_us();
         pos += n;
         db->watermark = pos - 1;
@@ -237,6 +239,7 @@ struct GpuFilterLogic : GpuLogicBase {
         ob->watermark = db->watermark;
         ob->stream_tag = db->stream_tag;
         ob->born_us = db->born_us;
+        ob->ts_mono = db->ts_mono;  // compaction preserves row order
         if (ctx.stats) ctx.stats->num_kernels += 3;
         release_after_use(db);
         record_ready(ob);
@@ -510,6 +513,7 @@ struct GpuJitFilterLogic : GpuLogicBase {
         ob->watermark = db->watermark;
         ob->stream_tag = db->stream_tag;
         ob->born_us = db->born_us;
+        ob->ts_mono = db->ts_mono;  // compaction preserves row order
         if (ctx.stats) ctx.stats->num_kernels += 4;
         release_after_use(db);
         record_ready(ob);
@@ -652,7 +656,8 @@ struct GpuReduceLogic : GpuLogicBase {
         Batch* ob = get_dev();
         wfa_segment_reduce_wave(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                                 ks.v_as_f32, ks.v_dt, ks.idx_sorted, db->ts, comb,
-                                ks.slot_to_key, ob->key, ob->cols[0], ob->ts, d_on);
+                                db->ts_mono ? 1 : 0, ks.slot_to_key, ob->key,
+                                ob->cols[0], ob->ts, d_on);
         HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
         ob->count = -1;
         ob->watermark = db->watermark;
@@ -1697,14 +1702,16 @@ std::vector<std::pair<std::string, double>> debug_a2a_stage_times(
         wfa_key_to_slot(s, r_key, n, ks.tab, ks.d_nslots, ks.table_cap, ks.slot,
                         ks.slot_to_key);
         HIPCHK(hipEventRecord(ev[9], s));
-        uint32_t *os2, *oi2;
-        wfa_sort_pairs2(s, ks.slot, ks.idx, ks.slot_t, ks.idx_t, nullptr,
-                        nullptr, ks.hist, n, ks.bits, &os2, &oi2, nullptr, 1, 0);
+        uint32_t *os2, *oi2, *ov2;
+        wfa_cast(s, r_val, 2, ks.v_f32, 2, n);
+        wfa_sort_pairs2(s, ks.slot, ks.idx, ks.slot_t, ks.idx_t,
+                        (uint32_t*)ks.v_f32, (uint32_t*)ks.v_sorted, ks.hist, n,
+                        ks.bits, &os2, &oi2, &ov2, 1, 0);
         wfa_segments(s, os2, n, ks.hist, ks.seg_start, ks.seg_slot, ks.d_nseg, 0);
         HIPCHK(hipEventRecord(ev[10], s));
         wfa_segment_reduce_wave(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                                r_val, 2, oi2, r_ts, 0, ks.slot_to_key, rk, rv,
-                                rt, d_on);
+                                ov2, 7, oi2, r_ts, 0, /*ts_last=*/1,
+                                ks.slot_to_key, rk, rv, rt, d_on);
         HIPCHK(hipEventRecord(ev[11], s));
         HIPCHK(hipStreamSynchronize(s));
         if (it >= 0)

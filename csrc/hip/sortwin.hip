@@ -465,13 +465,15 @@ __global__ void k_rs8_scatter(const uint32_t* keys, const uint32_t* vals,
 __global__ void k_rs8_scatter_lds(const uint32_t* keys, const uint32_t* vals,
                                   int64_t n, int shift, const uint32_t* hist,
                                   int64_t nblocks, const uint32_t* dbase,
-                                  uint32_t* keys_out, uint32_t* vals_out) {
+                                  uint32_t* keys_out, uint32_t* vals_out,
+                                  const uint32_t* vals2, uint32_t* vals2_out) {
     __shared__ uint32_t gbase[256];
     __shared__ uint32_t waveCnt[WFA_THREADS / 64][256];
     __shared__ uint32_t wavePre[WFA_THREADS / 64][256];
     __shared__ uint32_t localBase[256];
     __shared__ uint32_t sk[RS8_PER_BLOCK];
     __shared__ uint32_t sv[RS8_PER_BLOCK];
+    __shared__ uint32_t sv2[RS8_PER_BLOCK];  // second payload (value carry)
     __shared__ uint8_t sd[RS8_PER_BLOCK];
     const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
     const uint64_t lt = ((uint64_t)1 << lane) - 1;
@@ -482,7 +484,7 @@ __global__ void k_rs8_scatter_lds(const uint32_t* keys, const uint32_t* vals,
     __syncthreads();
     int64_t blockStart = (int64_t)blockIdx.x * RS8_PER_BLOCK;
     int64_t waveBase = blockStart + wave * RS8_PER_WAVE;
-    uint32_t mk[RS8_IPT], mv[RS8_IPT], mr[RS8_IPT];
+    uint32_t mk[RS8_IPT], mv[RS8_IPT], mv2[RS8_IPT], mr[RS8_IPT];
     int nit = 0;
 #pragma unroll
     for (int j = 0; j < RS8_IPT; ++j) {
@@ -502,6 +504,7 @@ __global__ void k_rs8_scatter_lds(const uint32_t* keys, const uint32_t* vals,
             uint32_t rank = pre + (uint32_t)__popcll(mask & lt);
             mk[j] = k;
             mv[j] = v;
+            if (vals2) mv2[j] = vals2[i];
             mr[j] = (d << 24) | (rank & 0xFFFFFF);
             if ((mask & lt) == 0) waveCnt[wave][d] = pre + (uint32_t)__popcll(mask);
             nit = j + 1;
@@ -538,6 +541,7 @@ __global__ void k_rs8_scatter_lds(const uint32_t* keys, const uint32_t* vals,
         uint32_t lpos = localBase[d] + wavePre[wave][d] + (mr[j] & 0xFFFFFF);
         sk[lpos] = mk[j];
         sv[lpos] = mv[j];
+        if (vals2) sv2[lpos] = mv2[j];
         sd[lpos] = (uint8_t)d;
     }
     __syncthreads();
@@ -547,6 +551,7 @@ __global__ void k_rs8_scatter_lds(const uint32_t* keys, const uint32_t* vals,
         uint32_t g = gbase[d] + (uint32_t)p - localBase[d];
         keys_out[g] = sk[p];
         vals_out[g] = sv[p];
+        if (vals2) vals2_out[g] = sv2[p];
     }
 }
 
@@ -577,12 +582,9 @@ extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
             hipLaunchKernelGGL(k_rs8_scan, dim3(256), dim3(256), 0, st, hist, nb, dt);
             hipLaunchKernelGGL(k_rs8_dbase, dim3(1), dim3(256), 0, st, dt, dbase);
             uint32_t* va_eff = (first && implicit_iota) ? nullptr : va;
-            if (val2)
-                hipLaunchKernelGGL(k_rs8_scatter, dim3(nb), dim3(WFA_THREADS), 0, st,
-                                   ka, va_eff, n, shift, hist, nb, dbase, kb, vb, wa, wb);
-            else
-                hipLaunchKernelGGL(k_rs8_scatter_lds, dim3(nb), dim3(WFA_THREADS), 0,
-                                   st, ka, va_eff, n, shift, hist, nb, dbase, kb, vb);
+            hipLaunchKernelGGL(k_rs8_scatter_lds, dim3(nb), dim3(WFA_THREADS), 0,
+                               st, ka, va_eff, n, shift, hist, nb, dbase, kb, vb,
+                               wa, wb);
             first = false;
             uint32_t* t;
             t = ka; ka = kb; kb = t;
@@ -766,7 +768,8 @@ __global__ void k_seg_reduce_wave(const uint32_t* seg_start, const uint32_t* seg
                                   const int64_t* d_nseg, int64_t n,
                                   const void* v_orig, const uint32_t* idx_sorted,
                                   const int64_t* ts_orig, int vdt, int comb,
-                                  const uint64_t* slot_to_key, uint64_t* out_key,
+                                  int ts_last, const uint64_t* slot_to_key,
+                                  uint64_t* out_key,
                                   void* out_val, int64_t* out_ts, int64_t* d_out_n) {
     const int64_t nseg = *d_nseg;
     const int lane = threadIdx.x & 63;
@@ -785,7 +788,10 @@ __global__ void k_seg_reduce_wave(const uint32_t* seg_start, const uint32_t* seg
             acc = (comb == 0 || comb == 3)
                       ? acc + x
                       : (comb == 1 ? fminf(acc, x) : fmaxf(acc, x));
-            if (ts_orig) tmax = max(tmax, ts_orig[r]);
+            // ts_last: the batch's ts column is nondecreasing in row order
+            // and the stable sort keeps per-segment rows in arrival order,
+            // so the segment max is its LAST row — skip the random gather
+            if (ts_orig && !ts_last) tmax = max(tmax, ts_orig[r]);
         }
         for (int o = 32; o; o >>= 1) {
             float ov = __shfl_down(acc, o, 64);
@@ -797,7 +803,9 @@ __global__ void k_seg_reduce_wave(const uint32_t* seg_start, const uint32_t* seg
         if (lane == 0) {
             ((float*)out_val)[j] = (comb == 3) ? (float)(e - b) : acc;
             out_key[j] = slot_to_key ? slot_to_key[seg_slot[j]] : (uint64_t)seg_slot[j];
-            if (out_ts) out_ts[j] = tmax;
+            if (out_ts)
+                out_ts[j] = (ts_orig && ts_last) ? ts_orig[idx_sorted[e - 1]]
+                                                 : tmax;
         }
     }
     if (blockIdx.x == 0 && threadIdx.x == 0) *d_out_n = nseg;
@@ -806,13 +814,13 @@ __global__ void k_seg_reduce_wave(const uint32_t* seg_start, const uint32_t* seg
 extern "C" void wfa_segment_reduce_wave(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
     const int64_t* d_nseg, int64_t n, const void* v_orig, int vdt,
-    const uint32_t* idx_sorted, const int64_t* ts_orig, int comb,
+    const uint32_t* idx_sorted, const int64_t* ts_orig, int comb, int ts_last,
     const uint64_t* slot_to_key, uint64_t* out_key, void* out_val,
     int64_t* out_ts, int64_t* d_out_n) {
     hipLaunchKernelGGL(k_seg_reduce_wave, dim3(WFA_MAX_BLOCKS), dim3(WFA_THREADS), 0,
                        (hipStream_t)s, seg_start, seg_slot, d_nseg, n, v_orig,
-                       idx_sorted, ts_orig, vdt, comb, slot_to_key, out_key, out_val,
-                       out_ts, d_out_n);
+                       idx_sorted, ts_orig, vdt, comb, ts_last, slot_to_key,
+                       out_key, out_val, out_ts, d_out_n);
 }
 
 

@@ -89,10 +89,12 @@ void wfa_segment_reduce(wfa_stream_t s, const uint32_t* seg_start,
                         int64_t* d_out_n);
 // wave-per-segment variant (f32 values; one wave cooperatively reduces
 // one key's segment — use for batches with large segments)
+// ts_last: ts column nondecreasing in row order (Batch::ts_mono) => the
+// per-segment ts max is its LAST row; skips the per-row random ts gather
 void wfa_segment_reduce_wave(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
     const int64_t* d_nseg, int64_t n, const void* v_orig, int vdt,
-    const uint32_t* idx_sorted, const int64_t* ts_orig, int comb,
+    const uint32_t* idx_sorted, const int64_t* ts_orig, int comb, int ts_last,
     const uint64_t* slot_to_key, uint64_t* out_key, void* out_val,
     int64_t* out_ts, int64_t* d_out_n);
 

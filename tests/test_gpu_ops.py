@@ -472,11 +472,12 @@ def test_gpu_reduce_keyed_sum():
            .withOutputSchema([2]).withOutputBatchSize(b).build())
     rd = (Reduce_GPU_Builder(native_gpu.gpu_keyed_reduce(native_gpu.COMB_SUM, 0, 256))
           .withOutputSchema([2]).withOutputBatchSize(b).build())
-    res = dict(s=0.0, n=0)
+    res = dict(s=0.0, n=0, rows=[])
 
     def pysink(cols):
         res['s'] += float(cols['c0'].astype(np.float64).sum())
         res['n'] += len(cols['c0'])
+        res['rows'].append((cols['key'].copy(), cols['ts'].copy()))
 
     g = wf.PipeGraph("red")
     p = g.add_source(src)
@@ -494,6 +495,15 @@ def test_gpu_reduce_keyed_sum():
     assert res['n'] == exp_n
     ref = float(val.astype(np.float64).sum())
     assert abs(res['s'] - ref) <= 2e-3 * max(1.0, abs(ref))
+    # output ts = max ts of that key within its batch (reference Reduce_GPU
+    # ts = max; exercises the monotonic-ts fast path: out_ts from the
+    # segment's LAST row must equal the true max)
+    assert len(res['rows']) == n // b
+    for bi, (ks_, ts_) in enumerate(res['rows']):
+        kb = key[bi * b:(bi + 1) * b]
+        tb = ts[bi * b:(bi + 1) * b]
+        for k, t in zip(ks_.tolist(), ts_.tolist()):
+            assert t == int(tb[kb == k].max()), (bi, k)
 
 
 def test_gpu_ffat_count_comb():
