@@ -462,6 +462,7 @@ __global__ void k_rs8_scatter(const uint32_t* keys, const uint32_t* vals,
 // the global write, so each wave stores contiguous runs per digit instead
 // of 4 B scattered words.  Used when there is no second payload (the hot
 // keyed-operator path).
+template <bool CARRY2>
 __global__ void k_rs8_scatter_lds(const uint32_t* keys, const uint32_t* vals,
                                   int64_t n, int shift, const uint32_t* hist,
                                   int64_t nblocks, const uint32_t* dbase,
@@ -473,7 +474,9 @@ __global__ void k_rs8_scatter_lds(const uint32_t* keys, const uint32_t* vals,
     __shared__ uint32_t localBase[256];
     __shared__ uint32_t sk[RS8_PER_BLOCK];
     __shared__ uint32_t sv[RS8_PER_BLOCK];
-    __shared__ uint32_t sv2[RS8_PER_BLOCK];  // second payload (value carry)
+    // second payload LDS only exists in the CARRY2 instantiation (a
+    // runtime-null variant still cost 16 KB static LDS = occupancy)
+    __shared__ uint32_t sv2[CARRY2 ? RS8_PER_BLOCK : 1];
     __shared__ uint8_t sd[RS8_PER_BLOCK];
     const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
     const uint64_t lt = ((uint64_t)1 << lane) - 1;
@@ -504,7 +507,7 @@ __global__ void k_rs8_scatter_lds(const uint32_t* keys, const uint32_t* vals,
             uint32_t rank = pre + (uint32_t)__popcll(mask & lt);
             mk[j] = k;
             mv[j] = v;
-            if (vals2) mv2[j] = vals2[i];
+            if (CARRY2) mv2[j] = vals2[i];
             mr[j] = (d << 24) | (rank & 0xFFFFFF);
             if ((mask & lt) == 0) waveCnt[wave][d] = pre + (uint32_t)__popcll(mask);
             nit = j + 1;
@@ -541,7 +544,7 @@ __global__ void k_rs8_scatter_lds(const uint32_t* keys, const uint32_t* vals,
         uint32_t lpos = localBase[d] + wavePre[wave][d] + (mr[j] & 0xFFFFFF);
         sk[lpos] = mk[j];
         sv[lpos] = mv[j];
-        if (vals2) sv2[lpos] = mv2[j];
+        if (CARRY2) sv2[lpos] = mv2[j];
         sd[lpos] = (uint8_t)d;
     }
     __syncthreads();
@@ -551,7 +554,7 @@ __global__ void k_rs8_scatter_lds(const uint32_t* keys, const uint32_t* vals,
         uint32_t g = gbase[d] + (uint32_t)p - localBase[d];
         keys_out[g] = sk[p];
         vals_out[g] = sv[p];
-        if (vals2) vals2_out[g] = sv2[p];
+        if (CARRY2) vals2_out[g] = sv2[p];
     }
 }
 
@@ -582,9 +585,14 @@ extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
             hipLaunchKernelGGL(k_rs8_scan, dim3(256), dim3(256), 0, st, hist, nb, dt);
             hipLaunchKernelGGL(k_rs8_dbase, dim3(1), dim3(256), 0, st, dt, dbase);
             uint32_t* va_eff = (first && implicit_iota) ? nullptr : va;
-            hipLaunchKernelGGL(k_rs8_scatter_lds, dim3(nb), dim3(WFA_THREADS), 0,
-                               st, ka, va_eff, n, shift, hist, nb, dbase, kb, vb,
-                               wa, wb);
+            if (val2)
+                hipLaunchKernelGGL(k_rs8_scatter_lds<true>, dim3(nb),
+                                   dim3(WFA_THREADS), 0, st, ka, va_eff, n, shift,
+                                   hist, nb, dbase, kb, vb, wa, wb);
+            else
+                hipLaunchKernelGGL(k_rs8_scatter_lds<false>, dim3(nb),
+                                   dim3(WFA_THREADS), 0, st, ka, va_eff, n, shift,
+                                   hist, nb, dbase, kb, vb, wa, wb);
             first = false;
             uint32_t* t;
             t = ka; ka = kb; kb = t;
