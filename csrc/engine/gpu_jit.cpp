@@ -813,10 +813,9 @@ struct GpuJitFfatLogic : GpuLogicBase {
 
     void cb_batch(Batch* db, Batch* ob, int64_t n, RuntimeCtx& ctx) {
         ks.group(stream, db, -1, ctx);
-        wfa_seg_last_ts(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                        ks.idx_sorted, db->ts, st_last);
         wfa_ffat_fire_offsets(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                              pane_len, P, S, st_fill, st_head, nf, d_on);
+                              pane_len, P, S, st_fill, st_head, nf, d_on,
+                              ks.idx_sorted, db->ts, st_last);
         ColArgs ca;
         ca.fill(db, fs);
         float* o[4];

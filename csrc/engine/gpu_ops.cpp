@@ -862,11 +862,10 @@ struct GpuFfatLogic : GpuLogicBase {
 
     void chain_cb(Batch* db, Batch* ob, int64_t n, RuntimeCtx& ctx) {
         ks.group(stream, db, vcol, ctx, /*want_vik=*/!use_tree);
-        wfa_seg_last_ts(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                        ks.idx_sorted, db->ts, st_last);
         uint32_t* nf = cb_nf;
         wfa_ffat_fire_offsets(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                              pane_len, P, S, st_fill, st_head, nf, d_on);
+                              pane_len, P, S, st_fill, st_head, nf, d_on,
+                              ks.idx_sorted, db->ts, st_last);
         if (use_tree)
             wfa_ffat_tree_fold(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                                ks.v_as_f32, ks.v_dt, ks.idx_sorted, db->ts, pane_len, P, S,
@@ -1065,7 +1064,8 @@ struct GpuGramLogic : GpuLogicBase {
         HIPCHK(hipMemcpyAsync(d_outcols, outp, 16 * 8, hipMemcpyHostToDevice, stream));
         // tumbling windows = panes of `win`, fire every pane (P=1, S=1)
         wfa_ffat_fire_offsets(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
-                              win, 1, 1, st_fill, st_head, nf, d_on);
+                              win, 1, 1, st_fill, st_head, nf, d_on, nullptr,
+                              nullptr, nullptr);
         wfa_gram_fold(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n, d_incols,
                       ks.idx_sorted, db->ts, win, st_fill, st_acc, st_head,
                       ks.slot_to_key, nf, ob->key, ob->col<int64_t>(0), d_outcols,
@@ -1802,7 +1802,8 @@ std::vector<std::pair<std::string, double>> debug_ffat_stage_times(
                      vik ? 16 : 0);
         HIPCHK(hipEventRecord(ev[3], s));
         wfa_ffat_fire_offsets(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n, pane, P,
-                              S, st_fill, st_head, nf, d_on);
+                              S, st_fill, st_head, nf, d_on, nullptr, nullptr,
+                              nullptr);
         HIPCHK(hipEventRecord(ev[4], s));
         wfa_ffat_cb_fold(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                          vik ? (const void*)os_ : (const void*)d_val, vik ? 6 : 5,

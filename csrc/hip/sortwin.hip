@@ -840,7 +840,9 @@ extern "C" void wfa_segment_reduce_wave(
 __global__ void k_fire_count(const uint32_t* seg_start, const uint32_t* seg_slot,
                              const int64_t* d_nseg, int64_t n, int64_t pane_len,
                              int64_t P, int64_t S, const uint32_t* st_fill,
-                             const uint32_t* st_head, uint32_t* nf) {
+                             const uint32_t* st_head, uint32_t* nf,
+                             const uint32_t* idx_sorted, const int64_t* ts_orig,
+                             int64_t* st_last) {
     const int64_t nseg = *d_nseg;
     for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nseg;
          j += gridDim.x * (int64_t)blockDim.x) {
@@ -854,6 +856,10 @@ __global__ void k_fire_count(const uint32_t* seg_start, const uint32_t* seg_slot
             return x < (uint64_t)P ? 0ull : (x - (uint64_t)P) / (uint64_t)S + 1ull;
         };
         nf[j] = (uint32_t)(F(head0 + ncomp) - F(head0));
+        if (st_last) {  // fused per-key last-arrival ts (EOS flush emit ts)
+            const int64_t t = ts_orig[idx_sorted[e - 1]];
+            if (t > st_last[slot]) st_last[slot] = t;
+        }
     }
 }
 
@@ -888,11 +894,12 @@ extern "C" void wfa_ffat_fire_offsets(wfa_stream_t s, const uint32_t* seg_start,
                                       int64_t n, int64_t pane_len, int64_t P,
                                       int64_t S, const uint32_t* st_fill,
                                       const uint32_t* st_head, uint32_t* nf,
-                                      int64_t* d_out_n) {
+                                      int64_t* d_out_n, const uint32_t* idx_sorted,
+                                      const int64_t* ts_orig, int64_t* st_last) {
     hipStream_t st = (hipStream_t)s;
     hipLaunchKernelGGL(k_fire_count, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
                        st, seg_start, seg_slot, d_nseg, n, pane_len, P, S, st_fill,
-                       st_head, nf);
+                       st_head, nf, idx_sorted, ts_orig, st_last);
     hipLaunchKernelGGL(k_fire_scan, dim3(1), dim3(1024), 0, st, nf, d_nseg, d_out_n);
 }
 

@@ -100,11 +100,15 @@ void wfa_segment_reduce_wave(
 
 // closed-form per-segment window-fire counts -> exclusive output offsets
 // (nf) + total (*d_out_n); run BEFORE the fold (reads pristine state)
+// st_last non-null: also max-folds each key's last-arrival ts (EOS flush
+// emit ts) — fused here to save a launch per batch
 void wfa_ffat_fire_offsets(wfa_stream_t s, const uint32_t* seg_start,
                            const uint32_t* seg_slot, const int64_t* d_nseg,
                            int64_t n, int64_t pane_len, int64_t P, int64_t S,
                            const uint32_t* st_fill, const uint32_t* st_head,
-                           uint32_t* nf, int64_t* d_out_n);
+                           uint32_t* nf, int64_t* d_out_n,
+                           const uint32_t* idx_sorted, const int64_t* ts_orig,
+                           int64_t* st_last);
 
 // ----- FFAT/pane sliding-window state machine (CB) -----
 // Batched multi-key redesign of the reference's per-key FlatFAT_GPU
