@@ -1060,3 +1060,60 @@ print("PEER_OK")
                        cwd=os.path.dirname(os.path.dirname(
                            os.path.abspath(__file__))))
     assert r.returncode == 0 and "PEER_OK" in r.stdout, r.stderr[-2000:]
+
+
+def test_gpu_split_merge_diamond():
+    """Device diamond (reference split_tests_gpu + merge_tests_gpu shapes):
+    source -> per-tuple split(2) -> branch maps (x+100 / x+200) -> merge ->
+    GPU sink; per-branch transforms verified through the merged sum."""
+    n, b = 100_000, 20_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, 31, b, vdt=0))
+           .withOutputSchema([0]).withOutputBatchSize(b).build())
+    g = wf.PipeGraph("gdia")
+    mp = g.add_source(src)
+    mp.split_gpu(2, expr="key & 1")
+    branches = []
+    for br in range(2):
+        bmp = mp.select(br)
+        bmp.add(Map_GPU_Builder(
+            native_gpu.gpu_affine_map(0, 1, 100 * (br + 1), dtype=0))
+            .withOutputSchema([0]).withOutputBatchSize(b).build())
+        branches.append(bmp)
+    merged = branches[0].merge(branches[1])
+    snk = wf.Sink_Builder(native.sum_sink(0)).withParallelism(1).build()
+    snk.out_schema = [0]
+    merged.add_sink(snk)
+    g.run()
+    ts, key, val = gen_batch(n, 0, 42, 31, 0)
+    exp = int(val[key % 2 == 0].sum() + 100 * (key % 2 == 0).sum()
+              + val[key % 2 == 1].sum() + 200 * (key % 2 == 1).sum())
+    assert g.sink_sum(snk) == exp
+    assert g.sink_count(snk) == n
+
+
+def test_gpu_broadcast_into_parallel_windows_shape():
+    """Broadcast device batches into 2 replicas of a keyed GPU window op:
+    each replica sees the full stream (D2D clones), so each fires the full
+    window set — 2x the single-replica count (reference
+    Parallel_Windows-on-GPU broadcast shape)."""
+    n, n_keys, b, win, slide = 200_000, 101, 50_000, 500, 100
+    counts = []
+    for par in (1, 2):
+        src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=5))
+               .withOutputSchema([5]).withOutputBatchSize(b).build())
+        ff = (Ffat_Windows_GPU_Builder(
+            native_gpu.gpu_ffat_windows(native_gpu.COMB_SUM, 0, win, slide,
+                                        max_keys=1024))
+            .withOutputSchema([2]).withOutputBatchSize(2 * b)
+            .withParallelism(par).build())
+        if par > 1:
+            ff.broadcast_input = True
+        snk = (Sink_GPU_Builder(native_gpu.gpu_count_sink())
+               .withParallelism(par).build())
+        g = wf.PipeGraph(f"gbw{par}")
+        mp = g.add_source(src)
+        mp.add(ff)
+        mp.add(snk)
+        g.run()
+        counts.append(g.sink_count(snk))
+    assert counts[1] == 2 * counts[0]
