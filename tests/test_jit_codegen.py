@@ -95,3 +95,12 @@ def test_split_source_compiles():
     _hipcc_compiles(_src_for(logic))
     logic2 = native_gpu.gpu_jit_split("v0 > 0.5f ? 1 : 0", ncols=2)
     _hipcc_compiles(_src_for(logic2))
+
+
+def test_geomean_window_source_compiles():
+    # device intrinsics (__logf/__expf) in user fold expressions
+    logic = native_gpu.gpu_jit_ffat_windows(
+        1000, 100, lift="__logf(v0);1.0f", comb="a0+b0;a1+b1",
+        finalize="(f1 > 0.0f) ? __expf(f0 / f1) : 0.0f",
+        identity=(0.0, 0.0), invertible=True)
+    _hipcc_compiles(_src_for(logic))
