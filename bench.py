@@ -148,8 +148,12 @@ def build_join_graph(n_tuples, batch, n_keys, par):
     sa = (wf.Source_Builder(native.seq_source(n_tuples, n_keys, batch))
           .withParallelism(1).withOutputSchema([0])
           .withOutputBatchSize(batch).build())
+    # B's offset must be a multiple of n_keys: key = v % n_keys, so a
+    # misaligned offset shifts every B key off A's and the join matches
+    # NOTHING (round-2 validity fix — the join was running empty)
+    off_b = (1_000_000_000 // n_keys) * n_keys
     sb = (wf.Source_Builder(native.seq_source(n_tuples, n_keys, batch,
-                                              value_offset=1_000_000_000))
+                                              value_offset=off_b))
           .withParallelism(1).withOutputSchema([0])
           .withOutputBatchSize(batch).build())
     mpA = g.add_source(sa)

@@ -434,9 +434,11 @@ def test_interval_join_user_function():
     import numpy as np
     n, n_keys, batch = 4000, 7, 128
 
+    OFF = 1_000_000_001  # divisible by n_keys=7: A and B keys stay aligned
+
     def joinfn(p):
         a = p['a'].astype(np.int64)
-        b = p['b'].astype(np.int64) - 1_000_000_000
+        b = p['b'].astype(np.int64) - OFF
         keep = ((a + b) % 2) == 0
         return keep, (a * 1000 + b).astype(np.float64)
 
@@ -445,7 +447,7 @@ def test_interval_join_user_function():
           .withParallelism(1).withOutputSchema([0])
           .withOutputBatchSize(batch).build())
     sb = (wf.Source_Builder(native.seq_source(n, n_keys, batch,
-                                              value_offset=1_000_000_000))
+                                              value_offset=OFF))
           .withParallelism(1).withOutputSchema([0])
           .withOutputBatchSize(batch).build())
     mpA = g.add_source(sa)
@@ -471,13 +473,14 @@ def test_interval_join_user_function():
     exp_s = 0.0
     exp_n = 0
     for d in (-1, 0, 1):                            # b.ts - a.ts in [-1, 1]
-        j = i + d                                   # B rows: ts=j, b=j+1e9
+        j = i + d                                   # B rows: ts=j, b=j+OFF
         ok = (j >= 1) & (j <= n)
         ia, jb = i[ok], j[ok]
-        m = (ia % n_keys) == ((jb + 1_000_000_000) % n_keys)
+        m = (ia % n_keys) == ((jb + OFF) % n_keys)
         ia, jb = ia[m], jb[m]
         keep = ((ia + jb) % 2) == 0
         exp_n += int(keep.sum())
         exp_s += float((ia[keep] * 1000 + jb[keep]).sum())
+    assert exp_n > 0, "vacuous oracle (no joinable pairs)"
     assert got['n'] == exp_n
     assert abs(got['s'] - exp_s) <= 1e-6 * max(1.0, exp_s)
