@@ -561,7 +561,10 @@ void Replica::run() {
             if (b->count > 0) stats.tuples_received += b->count;
             rctx.current_wm = collector->delivered_wm;
             rctx.current_tag = collector->delivered_tag;
-            if (b->count) rctx.current_ts = b->ts[b->count - 1];
+            // host batches only: device ts is not host memory, and lazy
+            // batches (count == -1) have no resolved row count yet
+            if (b->loc == Loc::HOST && b->count > 0)
+                rctx.current_ts = b->ts[b->count - 1];
             if (b->punct) {
                 int64_t wm = collector->delivered_wm;
                 release(b);

@@ -55,6 +55,9 @@ struct DeviceArena {
         void* p = nullptr;
         HIPCHK(hipSetDevice(device));
         HIPCHK(hipMalloc(&p, bytes));
+        if (getenv("WFA_DEBUG_SYNC"))
+            fprintf(stderr, "[alloc] %p..%p (%zu)\n", p, (char*)p + bytes,
+                    bytes);
         std::lock_guard<std::mutex> g(mu);
         allocated += bytes;
         return p;
@@ -77,6 +80,24 @@ inline bool wfa_prof() {
     static int v = -1;
     if (v < 0) v = getenv("WFA_PROF") ? 1 : 0;
     return v;
+}
+
+// WFA_DEBUG_SYNC=1: synchronize + error-check after every kernel group and
+// print a marker — turns an async "Memory access fault" abort into a
+// pinpointed stage (the poor man's compute-sanitizer for this pool)
+inline bool wfa_debug_sync() {
+    static int v = -1;
+    if (v < 0) v = getenv("WFA_DEBUG_SYNC") ? 1 : 0;
+    return v;
+}
+
+inline void dbg_sync(hipStream_t s, const char* where) {
+    if (!wfa_debug_sync()) return;
+    hipError_t e = hipStreamSynchronize(s);
+    hipError_t e2 = hipGetLastError();
+    fprintf(stderr, "[dbgsync] %s sync=%s last=%s\n", where,
+            hipGetErrorString(e), hipGetErrorString(e2));
+    fflush(stderr);
 }
 
 // ===== base for GPU logics =====

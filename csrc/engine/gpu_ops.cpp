@@ -154,6 +154,7 @@ struct GpuSourceLogic : GpuLogicBase {
             fprintf(stderr, "[prof] source get_dev wait %ld us\n", (long)tg);
         int64_t n = std::min<int64_t>(bsz, len - pos);
         wfa_gen_batch(stream, db->ts, db->key, db->cols[0], vdt, n, pos, seed, n_keys);
+        dbg_sync(stream, "gen");
         db->count = n;
         db->ts_mono = true;  // ts = start + i
         db->born_us = now_us();
@@ -306,6 +307,7 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
         }
         wfa_key_to_slot(s, db->key, n, tab, d_nslots, table_cap, slot,
                         slot_to_key);
+        dbg_sync(s, "key_to_slot");
         if (want_carry && vcol >= 0 && (int)db->schema.payload[vcol] == 2) {
             // carry the f32 value bits as a second sort payload: the fold
             // then reads values SEQUENTIALLY at the sorted position instead
@@ -346,7 +348,9 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
                 v_dt = 2;
             }
         }
+        dbg_sync(s, "sort");
         wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg, /*shr=*/0);
+        dbg_sync(s, "segments");
         if (ctx.stats) ctx.stats->num_kernels += 5 + 3 * ((bits + 3) / 4);
 }
 
@@ -552,6 +556,7 @@ struct GpuStatefulMapLogic : GpuLogicBase {
                            ks.idx_sorted, db->cols[col],
                            (int)db->schema.payload[col], spec, 0, a, b, d_state,
                            nullptr);
+        dbg_sync(stream, "stateful_apply");
         record_ready(db);
         if (ctx.stats) ctx.stats->num_kernels += 1;
         out.emit(db);

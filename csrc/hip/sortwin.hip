@@ -14,7 +14,26 @@
 //       per-key host loop is its main scaling flaw — SURVEY.md §7 step 6).
 #include <hip/hip_runtime.h>
 
+#include <cstdio>
+#include <cstdlib>
+
 #include "wfa_kernels.h"
+
+// WFA_DEBUG_SYNC: sync + error-print after each sort sub-kernel (fault
+// localization; see gpu_common.hpp dbg_sync)
+static int wfa_sort_dbg() {
+    static int v = -1;
+    if (v < 0) v = getenv("WFA_DEBUG_SYNC") ? 1 : 0;
+    return v;
+}
+static void sdbg(hipStream_t st, const char* what, int pass) {
+    if (!wfa_sort_dbg()) return;
+    hipError_t e = hipStreamSynchronize(st);
+    hipError_t e2 = hipGetLastError();
+    fprintf(stderr, "[sortdbg] p%d %s sync=%s last=%s\n", pass, what,
+            hipGetErrorString(e), hipGetErrorString(e2));
+    fflush(stderr);
+}
 
 #define WFA_THREADS 256
 #define WFA_MAX_BLOCKS 2048
@@ -582,8 +601,11 @@ extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
             int shift = base_shift + 8 * p;
             hipLaunchKernelGGL(k_rs8_hist, dim3(nb), dim3(WFA_THREADS), 0, st, ka, n,
                                shift, hist, nb);
+            sdbg(st, "hist", p);
             hipLaunchKernelGGL(k_rs8_scan, dim3(256), dim3(256), 0, st, hist, nb, dt);
+            sdbg(st, "scan", p);
             hipLaunchKernelGGL(k_rs8_dbase, dim3(1), dim3(256), 0, st, dt, dbase);
+            sdbg(st, "dbase", p);
             uint32_t* va_eff = (first && implicit_iota) ? nullptr : va;
             if (val2)
                 hipLaunchKernelGGL(k_rs8_scatter_lds<true>, dim3(nb),
@@ -593,6 +615,7 @@ extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                 hipLaunchKernelGGL(k_rs8_scatter_lds<false>, dim3(nb),
                                    dim3(WFA_THREADS), 0, st, ka, va_eff, n, shift,
                                    hist, nb, dbase, kb, vb, wa, wb);
+            sdbg(st, "scatter", p);
             first = false;
             uint32_t* t;
             t = ka; ka = kb; kb = t;
@@ -1212,6 +1235,40 @@ extern "C" void wfa_gram_fold(wfa_stream_t s, const uint32_t* seg_start,
 //               3 running_count (x=++state)
 // filter specs: 1 dedup_consecutive (keep if x != state; state=x)
 //               2 every_kth (keep when ++state % k == 0)
+// typed load/store for the stateful walk (round-2 fix: round 1 treated
+// every non-i64 dtype as f32 — a bf16 column got 4-byte accesses, twice
+// the buffer, silently corrupting neighboring device allocations)
+__device__ __forceinline__ double st_load(const void* col, int dt, uint32_t r) {
+    switch (dt) {
+        case 0: return (double)((const int64_t*)col)[r];
+        case 1: return ((const double*)col)[r];
+        case 2: return (double)((const float*)col)[r];
+        case 3: return (double)((const uint64_t*)col)[r];
+        case 4: return (double)((const int*)col)[r];
+        default: {  // bf16
+            union { uint32_t u; float f; } c;
+            c.u = (uint32_t)((const uint16_t*)col)[r] << 16;
+            return (double)c.f;
+        }
+    }
+}
+__device__ __forceinline__ void st_store(void* col, int dt, uint32_t r, double x) {
+    switch (dt) {
+        case 0: ((int64_t*)col)[r] = (int64_t)x; break;
+        case 1: ((double*)col)[r] = x; break;
+        case 2: ((float*)col)[r] = (float)x; break;
+        case 3: ((uint64_t*)col)[r] = (uint64_t)x; break;
+        case 4: ((int*)col)[r] = (int)x; break;
+        default: {  // bf16 round-to-nearest-even
+            union { uint32_t u; float f; } c;
+            c.f = (float)x;
+            uint32_t lsb = (c.u >> 16) & 1;
+            c.u += 0x7fff + lsb;
+            ((uint16_t*)col)[r] = (uint16_t)(c.u >> 16);
+        }
+    }
+}
+
 __global__ void k_stateful(const uint32_t* seg_start, const uint32_t* seg_slot,
                            const int64_t* d_nseg, int64_t n,
                            const uint32_t* idx_sorted, void* col, int dt,
@@ -1225,18 +1282,14 @@ __global__ void k_stateful(const uint32_t* seg_start, const uint32_t* seg_slot,
         double st = state[slot];
         for (int64_t i = seg_start[j]; i < e; ++i) {
             const uint32_t r = idx_sorted[i];
-            double x = (dt == 0) ? (double)((int64_t*)col)[r]
-                                 : (double)((float*)col)[r];
+            double x = st_load(col, dt, r);
             if (!is_filter) {
                 switch (spec) {
                     case 1: st += x; x = st; break;
                     case 2: st = a * st + (1.0 - a) * x; x = st; break;
                     case 3: st += 1.0; x = st; break;
                 }
-                if (dt == 0)
-                    ((int64_t*)col)[r] = (int64_t)x;
-                else
-                    ((float*)col)[r] = (float)x;
+                st_store(col, dt, r, x);
             } else {
                 uint32_t keep = 1;
                 switch (spec) {

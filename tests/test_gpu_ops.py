@@ -1117,3 +1117,40 @@ def test_gpu_broadcast_into_parallel_windows_shape():
         g.run()
         counts.append(g.sink_count(snk))
     assert counts[1] == 2 * counts[0]
+
+
+def test_gpu_stateful_map_bf16_column():
+    """Regression: keyed stateful ops on a bf16 column (round 2 found the
+    kernel doing 4-byte f32 accesses on the 2-byte column — OOB writes
+    corrupting neighboring device allocations).  EMA on bf16 vs oracle."""
+    n, n_keys, b = 100_000, 64, 25_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=5))
+           .withOutputSchema([5]).withOutputBatchSize(b).build())
+    sm = (Map_GPU_Builder(native_gpu.gpu_keyed_ema(0, alpha=0.5, max_keys=256))
+          .withOutputSchema([5]).withOutputBatchSize(b).build())
+    got = dict(s=0.0, n=0)
+
+    def pysink(cols):
+        got['s'] += float(bf16_to_f32_np(cols['c0']).astype(np.float64).sum())
+        got['n'] += len(cols['c0'])
+
+    g = wf.PipeGraph("bf16ema")
+    p = g.add_source(src)
+    p.chain(sm)
+    snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+    snk.out_schema = [5]
+    p.add_sink(snk)
+    g.run()
+    _, key, val = gen_batch(n, 0, 42, n_keys, 5)
+    from windflow_amd.synth import f32_to_bf16_np
+    vals = bf16_to_f32_np(val).astype(np.float64)
+    ema = {}
+    exp = 0.0
+    for k, v in zip(key.tolist(), vals.tolist()):
+        e = 0.5 * ema.get(k, 0.0) + 0.5 * v
+        ema[k] = e
+        # output column stores the bf16-rounded state
+        exp += float(bf16_to_f32_np(f32_to_bf16_np(
+            np.array([e], dtype=np.float32)))[0])
+    assert got['n'] == n
+    assert abs(got['s'] - exp) <= 5e-3 * max(1.0, abs(exp))
