@@ -294,8 +294,13 @@ struct KeyedScratch {
     int64_t max_keys = 0;
     int64_t cap = 0;
     int bits = 20;
+    // user-asserted dense integer keys (< max_keys): slot = key, the hash
+    // probe is skipped entirely (withDenseKeys)
+    bool dense = false;
+    uint32_t* d_overflow = nullptr;
 
     void alloc(int dev, int64_t cap_, int64_t mk, hipStream_t s);
+    void check_dense_overflow();
 
     uint32_t* idx_sorted = nullptr;  // valid after group()
     const void* v_as_f32 = nullptr;  // f32 or bf16, per v_dt

@@ -273,6 +273,21 @@ void KeyedScratch::alloc(int dev, int64_t cap_, int64_t mk, hipStream_t s) {
     v_f32 = (float*)A.get(4 * cap);
     wfa_fill_u64(s, tab, ~0ULL, 2 * table_cap);
     wfa_fill_u32(s, d_nslots, 0, 1);
+    if (dense) {
+        d_overflow = (uint32_t*)A.get(64);
+        wfa_fill_u32(s, d_overflow, 0, 1);
+        wfa_iota_u64(s, slot_to_key, mk);  // identity mapping
+    }
+}
+
+void KeyedScratch::check_dense_overflow() {
+    if (!dense) return;
+    uint32_t ovf = 0;
+    HIPCHK(hipMemcpy(&ovf, d_overflow, 4, hipMemcpyDeviceToHost));
+    if (ovf)
+        throw std::runtime_error(
+            "withDenseKeys: a key >= max_keys arrived — dense mode requires "
+            "integer keys in [0, max_keys); use the default hashed mode");
 }
 
 // sorts (slot, idx) pairs and fills segments; values stay unsorted and
@@ -292,8 +307,12 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
         bool vik = want_vik && vcol >= 0 && max_keys <= 65535 &&
                    (int)db->schema.payload[vcol] == 5;
         if (vik) {
-            wfa_key_to_slot_v(s, db->key, n, tab, d_nslots, table_cap, slot,
-                              slot_to_key, (const uint16_t*)db->cols[vcol]);
+            if (dense)
+                wfa_key_dense(s, db->key, n, max_keys, slot, d_nslots,
+                              d_overflow, (const uint16_t*)db->cols[vcol]);
+            else
+                wfa_key_to_slot_v(s, db->key, n, tab, d_nslots, table_cap, slot,
+                                  slot_to_key, (const uint16_t*)db->cols[vcol]);
             uint32_t *os, *oi;
             wfa_sort_pairs2(s, slot, idx, slot_t, idx_t, nullptr, nullptr, hist,
                             n, bits, &os, &oi, nullptr, /*implicit_iota=*/1,
@@ -305,8 +324,12 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
             if (ctx.stats) ctx.stats->num_kernels += 5 + 3 * ((bits + 3) / 4);
             return;
         }
-        wfa_key_to_slot(s, db->key, n, tab, d_nslots, table_cap, slot,
-                        slot_to_key);
+        if (dense)
+            wfa_key_dense(s, db->key, n, max_keys, slot, d_nslots, d_overflow,
+                          nullptr);
+        else
+            wfa_key_to_slot(s, db->key, n, tab, d_nslots, table_cap, slot,
+                            slot_to_key);
         dbg_sync(s, "key_to_slot");
         if (want_carry && vcol >= 0 && (int)db->schema.payload[vcol] == 2) {
             // carry the f32 value bits as a second sort payload: the fold
@@ -639,7 +662,9 @@ struct GpuReduceLogic : GpuLogicBase {
     float* d_oval = nullptr;
     int64_t* d_ots = nullptr;
     int64_t* d_on = nullptr;
-    GpuReduceLogic(int comb_, int vc, int64_t mk, int dev, Schema os, int64_t cap) {
+    GpuReduceLogic(int comb_, int vc, int64_t mk, int dev, Schema os,
+                   int64_t cap, bool dense = false) {
+        ks.dense = dense;
         comb = comb_; vcol = vc; max_keys = mk;
         device = dev;
         out_schema = os;  // payload [F32]
@@ -770,7 +795,9 @@ struct GpuFfatLogic : GpuLogicBase {
     }
 
     GpuFfatLogic(int comb_, int vc, int64_t w, int64_t sl, int64_t mk, bool tree,
-                 int dev, Schema os, int64_t cap, bool tb_, int64_t lat, int plog2) {
+                 int dev, Schema os, int64_t cap, bool tb_, int64_t lat, int plog2,
+                 bool dense = false) {
+        ks.dense = dense;
         comb = comb_; vcol = vc; win = w; slide = sl; max_keys = mk; use_tree = tree;
         tb = tb_; lateness = lat;
         if (plog2 > 0) pend_log2 = plog2;
@@ -932,6 +959,7 @@ struct GpuFfatLogic : GpuLogicBase {
         release_after_use(db);
         record_ready(ob);
         out.emit(ob);
+        if ((++batches & 63) == 0) ks.check_dense_overflow();
     }
     void destroy_graphs() {
         for (auto& [k, e] : graphs) (void)hipGraphExecDestroy(e);
@@ -955,7 +983,10 @@ struct GpuFfatLogic : GpuLogicBase {
         if (ctx.stats) ctx.stats->num_kernels += 4;
         record_ready(ob);
         out.emit(ob);
-        if ((++batches & 63) == 0) check_tb_flags();
+        if ((++batches & 63) == 0) {
+            check_tb_flags();
+            ks.check_dense_overflow();
+        }
     }
 
     bool on_punct(int64_t wm, EmitCtx& out, RuntimeCtx& ctx) override {
@@ -1593,16 +1624,17 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
         return std::make_shared<GpuReduceAllLogic>((int)ip[0], (int)ip[1], device,
                                                    os, out_batch);
     if (kind == "gpu_reduce")
-        // ip: [comb, vcol, max_keys]
+        // ip: [comb, vcol, max_keys, dense]
         return std::make_shared<GpuReduceLogic>((int)ip[0], (int)ip[1], ip[2], device,
-                                                os, out_batch);
+                                                os, out_batch,
+                                                ip.size() > 3 && ip[3] != 0);
     if (kind == "gpu_ffat") {
         // ip: [comb, vcol, win, slide, max_keys, use_tree,
         //      wintype(0 CB/1 TB), lateness, pend_ring_log2]
         auto l = std::make_shared<GpuFfatLogic>(
             (int)ip[0], (int)ip[1], ip[2], ip[3], ip[4], ip[5] != 0, device, os,
             out_batch, ip.size() > 6 && ip[6] != 0, ip.size() > 7 ? ip[7] : 0,
-            ip.size() > 8 ? (int)ip[8] : 0);
+            ip.size() > 8 ? (int)ip[8] : 0, ip.size() > 9 && ip[9] != 0);
         l->eng_ = eng;
         return l;
     }

@@ -177,6 +177,55 @@ extern "C" void wfa_key_to_slot_v(wfa_stream_t s, const uint64_t* key, int64_t n
                        slot_out, slot_to_key, val);
 }
 
+// Dense-key fast path (user-asserted integer keys < max_keys): slot = key,
+// no hash probe (the probe is ~25% of the flagship chain).  slot_to_key is
+// identity (prefilled); n_slots tracked as max(key)+1; keys out of range
+// raise the overflow flag (host checks and fails loudly).
+__global__ void k_key_dense(const uint64_t* key, int64_t n, int64_t max_keys,
+                            uint32_t* slot_out, uint32_t* n_slots,
+                            uint32_t* overflow, const uint16_t* val) {
+    __shared__ uint32_t blk_max;
+    if (threadIdx.x == 0) blk_max = 0;
+    __syncthreads();
+    uint32_t my_max = 0;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * (int64_t)blockDim.x) {
+        uint64_t k = key[i];
+        if (k >= (uint64_t)max_keys) {
+            atomicOr(overflow, 1u);
+            k = 0;
+        }
+        uint32_t sl = (uint32_t)k;
+        slot_out[i] = val ? ((sl << 16) | (uint32_t)val[i]) : sl;
+        if (sl + 1 > my_max) my_max = sl + 1;
+    }
+    for (int o = 32; o; o >>= 1)
+        my_max = max(my_max, (uint32_t)__shfl_down(my_max, o, 64));
+    if ((threadIdx.x & 63) == 0) atomicMax(&blk_max, my_max);
+    __syncthreads();
+    if (threadIdx.x == 0) atomicMax(n_slots, blk_max);
+}
+
+__global__ void k_iota_u64(uint64_t* p, int64_t n) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * (int64_t)blockDim.x)
+        p[i] = (uint64_t)i;
+}
+
+extern "C" void wfa_key_dense(wfa_stream_t s, const uint64_t* key, int64_t n,
+                              int64_t max_keys, uint32_t* slot_out,
+                              uint32_t* n_slots, uint32_t* overflow,
+                              const uint16_t* val) {
+    hipLaunchKernelGGL(k_key_dense, dim3(nblk(n)), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, key, n, max_keys, slot_out, n_slots,
+                       overflow, val);
+}
+
+extern "C" void wfa_iota_u64(wfa_stream_t s, uint64_t* p, int64_t n) {
+    hipLaunchKernelGGL(k_iota_u64, dim3(nblk(n)), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, p, n);
+}
+
 extern "C" void wfa_key_to_slot(wfa_stream_t s, const uint64_t* key, int64_t n,
                                 uint64_t* table_packed, uint32_t* n_slots,
                                 int64_t table_cap, uint32_t* slot_out,

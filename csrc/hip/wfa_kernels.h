@@ -48,6 +48,12 @@ void wfa_key_to_slot(wfa_stream_t s, const uint64_t* key, int64_t n,
                      uint64_t* table_packed, uint32_t* n_slots,
                      int64_t table_cap, uint32_t* slot_out,
                      uint64_t* slot_to_key);
+// dense-key fast path: slot = key (user-asserted key < max_keys); val
+// non-null packs VIK (slot<<16|bf16)
+void wfa_key_dense(wfa_stream_t s, const uint64_t* key, int64_t n,
+                   int64_t max_keys, uint32_t* slot_out, uint32_t* n_slots,
+                   uint32_t* overflow, const uint16_t* val);
+void wfa_iota_u64(wfa_stream_t s, uint64_t* p, int64_t n);
 
 // ----- stable LSD radix sort of (slot, iota idx) pairs, 4-bit digits -----
 // bits: how many low bits of slot to sort on. tmp arrays sized n (u32 each).

@@ -175,9 +175,13 @@ def gpu_keyed_every_kth(k=2, col=0, max_keys=1 << 16):
     return NativeLogic("gpu_filter_keyed", "", [float(k), 0.0], [2, col, max_keys])
 
 
-def gpu_keyed_reduce(comb=COMB_SUM, col=0, max_keys=1 << 16):
-    """per-batch keyed reduction -> one (key, agg, ts_max) per distinct key."""
-    return NativeLogic("gpu_reduce", "", [], [comb, col, max_keys])
+def gpu_keyed_reduce(comb=COMB_SUM, col=0, max_keys=1 << 16, dense_keys=False):
+    """per-batch keyed reduction -> one (key, agg, ts_max) per distinct key.
+    dense_keys: user-asserted integer keys in [0, max_keys) — slot = key,
+    skipping the hash probe (~25% of the keyed chain); out-of-range keys
+    fail loudly."""
+    return NativeLogic("gpu_reduce", "", [],
+                       [comb, col, max_keys, 1 if dense_keys else 0])
 
 
 def gpu_reduce_all(comb=COMB_SUM, col=0):
@@ -189,7 +193,7 @@ def gpu_reduce_all(comb=COMB_SUM, col=0):
 
 def gpu_ffat_windows(comb=COMB_SUM, col=0, win=1000, slide=100,
                      max_keys=1 << 16, use_tree=False, tb=False, lateness=0,
-                     pend_ring_log2=0):
+                     pend_ring_log2=0, dense_keys=False):
     """Keyed sliding window over panes (pane = gcd(win, slide)).
     CB (default): windows fire every `slide` tuples per key; use_tree
     selects the FlatFAT-arena path (O(log) combine for large win/slide
@@ -201,7 +205,8 @@ def gpu_ffat_windows(comb=COMB_SUM, col=0, win=1000, slide=100,
         raise ValueError("window length and slide must be >= 1")
     return NativeLogic("gpu_ffat", "", [],
                        [comb, col, win, slide, max_keys, 1 if use_tree else 0,
-                        1 if tb else 0, lateness, pend_ring_log2])
+                        1 if tb else 0, lateness, pend_ring_log2,
+                        1 if dense_keys else 0])
 
 
 def gpu_keyby_exchange():
