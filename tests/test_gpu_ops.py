@@ -1154,3 +1154,54 @@ def test_gpu_stateful_map_bf16_column():
             np.array([e], dtype=np.float32)))[0])
     assert got['n'] == n
     assert abs(got['s'] - exp) <= 5e-3 * max(1.0, abs(exp))
+
+
+def test_gpu_ffat_dense_keys_matches_hashed():
+    """withDenseKeys fast path (slot = key, no hash probe): identical
+    results to the hashed mode on the same stream (gpu_source keys are
+    already dense integers in [0, n_keys))."""
+    n, n_keys, win, slide, b = 200_000, 1024, 500, 100, 50_000
+    sums = []
+    for dense in (False, True):
+        src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=5))
+               .withOutputSchema([5]).withOutputBatchSize(b).build())
+        ff = (Ffat_Windows_GPU_Builder(
+            native_gpu.gpu_ffat_windows(native_gpu.COMB_SUM, 0, win, slide,
+                                        max_keys=n_keys, dense_keys=dense))
+              .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
+        acc = dict(s=0.0, n=0, k=0)
+
+        def pysink(cols):
+            acc['s'] += float(cols['c0'].astype(np.float64).sum())
+            acc['n'] += len(cols['c0'])
+            acc['k'] += int(cols['key'].astype(np.int64).sum())
+
+        g = wf.PipeGraph(f"dense{dense}")
+        p = g.add_source(src)
+        p.chain(ff)
+        snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+        snk.out_schema = [2]
+        p.add_sink(snk)
+        g.run()
+        sums.append((acc['n'], acc['k'], acc['s']))
+    assert sums[0][0] == sums[1][0]
+    assert sums[0][1] == sums[1][1]  # same keys emitted
+    assert abs(sums[0][2] - sums[1][2]) <= 1e-6 * max(1.0, abs(sums[0][2]))
+
+
+def test_gpu_dense_keys_overflow_raises():
+    """A key >= max_keys in dense mode must fail loudly, not corrupt."""
+    n, b = 200_000, 25_000
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, 1024, b, vdt=5))
+           .withOutputSchema([5]).withOutputBatchSize(b).build())
+    ff = (Ffat_Windows_GPU_Builder(
+        native_gpu.gpu_ffat_windows(native_gpu.COMB_SUM, 0, 500, 100,
+                                    max_keys=128, dense_keys=True))
+          .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
+    snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).build()
+    g = wf.PipeGraph("denseovf")
+    p = g.add_source(src)
+    p.chain(ff)
+    p.chain_sink(snk)
+    with pytest.raises(RuntimeError, match="withDenseKeys|dense"):
+        g.run()
