@@ -697,6 +697,10 @@ struct GpuReduceLogic : GpuLogicBase {
         record_ready(ob);
         out.emit(ob);
     }
+
+    void on_eos(EmitCtx&, RuntimeCtx&) override {
+        if (inited) ks.check_dense_overflow();
+    }
 };
 
 // ----- unkeyed full-batch reduce (reference reduce_gpu.hpp:269
@@ -998,6 +1002,7 @@ struct GpuFfatLogic : GpuLogicBase {
 
     void on_eos(EmitCtx& out, RuntimeCtx& ctx) override {
         ensure_init();
+        ks.check_dense_overflow();
         if (tb) {
             // complete every remaining data pane (wm -> +inf) ...
             tb_round_with_count(nullptr, 0, INT64_MAX / 4, out, ctx);
