@@ -431,7 +431,7 @@ void wfa_pack_meta(wfa_stream_t s, const uint32_t* counts, int world, int64_t wm
 }
 void wfa_count_u32(wfa_stream_t s, const uint32_t* v, int64_t n, uint32_t* counts,
                    int n_bins) {
-    hipMemsetAsync(counts, 0, 4 * n_bins, (hipStream_t)s);
+    (void)hipMemsetAsync(counts, 0, 4 * n_bins, (hipStream_t)s);
     hipLaunchKernelGGL(k_count_u32, dim3(nblk(n, 8)), dim3(WFA_THREADS), 0,
                        (hipStream_t)s, v, n, counts, n_bins);
 }
