@@ -51,7 +51,8 @@ def build_ffat_graph(n_tuples, batch, n_keys, win, slide, rank, world, device,
     return g, snk
 
 
-def build_a2a_graph(n_tuples, batch, n_keys, rank, world, device, dist_cfg):
+def build_a2a_graph(n_tuples, batch, n_keys, rank, world, device, dist_cfg,
+                    dense=True):
     """Driver config #4: Map_GPU -> RCCL keyby all-to-all -> Reduce_GPU,
     one rank per GPU."""
     import windflow_amd as wf
@@ -70,7 +71,8 @@ def build_a2a_graph(n_tuples, batch, n_keys, rank, world, device, dist_cfg):
           .withOutputSchema([2]).withOutputBatchSize(3 * batch)
           .withDevice(device).build())
     rd = (Reduce_GPU_Builder(
-        native_gpu.gpu_keyed_reduce(native_gpu.COMB_SUM, 0, 4 * n_keys))
+        native_gpu.gpu_keyed_reduce(native_gpu.COMB_SUM, 0, 4 * n_keys,
+                                    dense_keys=dense))
         .withOutputSchema([2]).withOutputBatchSize(3 * batch)
         .withDevice(device).build())
     snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).withDevice(device).build()
@@ -85,7 +87,7 @@ def build_a2a_graph(n_tuples, batch, n_keys, rank, world, device, dist_cfg):
 
 
 def build_ffat_x_graph(n_tuples, batch, n_keys, win, slide, rank, world, device,
-                       dist_cfg):
+                       dist_cfg, dense=True):
     """Keyed FFAT with a true cross-GPU keyby: source -> RCCL all-to-allv
     exchange -> FFAT window.  Every key's window state lives on exactly one
     rank (hash(key) % world)."""
@@ -104,7 +106,7 @@ def build_ffat_x_graph(n_tuples, batch, n_keys, win, slide, rank, world, device,
           .withDevice(device).build())
     ff = (Ffat_Windows_GPU_Builder(
         native_gpu.gpu_ffat_windows(native_gpu.COMB_SUM, 0, win, slide,
-                                    max_keys=2 * n_keys))
+                                    max_keys=2 * n_keys, dense_keys=dense))
         .withOutputSchema([2]).withOutputBatchSize(3 * batch)
         .withDevice(device).build())
     snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).withDevice(device).build()
@@ -284,11 +286,13 @@ def main():
                                              world, local_rank)
             if args.config == "a2a":
                 return build_a2a_graph(steps * B, B, args.keys, rank, world,
-                                       local_rank, dist_cfg)
+                                       local_rank, dist_cfg,
+                                       dense=not args.no_dense_keys)
             if args.config == "ffat_x":
                 return build_ffat_x_graph(steps * B, B, args.keys, args.win,
                                           args.slide, rank, world, local_rank,
-                                          dist_cfg)
+                                          dist_cfg,
+                                          dense=not args.no_dense_keys)
             return build_ffat_graph(steps * B, B, args.keys, args.win,
                                     args.slide, rank, world, local_rank,
                                     dense=not args.no_dense_keys)
@@ -348,7 +352,7 @@ def main():
                 "keys_per_rank": args.keys,
                 "parallelism": f"keyed-dp{max(world,1)}",
                 "p99_batch_latency_us": p99_us,
-                "dense_keys": (args.config == "ffat"
+                "dense_keys": (args.config in ("ffat", "ffat_x", "a2a")
                                and not args.no_dense_keys),
             },
         }
