@@ -1205,3 +1205,51 @@ def test_gpu_dense_keys_overflow_raises():
     p.chain_sink(snk)
     with pytest.raises(RuntimeError, match="withDenseKeys|dense"):
         g.run()
+
+
+def test_gpu_backpressure_tiny_queues():
+    """Un-chained GPU pipeline under 2-deep queues (WFA_QUEUE_CAP=2):
+    source -> map -> ffat -> sink in four threads with full backpressure.
+    Deadlock-freedom + exact counts (the regime where the round-2
+    standalone-op bugs lived)."""
+    import os
+    import subprocess
+    import sys
+    script = r"""
+import sys
+sys.path.insert(0, ".")
+import windflow_amd as wf
+from windflow_amd import native_gpu
+from windflow_amd.builders_gpu import (Source_GPU_Builder, Map_GPU_Builder,
+                                       Ffat_Windows_GPU_Builder,
+                                       Sink_GPU_Builder)
+n, n_keys, b = 400_000, 101, 10_000   # 40 batches through 2-deep queues
+src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=5))
+       .withOutputSchema([5]).withOutputBatchSize(b).build())
+mp_ = (Map_GPU_Builder(native_gpu.gpu_affine_map(0, 1.0, 0.0, dtype=5))
+       .withOutputSchema([5]).withOutputBatchSize(b).build())
+ff = (Ffat_Windows_GPU_Builder(
+    native_gpu.gpu_ffat_windows(native_gpu.COMB_COUNT, 0, 300, 100,
+                                max_keys=256))
+    .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
+snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).build()
+g = wf.PipeGraph("bp")
+p = g.add_source(src)
+p.add(mp_)   # add: separate threads + queues
+p.add(ff)
+p.add(snk)
+g.run()
+from windflow_amd.synth import gen_batch
+from collections import Counter
+ts, key, val = gen_batch(n, 0, 42, n_keys, 5)
+per = Counter(key.tolist())
+exp = sum((c - 1) // 100 + 1 for c in per.values())  # full + EOS partials
+assert g.sink_count(snk) == exp, (g.sink_count(snk), exp)
+print("BP_OK")
+"""
+    env = dict(os.environ, WFA_QUEUE_CAP="2")
+    r = subprocess.run([sys.executable, "-c", script], env=env,
+                       capture_output=True, text=True, timeout=300,
+                       cwd=os.path.dirname(os.path.dirname(
+                           os.path.abspath(__file__))))
+    assert r.returncode == 0 and "BP_OK" in r.stdout, r.stderr[-2000:]
