@@ -122,6 +122,11 @@ struct GpuLogicBase : OpLogic {
         inited = true;
     }
     ~GpuLogicBase() override {
+        for (auto& [hb, ev] : h2d_inflight) {  // host pools outlive logics
+            (void)hipEventSynchronize(ev);
+            release(hb);
+        }
+        h2d_inflight.clear();
         if (stream) (void)hipStreamDestroy(stream);
     }
 
@@ -182,8 +187,25 @@ struct GpuLogicBase : OpLogic {
     }
 
     // H2D staging: host batch -> fresh device batch (reference
-    // forward_emitter_gpu.hpp CPU->GPU path, redesigned: whole-SoA copies)
+    // forward_emitter_gpu.hpp CPU->GPU 1-batch overlap, redesigned:
+    // whole-SoA copies).  The host batch may be recycled by a CPU producer
+    // the moment we release it, so it is held until its copies' event
+    // fires — released OPPORTUNISTICALLY on later calls instead of a
+    // per-batch stream sync (round 1 synced every batch).
+    std::deque<std::pair<Batch*, hipEvent_t>> h2d_inflight;
+
+    void drain_h2d(bool block) {
+        while (!h2d_inflight.empty()) {
+            auto [old_hb, ev] = h2d_inflight.front();
+            if (!block && hipEventQuery(ev) != hipSuccess) break;
+            if (block) HIPCHK(hipEventSynchronize(ev));
+            release(old_hb);
+            h2d_inflight.pop_front();
+        }
+    }
+
     Batch* to_device(Batch* hb, RuntimeCtx& ctx) {
+        drain_h2d(false);
         Batch* db = get_dev();
         int64_t n = hb->count;
         HIPCHK(hipMemcpyAsync(db->ts, hb->ts, 8 * n, hipMemcpyHostToDevice, stream));
@@ -198,11 +220,13 @@ struct GpuLogicBase : OpLogic {
         db->count = n;
         db->watermark = ctx.current_wm;   // folded (hb may be a shared batch)
         db->stream_tag = ctx.current_tag;
+        db->ts_mono = hb->ts_mono;
         if (ctx.stats) ctx.stats->bytes_h2d += bytes;
-        // the host batch may be recycled by a CPU producer the moment we
-        // release it — the copies must have landed first
-        HIPCHK(hipStreamSynchronize(stream));
-        release(hb);
+        HIPCHK(hipEventRecord((hipEvent_t)db->ready_event, stream));
+        // db's event may be re-recorded later (recycling protocol) — that
+        // only delays the host batch's release, never un-orders it
+        h2d_inflight.push_back({hb, (hipEvent_t)db->ready_event});
+        if (h2d_inflight.size() > 4) drain_h2d(true);  // bounded holding
         return db;
     }
 
