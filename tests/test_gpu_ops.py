@@ -1253,3 +1253,64 @@ print("BP_OK")
                        cwd=os.path.dirname(os.path.dirname(
                            os.path.abspath(__file__))))
     assert r.returncode == 0 and "BP_OK" in r.stdout, r.stderr[-2000:]
+
+
+def test_gpu_h2d_overlap_recycling_exact():
+    """CPU source -> GPU map -> CPU sink under 2-deep queues: the round-2
+    H2D overlap holds each host batch on its copy event instead of a
+    per-batch stream sync (reference forward_emitter_gpu.hpp one-batch
+    overlap).  If a host batch were recycled and refilled before its
+    copies landed, the exact i64 sum below would be corrupted."""
+    import os
+    import subprocess
+    import sys
+    script = r"""
+import sys
+sys.path.insert(0, ".")
+import numpy as np
+import windflow_amd as wf
+from windflow_amd import native_gpu
+from windflow_amd.builders_gpu import Map_GPU_Builder
+
+n, b = 2_000_000, 8192
+state = dict(pos=0)
+
+def src(replica, par):
+    p = state['pos']
+    if p >= n:
+        return None
+    m = min(b, n - p)
+    state['pos'] += m
+    idx = np.arange(p, p + m, dtype=np.int64)
+    return dict(ts=idx, key=(idx % 257).astype(np.uint64),
+                c0=(idx * 3) % 1001, watermark=p + m)
+
+got = dict(s=0, n=0)
+
+def sink(cols):
+    got['s'] += int(cols['c0'].sum())
+    got['n'] += len(cols['c0'])
+
+g = wf.PipeGraph("h2d")
+p = g.add_source(wf.Source_Builder(src).withParallelism(1)
+                 .withOutputSchema([0]).withOutputBatchSize(b).build())
+mp_ = (Map_GPU_Builder(native_gpu.gpu_affine_map(0, 2, 7, dtype=0))
+       .withOutputSchema([0]).withOutputBatchSize(b).build())
+p.add(mp_)   # separate threads + 2-deep queues: fast host-batch recycling
+snk = wf.Sink_Builder(sink).withParallelism(1).build()
+snk.out_schema = [0]
+p.add_sink(snk)
+g.run()
+
+idx = np.arange(n, dtype=np.int64)
+exp = int(((idx * 3) % 1001 * 2 + 7).sum())
+assert got['n'] == n, (got['n'], n)
+assert got['s'] == exp, (got['s'], exp)
+print("H2D_OK")
+"""
+    env = dict(os.environ, WFA_QUEUE_CAP="2")
+    r = subprocess.run([sys.executable, "-c", script], env=env,
+                       capture_output=True, text=True, timeout=300,
+                       cwd=os.path.dirname(os.path.dirname(
+                           os.path.abspath(__file__))))
+    assert r.returncode == 0 and "H2D_OK" in r.stdout, r.stderr[-2000:]
