@@ -12,8 +12,10 @@
 // same compiled code the native catalog uses; only the value-dependent
 // kernels (lift/fold/advance/fire) are JIT-ed.
 //
-// Accumulator model: a window/reduce accumulator is a struct of NF f32
-// fields.  The user supplies, as ';'-separated C expressions:
+// Accumulator model: a window/reduce accumulator is a struct of NF fields
+// in the accumulator type `acct` (f32 default; "f64" spec flag switches
+// the fields, column loads and state arenas to double — output columns
+// stay F32).  The user supplies, as ';'-separated C expressions:
 //   lift:     NF exprs over v0..v{ncols-1} (column values as float),
 //             ts (i64), key (u64)          -> initial Acc of one tuple
 //   comb:     NF exprs over a0.., b0..     -> combine two Accs
@@ -45,6 +47,7 @@ struct JitFoldSpec {
     std::vector<std::string> lift, comb, fin;
     std::vector<double> ident;
     bool invertible = false;
+    bool acc64 = false;    // f64 accumulator fields (outputs stay F32 cols)
 };
 
 static std::vector<std::string> split_str(const std::string& s, char sep) {
@@ -66,9 +69,14 @@ static JitFoldSpec parse_fold_spec(const std::string& spec,
                                    const std::vector<int64_t>& ip) {
     JitFoldSpec fs;
     auto parts = split_str(spec, '\x1e');
-    if (parts.size() != 3)
+    if (parts.size() != 3 && parts.size() != 4)
         throw std::runtime_error(
-            "jit fold spec must be lift\\x1e comb\\x1e finalize");
+            "jit fold spec must be lift\\x1e comb\\x1e finalize[\\x1e flags]");
+    if (parts.size() == 4) {
+        if (parts[3] == "f64") fs.acc64 = true;
+        else if (!parts[3].empty() && parts[3] != "f32")
+            throw std::runtime_error("jit fold flags: '' | f32 | f64");
+    }
     fs.lift = split_str(parts[0], ';');
     fs.comb = split_str(parts[1], ';');
     fs.fin = split_str(parts[2], ';');
@@ -98,6 +106,16 @@ static std::string fmt_f32(double v) {
     return buf;
 }
 
+// identity literal in the accumulator type
+static std::string fmt_acc(double v, bool acc64) {
+    if (!acc64) return fmt_f32(v);
+    if (v == INFINITY) return "(1.0/0.0)";
+    if (v == -INFINITY) return "(-1.0/0.0)";
+    char buf[64];
+    snprintf(buf, sizeof buf, "%.17e", v);
+    return buf;
+}
+
 // ===== source generation =====
 // Common prelude: types, per-dtype loader, Acc struct + user functions.
 static std::string gen_prelude(const JitFoldSpec& fs) {
@@ -106,32 +124,35 @@ static std::string gen_prelude(const JitFoldSpec& fs) {
          "typedef unsigned int u32; typedef unsigned long long usz;\n";
     s += "#define NF " + std::to_string(fs.nf) + "\n";
     s += "#define NOUT " + std::to_string(fs.nout) + "\n";
-    s += "struct Acc { float f[NF]; };\n"
+    s += fs.acc64 ? "typedef double acct;\n" : "typedef float acct;\n";
+    s += "struct Acc { acct f[NF]; };\n"
          "__device__ __forceinline__ i64 jmin64(i64 a, i64 b) { return a < b ? a : b; }\n"
          "__device__ __forceinline__ i64 jmax64(i64 a, i64 b) { return a > b ? a : b; }\n";
-    // runtime-dtype column loader (dtype uniform per launch: scalar branch)
-    s += "__device__ __forceinline__ float jld(const void* p, int dt, i64 i) {\n"
+    // runtime-dtype column loader (dtype uniform per launch: scalar branch);
+    // loads in the ACCUMULATOR type so f64 folds keep i64/f64 column precision
+    s += "__device__ __forceinline__ acct jld(const void* p, int dt, i64 i) {\n"
          "    switch (dt) {\n"
-         "        case 0: return (float)((const i64*)p)[i];\n"
-         "        case 1: return (float)((const double*)p)[i];\n"
-         "        case 2: return ((const float*)p)[i];\n"
-         "        case 3: return (float)((const u64*)p)[i];\n"
-         "        case 4: return (float)((const int*)p)[i];\n"
+         "        case 0: return (acct)((const i64*)p)[i];\n"
+         "        case 1: return (acct)((const double*)p)[i];\n"
+         "        case 2: return (acct)((const float*)p)[i];\n"
+         "        case 3: return (acct)((const u64*)p)[i];\n"
+         "        case 4: return (acct)((const int*)p)[i];\n"
          "        default: {\n"
          "            union { u32 u; float f; } c;\n"
          "            c.u = (u32)((const unsigned short*)p)[i] << 16;\n"
-         "            return c.f;\n"
+         "            return (acct)c.f;\n"
          "        }\n"
          "    }\n"
          "}\n";
     s += "__device__ __forceinline__ Acc jident() { Acc r;\n";
     for (int f = 0; f < fs.nf; ++f)
-        s += "    r.f[" + std::to_string(f) + "] = " + fmt_f32(fs.ident[f]) + ";\n";
+        s += "    r.f[" + std::to_string(f) + "] = " +
+             fmt_acc(fs.ident[f], fs.acc64) + ";\n";
     s += "    return r; }\n";
     s += "__device__ __forceinline__ Acc jcomb(Acc A, Acc B) {\n";
     for (int f = 0; f < fs.nf; ++f) {
         auto fi = std::to_string(f);
-        s += "    float a" + fi + " = A.f[" + fi + "]; float b" + fi +
+        s += "    acct a" + fi + " = A.f[" + fi + "]; acct b" + fi +
              " = B.f[" + fi + "];\n";
     }
     s += "    Acc r;\n";
@@ -144,22 +165,22 @@ static std::string gen_prelude(const JitFoldSpec& fs) {
              "    for (int q = 0; q < NF; ++q) r.f[q] = A.f[q] - B.f[q];\n"
              "    return r; }\n";
     s += "__device__ __forceinline__ Acc jlift(";
-    for (int c = 0; c < fs.ncols; ++c) s += "float v" + std::to_string(c) + ", ";
+    for (int c = 0; c < fs.ncols; ++c) s += "acct v" + std::to_string(c) + ", ";
     s += "i64 ts, u64 key) {\n    (void)ts; (void)key;\n    Acc r;\n";
     for (int f = 0; f < fs.nf; ++f)
         s += "    r.f[" + std::to_string(f) + "] = (" + fs.lift[f] + ");\n";
     s += "    return r; }\n";
     s += "__device__ __forceinline__ void jfin(Acc A, float* o) {\n";
     for (int f = 0; f < fs.nf; ++f)
-        s += "    float f" + std::to_string(f) + " = A.f[" + std::to_string(f) +
+        s += "    acct f" + std::to_string(f) + " = A.f[" + std::to_string(f) +
              "]; (void)f" + std::to_string(f) + ";\n";
     for (int m = 0; m < fs.nout; ++m)
-        s += "    o[" + std::to_string(m) + "] = (" + fs.fin[m] + ");\n";
+        s += "    o[" + std::to_string(m) + "] = (float)(" + fs.fin[m] + ");\n";
     s += "}\n";
-    s += "__device__ __forceinline__ Acc acc_load(const float* p) { Acc r;\n"
+    s += "__device__ __forceinline__ Acc acc_load(const acct* p) { Acc r;\n"
          "#pragma unroll\n"
          "    for (int q = 0; q < NF; ++q) r.f[q] = p[q];\n    return r; }\n";
-    s += "__device__ __forceinline__ void acc_store(float* p, Acc a) {\n"
+    s += "__device__ __forceinline__ void acc_store(acct* p, Acc a) {\n"
          "#pragma unroll\n"
          "    for (int q = 0; q < NF; ++q) p[q] = a.f[q];\n}\n";
     // lane-uniform wave reduce (comb must be commutative+associative)
@@ -234,8 +255,8 @@ static std::string gen_cb_thread(const JitFoldSpec& fs) {
     s += KARGS_COLS;
     s += "    const u32* idx_sorted, const i64* ts_orig,\n"
          "    i64 pane_len, i64 P, i64 S, int ring_log2,\n"
-         "    float* st_acc, u32* st_fill, float* ring, u32* st_head,\n"
-         "    float* st_wsum, const u64* slot_to_key, const u32* fire_base,\n"
+         "    acct* st_acc, u32* st_fill, acct* ring, u32* st_head,\n"
+         "    acct* st_wsum, const u64* slot_to_key, const u32* fire_base,\n"
          "    u64* out_key, float* o0, float* o1, float* o2, float* o3,\n"
          "    i64* out_ts, i64 out_cap) {\n"
          "    const i64 nseg = *d_nseg;\n"
@@ -253,7 +274,7 @@ static std::string gen_cb_thread(const JitFoldSpec& fs) {
     if (inv)
         s += "        Acc wsum = acc_load(st_wsum + (usz)slot * NF);\n";
     s += "        i64 w = fire_base[j];\n"
-         "        float* rg = ring + (usz)slot * R * NF;\n"
+         "        acct* rg = ring + (usz)slot * R * NF;\n"
          "        for (; i < e; ++i) {\n"
          "            const u32 r = idx_sorted[i];\n"
          "            acc = jcomb(acc, JLOADS(r));\n"
@@ -297,8 +318,8 @@ static std::string gen_cb_wave(const JitFoldSpec& fs) {
     s += KARGS_COLS;
     s += "    const u32* idx_sorted, const i64* ts_orig,\n"
          "    i64 pane_len, i64 P, i64 S, int ring_log2,\n"
-         "    float* st_acc, u32* st_fill, float* ring, u32* st_head,\n"
-         "    float* st_wsum, const u64* slot_to_key, const u32* fire_base,\n"
+         "    acct* st_acc, u32* st_fill, acct* ring, u32* st_head,\n"
+         "    acct* st_wsum, const u64* slot_to_key, const u32* fire_base,\n"
          "    u64* out_key, float* o0, float* o1, float* o2, float* o3,\n"
          "    i64* out_ts, i64 out_cap) {\n"
          "    const i64 nseg = *d_nseg;\n"
@@ -319,7 +340,7 @@ static std::string gen_cb_wave(const JitFoldSpec& fs) {
     if (inv)
         s += "        Acc wsum = acc_load(st_wsum + (usz)slot * NF);\n";
     s += "        i64 w = fire_base[j];\n"
-         "        float* rg = ring + (usz)slot * R * NF;\n"
+         "        acct* rg = ring + (usz)slot * R * NF;\n"
          "        for (i64 pos = i0; pos < e; pos += 64) {\n"
          "            const u32 nchunk = (u32)jmin64((i64)64, e - pos);\n"
          "            Acc v;\n"
@@ -380,7 +401,7 @@ static std::string gen_cb_flush(const JitFoldSpec&) {
     return
         "extern \"C\" __global__ void jit_cb_flush(\n"
         "    const u32* n_slots, i64 P, i64 S, int ring_log2,\n"
-        "    const u32* st_fill, const float* st_acc, const float* ring,\n"
+        "    const u32* st_fill, const acct* st_acc, const acct* ring,\n"
         "    const u32* st_head, const i64* st_last, const u64* slot_to_key,\n"
         "    const u32* nf, i64 out_base, u64* out_key,\n"
         "    float* o0, float* o1, float* o2, float* o3, i64* out_ts,\n"
@@ -394,7 +415,7 @@ static std::string gen_cb_flush(const JitFoldSpec&) {
         "        const bool part = st_fill && st_fill[s];\n"
         "        const u64 H = head + (part ? 1 : 0);\n"
         "        u64 q0 = head < (u64)P ? 0 : (head - (u64)P) / (u64)S + 1;\n"
-        "        const float* rg = ring + (usz)s * R * NF;\n"
+        "        const acct* rg = ring + (usz)s * R * NF;\n"
         "        const u64 key = slot_to_key[s];\n"
         "        i64 w = nf[s];\n"
         "        for (u64 q = q0; q * (u64)S < H; ++q) {\n"
@@ -419,7 +440,7 @@ static std::string gen_tb_kernels(const JitFoldSpec& fs) {
          "    i64 n, ";
     s += KARGS_COLS;
     s += "    const u32* idx_sorted, const i64* ts_orig,\n"
-         "    i64 pane_len, i64 P, i64 S, int pend_log2, float* pend,\n"
+         "    i64 pane_len, i64 P, i64 S, int pend_log2, acct* pend,\n"
          "    i64* pend_base, i64* last_pane, const u64* slot_to_key,\n"
          "    u32* ignored, u32* overflow) {\n"
          "    const i64 nseg = *d_nseg;\n"
@@ -430,7 +451,7 @@ static std::string gen_tb_kernels(const JitFoldSpec& fs) {
          "        const u32 slot = seg_slot[j];\n"
          "        const u64 key = slot_to_key[slot];\n"
          "        const i64 e = (j + 1 < nseg) ? seg_start[j + 1] : n;\n"
-         "        float* pd = pend + (usz)slot * Rp * NF;\n"
+         "        acct* pd = pend + (usz)slot * Rp * NF;\n"
          "        i64 base = pend_base[slot];\n"
          "        i64 lastp = last_pane[slot];\n"
          "        u32 ign = 0;\n"
@@ -446,7 +467,7 @@ static std::string gen_tb_kernels(const JitFoldSpec& fs) {
          "            if (p < base) { ++ign; continue; }\n"
          "            if (p - base >= (i64)Rp) { atomicAdd(overflow, 1u); continue; }\n"
          "            Acc x = JLOADS(r);\n"
-         "            float* cell = pd + (usz)((u64)p & Pm) * NF;\n"
+         "            acct* cell = pd + (usz)((u64)p & Pm) * NF;\n"
          "            acc_store(cell, jcomb(acc_load(cell), x));\n"
          "            if (p > lastp) lastp = p;\n"
          "        }\n"
@@ -456,8 +477,8 @@ static std::string gen_tb_kernels(const JitFoldSpec& fs) {
          "    }\n}\n";
     s += "extern \"C\" __global__ void jit_tb_advance(\n"
          "    const u32* n_slots, i64 limit_pane, i64 pane_len, i64 P, i64 S,\n"
-         "    int ring_log2, int pend_log2, float* pend, i64* pend_base,\n"
-         "    const i64* last_pane, u32* st_head, float* st_wsum, float* ring,\n"
+         "    int ring_log2, int pend_log2, acct* pend, i64* pend_base,\n"
+         "    const i64* last_pane, u32* st_head, acct* st_wsum, acct* ring,\n"
          "    const u64* slot_to_key, const u32* nf, u64* out_key,\n"
          "    float* o0, float* o1, float* o2, float* o3, i64* out_ts,\n"
          "    i64 out_cap) {\n"
@@ -473,14 +494,14 @@ static std::string gen_tb_kernels(const JitFoldSpec& fs) {
          "        const i64 hi = jmin64(limit_pane, last_pane[s]);\n"
          "        if (hi < base) continue;\n"
          "        const u64 key = slot_to_key[s];\n"
-         "        float* pd = pend + (usz)s * Rp * NF;\n"
-         "        float* rg = ring + (usz)s * R * NF;\n"
+         "        acct* pd = pend + (usz)s * Rp * NF;\n"
+         "        acct* rg = ring + (usz)s * R * NF;\n"
          "        u32 head = st_head[s];\n";
     if (inv)
         s += "        Acc wsum = acc_load(st_wsum + (usz)s * NF);\n";
     s += "        i64 w = nf[s];\n"
          "        for (i64 q = base; q <= hi; ++q) {\n"
-         "            float* cell = pd + (usz)((u64)q & Pm) * NF;\n"
+         "            acct* cell = pd + (usz)((u64)q & Pm) * NF;\n"
          "            Acc pane = acc_load(cell);\n"
          "            acc_store(cell, jident());\n"
          "            acc_store(rg + (usz)(head & Rm) * NF, pane);\n";
@@ -514,6 +535,13 @@ static std::string gen_fold_source(const JitFoldSpec& fs, bool windows) {
         s += gen_reduce_kernel(fs);
         return s;
     }
+    // state-arena identity fill in the accumulator type (host fill helpers
+    // are f32-only)
+    s += "extern \"C\" __global__ void jit_fill_ident(acct* p, i64 ncells) {\n"
+         "    for (i64 i = blockIdx.x * (i64)blockDim.x + threadIdx.x;\n"
+         "         i < ncells; i += gridDim.x * (i64)blockDim.x)\n"
+         "        acc_store(p + (usz)i * NF, jident());\n"
+         "}\n";
     s += gen_cb_thread(fs);
     s += gen_cb_wave(fs);
     s += gen_cb_flush(fs);
@@ -695,7 +723,7 @@ struct GpuJitFfatLogic : GpuLogicBase {
     int ring_log2;
     KeyedScratch ks;
     hipFunction_t f_cb = nullptr, f_cb_wave = nullptr, f_flush = nullptr;
-    hipFunction_t f_tb_lift = nullptr, f_tb_adv = nullptr;
+    hipFunction_t f_tb_lift = nullptr, f_tb_adv = nullptr, f_fill = nullptr;
     // state arenas
     uint32_t* st_fill = nullptr;
     float* st_acc = nullptr;
@@ -742,15 +770,17 @@ struct GpuJitFfatLogic : GpuLogicBase {
         f_flush = jit_fn(mod, "jit_cb_flush");
         f_tb_lift = jit_fn(mod, "jit_tb_lift");
         f_tb_adv = jit_fn(mod, "jit_tb_advance");
+        f_fill = jit_fn(mod, "jit_fill_ident");
         ks.alloc(device, out_cap, max_keys, stream);
         auto& A = arena(device);
         int64_t R = 1ll << ring_log2;
         const int NF = fs.nf;
+        const int64_t esz = fs.acc64 ? 8 : 4;  // accumulator scalar bytes
         st_fill = (uint32_t*)A.get(4 * max_keys);
-        st_acc = (float*)A.get(4 * max_keys * NF);
-        ring = (float*)A.get(4 * max_keys * R * NF);
+        st_acc = (float*)A.get(esz * max_keys * NF);
+        ring = (float*)A.get(esz * max_keys * R * NF);
         st_head = (uint32_t*)A.get(4 * max_keys);
-        st_wsum = (float*)A.get(4 * max_keys * NF);
+        st_wsum = (float*)A.get(esz * max_keys * NF);
         st_last = (int64_t*)A.get(8 * max_keys);
         d_on = (int64_t*)A.get(64);
         nf = (uint32_t*)A.get(4 * (std::max(out_cap, max_keys) + 1));
@@ -762,7 +792,7 @@ struct GpuJitFfatLogic : GpuLogicBase {
         fill_ident(ring, max_keys * R);
         if (tb) {
             int64_t Rp = 1ll << pend_log2;
-            tb_pend = (float*)A.get(4 * max_keys * Rp * NF);
+            tb_pend = (float*)A.get(esz * max_keys * Rp * NF);
             tb_base = (int64_t*)A.get(8 * max_keys);
             tb_last_pane = (int64_t*)A.get(8 * max_keys);
             tb_flags = (uint32_t*)A.get(64);
@@ -778,23 +808,12 @@ struct GpuJitFfatLogic : GpuLogicBase {
         if (h_flags) (void)hipHostFree(h_flags);
     }
 
-    // cells are Acc-strided: fill each field with its identity value
+    // cells are Acc-strided: the generated jit_fill_ident stores jident()
+    // per cell in the accumulator type (f32 or f64)
     void fill_ident(float* p, int64_t n_cells) {
-        for (int f = 0; f < fs.nf; ++f) {
-            // strided fill: one fill per field value when fields share the
-            // identity, else per-field strided kernel would be needed; all
-            // cells are contiguous [cell][field], so fill field-by-field
-            // only when identities differ
-            bool uniform = true;
-            for (int g = 1; g < fs.nf; ++g)
-                if (fs.ident[g] != fs.ident[0]) uniform = false;
-            if (uniform) {
-                wfa_fill_f32(stream, p, (float)fs.ident[0], n_cells * fs.nf);
-                return;
-            }
-            wfa_fill_f32_strided(stream, p + f, (float)fs.ident[f], n_cells,
-                                 fs.nf);
-        }
+        ArgPack a;
+        a.add(p); a.add(n_cells);
+        launch(f_fill, stream, 512, a);
     }
 
     void check_tb_flags() {

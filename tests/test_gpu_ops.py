@@ -1314,3 +1314,64 @@ print("H2D_OK")
                        cwd=os.path.dirname(os.path.dirname(
                            os.path.abspath(__file__))))
     assert r.returncode == 0 and "H2D_OK" in r.stdout, r.stderr[-2000:]
+
+
+def test_gpu_jit_ffat_avg_f64_accumulator():
+    """acc="f64": double-precision accumulator fields/arenas.  Values are
+    2^24 + (i % 7) — an f32 accumulator loses the small addends against
+    the 2^24 offset, an f64 one keeps them exactly (output is still an F32
+    column, quantum 2 at this magnitude)."""
+    n, n_keys, b, win, slide = 120_000, 31, 15_000, 300, 100
+    state = dict(pos=0)
+
+    def src(replica, par):
+        p = state['pos']
+        if p >= n:
+            return None
+        m = min(b, n - p)
+        state['pos'] += m
+        idx = np.arange(p, p + m, dtype=np.int64)
+        return dict(ts=idx, key=(idx % n_keys).astype(np.uint64),
+                    c0=(2.0**24 + (idx % 7)).astype(np.float64),
+                    watermark=p + m)
+
+    ff = (Ffat_Windows_GPU_Builder(
+        native_gpu.gpu_avg_ffat_windows(win, slide, col=0, max_keys=64,
+                                        acc="f64"))
+          .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
+    res = dict(rows=[])
+
+    def pysink(cols):
+        res['rows'].append((cols['key'].copy(), cols['c0'].copy()))
+
+    g = wf.PipeGraph("jf64")
+    p = g.add_source(wf.Source_Builder(src).withParallelism(1)
+                     .withOutputSchema([1]).withOutputBatchSize(b).build())
+    p.add(ff)
+    snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+    snk.out_schema = [2]
+    p.add_sink(snk)
+    g.run()
+
+    from collections import defaultdict
+    idx = np.arange(n, dtype=np.int64)
+    key = idx % n_keys
+    val = (2.0**24 + (idx % 7)).astype(np.float64)
+    per = defaultdict(list)
+    for k, v in zip(key.tolist(), val.tolist()):
+        per[k].append(v)
+    exp = defaultdict(list)
+    for k, vs in per.items():
+        w = 0
+        while w * slide < len(vs):
+            exp[k].append(float(np.mean(vs[w * slide: w * slide + win])))
+            w += 1
+    got = defaultdict(list)
+    for k_arr, v_arr in res['rows']:
+        for k, v in zip(k_arr.tolist(), v_arr.tolist()):
+            got[k].append(v)
+    assert sum(map(len, got.values())) == sum(map(len, exp.values()))
+    for k in exp:
+        for a, bb in zip(sorted(got[k]), sorted(exp[k])):
+            # f64-exact value rounded once to the F32 output column
+            assert abs(a - bb) <= 1.5, (k, a, bb)

@@ -104,3 +104,24 @@ def test_geomean_window_source_compiles():
         finalize="(f1 > 0.0f) ? __expf(f0 / f1) : 0.0f",
         identity=(0.0, 0.0), invertible=True)
     _hipcc_compiles(_src_for(logic))
+
+
+def test_f64_accumulator_sources_compile():
+    # double-precision accumulators: reduce + CB + TB variants
+    r = native_gpu.gpu_jit_reduce(lift="v0", comb="a0+b0", finalize="f0",
+                                  acc="f64")
+    _hipcc_compiles(_src_for(r))
+    w = native_gpu.gpu_jit_ffat_windows(
+        1000, 100, lift="v0;1.0", comb="a0+b0;a1+b1",
+        finalize="(f1 > 0.0) ? (f0 / f1) : 0.0", identity=(0.0, 0.0),
+        invertible=True, acc="f64")
+    _hipcc_compiles(_src_for(w))
+    t = native_gpu.gpu_jit_ffat_windows(
+        1000, 100, comb="fmin(a0, b0)", identity=(float("inf"),),
+        tb=True, lateness=10, acc="f64")
+    _hipcc_compiles(_src_for(t))
+
+
+def test_f64_flag_validation():
+    with pytest.raises(ValueError):
+        native_gpu.gpu_jit_reduce(acc="f16")

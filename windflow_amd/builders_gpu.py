@@ -65,7 +65,7 @@ class Ffat_Windows_GPU_Builder(_GpuBuilder):
 
     def __init__(self, func=None, lift=None, comb=None, finalize=None,
                  identity=None, cols=(0,), invertible=False,
-                 max_keys=1 << 16):
+                 max_keys=1 << 16, acc="f32"):
         if func is None and comb is not None:
             from . import native_gpu
             lift = lift if lift is not None else "v0"
@@ -74,7 +74,7 @@ class Ffat_Windows_GPU_Builder(_GpuBuilder):
             func = native_gpu.gpu_jit_ffat_windows(
                 1, 1, lift=lift, comb=comb, finalize=finalize,
                 identity=identity, cols=cols, invertible=invertible,
-                max_keys=max_keys)
+                max_keys=max_keys, acc=acc)
         super().__init__(func)
         self._op.window = dict(type=0, win=0, slide=0, lateness=0)
 

@@ -53,12 +53,14 @@ def _expr_list(x):
     return list(x)
 
 
-def _jit_fold_spec(lift, comb, finalize, identity, cols):
+def _jit_fold_spec(lift, comb, finalize, identity, cols, acc="f32"):
     """Normalize a user fold into (spec, fparams, iparam-prefix).
 
     lift/comb/finalize: str (';'-joined C expressions) or list of str.
-    lift sees v0..v{len(cols)-1} (float), ts (i64), key (u64); comb sees
-    a0.., b0..; finalize sees f0..  identity: one float per field.
+    lift sees v0..v{len(cols)-1} (in the accumulator type), ts (i64),
+    key (u64); comb sees a0.., b0..; finalize sees f0..  identity: one
+    float per field.  acc="f64" keeps the accumulator fields (and column
+    loads) in double precision — output columns stay F32.
     Parity with the reference's arbitrary __device__ lift/combine functors
     (wf/builders_gpu.hpp:466-620, meta_gpu.hpp).  Combine must be
     associative AND commutative (folds are tree/wave shaped)."""
@@ -72,18 +74,21 @@ def _jit_fold_spec(lift, comb, finalize, identity, cols):
         raise ValueError("1..4 value columns")
     if not 1 <= nout <= 4:
         raise ValueError("1..4 outputs")
-    spec = "\x1e".join([";".join(lift), ";".join(comb), ";".join(fin)])
+    if acc not in ("f32", "f64"):
+        raise ValueError("acc must be 'f32' or 'f64'")
+    spec = "\x1e".join([";".join(lift), ";".join(comb), ";".join(fin), acc])
     cols4 = (cols + [0, 0, 0, 0])[:4]
     return spec, identity, [nf, nout, nc] + cols4
 
 
 def gpu_jit_reduce(lift="v0", comb="a0+b0", finalize="f0", identity=(0.0,),
-                   cols=(0,), max_keys=1 << 16):
+                   cols=(0,), max_keys=1 << 16, acc="f32"):
     """Per-batch keyed reduction with an ARBITRARY user fold, runtime-
     compiled with hiprtc (reference Reduce_GPU accepts any __device__
     combine, builders_gpu.hpp:350).  Emits one row per distinct key:
-    (key, finalize(acc) as F32 columns, ts_max)."""
-    spec, fp, pre = _jit_fold_spec(lift, comb, finalize, identity, cols)
+    (key, finalize(acc) as F32 columns, ts_max).  acc="f64" for double-
+    precision accumulation (large sums, i64/f64 columns)."""
+    spec, fp, pre = _jit_fold_spec(lift, comb, finalize, identity, cols, acc)
     return NativeLogic("gpu_jit_reduce", spec, fp, pre + [int(max_keys)])
 
 
@@ -97,7 +102,7 @@ def gpu_avg_reduce(col=0, max_keys=1 << 16):
 def gpu_jit_ffat_windows(win=1000, slide=100, lift="v0", comb="a0+b0",
                          finalize="f0", identity=(0.0,), cols=(0,),
                          max_keys=1 << 16, tb=False, lateness=0,
-                         pend_ring_log2=0, invertible=False):
+                         pend_ring_log2=0, invertible=False, acc="f32"):
     """Keyed sliding window with an ARBITRARY user lift+combine+finalize
     fold (reference Ffat_Windows_GPU arbitrary lift/comb functors,
     ffat_windows_gpu.hpp:60).  Multi-column values (cols), multi-column
@@ -106,7 +111,7 @@ def gpu_jit_ffat_windows(win=1000, slide=100, lift="v0", comb="a0+b0",
     instead of P-pane recombines — use for sum/avg-shaped folds)."""
     if win < 1 or slide < 1:
         raise ValueError("window length and slide must be >= 1")
-    spec, fp, pre = _jit_fold_spec(lift, comb, finalize, identity, cols)
+    spec, fp, pre = _jit_fold_spec(lift, comb, finalize, identity, cols, acc)
     ip = pre + [int(max_keys), int(win), int(slide), 1 if tb else 0,
                 int(lateness), int(pend_ring_log2), 1 if invertible else 0]
     return NativeLogic("gpu_jit_ffat", spec, fp, ip)
