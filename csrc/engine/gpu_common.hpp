@@ -159,13 +159,13 @@ struct GpuLogicBase : OpLogic {
     // never serialize against consumers still reading a recycled batch
     // (288 GB HBM makes a deep rotation cheap; reference recycling_gpu
     // instead spin-waits under memory pressure).
-    Batch* get_dev() {
+    Batch* get_from(Pool& pool) {
         constexpr int MAX_DEPTH = 8;
         Batch* chosen = nullptr;
         Batch* skipped[MAX_DEPTH];
         int nskip = 0;
         while (nskip < MAX_DEPTH) {
-            Batch* b = dev_pool->try_pop();
+            Batch* b = pool.try_pop();
             if (!b) break;
             if (!b->ready_event ||
                 hipEventQuery((hipEvent_t)b->ready_event) == hipSuccess) {
@@ -174,16 +174,38 @@ struct GpuLogicBase : OpLogic {
             }
             skipped[nskip++] = b;
         }
-        for (int i = 0; i < nskip; ++i) dev_pool->put(skipped[i]);
+        for (int i = 0; i < nskip; ++i) pool.put(skipped[i]);
         if (!chosen) {
-            if (nskip > 0 && dev_pool->live.load(std::memory_order_relaxed) >= MAX_DEPTH)
-                chosen = dev_pool->get();  // bounded: reuse, stream-waits below
+            if (nskip > 0 && pool.live.load(std::memory_order_relaxed) >= MAX_DEPTH)
+                chosen = pool.get();  // bounded: reuse, stream-waits below
             else
-                chosen = dev_pool->make_new();
+                chosen = pool.make_new();
         }
         if (chosen->ready_event)
             HIPCHK(hipStreamWaitEvent(stream, (hipEvent_t)chosen->ready_event, 0));
         return chosen;
+    }
+
+    Batch* get_dev() { return get_from(*dev_pool); }
+
+    // Input-side device pool: H2D/peer clones must carry the SOURCE batch's
+    // schema, not this logic's output schema — a logic whose in-dtype
+    // differs from its out-dtype (e.g. f64 column in, f32 results out)
+    // would otherwise mislabel (and under-size) the staged columns.
+    std::unique_ptr<Pool> in_pool;
+    std::vector<std::unique_ptr<Pool>> retired_in_pools;
+
+    Batch* get_in(const Batch* src) {
+        if (!in_pool || in_pool->schema.payload != src->schema.payload ||
+            in_pool->capacity < src->count) {
+            if (in_pool) retired_in_pools.push_back(std::move(in_pool));
+            in_pool = std::make_unique<Pool>(
+                src->schema, std::max({out_cap, src->capacity, src->count}),
+                false);
+            in_pool->loc = Loc::DEVICE;
+            in_pool->device = device;
+        }
+        return get_from(*in_pool);
     }
 
     // H2D staging: host batch -> fresh device batch (reference
@@ -206,7 +228,7 @@ struct GpuLogicBase : OpLogic {
 
     Batch* to_device(Batch* hb, RuntimeCtx& ctx) {
         drain_h2d(false);
-        Batch* db = get_dev();
+        Batch* db = get_in(hb);
         int64_t n = hb->count;
         HIPCHK(hipMemcpyAsync(db->ts, hb->ts, 8 * n, hipMemcpyHostToDevice, stream));
         HIPCHK(hipMemcpyAsync(db->key, hb->key, 8 * n, hipMemcpyHostToDevice, stream));
@@ -262,7 +284,7 @@ struct GpuLogicBase : OpLogic {
         // from the consumer device: wait on the producing event (legal
         // cross-device), then copy on OUR stream
         wait_ready(b);
-        Batch* db = get_dev();
+        Batch* db = get_in(b);
         const int64_t n = b->count;
         HIPCHK(hipMemcpyPeerAsync(db->ts, device, b->ts, b->device, 8 * n,
                                   stream));
