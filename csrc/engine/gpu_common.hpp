@@ -344,9 +344,13 @@ struct KeyedScratch {
     // probe is skipped entirely (withDenseKeys)
     bool dense = false;
     uint32_t* d_overflow = nullptr;
+    // bounded-slot segments table (allocated when max_keys <= 64 K):
+    // single-pass boundary scatter replaces the two-pass count+scan+scatter
+    uint32_t* by_slot = nullptr;
 
     void alloc(int dev, int64_t cap_, int64_t mk, hipStream_t s);
     void check_dense_overflow();
+    void segs(hipStream_t s, const uint32_t* slot_sorted, int64_t n, int shr);
 
     uint32_t* idx_sorted = nullptr;  // valid after group()
     const void* v_as_f32 = nullptr;  // f32 or bf16, per v_dt

@@ -278,6 +278,17 @@ void KeyedScratch::alloc(int dev, int64_t cap_, int64_t mk, hipStream_t s) {
         wfa_fill_u32(s, d_overflow, 0, 1);
         wfa_iota_u64(s, slot_to_key, mk);  // identity mapping
     }
+    // table_cap (>= 2*mk) entries: covers every legal slot id with margin
+    if (max_keys <= 65536) by_slot = (uint32_t*)A.get(4 * table_cap);
+}
+
+void KeyedScratch::segs(hipStream_t s, const uint32_t* slot_sorted, int64_t n,
+                        int shr) {
+    if (by_slot)
+        wfa_segments_dense(s, slot_sorted, n, by_slot, table_cap, seg_start,
+                           seg_slot, d_nseg, shr);
+    else
+        wfa_segments(s, slot_sorted, n, hist, seg_start, seg_slot, d_nseg, shr);
 }
 
 void KeyedScratch::check_dense_overflow() {
@@ -320,7 +331,7 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
             idx_sorted = oi;
             v_as_f32 = os;
             v_dt = 6;
-            wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg, /*shr=*/16);
+            segs(s, os, n, /*shr=*/16);
             if (ctx.stats) ctx.stats->num_kernels += 5 + 3 * ((bits + 3) / 4);
             return;
         }
@@ -345,7 +356,7 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
             idx_sorted = oi;
             v_as_f32 = ov;
             v_dt = 7;
-            wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg, 0);
+            segs(s, os, n, 0);
             if (ctx.stats) ctx.stats->num_kernels += 6 + 3 * ((bits + 3) / 4);
             return;
         }
@@ -372,7 +383,7 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
             }
         }
         dbg_sync(s, "sort");
-        wfa_segments(s, os, n, hist, seg_start, seg_slot, d_nseg, /*shr=*/0);
+        segs(s, os, n, /*shr=*/0);
         dbg_sync(s, "segments");
         if (ctx.stats) ctx.stats->num_kernels += 5 + 3 * ((bits + 3) / 4);
 }
@@ -1749,7 +1760,7 @@ std::vector<std::pair<std::string, double>> debug_a2a_stage_times(
         wfa_sort_pairs2(s, ks.slot, ks.idx, ks.slot_t, ks.idx_t,
                         (uint32_t*)ks.v_f32, (uint32_t*)ks.v_sorted, ks.hist, n,
                         ks.bits, &os2, &oi2, &ov2, 1, 0);
-        wfa_segments(s, os2, n, ks.hist, ks.seg_start, ks.seg_slot, ks.d_nseg, 0);
+        ks.segs(s, os2, n, 0);
         HIPCHK(hipEventRecord(ev[10], s));
         wfa_segment_reduce_wave(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                                 ov2, 7, oi2, r_ts, 0, /*ts_last=*/1,
@@ -1840,8 +1851,7 @@ std::vector<std::pair<std::string, double>> debug_ffat_stage_times(
         wfa_sort_pairs2(s, ks.slot, ks.idx, ks.slot_t, ks.idx_t, nullptr, nullptr,
                         ks.hist, n, ks.bits, &os_, &oi, nullptr, 1, vik ? 16 : 0);
         HIPCHK(hipEventRecord(ev[2], s));
-        wfa_segments(s, os_, n, ks.hist, ks.seg_start, ks.seg_slot, ks.d_nseg,
-                     vik ? 16 : 0);
+        ks.segs(s, os_, n, vik ? 16 : 0);
         HIPCHK(hipEventRecord(ev[3], s));
         wfa_ffat_fire_offsets(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n, pane, P,
                               S, st_fill, st_head, nf, d_on, nullptr, nullptr,

@@ -772,6 +772,65 @@ __global__ void k_nseg_total(const uint32_t* slot_sorted, int64_t n,
     }
 }
 
+// Bounded-slot fast path: every slot has at most ONE boundary in the
+// sorted sequence, so boundaries scatter directly into a per-slot table
+// (one 4 B/elem read of the sorted slots, no second pass, no grid scan),
+// then a single-workgroup ordered compaction over max_slots entries
+// builds the dense (seg_start, seg_slot) lists.  Used when max_slots is
+// small (<= 64 K: table fits L2); the general two-pass path below covers
+// unbounded slot counts (e.g. all-unique keys).
+__global__ void k_seg_by_slot(const uint32_t* slot, int64_t n, uint32_t* by_slot,
+                              int shr) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * (int64_t)blockDim.x) {
+        uint32_t sl = slot[i] >> shr;
+        if (i == 0 || (slot[i - 1] >> shr) != sl) by_slot[sl] = (uint32_t)i;
+    }
+}
+
+__global__ void k_seg_compact_dense(const uint32_t* by_slot, int64_t max_slots,
+                                    uint32_t* seg_start, uint32_t* seg_slot,
+                                    int64_t* d_nseg) {
+    __shared__ uint32_t tc[1024];
+    __shared__ uint32_t base;
+    if (threadIdx.x == 0) base = 0;
+    __syncthreads();
+    for (int64_t c = 0; c < max_slots; c += blockDim.x) {
+        int64_t s = c + threadIdx.x;
+        uint32_t v = (s < max_slots) ? by_slot[s] : 0xFFFFFFFFu;
+        uint32_t pred = (v != 0xFFFFFFFFu) ? 1u : 0u;
+        tc[threadIdx.x] = pred;
+        __syncthreads();
+        for (int off = 1; off < (int)blockDim.x; off <<= 1) {
+            uint32_t t = (threadIdx.x >= (unsigned)off) ? tc[threadIdx.x - off] : 0;
+            __syncthreads();
+            tc[threadIdx.x] += t;
+            __syncthreads();
+        }
+        if (pred) {
+            uint32_t w = base + tc[threadIdx.x] - 1;
+            seg_start[w] = v;
+            seg_slot[w] = (uint32_t)s;
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) base += tc[blockDim.x - 1];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) *d_nseg = (int64_t)base;
+}
+
+extern "C" void wfa_segments_dense(wfa_stream_t s, const uint32_t* slot_sorted,
+                                   int64_t n, uint32_t* by_slot, int64_t max_slots,
+                                   uint32_t* seg_start, uint32_t* seg_slot,
+                                   int64_t* d_nseg, int shr) {
+    hipStream_t st = (hipStream_t)s;
+    (void)hipMemsetAsync(by_slot, 0xFF, 4 * max_slots, st);
+    hipLaunchKernelGGL(k_seg_by_slot, dim3(nblk(n)), dim3(WFA_THREADS), 0, st,
+                       slot_sorted, n, by_slot, shr);
+    hipLaunchKernelGGL(k_seg_compact_dense, dim3(1), dim3(1024), 0, st, by_slot,
+                       max_slots, seg_start, seg_slot, d_nseg);
+}
+
 extern "C" void wfa_segments(wfa_stream_t s, const uint32_t* slot_sorted, int64_t n,
                              uint32_t* scan_tmp, uint32_t* seg_start,
                              uint32_t* seg_slot, int64_t* d_nseg, int shr) {
