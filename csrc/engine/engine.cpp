@@ -588,6 +588,7 @@ void Replica::run() {
         e->flush();
         e->eos();
     }
+    logic->post_eos();
     stats.end_us = now_us();
 }
 

@@ -301,6 +301,10 @@ struct OpLogic {
     virtual bool accepts_device() const { return false; }
     // Source: fill-and-emit loop; return false when exhausted.
     virtual bool source_step(EmitCtx& out, RuntimeCtx& ctx) { return false; }
+    // Called by the replica thread AFTER on_eos/flush, while the engine's
+    // pools are still alive — the place to return held batches (logic
+    // destructors run after Engine::pools is destroyed).
+    virtual void post_eos() {}
 };
 
 // Emission context handed to operator logic: wraps the replica's emitters.
@@ -358,6 +362,9 @@ struct ChainLogic : OpLogic {
     bool on_punct(int64_t wm, EmitCtx&, RuntimeCtx& ctx) override;
     void on_eos(EmitCtx&, RuntimeCtx& ctx) override;
     void warm(RuntimeCtx& ctx) override;
+    void post_eos() override {
+        for (auto& st : stages) st->post_eos();
+    }
 };
 
 struct EdgeSpec {

@@ -121,12 +121,11 @@ struct GpuLogicBase : OpLogic {
         init_device();
         inited = true;
     }
+    // held host batches go back at post_eos() — Engine::pools is destroyed
+    // BEFORE replica logics, so the destructor must not release into pools
+    void post_eos() override { drain_h2d(true); }
     ~GpuLogicBase() override {
-        for (auto& [hb, ev] : h2d_inflight) {  // host pools outlive logics
-            (void)hipEventSynchronize(ev);
-            release(hb);
-        }
-        h2d_inflight.clear();
+        h2d_inflight.clear();  // post_eos drained; abort paths just leak
         if (stream) (void)hipStreamDestroy(stream);
     }
 

@@ -791,29 +791,40 @@ __global__ void k_seg_by_slot(const uint32_t* slot, int64_t n, uint32_t* by_slot
 __global__ void k_seg_compact_dense(const uint32_t* by_slot, int64_t max_slots,
                                     uint32_t* seg_start, uint32_t* seg_slot,
                                     int64_t* d_nseg) {
-    __shared__ uint32_t tc[1024];
+    // one workgroup (1024 = 16 waves); ballot wave-scan + serial wave-total
+    // scan per chunk (the Hillis-Steele version cost more than the whole
+    // two-pass path it replaced)
+    __shared__ uint32_t wsum[17];
     __shared__ uint32_t base;
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
     if (threadIdx.x == 0) base = 0;
     __syncthreads();
     for (int64_t c = 0; c < max_slots; c += blockDim.x) {
         int64_t s = c + threadIdx.x;
         uint32_t v = (s < max_slots) ? by_slot[s] : 0xFFFFFFFFu;
-        uint32_t pred = (v != 0xFFFFFFFFu) ? 1u : 0u;
-        tc[threadIdx.x] = pred;
+        bool pred = v != 0xFFFFFFFFu;
+        uint64_t m = __ballot(pred);
+        uint32_t before = __popcll(m & ((1ULL << lane) - 1));
+        if (lane == 0) wsum[wid] = (uint32_t)__popcll(m);
         __syncthreads();
-        for (int off = 1; off < (int)blockDim.x; off <<= 1) {
-            uint32_t t = (threadIdx.x >= (unsigned)off) ? tc[threadIdx.x - off] : 0;
-            __syncthreads();
-            tc[threadIdx.x] += t;
-            __syncthreads();
+        if (threadIdx.x == 0) {
+            uint32_t acc = 0;
+            for (int w = 0; w < 16; ++w) {
+                uint32_t t = wsum[w];
+                wsum[w] = acc;
+                acc += t;
+            }
+            wsum[16] = acc;
         }
+        __syncthreads();
         if (pred) {
-            uint32_t w = base + tc[threadIdx.x] - 1;
+            uint32_t w = base + wsum[wid] + before;
             seg_start[w] = v;
             seg_slot[w] = (uint32_t)s;
         }
         __syncthreads();
-        if (threadIdx.x == 0) base += tc[blockDim.x - 1];
+        if (threadIdx.x == 0) base += wsum[16];
         __syncthreads();
     }
     if (threadIdx.x == 0) *d_nseg = (int64_t)base;
