@@ -709,6 +709,11 @@ struct GpuJitReduceLogic : GpuLogicBase {
         release_after_use(db);
         record_ready(ob);
         out.emit(ob);
+        if ((++batches_ & 63) == 0) ks.check_dense_overflow();
+    }
+    int64_t batches_ = 0;
+    void on_eos(EmitCtx&, RuntimeCtx&) override {
+        if (inited) ks.check_dense_overflow();
     }
 };
 
@@ -976,6 +981,7 @@ struct GpuJitFfatLogic : GpuLogicBase {
 
     void on_eos(EmitCtx& out, RuntimeCtx& ctx) override {
         ensure_init();
+        ks.check_dense_overflow();
         if (tb) {
             tb_round(nullptr, 0, INT64_MAX / 4, out, ctx);
             check_tb_flags();
@@ -1237,18 +1243,21 @@ std::shared_ptr<OpLogic> make_gpu_jit_logic(const std::string& kind,
             spec, ip.empty() ? 1 : (int)ip.at(0), device, os, out_batch);
     JitFoldSpec fs = parse_fold_spec(spec, fp, ip);
     if (kind == "gpu_jit_reduce") {
-        // ip: [nf, nout, ncols, col0..3, max_keys]
-        return std::make_shared<GpuJitReduceLogic>(std::move(fs), ip.at(7),
-                                                   device, os, out_batch);
+        // ip: [nf, nout, ncols, col0..3, max_keys, dense]
+        auto l = std::make_shared<GpuJitReduceLogic>(std::move(fs), ip.at(7),
+                                                     device, os, out_batch);
+        l->ks.dense = ip.size() > 8 && ip[8] != 0;
+        return l;
     }
     if (kind == "gpu_jit_ffat") {
         // ip: [nf, nout, ncols, col0..3, max_keys, win, slide, wintype,
-        //      lateness, pend_log2, invertible]
+        //      lateness, pend_log2, invertible, dense]
         fs.invertible = ip.size() > 13 && ip[13] != 0;
         auto l = std::make_shared<GpuJitFfatLogic>(
             std::move(fs), ip.at(8), ip.at(9), ip.at(7), device, os, out_batch,
             ip.size() > 10 && ip[10] != 0, ip.size() > 11 ? ip[11] : 0,
             ip.size() > 12 ? (int)ip[12] : 0);
+        l->ks.dense = ip.size() > 14 && ip[14] != 0;
         l->eng_ = eng;
         return l;
     }

@@ -106,3 +106,14 @@ class Ffat_Windows_GPU_Builder(_GpuBuilder):
     def withNumWinPerBatch(self, n):
         self._op.extra['num_win_per_batch'] = int(n)
         return self
+
+    def withDenseKeys(self):
+        """User-asserted integer keys in [0, max_keys): slot = key, the
+        hash probe is skipped (loud overflow check).  Native
+        catalog combines only (the JIT fold path takes dense via
+        KeyedScratch the same way once supported end-to-end)."""
+        if self._op.logic is None or self._op.logic.kind != "gpu_ffat":
+            raise ValueError("withDenseKeys: native gpu_ffat combines only "
+                             "(use dense_keys=True on the factory)")
+        self._op.logic.iparams[9] = 1
+        return self
