@@ -1375,3 +1375,46 @@ def test_gpu_jit_ffat_avg_f64_accumulator():
         for a, bb in zip(sorted(got[k]), sorted(exp[k])):
             # f64-exact value rounded once to the F32 output column
             assert abs(a - bb) <= 1.5, (k, a, bb)
+
+
+def test_gpu_jit_ffat_avg_dense_keys_matches_oracle():
+    """JIT window fold with dense_keys=True (slot = key, no hash probe)
+    must match the same oracle as the hashed path."""
+    n, n_keys, b, win, slide = 120_000, 101, 17_000, 400, 100
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=2))
+           .withOutputSchema([2]).withOutputBatchSize(b).build())
+    ff = (Ffat_Windows_GPU_Builder(
+        native_gpu.gpu_avg_ffat_windows(win, slide, col=0, max_keys=256,
+                                        dense_keys=True))
+          .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
+    res = dict(rows=[])
+
+    def pysink(cols):
+        res['rows'].append((cols['key'].copy(), cols['c0'].copy()))
+
+    g = wf.PipeGraph("jdense")
+    p = g.add_source(src)
+    p.chain(ff)
+    snk = wf.Sink_Builder(pysink).withParallelism(1).build()
+    snk.out_schema = [2]
+    p.add_sink(snk)
+    g.run()
+    ts, key, val = gen_batch(n, 0, 42, n_keys, 2)
+    from collections import defaultdict
+    per = defaultdict(list)
+    for k, v in zip(key.tolist(), val.astype(np.float64).tolist()):
+        per[k].append(v)
+    exp = defaultdict(list)
+    for k, vs in per.items():
+        w = 0
+        while w * slide < len(vs):
+            exp[k].append(float(np.mean(vs[w * slide: w * slide + win])))
+            w += 1
+    got = defaultdict(list)
+    for k_arr, v_arr in res['rows']:
+        for k, v in zip(k_arr.tolist(), v_arr.tolist()):
+            got[k].append(v)
+    assert sum(map(len, got.values())) == sum(map(len, exp.values()))
+    for k in exp:
+        for a, bb in zip(sorted(got[k]), sorted(exp[k])):
+            assert abs(a - bb) <= 2e-3 * max(1.0, abs(bb)), (k, a, bb)

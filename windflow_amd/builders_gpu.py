@@ -109,11 +109,14 @@ class Ffat_Windows_GPU_Builder(_GpuBuilder):
 
     def withDenseKeys(self):
         """User-asserted integer keys in [0, max_keys): slot = key, the
-        hash probe is skipped (loud overflow check).  Native
-        catalog combines only (the JIT fold path takes dense via
-        KeyedScratch the same way once supported end-to-end)."""
-        if self._op.logic is None or self._op.logic.kind != "gpu_ffat":
-            raise ValueError("withDenseKeys: native gpu_ffat combines only "
-                             "(use dense_keys=True on the factory)")
-        self._op.logic.iparams[9] = 1
+        hash probe is skipped (loud overflow check)."""
+        lg = self._op.logic
+        if lg is None or lg.kind not in ("gpu_ffat", "gpu_jit_ffat"):
+            raise ValueError("withDenseKeys: gpu_ffat / gpu_jit_ffat only")
+        if lg.kind == "gpu_ffat":
+            lg.iparams[9] = 1
+        else:
+            while len(lg.iparams) < 15:
+                lg.iparams.append(0)
+            lg.iparams[14] = 1
         return self

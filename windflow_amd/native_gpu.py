@@ -82,14 +82,15 @@ def _jit_fold_spec(lift, comb, finalize, identity, cols, acc="f32"):
 
 
 def gpu_jit_reduce(lift="v0", comb="a0+b0", finalize="f0", identity=(0.0,),
-                   cols=(0,), max_keys=1 << 16, acc="f32"):
+                   cols=(0,), max_keys=1 << 16, acc="f32", dense_keys=False):
     """Per-batch keyed reduction with an ARBITRARY user fold, runtime-
     compiled with hiprtc (reference Reduce_GPU accepts any __device__
     combine, builders_gpu.hpp:350).  Emits one row per distinct key:
     (key, finalize(acc) as F32 columns, ts_max).  acc="f64" for double-
     precision accumulation (large sums, i64/f64 columns)."""
     spec, fp, pre = _jit_fold_spec(lift, comb, finalize, identity, cols, acc)
-    return NativeLogic("gpu_jit_reduce", spec, fp, pre + [int(max_keys)])
+    return NativeLogic("gpu_jit_reduce", spec, fp,
+                       pre + [int(max_keys), 1 if dense_keys else 0])
 
 
 def gpu_avg_reduce(col=0, max_keys=1 << 16):
@@ -102,7 +103,8 @@ def gpu_avg_reduce(col=0, max_keys=1 << 16):
 def gpu_jit_ffat_windows(win=1000, slide=100, lift="v0", comb="a0+b0",
                          finalize="f0", identity=(0.0,), cols=(0,),
                          max_keys=1 << 16, tb=False, lateness=0,
-                         pend_ring_log2=0, invertible=False, acc="f32"):
+                         pend_ring_log2=0, invertible=False, acc="f32",
+                         dense_keys=False):
     """Keyed sliding window with an ARBITRARY user lift+combine+finalize
     fold (reference Ffat_Windows_GPU arbitrary lift/comb functors,
     ffat_windows_gpu.hpp:60).  Multi-column values (cols), multi-column
@@ -113,7 +115,8 @@ def gpu_jit_ffat_windows(win=1000, slide=100, lift="v0", comb="a0+b0",
         raise ValueError("window length and slide must be >= 1")
     spec, fp, pre = _jit_fold_spec(lift, comb, finalize, identity, cols, acc)
     ip = pre + [int(max_keys), int(win), int(slide), 1 if tb else 0,
-                int(lateness), int(pend_ring_log2), 1 if invertible else 0]
+                int(lateness), int(pend_ring_log2), 1 if invertible else 0,
+                1 if dense_keys else 0]
     return NativeLogic("gpu_jit_ffat", spec, fp, ip)
 
 
