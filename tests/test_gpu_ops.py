@@ -1418,3 +1418,49 @@ def test_gpu_jit_ffat_avg_dense_keys_matches_oracle():
     for k in exp:
         for a, bb in zip(sorted(got[k]), sorted(exp[k])):
             assert abs(a - bb) <= 2e-3 * max(1.0, abs(bb)), (k, a, bb)
+
+
+def test_gpu_ffat_single_key_giant_segment():
+    """n_keys=1: the whole batch is ONE segment (single boundary through
+    the dense segment-compaction path; longest possible per-segment fold
+    chain)."""
+    n, b, win, slide = 300_000, 60_000, 1000, 100
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, 1, b, vdt=5))
+           .withOutputSchema([5]).withOutputBatchSize(b).build())
+    ff = (Ffat_Windows_GPU_Builder(
+        native_gpu.gpu_ffat_windows(native_gpu.COMB_COUNT, 0, win, slide,
+                                    max_keys=16))
+          .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
+    g = wf.PipeGraph("onekey")
+    p = g.add_source(src)
+    p.chain(ff)
+    snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).build()
+    p.chain_sink(snk)
+    g.run()
+    # one key, n tuples: a window fires every `slide` tuples (+ EOS tails)
+    exp = (n - 1) // slide + 1
+    assert g.sink_count(snk) == exp, (g.sink_count(snk), exp)
+
+
+def test_gpu_ffat_at_64k_boundary():
+    """max_keys exactly 65536: the bounded-slot segments table still
+    engages (its boundary), VIK does not (needs <= 65535) — fallback
+    gather fold + dense segments together."""
+    n, n_keys, b, win, slide = 600_000, 60_000, 120_000, 40, 10
+    src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=5))
+           .withOutputSchema([5]).withOutputBatchSize(b).build())
+    ff = (Ffat_Windows_GPU_Builder(
+        native_gpu.gpu_ffat_windows(native_gpu.COMB_COUNT, 0, win, slide,
+                                    max_keys=65536))
+          .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
+    g = wf.PipeGraph("b64k")
+    p = g.add_source(src)
+    p.chain(ff)
+    snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).build()
+    p.chain_sink(snk)
+    g.run()
+    ts, key, val = gen_batch(n, 0, 42, n_keys, 5)
+    from collections import Counter
+    per = Counter(key.tolist())
+    exp = sum((c - 1) // slide + 1 for c in per.values())
+    assert g.sink_count(snk) == exp, (g.sink_count(snk), exp)
