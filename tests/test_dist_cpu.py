@@ -68,3 +68,30 @@ def test_cpu_keyby_exchange_world2():
     for p, o in zip(procs, outs):
         assert p.returncode == 0, o
     assert "DIST_CPU_OK" in outs[0], outs[0]
+
+
+@pytest.mark.timeout(180)
+def test_bench_driver_contract_world2_cpu():
+    """The driver launches bench.py via torch.distributed.run with one rank
+    per GPU; this runs the same launch shape on CPU (gloo, world=2,
+    --config cpu) and checks the contract: exactly ONE JSON line from
+    rank 0 with the required fields and whole-job aggregation."""
+    import json
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29573", "bench.py", "--config", "cpu",
+         "--gpus", "2", "--steps", "8", "--warmup", "2"],
+        cwd=root, env=dict(os.environ, GLOO_SOCKET_IFNAME="lo"),
+        capture_output=True, text=True, timeout=170)
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = [l for l in r.stdout.splitlines() if l.startswith('{"metric"')]
+    assert len(lines) == 1, r.stdout[-2000:]
+    d = json.loads(lines[0])
+    for k in ("metric", "value", "unit", "steps", "warmup", "ms_per_step",
+              "higher_is_better", "scaling", "vs_baseline", "dtype", "data",
+              "config"):
+        assert k in d, k
+    assert d["steps"] == 8 and d["warmup"] == 2
+    assert d["value"] > 0 and d["data"] == "synthetic"
