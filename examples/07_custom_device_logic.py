@@ -77,3 +77,23 @@ for br in range(2):
     sinks.append(s)
 g3.run()
 print(f"split: low={g3.sink_count(sinks[0])} high={g3.sink_count(sinks[1])}")
+
+# 5. round-2 extras on the same fold surface:
+#    - acc="f64": double-precision accumulator fields + column loads
+#      (large-magnitude sums; output columns stay F32)
+#    - withDenseKeys(): user-asserted integer keys < max_keys skip the
+#      hash probe (flagship: 18.1 -> 22 B tuples/s)
+g4 = wf.PipeGraph("f64_dense")
+src4 = (Source_GPU_Builder(native_gpu.gpu_source(N, KEYS, BATCH, vdt=2))
+        .withOutputSchema([2]).withOutputBatchSize(BATCH).build())
+mp4 = g4.add_source(src4)
+mp4.chain(Ffat_Windows_GPU_Builder(lift="v0;1.0", comb="a0+b0;a1+b1",
+                                   finalize="(f1 > 0.0) ? (f0 / f1) : 0.0",
+                                   identity=(0, 0), invertible=True,
+                                   max_keys=2 * KEYS, acc="f64")
+          .withCBWindows(1000, 100).withDenseKeys()
+          .withOutputSchema([2]).withOutputBatchSize(2 * BATCH).build())
+snk4 = Sink_GPU_Builder(native_gpu.gpu_count_sink()).build()
+mp4.chain_sink(snk4)
+g4.run()
+print(f"f64-accumulator dense-key AVG windows: {g4.sink_count(snk4)}")
