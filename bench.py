@@ -215,7 +215,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=20)
     # default micro-batch = measured throughput plateau (BASELINE.md sweep:
     # 16.2 B t/s @4M, 19.8 @8M, 19.8 @16M; p99 2.2 ms at 8M)
-    ap.add_argument("--batch", type=int, default=8_388_608)
+    ap.add_argument("--batch", type=int, default=16_777_216)
     ap.add_argument("--keys", type=int, default=8192,
                     help="distinct keys per rank")
     ap.add_argument("--par", type=int, default=1,
