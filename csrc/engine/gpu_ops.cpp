@@ -285,8 +285,8 @@ void KeyedScratch::alloc(int dev, int64_t cap_, int64_t mk, hipStream_t s) {
 void KeyedScratch::segs(hipStream_t s, const uint32_t* slot_sorted, int64_t n,
                         int shr) {
     if (by_slot)
-        wfa_segments_dense(s, slot_sorted, n, by_slot, table_cap, seg_start,
-                           seg_slot, d_nseg, shr);
+        wfa_segments_dense(s, slot_sorted, n, by_slot, table_cap, hist,
+                           seg_start, seg_slot, d_nseg, shr);
     else
         wfa_segments(s, slot_sorted, n, hist, seg_start, seg_slot, d_nseg, shr);
 }

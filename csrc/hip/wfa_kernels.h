@@ -84,11 +84,12 @@ void wfa_segments(wfa_stream_t s, const uint32_t* slot_sorted, int64_t n,
                   uint32_t* scan_tmp, uint32_t* seg_start, uint32_t* seg_slot,
                   int64_t* d_nseg, int shr);
 // bounded-slot fast path: one read + per-slot boundary table (by_slot,
-// max_slots u32 entries) + single-block ordered compaction
+// max_slots u32 entries) + multi-block ordered compaction (scratch32 >=
+// 32 u32, e.g. the sort hist)
 void wfa_segments_dense(wfa_stream_t s, const uint32_t* slot_sorted, int64_t n,
                         uint32_t* by_slot, int64_t max_slots,
-                        uint32_t* seg_start, uint32_t* seg_slot,
-                        int64_t* d_nseg, int shr);
+                        uint32_t* scratch32, uint32_t* seg_start,
+                        uint32_t* seg_slot, int64_t* d_nseg, int shr);
 
 // ----- per-batch keyed reduction (Reduce_GPU semantics) -----
 // comb: 0 sum 1 min 2 max 3 count ; vdt: dtype of v (F32 or I64 accum f64/i64)
