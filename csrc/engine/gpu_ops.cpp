@@ -257,6 +257,9 @@ void KeyedScratch::alloc(int dev, int64_t cap_, int64_t mk, hipStream_t s) {
     while (table_cap < 2 * mk) table_cap <<= 1;
     bits = 1;
     while ((1ll << bits) < mk + 1) ++bits;
+    // floor 5: the 4-bit sort path ignores base_shift (VIK packs the value
+    // in the low 16 bits) — always take the 8-bit machinery
+    if (bits < 5) bits = 5;
     auto& A = arena(dev);
     tab = (uint64_t*)A.get(16 * table_cap);
     d_nslots = (uint32_t*)A.get(64);
@@ -318,16 +321,22 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
         bool vik = want_vik && vcol >= 0 && max_keys <= 65535 &&
                    (int)db->schema.payload[vcol] == 5;
         if (vik) {
-            if (dense)
-                wfa_key_dense(s, db->key, n, max_keys, slot, d_nslots,
-                              d_overflow, (const uint16_t*)db->cols[vcol]);
-            else
+            uint32_t *os, *oi;
+            if (dense) {
+                // fused: slot write + sort pass-0 histogram in one kernel
+                wfa_key_dense_h(s, db->key, n, max_keys, slot, d_nslots,
+                                d_overflow, (const uint16_t*)db->cols[vcol],
+                                hist, /*shift=*/16);
+                wfa_sort_pairs2_ph(s, slot, idx, slot_t, idx_t, nullptr,
+                                   nullptr, hist, n, bits, &os, &oi, nullptr,
+                                   /*implicit_iota=*/1, /*base_shift=*/16);
+            } else {
                 wfa_key_to_slot_v(s, db->key, n, tab, d_nslots, table_cap, slot,
                                   slot_to_key, (const uint16_t*)db->cols[vcol]);
-            uint32_t *os, *oi;
-            wfa_sort_pairs2(s, slot, idx, slot_t, idx_t, nullptr, nullptr, hist,
-                            n, bits, &os, &oi, nullptr, /*implicit_iota=*/1,
-                            /*base_shift=*/16);
+                wfa_sort_pairs2(s, slot, idx, slot_t, idx_t, nullptr, nullptr,
+                                hist, n, bits, &os, &oi, nullptr,
+                                /*implicit_iota=*/1, /*base_shift=*/16);
+            }
             idx_sorted = oi;
             v_as_f32 = os;
             v_dt = 6;
@@ -336,8 +345,8 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
             return;
         }
         if (dense)
-            wfa_key_dense(s, db->key, n, max_keys, slot, d_nslots, d_overflow,
-                          nullptr);
+            wfa_key_dense_h(s, db->key, n, max_keys, slot, d_nslots, d_overflow,
+                            nullptr, hist, /*shift=*/0);
         else
             wfa_key_to_slot(s, db->key, n, tab, d_nslots, table_cap, slot,
                             slot_to_key);
@@ -350,9 +359,10 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
             // gather and keep the cheaper LDS scatter instead)
             wfa_cast(s, db->cols[vcol], 2, v_f32, 2, n);  // sort ping buffer
             uint32_t *os, *oi, *ov;
-            wfa_sort_pairs2(s, slot, idx, slot_t, idx_t, (uint32_t*)v_f32,
-                            (uint32_t*)v_sorted, hist, n, bits, &os, &oi, &ov,
-                            /*implicit_iota=*/1, /*base_shift=*/0);
+            (dense ? wfa_sort_pairs2_ph : wfa_sort_pairs2)(
+                s, slot, idx, slot_t, idx_t, (uint32_t*)v_f32,
+                (uint32_t*)v_sorted, hist, n, bits, &os, &oi, &ov,
+                /*implicit_iota=*/1, /*base_shift=*/0);
             idx_sorted = oi;
             v_as_f32 = ov;
             v_dt = 7;
@@ -361,9 +371,9 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
             return;
         }
         uint32_t *os, *oi;
-        wfa_sort_pairs2(s, slot, idx, slot_t, idx_t, nullptr, nullptr, hist, n,
-                        bits, &os, &oi, nullptr, /*implicit_iota=*/1,
-                        /*base_shift=*/0);
+        (dense ? wfa_sort_pairs2_ph : wfa_sort_pairs2)(
+            s, slot, idx, slot_t, idx_t, nullptr, nullptr, hist, n, bits, &os,
+            &oi, nullptr, /*implicit_iota=*/1, /*base_shift=*/0);
         idx_sorted = oi;
         if (vcol >= 0) {
             // A/B-measured: reading values THROUGH idx_sorted in the folds

@@ -206,6 +206,62 @@ __global__ void k_key_dense(const uint64_t* key, int64_t n, int64_t max_keys,
     if (threadIdx.x == 0) atomicMax(n_slots, blk_max);
 }
 
+// dense key->slot FUSED with the radix sort's pass-0 per-block histogram:
+// the sort's first count pass re-reads the whole slot array; since this
+// kernel just wrote it, accumulate the same LDS histogram here (sort
+// blocking: RS8_PER_BLOCK elements per block, hist[d*nblocks + b]) and
+// let the sort skip its pass-0 hist kernel (one 4n-byte read saved).
+__global__ void k_key_dense_h(const uint64_t* key, int64_t n, int64_t max_keys,
+                              uint32_t* slot_out, uint32_t* n_slots,
+                              uint32_t* overflow, const uint16_t* val,
+                              uint32_t* hist, int64_t nblocks, int shift) {
+    __shared__ uint32_t h[WFA_THREADS / 64][256];
+    __shared__ uint32_t blk_max;
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    if (threadIdx.x == 0) blk_max = 0;
+    for (int d = threadIdx.x; d < 256; d += blockDim.x)
+        for (int w = 0; w < WFA_THREADS / 64; ++w) h[w][d] = 0;
+    __syncthreads();
+    uint32_t my_max = 0;
+    int64_t waveBase = (int64_t)blockIdx.x * RS8_PER_BLOCK + wave * RS8_PER_WAVE;
+#pragma unroll
+    for (int j = 0; j < RS8_IPT; ++j) {
+        int64_t i = waveBase + j * 64 + lane;
+        if (i >= n) continue;
+        uint64_t k = key[i];
+        if (k >= (uint64_t)max_keys) {
+            atomicOr(overflow, 1u);
+            k = 0;
+        }
+        uint32_t sl = (uint32_t)k;
+        uint32_t packed = val ? ((sl << 16) | (uint32_t)val[i]) : sl;
+        slot_out[i] = packed;
+        atomicAdd(&h[wave][(packed >> shift) & 255], 1u);
+        if (sl + 1 > my_max) my_max = sl + 1;
+    }
+    for (int o = 32; o; o >>= 1)
+        my_max = max(my_max, (uint32_t)__shfl_down(my_max, o, 64));
+    if (lane == 0) atomicMax(&blk_max, my_max);
+    __syncthreads();
+    for (int d = threadIdx.x; d < 256; d += blockDim.x) {
+        uint32_t t = 0;
+        for (int w = 0; w < WFA_THREADS / 64; ++w) t += h[w][d];
+        hist[(int64_t)d * nblocks + blockIdx.x] = t;
+    }
+    if (threadIdx.x == 0) atomicMax(n_slots, blk_max);
+}
+
+extern "C" void wfa_key_dense_h(wfa_stream_t s, const uint64_t* key, int64_t n,
+                                int64_t max_keys, uint32_t* slot_out,
+                                uint32_t* n_slots, uint32_t* overflow,
+                                const uint16_t* val, uint32_t* hist,
+                                int shift) {
+    int64_t nb = (n + RS8_PER_BLOCK - 1) / RS8_PER_BLOCK;
+    hipLaunchKernelGGL(k_key_dense_h, dim3(nb), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, key, n, max_keys, slot_out, n_slots,
+                       overflow, val, hist, nb, shift);
+}
+
 __global__ void k_iota_u64(uint64_t* p, int64_t n) {
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += gridDim.x * (int64_t)blockDim.x)
@@ -626,13 +682,13 @@ __global__ void k_rs8_scatter_lds(const uint32_t* keys, const uint32_t* vals,
     }
 }
 
-extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
-                                uint32_t* slot_tmp, uint32_t* idx_tmp,
-                                uint32_t* val2, uint32_t* val2_tmp,
-                                uint32_t* hist, int64_t n, int bits,
-                                uint32_t** out_slot, uint32_t** out_idx,
-                                uint32_t** out_val2, int implicit_iota,
-                                int base_shift) {
+static void sort_pairs2_impl(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
+                             uint32_t* slot_tmp, uint32_t* idx_tmp,
+                             uint32_t* val2, uint32_t* val2_tmp,
+                             uint32_t* hist, int64_t n, int bits,
+                             uint32_t** out_slot, uint32_t** out_idx,
+                             uint32_t** out_val2, int implicit_iota,
+                             int base_shift, int skip_hist0) {
     hipStream_t st = (hipStream_t)s;
     uint32_t *ka = slot, *va = idx, *kb = slot_tmp, *vb = idx_tmp;
     uint32_t *wa = val2, *wb = val2_tmp;
@@ -648,8 +704,10 @@ extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
         int passes = (bits + 7) / 8;
         for (int p = 0; p < passes; ++p) {
             int shift = base_shift + 8 * p;
-            hipLaunchKernelGGL(k_rs8_hist, dim3(nb), dim3(WFA_THREADS), 0, st, ka, n,
-                               shift, hist, nb);
+            if (!(p == 0 && skip_hist0)) {
+                hipLaunchKernelGGL(k_rs8_hist, dim3(nb), dim3(WFA_THREADS), 0, st,
+                                   ka, n, shift, hist, nb);
+            }
             sdbg(st, "hist", p);
             hipLaunchKernelGGL(k_rs8_scan, dim3(256), dim3(256), 0, st, hist, nb, dt);
             sdbg(st, "scan", p);
@@ -697,6 +755,32 @@ extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
     *out_slot = ka;
     *out_idx = va;
     if (out_val2) *out_val2 = wa;
+}
+
+extern "C" void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
+                                uint32_t* slot_tmp, uint32_t* idx_tmp,
+                                uint32_t* val2, uint32_t* val2_tmp,
+                                uint32_t* hist, int64_t n, int bits,
+                                uint32_t** out_slot, uint32_t** out_idx,
+                                uint32_t** out_val2, int implicit_iota,
+                                int base_shift) {
+    sort_pairs2_impl(s, slot, idx, slot_tmp, idx_tmp, val2, val2_tmp, hist, n,
+                     bits, out_slot, out_idx, out_val2, implicit_iota,
+                     base_shift, 0);
+}
+
+// pass-0 histogram already produced by the slot-writing kernel
+// (k_key_dense_h) with the same blocking — skip the first count pass
+extern "C" void wfa_sort_pairs2_ph(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
+                                   uint32_t* slot_tmp, uint32_t* idx_tmp,
+                                   uint32_t* val2, uint32_t* val2_tmp,
+                                   uint32_t* hist, int64_t n, int bits,
+                                   uint32_t** out_slot, uint32_t** out_idx,
+                                   uint32_t** out_val2, int implicit_iota,
+                                   int base_shift) {
+    sort_pairs2_impl(s, slot, idx, slot_tmp, idx_tmp, val2, val2_tmp, hist, n,
+                     bits, out_slot, out_idx, out_val2, implicit_iota,
+                     base_shift, bits > 4 ? 1 : 0);
 }
 
 extern "C" void wfa_sort_pairs(wfa_stream_t s, uint32_t* slot, uint32_t* idx,

@@ -53,6 +53,12 @@ void wfa_key_to_slot(wfa_stream_t s, const uint64_t* key, int64_t n,
 void wfa_key_dense(wfa_stream_t s, const uint64_t* key, int64_t n,
                    int64_t max_keys, uint32_t* slot_out, uint32_t* n_slots,
                    uint32_t* overflow, const uint16_t* val);
+// fused variant: also writes the radix sort's pass-0 per-block histogram
+// (sort blocking) so wfa_sort_pairs2_ph can skip its first count pass
+void wfa_key_dense_h(wfa_stream_t s, const uint64_t* key, int64_t n,
+                     int64_t max_keys, uint32_t* slot_out, uint32_t* n_slots,
+                     uint32_t* overflow, const uint16_t* val, uint32_t* hist,
+                     int shift);
 void wfa_iota_u64(wfa_stream_t s, uint64_t* p, int64_t n);
 
 // ----- stable LSD radix sort of (slot, iota idx) pairs, 4-bit digits -----
@@ -70,6 +76,13 @@ void wfa_sort_pairs2(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
                      int64_t n, int bits, uint32_t** out_slot,
                      uint32_t** out_idx, uint32_t** out_val2,
                      int implicit_iota, int base_shift);
+// wfa_sort_pairs2 with the pass-0 histogram precomputed (wfa_key_dense_h)
+void wfa_sort_pairs2_ph(wfa_stream_t s, uint32_t* slot, uint32_t* idx,
+                        uint32_t* slot_tmp, uint32_t* idx_tmp,
+                        uint32_t* val2, uint32_t* val2_tmp, uint32_t* hist,
+                        int64_t n, int bits, uint32_t** out_slot,
+                        uint32_t** out_idx, uint32_t** out_val2,
+                        int implicit_iota, int base_shift);
 int64_t wfa_sort_nblocks(int64_t n);
 int64_t wfa_sort_hist_u32(int64_t cap);  // hist scratch size in u32
 
