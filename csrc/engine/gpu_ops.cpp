@@ -331,11 +331,12 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
                                    nullptr, hist, n, bits, &os, &oi, nullptr,
                                    /*implicit_iota=*/1, /*base_shift=*/16);
             } else {
-                wfa_key_to_slot_v(s, db->key, n, tab, d_nslots, table_cap, slot,
-                                  slot_to_key, (const uint16_t*)db->cols[vcol]);
-                wfa_sort_pairs2(s, slot, idx, slot_t, idx_t, nullptr, nullptr,
-                                hist, n, bits, &os, &oi, nullptr,
-                                /*implicit_iota=*/1, /*base_shift=*/16);
+                wfa_key_to_slot_h(s, db->key, n, tab, d_nslots, table_cap, slot,
+                                  slot_to_key, (const uint16_t*)db->cols[vcol],
+                                  hist, /*shift=*/16);
+                wfa_sort_pairs2_ph(s, slot, idx, slot_t, idx_t, nullptr,
+                                   nullptr, hist, n, bits, &os, &oi, nullptr,
+                                   /*implicit_iota=*/1, /*base_shift=*/16);
             }
             idx_sorted = oi;
             v_as_f32 = os;
@@ -348,8 +349,8 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
             wfa_key_dense_h(s, db->key, n, max_keys, slot, d_nslots, d_overflow,
                             nullptr, hist, /*shift=*/0);
         else
-            wfa_key_to_slot(s, db->key, n, tab, d_nslots, table_cap, slot,
-                            slot_to_key);
+            wfa_key_to_slot_h(s, db->key, n, tab, d_nslots, table_cap, slot,
+                              slot_to_key, nullptr, hist, /*shift=*/0);
         dbg_sync(s, "key_to_slot");
         if (want_carry && vcol >= 0 && (int)db->schema.payload[vcol] == 2) {
             // carry the f32 value bits as a second sort payload: the fold
@@ -359,10 +360,9 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
             // gather and keep the cheaper LDS scatter instead)
             wfa_cast(s, db->cols[vcol], 2, v_f32, 2, n);  // sort ping buffer
             uint32_t *os, *oi, *ov;
-            (dense ? wfa_sort_pairs2_ph : wfa_sort_pairs2)(
-                s, slot, idx, slot_t, idx_t, (uint32_t*)v_f32,
-                (uint32_t*)v_sorted, hist, n, bits, &os, &oi, &ov,
-                /*implicit_iota=*/1, /*base_shift=*/0);
+            wfa_sort_pairs2_ph(s, slot, idx, slot_t, idx_t, (uint32_t*)v_f32,
+                               (uint32_t*)v_sorted, hist, n, bits, &os, &oi,
+                               &ov, /*implicit_iota=*/1, /*base_shift=*/0);
             idx_sorted = oi;
             v_as_f32 = ov;
             v_dt = 7;
@@ -371,9 +371,9 @@ void KeyedScratch::group(hipStream_t s, Batch* db, int vcol, RuntimeCtx& ctx,
             return;
         }
         uint32_t *os, *oi;
-        (dense ? wfa_sort_pairs2_ph : wfa_sort_pairs2)(
-            s, slot, idx, slot_t, idx_t, nullptr, nullptr, hist, n, bits, &os,
-            &oi, nullptr, /*implicit_iota=*/1, /*base_shift=*/0);
+        wfa_sort_pairs2_ph(s, slot, idx, slot_t, idx_t, nullptr, nullptr,
+                           hist, n, bits, &os, &oi, nullptr,
+                           /*implicit_iota=*/1, /*base_shift=*/0);
         idx_sorted = oi;
         if (vcol >= 0) {
             // A/B-measured: reading values THROUGH idx_sorted in the folds

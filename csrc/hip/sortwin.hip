@@ -206,6 +206,78 @@ __global__ void k_key_dense(const uint64_t* key, int64_t n, int64_t max_keys,
     if (threadIdx.x == 0) atomicMax(n_slots, blk_max);
 }
 
+
+// hashed key->slot probe FUSED with the sort's pass-0 histogram (same
+// trick as k_key_dense_h; the probe is latency-bound, the LDS histogram
+// rides along for free and the sort skips one full read of slot_out)
+__global__ void k_key_to_slot_h(const uint64_t* key, int64_t n, uint64_t* tab,
+                                uint32_t* n_slots, int64_t cap,
+                                uint32_t* slot_out, uint64_t* slot_to_key,
+                                const uint16_t* val, uint32_t* hist,
+                                int64_t nblocks, int shift) {
+    __shared__ uint32_t h[WFA_THREADS / 64][256];
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    for (int d = threadIdx.x; d < 256; d += blockDim.x)
+        for (int w = 0; w < WFA_THREADS / 64; ++w) h[w][d] = 0;
+    __syncthreads();
+    const uint64_t mask = (uint64_t)cap - 1;
+    int64_t waveBase = (int64_t)blockIdx.x * RS8_PER_BLOCK + wave * RS8_PER_WAVE;
+    for (int j = 0; j < RS8_IPT; ++j) {
+        int64_t i = waveBase + j * 64 + lane;
+        if (i >= n) continue;
+        uint64_t k = key[i];
+        uint64_t p = mix64s(k) & mask;
+        for (;;) {
+            uint64_t cur = __hip_atomic_load(&tab[2 * p], __ATOMIC_RELAXED,
+                                             __HIP_MEMORY_SCOPE_AGENT);
+            if (cur == k) break;
+            if (cur == WFA_EMPTY_KEY) {
+                uint64_t expected = WFA_EMPTY_KEY;
+                bool won = __hip_atomic_compare_exchange_strong(
+                    &tab[2 * p], &expected, k, __ATOMIC_RELAXED,
+                    __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (won) {
+                    uint32_t slot = atomicAdd(n_slots, 1u);
+                    slot_to_key[slot] = k;
+                    __hip_atomic_store(&tab[2 * p + 1], (uint64_t)slot,
+                                       __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+                    break;
+                }
+                if (expected == k) break;
+            }
+            p = (p + 1) & mask;
+        }
+        uint64_t sl;
+        do {
+            sl = __hip_atomic_load(&tab[2 * p + 1], __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+        } while (sl == ~0ULL);
+        uint32_t packed = val ? (((uint32_t)sl << 16) | (uint32_t)val[i])
+                              : (uint32_t)sl;
+        slot_out[i] = packed;
+        atomicAdd(&h[wave][(packed >> shift) & 255], 1u);
+    }
+    __syncthreads();
+    for (int d = threadIdx.x; d < 256; d += blockDim.x) {
+        uint32_t t = 0;
+        for (int w = 0; w < WFA_THREADS / 64; ++w) t += h[w][d];
+        hist[(int64_t)d * nblocks + blockIdx.x] = t;
+    }
+}
+
+extern "C" void wfa_key_to_slot_h(wfa_stream_t s, const uint64_t* key,
+                                  int64_t n, uint64_t* table_packed,
+                                  uint32_t* n_slots, int64_t table_cap,
+                                  uint32_t* slot_out, uint64_t* slot_to_key,
+                                  const uint16_t* val, uint32_t* hist,
+                                  int shift) {
+    int64_t nb = (n + RS8_PER_BLOCK - 1) / RS8_PER_BLOCK;
+    hipLaunchKernelGGL(k_key_to_slot_h, dim3(nb), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, key, n, table_packed, n_slots,
+                       table_cap, slot_out, slot_to_key, val, hist, nb, shift);
+}
+
 // dense key->slot FUSED with the radix sort's pass-0 per-block histogram:
 // the sort's first count pass re-reads the whole slot array; since this
 // kernel just wrote it, accumulate the same LDS histogram here (sort

@@ -53,8 +53,13 @@ void wfa_key_to_slot(wfa_stream_t s, const uint64_t* key, int64_t n,
 void wfa_key_dense(wfa_stream_t s, const uint64_t* key, int64_t n,
                    int64_t max_keys, uint32_t* slot_out, uint32_t* n_slots,
                    uint32_t* overflow, const uint16_t* val);
-// fused variant: also writes the radix sort's pass-0 per-block histogram
+// fused variants: also write the radix sort's pass-0 per-block histogram
 // (sort blocking) so wfa_sort_pairs2_ph can skip its first count pass
+void wfa_key_to_slot_h(wfa_stream_t s, const uint64_t* key, int64_t n,
+                       uint64_t* table_packed, uint32_t* n_slots,
+                       int64_t table_cap, uint32_t* slot_out,
+                       uint64_t* slot_to_key, const uint16_t* val,
+                       uint32_t* hist, int shift);
 void wfa_key_dense_h(wfa_stream_t s, const uint64_t* key, int64_t n,
                      int64_t max_keys, uint32_t* slot_out, uint32_t* n_slots,
                      uint32_t* overflow, const uint16_t* val, uint32_t* hist,
