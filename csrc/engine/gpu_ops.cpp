@@ -1887,6 +1887,85 @@ std::vector<std::pair<std::string, double>> debug_ffat_stage_times(
     return out;
 }
 
+// MFMA Gram-window chain harness: per-stage hipEvent times for the
+// matrix-core combine (k_gram).  Also reports derived MFMA throughput:
+// each tuple contributes one 16-vector to C += v*v^T (512 FLOP/tuple,
+// one v_mfma_f32_16x16x4_f32 per 4 tuples per key segment).
+std::vector<std::pair<std::string, double>> debug_gram_stage_times(
+    int64_t n, int64_t n_keys, int64_t win, int iters) {
+    HIPCHK(hipSetDevice(0));
+    hipStream_t s = nullptr;
+    KeyedScratch ks;
+    ks.alloc(0, n, n_keys, s);
+    auto& A = arena(0);
+    uint64_t* d_key = (uint64_t*)A.get(8 * n);
+    int64_t* d_ts = (int64_t*)A.get(8 * n);
+    float* d_cols[16];
+    for (int c = 0; c < 16; ++c) {
+        d_cols[c] = (float*)A.get(4 * n);
+        wfa_gen_batch(s, d_ts, d_key, d_cols[c], 2, n, 0, 1000 + c, n_keys);
+    }
+    // keys/ts from the last gen call are fine (random keys, ts = i)
+    int64_t fires16 = (n / std::max<int64_t>(win, 1) + n_keys + 64) * 16;
+    uint32_t* st_fill = (uint32_t*)A.get(4 * n_keys);
+    uint32_t* st_head = (uint32_t*)A.get(4 * n_keys);
+    float* st_acc = (float*)A.get(4 * n_keys * 256);
+    uint32_t* nf = (uint32_t*)A.get(4 * (fires16 + 1));
+    uint64_t* o_key = (uint64_t*)A.get(8 * fires16);
+    int64_t* o_gwid = (int64_t*)A.get(8 * fires16);
+    int64_t* o_ts = (int64_t*)A.get(8 * fires16);
+    float* o_cols[16];
+    for (int c = 0; c < 16; ++c) o_cols[c] = (float*)A.get(4 * fires16);
+    int64_t* d_on = (int64_t*)A.get(64);
+    const float** d_in = (const float**)A.get(16 * 8);
+    float** d_out = (float**)A.get(16 * 8);
+    HIPCHK(hipMemcpyAsync(d_in, d_cols, 16 * 8, hipMemcpyHostToDevice, s));
+    HIPCHK(hipMemcpyAsync(d_out, o_cols, 16 * 8, hipMemcpyHostToDevice, s));
+    HIPCHK(hipMemsetAsync(st_fill, 0, 4 * n_keys, s));
+    HIPCHK(hipMemsetAsync(st_head, 0, 4 * n_keys, s));
+    HIPCHK(hipMemsetAsync(st_acc, 0, 4 * n_keys * 256, s));
+
+    constexpr int NS = 4;
+    const char* names[NS] = {"group", "fire_offsets", "mfma_fold", "gflops"};
+    hipEvent_t ev[NS];
+    for (auto& e : ev) HIPCHK(hipEventCreate(&e));
+    double acc[NS] = {0};
+    RuntimeCtx rctx;
+    for (int it = -2; it < iters; ++it) {
+        HIPCHK(hipEventRecord(ev[0], s));
+        wfa_key_to_slot(s, d_key, n, ks.tab, ks.d_nslots, ks.table_cap,
+                        ks.slot, ks.slot_to_key);
+        uint32_t *os_, *oi;
+        wfa_sort_pairs2(s, ks.slot, ks.idx, ks.slot_t, ks.idx_t, nullptr,
+                        nullptr, ks.hist, n, ks.bits, &os_, &oi, nullptr, 1, 0);
+        ks.segs(s, os_, n, 0);
+        HIPCHK(hipEventRecord(ev[1], s));
+        wfa_ffat_fire_offsets(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n, win,
+                              1, 1, st_fill, st_head, nf, d_on, nullptr,
+                              nullptr, nullptr);
+        HIPCHK(hipEventRecord(ev[2], s));
+        wfa_gram_fold(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n, d_in, oi,
+                      d_ts, win, st_fill, st_acc, st_head, ks.slot_to_key, nf,
+                      o_key, o_gwid, d_out, o_ts, fires16, d_on);
+        HIPCHK(hipEventRecord(ev[3], s));
+        HIPCHK(hipStreamSynchronize(s));
+        if (it >= 0)
+            for (int k = 0; k < NS - 1; ++k) {
+                float ms = 0;
+                HIPCHK(hipEventElapsedTime(&ms, ev[k], ev[k + 1]));
+                acc[k] += ms * 1000.0;
+            }
+    }
+    for (auto& e : ev) (void)hipEventDestroy(e);
+    std::vector<std::pair<std::string, double>> out;
+    for (int k = 0; k < NS - 1; ++k) out.push_back({names[k], acc[k] / iters});
+    // FLOPs through the matrix cores per batch: 512 per tuple (v*v^T into
+    // a 16x16 accumulator), over the fold stage time
+    double fold_us = acc[2] / iters;
+    out.push_back({"gflops", fold_us > 0 ? (512.0 * n) / (fold_us * 1e3) : 0});
+    return out;
+}
+
 }  // namespace wfa
 
 #else  // !WFA_WITH_HIP
@@ -1911,6 +1990,12 @@ std::vector<std::pair<std::string, double>> debug_ffat_stage_times(int64_t, int6
 }
 std::vector<std::pair<std::string, double>> debug_a2a_stage_times(int64_t, int64_t,
                                                                   int) {
+    throw std::runtime_error("built without HIP");
+}
+std::vector<std::pair<std::string, double>> debug_gram_stage_times(int64_t,
+                                                                   int64_t,
+                                                                   int64_t,
+                                                                   int) {
     throw std::runtime_error("built without HIP");
 }
 std::shared_ptr<OpLogic> make_gpu_logic(const std::string&, const std::string&,
