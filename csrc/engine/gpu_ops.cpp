@@ -1084,6 +1084,9 @@ struct GpuGramLogic : GpuLogicBase {
     int64_t* d_on = nullptr;
     const float** d_incols = nullptr;
     float** d_outcols = nullptr;
+    uint32_t* d_inv = nullptr;
+    float* d_staged = nullptr;
+    int64_t staged_cap = 0;
 
     GpuGramLogic(int64_t w, int64_t mk, int dev, Schema os, int64_t cap) {
         win = w;
@@ -1128,10 +1131,20 @@ struct GpuGramLogic : GpuLogicBase {
         wfa_ffat_fire_offsets(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                               win, 1, 1, st_fill, st_head, nf, d_on, nullptr,
                               nullptr, nullptr);
+        if (n > staged_cap) {  // sorted row-major staging (see k_gram_stage)
+            auto& A = arena(device);
+            if (d_staged) {
+                A.put(d_staged, 64 * staged_cap);
+                A.put(d_inv, 4 * staged_cap);
+            }
+            staged_cap = n;
+            d_staged = (float*)A.get(64 * staged_cap);
+            d_inv = (uint32_t*)A.get(4 * staged_cap);
+        }
         wfa_gram_fold(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n, d_incols,
                       ks.idx_sorted, db->ts, win, st_fill, st_acc, st_head,
                       ks.slot_to_key, nf, ob->key, ob->col<int64_t>(0), d_outcols,
-                      ob->ts, ob->capacity, d_on);
+                      ob->ts, ob->capacity, d_on, d_inv, d_staged);
         HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
         ob->count = -1;
         ob->watermark = db->watermark;
@@ -1917,6 +1930,8 @@ std::vector<std::pair<std::string, double>> debug_gram_stage_times(
     float* o_cols[16];
     for (int c = 0; c < 16; ++c) o_cols[c] = (float*)A.get(4 * fires16);
     int64_t* d_on = (int64_t*)A.get(64);
+    uint32_t* d_inv = (uint32_t*)A.get(4 * n);
+    float* d_staged = (float*)A.get(64 * n);
     const float** d_in = (const float**)A.get(16 * 8);
     float** d_out = (float**)A.get(16 * 8);
     HIPCHK(hipMemcpyAsync(d_in, d_cols, 16 * 8, hipMemcpyHostToDevice, s));
@@ -1946,7 +1961,8 @@ std::vector<std::pair<std::string, double>> debug_gram_stage_times(
         HIPCHK(hipEventRecord(ev[2], s));
         wfa_gram_fold(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n, d_in, oi,
                       d_ts, win, st_fill, st_acc, st_head, ks.slot_to_key, nf,
-                      o_key, o_gwid, d_out, o_ts, fires16, d_on);
+                      o_key, o_gwid, d_out, o_ts, fires16, d_on, d_inv,
+                      d_staged);
         HIPCHK(hipEventRecord(ev[3], s));
         HIPCHK(hipStreamSynchronize(s));
         if (it >= 0)

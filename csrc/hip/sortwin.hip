@@ -1452,9 +1452,32 @@ extern "C" void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,
 // in a per-slot arena laid out in the SAME lane mapping.
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
+// inverse permutation: inv[idx_sorted[j]] = j (sorted position of row r)
+__global__ void k_inv_perm(const uint32_t* idx_sorted, int64_t n, uint32_t* inv) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * (int64_t)blockDim.x)
+        inv[idx_sorted[i]] = (uint32_t)i;
+}
+
+// transpose-gather the 16 value columns into SORTED row-major staging:
+// staged[inv[r]*16 + c] = colp[c][r].  Thread per SOURCE row: the 16
+// column reads are coalesced across the wave, and the 64 B row write is
+// one full cache line at a random offset — this replaces the per-MFMA-step
+// random gather that made the fold memory-bound (measured 2.4 ms/8M,
+// ~1.8 TFLOP/s; see profiles/gram_mfma_r02.md).
+__global__ void k_gram_stage(const float* const* colp, const uint32_t* inv,
+                             int64_t n, float* staged) {
+    for (int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; r < n;
+         r += gridDim.x * (int64_t)blockDim.x) {
+        const int64_t w = (int64_t)inv[r] * 16;
+#pragma unroll
+        for (int c = 0; c < 16; ++c) staged[w + c] = colp[c][r];
+    }
+}
+
 __global__ void k_gram(const uint32_t* seg_start, const uint32_t* seg_slot,
                        const int64_t* d_nseg, int64_t n,
-                       const float* const* colp,  // [16] input column ptrs
+                       const float* staged,  // [n][16] sorted row-major
                        const uint32_t* idx_sorted, const int64_t* ts_orig,
                        int64_t win, uint32_t* st_fill, float* st_acc /*256/slot*/,
                        uint32_t* st_head, const uint64_t* slot_to_key,
@@ -1467,7 +1490,6 @@ __global__ void k_gram(const uint32_t* seg_start, const uint32_t* seg_slot,
     const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
     const int krow = lane >> 4;   // 0..3 (K slice of this step)
     const int dim = lane & 15;    // matrix column
-    const float* mycol = colp[dim];
     for (int64_t j = wid; j < nseg; j += nw) {
         const uint32_t slot = seg_slot[j];
         const int64_t i0 = seg_start[j];
@@ -1484,7 +1506,7 @@ __global__ void k_gram(const uint32_t* seg_start, const uint32_t* seg_slot,
         while (pos < e) {
             int64_t chunk = min((int64_t)4, min(e - pos, win - cnt));
             float v = 0.f;
-            if (krow < chunk) v = mycol[idx_sorted[pos + krow]];
+            if (krow < chunk) v = staged[(pos + krow) * 16 + dim];
             acc = __builtin_amdgcn_mfma_f32_16x16x4f32(v, v, acc, 0, 0, 0);
             last_ts = ts_orig[idx_sorted[pos + chunk - 1]];
             pos += chunk;
@@ -1529,10 +1551,15 @@ extern "C" void wfa_gram_fold(wfa_stream_t s, const uint32_t* seg_start,
                               const uint32_t* fire_base, uint64_t* out_key,
                               int64_t* out_gwid, float* const* out_colp,
                               int64_t* out_ts, int64_t out_cap,
-                              int64_t* d_out_n) {
+                              int64_t* d_out_n, uint32_t* inv_scratch,
+                              float* staged) {
     hipStream_t st = (hipStream_t)s;
+    hipLaunchKernelGGL(k_inv_perm, dim3(nblk(n)), dim3(WFA_THREADS), 0, st,
+                       idx_sorted, n, inv_scratch);
+    hipLaunchKernelGGL(k_gram_stage, dim3(nblk(n)), dim3(WFA_THREADS), 0, st,
+                       colp, inv_scratch, n, staged);
     hipLaunchKernelGGL(k_gram, dim3(WFA_MAX_BLOCKS / 2), dim3(WFA_THREADS), 0, st,
-                       seg_start, seg_slot, d_nseg, n, colp, idx_sorted, ts_orig,
+                       seg_start, seg_slot, d_nseg, n, staged, idx_sorted, ts_orig,
                        win, st_fill, st_acc, st_head, slot_to_key, fire_base,
                        out_key, out_gwid, out_colp, out_ts, out_cap);
     hipLaunchKernelGGL(k_scale16, dim3(1), dim3(1), 0, st, d_out_n);

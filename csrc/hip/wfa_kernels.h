@@ -186,7 +186,8 @@ void wfa_gram_fold(wfa_stream_t s, const uint32_t* seg_start,
                    uint32_t* st_head, const uint64_t* slot_to_key,
                    const uint32_t* fire_base, uint64_t* out_key,
                    int64_t* out_gwid, float* const* out_colp,
-                   int64_t* out_ts, int64_t out_cap, int64_t* d_out_n);
+                   int64_t* out_ts, int64_t out_cap, int64_t* d_out_n, uint32_t* inv_scratch,
+                   float* staged);
 
 // stateful map/filter: per-key (slot) f64 state, key-order segment walk;
 // map writes results to original positions in place, filter fills flags
