@@ -1141,6 +1141,7 @@ struct GpuGramLogic : GpuLogicBase {
             d_staged = (float*)A.get(64 * staged_cap);
             d_inv = (uint32_t*)A.get(4 * staged_cap);
         }
+        wfa_gram_prep(stream, d_incols, ks.idx_sorted, n, d_inv, d_staged);
         wfa_gram_fold(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n, d_incols,
                       ks.idx_sorted, db->ts, win, st_fill, st_acc, st_head,
                       ks.slot_to_key, nf, ob->key, ob->col<int64_t>(0), d_outcols,
@@ -1940,8 +1941,9 @@ std::vector<std::pair<std::string, double>> debug_gram_stage_times(
     HIPCHK(hipMemsetAsync(st_head, 0, 4 * n_keys, s));
     HIPCHK(hipMemsetAsync(st_acc, 0, 4 * n_keys * 256, s));
 
-    constexpr int NS = 4;
-    const char* names[NS] = {"group", "fire_offsets", "mfma_fold", "gflops"};
+    constexpr int NS = 5;
+    const char* names[NS] = {"group", "fire_offsets", "stage_rows", "mfma_fold",
+                             "gflops"};
     hipEvent_t ev[NS];
     for (auto& e : ev) HIPCHK(hipEventCreate(&e));
     double acc[NS] = {0};
@@ -1959,11 +1961,13 @@ std::vector<std::pair<std::string, double>> debug_gram_stage_times(
                               1, 1, st_fill, st_head, nf, d_on, nullptr,
                               nullptr, nullptr);
         HIPCHK(hipEventRecord(ev[2], s));
+        wfa_gram_prep(s, d_in, oi, n, d_inv, d_staged);
+        HIPCHK(hipEventRecord(ev[3], s));
         wfa_gram_fold(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n, d_in, oi,
                       d_ts, win, st_fill, st_acc, st_head, ks.slot_to_key, nf,
                       o_key, o_gwid, d_out, o_ts, fires16, d_on, d_inv,
                       d_staged);
-        HIPCHK(hipEventRecord(ev[3], s));
+        HIPCHK(hipEventRecord(ev[4], s));
         HIPCHK(hipStreamSynchronize(s));
         if (it >= 0)
             for (int k = 0; k < NS - 1; ++k) {

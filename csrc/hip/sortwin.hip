@@ -1542,6 +1542,16 @@ __global__ void k_gram(const uint32_t* seg_start, const uint32_t* seg_slot,
 
 __global__ void k_scale16(int64_t* d_out_n) { *d_out_n *= 16; }
 
+extern "C" void wfa_gram_prep(wfa_stream_t s, const float* const* colp,
+                              const uint32_t* idx_sorted, int64_t n,
+                              uint32_t* inv_scratch, float* staged) {
+    hipStream_t st = (hipStream_t)s;
+    hipLaunchKernelGGL(k_inv_perm, dim3(nblk(n)), dim3(WFA_THREADS), 0, st,
+                       idx_sorted, n, inv_scratch);
+    hipLaunchKernelGGL(k_gram_stage, dim3(nblk(n)), dim3(WFA_THREADS), 0, st,
+                       colp, inv_scratch, n, staged);
+}
+
 extern "C" void wfa_gram_fold(wfa_stream_t s, const uint32_t* seg_start,
                               const uint32_t* seg_slot, const int64_t* d_nseg,
                               int64_t n, const float* const* colp,
@@ -1554,10 +1564,7 @@ extern "C" void wfa_gram_fold(wfa_stream_t s, const uint32_t* seg_start,
                               int64_t* d_out_n, uint32_t* inv_scratch,
                               float* staged) {
     hipStream_t st = (hipStream_t)s;
-    hipLaunchKernelGGL(k_inv_perm, dim3(nblk(n)), dim3(WFA_THREADS), 0, st,
-                       idx_sorted, n, inv_scratch);
-    hipLaunchKernelGGL(k_gram_stage, dim3(nblk(n)), dim3(WFA_THREADS), 0, st,
-                       colp, inv_scratch, n, staged);
+    (void)colp; (void)inv_scratch;  // staging done by wfa_gram_prep
     hipLaunchKernelGGL(k_gram, dim3(WFA_MAX_BLOCKS / 2), dim3(WFA_THREADS), 0, st,
                        seg_start, seg_slot, d_nseg, n, staged, idx_sorted, ts_orig,
                        win, st_fill, st_acc, st_head, slot_to_key, fire_base,

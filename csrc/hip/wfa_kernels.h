@@ -178,6 +178,9 @@ void wfa_ffat_cb_fold_fused(
 // (v_mfma_f32_16x16x4_f32).  Output: 16 rows per fired window
 // (key, gwid, ts, Gram row).  d_out_n (from the fire-offset scan) is
 // scaled to rows (×16) by the wrapper.
+void wfa_gram_prep(wfa_stream_t s, const float* const* colp,
+                   const uint32_t* idx_sorted, int64_t n,
+                   uint32_t* inv_scratch, float* staged);
 void wfa_gram_fold(wfa_stream_t s, const uint32_t* seg_start,
                    const uint32_t* seg_slot, const int64_t* d_nseg,
                    int64_t n, const float* const* colp,
