@@ -1981,7 +1981,7 @@ std::vector<std::pair<std::string, double>> debug_gram_stage_times(
     for (int k = 0; k < NS - 1; ++k) out.push_back({names[k], acc[k] / iters});
     // FLOPs through the matrix cores per batch: 512 per tuple (v*v^T into
     // a 16x16 accumulator), over the fold stage time
-    double fold_us = acc[2] / iters;
+    double fold_us = acc[3] / iters;
     out.push_back({"gflops", fold_us > 0 ? (512.0 * n) / (fold_us * 1e3) : 0});
     return out;
 }

@@ -1465,13 +1465,25 @@ __global__ void k_inv_perm(const uint32_t* idx_sorted, int64_t n, uint32_t* inv)
 // one full cache line at a random offset — this replaces the per-MFMA-step
 // random gather that made the fold memory-bound (measured 2.4 ms/8M,
 // ~1.8 TFLOP/s; see profiles/gram_mfma_r02.md).
+// 4 lanes cooperate per row: each writes one float4, so the 4 stores of
+// one row land in ONE 64 B line transaction (a scalar version did 16
+// separate 4 B writes per row -> L2 transaction-bound, 2.7 ms/8M)
 __global__ void k_gram_stage(const float* const* colp, const uint32_t* inv,
                              int64_t n, float* staged) {
-    for (int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; r < n;
-         r += gridDim.x * (int64_t)blockDim.x) {
-        const int64_t w = (int64_t)inv[r] * 16;
-#pragma unroll
-        for (int c = 0; c < 16; ++c) staged[w + c] = colp[c][r];
+    const int sub = threadIdx.x & 3;  // which float4 of the 16-f row
+    const float* c0 = colp[4 * sub + 0];
+    const float* c1 = colp[4 * sub + 1];
+    const float* c2 = colp[4 * sub + 2];
+    const float* c3 = colp[4 * sub + 3];
+    const int64_t rbase = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 2;
+    const int64_t rstride = ((int64_t)gridDim.x * blockDim.x) >> 2;
+    for (int64_t r = rbase; r < n; r += rstride) {
+        float4 v;
+        v.x = c0[r];
+        v.y = c1[r];
+        v.z = c2[r];
+        v.w = c3[r];
+        ((float4*)staged)[(int64_t)inv[r] * 4 + sub] = v;
     }
 }
 
