@@ -1552,6 +1552,122 @@ __global__ void k_gram(const uint32_t* seg_start, const uint32_t* seg_slot,
     }
 }
 
+// window-per-wave variant: one wave per FIRED WINDOW (k_gram_win, reads
+// per-slot state, never writes) then one wave per segment for the tail
+// (k_gram_tail, updates state) — TWO launches so state reads order before
+// writes.  At low key counts the per-segment serial MFMA chain starves
+// the chip (profiles/gram_mfma_r02.md); fired windows have independent
+// accumulators and spread across all wave slots.
+__global__ void k_gram_win(const uint32_t* seg_start, const uint32_t* seg_slot,
+                           const int64_t* d_nseg, const int64_t* d_total,
+                           int64_t n, const float* staged,
+                           const uint32_t* idx_sorted, const int64_t* ts_orig,
+                           int64_t win, const uint32_t* st_fill,
+                           const float* st_acc, const uint32_t* st_head,
+                           const uint64_t* slot_to_key,
+                           const uint32_t* fire_base, uint64_t* out_key,
+                           int64_t* out_gwid, float* const* out_colp,
+                           int64_t* out_ts, int64_t out_cap) {
+    const int64_t nseg = *d_nseg;
+    const int64_t total = *d_total;
+    const int lane = threadIdx.x & 63;
+    const int64_t wid = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
+    const int krow = lane >> 4;
+    const int dim = lane & 15;
+    for (int64_t p = wid; p < total; p += nw) {
+        // binary-search the owning segment in the fire-offset scan
+        int64_t lo = 0, hi = nseg - 1;
+        while (lo < hi) {
+            int64_t mid = (lo + hi + 1) >> 1;
+            if ((int64_t)fire_base[mid] <= p) lo = mid;
+            else hi = mid - 1;
+        }
+        const int64_t j = lo;
+        const int64_t w = p - (int64_t)fire_base[j];
+        const uint32_t slot = seg_slot[j];
+        const int64_t i0 = seg_start[j];
+        const int64_t fill = st_fill[slot];
+        f32x4 acc;
+        int64_t start, cnt;
+        if (w == 0) {
+            start = i0;
+            cnt = win - fill;
+            const float* sa = st_acc + (size_t)slot * 256;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) acc[r] = sa[lane * 4 + r];
+        } else {
+            start = i0 + w * win - fill;
+            cnt = win;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) acc[r] = 0.f;
+        }
+        for (int64_t q = 0; q < cnt; q += 4) {
+            float v = 0.f;
+            if (krow < cnt - q) v = staged[(start + q + krow) * 16 + dim];
+            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(v, v, acc, 0, 0, 0);
+        }
+        const int64_t base = p * 16;
+        if (base + 16 <= out_cap) {
+            const int64_t last_ts = ts_orig[idx_sorted[start + cnt - 1]];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = krow * 4 + r;
+                out_colp[dim][base + row] = acc[r];
+                if (dim == 0) {
+                    out_key[base + row] = slot_to_key[slot];
+                    out_gwid[base + row] = (int64_t)st_head[slot] + w;
+                    out_ts[base + row] = last_ts;
+                }
+            }
+        }
+    }
+}
+
+__global__ void k_gram_tail(const uint32_t* seg_start, const uint32_t* seg_slot,
+                            const int64_t* d_nseg, const int64_t* d_total,
+                            int64_t n, const float* staged, int64_t win,
+                            uint32_t* st_fill, float* st_acc, uint32_t* st_head,
+                            const uint32_t* fire_base) {
+    const int64_t nseg = *d_nseg;
+    const int64_t total = *d_total;
+    const int lane = threadIdx.x & 63;
+    const int64_t wid = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
+    const int krow = lane >> 4;
+    const int dim = lane & 15;
+    for (int64_t j = wid; j < nseg; j += nw) {
+        const uint32_t slot = seg_slot[j];
+        const int64_t i0 = seg_start[j];
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        const int64_t fill = st_fill[slot];
+        const int64_t fires =
+            ((j + 1 < nseg) ? (int64_t)fire_base[j + 1] : total) -
+            (int64_t)fire_base[j];
+        const int64_t consumed = fires == 0 ? 0 : fires * win - fill;
+        float* sa = st_acc + (size_t)slot * 256;
+        f32x4 acc;
+        if (fires == 0) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) acc[r] = sa[lane * 4 + r];
+        } else {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) acc[r] = 0.f;
+        }
+        for (int64_t q = i0 + consumed; q < e; q += 4) {
+            float v = 0.f;
+            if (q + krow < e) v = staged[(q + krow) * 16 + dim];
+            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(v, v, acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) sa[lane * 4 + r] = acc[r];
+        if (lane == 0) {
+            st_fill[slot] = (uint32_t)(fill + (e - i0) - fires * win);
+            st_head[slot] += (uint32_t)fires;
+        }
+    }
+}
+
 __global__ void k_scale16(int64_t* d_out_n) { *d_out_n *= 16; }
 
 extern "C" void wfa_gram_prep(wfa_stream_t s, const float* const* colp,
@@ -1577,10 +1693,29 @@ extern "C" void wfa_gram_fold(wfa_stream_t s, const uint32_t* seg_start,
                               float* staged) {
     hipStream_t st = (hipStream_t)s;
     (void)colp; (void)inv_scratch;  // staging done by wfa_gram_prep
-    hipLaunchKernelGGL(k_gram, dim3(WFA_MAX_BLOCKS / 2), dim3(WFA_THREADS), 0, st,
-                       seg_start, seg_slot, d_nseg, n, staged, idx_sorted, ts_orig,
-                       win, st_fill, st_acc, st_head, slot_to_key, fire_base,
-                       out_key, out_gwid, out_colp, out_ts, out_cap);
+    static int seg_mode = -1;
+    if (seg_mode < 0) {
+        const char* e = getenv("WFA_GRAM_SEG");
+        seg_mode = (e && e[0] == '1') ? 1 : 0;
+    }
+    if (seg_mode) {
+        // legacy wave-per-segment fold (A/B reference)
+        hipLaunchKernelGGL(k_gram, dim3(WFA_MAX_BLOCKS / 2), dim3(WFA_THREADS),
+                           0, st, seg_start, seg_slot, d_nseg, n, staged,
+                           idx_sorted, ts_orig, win, st_fill, st_acc, st_head,
+                           slot_to_key, fire_base, out_key, out_gwid, out_colp,
+                           out_ts, out_cap);
+    } else {
+        hipLaunchKernelGGL(k_gram_win, dim3(WFA_MAX_BLOCKS), dim3(WFA_THREADS),
+                           0, st, seg_start, seg_slot, d_nseg, d_out_n, n,
+                           staged, idx_sorted, ts_orig, win, st_fill, st_acc,
+                           st_head, slot_to_key, fire_base, out_key, out_gwid,
+                           out_colp, out_ts, out_cap);
+        hipLaunchKernelGGL(k_gram_tail, dim3(WFA_MAX_BLOCKS / 2),
+                           dim3(WFA_THREADS), 0, st, seg_start, seg_slot,
+                           d_nseg, d_out_n, n, staged, win, st_fill, st_acc,
+                           st_head, fire_base);
+    }
     hipLaunchKernelGGL(k_scale16, dim3(1), dim3(1), 0, st, d_out_n);
 }
 
