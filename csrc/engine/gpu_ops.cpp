@@ -1155,6 +1155,9 @@ struct GpuGramLogic : GpuLogicBase {
         record_ready(ob);
         out.emit(ob);
     }
+    void on_eos(EmitCtx&, RuntimeCtx&) override {
+        if (inited) ks.check_dense_overflow();
+    }
 };
 
 // ===== GPU -> host stage (D2H) =====
@@ -1678,9 +1681,13 @@ std::shared_ptr<OpLogic> make_gpu_logic(const std::string& kind, const std::stri
         l->eng_ = eng;
         return l;
     }
-    if (kind == "gpu_gram")
-        // ip: [win, max_keys]
-        return std::make_shared<GpuGramLogic>(ip[0], ip[1], device, os, out_batch);
+    if (kind == "gpu_gram") {
+        // ip: [win, max_keys, dense]
+        auto l = std::make_shared<GpuGramLogic>(ip[0], ip[1], device, os,
+                                                out_batch);
+        l->ks.dense = ip.size() > 2 && ip[2] != 0;
+        return l;
+    }
     if (kind == "gpu_exchange")
         return std::make_shared<GpuExchangeLogic>(eng, device, os, out_batch);
     if (kind == "gpu_to_host")

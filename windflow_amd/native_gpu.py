@@ -224,14 +224,15 @@ def gpu_keyby_exchange():
     return NativeLogic("gpu_exchange", "", [], [])
 
 
-def gpu_gram_windows(win, max_keys=1 << 14):
+def gpu_gram_windows(win, max_keys=1 << 14, dense_keys=False):
     """MFMA windowed Gram aggregator (MI355X extension): per-key tumbling
     windows of `win` tuples over 16 F32 payload columns; the window
     aggregate sum(v * v^T) — online covariance / Gram — is GEMM-shaped and
     runs on the matrix cores (v_mfma_f32_16x16x4_f32).  Emits 16 rows per
     fired window: (key, ts=last tuple, c0=gwid, c1..c16 = one Gram row).
     Output schema must be [I64] + [F32]*16."""
-    return NativeLogic("gpu_gram", "", [], [win, max_keys])
+    return NativeLogic("gpu_gram", "", [],
+                       [win, max_keys, 1 if dense_keys else 0])
 
 
 def gpu_count_sink():
