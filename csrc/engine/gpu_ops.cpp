@@ -794,6 +794,11 @@ struct GpuFfatLogic : GpuLogicBase {
     int64_t* st_last = nullptr;  // per-slot last-arrival ts (EOS flush)
     int64_t* d_on = nullptr;
     uint32_t* cb_nf = nullptr;  // CB fire-offset scratch
+    // pane-wave fold scratch (pane_len >= 32): completed-pane scan, pane
+    // partials, next open-pane accumulators
+    uint32_t* pw_base = nullptr;
+    float* pw_temp = nullptr;
+    float* st_acc_new = nullptr;
     // TB state
     float* tb_pend = nullptr;
     int64_t* tb_base = nullptr;
@@ -862,6 +867,12 @@ struct GpuFfatLogic : GpuLogicBase {
                      max_keys * (use_tree ? 2 * (1ll << ring_log2) : (1ll << ring_log2)));
         // fire-offset scratch doubles as the EOS-flush per-slot counter
         cb_nf = (uint32_t*)A.get(4 * (std::max(out_cap, max_keys) + 1));
+        if (!tb && !use_tree && pane_len >= 32 && pane_wave_enabled()) {
+            int64_t np = out_cap / pane_len + max_keys + 64;
+            pw_base = (uint32_t*)A.get(4 * (np + 1));
+            pw_temp = (float*)A.get(4 * (np + 1));
+            st_acc_new = (float*)A.get(4 * max_keys);
+        }
         use_pane2 = !tb && !use_tree && pane2_enabled() && pane_len >= 32;
         if (tb) {
             int64_t Rp = 1ll << pend_log2;
@@ -917,9 +928,36 @@ struct GpuFfatLogic : GpuLogicBase {
         return v;
     }
 
+    // pane-wave fold (see wfa_ffat_cb_fold_pw): default for pane_len >= 32,
+    // WFA_PANE_WAVE=0 restores the wave-per-segment fold
+    static bool pane_wave_enabled() {
+        static int v = -1;
+        if (v < 0) {
+            const char* e = getenv("WFA_PANE_WAVE");
+            v = (e && e[0] == '0') ? 0 : 1;
+        }
+        return v;
+    }
+
     void chain_cb(Batch* db, Batch* ob, int64_t n, RuntimeCtx& ctx) {
         ks.group(stream, db, vcol, ctx, /*want_vik=*/!use_tree);
         uint32_t* nf = cb_nf;
+        if (pw_base && !use_pane2) {
+            wfa_ffat_fire_offsets_pw(stream, ks.seg_start, ks.seg_slot,
+                                     ks.d_nseg, n, pane_len, P, S, st_fill,
+                                     st_head, nf, d_on, ks.idx_sorted, db->ts,
+                                     st_last, pw_base, d_on + 1);
+            wfa_ffat_cb_fold_pw(stream, ks.seg_start, ks.seg_slot, ks.d_nseg,
+                                d_on + 1, n, ks.v_as_f32, ks.v_dt,
+                                ks.idx_sorted, db->ts, pane_len, P, S, comb,
+                                ring_log2, st_count, st_fill, st_acc,
+                                st_acc_new, ring_or_tree, st_head, st_wsum,
+                                ks.slot_to_key, nf, pw_base, pw_temp, ob->key,
+                                (float*)ob->cols[0], ob->ts, ob->capacity);
+            HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8,
+                                  hipMemcpyDeviceToHost, stream));
+            return;
+        }
         wfa_ffat_fire_offsets(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                               pane_len, P, S, st_fill, st_head, nf, d_on,
                               ks.idx_sorted, db->ts, st_last);

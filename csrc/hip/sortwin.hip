@@ -1179,7 +1179,7 @@ __global__ void k_fire_count(const uint32_t* seg_start, const uint32_t* seg_slot
                              int64_t P, int64_t S, const uint32_t* st_fill,
                              const uint32_t* st_head, uint32_t* nf,
                              const uint32_t* idx_sorted, const int64_t* ts_orig,
-                             int64_t* st_last) {
+                             int64_t* st_last, uint32_t* pane_nf) {
     const int64_t nseg = *d_nseg;
     for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nseg;
          j += gridDim.x * (int64_t)blockDim.x) {
@@ -1193,6 +1193,7 @@ __global__ void k_fire_count(const uint32_t* seg_start, const uint32_t* seg_slot
             return x < (uint64_t)P ? 0ull : (x - (uint64_t)P) / (uint64_t)S + 1ull;
         };
         nf[j] = (uint32_t)(F(head0 + ncomp) - F(head0));
+        if (pane_nf) pane_nf[j] = (uint32_t)ncomp;
         if (st_last) {  // fused per-key last-arrival ts (EOS flush emit ts)
             const int64_t t = ts_orig[idx_sorted[e - 1]];
             if (t > st_last[slot]) st_last[slot] = t;
@@ -1236,8 +1237,27 @@ extern "C" void wfa_ffat_fire_offsets(wfa_stream_t s, const uint32_t* seg_start,
     hipStream_t st = (hipStream_t)s;
     hipLaunchKernelGGL(k_fire_count, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
                        st, seg_start, seg_slot, d_nseg, n, pane_len, P, S, st_fill,
-                       st_head, nf, idx_sorted, ts_orig, st_last);
+                       st_head, nf, idx_sorted, ts_orig, st_last,
+                       (uint32_t*)nullptr);
     hipLaunchKernelGGL(k_fire_scan, dim3(1), dim3(1024), 0, st, nf, d_nseg, d_out_n);
+}
+
+// variant also producing the per-segment COMPLETED-PANE scan (pane_base)
+// + total (*d_total_panes) for the pane-wave fold below
+extern "C" void wfa_ffat_fire_offsets_pw(
+    wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
+    const int64_t* d_nseg, int64_t n, int64_t pane_len, int64_t P, int64_t S,
+    const uint32_t* st_fill, const uint32_t* st_head, uint32_t* nf,
+    int64_t* d_out_n, const uint32_t* idx_sorted, const int64_t* ts_orig,
+    int64_t* st_last, uint32_t* pane_base, int64_t* d_total_panes) {
+    hipStream_t st = (hipStream_t)s;
+    hipLaunchKernelGGL(k_fire_count, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS),
+                       0, st, seg_start, seg_slot, d_nseg, n, pane_len, P, S,
+                       st_fill, st_head, nf, idx_sorted, ts_orig, st_last,
+                       pane_base);
+    hipLaunchKernelGGL(k_fire_scan, dim3(1), dim3(1024), 0, st, nf, d_nseg, d_out_n);
+    hipLaunchKernelGGL(k_fire_scan, dim3(1), dim3(1024), 0, st, pane_base, d_nseg,
+                       d_total_panes);
 }
 
 // ===== batched multi-key CB sliding-window fold (pane ring) =====
@@ -1414,6 +1434,173 @@ __global__ void k_ffat_cb_wave(const uint32_t* seg_start, const uint32_t* seg_sl
         }
     }
 #undef WCOMB
+}
+
+// ===== pane-wave CB fold =====
+// The serial per-key state machine above walks TUPLES; at pane_len >= 32
+// almost all of that walk is pane-partial accumulation, which is
+// independent per pane.  Decomposition: (A) one wave per completed pane
+// computes its partial from the sorted values into a dense temp array
+// (pane 0 folds the carried open-pane accumulator), plus one wave per
+// segment for the new open tail -> st_acc_new; (B) one thread per segment
+// replays the tiny per-PANE machine (ring/wsum/fires) over the temps —
+// 1/pane_len of the old serial work.  A only READS state; B writes it.
+__global__ void k_pane_partials(const uint32_t* seg_start,
+                                const uint32_t* seg_slot,
+                                const int64_t* d_nseg,
+                                const int64_t* d_total_panes, int64_t n,
+                                const void* v_f32, int vdt,
+                                const uint32_t* idx_sorted, int64_t pane_len,
+                                int comb, const uint32_t* st_fill,
+                                const float* st_acc, const uint32_t* pane_base,
+                                float* temp, float* st_acc_new) {
+    const int64_t nseg = *d_nseg;
+    const int64_t total = *d_total_panes;
+    const float ident = (comb == 1) ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
+    const int lane = threadIdx.x & 63;
+    const int64_t wid = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
+#define PWCOMB(a, b) ((comb == 1) ? fminf(a, b) : (comb == 2 ? fmaxf(a, b) : (a) + (b)))
+    for (int64_t g = wid; g < total + nseg; g += nw) {
+        int64_t j, p;
+        bool tail = g >= total;
+        if (tail) {
+            j = g - total;
+            p = -1;
+        } else {
+            int64_t lo = 0, hi = nseg - 1;
+            while (lo < hi) {
+                int64_t mid = (lo + hi + 1) >> 1;
+                if ((int64_t)pane_base[mid] <= g) lo = mid;
+                else hi = mid - 1;
+            }
+            j = lo;
+            p = g - (int64_t)pane_base[j];
+        }
+        const uint32_t slot = seg_slot[j];
+        const int64_t i0 = seg_start[j];
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        const int64_t fill = st_fill[slot];
+        const int64_t ncomp = (fill + (e - i0)) / pane_len;
+        int64_t start, cnt;
+        float carry = ident;
+        if (tail) {
+            // new open pane: rows after the last completed pane
+            start = i0 + (ncomp == 0 ? 0 : ncomp * pane_len - fill);
+            cnt = e - start;
+            if (ncomp == 0) carry = st_acc[slot];  // pane still open
+        } else if (p == 0) {
+            start = i0;
+            cnt = pane_len - fill;
+            carry = st_acc[slot];
+        } else {
+            start = i0 + p * pane_len - fill;
+            cnt = pane_len;
+        }
+        float part = ident;
+        for (int64_t q = lane; q < cnt; q += 64) {
+            float x = (comb == 3)
+                          ? 1.0f
+                          : wfa_val_at(v_f32, vdt,
+                                       vdt == 6 ? start + q : idx_sorted[start + q]);
+            part = PWCOMB(part, x);
+        }
+        for (int o = 32; o; o >>= 1) part = PWCOMB(part, __shfl_xor(part, o, 64));
+        part = PWCOMB(part, carry);
+        if (lane == 0) {
+            if (tail) st_acc_new[slot] = part;
+            else temp[g] = part;
+        }
+    }
+#undef PWCOMB
+}
+
+__global__ void k_pane_advance(const uint32_t* seg_start,
+                               const uint32_t* seg_slot, const int64_t* d_nseg,
+                               int64_t n, const uint32_t* idx_sorted,
+                               const int64_t* ts_orig, int64_t pane_len,
+                               int64_t P, int64_t S, int comb, int ring_log2,
+                               int64_t* st_count, uint32_t* st_fill,
+                               float* st_acc, const float* st_acc_new,
+                               float* ring, uint32_t* st_head, float* st_wsum,
+                               const uint64_t* slot_to_key,
+                               const uint32_t* fire_base,
+                               const uint32_t* pane_base, const float* temp,
+                               uint64_t* out_key, float* out_val,
+                               int64_t* out_ts, int64_t out_cap) {
+    const int64_t nseg = *d_nseg;
+    const uint32_t R = 1u << ring_log2;
+    const uint32_t Rm = R - 1;
+    for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nseg;
+         j += gridDim.x * (int64_t)blockDim.x) {
+        const uint32_t slot = seg_slot[j];
+        const int64_t i0 = seg_start[j];
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        const int64_t fill = st_fill[slot];
+        const int64_t len = e - i0;
+        const int64_t ncomp = (fill + len) / pane_len;
+        uint32_t head = st_head[slot];
+        float wsum = st_wsum[slot];
+        int64_t w = fire_base[j];
+        float* rg = ring + (size_t)slot * R;
+        const float* tp = temp + pane_base[j];
+        for (int64_t q = 0; q < ncomp; ++q) {
+            const float acc = tp[q];
+            if (comb == 0 || comb == 3) {
+                wsum += acc;
+                if (head >= (uint32_t)P) wsum -= rg[(head - (uint32_t)P) & Rm];
+            }
+            rg[head & Rm] = acc;
+            ++head;
+            if (head >= (uint32_t)P && ((head - (uint32_t)P) % (uint32_t)S) == 0) {
+                float res;
+                if (comb == 0 || comb == 3) {
+                    res = wsum;
+                } else {
+                    res = rg[(head - 1) & Rm];
+                    for (uint32_t z = 2; z <= (uint32_t)P; ++z) {
+                        float pv = rg[(head - z) & Rm];
+                        res = (comb == 1) ? fminf(res, pv) : fmaxf(res, pv);
+                    }
+                }
+                if (w < out_cap) {
+                    out_key[w] = slot_to_key[slot];
+                    out_val[w] = res;
+                    const int64_t last = i0 + (q + 1) * pane_len - fill - 1;
+                    out_ts[w] = ts_orig ? ts_orig[idx_sorted[last]] : 0;
+                }
+                ++w;
+            }
+        }
+        st_fill[slot] = (uint32_t)(fill + len - ncomp * pane_len);
+        st_acc[slot] = st_acc_new[slot];
+        st_head[slot] = head;
+        st_wsum[slot] = wsum;
+        st_count[slot] += len;
+    }
+}
+
+extern "C" void wfa_ffat_cb_fold_pw(
+    wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
+    const int64_t* d_nseg, const int64_t* d_total_panes, int64_t n,
+    const void* v_f32, int vdt, const uint32_t* idx_sorted,
+    const int64_t* ts_orig, int64_t pane_len, int64_t P, int64_t S, int comb,
+    int ring_log2, int64_t* st_count, uint32_t* st_fill, float* st_acc,
+    float* st_acc_new, float* ring, uint32_t* st_head, float* st_wsum,
+    const uint64_t* slot_to_key, const uint32_t* fire_base,
+    const uint32_t* pane_base, float* temp, uint64_t* out_key, float* out_val,
+    int64_t* out_ts, int64_t out_cap) {
+    hipStream_t st = (hipStream_t)s;
+    hipLaunchKernelGGL(k_pane_partials, dim3(WFA_MAX_BLOCKS), dim3(WFA_THREADS),
+                       0, st, seg_start, seg_slot, d_nseg, d_total_panes, n,
+                       v_f32, vdt, idx_sorted, pane_len, comb, st_fill, st_acc,
+                       pane_base, temp, st_acc_new);
+    hipLaunchKernelGGL(k_pane_advance, dim3(WFA_MAX_BLOCKS / 8),
+                       dim3(WFA_THREADS), 0, st, seg_start, seg_slot, d_nseg, n,
+                       idx_sorted, ts_orig, pane_len, P, S, comb, ring_log2,
+                       st_count, st_fill, st_acc, st_acc_new, ring, st_head,
+                       st_wsum, slot_to_key, fire_base, pane_base, temp,
+                       out_key, out_val, out_ts, out_cap);
 }
 
 extern "C" void wfa_ffat_cb_fold(wfa_stream_t s, const uint32_t* seg_start,

@@ -141,6 +141,28 @@ void wfa_ffat_fire_offsets(wfa_stream_t s, const uint32_t* seg_start,
                            const uint32_t* idx_sorted, const int64_t* ts_orig,
                            int64_t* st_last);
 
+// pane-wave variants: fire offsets also emit the completed-pane scan
+// (pane_base + *d_total_panes); the fold then runs (A) one wave per
+// completed pane -> temp partials + tail -> st_acc_new, (B) one thread per
+// segment replaying the per-PANE ring/wsum/fire machine (1/pane_len of
+// the per-tuple serial walk).  Use when pane_len >= 32.
+void wfa_ffat_fire_offsets_pw(
+    wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
+    const int64_t* d_nseg, int64_t n, int64_t pane_len, int64_t P, int64_t S,
+    const uint32_t* st_fill, const uint32_t* st_head, uint32_t* nf,
+    int64_t* d_out_n, const uint32_t* idx_sorted, const int64_t* ts_orig,
+    int64_t* st_last, uint32_t* pane_base, int64_t* d_total_panes);
+void wfa_ffat_cb_fold_pw(
+    wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
+    const int64_t* d_nseg, const int64_t* d_total_panes, int64_t n,
+    const void* v_f32, int vdt, const uint32_t* idx_sorted,
+    const int64_t* ts_orig, int64_t pane_len, int64_t P, int64_t S, int comb,
+    int ring_log2, int64_t* st_count, uint32_t* st_fill, float* st_acc,
+    float* st_acc_new, float* ring, uint32_t* st_head, float* st_wsum,
+    const uint64_t* slot_to_key, const uint32_t* fire_base,
+    const uint32_t* pane_base, float* temp, uint64_t* out_key, float* out_val,
+    int64_t* out_ts, int64_t out_cap);
+
 // ----- FFAT/pane sliding-window state machine (CB) -----
 // Batched multi-key redesign of the reference's per-key FlatFAT_GPU
 // (SURVEY.md §7 step 6): all keys' folds advance in ONE kernel over the
