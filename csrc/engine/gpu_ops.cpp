@@ -928,13 +928,17 @@ struct GpuFfatLogic : GpuLogicBase {
         return v;
     }
 
-    // pane-wave fold (see wfa_ffat_cb_fold_pw): default for pane_len >= 32,
-    // WFA_PANE_WAVE=0 restores the wave-per-segment fold
+    // pane-wave fold (see wfa_ffat_cb_fold_pw) — measured SLOWER than the
+    // wave-per-segment fold at the flagship shape (23.7 vs 25.7 B t/s,
+    // 16M batch, 8192 keys: the extra pane-scan + temp round-trip +
+    // low-occupancy advance outweigh the shorter serial walk), so it is
+    // OPT-IN via WFA_PANE_WAVE=1 (kept: correct per full suite + fuzz,
+    // and the decomposition is the right shape for much longer panes)
     static bool pane_wave_enabled() {
         static int v = -1;
         if (v < 0) {
             const char* e = getenv("WFA_PANE_WAVE");
-            v = (e && e[0] == '0') ? 0 : 1;
+            v = (e && e[0] == '1') ? 1 : 0;
         }
         return v;
     }
