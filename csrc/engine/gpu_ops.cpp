@@ -1324,6 +1324,7 @@ struct GpuExchangeLogic : GpuLogicBase {
     uint32_t *idx = nullptr, *dest_t = nullptr, *idx_t = nullptr;
     int dest_flip = 0;
     uint32_t* hist = nullptr;
+    uint32_t* hist_b = nullptr;  // ping-pongs with dest2 (meta vs main stream)
     uint32_t* d_counts = nullptr;
     int64_t* d_meta = nullptr;      // [world+2] send metadata
     int64_t* d_meta_all = nullptr;  // [world*(world+2)] allgathered
@@ -1352,6 +1353,7 @@ struct GpuExchangeLogic : GpuLogicBase {
                 "gpu_keyby_exchange supports up to 8 ranks (one xGMI node)");
         bits = 1;
         while ((1 << bits) < world) ++bits;
+        if (bits < 5) bits = 5;  // 8-bit sort machinery (1 pass, fused hist)
     }
     void init_device() override {
         comm = get_rccl_comm(eng->rccl_id, rank, world, device);
@@ -1362,6 +1364,7 @@ struct GpuExchangeLogic : GpuLogicBase {
         dest_t = (uint32_t*)A.get(4 * out_cap);
         idx_t = (uint32_t*)A.get(4 * out_cap);
         hist = (uint32_t*)A.get(4 * wfa_sort_hist_u32(out_cap));
+        hist_b = (uint32_t*)A.get(4 * wfa_sort_hist_u32(out_cap));
         d_counts = (uint32_t*)A.get(4 * world + 64);
         d_meta = (int64_t*)A.get(8 * (world + 2));
         d_meta_all = (int64_t*)A.get(8 * world * (world + 2));
@@ -1550,10 +1553,11 @@ struct GpuExchangeLogic : GpuLogicBase {
         // batch's sort/gather/sends): the per-batch host sync then waits
         // only for THIS batch's input + tiny kernels, not the pipeline
         uint32_t* dest = dest2[dest_flip];
+        uint32_t* hi = dest_flip ? hist_b : hist;
         dest_flip ^= 1;
         if (db->ready_event)
             HIPCHK(hipStreamWaitEvent(meta_stream, (hipEvent_t)db->ready_event, 0));
-        wfa_bucket_by_key(meta_stream, db->key, n, world, dest);
+        wfa_bucket_by_key_h(meta_stream, db->key, n, world, dest, hi);
         wfa_count_u32(meta_stream, dest, n, d_counts, world);
         HIPCHK(hipEventRecord(ev_counts, meta_stream));
         start_meta(true, wm, false);
@@ -1583,9 +1587,9 @@ struct GpuExchangeLogic : GpuLogicBase {
         // stream; waits for this batch's bucket output)
         HIPCHK(hipStreamWaitEvent(stream, ev_counts, 0));
         uint32_t *od, *oi;
-        wfa_sort_pairs2(stream, dest, idx, dest_t, idx_t, nullptr, nullptr, hist,
-                        n, bits, &od, &oi, nullptr, /*implicit_iota=*/1,
-                        /*base_shift=*/0);
+        wfa_sort_pairs2_ph(stream, dest, idx, dest_t, idx_t, nullptr, nullptr,
+                           hi, n, bits, &od, &oi, nullptr, /*implicit_iota=*/1,
+                           /*base_shift=*/0);
         size_t nc = db->cols.size();
         std::vector<void*> ptrs(2 * nc);
         for (size_t c = 0; c < nc; ++c) {

@@ -278,6 +278,43 @@ extern "C" void wfa_key_to_slot_h(wfa_stream_t s, const uint64_t* key,
                        table_cap, slot_out, slot_to_key, val, hist, nb, shift);
 }
 
+// keyby-exchange dest bucketing FUSED with the dest-sort's pass-0
+// histogram (same pattern): dest = splitmix64(key) % world, and the
+// single 8-bit dest-sort pass then skips its count read entirely.
+// Runs on the exchange's META stream, overlapped with the previous
+// batch's sort/gather (hist ping-pongs with the dest array).
+__global__ void k_bucket_h(const uint64_t* key, int64_t n, int world,
+                           uint32_t* dest, uint32_t* hist, int64_t nblocks) {
+    __shared__ uint32_t h[WFA_THREADS / 64][256];
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    for (int d = threadIdx.x; d < 256; d += blockDim.x)
+        for (int w = 0; w < WFA_THREADS / 64; ++w) h[w][d] = 0;
+    __syncthreads();
+    int64_t waveBase = (int64_t)blockIdx.x * RS8_PER_BLOCK + wave * RS8_PER_WAVE;
+#pragma unroll
+    for (int j = 0; j < RS8_IPT; ++j) {
+        int64_t i = waveBase + j * 64 + lane;
+        if (i >= n) continue;
+        uint32_t d = (uint32_t)(mix64s(key[i]) % (uint64_t)world);
+        dest[i] = d;
+        atomicAdd(&h[wave][d], 1u);
+    }
+    __syncthreads();
+    for (int d = threadIdx.x; d < 256; d += blockDim.x) {
+        uint32_t t = 0;
+        for (int w = 0; w < WFA_THREADS / 64; ++w) t += h[w][d];
+        hist[(int64_t)d * nblocks + blockIdx.x] = t;
+    }
+}
+
+extern "C" void wfa_bucket_by_key_h(wfa_stream_t s, const uint64_t* key,
+                                    int64_t n, int world, uint32_t* dest,
+                                    uint32_t* hist) {
+    int64_t nb = (n + RS8_PER_BLOCK - 1) / RS8_PER_BLOCK;
+    hipLaunchKernelGGL(k_bucket_h, dim3(nb), dim3(WFA_THREADS), 0,
+                       (hipStream_t)s, key, n, world, dest, hist, nb);
+}
+
 // dense key->slot FUSED with the radix sort's pass-0 per-block histogram:
 // the sort's first count pass re-reads the whole slot array; since this
 // kernel just wrote it, accumulate the same LDS histogram here (sort

@@ -65,6 +65,9 @@ void wfa_key_dense_h(wfa_stream_t s, const uint64_t* key, int64_t n,
                      uint32_t* overflow, const uint16_t* val, uint32_t* hist,
                      int shift);
 void wfa_iota_u64(wfa_stream_t s, uint64_t* p, int64_t n);
+// exchange dest bucketing fused with the dest-sort pass-0 histogram
+void wfa_bucket_by_key_h(wfa_stream_t s, const uint64_t* key, int64_t n,
+                         int world, uint32_t* dest, uint32_t* hist);
 
 // ----- stable LSD radix sort of (slot, iota idx) pairs, 4-bit digits -----
 // bits: how many low bits of slot to sort on. tmp arrays sized n (u32 each).
