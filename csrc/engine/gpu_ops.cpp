@@ -1341,6 +1341,11 @@ struct GpuExchangeLogic : GpuLogicBase {
     // serialization attributed to the first kernel after the sync)
     hipStream_t meta_stream = nullptr;
     hipEvent_t ev_counts = nullptr;
+    // per-ping-pong-buffer liveness: bucket(N+2) on the meta stream must
+    // not overwrite dest2[i]/hist[i] before the MAIN-stream sort that
+    // reads them has executed (a slow peer's RCCL group can let the host
+    // run several batches ahead)
+    hipEvent_t ev_dest[2] = {nullptr, nullptr};
 
     GpuExchangeLogic(Engine* e, int dev, Schema os, int64_t cap) : eng(e) {
         device = dev;
@@ -1382,12 +1387,18 @@ struct GpuExchangeLogic : GpuLogicBase {
         send_pool->device = device;
         HIPCHK(hipStreamCreateWithFlags(&meta_stream, hipStreamNonBlocking));
         HIPCHK(hipEventCreateWithFlags(&ev_counts, hipEventDisableTiming));
+        for (int i = 0; i < 2; ++i) {
+            HIPCHK(hipEventCreateWithFlags(&ev_dest[i], hipEventDisableTiming));
+            HIPCHK(hipEventRecord(ev_dest[i], stream));  // initially free
+        }
     }
     ~GpuExchangeLogic() override {
         if (h_meta) (void)hipHostFree(h_meta);
         if (h_counts) (void)hipHostFree(h_counts);
         if (meta_stream) (void)hipStreamDestroy(meta_stream);
         if (ev_counts) (void)hipEventDestroy(ev_counts);
+        for (int i = 0; i < 2; ++i)
+            if (ev_dest[i]) (void)hipEventDestroy(ev_dest[i]);
     }
 
     // kick the metadata rendezvous on meta_stream: pack counts -> allgather
@@ -1554,9 +1565,11 @@ struct GpuExchangeLogic : GpuLogicBase {
         // only for THIS batch's input + tiny kernels, not the pipeline
         uint32_t* dest = dest2[dest_flip];
         uint32_t* hi = dest_flip ? hist_b : hist;
+        hipEvent_t ev_free = ev_dest[dest_flip];
         dest_flip ^= 1;
         if (db->ready_event)
             HIPCHK(hipStreamWaitEvent(meta_stream, (hipEvent_t)db->ready_event, 0));
+        HIPCHK(hipStreamWaitEvent(meta_stream, ev_free, 0));
         wfa_bucket_by_key_h(meta_stream, db->key, n, world, dest, hi);
         wfa_count_u32(meta_stream, dest, n, d_counts, world);
         HIPCHK(hipEventRecord(ev_counts, meta_stream));
@@ -1590,6 +1603,7 @@ struct GpuExchangeLogic : GpuLogicBase {
         wfa_sort_pairs2_ph(stream, dest, idx, dest_t, idx_t, nullptr, nullptr,
                            hi, n, bits, &od, &oi, nullptr, /*implicit_iota=*/1,
                            /*base_shift=*/0);
+        HIPCHK(hipEventRecord(ev_free, stream));  // dest/hist consumed
         size_t nc = db->cols.size();
         std::vector<void*> ptrs(2 * nc);
         for (size_t c = 0; c < nc; ++c) {
