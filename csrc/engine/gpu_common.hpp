@@ -207,12 +207,12 @@ struct GpuLogicBase : OpLogic {
         return get_from(*in_pool);
     }
 
-    // H2D staging: host batch -> fresh device batch (reference
-    // forward_emitter_gpu.hpp CPU->GPU 1-batch overlap, redesigned:
-    // whole-SoA copies).  The host batch may be recycled by a CPU producer
-    // the moment we release it, so it is held until its copies' event
-    // fires — released OPPORTUNISTICALLY on later calls instead of a
-    // per-batch stream sync (round 1 synced every batch).
+    // Staging in-flight window (H2D and peer D2D): the source batch may
+    // be recycled by its producer the moment we release it, so it is held
+    // until its copies' event fires — released OPPORTUNISTICALLY on later
+    // calls instead of a per-batch stream sync (round 1 synced every
+    // batch; reference forward_emitter_gpu.hpp has the same 1-batch
+    // overlap on its pinned staging buffers).
     std::deque<std::pair<Batch*, hipEvent_t>> h2d_inflight;
 
     void drain_h2d(bool block) {
@@ -278,6 +278,7 @@ struct GpuLogicBase : OpLogic {
     // otherwise — the reference has no P2P path at all
     // (wf/forward_emitter_gpu.hpp:296-328 always stages via pinned host).
     Batch* peer_copy(Batch* b, RuntimeCtx& ctx) {
+        drain_h2d(false);
         enable_peer(device, b->device);
         // producer-side contents must be valid before the engine reads them
         // from the consumer device: wait on the producing event (legal
@@ -302,9 +303,11 @@ struct GpuLogicBase : OpLogic {
         db->born_us = b->born_us;
         if (ctx.stats) ctx.stats->bytes_d2h += bytes;  // inter-device traffic
         // the source batch may be recycled by its (other-device) producer
-        // the moment we release: the copies must have read it first
-        HIPCHK(hipStreamSynchronize(stream));
-        release(b);
+        // the moment we release: hold it on the copies' event instead of a
+        // per-batch stream sync (same 1-batch-overlap protocol as H2D)
+        HIPCHK(hipEventRecord((hipEvent_t)db->ready_event, stream));
+        h2d_inflight.push_back({b, (hipEvent_t)db->ready_event});
+        if (h2d_inflight.size() > 4) drain_h2d(true);
         return db;
     }
 
