@@ -17,12 +17,16 @@ from windflow_amd.builders_gpu import (Source_GPU_Builder,   # noqa: E402
 
 def run(n_steps=100, batch=16_777_216, keys=8192, dense=True):
     n = n_steps * batch
+    # global ts advances 1/tuple, so each key sees ts every ~keys units:
+    # scale the window extent so a key-window holds ~1000 tuples (the CB
+    # flagship shape) instead of firing the empty inter-tuple window grid
+    win, slide = 1000 * keys, 100 * keys
     src = (Source_GPU_Builder(native_gpu.gpu_source(n, keys, batch, vdt=5))
            .withOutputSchema([5]).withOutputBatchSize(batch).build())
     ff = (Ffat_Windows_GPU_Builder(
-        native_gpu.gpu_ffat_windows(native_gpu.COMB_SUM, 0, 1000, 100,
+        native_gpu.gpu_ffat_windows(native_gpu.COMB_SUM, 0, win, slide,
                                     max_keys=keys, tb=True, lateness=0,
-                                    dense_keys=dense))
+                                    pend_ring_log2=18, dense_keys=dense))
           .withOutputSchema([2]).withOutputBatchSize(2 * batch).build())
     snk = Sink_GPU_Builder(native_gpu.gpu_count_sink()).build()
     g = wf.PipeGraph("tb_bench")
