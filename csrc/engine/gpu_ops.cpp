@@ -992,7 +992,7 @@ struct GpuFfatLogic : GpuLogicBase {
         Batch* db = input_on_device(in, ctx);
         int64_t n = db->count;
         if (tb) {
-            ks.group(stream, db, vcol, ctx);
+            ks.group(stream, db, vcol, ctx, /*want_vik=*/true);
             wfa_seg_last_ts(stream, ks.seg_start, ks.seg_slot, ks.d_nseg, n,
                             ks.idx_sorted, db->ts, st_last);
             tb_round_with_count(db, n, db->watermark, out, ctx);
@@ -1043,7 +1043,8 @@ struct GpuFfatLogic : GpuLogicBase {
                           tb_pend, tb_base, tb_last, st_head, st_wsum,
                           ring_or_tree, ks.d_nslots, ks.slot_to_key, tb_nf,
                           tb_flags, tb_flags + 1, ob->key, (float*)ob->cols[0],
-                          ob->ts, ob->capacity, d_on);
+                          ob->ts, ob->capacity, d_on,
+                          db && db->ts_mono ? 1 : 0);
         HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
         ob->count = -1;
         ob->watermark = wm - lateness;

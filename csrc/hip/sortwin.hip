@@ -2078,7 +2078,10 @@ __global__ void k_tb_lift(const uint32_t* seg_start, const uint32_t* seg_slot,
                 atomicAdd(overflow, 1u);
                 continue;
             }
-            const float x = (comb == 3) ? 1.0f : wfa_val_at(v_f32, vdt, idx_sorted[i]);
+            const float x = (comb == 3)
+                                ? 1.0f
+                                : wfa_val_at(v_f32, vdt,
+                                             vdt == 6 ? i : idx_sorted[i]);
             float* cell = &pd[(uint64_t)p & Pm];
             *cell = (comb == 1) ? fminf(*cell, x)
                                 : (comb == 2 ? fmaxf(*cell, x) : *cell + x);
@@ -2197,6 +2200,92 @@ __global__ void k_tb_advance(const uint32_t* n_slots, int64_t limit_pane,
     }
 }
 
+// Wave-per-segment TB lift for MONOTONIC-ts batches: pane boundaries are
+// found by lane-uniform binary search (log2(seg_len) ts reads per PANE
+// instead of one random gather per ROW), and pane partials come from
+// coalesced wave reduces.  The thread-per-segment kernel below walks
+// tuples serially — TB measured 11x slower than CB at the flagship shape
+// because of it (tools/tb_bench.py).  Gated on Batch::ts_mono; arbitrary
+// ts orders take the serial kernel.
+__global__ void k_tb_lift_wave(const uint32_t* seg_start, const uint32_t* seg_slot,
+                               const int64_t* d_nseg, int64_t n,
+                               const void* v_f32, int vdt,
+                               const uint32_t* idx_sorted,
+                               const int64_t* ts_orig, int64_t pane_len,
+                               int64_t P, int64_t S, int comb, int pend_log2,
+                               float* pend, int64_t* pend_base,
+                               int64_t* last_pane, uint32_t* ignored,
+                               uint32_t* overflow) {
+    const int64_t nseg = *d_nseg;
+    const uint32_t Rp = 1u << pend_log2;
+    const uint32_t Pm = Rp - 1;
+    const float ident = (comb == 1) ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
+    const int lane = threadIdx.x & 63;
+    const int64_t wid = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
+#define TLCOMB(a, b) ((comb == 1) ? fminf(a, b) : (comb == 2 ? fmaxf(a, b) : (a) + (b)))
+    for (int64_t j = wid; j < nseg; j += nw) {
+        const uint32_t slot = seg_slot[j];
+        int64_t i = seg_start[j];
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        float* pd = pend + (size_t)slot * Rp;
+        int64_t base = pend_base[slot];
+        int64_t lastp = last_pane[slot];
+        uint32_t ign = 0;
+        while (i < e) {
+            const int64_t t0 = ts_orig[idx_sorted[i]];
+            const int64_t p = t0 / pane_len;
+            if (base < 0) {
+                int64_t w0 = t0 - P * pane_len + 1;
+                w0 = w0 <= 0 ? 0 : (w0 + S * pane_len - 1) / (S * pane_len);
+                base = w0 * S;
+            }
+            // first index with ts >= (p+1)*pane_len (ts nondecreasing in
+            // arrival order; stable sort preserved it per segment)
+            int64_t lo = i + 1, hi = e;
+            const int64_t tlim = (p + 1) * pane_len;
+            while (lo < hi) {
+                int64_t mid = (lo + hi) >> 1;
+                if (ts_orig[idx_sorted[mid]] < tlim) lo = mid + 1;
+                else hi = mid;
+            }
+            const int64_t pe = lo;
+            if (p < base) {
+                ign += (uint32_t)(pe - i);
+                i = pe;
+                continue;
+            }
+            if (p - base >= (int64_t)Rp) {
+                if (lane == 0) atomicAdd(overflow, 1u);
+                i = pe;
+                continue;
+            }
+            float part = ident;
+            for (int64_t q = i + lane; q < pe; q += 64) {
+                float x = (comb == 3)
+                              ? 1.0f
+                              : wfa_val_at(v_f32, vdt,
+                                           vdt == 6 ? q : idx_sorted[q]);
+                part = TLCOMB(part, x);
+            }
+            for (int o = 32; o; o >>= 1)
+                part = TLCOMB(part, __shfl_xor(part, o, 64));
+            if (lane == 0) {
+                float* cell = &pd[(uint64_t)p & Pm];
+                *cell = TLCOMB(*cell, part);
+            }
+            if (p > lastp) lastp = p;
+            i = pe;
+        }
+        if (lane == 0) {
+            pend_base[slot] = base;
+            last_pane[slot] = lastp;
+            if (ign) atomicAdd(ignored, ign);
+        }
+    }
+#undef TLCOMB
+}
+
 extern "C" void wfa_ffat_tb_round(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
     const int64_t* d_nseg, int64_t n, const void* v_f32, int vdt,
@@ -2206,13 +2295,22 @@ extern "C" void wfa_ffat_tb_round(
     uint32_t* st_head, float* st_wsum, float* ring, const uint32_t* n_slots,
     const uint64_t* slot_to_key, uint32_t* nf, uint32_t* ignored,
     uint32_t* overflow, uint64_t* out_key, float* out_val, int64_t* out_ts,
-    int64_t out_cap, int64_t* d_out_n) {
+    int64_t out_cap, int64_t* d_out_n, int ts_mono) {
     hipStream_t st = (hipStream_t)s;
-    if (n > 0)
-        hipLaunchKernelGGL(k_tb_lift, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0,
-                           st, seg_start, seg_slot, d_nseg, n, v_f32, vdt, idx_sorted,
-                           ts_orig, pane_len, P, S, comb, pend_log2, pend, pend_base,
-                           last_pane, ignored, overflow);
+    if (n > 0) {
+        if (ts_mono)
+            hipLaunchKernelGGL(k_tb_lift_wave, dim3(WFA_MAX_BLOCKS / 2),
+                               dim3(WFA_THREADS), 0, st, seg_start, seg_slot,
+                               d_nseg, n, v_f32, vdt, idx_sorted, ts_orig,
+                               pane_len, P, S, comb, pend_log2, pend, pend_base,
+                               last_pane, ignored, overflow);
+        else
+            hipLaunchKernelGGL(k_tb_lift, dim3(WFA_MAX_BLOCKS / 8),
+                               dim3(WFA_THREADS), 0, st, seg_start, seg_slot,
+                               d_nseg, n, v_f32, vdt, idx_sorted, ts_orig,
+                               pane_len, P, S, comb, pend_log2, pend, pend_base,
+                               last_pane, ignored, overflow);
+    }
     hipLaunchKernelGGL(k_tb_count, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS), 0, st,
                        n_slots, limit_pane, pend_base, last_pane, st_head, P, S, nf);
     hipLaunchKernelGGL(k_tb_scan, dim3(1), dim3(1024), 0, st, nf, n_slots, d_out_n);
