@@ -475,6 +475,74 @@ static std::string gen_tb_kernels(const JitFoldSpec& fs) {
          "        last_pane[slot] = lastp;\n"
          "        if (ign) atomicAdd(ignored, ign);\n"
          "    }\n}\n";
+    // wave-per-segment lift for MONOTONIC-ts batches (mirrors the native
+    // k_tb_lift_wave): pane boundaries by lane-uniform binary search, pane
+    // partials by wave jcomb — the serial kernel above stays the fallback
+    // for arbitrary ts orders.
+    s += "extern \"C\" __global__ void jit_tb_lift_wave(\n"
+         "    const u32* seg_start, const u32* seg_slot, const i64* d_nseg,\n"
+         "    i64 n, ";
+    s += KARGS_COLS;
+    s += "    const u32* idx_sorted, const i64* ts_orig,\n"
+         "    i64 pane_len, i64 P, i64 S, int pend_log2, acct* pend,\n"
+         "    i64* pend_base, i64* last_pane, const u64* slot_to_key,\n"
+         "    u32* ignored, u32* overflow) {\n"
+         "    const i64 nseg = *d_nseg;\n"
+         "    const u32 Rp = 1u << pend_log2;\n"
+         "    const u32 Pm = Rp - 1;\n"
+         "    const int lane = threadIdx.x & 63;\n"
+         "    const i64 wid = ((i64)blockIdx.x * blockDim.x + threadIdx.x) >> 6;\n"
+         "    const i64 nw = ((i64)gridDim.x * blockDim.x) >> 6;\n"
+         "    for (i64 j = wid; j < nseg; j += nw) {\n"
+         "        const u32 slot = seg_slot[j];\n"
+         "        const u64 key = slot_to_key[slot]; (void)key;\n"
+         "        i64 i = seg_start[j];\n"
+         "        const i64 e = (j + 1 < nseg) ? seg_start[j + 1] : n;\n"
+         "        acct* pd = pend + (usz)slot * Rp * NF;\n"
+         "        i64 base = pend_base[slot];\n"
+         "        i64 lastp = last_pane[slot];\n"
+         "        u32 ign = 0;\n"
+         "        while (i < e) {\n"
+         "            const i64 t0 = ts_orig[idx_sorted[i]];\n"
+         "            const i64 p = t0 / pane_len;\n"
+         "            if (base < 0) {\n"
+         "                i64 w0 = t0 - P * pane_len + 1;\n"
+         "                w0 = w0 <= 0 ? 0 : (w0 + S * pane_len - 1) / (S * pane_len);\n"
+         "                base = w0 * S;\n"
+         "            }\n"
+         "            i64 lo = i + 1, hi = e;\n"
+         "            const i64 tlim = (p + 1) * pane_len;\n"
+         "            while (lo < hi) {\n"
+         "                i64 mid = (lo + hi) >> 1;\n"
+         "                if (ts_orig[idx_sorted[mid]] < tlim) lo = mid + 1;\n"
+         "                else hi = mid;\n"
+         "            }\n"
+         "            const i64 pe = lo;\n"
+         "            if (p < base) { ign += (u32)(pe - i); i = pe; continue; }\n"
+         "            if (p - base >= (i64)Rp) {\n"
+         "                if (lane == 0) atomicAdd(overflow, 1u);\n"
+         "                i = pe;\n"
+         "                continue;\n"
+         "            }\n"
+         "            Acc part = jident();\n"
+         "            for (i64 q = i + lane; q < pe; q += 64) {\n"
+         "                const u32 r = idx_sorted[q];\n"
+         "                part = jcomb(part, JLOADS(r));\n"
+         "            }\n"
+         "            part = jwave(part);\n"
+         "            if (lane == 0) {\n"
+         "                acct* cell = pd + (usz)((u64)p & Pm) * NF;\n"
+         "                acc_store(cell, jcomb(acc_load(cell), part));\n"
+         "            }\n"
+         "            if (p > lastp) lastp = p;\n"
+         "            i = pe;\n"
+         "        }\n"
+         "        if (lane == 0) {\n"
+         "            pend_base[slot] = base;\n"
+         "            last_pane[slot] = lastp;\n"
+         "            if (ign) atomicAdd(ignored, ign);\n"
+         "        }\n"
+         "    }\n}\n";
     s += "extern \"C\" __global__ void jit_tb_advance(\n"
          "    const u32* n_slots, i64 limit_pane, i64 pane_len, i64 P, i64 S,\n"
          "    int ring_log2, int pend_log2, acct* pend, i64* pend_base,\n"
@@ -728,7 +796,8 @@ struct GpuJitFfatLogic : GpuLogicBase {
     int ring_log2;
     KeyedScratch ks;
     hipFunction_t f_cb = nullptr, f_cb_wave = nullptr, f_flush = nullptr;
-    hipFunction_t f_tb_lift = nullptr, f_tb_adv = nullptr, f_fill = nullptr;
+    hipFunction_t f_tb_lift = nullptr, f_tb_lift_wave = nullptr;
+    hipFunction_t f_tb_adv = nullptr, f_fill = nullptr;
     // state arenas
     uint32_t* st_fill = nullptr;
     float* st_acc = nullptr;
@@ -774,6 +843,7 @@ struct GpuJitFfatLogic : GpuLogicBase {
         f_cb_wave = jit_fn(mod, "jit_cb_fold_wave");
         f_flush = jit_fn(mod, "jit_cb_flush");
         f_tb_lift = jit_fn(mod, "jit_tb_lift");
+        f_tb_lift_wave = jit_fn(mod, "jit_tb_lift_wave");
         f_tb_adv = jit_fn(mod, "jit_tb_advance");
         f_fill = jit_fn(mod, "jit_fill_ident");
         ks.alloc(device, out_cap, max_keys, stream);
@@ -866,6 +936,7 @@ struct GpuJitFfatLogic : GpuLogicBase {
 
     void tb_round(Batch* db, int64_t n, int64_t wm, EmitCtx& out,
                   RuntimeCtx& ctx) {
+        const bool mono = db && db->ts_mono;
         Batch* ob = get_dev();
         int64_t limit = (wm - lateness) / pane_len - 1;
         if (n > 0) {
@@ -882,7 +953,10 @@ struct GpuJitFfatLogic : GpuLogicBase {
             uint32_t* ign = tb_flags;
             uint32_t* ovf = tb_flags + 1;
             a.add(ign); a.add(ovf);
-            launch(f_tb_lift, stream, 256, a);
+            if (mono)
+                launch(f_tb_lift_wave, stream, 2048, a);
+            else
+                launch(f_tb_lift, stream, 256, a);
         }
         wfa_tb_count(stream, ks.d_nslots, limit, tb_base, tb_last_pane, st_head,
                      P, S, nf);
