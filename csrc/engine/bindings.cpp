@@ -652,6 +652,15 @@ PYBIND11_MODULE(_core, m) {
           },
           py::arg("n"), py::arg("n_keys"), py::arg("win") = 1000,
           py::arg("slide") = 100, py::arg("iters") = 20, py::arg("vik") = 0);
+    m.def("debug_tb_stage_times",
+          [](int64_t n, int64_t n_keys, int iters, int mono) {
+              std::map<std::string, double> m2;
+              for (auto& kv : debug_tb_stage_times(n, n_keys, iters, mono))
+                  m2[kv.first] = kv.second;
+              return m2;
+          },
+          py::arg("n"), py::arg("n_keys"), py::arg("iters") = 20,
+          py::arg("mono") = 1);
     m.def("debug_gram_stage_times",
           [](int64_t n, int64_t n_keys, int64_t win, int iters) {
               std::map<std::string, double> m2;

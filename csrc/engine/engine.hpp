@@ -508,6 +508,8 @@ std::vector<std::pair<std::string, double>> debug_ffat_stage_times(
     int64_t n, int64_t n_keys, int64_t win, int64_t slide, int iters, int vik);
 std::vector<std::pair<std::string, double>> debug_gram_stage_times(
     int64_t n, int64_t n_keys, int64_t win, int iters);
+std::vector<std::pair<std::string, double>> debug_tb_stage_times(
+    int64_t n, int64_t n_keys, int iters, int mono);
 std::vector<std::pair<std::string, double>> debug_a2a_stage_times(
     int64_t n, int64_t n_keys, int iters);
 

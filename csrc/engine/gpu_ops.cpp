@@ -1969,6 +1969,88 @@ std::vector<std::pair<std::string, double>> debug_ffat_stage_times(
     return out;
 }
 
+// TB (event-time) window chain harness: group / lift / count+scan /
+// advance per-stage hipEvent times.  Window extents scale with key count
+// (per-key ts density 1/n_keys) like tools/tb_bench.py.
+std::vector<std::pair<std::string, double>> debug_tb_stage_times(
+    int64_t n, int64_t n_keys, int iters, int mono) {
+    HIPCHK(hipSetDevice(0));
+    hipStream_t s = nullptr;
+    KeyedScratch ks;
+    ks.alloc(0, n, n_keys, s);
+    auto& A = arena(0);
+    uint64_t* d_key = (uint64_t*)A.get(8 * n);
+    int64_t* d_ts = (int64_t*)A.get(8 * n);
+    uint16_t* d_val = (uint16_t*)A.get(2 * n);
+    wfa_gen_batch(s, d_ts, d_key, d_val, 5, n, 0, 42, n_keys);
+    const int64_t pane = 100 * n_keys, P = 10, S = 1;
+    int ring_log2 = 5, pend_log2 = 18;
+    int64_t R = 1ll << ring_log2;
+    int64_t fires_cap = n / pane + n_keys + 64;
+    float* pend = (float*)A.get(4 * n_keys * (1ll << pend_log2));
+    int64_t* pend_base = (int64_t*)A.get(8 * n_keys);
+    int64_t* last_pane = (int64_t*)A.get(8 * n_keys);
+    uint32_t* st_head = (uint32_t*)A.get(4 * n_keys);
+    float* st_wsum = (float*)A.get(4 * n_keys);
+    float* ring = (float*)A.get(4 * n_keys * R);
+    uint32_t* nf = (uint32_t*)A.get(4 * (n_keys + 1));
+    uint32_t* flags = (uint32_t*)A.get(64);
+    uint64_t* o_key = (uint64_t*)A.get(8 * fires_cap);
+    float* o_val = (float*)A.get(4 * fires_cap);
+    int64_t* o_ts = (int64_t*)A.get(8 * fires_cap);
+    int64_t* d_on = (int64_t*)A.get(64);
+    wfa_fill_f32(s, pend, 0.f, n_keys * (1ll << pend_log2));
+    wfa_fill_f32(s, ring, 0.f, n_keys * R);
+    wfa_fill_f32(s, st_wsum, 0.f, n_keys);
+    wfa_fill_u64(s, (uint64_t*)pend_base, (uint64_t)-1ll, n_keys);
+    wfa_fill_u64(s, (uint64_t*)last_pane, (uint64_t)-1ll, n_keys);
+    HIPCHK(hipMemsetAsync(st_head, 0, 4 * n_keys, s));
+    HIPCHK(hipMemsetAsync(flags, 0, 64, s));
+
+    constexpr int NS = 4;
+    const char* names[NS] = {"group", "lift", "count_scan", "advance"};
+    hipEvent_t ev[NS + 1];
+    for (auto& e : ev) HIPCHK(hipEventCreate(&e));
+    double acc[NS] = {0};
+    RuntimeCtx rctx;
+    for (int it = -2; it < iters; ++it) {
+        wfa_gen_batch(s, d_ts, d_key, d_val, 5, n, (int64_t)(it + 2) * n, 42,
+                      n_keys);
+        HIPCHK(hipEventRecord(ev[0], s));
+        wfa_key_to_slot(s, d_key, n, ks.tab, ks.d_nslots, ks.table_cap,
+                        ks.slot, ks.slot_to_key);
+        uint32_t *os_, *oi;
+        wfa_sort_pairs2(s, ks.slot, ks.idx, ks.slot_t, ks.idx_t, nullptr,
+                        nullptr, ks.hist, n, ks.bits, &os_, &oi, nullptr, 1, 0);
+        ks.segs(s, os_, n, 0);
+        HIPCHK(hipEventRecord(ev[1], s));
+        wfa_tb_lift_only(s, ks.seg_start, ks.seg_slot, ks.d_nseg, n, d_val, 5,
+                         oi, d_ts, pane, P, S, 0, pend_log2, pend, pend_base,
+                         last_pane, flags, flags + 1, mono);
+        HIPCHK(hipEventRecord(ev[2], s));
+        int64_t limit = ((int64_t)(it + 3) * n) / pane - 1;
+        wfa_tb_countscan_only(s, ks.d_nslots, limit, pend_base, last_pane,
+                              st_head, P, S, nf, d_on);
+        HIPCHK(hipEventRecord(ev[3], s));
+        wfa_tb_advance_only(s, ks.d_nslots, limit, pane, P, S, 0, ring_log2,
+                            pend_log2, pend, pend_base, last_pane, st_head,
+                            st_wsum, ring, ks.slot_to_key, nf, o_key, o_val,
+                            o_ts, fires_cap);
+        HIPCHK(hipEventRecord(ev[4], s));
+        HIPCHK(hipStreamSynchronize(s));
+        if (it >= 0)
+            for (int k = 0; k < NS; ++k) {
+                float ms = 0;
+                HIPCHK(hipEventElapsedTime(&ms, ev[k], ev[k + 1]));
+                acc[k] += ms * 1000.0;
+            }
+    }
+    for (auto& e : ev) (void)hipEventDestroy(e);
+    std::vector<std::pair<std::string, double>> out;
+    for (int k = 0; k < NS; ++k) out.push_back({names[k], acc[k] / iters});
+    return out;
+}
+
 // MFMA Gram-window chain harness: per-stage hipEvent times for the
 // matrix-core combine (k_gram).  Also reports derived MFMA throughput:
 // each tuple contributes one 16-vector to C += v*v^T (512 FLOP/tuple,
@@ -2084,6 +2166,11 @@ std::vector<std::pair<std::string, double>> debug_gram_stage_times(int64_t,
                                                                    int64_t,
                                                                    int64_t,
                                                                    int) {
+    throw std::runtime_error("built without HIP");
+}
+std::vector<std::pair<std::string, double>> debug_tb_stage_times(int64_t,
+                                                                 int64_t, int,
+                                                                 int) {
     throw std::runtime_error("built without HIP");
 }
 std::shared_ptr<OpLogic> make_gpu_logic(const std::string&, const std::string&,

@@ -2286,6 +2286,57 @@ __global__ void k_tb_lift_wave(const uint32_t* seg_start, const uint32_t* seg_sl
 #undef TLCOMB
 }
 
+// stage-split entries (hipEvent harness; production uses wfa_ffat_tb_round)
+extern "C" void wfa_tb_lift_only(
+    wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
+    const int64_t* d_nseg, int64_t n, const void* v_f32, int vdt,
+    const uint32_t* idx_sorted, const int64_t* ts_orig, int64_t pane_len,
+    int64_t P, int64_t S, int comb, int pend_log2, float* pend,
+    int64_t* pend_base, int64_t* last_pane, uint32_t* ignored,
+    uint32_t* overflow, int ts_mono) {
+    hipStream_t st = (hipStream_t)s;
+    if (ts_mono)
+        hipLaunchKernelGGL(k_tb_lift_wave, dim3(WFA_MAX_BLOCKS / 2),
+                           dim3(WFA_THREADS), 0, st, seg_start, seg_slot,
+                           d_nseg, n, v_f32, vdt, idx_sorted, ts_orig, pane_len,
+                           P, S, comb, pend_log2, pend, pend_base, last_pane,
+                           ignored, overflow);
+    else
+        hipLaunchKernelGGL(k_tb_lift, dim3(WFA_MAX_BLOCKS / 8),
+                           dim3(WFA_THREADS), 0, st, seg_start, seg_slot,
+                           d_nseg, n, v_f32, vdt, idx_sorted, ts_orig, pane_len,
+                           P, S, comb, pend_log2, pend, pend_base, last_pane,
+                           ignored, overflow);
+}
+
+extern "C" void wfa_tb_countscan_only(wfa_stream_t s, const uint32_t* n_slots,
+                                      int64_t limit_pane, int64_t* pend_base,
+                                      int64_t* last_pane, uint32_t* st_head,
+                                      int64_t P, int64_t S, uint32_t* nf,
+                                      int64_t* d_out_n) {
+    hipStream_t st = (hipStream_t)s;
+    hipLaunchKernelGGL(k_tb_count, dim3(WFA_MAX_BLOCKS / 8), dim3(WFA_THREADS),
+                       0, st, n_slots, limit_pane, pend_base, last_pane,
+                       st_head, P, S, nf);
+    hipLaunchKernelGGL(k_tb_scan, dim3(1), dim3(1024), 0, st, nf, n_slots,
+                       d_out_n);
+}
+
+extern "C" void wfa_tb_advance_only(
+    wfa_stream_t s, const uint32_t* n_slots, int64_t limit_pane,
+    int64_t pane_len, int64_t P, int64_t S, int comb, int ring_log2,
+    int pend_log2, float* pend, int64_t* pend_base, int64_t* last_pane,
+    uint32_t* st_head, float* st_wsum, float* ring,
+    const uint64_t* slot_to_key, uint32_t* nf, uint64_t* out_key,
+    float* out_val, int64_t* out_ts, int64_t out_cap) {
+    hipStream_t st = (hipStream_t)s;
+    hipLaunchKernelGGL(k_tb_advance, dim3(WFA_MAX_BLOCKS / 8),
+                       dim3(WFA_THREADS), 0, st, n_slots, limit_pane, pane_len,
+                       P, S, comb, ring_log2, pend_log2, pend, pend_base,
+                       last_pane, st_head, st_wsum, ring, slot_to_key, nf,
+                       out_key, out_val, out_ts, out_cap);
+}
+
 extern "C" void wfa_ffat_tb_round(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
     const int64_t* d_nseg, int64_t n, const void* v_f32, int vdt,

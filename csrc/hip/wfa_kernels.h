@@ -230,6 +230,27 @@ void wfa_stateful_apply(wfa_stream_t s, const uint32_t* seg_start,
 // (pane id = ts / pane_len, absolute; pend ring of 2^pend_log2 panes), then
 // complete every slot's panes up to limit_pane and fire windows.  Call with
 // n == 0 (and null segment args) for a pure watermark/EOS advance.
+// stage-split TB entries for the hipEvent harness
+void wfa_tb_lift_only(wfa_stream_t s, const uint32_t* seg_start,
+                      const uint32_t* seg_slot, const int64_t* d_nseg,
+                      int64_t n, const void* v_f32, int vdt,
+                      const uint32_t* idx_sorted, const int64_t* ts_orig,
+                      int64_t pane_len, int64_t P, int64_t S, int comb,
+                      int pend_log2, float* pend, int64_t* pend_base,
+                      int64_t* last_pane, uint32_t* ignored,
+                      uint32_t* overflow, int ts_mono);
+void wfa_tb_countscan_only(wfa_stream_t s, const uint32_t* n_slots,
+                           int64_t limit_pane, int64_t* pend_base,
+                           int64_t* last_pane, uint32_t* st_head, int64_t P,
+                           int64_t S, uint32_t* nf, int64_t* d_out_n);
+void wfa_tb_advance_only(wfa_stream_t s, const uint32_t* n_slots,
+                         int64_t limit_pane, int64_t pane_len, int64_t P,
+                         int64_t S, int comb, int ring_log2, int pend_log2,
+                         float* pend, int64_t* pend_base, int64_t* last_pane,
+                         uint32_t* st_head, float* st_wsum, float* ring,
+                         const uint64_t* slot_to_key, uint32_t* nf,
+                         uint64_t* out_key, float* out_val, int64_t* out_ts,
+                         int64_t out_cap);
 void wfa_ffat_tb_round(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
     const int64_t* d_nseg, int64_t n, const void* v_f32, int vdt,
