@@ -28,6 +28,10 @@ CASES = [
     ("avg",   "tb",   600, 300, 64,     400_000, 100_000, 2),
     ("jmin",  "cb",   300, 60,  101,    400_000, 100_000, 2),
     ("avg",   "cb",    24, 8,   64,     400_000, 100_000, 2),  # thread kernel
+    # round-2 dense-key variants (slot = key; same oracles)
+    ("sum",   "cb",   500, 100, 101,  1_000_000, 250_000, 5, True),
+    ("avg",   "tb",   600, 300, 64,     400_000, 100_000, 2, True),
+    ("count", "tb",   777, 111, 128,  1_000_000, 250_000, 5, True),
 ]
 
 
@@ -38,24 +42,27 @@ def engine():
                                            Ffat_Windows_GPU_Builder)
     COMBS = {"sum": native_gpu.COMB_SUM, "min": native_gpu.COMB_MIN,
              "max": native_gpu.COMB_MAX, "count": native_gpu.COMB_COUNT}
-    for ci, (comb, form, win, slide, n_keys, n, b, vdt) in enumerate(CASES):
+    for ci, case in enumerate(CASES):
+        comb, form, win, slide, n_keys, n, b, vdt = case[:8]
+        dense = bool(case[8]) if len(case) > 8 else False
         src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=vdt))
                .withOutputSchema([vdt]).withOutputBatchSize(b).build())
         if comb == "avg":
             logic = native_gpu.gpu_avg_ffat_windows(
                 win, slide, col=0, max_keys=2 * n_keys, tb=form == "tb",
-                pend_ring_log2=13)
+                pend_ring_log2=13, dense_keys=dense)
         elif comb == "jmin":
             logic = native_gpu.gpu_jit_ffat_windows(
                 win, slide, lift="v0", comb="fminf(a0, b0)", finalize="f0",
                 identity=(float("inf"),), max_keys=2 * n_keys,
-                tb=form == "tb", pend_ring_log2=13)
+                tb=form == "tb", pend_ring_log2=13, dense_keys=dense)
         else:
             logic = native_gpu.gpu_ffat_windows(COMBS[comb], 0, win, slide,
                                                 max_keys=2 * n_keys,
                                                 use_tree=form == "tree",
                                                 tb=form == "tb",
-                                                pend_ring_log2=13)
+                                                pend_ring_log2=13,
+                                                dense_keys=dense)
         ff = (Ffat_Windows_GPU_Builder(logic)
             .withOutputSchema([2]).withOutputBatchSize(4 * b).build())
         keys, vals = [], []
@@ -80,7 +87,8 @@ def engine():
 def check():
     from windflow_amd.synth import gen_batch
     fails = 0
-    for ci, (comb, form, win, slide, n_keys, n, b, vdt) in enumerate(CASES):
+    for ci, case in enumerate(CASES):
+        comb, form, win, slide, n_keys, n, b, vdt = case[:8]
         d = np.load(f"gpurun_out/wfz_{ci}.npz")
         got = defaultdict(list)
         for k, v in zip(d['key'].tolist(), d['val'].tolist()):
