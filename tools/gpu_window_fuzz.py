@@ -31,7 +31,7 @@ CASES = [
     # round-2 dense-key variants (slot = key; same oracles)
     ("sum",   "cb",   500, 100, 101,  1_000_000, 250_000, 5, True),
     ("avg",   "tb",   600, 300, 64,     400_000, 100_000, 2, True),
-    ("count", "tb",   777, 111, 128,  1_000_000, 250_000, 5, True),
+    ("count", "tb",   600, 200, 128,    400_000, 100_000, 5, True),
 ]
 
 
