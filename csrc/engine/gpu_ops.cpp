@@ -805,6 +805,10 @@ struct GpuFfatLogic : GpuLogicBase {
     int64_t* tb_last = nullptr;
     uint32_t* tb_flags = nullptr;  // [ignored, overflow]
     uint32_t* tb_nf = nullptr;
+    // pane-parallel lift scratch (allocated when max_keys <= 4096: span
+    // scan + first-pane per segment; null disables the path)
+    uint32_t* tb_span_nf = nullptr;
+    int64_t* tb_pfirst = nullptr;
     uint32_t* h_flags = nullptr;   // pinned readback
     Engine* eng_ = nullptr;        // central dropped-tuple accounting
     int64_t batches = 0;
@@ -881,6 +885,10 @@ struct GpuFfatLogic : GpuLogicBase {
             tb_last = (int64_t*)A.get(8 * max_keys);
             tb_flags = (uint32_t*)A.get(64);
             tb_nf = (uint32_t*)A.get(4 * (max_keys + 1));
+            if (max_keys <= 4096) {  // pane-parallel lift (few segments)
+                tb_span_nf = (uint32_t*)A.get(4 * (max_keys + 1));
+                tb_pfirst = (int64_t*)A.get(8 * (max_keys + 1));
+            }
             wfa_fill_f32(stream, tb_pend, ident, max_keys * Rp);
             wfa_fill_u64(stream, (uint64_t*)tb_base, (uint64_t)-1ll, max_keys);
             wfa_fill_u64(stream, (uint64_t*)tb_last, (uint64_t)-1ll, max_keys);
@@ -1044,7 +1052,8 @@ struct GpuFfatLogic : GpuLogicBase {
                           ring_or_tree, ks.d_nslots, ks.slot_to_key, tb_nf,
                           tb_flags, tb_flags + 1, ob->key, (float*)ob->cols[0],
                           ob->ts, ob->capacity, d_on,
-                          db && db->ts_mono ? 1 : 0);
+                          db && db->ts_mono ? 1 : 0, tb_span_nf, tb_pfirst,
+                          d_on + 1);
         HIPCHK(hipMemcpyAsync(ob->lazy_count, d_on, 8, hipMemcpyDeviceToHost, stream));
         ob->count = -1;
         ob->watermark = wm - lateness;

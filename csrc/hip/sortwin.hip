@@ -2286,6 +2286,112 @@ __global__ void k_tb_lift_wave(const uint32_t* seg_start, const uint32_t* seg_sl
 #undef TLCOMB
 }
 
+// Pane-parallel TB lift for LOW segment counts (< ~4096): wave-per-segment
+// starves the chip, so assign one wave per (segment, pane-in-span).
+// k_tb_span: per segment, pane span + first pane from the first/last ts
+// (monotonic).  k_tb_lift_pw: wave g -> binary search owning segment in
+// the span scan, then binary search its pane's row range and wave-reduce.
+// Panes with no rows write nothing.  Base/lastp/ignored bookkeeping runs
+// in the span kernel (one thread per segment, reads only boundary ts).
+__global__ void k_tb_span(const uint32_t* seg_start, const uint32_t* seg_slot,
+                          const int64_t* d_nseg, int64_t n,
+                          const uint32_t* idx_sorted, const int64_t* ts_orig,
+                          int64_t pane_len, int64_t P, int64_t S,
+                          int64_t* pend_base, int64_t* last_pane,
+                          uint32_t* span_nf, int64_t* pfirst) {
+    const int64_t nseg = *d_nseg;
+    for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nseg;
+         j += gridDim.x * (int64_t)blockDim.x) {
+        const uint32_t slot = seg_slot[j];
+        const int64_t b = seg_start[j];
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        const int64_t t0 = ts_orig[idx_sorted[b]];
+        const int64_t t1 = ts_orig[idx_sorted[e - 1]];
+        int64_t base = pend_base[slot];
+        if (base < 0) {
+            int64_t w0 = t0 - P * pane_len + 1;
+            w0 = w0 <= 0 ? 0 : (w0 + S * pane_len - 1) / (S * pane_len);
+            base = w0 * S;
+            pend_base[slot] = base;
+        }
+        const int64_t p0 = t0 / pane_len, p1 = t1 / pane_len;
+        pfirst[j] = p0;
+        span_nf[j] = (uint32_t)(p1 - p0 + 1);
+        if (p1 > last_pane[slot]) last_pane[slot] = p1;
+    }
+}
+
+__global__ void k_tb_lift_pw(const uint32_t* seg_start, const uint32_t* seg_slot,
+                             const int64_t* d_nseg, const int64_t* d_total,
+                             int64_t n, const void* v_f32, int vdt,
+                             const uint32_t* idx_sorted, const int64_t* ts_orig,
+                             int64_t pane_len, int comb, int pend_log2,
+                             float* pend, const int64_t* pend_base,
+                             const uint32_t* span_base, const int64_t* pfirst,
+                             uint32_t* ignored, uint32_t* overflow) {
+    const int64_t nseg = *d_nseg;
+    const int64_t total = *d_total;
+    const uint32_t Rp = 1u << pend_log2;
+    const uint32_t Pm = Rp - 1;
+    const float ident = (comb == 1) ? INFINITY : (comb == 2 ? -INFINITY : 0.f);
+    const int lane = threadIdx.x & 63;
+    const int64_t wid = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
+#define TPCOMB(a, b) ((comb == 1) ? fminf(a, b) : (comb == 2 ? fmaxf(a, b) : (a) + (b)))
+    for (int64_t g = wid; g < total; g += nw) {
+        int64_t lo = 0, hi = nseg - 1;
+        while (lo < hi) {
+            int64_t mid = (lo + hi + 1) >> 1;
+            if ((int64_t)span_base[mid] <= g) lo = mid;
+            else hi = mid - 1;
+        }
+        const int64_t j = lo;
+        const int64_t p = pfirst[j] + (g - (int64_t)span_base[j]);
+        const uint32_t slot = seg_slot[j];
+        const int64_t b = seg_start[j];
+        const int64_t e = (j + 1 < nseg) ? seg_start[j + 1] : n;
+        // row range of pane p: [lower_bound(p*L), lower_bound((p+1)*L))
+        int64_t s0 = b, s1 = e;
+        const int64_t tlo = p * pane_len;
+        while (s0 < s1) {
+            int64_t mid = (s0 + s1) >> 1;
+            if (ts_orig[idx_sorted[mid]] < tlo) s0 = mid + 1;
+            else s1 = mid;
+        }
+        int64_t e0 = s0, e1 = e;
+        const int64_t thi = (p + 1) * pane_len;
+        while (e0 < e1) {
+            int64_t mid = (e0 + e1) >> 1;
+            if (ts_orig[idx_sorted[mid]] < thi) e0 = mid + 1;
+            else e1 = mid;
+        }
+        const int64_t pe = e0;
+        if (pe == s0) continue;   // empty pane: no write
+        const int64_t base = pend_base[slot];
+        if (p < base) {
+            if (lane == 0) atomicAdd(ignored, (uint32_t)(pe - s0));
+            continue;
+        }
+        if (p - base >= (int64_t)Rp) {
+            if (lane == 0) atomicAdd(overflow, 1u);
+            continue;
+        }
+        float part = ident;
+        for (int64_t q = s0 + lane; q < pe; q += 64) {
+            float x = (comb == 3)
+                          ? 1.0f
+                          : wfa_val_at(v_f32, vdt, vdt == 6 ? q : idx_sorted[q]);
+            part = TPCOMB(part, x);
+        }
+        for (int o = 32; o; o >>= 1) part = TPCOMB(part, __shfl_xor(part, o, 64));
+        if (lane == 0) {
+            float* cell = &pend[(size_t)slot * Rp + ((uint64_t)p & Pm)];
+            *cell = TPCOMB(*cell, part);
+        }
+    }
+#undef TPCOMB
+}
+
 // stage-split entries (hipEvent harness; production uses wfa_ffat_tb_round)
 extern "C" void wfa_tb_lift_only(
     wfa_stream_t s, const uint32_t* seg_start, const uint32_t* seg_slot,
@@ -2346,10 +2452,25 @@ extern "C" void wfa_ffat_tb_round(
     uint32_t* st_head, float* st_wsum, float* ring, const uint32_t* n_slots,
     const uint64_t* slot_to_key, uint32_t* nf, uint32_t* ignored,
     uint32_t* overflow, uint64_t* out_key, float* out_val, int64_t* out_ts,
-    int64_t out_cap, int64_t* d_out_n, int ts_mono) {
+    int64_t out_cap, int64_t* d_out_n, int ts_mono, uint32_t* span_nf,
+    int64_t* pfirst, int64_t* d_total_spans) {
     hipStream_t st = (hipStream_t)s;
     if (n > 0) {
-        if (ts_mono)
+        if (ts_mono && span_nf) {
+            // low-segment-count shape: pane-parallel lift (wave per pane)
+            hipLaunchKernelGGL(k_tb_span, dim3(WFA_MAX_BLOCKS / 8),
+                               dim3(WFA_THREADS), 0, st, seg_start, seg_slot,
+                               d_nseg, n, idx_sorted, ts_orig, pane_len, P, S,
+                               pend_base, last_pane, span_nf, pfirst);
+            hipLaunchKernelGGL(k_fire_scan, dim3(1), dim3(1024), 0, st,
+                               span_nf, d_nseg, d_total_spans);
+            hipLaunchKernelGGL(k_tb_lift_pw, dim3(WFA_MAX_BLOCKS),
+                               dim3(WFA_THREADS), 0, st, seg_start, seg_slot,
+                               d_nseg, d_total_spans, n, v_f32, vdt,
+                               idx_sorted, ts_orig, pane_len, comb, pend_log2,
+                               pend, pend_base, span_nf, pfirst, ignored,
+                               overflow);
+        } else if (ts_mono)
             hipLaunchKernelGGL(k_tb_lift_wave, dim3(WFA_MAX_BLOCKS / 2),
                                dim3(WFA_THREADS), 0, st, seg_start, seg_slot,
                                d_nseg, n, v_f32, vdt, idx_sorted, ts_orig,

@@ -260,7 +260,8 @@ void wfa_ffat_tb_round(
     uint32_t* st_head, float* st_wsum, float* ring, const uint32_t* n_slots,
     const uint64_t* slot_to_key, uint32_t* nf, uint32_t* ignored,
     uint32_t* overflow, uint64_t* out_key, float* out_val, int64_t* out_ts,
-    int64_t out_cap, int64_t* d_out_n, int ts_mono);
+    int64_t out_cap, int64_t* d_out_n, int ts_mono, uint32_t* span_nf,
+                       int64_t* pfirst, int64_t* d_total_spans);
 
 // ----- FlatFAT arena path (non-invertible combines over many panes) -----
 // Per-slot complete binary tree over ring of 2^ring_log2 pane leaves,
