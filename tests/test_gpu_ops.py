@@ -828,15 +828,17 @@ def test_gpu_jit_ffat_minmax_multicol():
             assert abs(a0 - b0) <= 1e-5 and abs(a1 - b1) <= 1e-5, (k,)
 
 
-def test_gpu_jit_ffat_tb_avg_vs_oracle():
+@pytest.mark.parametrize("acc", ["f32", "f64"])
+def test_gpu_jit_ffat_tb_avg_vs_oracle(acc):
     """Event-time JIT windows: AVG over TB panes + watermark advance +
-    EOS flush, vs brute-force oracle."""
+    EOS flush, vs brute-force oracle — in both accumulator precisions
+    (f64 exercises the double pend arena through the TB wave lift)."""
     n, n_keys, win, slide, b = 120_000, 101, 400, 100, 17_000
     src = (Source_GPU_Builder(native_gpu.gpu_source(n, n_keys, b, vdt=2))
            .withOutputSchema([2]).withOutputBatchSize(b).build())
     ff = (Ffat_Windows_GPU_Builder(
         native_gpu.gpu_avg_ffat_windows(win, slide, col=0, max_keys=1024,
-                                        tb=True, pend_ring_log2=10))
+                                        tb=True, pend_ring_log2=10, acc=acc))
           .withOutputSchema([2]).withOutputBatchSize(2 * b).build())
     res = dict(rows=[])
 
