@@ -1565,6 +1565,17 @@ struct GpuExchangeLogic : GpuLogicBase {
         int64_t n = db->count;
         int64_t wm = db->watermark;
         if (n > out_cap) throw std::runtime_error("exchange input > out_batch");
+        // world=1: the rendezvous is the identity (counts/watermark/done
+        // are all local) — forward without the per-batch meta D2H + host
+        // sync.  Besides the cycles, this makes the world=1 exchange
+        // immune to the pool-side host<->device stalls that intermittently
+        // inflated a2a p99 to ~200 ms (BASELINE.md straggler note).
+        // WFA_XCHG_NO_SELFPASS keeps the full machinery testable.
+        if (world == 1 && !getenv("WFA_XCHG_NO_SELFPASS")) {
+            cur_wm = std::max(cur_wm, wm);
+            out.emit(db);
+            return;
+        }
         Batch* sb = send_pool->get();
         int64_t t1 = now_us();
         if (sb->ready_event)
@@ -1644,6 +1655,7 @@ struct GpuExchangeLogic : GpuLogicBase {
 
     void on_eos(EmitCtx& out, RuntimeCtx& ctx) override {
         ensure_init();
+        if (world == 1 && !getenv("WFA_XCHG_NO_SELFPASS")) return;
         // EOS rounds: keep matching other ranks' collectives until all done
         for (;;) {
             start_meta(false, cur_wm, true);
